@@ -1,0 +1,132 @@
+"""Golden query suite over the committed fixtures (tests/golden/data).
+
+Query IR is the plan-level surface of SURVEY.md §8b: conjunctive predicates
+(col op literal | BETWEEN | LIKE-'%x%'), optional GROUP BY keys, aggregate
+list, plus the injected time range [start, end) in ms
+(src/query/mod.rs:829-888). Edge cases mirror what the reference's own tests
+exercise at planning level (stream_schema_provider.rs:1139-1629) plus
+value-level cases the reference leaves to its engine: empty results,
+all-match, NULL-heavy columns (c4), dict→PLAIN fallback pages (c1 latency,
+c3 message), multi-file merges, single-row files.
+"""
+
+# time base must match datagen.gen.BASE_TS_MS
+BASE = 1756684800000  # 2025-09-01T00:00:00Z in ms
+MIN = 60_000
+
+# fixture name -> list of (query_name, query_dict)
+GOLDEN_QUERIES = {
+    "g_c1": [
+        ("count_by_level", {
+            "select": [{"agg": "count_star"}],
+            "group_by": ["level"],
+        }),
+        ("count_max_by_host_between", {
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["host"],
+            "preds": [{"col": "p_timestamp", "op": "between",
+                       "lo": BASE + MIN // 2, "hi": BASE + 2 * MIN}],
+        }),
+        ("sum_min_where_host_eq", {
+            "select": [{"agg": "sum", "col": "latency"},
+                       {"agg": "min", "col": "latency"},
+                       {"agg": "count_star"}],
+            "preds": [{"col": "host", "op": "eq", "lit": "host-0001"}],
+        }),
+        ("empty_result", {
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+            "preds": [{"col": "host", "op": "eq", "lit": "no-such-host"}],
+        }),
+        ("all_match", {
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+            "preds": [{"col": "latency", "op": "ge", "lit": 0}],
+        }),
+        ("time_range_injected", {
+            "select": [{"agg": "count_star"}],
+            "group_by": ["level"],
+            "time_range": [BASE + MIN, BASE + 2 * MIN],
+        }),
+        ("lt_on_i64", {
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "f_i64"},
+                       {"agg": "min", "col": "f_i64"}],
+            "preds": [{"col": "latency", "op": "lt", "lit": 1000}],
+        }),
+        ("ne_and_two_keys", {
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+            "group_by": ["level", "f_str2"],
+            "preds": [{"col": "f_str1", "op": "ne", "lit": "region-0"}],
+        }),
+        ("sum_f64", {
+            "select": [{"agg": "sum", "col": "f_f64"}, {"agg": "count_star"}],
+            "group_by": ["level"],
+            "preds": [{"col": "level", "op": "ge", "lit": "INFO"}],
+        }),
+    ],
+    "g_c1_pages": [
+        ("count_by_level", {
+            "select": [{"agg": "count_star"}],
+            "group_by": ["level"],
+        }),
+        ("between_half", {
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["host"],
+            "preds": [{"col": "p_timestamp", "op": "between",
+                       "lo": BASE, "hi": BASE + MIN // 2}],
+        }),
+    ],
+    "g_c3": [
+        ("like_error", {
+            "select": [{"agg": "count_star"}],
+            "preds": [{"col": "message", "op": "contains", "lit": "error"}],
+        }),
+        ("like_error_by_level", {
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+            "group_by": ["level"],
+            "preds": [{"col": "message", "op": "contains", "lit": "error"}],
+        }),
+        ("like_no_match", {
+            "select": [{"agg": "count_star"}],
+            "preds": [{"col": "message", "op": "contains", "lit": "zzzzzzzzzzzz"}],
+        }),
+    ],
+    "g_c4": [
+        ("three_key_groupby", {
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["service", "span_kind", "status"],
+        }),
+        ("count_nullable_col", {
+            "select": [{"agg": "count_star"}, {"agg": "count", "col": "attr_s0"},
+                       {"agg": "count", "col": "attr_i5"}],
+        }),
+        ("group_by_nullable", {
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "attr_i0"}],
+            "group_by": ["attr_s1"],
+        }),
+        ("pred_on_nullable", {
+            "select": [{"agg": "count_star"}],
+            "preds": [{"col": "attr_i1", "op": "ge", "lit": 500000}],
+            "group_by": ["status"],
+        }),
+    ],
+    "g_edge": [
+        ("tiny_files_count", {
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"},
+                       {"agg": "min", "col": "latency"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["level"],
+        }),
+        ("tiny_between_none", {
+            "select": [{"agg": "count_star"}],
+            "time_range": [0, 1],
+        }),
+    ],
+}
+
+# fixture name -> datagen parameters
+GOLDEN_FIXTURES = {
+    "g_c1":       dict(config="c1", rows=120_000, rows_per_file=40_000, seed=1001),
+    "g_c1_pages": dict(config="c1", rows=40_000, rows_per_file=40_000, seed=1002,
+                       data_page_size=8 * 1024),
+    "g_c3":       dict(config="c3", rows=25_000, rows_per_file=25_000, seed=1003),
+    "g_c4":       dict(config="c4", rows=20_000, rows_per_file=20_000, seed=1004),
+    "g_edge":     dict(config="c1", rows=9, rows_per_file=4, seed=1005),
+}
