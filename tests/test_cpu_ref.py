@@ -1,0 +1,29 @@
+"""Pin the scalar C full-path restatement (oracle/cpu_ref.c: thrift footer ->
+LZ4_RAW -> PLAIN/RLE_DICT/DELTA_BP decode -> filter -> group-by) against the
+committed golden vectors. This is the independent check that our
+understanding of the parquet dialect — the same understanding the HIP
+decoders implement — is correct, without pyarrow in the loop."""
+
+import pytest
+
+from oracle import cpu_ref_runner
+from oracle.compare import assert_rows_equal
+from tests.golden_queries import GOLDEN_QUERIES
+
+
+def _all_cases():
+    return [f"{fx}/{q}" for fx, qs in GOLDEN_QUERIES.items() for q, _ in qs]
+
+
+@pytest.fixture(scope="session", autouse=True)
+def built():
+    cpu_ref_runner.build()
+
+
+@pytest.mark.parametrize("case", _all_cases())
+def test_cpu_ref_matches_golden(golden, case):
+    fx = case.split("/")[0]
+    entry = golden["answers"][case]
+    files = golden["fixtures"][fx]["files"]
+    r = cpu_ref_runner.execute(files, entry["query"])
+    assert_rows_equal(r["rows"], entry["result"]["rows"], case)
