@@ -268,14 +268,15 @@ def gen_stream(
         by_day.setdefault(day, []).append(f)
     manifest_list = []
     for day, files in sorted(by_day.items()):
-        mpath = os.path.join(stream_dir, f"date={day}", "manifest.json")
+        rel_mpath = f"{stream}/date={day}/manifest.json"
+        mpath = os.path.join(root, rel_mpath)
         with open(mpath, "w") as fh:
             json.dump({"version": "v2", "files": files}, fh)
         d0 = datetime.strptime(day, "%Y-%m-%d").replace(tzinfo=timezone.utc)
         lo = int(d0.timestamp() * 1000)
         manifest_list.append(
             {
-                "manifest_path": mpath,
+                "manifest_path": rel_mpath,
                 "time_lower_bound": _ts_iso(lo),
                 "time_upper_bound": _ts_iso(lo + 86_400_000 - 1),
                 "events_ingested": sum(f["num_rows"] for f in files),

@@ -1,0 +1,53 @@
+// Shared host/device types for the gpuq execution kernels (gfx950).
+#pragma once
+#include <cstdint>
+
+namespace gpuq {
+
+// One parquet page staged in device memory.
+// raw bytes at d_raw + src_off; decompressed image at d_dec + dst_off.
+struct DevPage {
+  uint64_t src_off;
+  uint64_t dst_off;
+  uint32_t comp_size;
+  uint32_t uncomp_size;
+  uint32_t num_values;   // rows incl. nulls (data pages)
+  uint32_t row_start;    // partition-global row index of first row
+  uint32_t aux;          // per-kind index into an aux pool (remap/LUT/dict values)
+  uint8_t  optional;     // has def levels (max_def_level == 1)
+  uint8_t  raw_copy;     // stored uncompressed: memcpy instead of LZ4
+  uint8_t  encoding;     // ENC_* (meta.h)
+  uint8_t  phys;         // PT_* (meta.h)
+};
+
+// comparison kernel ops (matches gpuq_op order where applicable)
+enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };
+
+enum { AGGK_COUNT_STAR = 0, AGGK_COUNT, AGGK_SUM_I64, AGGK_SUM_F64,
+       AGGK_MIN_I64, AGGK_MAX_I64, AGGK_MIN_F64, AGGK_MAX_F64 };
+
+constexpr int MAX_KEYS = 4;
+constexpr int MAX_AGGS = 8;
+
+// Aggregate table layout: G groups x (1 + 2*n_aggs) u64 slots:
+//   slot 0: presence (selected-row count)
+//   per agg a: slot 1+2a = value (i64 / f64 bits), slot 2+2a = non-null count
+struct AggArgs {
+  const uint8_t* mask;          // selection mask (1 byte/row), may be null (=all)
+  int64_t n_rows;               // partition rows
+  int n_keys;
+  const int32_t* key_gid[MAX_KEYS];  // 0 = NULL group
+  int32_t key_size[MAX_KEYS];        // global dict size incl. null slot
+  int n_aggs;
+  int32_t agg_kind[MAX_AGGS];        // AGGK_*
+  const int64_t* agg_val[MAX_AGGS];  // i64 array or f64 bits (same width)
+  const uint8_t* agg_valid[MAX_AGGS];// may be null (= all valid)
+  uint64_t* table;
+  int32_t n_groups;
+};
+
+// error codes written to *d_error by kernels
+enum { ERR_NONE = 0, ERR_LZ4 = 1, ERR_RLE = 2, ERR_DELTA = 3, ERR_DICT_RANGE = 4,
+       ERR_PAGE = 5 };
+
+}  // namespace gpuq

@@ -1,0 +1,1126 @@
+// gpuq host runtime: the C-ABI behind include/gpuq.h.
+//
+// Replaces the reference's DataFusion physical-plan execution for
+// filter/aggregate scans: plan building mirrors
+// StandardTableProvider::scan (stream_schema_provider.rs:616-753) — row-group
+// pruning via footer min/max stats (the row-group analog of
+// can_be_pruned/satisfy_constraints, :1049-1137), byte-balanced shard
+// assignment (balanced_file_groups, :146-165) — and execution mirrors the
+// DataSourceExec -> FilterExec -> AggregateExec(Partial) pipeline
+// (SURVEY.md §3a step 7), with the per-row work on the GPU and the
+// Partial->Final merge left to the caller (one RCCL reduce across GPUs).
+#include "../../include/gpuq.h"
+#include "meta.h"
+#include "kernels_api.h"
+#include "dev_types.h"
+
+#include <hip/hip_runtime.h>
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <cstring>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+using namespace gpuq;
+
+#define HIP_TRY(call)                                                        \
+  do {                                                                       \
+    hipError_t _e = (call);                                                  \
+    if (_e != hipSuccess)                                                    \
+      throw std::runtime_error(std::string("HIP error: ") +                  \
+                               hipGetErrorString(_e) + " at " #call);        \
+  } while (0)
+
+// ------------------------------------------------------------------
+struct gpuq_ctx {
+  std::vector<int> devices;
+  std::mutex mu;
+  std::string last_error;
+  void set_error(const std::string& e) {
+    std::lock_guard<std::mutex> g(mu);
+    last_error = e;
+  }
+};
+
+namespace {
+
+struct MappedFile {
+  std::string path;
+  int fd = -1;
+  const uint8_t* data = nullptr;
+  size_t size = 0;
+  FileMeta meta;
+  ~MappedFile() {
+    if (data) munmap((void*)data, size);
+    if (fd >= 0) close(fd);
+  }
+};
+
+// plan-level column info
+struct ColPlan {
+  std::string name;
+  int phys = -1;
+  bool optional = false;
+  // required representations
+  bool need_gid = false;    // group key (utf8 dict) or COUNT(utf8 col)
+  bool need_val = false;    // i64/f64 row-aligned values + valid
+  bool need_gid_valid = false;  // validity bytes alongside gid (COUNT(utf8))
+  // predicate routing
+  std::vector<int> lut_preds;      // preds evaluated per-dict-entry (utf8)
+  std::vector<int> cmp_preds;      // preds on the decoded i64 array
+  std::vector<int> contains_preds; // CONTAINS on byte_array pages
+  // global dictionary (need_gid): gid 1.. ; 0 = NULL
+  std::vector<std::string> gdict;
+  std::unordered_map<std::string, int32_t> gmap;
+  int32_t gid_of(const std::string& s) {
+    auto it = gmap.find(s);
+    if (it != gmap.end()) return it->second;
+    gdict.push_back(s);
+    int32_t id = (int32_t)gdict.size();  // 1-based
+    gmap.emplace(gdict.back(), id);
+    return id;
+  }
+};
+
+struct PredPlan {
+  gpuq_pred p;
+  std::string col, str_lit;
+};
+
+struct AggPlan {
+  int op;             // gpuq_agg_op
+  std::string col;    // empty for count_star
+  int col_idx = -1;   // into plan.cols
+  int kind = 0;       // AGGK_*
+  bool is_f64 = false;
+};
+
+// one column chunk of one selected row group
+struct ChunkTask {
+  int file_idx, rg_idx, col_idx;
+  uint64_t raw_off = 0;            // into partition d_raw
+  const ColumnChunkMeta* cm = nullptr;
+  std::vector<PageInfo> pages;
+  // dict-derived per-chunk aux (host-built)
+  std::vector<int32_t> remap;      // local dict id -> gid
+  std::vector<int64_t> dictv;      // local dict values (i64 / f64 bits)
+  std::vector<uint8_t> lut;        // combined pred LUT (AND of lut_preds)
+  bool has_plain_data_pages = false;
+};
+
+struct RgRef {
+  int file_idx, rg_idx;
+  int64_t rows = 0, needed_bytes = 0, total_bytes = 0;
+  uint32_t row_start = 0;  // partition-global
+};
+
+// decode-task lists per (kind,col); ids index into the partition's page array
+enum TaskKind { TK_DICT_GID, TK_DICT_VAL, TK_PLAIN_VAL, TK_DELTA_VAL,
+                TK_DICT_MASK, TK_BYTES_CONTAINS, TK_N };
+
+struct Partition {
+  int device = -1;
+  std::vector<RgRef> rgs;
+  int64_t n_rows = 0;
+  std::vector<ChunkTask> chunks;
+  // host-built device images
+  std::vector<DevPage> pages;                 // data pages only
+  std::vector<int32_t> ids[TK_N];             // per task kind... per column!
+  std::map<std::pair<int,int>, std::vector<int32_t>> tasks;  // (kind,col)->page ids
+  std::vector<int32_t> remap_pool;
+  std::vector<int64_t> dictv_pool;
+  std::vector<uint8_t> lut_pool;
+  uint64_t raw_bytes = 0, dec_bytes = 0;
+  int64_t bytes_scanned = 0, rowgroup_bytes_total = 0;
+
+  // device state
+  bool loaded = false;
+  hipStream_t stream = nullptr;
+  uint8_t *d_raw = nullptr, *d_dec = nullptr;
+  DevPage* d_pages = nullptr;
+  int32_t* d_remap = nullptr;
+  int64_t* d_dictv = nullptr;
+  uint8_t* d_lut = nullptr;
+  uint8_t* d_mask = nullptr;
+  int32_t* d_err = nullptr;
+  uint64_t* d_table = nullptr;
+  int32_t* d_agg_kind = nullptr;
+  uint8_t* d_needle = nullptr;
+  std::map<std::pair<int,int>, int32_t*> d_ids;     // (kind,col) -> ids
+  std::map<int, int32_t*> d_gid;                    // col -> gid array
+  std::map<int, int64_t*> d_val;                    // col -> value array
+  std::map<int, uint8_t*> d_valid;                  // col -> valid array
+  int64_t load_ns = 0;
+};
+
+}  // namespace
+
+struct gpuq_plan {
+  gpuq_ctx* ctx = nullptr;
+  std::vector<std::unique_ptr<MappedFile>> files;
+  std::vector<ColPlan> cols;
+  std::vector<PredPlan> preds;
+  std::vector<AggPlan> aggs;
+  std::vector<int> group_cols;   // indices into cols
+  int64_t limit = -1;
+  std::vector<Partition> parts;
+  int32_t n_groups = 0;          // product of key sizes (incl null slots)
+  std::mutex mu;
+  // metrics
+  int64_t m_rows_scanned = 0, m_rows_out = 0, m_kernel_ns = 0, m_exec_ns = 0,
+          m_load_ns = 0, m_decomp_ns = 0, m_hbm_est = 0;
+  ~gpuq_plan();
+};
+
+// ------------------------------------------------------------------
+// C ABI: session
+// ------------------------------------------------------------------
+extern "C" gpuq_ctx* gpuq_session_create(uint64_t device_mask) {
+  auto* c = new gpuq_ctx();
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
+  for (int i = 0; i < 64 && i < n; i++)
+    if (device_mask & (1ull << i)) c->devices.push_back(i);
+  if (c->devices.empty() && device_mask == 0 && n > 0) c->devices.push_back(0);
+  return c;
+}
+extern "C" void gpuq_session_destroy(gpuq_ctx* c) { delete c; }
+extern "C" const char* gpuq_last_error(gpuq_ctx* c) {
+  return c ? c->last_error.c_str() : "null ctx";
+}
+extern "C" int32_t gpuq_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return -1;
+  return n;
+}
+
+// ------------------------------------------------------------------
+// plan building
+// ------------------------------------------------------------------
+namespace {
+
+int find_or_add_col(std::vector<ColPlan>& cols, const std::string& name) {
+  for (size_t i = 0; i < cols.size(); i++)
+    if (cols[i].name == name) return (int)i;
+  cols.push_back(ColPlan{});
+  cols.back().name = name;
+  return (int)cols.size() - 1;
+}
+
+// row-group-level min/max pruning for i64 comparisons — the row-group analog
+// of ManifestExt::can_be_pruned + satisfy_constraints
+// (stream_schema_provider.rs:1049-1137): prune when the predicate can match
+// NO value in [min,max].
+bool rg_pruned_by_stats(const gpuq_plan& plan, const FileMeta& fm,
+                        const RowGroupMeta& rg) {
+  for (const auto& pp : plan.preds) {
+    if (pp.p.lit_kind != GPUQ_LIT_I64) continue;
+    int ci = fm.col_index(pp.col);
+    if (ci < 0) continue;
+    const auto& cm = rg.chunks[ci];
+    if (!cm.has_i64_stats) continue;
+    int64_t mn = cm.stat_min, mx = cm.stat_max;
+    bool can_match = true;
+    switch (pp.p.op) {
+      case GPUQ_EQ: can_match = pp.p.i64[0] >= mn && pp.p.i64[0] <= mx; break;
+      case GPUQ_LT: can_match = mn < pp.p.i64[0]; break;
+      case GPUQ_LE: can_match = mn <= pp.p.i64[0]; break;
+      case GPUQ_GT: can_match = mx > pp.p.i64[0]; break;
+      case GPUQ_GE: can_match = mx >= pp.p.i64[0]; break;
+      case GPUQ_BETWEEN:
+        can_match = (pp.p.hi_exclusive ? mn < pp.p.i64[1] : mn <= pp.p.i64[1]) &&
+                    mx >= pp.p.i64[0];
+        break;
+      default: break;
+    }
+    if (!can_match) return true;
+  }
+  return false;
+}
+
+// evaluate a string predicate against one dict entry (host; LUT build)
+bool eval_str_pred(const gpuq_pred& p, const std::string& lit,
+                   const uint8_t* s, uint32_t len) {
+  if (p.op == GPUQ_CONTAINS) {
+    if (lit.empty()) return true;
+    if (len < lit.size()) return false;
+    return memmem(s, len, lit.data(), lit.size()) != nullptr;
+  }
+  int c = memcmp(s, lit.data(), std::min((size_t)len, lit.size()));
+  if (c == 0) c = (len < lit.size()) ? -1 : (len > lit.size() ? 1 : 0);
+  switch (p.op) {
+    case GPUQ_EQ: return c == 0;
+    case GPUQ_NE: return c != 0;
+    case GPUQ_LT: return c < 0;
+    case GPUQ_LE: return c <= 0;
+    case GPUQ_GT: return c > 0;
+    case GPUQ_GE: return c >= 0;
+  }
+  return false;
+}
+
+int64_t now_ns() {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return (int64_t)ts.tv_sec * 1000000000 + ts.tv_nsec;
+}
+
+}  // namespace
+
+extern "C" gpuq_plan* gpuq_plan_build(
+    gpuq_ctx* ctx, const gpuq_file* files, int32_t n_files,
+    const char* const* projection, int32_t n_projection,
+    const gpuq_pred* preds, int32_t n_preds,
+    const char* const* group_by, int32_t n_group_by,
+    const gpuq_agg* aggs, int32_t n_aggs, int64_t limit) try {
+  (void)projection; (void)n_projection;
+  auto plan = std::make_unique<gpuq_plan>();
+  plan->ctx = ctx;
+  plan->limit = limit;
+  if (ctx->devices.empty())
+    throw std::runtime_error("no GPU devices in session (gpuq never falls back to CPU)");
+
+  // --- map files + parse footers ---
+  for (int32_t i = 0; i < n_files; i++) {
+    auto mf = std::make_unique<MappedFile>();
+    mf->path = files[i].path;
+    mf->fd = open(files[i].path, O_RDONLY);
+    if (mf->fd < 0) throw std::runtime_error("cannot open " + mf->path);
+    struct stat st;
+    fstat(mf->fd, &st);
+    mf->size = st.st_size;
+    mf->data = (const uint8_t*)mmap(nullptr, mf->size, PROT_READ, MAP_PRIVATE, mf->fd, 0);
+    if (mf->data == MAP_FAILED) throw std::runtime_error("mmap failed: " + mf->path);
+    mf->meta = parse_footer(mf->data, mf->size);
+    plan->files.push_back(std::move(mf));
+  }
+  const FileMeta& fm0 = plan->files[0]->meta;
+
+  // --- query spec -> column roles ---
+  for (int32_t i = 0; i < n_preds; i++) {
+    PredPlan pp;
+    pp.p = preds[i];
+    pp.col = preds[i].column;
+    if (preds[i].str) pp.str_lit = preds[i].str;
+    plan->preds.push_back(std::move(pp));
+  }
+  for (int32_t i = 0; i < n_group_by; i++) {
+    int ci = find_or_add_col(plan->cols, group_by[i]);
+    plan->cols[ci].need_gid = true;
+    plan->group_cols.push_back(ci);
+  }
+  for (int32_t i = 0; i < n_aggs; i++) {
+    AggPlan ap;
+    ap.op = aggs[i].op;
+    if (aggs[i].op != GPUQ_AGG_COUNT_STAR) {
+      ap.col = aggs[i].column;
+      ap.col_idx = find_or_add_col(plan->cols, ap.col);
+    }
+    plan->aggs.push_back(std::move(ap));
+  }
+  // resolve column phys types from schema + route predicates
+  for (auto& c : plan->cols) {
+    int si = fm0.col_index(c.name);
+    if (si < 0) throw std::runtime_error("no such column: " + c.name);
+    c.phys = fm0.columns[si].phys_type;
+    c.optional = fm0.columns[si].optional;
+    if (c.need_gid && c.phys != PT_BYTE_ARRAY)
+      throw std::runtime_error("group key must be utf8: " + c.name);
+  }
+  for (size_t pi = 0; pi < plan->preds.size(); pi++) {
+    auto& pp = plan->preds[pi];
+    int ci = find_or_add_col(plan->cols, pp.col);
+    auto& c = plan->cols[ci];
+    if (c.phys == -1) {
+      int si = fm0.col_index(c.name);
+      if (si < 0) throw std::runtime_error("no such column: " + c.name);
+      c.phys = fm0.columns[si].phys_type;
+      c.optional = fm0.columns[si].optional;
+    }
+    if (c.phys == PT_BYTE_ARRAY) {
+      if (pp.p.lit_kind != GPUQ_LIT_STR)
+        throw std::runtime_error("string column needs string literal: " + pp.col);
+      if (pp.p.op == GPUQ_CONTAINS) c.contains_preds.push_back((int)pi);
+      c.lut_preds.push_back((int)pi);   // dict pages via LUT (incl. contains)
+    } else if (c.phys == PT_INT64 || c.phys == PT_INT32) {
+      if (pp.p.lit_kind != GPUQ_LIT_I64)
+        throw std::runtime_error("int column needs int literal: " + pp.col);
+      c.cmp_preds.push_back((int)pi);
+      c.need_val = true;
+    } else {
+      throw std::runtime_error("predicates on f64 columns: next row (SURVEY §8f)");
+    }
+  }
+  // aggregate kinds
+  for (auto& ap : plan->aggs) {
+    if (ap.op == GPUQ_AGG_COUNT_STAR) { ap.kind = AGGK_COUNT_STAR; continue; }
+    auto& c = plan->cols[ap.col_idx];
+    ap.is_f64 = (c.phys == PT_DOUBLE || c.phys == PT_FLOAT);
+    if (c.phys == PT_BYTE_ARRAY && ap.op != GPUQ_AGG_COUNT)
+      throw std::runtime_error("utf8 min/max: next row (SURVEY §8f)");
+    if (c.phys == PT_BYTE_ARRAY) {          // COUNT(utf8): validity only
+      c.need_gid = true;
+      c.need_gid_valid = true;
+    } else {
+      c.need_val = true;
+    }
+    switch (ap.op) {
+      case GPUQ_AGG_COUNT: ap.kind = AGGK_COUNT; break;
+      case GPUQ_AGG_SUM: ap.kind = ap.is_f64 ? AGGK_SUM_F64 : AGGK_SUM_I64; break;
+      case GPUQ_AGG_MIN:
+        if (ap.is_f64) throw std::runtime_error("f64 min/max: next row");
+        ap.kind = AGGK_MIN_I64; break;
+      case GPUQ_AGG_MAX:
+        if (ap.is_f64) throw std::runtime_error("f64 min/max: next row");
+        ap.kind = AGGK_MAX_I64; break;
+      default: throw std::runtime_error("bad agg op");
+    }
+  }
+  if (plan->aggs.empty())
+    throw std::runtime_error("projection-only scans: next row (SURVEY §8f #4); give aggregates");
+  if (plan->group_cols.size() > (size_t)MAX_KEYS || plan->aggs.size() > (size_t)MAX_AGGS)
+    throw std::runtime_error("too many keys/aggregates");
+
+  // --- select row groups (caller's list + footer-stats pruning) ---
+  std::vector<RgRef> selected;
+  for (int32_t i = 0; i < n_files; i++) {
+    const auto& mf = *plan->files[i];
+    std::vector<int32_t> rg_list;
+    if (files[i].row_groups && files[i].n_row_groups >= 0)
+      rg_list.assign(files[i].row_groups, files[i].row_groups + files[i].n_row_groups);
+    else
+      for (size_t g = 0; g < mf.meta.row_groups.size(); g++) rg_list.push_back((int32_t)g);
+    for (int32_t g : rg_list) {
+      const auto& rg = mf.meta.row_groups[g];
+      if (rg_pruned_by_stats(*plan, mf.meta, rg)) continue;
+      RgRef r;
+      r.file_idx = i;
+      r.rg_idx = g;
+      r.rows = rg.num_rows;
+      r.total_bytes = rg.total_compressed_size;
+      for (auto& c : plan->cols) {
+        int si = mf.meta.col_index(c.name);
+        if (si >= 0) r.needed_bytes += rg.chunks[si].total_compressed_size;
+      }
+      selected.push_back(r);
+    }
+  }
+
+  // --- byte-balanced assignment to devices (balanced_file_groups,
+  //     stream_schema_provider.rs:146-165: greedy into least-loaded group) ---
+  int n_parts = (int)ctx->devices.size();
+  plan->parts.resize(n_parts);
+  for (int p = 0; p < n_parts; p++) plan->parts[p].device = ctx->devices[p];
+  std::vector<int64_t> load(n_parts, 0);
+  std::stable_sort(selected.begin(), selected.end(),
+                   [](const RgRef& a, const RgRef& b) { return a.needed_bytes > b.needed_bytes; });
+  for (auto& r : selected) {
+    int best = 0;
+    for (int p = 1; p < n_parts; p++) if (load[p] < load[best]) best = p;
+    load[best] += r.needed_bytes;
+    plan->parts[best].rgs.push_back(r);
+  }
+
+  // --- per-partition host build: page walks, dict processing, device images ---
+  for (auto& part : plan->parts) {
+    uint32_t row_cursor = 0;
+    for (auto& r : part.rgs) {
+      r.row_start = row_cursor;
+      row_cursor += (uint32_t)r.rows;
+      part.rowgroup_bytes_total += r.total_bytes;
+    }
+    part.n_rows = row_cursor;
+
+    for (auto& r : part.rgs) {
+      const auto& mf = *plan->files[r.file_idx];
+      const auto& rg = mf.meta.row_groups[r.rg_idx];
+      for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+        auto& c = plan->cols[ci];
+        int si = mf.meta.col_index(c.name);
+        if (si < 0) throw std::runtime_error("column missing in file: " + c.name);
+        const auto& cm = rg.chunks[si];
+        ChunkTask t;
+        t.file_idx = r.file_idx; t.rg_idx = r.rg_idx; t.col_idx = (int)ci;
+        t.cm = &cm;
+        t.raw_off = part.raw_bytes;
+        part.raw_bytes += cm.total_compressed_size;
+        part.bytes_scanned += cm.total_compressed_size;
+        t.pages = walk_pages(mf.data, cm, rg.num_rows);
+
+        // host-side dictionary processing
+        for (auto& pi : t.pages) {
+          if (pi.type != PAGE_DICT) continue;
+          std::vector<uint8_t> dbuf(pi.uncomp_size);
+          const uint8_t* d;
+          if (cm.codec == CODEC_UNCOMPRESSED || pi.comp_size == pi.uncomp_size) {
+            d = mf.data + pi.payload_off;
+          } else if (cm.codec == CODEC_LZ4_RAW) {
+            int n = lz4_decompress_host(mf.data + pi.payload_off, pi.comp_size,
+                                        dbuf.data(), dbuf.size());
+            if (n != pi.uncomp_size) throw std::runtime_error("dict page lz4 failure");
+            d = dbuf.data();
+          } else throw std::runtime_error("unsupported codec");
+          if (c.phys == PT_BYTE_ARRAY) {
+            const uint8_t* q = d;
+            t.remap.reserve(pi.num_values);
+            t.lut.reserve(pi.num_values);
+            for (int32_t k = 0; k < pi.num_values; k++) {
+              uint32_t l; memcpy(&l, q, 4); q += 4;
+              if (c.need_gid)
+                t.remap.push_back(c.gid_of(std::string((const char*)q, l)));
+              if (!c.lut_preds.empty()) {
+                uint8_t ok = 1;
+                for (int pidx : c.lut_preds) {
+                  const auto& pp = plan->preds[pidx];
+                  ok &= (uint8_t)eval_str_pred(pp.p, pp.str_lit, q, l);
+                }
+                t.lut.push_back(ok);
+              }
+              q += l;
+            }
+          } else if (c.phys == PT_INT64) {
+            t.dictv.resize(pi.num_values);
+            memcpy(t.dictv.data(), d, 8 * (size_t)pi.num_values);
+          } else if (c.phys == PT_INT32) {
+            t.dictv.resize(pi.num_values);
+            for (int32_t k = 0; k < pi.num_values; k++) {
+              int32_t v; memcpy(&v, d + 4 * (size_t)k, 4); t.dictv[k] = v;
+            }
+          } else if (c.phys == PT_DOUBLE) {
+            t.dictv.resize(pi.num_values);
+            memcpy(t.dictv.data(), d, 8 * (size_t)pi.num_values);
+          }
+        }
+        for (auto& pi : t.pages)
+          if (pi.type == PAGE_DATA && pi.encoding == ENC_PLAIN)
+            t.has_plain_data_pages = true;
+        if (t.has_plain_data_pages && c.need_gid)
+          throw std::runtime_error("group key with PLAIN fallback pages (high-cardinality "
+                                   "key hashing): next row (SURVEY §8f) — " + c.name);
+        part.chunks.push_back(std::move(t));
+      }
+    }
+
+    // device page descriptors + aux pools + task lists
+    for (auto& t : part.chunks) {
+      auto& c = plan->cols[t.col_idx];
+      uint32_t remap_base = (uint32_t)part.remap_pool.size();
+      uint32_t dictv_base = (uint32_t)part.dictv_pool.size();
+      uint32_t lut_base = (uint32_t)part.lut_pool.size();
+      part.remap_pool.insert(part.remap_pool.end(), t.remap.begin(), t.remap.end());
+      part.dictv_pool.insert(part.dictv_pool.end(), t.dictv.begin(), t.dictv.end());
+      part.lut_pool.insert(part.lut_pool.end(), t.lut.begin(), t.lut.end());
+
+      const auto& mf = *plan->files[t.file_idx];
+      uint32_t rstart = 0;
+      for (auto& r : part.rgs)
+        if (r.file_idx == t.file_idx && r.rg_idx == t.rg_idx) { rstart = r.row_start; break; }
+
+      uint32_t row_in_rg = 0;
+      for (auto& pi : t.pages) {
+        if (pi.type != PAGE_DATA) continue;
+        DevPage dp{};
+        dp.src_off = t.raw_off + (uint64_t)(pi.payload_off - t.cm->start_offset());
+        dp.dst_off = part.dec_bytes;
+        part.dec_bytes += (uint64_t)pi.uncomp_size;
+        dp.comp_size = pi.comp_size;
+        dp.uncomp_size = pi.uncomp_size;
+        dp.num_values = pi.num_values;
+        dp.row_start = rstart + row_in_rg;
+        row_in_rg += pi.num_values;
+        dp.optional = mf.meta.columns[mf.meta.col_index(c.name)].optional;
+        dp.raw_copy = (t.cm->codec == CODEC_UNCOMPRESSED);
+        dp.encoding = (uint8_t)pi.encoding;
+        dp.phys = (uint8_t)c.phys;
+        int32_t page_id = (int32_t)part.pages.size();
+
+        bool dict_enc = (pi.encoding == ENC_RLE_DICT || pi.encoding == ENC_PLAIN_DICT);
+        if (c.need_gid && dict_enc) {
+          dp.aux = remap_base;
+          part.tasks[{TK_DICT_GID, t.col_idx}].push_back(page_id);
+        }
+        if (c.need_val) {
+          if (dict_enc) {
+            DevPage dv = dp; dv.aux = dictv_base;
+            part.tasks[{TK_DICT_VAL, t.col_idx}].push_back(page_id);
+            dp.aux = dictv_base;
+          } else if (pi.encoding == ENC_PLAIN) {
+            part.tasks[{TK_PLAIN_VAL, t.col_idx}].push_back(page_id);
+          } else if (pi.encoding == ENC_DELTA_BP) {
+            part.tasks[{TK_DELTA_VAL, t.col_idx}].push_back(page_id);
+          } else throw std::runtime_error("unsupported encoding for values");
+        }
+        if (!c.lut_preds.empty()) {
+          if (dict_enc) {
+            dp.aux = lut_base;
+            part.tasks[{TK_DICT_MASK, t.col_idx}].push_back(page_id);
+          } else if (pi.encoding == ENC_PLAIN) {
+            // PLAIN fallback page of a string column: only CONTAINS supported
+            if ((int)c.contains_preds.size() != (int)c.lut_preds.size())
+              throw std::runtime_error("non-LIKE predicate on PLAIN utf8 pages: next row");
+            part.tasks[{TK_BYTES_CONTAINS, t.col_idx}].push_back(page_id);
+          } else throw std::runtime_error("unsupported encoding for string predicate");
+        }
+        part.pages.push_back(dp);
+      }
+      if (row_in_rg != (uint32_t)mf.meta.row_groups[t.rg_idx].num_rows)
+        throw std::runtime_error("page rows mismatch");
+    }
+    part.dec_bytes += 64;  // over-read pad for bit unpackers
+  }
+
+  // group table size
+  int64_t g = 1;
+  for (int ci : plan->group_cols) g *= (int64_t)plan->cols[ci].gdict.size() + 1;
+  if (g > (int64_t)(1 << 22))
+    throw std::runtime_error("group-key cardinality product too large: hash fallback is a next row");
+  plan->n_groups = (int32_t)g;
+
+  for (auto& part : plan->parts)
+    plan->m_rows_scanned += part.n_rows;
+
+  return plan.release();
+} catch (const std::exception& e) {
+  if (ctx) ctx->set_error(e.what());
+  return nullptr;
+}
+
+extern "C" int32_t gpuq_plan_partition_count(gpuq_plan* p) {
+  return p ? (int32_t)p->parts.size() : 0;
+}
+
+// ------------------------------------------------------------------
+// load: stage raw chunks + aux pools into HBM
+// ------------------------------------------------------------------
+extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
+  if (!plan || pi < 0 || pi >= (int32_t)plan->parts.size()) return -1;
+  Partition& part = plan->parts[pi];
+  if (part.loaded) return 0;
+  int64_t t0 = now_ns();
+  HIP_TRY(hipSetDevice(part.device));
+  HIP_TRY(hipStreamCreateWithFlags(&part.stream, hipStreamNonBlocking));
+
+  HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes, 16)));
+  HIP_TRY(hipMalloc(&part.d_dec, std::max<uint64_t>(part.dec_bytes, 16)));
+  for (auto& t : part.chunks) {
+    const auto& mf = *plan->files[t.file_idx];
+    HIP_TRY(hipMemcpyAsync(part.d_raw + t.raw_off, mf.data + t.cm->start_offset(),
+                           t.cm->total_compressed_size, hipMemcpyHostToDevice,
+                           part.stream));
+  }
+  HIP_TRY(hipMalloc(&part.d_pages, std::max<size_t>(part.pages.size() * sizeof(DevPage), 16)));
+  HIP_TRY(hipMemcpyAsync(part.d_pages, part.pages.data(),
+                         part.pages.size() * sizeof(DevPage), hipMemcpyHostToDevice,
+                         part.stream));
+  auto upload_pool = [&](const void* src, size_t bytes, void** dst) {
+    HIP_TRY(hipMalloc(dst, std::max<size_t>(bytes, 16)));
+    if (bytes)
+      HIP_TRY(hipMemcpyAsync(*dst, src, bytes, hipMemcpyHostToDevice, part.stream));
+  };
+  upload_pool(part.remap_pool.data(), part.remap_pool.size() * 4, (void**)&part.d_remap);
+  upload_pool(part.dictv_pool.data(), part.dictv_pool.size() * 8, (void**)&part.d_dictv);
+  upload_pool(part.lut_pool.data(), part.lut_pool.size(), (void**)&part.d_lut);
+  for (auto& kv : part.tasks) {
+    int32_t* d = nullptr;
+    upload_pool(kv.second.data(), kv.second.size() * 4, (void**)&d);
+    part.d_ids[kv.first] = d;
+  }
+  // column outputs
+  for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+    auto& c = plan->cols[ci];
+    if (c.need_gid) {
+      int32_t* d; HIP_TRY(hipMalloc(&d, std::max<int64_t>(part.n_rows * 4, 16)));
+      part.d_gid[(int)ci] = d;
+      if (c.need_gid_valid) {
+        uint8_t* v; HIP_TRY(hipMalloc(&v, std::max<int64_t>(part.n_rows, 16)));
+        part.d_valid[(int)ci] = v;
+      }
+    }
+    if (c.need_val) {
+      int64_t* d; HIP_TRY(hipMalloc(&d, std::max<int64_t>(part.n_rows * 8, 16)));
+      part.d_val[(int)ci] = d;
+      uint8_t* v; HIP_TRY(hipMalloc(&v, std::max<int64_t>(part.n_rows, 16)));
+      part.d_valid[(int)ci] = v;
+    }
+  }
+  HIP_TRY(hipMalloc(&part.d_mask, std::max<int64_t>(part.n_rows, 16)));
+  HIP_TRY(hipMalloc(&part.d_err, 4));
+  size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size()) * 8;
+  HIP_TRY(hipMalloc(&part.d_table, std::max<size_t>(tsz, 16)));
+  std::vector<int32_t> kinds;
+  for (auto& a : plan->aggs) kinds.push_back(a.kind);
+  upload_pool(kinds.data(), kinds.size() * 4, (void**)&part.d_agg_kind);
+  // needle buffer (first CONTAINS pred; one per plan supported per column set)
+  std::string needle;
+  for (auto& pp : plan->preds)
+    if (pp.p.op == GPUQ_CONTAINS) { needle = pp.str_lit; break; }
+  upload_pool(needle.data(), needle.size() + 1, (void**)&part.d_needle);
+
+  HIP_TRY(hipStreamSynchronize(part.stream));
+  part.loaded = true;
+  part.load_ns = now_ns() - t0;
+  {
+    std::lock_guard<std::mutex> g(plan->mu);
+    plan->m_load_ns += part.load_ns;
+  }
+  return 0;
+} catch (const std::exception& e) {
+  if (plan && plan->ctx) plan->ctx->set_error(e.what());
+  return -1;
+}
+
+// ------------------------------------------------------------------
+// execute
+// ------------------------------------------------------------------
+namespace {
+
+// Arrow C export helpers ------------------------------------------------
+struct ExportedBatch {
+  // buffers we hand to Arrow; freed on release
+  std::vector<void*> allocs;
+  void* grab(size_t n) {
+    void* p = malloc(n ? n : 1);
+    allocs.push_back(p);
+    return p;
+  }
+};
+
+void release_schema(struct ArrowSchema* s) {
+  if (!s || !s->release) return;
+  for (int64_t i = 0; i < s->n_children; i++)
+    if (s->children[i]) { release_schema(s->children[i]); free(s->children[i]); }
+  free(s->children);
+  free((void*)s->format);
+  free((void*)s->name);
+  s->release = nullptr;
+}
+void release_array(struct ArrowArray* a) {
+  if (!a || !a->release) return;
+  for (int64_t i = 0; i < a->n_children; i++)
+    if (a->children[i]) { release_array(a->children[i]); free(a->children[i]); }
+  free(a->children);
+  if (a->private_data) {
+    auto* eb = (ExportedBatch*)a->private_data;
+    for (void* p : eb->allocs) free(p);
+    delete eb;
+  }
+  free(a->buffers);
+  a->release = nullptr;
+}
+
+void make_schema_field(struct ArrowSchema* s, const char* fmt, const std::string& name) {
+  memset(s, 0, sizeof(*s));
+  s->format = strdup(fmt);
+  s->name = strdup(name.c_str());
+  s->flags = 2;  // ARROW_FLAG_NULLABLE
+  s->release = release_schema;
+}
+
+struct StreamState {
+  struct ArrowSchema schema;
+  bool schema_moved = false;
+  struct ArrowArray batch;
+  bool batch_taken = false;
+  std::string err;
+};
+
+int ss_get_schema(struct ArrowArrayStream* st, struct ArrowSchema* out) {
+  auto* s = (StreamState*)st->private_data;
+  // deep-ish copy: rebuild (simplest: move once; pyarrow calls once)
+  *out = s->schema;
+  s->schema_moved = true;
+  s->schema.release = nullptr;
+  return 0;
+}
+int ss_get_next(struct ArrowArrayStream* st, struct ArrowArray* out) {
+  auto* s = (StreamState*)st->private_data;
+  if (s->batch_taken) {
+    memset(out, 0, sizeof(*out));
+    out->release = nullptr;
+    return 0;
+  }
+  *out = s->batch;
+  s->batch.release = nullptr;
+  s->batch_taken = true;
+  return 0;
+}
+const char* ss_get_last_error(struct ArrowArrayStream* st) {
+  auto* s = (StreamState*)st->private_data;
+  return s->err.empty() ? nullptr : s->err.c_str();
+}
+void ss_release(struct ArrowArrayStream* st) {
+  auto* s = (StreamState*)st->private_data;
+  if (s) {
+    if (s->schema.release) release_schema(&s->schema);
+    if (s->batch.release) release_array(&s->batch);
+    delete s;
+  }
+  st->release = nullptr;
+}
+
+}  // namespace
+
+extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
+                                     struct ArrowArrayStream* out) try {
+  if (!plan || pi < 0 || pi >= (int32_t)plan->parts.size()) return -1;
+  Partition& part = plan->parts[pi];
+  if (!part.loaded) {
+    if (gpuq_plan_load(plan, pi) != 0) return -1;
+  }
+  HIP_TRY(hipSetDevice(part.device));
+  hipStream_t st = part.stream;
+  int64_t t0 = now_ns();
+
+  hipEvent_t ev0, ev1, ev_decomp;
+  HIP_TRY(hipEventCreate(&ev0));
+  HIP_TRY(hipEventCreate(&ev1));
+  HIP_TRY(hipEventCreate(&ev_decomp));
+
+  HIP_TRY(hipMemsetAsync(part.d_err, 0, 4, st));
+  HIP_TRY(hipMemsetAsync(part.d_mask, 1, part.n_rows, st));
+  launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
+                    part.d_agg_kind);
+
+  HIP_TRY(hipEventRecord(ev0, st));
+  // 1. decompress every data page
+  {
+    std::vector<int32_t> all_ids(part.pages.size());
+    for (size_t i = 0; i < all_ids.size(); i++) all_ids[i] = (int32_t)i;
+    int32_t* d_all = nullptr;
+    HIP_TRY(hipMalloc(&d_all, std::max<size_t>(all_ids.size() * 4, 16)));
+    HIP_TRY(hipMemcpyAsync(d_all, all_ids.data(), all_ids.size() * 4,
+                           hipMemcpyHostToDevice, st));
+    launch_lz4(st, part.d_raw, part.d_dec, part.d_pages, d_all,
+               (int)all_ids.size(), part.d_err);
+    HIP_TRY(hipEventRecord(ev_decomp, st));
+    // free after stream drains (defer: keep till end of exec and free then)
+    HIP_TRY(hipStreamSynchronize(st));
+    HIP_TRY(hipFree(d_all));
+  }
+
+  // 2. decode + predicate kernels
+  for (auto& kv : part.tasks) {
+    int kind = kv.first.first, col = kv.first.second;
+    int n = (int)kv.second.size();
+    int32_t* ids = part.d_ids[kv.first];
+    switch (kind) {
+      case TK_DICT_GID: {
+        auto it = part.d_valid.find(col);
+        launch_dict_gid(st, part.d_dec, part.d_pages, ids, n, part.d_remap,
+                        part.d_gid[col],
+                        it != part.d_valid.end() ? it->second : nullptr,
+                        part.d_err);
+        break;
+      }
+      case TK_DICT_VAL:
+        launch_dict_i64(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
+                        part.d_val[col], part.d_valid[col], part.d_err);
+        break;
+      case TK_PLAIN_VAL:
+        launch_plain_fixed(st, part.d_dec, part.d_pages, ids, n,
+                           part.d_val[col], part.d_valid[col], part.d_err);
+        break;
+      case TK_DELTA_VAL:
+        launch_delta_i64(st, part.d_dec, part.d_pages, ids, n,
+                         part.d_val[col], part.d_valid[col], part.d_err);
+        break;
+      case TK_DICT_MASK:
+        launch_dict_mask(st, part.d_dec, part.d_pages, ids, n, part.d_lut,
+                         part.d_mask, part.d_err);
+        break;
+      case TK_BYTES_CONTAINS: {
+        std::string needle;
+        for (int pidx : plan->cols[col].contains_preds)
+          needle = plan->preds[pidx].str_lit;  // single CONTAINS per col
+        launch_bytes_contains(st, part.d_dec, part.d_pages, ids, n,
+                              part.d_needle, (int)needle.size(),
+                              part.d_mask, part.d_err);
+        break;
+      }
+    }
+  }
+  // i64 comparisons on decoded arrays
+  for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+    auto& c = plan->cols[ci];
+    for (int pidx : c.cmp_preds) {
+      const auto& pp = plan->preds[pidx];
+      int mode;
+      switch (pp.p.op) {
+        case GPUQ_EQ: mode = CMP_EQ; break;
+        case GPUQ_NE: mode = CMP_NE; break;
+        case GPUQ_LT: mode = CMP_LT; break;
+        case GPUQ_LE: mode = CMP_LE; break;
+        case GPUQ_GT: mode = CMP_GT; break;
+        case GPUQ_GE: mode = CMP_GE; break;
+        case GPUQ_BETWEEN: mode = CMP_RANGE; break;
+        default: throw std::runtime_error("bad cmp op");
+      }
+      launch_cmp_i64(st, part.d_val[(int)ci], part.d_valid[(int)ci],
+                     pp.p.i64[0], pp.p.i64[1], mode, pp.p.hi_exclusive,
+                     part.d_mask, part.n_rows);
+    }
+  }
+
+  // 3. aggregate
+  AggArgs a{};
+  a.mask = part.d_mask;
+  a.n_rows = part.n_rows;
+  a.n_keys = (int)plan->group_cols.size();
+  for (int k = 0; k < a.n_keys; k++) {
+    int ci = plan->group_cols[k];
+    a.key_gid[k] = part.d_gid[ci];
+    a.key_size[k] = (int32_t)plan->cols[ci].gdict.size() + 1;
+  }
+  a.n_aggs = (int)plan->aggs.size();
+  for (int i = 0; i < a.n_aggs; i++) {
+    const auto& ap = plan->aggs[i];
+    a.agg_kind[i] = ap.kind;
+    if (ap.col_idx >= 0) {
+      auto itv = part.d_val.find(ap.col_idx);
+      a.agg_val[i] = itv != part.d_val.end() ? itv->second : nullptr;
+      auto itd = part.d_valid.find(ap.col_idx);
+      a.agg_valid[i] = itd != part.d_valid.end() ? itd->second : nullptr;
+    }
+  }
+  a.table = part.d_table;
+  a.n_groups = plan->n_groups;
+  launch_agg(st, a);
+  HIP_TRY(hipEventRecord(ev1, st));
+
+  // 4. D2H results
+  size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size());
+  std::vector<uint64_t> table(tsz);
+  HIP_TRY(hipMemcpyAsync(table.data(), part.d_table, tsz * 8,
+                         hipMemcpyDeviceToHost, st));
+  int32_t herr = 0;
+  HIP_TRY(hipMemcpyAsync(&herr, part.d_err, 4, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  if (herr != 0)
+    throw std::runtime_error("kernel error code " + std::to_string(herr));
+
+  float ms_total = 0, ms_decomp = 0;
+  HIP_TRY(hipEventElapsedTime(&ms_total, ev0, ev1));
+  HIP_TRY(hipEventElapsedTime(&ms_decomp, ev0, ev_decomp));
+  HIP_TRY(hipEventDestroy(ev0));
+  HIP_TRY(hipEventDestroy(ev1));
+  HIP_TRY(hipEventDestroy(ev_decomp));
+
+  // 5. assemble the PARTIAL aggregate batch
+  int n_keys = (int)plan->group_cols.size();
+  int n_aggs = (int)plan->aggs.size();
+  int slots = 1 + 2 * n_aggs;
+  std::vector<int64_t> live;  // group ids with presence > 0
+  for (int64_t g = 0; g < plan->n_groups; g++)
+    if (table[(size_t)g * slots]) live.push_back(g);
+  int64_t nr = (int64_t)live.size();
+  // no-group aggregate over empty selection: emit the single empty row
+  bool empty_aggregate_row = (n_keys == 0 && nr == 0);
+  if (empty_aggregate_row) { live.push_back(0); nr = 1; }
+
+  auto* ss = new StreamState();
+  memset(&ss->schema, 0, sizeof(ss->schema));
+  ss->schema.format = strdup("+s");
+  ss->schema.name = strdup("");
+  ss->schema.release = release_schema;
+  int n_fields = n_keys + 1 + 2 * n_aggs;
+  ss->schema.n_children = n_fields;
+  ss->schema.children = (struct ArrowSchema**)calloc(n_fields, sizeof(void*));
+  int f = 0;
+  for (int k = 0; k < n_keys; k++) {
+    ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
+    make_schema_field(ss->schema.children[f], "u", plan->cols[plan->group_cols[k]].name);
+    f++;
+  }
+  ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
+  make_schema_field(ss->schema.children[f], "l", "__presence");
+  f++;
+  for (int i = 0; i < n_aggs; i++) {
+    ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
+    make_schema_field(ss->schema.children[f],
+                      plan->aggs[i].kind == AGGK_SUM_F64 ? "g" : "l",
+                      "agg" + std::to_string(i));
+    f++;
+    ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
+    make_schema_field(ss->schema.children[f], "l", "agg" + std::to_string(i) + "_count");
+    f++;
+  }
+
+  auto* eb = new ExportedBatch();
+  memset(&ss->batch, 0, sizeof(ss->batch));
+  ss->batch.length = nr;
+  ss->batch.n_buffers = 1;
+  ss->batch.buffers = (const void**)calloc(1, sizeof(void*));
+  ss->batch.n_children = n_fields;
+  ss->batch.children = (struct ArrowArray**)calloc(n_fields, sizeof(void*));
+  ss->batch.release = release_array;
+  ss->batch.private_data = eb;
+
+  auto make_child = [&](int idx) {
+    auto* ch = (struct ArrowArray*)calloc(1, sizeof(struct ArrowArray));
+    ss->batch.children[idx] = ch;
+    ch->length = nr;
+    ch->release = release_array;
+    return ch;
+  };
+
+  // decode combined gid -> per-key local gids
+  std::vector<std::vector<int32_t>> key_gids(n_keys, std::vector<int32_t>(nr));
+  for (int64_t r = 0; r < nr; r++) {
+    int64_t g = live[r];
+    for (int k = n_keys - 1; k >= 0; k--) {
+      int ci = plan->group_cols[k];
+      int32_t sz = (int32_t)plan->cols[ci].gdict.size() + 1;
+      key_gids[k][r] = (int32_t)(g % sz);
+      g /= sz;
+    }
+  }
+  f = 0;
+  for (int k = 0; k < n_keys; k++) {
+    auto* ch = make_child(f++);
+    auto& c = plan->cols[plan->group_cols[k]];
+    ch->n_buffers = 3;
+    ch->buffers = (const void**)calloc(3, sizeof(void*));
+    uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
+    memset(validity, 0, (nr + 7) / 8);
+    int32_t* offs = (int32_t*)eb->grab((nr + 1) * 4);
+    size_t total = 0;
+    for (int64_t r = 0; r < nr; r++) {
+      int32_t gid = key_gids[k][r];
+      if (gid > 0) total += c.gdict[gid - 1].size();
+    }
+    char* data = (char*)eb->grab(total);
+    size_t off = 0;
+    int64_t nulls = 0;
+    for (int64_t r = 0; r < nr; r++) {
+      offs[r] = (int32_t)off;
+      int32_t gid = key_gids[k][r];
+      if (gid > 0) {
+        validity[r / 8] |= (uint8_t)(1 << (r % 8));
+        const auto& s = c.gdict[gid - 1];
+        memcpy(data + off, s.data(), s.size());
+        off += s.size();
+      } else nulls++;
+    }
+    offs[nr] = (int32_t)off;
+    ch->null_count = nulls;
+    ch->buffers[0] = nulls ? validity : nullptr;
+    ch->buffers[1] = offs;
+    ch->buffers[2] = data;
+  }
+  // presence
+  {
+    auto* ch = make_child(f++);
+    ch->n_buffers = 2;
+    ch->buffers = (const void**)calloc(2, sizeof(void*));
+    int64_t* v = (int64_t*)eb->grab(nr * 8);
+    for (int64_t r = 0; r < nr; r++)
+      v[r] = empty_aggregate_row ? 0 : (int64_t)table[(size_t)live[r] * slots];
+    ch->buffers[1] = v;
+  }
+  int64_t rows_out_total = 0;
+  for (int64_t r = 0; r < nr; r++)
+    rows_out_total += empty_aggregate_row ? 0 : (int64_t)table[(size_t)live[r] * slots];
+  // aggs
+  for (int i = 0; i < n_aggs; i++) {
+    auto* chv = make_child(f++);
+    chv->n_buffers = 2;
+    chv->buffers = (const void**)calloc(2, sizeof(void*));
+    int64_t* vv = (int64_t*)eb->grab(nr * 8);
+    uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
+    memset(validity, 0xff, (nr + 7) / 8);
+    int64_t nulls = 0;
+    for (int64_t r = 0; r < nr; r++) {
+      size_t base = (size_t)live[r] * slots;
+      uint64_t cnt = empty_aggregate_row ? 0 : table[base + 2 + 2 * i];
+      uint64_t val = empty_aggregate_row ? 0 : table[base + 1 + 2 * i];
+      int kind = plan->aggs[i].kind;
+      if (kind == AGGK_COUNT_STAR || kind == AGGK_COUNT) {
+        vv[r] = (int64_t)cnt;
+      } else if (cnt == 0) {
+        vv[r] = 0;
+        validity[r / 8] &= (uint8_t)~(1 << (r % 8));
+        nulls++;
+      } else {
+        vv[r] = (int64_t)val;
+      }
+    }
+    if (nulls) {
+      chv->n_buffers = 2;
+      chv->buffers = (const void**)realloc(chv->buffers, 2 * sizeof(void*));
+      chv->buffers[0] = validity;
+      chv->null_count = nulls;
+    }
+    chv->buffers[1] = vv;
+
+    auto* chc = make_child(f++);
+    chc->n_buffers = 2;
+    chc->buffers = (const void**)calloc(2, sizeof(void*));
+    int64_t* cv = (int64_t*)eb->grab(nr * 8);
+    for (int64_t r = 0; r < nr; r++)
+      cv[r] = empty_aggregate_row ? 0 : (int64_t)table[(size_t)live[r] * slots + 2 + 2 * i];
+    chc->buffers[1] = cv;
+  }
+
+  memset(out, 0, sizeof(*out));
+  out->get_schema = ss_get_schema;
+  out->get_next = ss_get_next;
+  out->get_last_error = ss_get_last_error;
+  out->release = ss_release;
+  out->private_data = ss;
+
+  {
+    std::lock_guard<std::mutex> g(plan->mu);
+    plan->m_kernel_ns += (int64_t)(ms_total * 1e6);
+    plan->m_decomp_ns += (int64_t)(ms_decomp * 1e6);
+    plan->m_exec_ns += now_ns() - t0;
+    plan->m_rows_out += rows_out_total;
+  }
+  return 0;
+} catch (const std::exception& e) {
+  if (plan && plan->ctx) plan->ctx->set_error(e.what());
+  return -1;
+}
+
+extern "C" int32_t gpuq_plan_metrics(gpuq_plan* p, gpuq_metrics* out) {
+  if (!p || !out) return -1;
+  std::lock_guard<std::mutex> g(p->mu);
+  memset(out, 0, sizeof(*out));
+  out->rows_scanned = p->m_rows_scanned;
+  out->rows_out = p->m_rows_out;
+  for (auto& part : p->parts) {
+    out->bytes_scanned += part.bytes_scanned;
+    out->rowgroup_bytes_total += part.rowgroup_bytes_total;
+    out->hbm_bytes_est += (int64_t)(part.raw_bytes + 2 * part.dec_bytes);
+  }
+  out->kernel_ns = p->m_kernel_ns;
+  out->exec_ns = p->m_exec_ns;
+  out->load_ns = p->m_load_ns;
+  out->decomp_ns = p->m_decomp_ns;
+  return 0;
+}
+
+gpuq_plan::~gpuq_plan() {
+  for (auto& part : parts) {
+    if (!part.loaded) continue;
+    hipSetDevice(part.device);
+    auto F = [](void* p) { if (p) hipFree(p); };
+    F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
+    F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
+    F(part.d_table); F(part.d_agg_kind); F(part.d_needle);
+    for (auto& kv : part.d_ids) F(kv.second);
+    for (auto& kv : part.d_gid) F(kv.second);
+    for (auto& kv : part.d_val) F(kv.second);
+    for (auto& kv : part.d_valid) F(kv.second);
+    if (part.stream) hipStreamDestroy(part.stream);
+  }
+}
+
+extern "C" void gpuq_plan_destroy(gpuq_plan* p) { delete p; }

@@ -1,0 +1,837 @@
+// gfx950 (CDNA4, MI355X) kernels for the Parseable hot path:
+// LZ4_RAW page decompression, parquet RLE/dictionary + DELTA_BINARY_PACKED
+// decode into device-resident column arrays, predicate masks, and hash
+// group-by with count/sum/min/max.
+//
+// Design notes (HBM-bound integer/byte work — no MFMA, per the north star):
+//  - wavefront = 64; one wave per page for the inherently-serial-per-page
+//    decode chains (LZ4 token stream, RLE run headers), with all 64 lanes
+//    doing the data movement of each run/sequence in parallel;
+//  - parsing is LANE-REDUNDANT (all lanes read the same few header bytes —
+//    a broadcast L1 load) instead of shfl choreography;
+//  - LZ4 match copies read the last-64-KiB window from an LDS ring, so no
+//    global store->load ordering waits are needed inside a page;
+//  - row-parallel kernels (comparisons, aggregation) are grid-stride with
+//    coalesced accesses; aggregation uses a per-block LDS table when the
+//    group table fits, flushed once per block with global atomics.
+#include <hip/hip_runtime.h>
+#include "dev_types.h"
+#include "meta.h"
+
+namespace gpuq {
+
+#define WAVE 64
+
+// ------------------------------------------------------------------
+// LZ4_RAW page decompression: one wave per page.
+// ------------------------------------------------------------------
+__global__ void __launch_bounds__(WAVE)
+k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
+            const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
+            int n, int32_t* __restrict__ d_error) {
+  __shared__ uint8_t ring[65536];  // last 64 KiB of output (LZ4 max offset)
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const uint8_t* src = raw + pg.src_off;
+  uint8_t* dst = dec + pg.dst_off;
+  const int lane = threadIdx.x;
+
+  if (pg.raw_copy || pg.comp_size == pg.uncomp_size) {
+    for (uint32_t i = lane * 16u; i < pg.uncomp_size; i += WAVE * 16u) {
+      uint32_t rem = pg.uncomp_size - i;
+      if (rem >= 16 && (((uintptr_t)(src + i)) & 15) == 0 && (((uintptr_t)(dst + i)) & 15) == 0) {
+        *(uint4*)(dst + i) = *(const uint4*)(src + i);
+      } else {
+        for (uint32_t b = 0; b < 16 && i + b < pg.uncomp_size; b++) dst[i + b] = src[i + b];
+      }
+    }
+    return;
+  }
+
+  uint32_t s = 0, d = 0;
+  const uint32_t comp = pg.comp_size, uncomp = pg.uncomp_size;
+  bool bad = false;
+  while (s < comp && d < uncomp) {
+    // ---- lane-redundant parse ----
+    uint32_t token = src[s++];
+    uint32_t lit = token >> 4;
+    if (lit == 15) {
+      uint32_t b;
+      do { if (s >= comp) { bad = true; break; } b = src[s++]; lit += b; } while (b == 255);
+      if (bad) break;
+    }
+    if (s + lit > comp || d + lit > uncomp) { bad = true; break; }
+    // ---- parallel literal copy (global->global + LDS ring) ----
+    for (uint32_t i = lane; i < lit; i += WAVE) {
+      uint8_t v = src[s + i];
+      dst[d + i] = v;
+      ring[(d + i) & 0xFFFF] = v;
+    }
+    s += lit; d += lit;
+    if (s >= comp) break;  // last sequence: literals only
+    if (s + 2 > comp) { bad = true; break; }
+    uint32_t off = src[s] | ((uint32_t)src[s + 1] << 8);
+    s += 2;
+    if (off == 0 || off > d) { bad = true; break; }
+    uint32_t ml = token & 0xf;
+    if (ml == 15) {
+      uint32_t b;
+      do { if (s >= comp) { bad = true; break; } b = src[s++]; ml += b; } while (b == 255);
+      if (bad) break;
+    }
+    ml += 4;
+    if (d + ml > uncomp) { bad = true; break; }
+    // ---- match copy from the LDS ring, chunked so ring writes never
+    //      overwrite the pattern region another lane still reads ----
+    uint32_t done = 0;
+    while (done < ml) {
+      uint32_t chunk;
+      __syncthreads();  // prior writes (literals / previous chunk) visible
+      if (off < WAVE) {
+        chunk = min(ml - done, 65536u - off);
+        for (uint32_t i = lane; i < chunk; i += WAVE) {
+          uint8_t v = ring[(d + done - off + (i % off)) & 0xFFFF];
+          dst[d + done + i] = v;
+          ring[(d + done + i) & 0xFFFF] = v;
+        }
+      } else {
+        chunk = min(ml - done, min(off, 65536u - off));
+        for (uint32_t i = lane; i < chunk; i += WAVE) {
+          uint8_t v = ring[(d + done - off + i) & 0xFFFF];
+          dst[d + done + i] = v;
+          ring[(d + done + i) & 0xFFFF] = v;
+        }
+      }
+      done += chunk;
+    }
+    __syncthreads();
+    d += ml;
+  }
+  if ((bad || d != uncomp) && lane == 0) atomicExch(d_error, ERR_LZ4);
+}
+
+// ------------------------------------------------------------------
+// serial readers (lane-redundant or lane0) for slow paths
+// ------------------------------------------------------------------
+struct SerialRle {
+  const uint8_t* p;
+  const uint8_t* end;
+  int bit_width, byte_w;
+  // run state
+  uint32_t run_left = 0;
+  bool packed = false;
+  uint32_t rle_val = 0;
+  uint64_t bit_acc = 0;
+  int bit_cnt = 0;
+  __device__ SerialRle(const uint8_t* p_, const uint8_t* end_, int bw)
+      : p(p_), end(end_), bit_width(bw), byte_w((bw + 7) / 8) {}
+  __device__ uint32_t varint() {
+    uint64_t v = 0; int sh = 0;
+    for (;;) {
+      uint8_t b = (p < end) ? *p++ : 0;
+      v |= (uint64_t)(b & 0x7f) << sh;
+      if (!(b & 0x80)) return (uint32_t)v;
+      sh += 7;
+    }
+  }
+  __device__ uint32_t next() {
+    if (!run_left) {
+      uint32_t hdr = varint();
+      if (hdr & 1) {
+        packed = true;
+        run_left = (hdr >> 1) * 8;
+        bit_acc = 0; bit_cnt = 0;
+      } else {
+        packed = false;
+        run_left = hdr >> 1;
+        rle_val = 0;
+        for (int b = 0; b < byte_w; b++)
+          rle_val |= (uint32_t)((p < end) ? *p++ : 0) << (8 * b);
+      }
+    }
+    run_left--;
+    if (!packed) return rle_val;
+    while (bit_cnt < bit_width) {
+      bit_acc |= (uint64_t)((p < end) ? *p++ : 0) << bit_cnt;
+      bit_cnt += 8;
+    }
+    uint32_t v = (uint32_t)(bit_acc & ((bit_width >= 32) ? 0xffffffffull
+                                                         : ((1ull << bit_width) - 1)));
+    bit_acc >>= bit_width;
+    bit_cnt -= bit_width;
+    return v;
+  }
+};
+
+// def-level helpers: v1 page payload = [u32 len][RLE runs] when optional.
+// Returns pointer to values; *all_valid set when a single RLE(1) run covers
+// the page (the overwhelmingly common case for Parseable streams).
+__device__ inline const uint8_t* def_levels(const DevPage& pg, const uint8_t* payload,
+                                            const uint8_t** def_start, uint32_t* def_len,
+                                            bool* all_valid) {
+  if (!pg.optional) { *def_start = nullptr; *def_len = 0; *all_valid = true; return payload; }
+  uint32_t dl;
+  memcpy(&dl, payload, 4);
+  *def_start = payload + 4;
+  *def_len = dl;
+  // quick probe: single RLE run of value 1 covering all values?
+  const uint8_t* q = payload + 4;
+  uint64_t hdr = 0; int sh = 0;
+  for (;;) {
+    uint8_t b = *q++;
+    hdr |= (uint64_t)(b & 0x7f) << sh;
+    if (!(b & 0x80)) break;
+    sh += 7;
+  }
+  *all_valid = (!(hdr & 1)) && ((hdr >> 1) >= pg.num_values) && (*q == 1);
+  return payload + 4 + dl;
+}
+
+// ------------------------------------------------------------------
+// dictionary-index decode (RLE/bit-packed hybrid), one wave per page.
+// Emit policy via template: remap-to-gid, gather-i64, gather-f64, LUT-mask.
+// ------------------------------------------------------------------
+struct EmitGid {
+  const int32_t* remap;
+  int32_t* out;        // row-aligned gid, 0 = NULL
+  uint8_t* valid;      // optional validity bytes (for COUNT(utf8_col))
+  __device__ void operator()(uint32_t row, uint32_t idx) const {
+    out[row] = remap[idx];
+    if (valid) valid[row] = 1;
+  }
+  __device__ void null_at(uint32_t row) const { out[row] = 0; if (valid) valid[row] = 0; }
+};
+struct EmitDictI64 {
+  const int64_t* dictv;
+  int64_t* out;
+  uint8_t* valid;
+  __device__ void operator()(uint32_t row, uint32_t idx) const { out[row] = dictv[idx]; valid[row] = 1; }
+  __device__ void null_at(uint32_t row) const { valid[row] = 0; }
+};
+struct EmitDictMask {
+  const uint8_t* lut;  // per-local-dict-id pred result
+  uint8_t* mask;
+  __device__ void operator()(uint32_t row, uint32_t idx) const { mask[row] &= lut[idx]; }
+  __device__ void null_at(uint32_t row) const { mask[row] = 0; }  // NULL never matches
+};
+
+template <class Emit>
+__device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit emit,
+                                 int32_t* d_error) {
+  const int lane = threadIdx.x;
+  const uint8_t* def_start; uint32_t def_len; bool all_valid;
+  const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
+  const uint32_t nv = pg.num_values;
+  const uint32_t row0 = pg.row_start;
+
+  if (all_valid) {
+    int bw = *vals++;
+    if (bw == 0) {  // all values are dict id 0
+      for (uint32_t i = lane; i < nv; i += WAVE) emit(row0 + i, 0);
+      return;
+    }
+    // lane-redundant run-header walk; data movement parallel per run
+    const uint8_t* p = vals;
+    uint32_t v = 0;
+    while (v < nv) {
+      // varint header (redundant on all lanes)
+      uint64_t hdr = 0; int sh = 0;
+      for (;;) {
+        uint8_t b = *p++;
+        hdr |= (uint64_t)(b & 0x7f) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+      }
+      if (hdr & 1) {
+        uint32_t groups = (uint32_t)(hdr >> 1);
+        // group g: 8 values packed in bw bytes at p + g*bw
+        for (uint32_t g = lane; g < groups; g += WAVE) {
+          const uint8_t* q = p + (size_t)g * bw;
+          uint64_t acc = 0;
+          for (int b = 0; b < bw; b++) acc |= (uint64_t)q[b] << (8 * b);
+          uint32_t base = v + g * 8;
+          uint32_t mask_v = (bw >= 32) ? 0xffffffffu : ((1u << bw) - 1);
+          for (int k = 0; k < 8; k++) {
+            uint32_t idx = (uint32_t)(acc >> (k * bw)) & mask_v;
+            if (bw > 8 && k * bw + bw > 64) {  // straddles u64 window: re-read
+              uint64_t acc2 = 0;
+              const uint8_t* q2 = q + (k * bw) / 8;
+              int shift = (k * bw) % 8;
+              for (int b = 0; b < 8 && q2 + b < p + (size_t)(g + 1) * bw + 8; b++)
+                acc2 |= (uint64_t)q2[b] << (8 * b);
+              idx = (uint32_t)(acc2 >> shift) & mask_v;
+            }
+            if (base + k < nv) emit(row0 + base + k, idx);
+          }
+        }
+        p += (size_t)groups * bw;
+        uint32_t add = groups * 8;
+        v += (add > nv - v) ? (nv - v) : add;
+      } else {
+        uint32_t cnt = (uint32_t)(hdr >> 1);
+        uint32_t val = 0;
+        int byte_w = (bw + 7) / 8;
+        for (int b = 0; b < byte_w; b++) val |= (uint32_t)p[b] << (8 * b);
+        p += byte_w;
+        if (cnt > nv - v) cnt = nv - v;
+        for (uint32_t i = lane; i < cnt; i += WAVE) emit(row0 + v + i, val);
+        v += cnt;
+      }
+    }
+  } else {
+    // nulls present: lane0 serial interleave of def levels and indices
+    if (lane != 0) return;
+    SerialRle def(def_start, def_start + def_len, 1);
+    int bw = *vals++;
+    SerialRle idx(vals, payload + pg.uncomp_size, bw);
+    for (uint32_t r = 0; r < nv; r++) {
+      if (def.next()) emit(row0 + r, bw ? idx.next() : 0);
+      else emit.null_at(row0 + r);
+    }
+  }
+}
+
+template <class Emit>
+__global__ void __launch_bounds__(WAVE)
+k_dict_pages(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+             const int32_t* __restrict__ ids, int n, Emit emit, int32_t* d_error) {
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  DevPage pg = pages[ids[pi]];
+  Emit e = emit;
+  // per-page aux offset
+  e.advance(pg.aux);
+  dict_page_decode(pg, dec + pg.dst_off, e, d_error);
+}
+
+// wrappers adding per-page aux advance
+struct EmitGidP : EmitGid {
+  const int32_t* pool;
+  __device__ void advance(uint32_t aux) { remap = pool + aux; }
+};
+struct EmitDictI64P : EmitDictI64 {
+  const int64_t* pool;
+  __device__ void advance(uint32_t aux) { dictv = pool + aux; }
+};
+struct EmitDictMaskP : EmitDictMask {
+  const uint8_t* pool;
+  __device__ void advance(uint32_t aux) { lut = pool + aux; }
+};
+
+// ------------------------------------------------------------------
+// PLAIN i64 / f64 pages -> row-aligned arrays
+// ------------------------------------------------------------------
+__global__ void __launch_bounds__(WAVE)
+k_plain_fixed(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+              const int32_t* __restrict__ ids, int n,
+              int64_t* __restrict__ out, uint8_t* __restrict__ valid,
+              int32_t* d_error) {
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const int lane = threadIdx.x;
+  const uint8_t* def_start; uint32_t def_len; bool all_valid;
+  const uint8_t* vals = def_levels(pg, dec + pg.dst_off, &def_start, &def_len, &all_valid);
+  const uint32_t nv = pg.num_values, row0 = pg.row_start;
+  if (all_valid) {
+    for (uint32_t i = lane; i < nv; i += WAVE) {
+      int64_t v;
+      memcpy(&v, vals + (size_t)i * 8, 8);
+      out[row0 + i] = v;
+      if (valid) valid[row0 + i] = 1;
+    }
+  } else {
+    if (lane != 0) return;
+    SerialRle def(def_start, def_start + def_len, 1);
+    const uint8_t* q = vals;
+    for (uint32_t r = 0; r < nv; r++) {
+      if (def.next()) {
+        int64_t v; memcpy(&v, q, 8); q += 8;
+        out[row0 + r] = v;
+        if (valid) valid[row0 + r] = 1;
+      } else if (valid) valid[row0 + r] = 0;
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// DELTA_BINARY_PACKED i64 -> row-aligned (one wave per page).
+// Phase A: lane-redundant block-header walk storing per-miniblock
+//   (data offset, bit width, block min_delta ref); Phase B: parallel
+//   per-miniblock delta sums; Phase C: serial scan of miniblock sums;
+//   Phase D: parallel value reconstruction.
+// LDS budget: 8192 miniblocks (262,144 values @ 32/miniblock).
+// ------------------------------------------------------------------
+#define MAX_MB 8192
+#define MAX_BLK 2048
+__global__ void __launch_bounds__(WAVE)
+k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+            const int32_t* __restrict__ ids, int n,
+            int64_t* __restrict__ out, uint8_t* __restrict__ valid,
+            int32_t* d_error) {
+  // LDS budget (fits 160 KiB/CU): 32K off + 8K bw + 16K md + 64K sum = 120K.
+  // mb_sum is re-used in place as the post-scan starting value; the block id
+  // of miniblock m is m / mpb (uniform miniblocks per block).
+  __shared__ uint32_t mb_off[MAX_MB];     // payload-relative offset of miniblock data
+  __shared__ uint8_t mb_bw[MAX_MB];
+  __shared__ int64_t blk_md[MAX_BLK];     // min_delta per block
+  __shared__ int64_t mb_sum[MAX_MB];      // phase B: delta sums; phase C: start values
+
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const int lane = threadIdx.x;
+  const uint8_t* def_start; uint32_t def_len; bool all_valid;
+  const uint8_t* payload = dec + pg.dst_off;
+  const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
+
+  if (!all_valid) {
+    // nulls in the time column: lane0 fully serial (never hit by Parseable
+    // streams — p_timestamp is always set by ingest, event/format/mod.rs:167)
+    if (lane != 0) return;
+    SerialRle def(def_start, def_start + def_len, 1);
+    const uint8_t* q = vals;
+    auto rv = [&]() { uint64_t v = 0; int sh = 0; for (;;) { uint8_t b = *q++; v |= (uint64_t)(b & 0x7f) << sh; if (!(b & 0x80)) return v; sh += 7; } };
+    auto rz = [&]() { uint64_t v = rv(); return (int64_t)(v >> 1) ^ -(int64_t)(v & 1); };
+    uint64_t blk = rv(), mpb = rv(), total = rv();
+    int64_t value = rz();
+    uint64_t per_mini = blk / mpb;
+    uint64_t emitted = 0;
+    int64_t cur_md = 0; uint8_t bws[256]; uint64_t mb = 0, in_mb = 0;
+    uint64_t acc = 0; int nbits = 0;
+    for (uint32_t r = 0; r < pg.num_values; r++) {
+      if (!def.next()) { if (valid) valid[pg.row_start + r] = 0; continue; }
+      int64_t v;
+      if (emitted == 0) v = value;
+      else {
+        if (((emitted - 1) % (per_mini * mpb)) == 0) {  // new block
+          cur_md = rz();
+          for (uint64_t m = 0; m < mpb; m++) bws[m] = *q++;
+          mb = 0; in_mb = 0; acc = 0; nbits = 0;
+        }
+        int bw = bws[mb];
+        uint64_t d = 0;
+        if (bw) {
+          while (nbits < bw) { acc |= (uint64_t)(*q++) << nbits; nbits += 8; }
+          d = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
+          acc >>= bw; nbits -= bw;
+        }
+        value += cur_md + (int64_t)d;
+        v = value;
+        if (++in_mb == per_mini) { in_mb = 0; mb++; acc = 0; nbits = 0; }
+      }
+      emitted++;
+      out[pg.row_start + r] = v;
+      if (valid) valid[pg.row_start + r] = 1;
+    }
+    (void)total;
+    return;
+  }
+
+  // ---- fast path: no nulls ----
+  // Phase A (lane-redundant): header + block walk
+  const uint8_t* q = vals;
+  auto rv = [&]() { uint64_t v = 0; int sh = 0; for (;;) { uint8_t b = *q++; v |= (uint64_t)(b & 0x7f) << sh; if (!(b & 0x80)) return v; sh += 7; } };
+  auto rz = [&]() { uint64_t v = rv(); return (int64_t)(v >> 1) ^ -(int64_t)(v & 1); };
+  uint64_t blk_size = rv(), mpb = rv(), total = rv();
+  int64_t first = rz();
+  uint64_t per_mini = blk_size / mpb;
+  if (per_mini % 8 || total > pg.num_values || mpb > 256) {
+    if (lane == 0) atomicExch(d_error, ERR_DELTA);
+    return;
+  }
+  uint64_t n_deltas = total ? total - 1 : 0;
+  uint32_t n_mb = (uint32_t)((n_deltas + per_mini - 1) / per_mini);
+  uint32_t n_blk = (uint32_t)((n_mb + mpb - 1) / mpb);
+  if (n_mb > MAX_MB || n_blk > MAX_BLK) {
+    if (lane == 0) atomicExch(d_error, ERR_DELTA);
+    return;
+  }
+  // walk blocks redundantly; every lane records into LDS identically
+  {
+    uint32_t mb = 0;
+    for (uint32_t b = 0; b < n_blk; b++) {
+      int64_t md = rz();
+      if (lane == 0) blk_md[b] = md;
+      const uint8_t* bws = q;
+      q += mpb;
+      for (uint64_t m = 0; m < mpb && mb < n_mb; m++, mb++) {
+        if (lane == 0) {
+          mb_off[mb] = (uint32_t)(q - vals);
+          mb_bw[mb] = bws[m];
+        }
+        q += (per_mini * bws[m]) / 8;
+      }
+    }
+  }
+  __syncthreads();
+
+  // Phase B: per-miniblock delta sums (parallel over miniblocks)
+  for (uint32_t m = lane; m < n_mb; m += WAVE) {
+    const uint8_t* p = vals + mb_off[m];
+    int bw = mb_bw[m];
+    int64_t md = blk_md[m / (uint32_t)mpb];
+    uint64_t cnt = per_mini;
+    if ((uint64_t)(m + 1) * per_mini > n_deltas) cnt = n_deltas - (uint64_t)m * per_mini;
+    int64_t s = 0;
+    uint64_t acc = 0; int nbits = 0;
+    for (uint64_t i = 0; i < cnt; i++) {
+      uint64_t dv = 0;
+      if (bw) {
+        while (nbits < bw) { acc |= (uint64_t)(*p++) << nbits; nbits += 8; }
+        dv = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
+        acc >>= bw; nbits -= bw;
+      }
+      s += md + (int64_t)dv;
+    }
+    mb_sum[m] = s;
+  }
+  __syncthreads();
+
+  // Phase C: serial exclusive scan of miniblock sums -> starting value
+  if (lane == 0) {
+    int64_t run = first;
+    for (uint32_t m = 0; m < n_mb; m++) { int64_t s = mb_sum[m]; mb_sum[m] = run; run += s; }
+  }
+  __syncthreads();
+
+  // Phase D: reconstruct values
+  if (total) {
+    if (lane == 0) { out[pg.row_start] = first; if (valid) valid[pg.row_start] = 1; }
+  }
+  for (uint32_t m = lane; m < n_mb; m += WAVE) {
+    const uint8_t* p = vals + mb_off[m];
+    int bw = mb_bw[m];
+    int64_t md = blk_md[m / (uint32_t)mpb];
+    uint64_t cnt = per_mini;
+    if ((uint64_t)(m + 1) * per_mini > n_deltas) cnt = n_deltas - (uint64_t)m * per_mini;
+    int64_t v = mb_sum[m];
+    uint64_t acc = 0; int nbits = 0;
+    uint64_t base = 1 + (uint64_t)m * per_mini;  // value index of first delta output
+    for (uint64_t i = 0; i < cnt; i++) {
+      uint64_t dv = 0;
+      if (bw) {
+        while (nbits < bw) { acc |= (uint64_t)(*p++) << nbits; nbits += 8; }
+        dv = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
+        acc >>= bw; nbits -= bw;
+      }
+      v += md + (int64_t)dv;
+      out[pg.row_start + base + i] = v;
+      if (valid) valid[pg.row_start + base + i] = 1;
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// PLAIN byte_array CONTAINS (LIKE '%needle%'): fused offsets walk +
+// substring scan, one wave per page. lane0 walks the length-prefixed
+// values into an LDS batch of offsets; all lanes then scan one value each.
+// ------------------------------------------------------------------
+#define CBATCH 1024
+__global__ void __launch_bounds__(WAVE)
+k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+                 const int32_t* __restrict__ ids, int n,
+                 const uint8_t* __restrict__ needle, int nlen,
+                 uint8_t* __restrict__ mask, int32_t* d_error) {
+  __shared__ uint32_t offs[CBATCH + 1];
+  __shared__ uint8_t nulls[CBATCH];
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const int lane = threadIdx.x;
+  const uint8_t* def_start; uint32_t def_len; bool all_valid;
+  const uint8_t* payload = dec + pg.dst_off;
+  const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
+  const uint32_t nv = pg.num_values, row0 = pg.row_start;
+
+  SerialRle def(def_start, def_start + (def_start ? def_len : 0), 1);
+  uint32_t walk = 0;  // byte offset within vals (maintained by lane0)
+  for (uint32_t b0 = 0; b0 < nv; b0 += CBATCH) {
+    uint32_t bn = min((uint32_t)CBATCH, nv - b0);
+    if (lane == 0) {
+      uint32_t w = walk;
+      for (uint32_t i = 0; i < bn; i++) {
+        int present = all_valid ? 1 : (int)def.next();
+        nulls[i] = (uint8_t)!present;
+        offs[i] = w;
+        if (present) {
+          uint32_t l;
+          memcpy(&l, vals + w, 4);
+          w += 4 + l;
+        }
+      }
+      offs[bn] = w;
+      walk = w;
+    }
+    __syncthreads();
+    for (uint32_t i = lane; i < bn; i += WAVE) {
+      uint8_t hit = 0;
+      if (!nulls[i]) {
+        uint32_t o = offs[i];
+        uint32_t vl;
+        memcpy(&vl, vals + o, 4);
+        const uint8_t* s = vals + o + 4;
+        if (nlen == 0) hit = 1;
+        else if (vl >= (uint32_t)nlen) {
+          uint8_t c0 = needle[0];
+          for (uint32_t j = 0; j + nlen <= vl; j++) {
+            if (s[j] == c0) {
+              uint32_t k = 1;
+              while (k < (uint32_t)nlen && s[j + k] == needle[k]) k++;
+              if (k == (uint32_t)nlen) { hit = 1; break; }
+            }
+          }
+        }
+      }
+      mask[row0 + b0 + i] &= hit;
+    }
+    __syncthreads();
+  }
+}
+
+// ------------------------------------------------------------------
+// row-parallel comparison on i64 arrays
+// ------------------------------------------------------------------
+__global__ void k_cmp_i64(const int64_t* __restrict__ col,
+                          const uint8_t* __restrict__ valid,
+                          int64_t lo, int64_t hi, int mode, int hi_exclusive,
+                          uint8_t* __restrict__ mask, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint8_t ok = valid ? valid[i] : 1;
+    if (ok) {
+      int64_t v = col[i];
+      switch (mode) {
+        case CMP_EQ: ok = (v == lo); break;
+        case CMP_NE: ok = (v != lo); break;
+        case CMP_LT: ok = (v < lo); break;
+        case CMP_LE: ok = (v <= lo); break;
+        case CMP_GT: ok = (v > lo); break;
+        case CMP_GE: ok = (v >= lo); break;
+        case CMP_RANGE: ok = (v >= lo) && (hi_exclusive ? (v < hi) : (v <= hi)); break;
+      }
+    }
+    mask[i] &= ok;
+  }
+}
+
+// ------------------------------------------------------------------
+// aggregation: grid-stride, per-block LDS table (when it fits) flushed
+// with global atomics.
+// ------------------------------------------------------------------
+__device__ inline void atomic_min_i64(uint64_t* addr, int64_t val) {
+  int64_t old = (int64_t)*addr;
+  while (val < old) {
+    uint64_t prev = atomicCAS((unsigned long long*)addr, (unsigned long long)old,
+                              (unsigned long long)val);
+    if ((int64_t)prev == old) break;
+    old = (int64_t)prev;
+  }
+}
+__device__ inline void atomic_max_i64(uint64_t* addr, int64_t val) {
+  int64_t old = (int64_t)*addr;
+  while (val > old) {
+    uint64_t prev = atomicCAS((unsigned long long*)addr, (unsigned long long)old,
+                              (unsigned long long)val);
+    if ((int64_t)prev == old) break;
+    old = (int64_t)prev;
+  }
+}
+__device__ inline void atomic_add_f64(uint64_t* addr, double val) {
+  uint64_t old = *addr;
+  for (;;) {
+    double cur = __longlong_as_double((long long)old);
+    uint64_t desired = (uint64_t)__double_as_longlong(cur + val);
+    uint64_t prev = atomicCAS((unsigned long long*)addr, (unsigned long long)old,
+                              (unsigned long long)desired);
+    if (prev == old) break;
+    old = prev;
+  }
+}
+// LDS variants (shared-memory atomics)
+__device__ inline void atomic_min_i64_s(uint64_t* addr, int64_t val) {
+  int64_t old = (int64_t)*addr;
+  while (val < old) {
+    uint64_t prev = atomicCAS((unsigned long long*)addr, (unsigned long long)old,
+                              (unsigned long long)val);
+    if ((int64_t)prev == old) break;
+    old = (int64_t)prev;
+  }
+}
+
+template <bool USE_LDS>
+__global__ void __launch_bounds__(256)
+k_agg(AggArgs a) {
+  extern __shared__ uint64_t lt[];
+  const int slots = 1 + 2 * a.n_aggs;
+  const int64_t tsz = (int64_t)a.n_groups * slots;
+  uint64_t* tab;
+  if (USE_LDS) {
+    tab = lt;
+    for (int64_t i = threadIdx.x; i < tsz; i += blockDim.x) {
+      int s = (int)(i % slots);
+      uint64_t init = 0;
+      if (s > 0 && ((s - 1) & 1) == 0) {  // value slot
+        int ai = (s - 1) / 2;
+        int k = a.agg_kind[ai];
+        if (k == AGGK_MIN_I64) init = (uint64_t)INT64_MAX;
+        else if (k == AGGK_MAX_I64) init = (uint64_t)INT64_MIN;
+      }
+      tab[i] = init;
+    }
+    __syncthreads();
+  } else {
+    tab = a.table;  // pre-initialized by k_init_table
+  }
+
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < a.n_rows; i += stride) {
+    if (a.mask && !a.mask[i]) continue;
+    int32_t g = 0;
+    for (int k = 0; k < a.n_keys; k++)
+      g = g * a.key_size[k] + a.key_gid[k][i];
+    uint64_t* row = tab + (int64_t)g * slots;
+    atomicAdd((unsigned long long*)&row[0], 1ull);
+    for (int ai = 0; ai < a.n_aggs; ai++) {
+      int k = a.agg_kind[ai];
+      if (k == AGGK_COUNT_STAR) {
+        atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+        continue;
+      }
+      if (a.agg_valid[ai] && !a.agg_valid[ai][i]) continue;
+      if (k == AGGK_COUNT) {  // validity only; no value array needed
+        atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+        continue;
+      }
+      int64_t v = a.agg_val[ai][i];
+      uint64_t* vs = &row[1 + 2 * ai];
+      switch (k) {
+        case AGGK_SUM_I64: atomicAdd((unsigned long long*)vs, (unsigned long long)v); break;
+        case AGGK_SUM_F64: atomic_add_f64(vs, __longlong_as_double((long long)v)); break;
+        case AGGK_MIN_I64: atomic_min_i64(vs, v); break;
+        case AGGK_MAX_I64: atomic_max_i64(vs, v); break;
+      }
+      atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+    }
+  }
+
+  if (USE_LDS) {
+    __syncthreads();
+    // flush block table to global
+    for (int64_t t = threadIdx.x; t < tsz; t += blockDim.x) {
+      int s = (int)(t % slots);
+      uint64_t v = tab[t];
+      uint64_t* g = &a.table[t];
+      if (s == 0) { if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v); continue; }
+      if (((s - 1) & 1) == 1) { if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v); continue; }
+      int ai = (s - 1) / 2;
+      switch (a.agg_kind[ai]) {
+        case AGGK_COUNT_STAR: case AGGK_COUNT:
+          if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v); break;
+        case AGGK_SUM_I64:
+          if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v); break;
+        case AGGK_SUM_F64: {
+          double d = __longlong_as_double((long long)v);
+          if (d != 0.0) atomic_add_f64(g, d);
+          break;
+        }
+        case AGGK_MIN_I64:
+          if ((int64_t)v != INT64_MAX) atomic_min_i64(g, (int64_t)v); break;
+        case AGGK_MAX_I64:
+          if ((int64_t)v != INT64_MIN) atomic_max_i64(g, (int64_t)v); break;
+      }
+    }
+  }
+}
+
+__global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
+                             const int32_t* agg_kind) {
+  int slots = 1 + 2 * n_aggs;
+  int64_t tsz = (int64_t)n_groups * slots;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < tsz; i += stride) {
+    int s = (int)(i % slots);
+    uint64_t init = 0;
+    if (s > 0 && ((s - 1) & 1) == 0) {
+      int k = agg_kind[(s - 1) / 2];
+      if (k == AGGK_MIN_I64) init = (uint64_t)INT64_MAX;
+      else if (k == AGGK_MAX_I64) init = (uint64_t)INT64_MIN;
+    }
+    table[i] = init;
+  }
+}
+
+// ------------------------------------------------------------------
+// host-side launchers (called from gpuq.cpp, same TU set)
+// ------------------------------------------------------------------
+void launch_lz4(hipStream_t st, const uint8_t* raw, uint8_t* dec,
+                const DevPage* pages, const int32_t* ids, int n, int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_lz4_pages, dim3(n), dim3(WAVE), 0, st, raw, dec, pages, ids, n, d_err);
+}
+void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                     const int32_t* ids, int n, const int32_t* remap_pool,
+                     int32_t* out, uint8_t* valid, int32_t* d_err) {
+  if (!n) return;
+  EmitGidP e{}; e.pool = remap_pool; e.out = out; e.valid = valid;
+  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, d_err);
+}
+void launch_dict_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                     const int32_t* ids, int n, const int64_t* dictv_pool,
+                     int64_t* out, uint8_t* valid, int32_t* d_err) {
+  if (!n) return;
+  EmitDictI64P e{}; e.pool = dictv_pool; e.out = out; e.valid = valid;
+  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, d_err);
+}
+void launch_dict_mask(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                      const int32_t* ids, int n, const uint8_t* lut_pool,
+                      uint8_t* mask, int32_t* d_err) {
+  if (!n) return;
+  EmitDictMaskP e{}; e.pool = lut_pool; e.mask = mask;
+  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, d_err);
+}
+void launch_plain_fixed(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                        const int32_t* ids, int n, int64_t* out, uint8_t* valid,
+                        int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_plain_fixed, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, d_err);
+}
+void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                      const int32_t* ids, int n, int64_t* out, uint8_t* valid,
+                      int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, d_err);
+}
+void launch_bytes_contains(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                           const int32_t* ids, int n, const uint8_t* needle, int nlen,
+                           uint8_t* mask, int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_bytes_contains, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, needle, nlen, mask, d_err);
+}
+void launch_cmp_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
+                    int64_t lo, int64_t hi, int mode, int hi_excl,
+                    uint8_t* mask, int64_t n) {
+  int blocks = (int)((n + 255) / 256);
+  if (blocks > 4096) blocks = 4096;
+  if (n) hipLaunchKernelGGL(k_cmp_i64, dim3(blocks), dim3(256), 0, st, col, valid, lo, hi, mode, hi_excl, mask, n);
+}
+void launch_init_table(hipStream_t st, uint64_t* table, int32_t n_groups,
+                       int n_aggs, const int32_t* d_agg_kind) {
+  int64_t tsz = (int64_t)n_groups * (1 + 2 * n_aggs);
+  int blocks = (int)((tsz + 255) / 256);
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(k_init_table, dim3(blocks), dim3(256), 0, st, table, n_groups, n_aggs, d_agg_kind);
+}
+void launch_agg(hipStream_t st, const AggArgs& a) {
+  size_t lds = (size_t)a.n_groups * (1 + 2 * a.n_aggs) * 8;
+  int blocks = (int)((a.n_rows + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  if (lds <= 32 * 1024) {
+    hipLaunchKernelGGL(k_agg<true>, dim3(blocks), dim3(256), lds, st, a);
+  } else {
+    hipLaunchKernelGGL(k_agg<false>, dim3(blocks), dim3(256), 0, st, a);
+  }
+}
+
+}  // namespace gpuq

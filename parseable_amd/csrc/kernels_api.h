@@ -1,0 +1,33 @@
+// Launch wrappers implemented in kernels.hip.
+#pragma once
+#include <hip/hip_runtime.h>
+#include "dev_types.h"
+
+namespace gpuq {
+void launch_lz4(hipStream_t, const uint8_t* raw, uint8_t* dec,
+                const DevPage*, const int32_t* ids, int n, int32_t* d_err);
+void launch_dict_gid(hipStream_t, const uint8_t* dec, const DevPage*,
+                     const int32_t* ids, int n, const int32_t* remap_pool,
+                     int32_t* out, uint8_t* valid, int32_t* d_err);
+void launch_dict_i64(hipStream_t, const uint8_t* dec, const DevPage*,
+                     const int32_t* ids, int n, const int64_t* dictv_pool,
+                     int64_t* out, uint8_t* valid, int32_t* d_err);
+void launch_dict_mask(hipStream_t, const uint8_t* dec, const DevPage*,
+                      const int32_t* ids, int n, const uint8_t* lut_pool,
+                      uint8_t* mask, int32_t* d_err);
+void launch_plain_fixed(hipStream_t, const uint8_t* dec, const DevPage*,
+                        const int32_t* ids, int n, int64_t* out, uint8_t* valid,
+                        int32_t* d_err);
+void launch_delta_i64(hipStream_t, const uint8_t* dec, const DevPage*,
+                      const int32_t* ids, int n, int64_t* out, uint8_t* valid,
+                      int32_t* d_err);
+void launch_bytes_contains(hipStream_t, const uint8_t* dec, const DevPage*,
+                           const int32_t* ids, int n, const uint8_t* needle,
+                           int nlen, uint8_t* mask, int32_t* d_err);
+void launch_cmp_i64(hipStream_t, const int64_t* col, const uint8_t* valid,
+                    int64_t lo, int64_t hi, int mode, int hi_excl,
+                    uint8_t* mask, int64_t n);
+void launch_init_table(hipStream_t, uint64_t* table, int32_t n_groups,
+                       int n_aggs, const int32_t* d_agg_kind);
+void launch_agg(hipStream_t, const AggArgs&);
+}  // namespace gpuq

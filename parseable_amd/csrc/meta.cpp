@@ -1,0 +1,280 @@
+#include "meta.h"
+#include <cstring>
+
+namespace gpuq {
+
+// ---- thrift compact protocol ----------------------------------------
+namespace {
+
+struct Reader {
+  const uint8_t *p, *end;
+  uint8_t u8() {
+    if (p >= end) throw std::runtime_error("thrift: eof");
+    return *p++;
+  }
+  uint64_t varint() {
+    uint64_t v = 0; int sh = 0;
+    for (;;) {
+      uint8_t b = u8();
+      v |= (uint64_t)(b & 0x7f) << sh;
+      if (!(b & 0x80)) return v;
+      sh += 7;
+    }
+  }
+  int64_t zigzag() { uint64_t v = varint(); return (int64_t)(v >> 1) ^ -(int64_t)(v & 1); }
+  // returns wire type, 0 = stop
+  int field(int16_t& fid) {
+    uint8_t b = u8();
+    if (!b) return 0;
+    int delta = b >> 4, t = b & 0xf;
+    if (delta) fid = (int16_t)(fid + delta); else fid = (int16_t)zigzag();
+    return t;
+  }
+  void list_head(int& etype, uint32_t& n) {
+    uint8_t h = u8(); etype = h & 0xf; n = h >> 4;
+    if (n == 15) n = (uint32_t)varint();
+  }
+  std::string binary() {
+    uint64_t n = varint();
+    if (p + n > end) throw std::runtime_error("thrift: binary overrun");
+    std::string s((const char*)p, n); p += n; return s;
+  }
+  const uint8_t* binary_view(uint32_t& n) {
+    n = (uint32_t)varint();
+    if (p + n > end) throw std::runtime_error("thrift: binary overrun");
+    const uint8_t* s = p; p += n; return s;
+  }
+  void skip(int t) {
+    switch (t) {
+      case 1: case 2: break;
+      case 3: u8(); break;
+      case 4: case 5: case 6: zigzag(); break;
+      case 7: p += 8; break;
+      case 8: { uint64_t n = varint(); p += n; break; }
+      case 9: case 10: {
+        int et; uint32_t n; list_head(et, n);
+        for (uint32_t i = 0; i < n; i++) skip(et);
+        break;
+      }
+      case 12: {
+        int16_t fid = 0;
+        for (;;) { int ft = field(fid); if (!ft) break; skip(ft); }
+        break;
+      }
+      default: throw std::runtime_error("thrift: bad wire type");
+    }
+    if (p > end) throw std::runtime_error("thrift: overrun");
+  }
+};
+
+void parse_stats(Reader& r, ColumnChunkMeta& cm, int phys_type) {
+  // parquet Statistics struct: 5=max_value, 6=min_value (new);
+  // 1=max, 2=min (deprecated) — all binary, little-endian for numerics.
+  int16_t fid = 0;
+  std::string mn, mx;
+  for (;;) {
+    int t = r.field(fid);
+    if (!t) break;
+    if ((fid == 5 || fid == 1) && t == 8) mx = r.binary();
+    else if ((fid == 6 || fid == 2) && t == 8) mn = r.binary();
+    else r.skip(t);
+  }
+  if ((phys_type == PT_INT64) && mn.size() == 8 && mx.size() == 8) {
+    memcpy(&cm.stat_min, mn.data(), 8);
+    memcpy(&cm.stat_max, mx.data(), 8);
+    cm.has_i64_stats = true;
+  }
+}
+
+void parse_column_meta(Reader& r, ColumnChunkMeta& cm, const FileMeta& fm, int& phys) {
+  int16_t fid = 0;
+  std::string name;
+  for (;;) {
+    int t = r.field(fid);
+    if (!t) break;
+    switch (fid) {
+      case 1: phys = (int)r.zigzag(); break;
+      case 3: {  // path_in_schema (flat: one element)
+        int et; uint32_t n; r.list_head(et, n);
+        for (uint32_t i = 0; i < n; i++) {
+          std::string s = r.binary();
+          if (i == 0) name = s;
+        }
+        break;
+      }
+      case 4: cm.codec = (int)r.zigzag(); break;
+      case 5: cm.num_values = r.zigzag(); break;
+      case 7: cm.total_compressed_size = r.zigzag(); break;
+      case 9: cm.data_page_offset = r.zigzag(); break;
+      case 11: cm.dict_page_offset = r.zigzag(); break;
+      case 12: parse_stats(r, cm, phys); break;
+      default: r.skip(t);
+    }
+  }
+  for (size_t i = 0; i < fm.columns.size(); i++)
+    if (fm.columns[i].name == name) { cm.schema_idx = (int)i; break; }
+}
+
+}  // namespace
+
+FileMeta parse_footer(const uint8_t* buf, size_t len) {
+  if (len < 12 || memcmp(buf + len - 4, "PAR1", 4))
+    throw std::runtime_error("not a parquet file");
+  uint32_t flen;
+  memcpy(&flen, buf + len - 8, 4);
+  if ((size_t)flen + 8 > len) throw std::runtime_error("bad footer length");
+  Reader r{buf + len - 8 - flen, buf + len - 8};
+  FileMeta fm;
+  int16_t fid = 0;
+  for (;;) {
+    int t = r.field(fid);
+    if (!t) break;
+    switch (fid) {
+      case 2: {  // schema: list<SchemaElement>
+        int et; uint32_t n; r.list_head(et, n);
+        for (uint32_t i = 0; i < n; i++) {
+          int16_t f2 = 0;
+          SchemaColumn sc;
+          int rep = 0;
+          for (;;) {
+            int t2 = r.field(f2);
+            if (!t2) break;
+            switch (f2) {
+              case 1: sc.phys_type = (int)r.zigzag(); break;
+              case 3: rep = (int)r.zigzag(); break;
+              case 4: sc.name = r.binary(); break;
+              default: r.skip(t2);
+            }
+          }
+          sc.optional = (rep == 1);
+          if (i > 0) fm.columns.push_back(sc);  // element 0 = root group
+        }
+        break;
+      }
+      case 3: fm.num_rows = r.zigzag(); break;
+      case 4: {  // row_groups
+        int et; uint32_t n; r.list_head(et, n);
+        for (uint32_t g = 0; g < n; g++) {
+          RowGroupMeta rg;
+          rg.chunks.resize(fm.columns.size());
+          int16_t f2 = 0;
+          for (;;) {
+            int t2 = r.field(f2);
+            if (!t2) break;
+            if (f2 == 1) {  // columns: list<ColumnChunk>
+              int et2; uint32_t nc; r.list_head(et2, nc);
+              for (uint32_t c = 0; c < nc; c++) {
+                ColumnChunkMeta cm;
+                int phys = -1;
+                int16_t f3 = 0;
+                for (;;) {
+                  int t3 = r.field(f3);
+                  if (!t3) break;
+                  if (f3 == 3) parse_column_meta(r, cm, fm, phys);
+                  else r.skip(t3);
+                }
+                if (cm.schema_idx >= 0 && cm.schema_idx < (int)fm.columns.size()) {
+                  rg.chunks[cm.schema_idx] = cm;
+                  rg.total_compressed_size += cm.total_compressed_size;
+                }
+              }
+            } else if (f2 == 2) rg.total_byte_size = r.zigzag();
+            else if (f2 == 3) rg.num_rows = r.zigzag();
+            else r.skip(t2);
+          }
+          fm.row_groups.push_back(std::move(rg));
+        }
+        break;
+      }
+      default: r.skip(t);
+    }
+  }
+  return fm;
+}
+
+std::vector<PageInfo> walk_pages(const uint8_t* buf, const ColumnChunkMeta& cm,
+                                 int64_t rg_rows) {
+  std::vector<PageInfo> out;
+  const uint8_t* p = buf + cm.start_offset();
+  const uint8_t* chunk_end = buf + cm.start_offset() + cm.total_compressed_size;
+  int64_t rows = 0;
+  while (rows < rg_rows && p < chunk_end) {
+    Reader r{p, chunk_end};
+    PageInfo pi{};
+    int16_t fid = 0;
+    for (;;) {
+      int t = r.field(fid);
+      if (!t) break;
+      switch (fid) {
+        case 1: pi.type = (int)r.zigzag(); break;
+        case 2: pi.uncomp_size = (int32_t)r.zigzag(); break;
+        case 3: pi.comp_size = (int32_t)r.zigzag(); break;
+        case 5: {  // DataPageHeader
+          int16_t f2 = 0;
+          for (;;) {
+            int t2 = r.field(f2);
+            if (!t2) break;
+            if (f2 == 1) pi.num_values = (int32_t)r.zigzag();
+            else if (f2 == 2) pi.encoding = (int)r.zigzag();
+            else r.skip(t2);
+          }
+          break;
+        }
+        case 7: {  // DictionaryPageHeader
+          int16_t f2 = 0;
+          for (;;) {
+            int t2 = r.field(f2);
+            if (!t2) break;
+            if (f2 == 1) pi.num_values = (int32_t)r.zigzag();
+            else if (f2 == 2) pi.encoding = (int)r.zigzag();
+            else r.skip(t2);
+          }
+          break;
+        }
+        case 8:
+          throw std::runtime_error("data page v2 not produced by this writer");
+        default: r.skip(t);
+      }
+    }
+    pi.payload_off = r.p - buf;
+    out.push_back(pi);
+    p = r.p + pi.comp_size;
+    if (pi.type == PAGE_DATA) rows += pi.num_values;
+  }
+  if (rows != rg_rows)
+    throw std::runtime_error("page walk row-count mismatch");
+  return out;
+}
+
+int lz4_decompress_host(const uint8_t* src, size_t src_len,
+                        uint8_t* dst, size_t dst_cap) {
+  const uint8_t *sp = src, *send = src + src_len;
+  uint8_t *dp = dst, *dend = dst + dst_cap;
+  while (sp < send) {
+    uint8_t token = *sp++;
+    size_t lit = token >> 4;
+    if (lit == 15) {
+      uint8_t b;
+      do { if (sp >= send) return -1; b = *sp++; lit += b; } while (b == 255);
+    }
+    if (sp + lit > send || dp + lit > dend) return -1;
+    memcpy(dp, sp, lit); sp += lit; dp += lit;
+    if (sp >= send) break;  // last sequence carries only literals
+    if (sp + 2 > send) return -1;
+    size_t off = sp[0] | ((size_t)sp[1] << 8); sp += 2;
+    if (off == 0 || (size_t)(dp - dst) < off) return -1;
+    size_t mlen = token & 0xf;
+    if (mlen == 15) {
+      uint8_t b;
+      do { if (sp >= send) return -1; b = *sp++; mlen += b; } while (b == 255);
+    }
+    mlen += 4;
+    if (dp + mlen > dend) return -1;
+    const uint8_t* mp = dp - off;
+    for (size_t i = 0; i < mlen; i++) dp[i] = mp[i];  // overlap repeats
+    dp += mlen;
+  }
+  return (int)(dp - dst);
+}
+
+}  // namespace gpuq

@@ -1,0 +1,153 @@
+"""Cross-GPU Final merge of partial aggregate tables over torch.distributed
+(RCCL on ROCm — backend "nccl" — over xGMI; gloo on CPU for tests).
+
+Design per SURVEY.md §8e: row-group shards are embarrassingly parallel; the
+ONLY data-path exchange is this merge of fixed-width aggregate tables —
+KB..MB payloads, latency-bound on xGMI. Key spaces can differ per rank (each
+rank's plan builds its global dictionary from its own shard's dict pages), so
+a one-time all_gather_object agrees on the union key space (setup, untimed);
+every step then reduces dense tensors: SUM for presence/counts/sums, MIN/MAX
+for min/max — mirroring the reference's AggregateExec Partial->Final split
+(SURVEY.md §3a step 7)."""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+I64_MAX = 2**63 - 1
+I64_MIN = -(2**63)
+
+
+def _batch_rows(batch, n_keys, n_aggs):
+    """partial batch (C-ABI schema: keys..., __presence, agg{i}, agg{i}_count)
+    -> {key_tuple: (presence, [(val, cnt), ...])}"""
+    out = {}
+    if batch is None or batch.num_rows == 0:
+        return out
+    cols = [batch.column(i).to_pylist() for i in range(batch.num_columns)]
+    for r in range(batch.num_rows):
+        key = tuple(cols[k][r] for k in range(n_keys))
+        presence = cols[n_keys][r]
+        aggs = []
+        for i in range(n_aggs):
+            aggs.append((cols[n_keys + 1 + 2 * i][r], cols[n_keys + 2 + 2 * i][r]))
+        out[key] = (presence, aggs)
+    return out
+
+
+class DistMerger:
+    """Final-merge helper. setup() once (untimed; agrees the key space),
+    then step(batch) per timed iteration (dense all_reduce merge)."""
+
+    def __init__(self, query: dict, device: str = "cpu"):
+        self.query = query
+        self.group_by = query.get("group_by", [])
+        self.aggs = query["select"]
+        self.device = device
+        self.keyspace: list[tuple] | None = None
+        self.key_index: dict[tuple, int] = {}
+        self.is_f64 = [False] * len(self.aggs)
+
+    @property
+    def world(self):
+        return dist.get_world_size() if dist.is_initialized() else 1
+
+    def setup(self, batch):
+        import pyarrow as pa
+
+        if batch is not None:
+            nk = len(self.group_by)
+            for i in range(len(self.aggs)):
+                self.is_f64[i] = pa.types.is_floating(batch.schema.field(nk + 1 + 2 * i).type)
+        local = sorted(_batch_rows(batch, len(self.group_by), len(self.aggs)).keys(),
+                       key=lambda k: tuple((v is None, v or "") for v in k))
+        if self.world > 1:
+            gathered: list = [None] * self.world
+            dist.all_gather_object(gathered, local)
+            allk = set()
+            for g in gathered:
+                allk.update(g)
+        else:
+            allk = set(local)
+        self.keyspace = sorted(allk, key=lambda k: tuple((v is None, v or "") for v in k))
+        self.key_index = {k: i for i, k in enumerate(self.keyspace)}
+
+    def step(self, batch) -> list:
+        """Merge this rank's partial with every other rank's; returns the
+        final rows (same normalized form as the oracle)."""
+        assert self.keyspace is not None, "call setup() first"
+        G = max(len(self.keyspace), 1)
+        n_aggs = len(self.aggs)
+        rows = _batch_rows(batch, len(self.group_by), n_aggs)
+
+        dev = self.device
+        presence = torch.zeros(G, dtype=torch.int64, device=dev)
+        counts = torch.zeros(G, n_aggs, dtype=torch.int64, device=dev)
+        sums_i = torch.zeros(G, n_aggs, dtype=torch.int64, device=dev)
+        sums_f = torch.zeros(G, n_aggs, dtype=torch.float64, device=dev)
+        mins = torch.full((G, n_aggs), I64_MAX, dtype=torch.int64, device=dev)
+        maxs = torch.full((G, n_aggs), I64_MIN, dtype=torch.int64, device=dev)
+
+        for key, (p, aggvals) in rows.items():
+            gi = self.key_index.get(key)
+            if gi is None:
+                # a key outside the agreed space would mean setup() raced a
+                # plan change — fail loudly rather than drop rows
+                raise RuntimeError(f"key {key} not in agreed key space")
+            presence[gi] += p
+            for i, a in enumerate(self.aggs):
+                v, c = aggvals[i]
+                if a["agg"] in ("count_star", "count"):
+                    counts[gi, i] += v
+                    continue
+                if c == 0 or v is None:
+                    continue
+                counts[gi, i] += c
+                if a["agg"] == "sum":
+                    if self.is_f64[i]:
+                        sums_f[gi, i] += v
+                    else:
+                        sums_i[gi, i] += v
+                elif a["agg"] == "min":
+                    mins[gi, i] = min(mins[gi, i].item(), v)
+                elif a["agg"] == "max":
+                    maxs[gi, i] = max(maxs[gi, i].item(), v)
+
+        if self.world > 1:
+            dist.all_reduce(presence, op=dist.ReduceOp.SUM)
+            dist.all_reduce(counts, op=dist.ReduceOp.SUM)
+            if any(a["agg"] == "sum" for a in self.aggs):
+                dist.all_reduce(sums_i, op=dist.ReduceOp.SUM)
+                dist.all_reduce(sums_f, op=dist.ReduceOp.SUM)
+            if any(a["agg"] == "min" for a in self.aggs):
+                dist.all_reduce(mins, op=dist.ReduceOp.MIN)
+            if any(a["agg"] == "max" for a in self.aggs):
+                dist.all_reduce(maxs, op=dist.ReduceOp.MAX)
+
+        presence = presence.cpu()
+        counts = counts.cpu()
+        sums_i, sums_f = sums_i.cpu(), sums_f.cpu()
+        mins, maxs = mins.cpu(), maxs.cpu()
+        out = []
+        for gi, key in enumerate(self.keyspace):
+            if self.group_by and presence[gi].item() == 0:
+                continue
+            row = list(key)
+            for i, a in enumerate(self.aggs):
+                c = counts[gi, i].item()
+                if a["agg"] in ("count_star", "count"):
+                    row.append(c)
+                elif c == 0:
+                    row.append(None)
+                elif a["agg"] == "sum":
+                    row.append(sums_f[gi, i].item() if self.is_f64[i]
+                               else sums_i[gi, i].item())
+                elif a["agg"] == "min":
+                    row.append(mins[gi, i].item())
+                elif a["agg"] == "max":
+                    row.append(maxs[gi, i].item())
+            out.append(row)
+        if not self.group_by and not out:
+            out = [[0 if a["agg"] in ("count_star", "count") else None for a in self.aggs]]
+        return out

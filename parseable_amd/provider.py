@@ -1,0 +1,430 @@
+"""The drop-in host surface of the MI355X query path, mirroring the
+reference's TableProvider/ExecutionPlan shape:
+
+  StandardTableProvider.scan(query)  ->  GpuExecutionPlan
+      mirrors StandardTableProvider::scan (stream_schema_provider.rs:616-753):
+      snapshot -> manifest selection by time (snapshot.rs:42-71) -> file-level
+      min/max pruning (can_be_pruned/satisfy_constraints,
+      stream_schema_provider.rs:1049-1137) -> count fast path
+      (query.rs:189-256, query/mod.rs:537-590) -> gpuq_plan_build.
+
+  GpuExecutionPlan.execute(partition) -> pyarrow.RecordBatch (partial agg)
+      mirrors ExecutionPlan::execute's pull-based per-partition streams
+      (query/mod.rs:341-363); merge_partials() is the AggregateExec(Final).
+
+  Query.execute() is the single execution funnel (query/mod.rs:152-165).
+
+Time-range injection (`p_timestamp >= start AND < end`) mirrors `transform`
+(query/mod.rs:829-888). All per-row compute runs in libgpuq.so on the GPU;
+this layer only handles metadata (manifest JSON, footers are parsed in C++).
+"""
+
+from __future__ import annotations
+
+import ctypes as C
+import json
+import os
+
+from . import _lib
+from ._lib import AGGS, OPS, GpuqAgg, GpuqFile, GpuqPred, GpuqMetrics
+
+
+class GpuqError(RuntimeError):
+    pass
+
+
+class GpuSession:
+    def __init__(self, device_mask: int = 0):
+        lib = _lib.load()
+        n = lib.gpuq_device_count()
+        if n <= 0:
+            raise GpuqError(
+                "no HIP devices visible — the gpuq path requires an MI355X; "
+                "it never falls back to CPU"
+            )
+        self._lib = lib
+        self._ctx = lib.gpuq_session_create(C.c_uint64(device_mask))
+        if not self._ctx:
+            raise GpuqError("gpuq_session_create failed")
+
+    def _err(self):
+        return self._lib.gpuq_last_error(self._ctx).decode()
+
+    def close(self):
+        if self._ctx:
+            self._lib.gpuq_session_destroy(self._ctx)
+            self._ctx = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def _stats_entry(col):
+    if not col or col.get("stats") is None:
+        return None
+    return col["stats"]
+
+
+def _file_pruned(file_entry, preds):
+    """Port of ManifestExt::can_be_pruned (stream_schema_provider.rs:1049-1078):
+    a file is pruned when some predicate provably matches no row. BETWEEN is
+    decomposed into ge(lo) + le/lt(hi) bounds as the reference's injected
+    filters are (query/mod.rs:829-888)."""
+    cols = {c["name"]: c for c in file_entry.get("columns", [])}
+
+    def check(col, op, value):
+        st = _stats_entry(cols.get(col))
+        if st is None:
+            return False  # no stats -> cannot prune
+        (kind, mm), = st.items()
+        mn, mx = mm["min"], mm["max"]
+        if kind == "Int":
+            if not isinstance(value, int) or isinstance(value, bool):
+                return False
+        elif kind == "Float":
+            if not isinstance(value, float):
+                return False
+        elif kind == "String":
+            if not isinstance(value, str):
+                return False
+        else:
+            return False
+        if op == "eq":
+            sat = mn <= value <= mx
+        elif op == "lt":
+            sat = mn < value
+        elif op == "le":
+            sat = mn <= value
+        elif op == "gt":
+            sat = mx > value
+        elif op == "ge":
+            sat = mx >= value
+        else:
+            return False  # ne/contains: never pruned (reference does the same)
+        return not sat
+
+    for p in preds:
+        op = p["op"]
+        col = p["col"]
+        if op == "between":
+            if check(col, "ge", p["lo"]) or check(col, "le" if not p.get("hi_exclusive") else "lt", p["hi"]):
+                return True
+        elif op == "contains" or op == "ne":
+            continue
+        else:
+            if check(col, op, p["lit"]):
+                return True
+    return False
+
+
+def _parse_iso_ms(s):
+    from datetime import datetime, timezone
+
+    return int(
+        datetime.strptime(s, "%Y-%m-%dT%H:%M:%S.%fZ")
+        .replace(tzinfo=timezone.utc)
+        .timestamp()
+        * 1000
+    )
+
+
+class ManifestCountResult:
+    """Count fast path: bare `SELECT count(*)` with no value filters is
+    answered from manifest num_rows sums and never reaches the scan engine
+    (query.rs:189-256, query/mod.rs:537-590)."""
+
+    def __init__(self, count):
+        self.count = count
+
+    def rows(self):
+        return [[self.count]]
+
+
+class StandardTableProvider:
+    def __init__(self, stream_dir: str, session: GpuSession | None = None):
+        self.stream_dir = stream_dir
+        self.session = session
+        snap_path = os.path.join(stream_dir, "stream.json")
+        with open(snap_path) as fh:
+            self.stream_json = json.load(fh)
+        self.snapshot = self.stream_json["snapshot"]
+
+    # Snapshot::manifests (catalog/snapshot.rs:42-71): retain manifests whose
+    # [lower,upper] overlaps the time predicates.
+    def _select_manifests(self, time_range):
+        items = self.snapshot["manifest_list"]
+        if time_range:
+            lo, hi = time_range
+            items = [
+                it
+                for it in items
+                if _parse_iso_ms(it["time_upper_bound"]) >= lo
+                and _parse_iso_ms(it["time_lower_bound"]) < hi
+            ]
+        return items
+
+    def _manifest_files(self, time_range):
+        files = []
+        for it in self._select_manifests(time_range):
+            mpath = it["manifest_path"]
+            if not os.path.isabs(mpath):
+                mpath = os.path.join(os.path.dirname(self.stream_dir), mpath)
+            with open(mpath) as fh:
+                m = json.load(fh)
+            files.extend(m["files"])
+        return files
+
+    def scan(self, query: dict):
+        """query: the IR of tests/golden_queries.py (select/group_by/preds/
+        time_range). Returns GpuExecutionPlan or ManifestCountResult."""
+        preds = list(query.get("preds", []))
+        time_range = query.get("time_range")
+        files = self._manifest_files(time_range)
+
+        # prune: time bounds as predicates on p_timestamp + value predicates
+        prune_preds = preds.copy()
+        if time_range:
+            prune_preds.append(
+                {
+                    "col": "p_timestamp",
+                    "op": "between",
+                    "lo": time_range[0],
+                    "hi": time_range[1],
+                    "hi_exclusive": True,
+                }
+            )
+        kept = [f for f in files if not _file_pruned(f, prune_preds)]
+
+        # count fast path (exact only when no value preds and the time range
+        # either is absent or fully covers every kept file's ts bounds)
+        sel = query["select"]
+        if (
+            not preds
+            and not query.get("group_by")
+            and len(sel) == 1
+            and sel[0]["agg"] == "count_star"
+        ):
+            exact = True
+            if time_range:
+                for fe in kept:
+                    st = _stats_entry(
+                        {c["name"]: c for c in fe["columns"]}.get("p_timestamp")
+                    )
+                    if not st or "Int" not in st:
+                        exact = False
+                        break
+                    mm = st["Int"]
+                    if not (time_range[0] <= mm["min"] and mm["max"] < time_range[1]):
+                        exact = False
+                        break
+            if exact:
+                return ManifestCountResult(sum(f["num_rows"] for f in kept))
+
+        paths = []
+        root = os.path.dirname(self.stream_dir)
+        for fe in kept:
+            p = fe["file_path"]
+            paths.append(p if os.path.isabs(p) else os.path.join(root, p))
+        return GpuExecutionPlan(self.session, paths, query)
+
+
+class GpuExecutionPlan:
+    def __init__(self, session: GpuSession, paths: list[str], query: dict):
+        if session is None:
+            raise GpuqError("a GpuSession (GPU) is required for scan execution")
+        self.session = session
+        self.query = query
+        lib = session._lib
+        self._lib = lib
+
+        n = len(paths)
+        self._path_bytes = [p.encode() for p in paths]
+        files = (GpuqFile * max(n, 1))()
+        for i, pb in enumerate(self._path_bytes):
+            files[i].path = pb
+            files[i].row_groups = None
+            files[i].n_row_groups = -1
+
+        preds = list(query.get("preds", []))
+        tr = query.get("time_range")
+        if tr:
+            preds.append(
+                {
+                    "col": "p_timestamp",
+                    "op": "between",
+                    "lo": tr[0],
+                    "hi": tr[1],
+                    "hi_exclusive": True,
+                }
+            )
+        self._pred_strs = []
+        cpreds = (GpuqPred * max(len(preds), 1))()
+        for i, p in enumerate(preds):
+            cpreds[i].column = p["col"].encode()
+            cpreds[i].op = OPS[p["op"]]
+            cpreds[i].hi_exclusive = 1 if p.get("hi_exclusive") else 0
+            if p["op"] == "between":
+                cpreds[i].lit_kind = 0
+                cpreds[i].i64[0] = p["lo"]
+                cpreds[i].i64[1] = p["hi"]
+            else:
+                lit = p["lit"]
+                if isinstance(lit, str):
+                    cpreds[i].lit_kind = 2
+                    b = lit.encode()
+                    self._pred_strs.append(b)
+                    cpreds[i].str = b
+                elif isinstance(lit, float):
+                    cpreds[i].lit_kind = 1
+                    cpreds[i].f64[0] = lit
+                else:
+                    cpreds[i].lit_kind = 0
+                    cpreds[i].i64[0] = lit
+
+        group_by = query.get("group_by", [])
+        self._gb_bytes = [g.encode() for g in group_by]
+        cgroup = (C.c_char_p * max(len(group_by), 1))(*self._gb_bytes)
+
+        aggs = query["select"]
+        self._agg_bytes = []
+        caggs = (GpuqAgg * max(len(aggs), 1))()
+        for i, a in enumerate(aggs):
+            caggs[i].op = AGGS[a["agg"]]
+            if a.get("col"):
+                b = a["col"].encode()
+                self._agg_bytes.append(b)
+                caggs[i].column = b
+
+        self._plan = lib.gpuq_plan_build(
+            session._ctx,
+            files, n,
+            None, 0,
+            cpreds, len(preds),
+            cgroup, len(group_by),
+            caggs, len(aggs),
+            C.c_int64(-1),
+        )
+        if not self._plan:
+            raise GpuqError(f"plan_build failed: {session._err()}")
+
+    def partition_count(self) -> int:
+        return self._lib.gpuq_plan_partition_count(self._plan)
+
+    def load(self, partition=None):
+        parts = range(self.partition_count()) if partition is None else [partition]
+        for p in parts:
+            if self._lib.gpuq_plan_load(self._plan, p) != 0:
+                raise GpuqError(f"plan_load failed: {self.session._err()}")
+
+    def execute(self, partition: int):
+        """-> pyarrow.RecordBatch of the PARTIAL aggregate for this shard."""
+        import pyarrow as pa
+
+        # allocate an ArrowArrayStream struct (5 ptr fields + private)
+        buf = C.create_string_buffer(6 * C.sizeof(C.c_void_p))
+        rc = self._lib.gpuq_plan_execute(self._plan, partition, C.cast(buf, C.c_void_p))
+        if rc != 0:
+            raise GpuqError(f"plan_execute failed: {self.session._err()}")
+        reader = pa.RecordBatchReader._import_from_c(C.addressof(buf))
+        batches = list(reader)
+        return batches[0] if batches else None
+
+    def execute_all(self):
+        """Run every partition and apply the Final aggregation merge
+        (the reference's AggregateExec(Final))."""
+        batches = [self.execute(p) for p in range(self.partition_count())]
+        return merge_partials([b for b in batches if b is not None], self.query)
+
+    def metrics(self) -> dict:
+        m = GpuqMetrics()
+        if self._lib.gpuq_plan_metrics(self._plan, C.byref(m)) != 0:
+            raise GpuqError("plan_metrics failed")
+        return {f: getattr(m, f) for f, _ in m._fields_}
+
+    def close(self):
+        if getattr(self, "_plan", None):
+            self._lib.gpuq_plan_destroy(self._plan)
+            self._plan = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def merge_partials(batches, query):
+    """Final merge of partial aggregate tables (cross-partition and
+    cross-rank). Output rows match the oracle's normalized form:
+    [key..., agg...] sorted by key tuple, NULL keys last."""
+    group_by = query.get("group_by", [])
+    aggs = query["select"]
+    acc = {}
+    for b in batches:
+        if b is None or b.num_rows == 0:
+            continue
+        cols = {name: b.column(i).to_pylist() for i, name in enumerate(b.schema.names)}
+        n = b.num_rows
+        for r in range(n):
+            key = tuple(cols[g][r] for g in group_by)
+            presence = cols["__presence"][r]
+            st = acc.get(key)
+            if st is None:
+                st = {"presence": 0, "vals": [None] * len(aggs), "cnts": [0] * len(aggs)}
+                acc[key] = st
+            st["presence"] += presence
+            for i, a in enumerate(aggs):
+                v = cols[f"agg{i}"][r]
+                c = cols[f"agg{i}_count"][r]
+                if a["agg"] in ("count_star", "count"):
+                    st["cnts"][i] += v
+                    continue
+                if c == 0 or v is None:
+                    continue
+                st["cnts"][i] += c
+                if st["vals"][i] is None:
+                    st["vals"][i] = v
+                elif a["agg"] == "sum":
+                    st["vals"][i] += v
+                elif a["agg"] == "min":
+                    st["vals"][i] = min(st["vals"][i], v)
+                elif a["agg"] == "max":
+                    st["vals"][i] = max(st["vals"][i], v)
+
+    rows = []
+    for key, st in acc.items():
+        if st["presence"] == 0 and group_by:
+            continue
+        row = list(key)
+        for i, a in enumerate(aggs):
+            if a["agg"] in ("count_star", "count"):
+                row.append(st["cnts"][i])
+            else:
+                row.append(st["vals"][i] if st["cnts"][i] > 0 else None)
+        rows.append(row)
+    rows.sort(key=lambda r: tuple(((1, "") if v is None else (0, v)) for v in r[: len(group_by)]))
+    if not group_by and not rows:
+        rows = [[0 if a["agg"] in ("count_star", "count") else None for a in aggs]]
+    return rows
+
+
+class Query:
+    """The single execution funnel, mirroring Query::execute
+    (query/mod.rs:152-165,291-372)."""
+
+    def __init__(self, provider: StandardTableProvider):
+        self.provider = provider
+
+    def execute(self, query: dict):
+        plan = self.provider.scan(query)
+        if isinstance(plan, ManifestCountResult):
+            return plan.rows(), None
+        try:
+            rows = plan.execute_all()
+            metrics = plan.metrics()
+            return rows, metrics
+        finally:
+            plan.close()
