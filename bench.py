@@ -1,0 +1,259 @@
+#!/usr/bin/env python3
+"""Benchmark for the MI355X query path (BASELINE.json metric: rows/sec +
+GB/s scanned for SELECT...WHERE...GROUP BY over a Parseable log stream).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`, launched
+for N>1 via torch.distributed.run with one rank per GPU (RCCL). A "step" is
+one pass of the hot path (LZ4_RAW decompress -> decode -> filter -> hash
+group-by -> cross-rank RCCL merge) over this rank's shard, with raw column
+chunks already resident in HBM when the timed region starts (plan.load() is
+untimed; the PCIe-inclusive rate is reported as load_gbps and discussed in
+DESIGN.md). Weak scaling: each rank owns its own shard of equal size.
+
+Default workload: BASELINE.json configs[1] ("c1": 100M rows x 8 cols,
+SELECT level,count(*) GROUP BY level) — the quoted single-GPU config.
+`--workload c2s|c3s` run the c2/c3-shaped queries at --rows scale for
+profiling (not the driver's line).
+
+Rank 0 prints ONE JSON line."""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+HBM_PEAK_GBS = 8000.0  # 8 TB/s spec (MI355X_MICROARCH.md)
+
+
+WORKLOADS = {
+    # name -> (datagen config, query builder)
+    "c1": ("c1", lambda a: {
+        "select": [{"agg": "count_star"}],
+        "group_by": ["level"],
+    }),
+    "c2s": ("c1", lambda a: {   # c2-shaped at reduced scale, single node
+        "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+        "group_by": ["host"],
+        "preds": [{"col": "p_timestamp", "op": "between",
+                   "lo": 1756684800000, "hi": 1756684800000 + a.between_ms}],
+    }),
+    "c3s": ("c3", lambda a: {   # c3-shaped LIKE byte scan
+        "select": [{"agg": "count_star"}],
+        "preds": [{"col": "message", "op": "contains", "lit": "error"}],
+    }),
+}
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--workload", default="c1")
+    ap.add_argument("--rows", type=int, default=100_000_000,
+                    help="rows per GPU (weak scaling)")
+    ap.add_argument("--between-ms", type=int, default=150 * 60_000,
+                    help="c2s: BETWEEN window width (~50%% of range)")
+    ap.add_argument("--data-dir", default=os.environ.get("GPUQ_DATA", "/tmp/gpuq_bench"))
+    ap.add_argument("--gen-workers", type=int, default=0)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+        merge_dev = f"cuda:{local_rank}"
+    else:
+        merge_dev = "cpu"
+
+    from datagen.gen import gen_stream
+    from parseable_amd import GpuSession, StandardTableProvider
+    from parseable_amd.dist import DistMerger
+
+    cfg, qbuild = WORKLOADS[args.workload]
+    query = qbuild(args)
+
+    # --- per-rank shard (generated once, cached in data-dir) ---
+    n_files = (args.rows + 262_143) // 262_144
+    shard = os.path.join(args.data_dir, f"{args.workload}_{args.rows}_r{rank}")
+    stream_dir = os.path.join(shard, "stream")
+    workers = args.gen_workers or max(1, (os.cpu_count() or 8) // max(1, min(world, 8)))
+    if not os.path.exists(os.path.join(stream_dir, "stream.json")):
+        log(f"[bench] generating {args.rows} rows/rank ({n_files} files, "
+            f"{workers} workers) under {shard} ...")
+        t0 = time.time()
+        gen_stream(shard, "stream", cfg, rows=args.rows, seed=42,
+                   workers=workers, minute_offset=rank * n_files)
+        log(f"[bench] datagen took {time.time() - t0:.1f}s")
+    if distributed:
+        dist.barrier()
+
+    # --- plan build + HBM residency (untimed) ---
+    session = GpuSession(device_mask=1 << local_rank)
+    provider = StandardTableProvider(stream_dir, session)
+    plan = provider.scan(query)
+    t0 = time.time()
+    plan.load()
+    log(f"[bench] plan built + loaded to HBM in {time.time() - t0:.1f}s")
+
+    merger = DistMerger(query, device=merge_dev)
+    first = plan.execute(0)
+    merger.setup(first)
+
+    # --- warmup ---
+    for _ in range(args.warmup):
+        merger.step(plan.execute(0))
+
+    m_before = plan.metrics()
+
+    # --- timed region ---
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+    t_start = time.perf_counter()
+    rows_final = None
+    for _ in range(args.steps):
+        batch = plan.execute(0)
+        rows_final = merger.step(batch)
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+    if distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t_start
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=merge_dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    m_after = plan.metrics()
+
+    # --- per-step aggregates over all ranks ---
+    rank_rows = m_after["rows_scanned"]  # rows per execute on this rank
+    rank_bytes = m_after["bytes_scanned"]
+    rank_rg_bytes = m_after["rowgroup_bytes_total"]
+    if distributed:
+        t = torch.tensor([rank_rows, rank_bytes, rank_rg_bytes],
+                         dtype=torch.float64, device=merge_dev)
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        total_rows, total_bytes, total_rg_bytes = (float(x) for x in t.tolist())
+    else:
+        total_rows, total_bytes, total_rg_bytes = (
+            float(rank_rows), float(rank_bytes), float(rank_rg_bytes))
+
+    sec_per_step = elapsed / args.steps
+    rows_per_sec = total_rows / sec_per_step
+    gbps_scanned = total_bytes / sec_per_step / 1e9
+    gbps_rowgroup = total_rg_bytes / sec_per_step / 1e9
+
+    # --- kernel-stage breakdown (this rank; HIP events inside libgpuq) ---
+    dk = (m_after["kernel_ns"] - m_before["kernel_ns"]) / args.steps
+    dd = (m_after["decomp_ns"] - m_before["decomp_ns"]) / args.steps
+    raw_b = rank_bytes  # compressed chunk bytes resident in HBM, read once/step
+    # decompressed arena bytes: est = hbm_bytes_est - raw - dec (we stored raw+2*dec)
+    dec_b = (m_after["hbm_bytes_est"] - raw_b) / 2
+    # dominant kernel: the LZ4 page-decompression sweep (algorithmic bytes =
+    # compressed read + decompressed write per launch; one launch per step)
+    decomp_algo_bytes = raw_b + dec_b
+    other_ns = max(dk - dd, 1.0)
+    decode_algo_bytes = dec_b + 9.0 * rank_rows  # dec read + gid(4)+idx reads + mask ~ coarse
+    if dd >= other_ns:
+        roof_kernel, roof_bytes, roof_ns = "lz4_page_decompress", decomp_algo_bytes, dd
+    else:
+        roof_kernel, roof_bytes, roof_ns = "decode+filter+groupby", decode_algo_bytes, other_ns
+    achieved_gbs = roof_bytes / max(roof_ns, 1.0)  # bytes/ns == GB/s
+    roofline = {
+        "bound": "hbm",
+        "kernel": roof_kernel,
+        "achieved": round(achieved_gbs, 2),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+        "traffic": None,  # filled from rocprofv3 --pmc runs (profiles/)
+    }
+
+    # --- CPU baseline: the oracle (kind=port) on a bounded sample ---
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.skip_cpu_baseline:
+        from oracle import query_oracle as qo
+        import pyarrow as pa
+
+        sample_files = sorted(
+            os.path.join(dp, f)
+            for dp, _, fs in os.walk(stream_dir)
+            for f in fs if f.endswith(".parquet")
+        )[: max(2, min(12, n_files // 8))]
+        import pyarrow.parquet as pq
+
+        sample_rows = sum(pq.read_metadata(f).num_rows for f in sample_files)
+        tcb = time.perf_counter()
+        qo.execute(sample_files, query)
+        tcb = time.perf_counter() - tcb
+        cpu_baseline = {
+            "value": round(sample_rows / tcb, 1),
+            "unit": "rows/s",
+            "cores": pa.cpu_count(),
+            "kind": "port",
+            "sample": f"oracle (pyarrow-decode + numpy agg) on {len(sample_files)} "
+                      f"files = {sample_rows} rows, {tcb:.1f}s",
+        }
+
+    if rank == 0:
+        out = {
+            "metric": "rows/sec scanned (SELECT...WHERE...GROUP BY, Parseable parquet dialect)",
+            "value": round(rows_per_sec, 1),
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(sec_per_step * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # no published reference numbers
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": args.workload,
+                "query": {
+                    "c1": "SELECT level,count(*) GROUP BY level",
+                    "c2s": "SELECT host,count(*),max(latency) WHERE ts BETWEEN ... GROUP BY host",
+                    "c3s": "SELECT count(*) WHERE message LIKE '%error%'",
+                }[args.workload],
+                "rows_per_gpu": args.rows,
+                "files_per_gpu": n_files,
+                "parallelism": f"dp{world}",
+                "gb_per_sec_scanned": round(gbps_scanned, 2),
+                "gb_per_sec_rowgroup_bytes": round(gbps_rowgroup, 2),
+                "kernel_ms_per_step": round(dk / 1e6, 3),
+                "decomp_ms_per_step": round(dd / 1e6, 3),
+                "load_gbps_pcie": round(
+                    rank_bytes / max(m_after["load_ns"], 1), 2),
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+        if rows_final is not None:
+            log(f"[bench] result rows (first 8): {rows_final[:8]}")
+
+    plan.close()
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

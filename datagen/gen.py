@@ -67,7 +67,9 @@ def _rand_strings(rng, n, lo, hi, alphabet=string.ascii_lowercase + string.digit
 
 
 def _dict_col(rng, n, values, s=1.2):
-    idx = rng.choice(len(values), size=n, p=_zipf_probs(len(values), s)).astype(np.int32)
+    cum = np.cumsum(_zipf_probs(len(values), s))
+    idx = np.searchsorted(cum, rng.random(n)).astype(np.int32)
+    np.clip(idx, 0, len(values) - 1, out=idx)
     return pa.DictionaryArray.from_arrays(pa.array(idx), pa.array(values)).cast(
         pa.string()
     )
@@ -169,6 +171,13 @@ def _column_stats_entry(col_meta, name):
     }
 
 
+def _gen_one_file(args):
+    (root, stream, config, rows, seed, rows_per_file, data_page_size, m,
+     minute_offset) = args
+    return _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
+                           data_page_size, m, minute_offset)
+
+
 def gen_stream(
     root: str,
     stream: str = "bench",
@@ -178,19 +187,54 @@ def gen_stream(
     rows_per_file: int = ROW_GROUP_SIZE,
     data_page_size: int | None = None,
     quiet: bool = True,
+    workers: int = 1,
+    minute_offset: int = 0,
 ):
-    """Write a synthetic stream. Returns dict with file list + manifest paths."""
+    """Write a synthetic stream. Returns dict with file list + manifest paths.
+    workers > 1 parallelizes per-file generation (deterministic: per-file rng
+    streams); minute_offset shifts the time range (per-rank shards)."""
     n_files = (rows + rows_per_file - 1) // rows_per_file
     stream_dir = os.path.join(root, stream)
     os.makedirs(stream_dir, exist_ok=True)
     manifest_files = []  # File entries (src/catalog/manifest.rs:143-152)
     file_paths = []
 
+    if workers > 1 and n_files > 2:
+        from concurrent.futures import ProcessPoolExecutor
+
+        argl = [
+            (root, stream, config, rows, seed, rows_per_file, data_page_size,
+             m, minute_offset)
+            for m in range(n_files)
+        ]
+        with ProcessPoolExecutor(max_workers=workers) as ex:
+            for i, (abs_path, rel_path, entry) in enumerate(ex.map(_gen_one_file, argl, chunksize=4)):
+                manifest_files.append(entry)
+                file_paths.append(abs_path)
+                if not quiet and (i % 50 == 0):
+                    print(f"  wrote {i + 1}/{n_files} files", flush=True)
+        return _finish_stream(root, stream, stream_dir, manifest_files,
+                              file_paths, rows, config)
+
     for m in range(n_files):
+        abs_path, rel_path, entry = _gen_file_inner(
+            root, stream, config, rows, seed, rows_per_file, data_page_size,
+            m, minute_offset)
+        manifest_files.append(entry)
+        file_paths.append(abs_path)
+        if not quiet and (m % 50 == 0):
+            print(f"  wrote {m + 1}/{n_files} files", flush=True)
+    return _finish_stream(root, stream, stream_dir, manifest_files, file_paths,
+                          rows, config)
+
+
+def _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
+                    data_page_size, m, minute_offset):
+    if True:
         n = min(rows_per_file, rows - m * rows_per_file)
         rng = np.random.default_rng([seed, m])  # per-file stream: parallel-safe
-        tbl = _minute_batch(config, rng, n, m)
-        t0 = BASE_TS_MS + m * MINUTE_MS
+        tbl = _minute_batch(config, rng, n, m + minute_offset)
+        t0 = BASE_TS_MS + (m + minute_offset) * MINUTE_MS
         dt = datetime.fromtimestamp(t0 / 1000, tz=timezone.utc)
         rel_dir = (
             f"{stream}/date={dt:%Y-%m-%d}/hour={dt:%H}/minute={dt:%M}"
@@ -241,26 +285,25 @@ def gen_stream(
                         prev["stats"] = None
                 else:
                     cols[name] = e
-        manifest_files.append(
-            {
-                "file_path": rel_path,
-                "num_rows": md.num_rows,
-                "file_size": os.path.getsize(abs_path),
-                "ingestion_size": ing,
-                "columns": list(cols.values()),
-                "sort_order_id": [
-                    {
-                        "field_name": "p_timestamp",
-                        "sort_kind": "AtTimestamp",
-                        "descending": True,
-                    }
-                ],
-            }
-        )
-        file_paths.append(abs_path)
-        if not quiet and (m % 50 == 0):
-            print(f"  wrote {m + 1}/{n_files} files", flush=True)
+        entry = {
+            "file_path": rel_path,
+            "num_rows": md.num_rows,
+            "file_size": os.path.getsize(abs_path),
+            "ingestion_size": ing,
+            "columns": list(cols.values()),
+            "sort_order_id": [
+                {
+                    "field_name": "p_timestamp",
+                    "sort_kind": "AtTimestamp",
+                    "descending": True,
+                }
+            ],
+        }
+        return abs_path, rel_path, entry
 
+
+def _finish_stream(root, stream, stream_dir, manifest_files, file_paths, rows,
+                   config):
     # daily manifests (partition bounds = whole UTC day, src/catalog/mod.rs:176-187)
     by_day = {}
     for f in manifest_files:

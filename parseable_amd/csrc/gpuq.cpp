@@ -543,15 +543,18 @@ extern "C" gpuq_plan* gpuq_plan_build(
         int32_t page_id = (int32_t)part.pages.size();
 
         bool dict_enc = (pi.encoding == ENC_RLE_DICT || pi.encoding == ENC_PLAIN_DICT);
+        // aux = remap pool (gid decode) XOR dict-value pool (value decode):
+        // a column never needs both (utf8 aggregates beyond COUNT are
+        // rejected above); aux_lut is independent so a group key may also
+        // carry a string predicate.
         if (c.need_gid && dict_enc) {
           dp.aux = remap_base;
           part.tasks[{TK_DICT_GID, t.col_idx}].push_back(page_id);
         }
         if (c.need_val) {
           if (dict_enc) {
-            DevPage dv = dp; dv.aux = dictv_base;
-            part.tasks[{TK_DICT_VAL, t.col_idx}].push_back(page_id);
             dp.aux = dictv_base;
+            part.tasks[{TK_DICT_VAL, t.col_idx}].push_back(page_id);
           } else if (pi.encoding == ENC_PLAIN) {
             part.tasks[{TK_PLAIN_VAL, t.col_idx}].push_back(page_id);
           } else if (pi.encoding == ENC_DELTA_BP) {
@@ -560,7 +563,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
         }
         if (!c.lut_preds.empty()) {
           if (dict_enc) {
-            dp.aux = lut_base;
+            dp.aux_lut = lut_base;
             part.tasks[{TK_DICT_MASK, t.col_idx}].push_back(page_id);
           } else if (pi.encoding == ENC_PLAIN) {
             // PLAIN fallback page of a string column: only CONTAINS supported

@@ -300,23 +300,22 @@ k_dict_pages(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
   if (pi >= n) return;
   DevPage pg = pages[ids[pi]];
   Emit e = emit;
-  // per-page aux offset
-  e.advance(pg.aux);
+  e.advance(pg);  // per-page aux pool offsets
   dict_page_decode(pg, dec + pg.dst_off, e, d_error);
 }
 
 // wrappers adding per-page aux advance
 struct EmitGidP : EmitGid {
   const int32_t* pool;
-  __device__ void advance(uint32_t aux) { remap = pool + aux; }
+  __device__ void advance(const DevPage& pg) { remap = pool + pg.aux; }
 };
 struct EmitDictI64P : EmitDictI64 {
   const int64_t* pool;
-  __device__ void advance(uint32_t aux) { dictv = pool + aux; }
+  __device__ void advance(const DevPage& pg) { dictv = pool + pg.aux; }
 };
 struct EmitDictMaskP : EmitDictMask {
   const uint8_t* pool;
-  __device__ void advance(uint32_t aux) { lut = pool + aux; }
+  __device__ void advance(const DevPage& pg) { lut = pool + pg.aux_lut; }
 };
 
 // ------------------------------------------------------------------
