@@ -8,6 +8,7 @@ JSON, pruning, partial-aggregate merge). There is NO CPU fallback: on a
 machine without a GPU every compute entry point raises."""
 
 from .provider import (  # noqa: F401
+    EmptyScanResult,
     GpuSession,
     GpuExecutionPlan,
     GpuqError,

@@ -187,6 +187,8 @@ extern "C" gpuq_ctx* gpuq_session_create(uint64_t device_mask) {
   auto* c = new gpuq_ctx();
   int n = 0;
   if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
+  if (n == 0 && getenv("GPUQ_FAKE_DEVICE")) n = 1;  // host-plan debugging ONLY:
+  // plan_build is pure host work; load/execute still fail at the first HIP call.
   for (int i = 0; i < 64 && i < n; i++)
     if (device_mask & (1ull << i)) c->devices.push_back(i);
   if (c->devices.empty() && device_mask == 0 && n > 0) c->devices.push_back(0);
@@ -287,6 +289,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
   plan->limit = limit;
   if (ctx->devices.empty())
     throw std::runtime_error("no GPU devices in session (gpuq never falls back to CPU)");
+  if (n_files <= 0)
+    throw std::runtime_error("empty file list: caller should use the empty-relation path "
+                             "(provider returns the empty aggregate without a scan)");
 
   // --- map files + parse footers ---
   for (int32_t i = 0; i < n_files; i++) {
@@ -460,13 +465,18 @@ extern "C" gpuq_plan* gpuq_plan_build(
           if (pi.type != PAGE_DICT) continue;
           std::vector<uint8_t> dbuf(pi.uncomp_size);
           const uint8_t* d;
-          if (cm.codec == CODEC_UNCOMPRESSED || pi.comp_size == pi.uncomp_size) {
+          if (cm.codec == CODEC_UNCOMPRESSED) {
             d = mf.data + pi.payload_off;
           } else if (cm.codec == CODEC_LZ4_RAW) {
+            // ALWAYS try LZ4 first: the writer compresses every v1 page, and
+            // comp_size may coincidentally equal uncomp_size (seen in golden
+            // data); only a failed decode of an equal-size page means the
+            // page was stored raw (robustness for other writers).
             int n = lz4_decompress_host(mf.data + pi.payload_off, pi.comp_size,
                                         dbuf.data(), dbuf.size());
-            if (n != pi.uncomp_size) throw std::runtime_error("dict page lz4 failure");
-            d = dbuf.data();
+            if (n == pi.uncomp_size) d = dbuf.data();
+            else if (pi.comp_size == pi.uncomp_size) d = mf.data + pi.payload_off;
+            else throw std::runtime_error("dict page lz4 failure");
           } else throw std::runtime_error("unsupported codec");
           if (c.phys == PT_BYTE_ARRAY) {
             const uint8_t* q = d;

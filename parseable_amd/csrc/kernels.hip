@@ -37,7 +37,10 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
   uint8_t* dst = dec + pg.dst_off;
   const int lane = threadIdx.x;
 
-  if (pg.raw_copy || pg.comp_size == pg.uncomp_size) {
+  // NOTE: comp_size == uncomp_size does NOT mean stored-raw — the writer
+  // always compresses v1 pages and sizes can coincide. Try LZ4 and only
+  // fall back to a raw copy if decode fails on an equal-size page.
+  if (pg.raw_copy) {
     for (uint32_t i = lane * 16u; i < pg.uncomp_size; i += WAVE * 16u) {
       uint32_t rem = pg.uncomp_size - i;
       if (rem >= 16 && (((uintptr_t)(src + i)) & 15) == 0 && (((uintptr_t)(dst + i)) & 15) == 0) {
@@ -108,7 +111,14 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     __syncthreads();
     d += ml;
   }
-  if ((bad || d != uncomp) && lane == 0) atomicExch(d_error, ERR_LZ4);
+  if (bad || d != uncomp) {
+    if (pg.comp_size == pg.uncomp_size) {
+      // stored raw (equal-size page that is not valid LZ4): plain copy
+      for (uint32_t i = lane; i < pg.uncomp_size; i += WAVE) dst[i] = src[i];
+    } else if (lane == 0) {
+      atomicExch(d_error, ERR_LZ4);
+    }
+  }
 }
 
 // ------------------------------------------------------------------

@@ -35,8 +35,12 @@ class GpuqError(RuntimeError):
 
 class GpuSession:
     def __init__(self, device_mask: int = 0):
+        import os
+
         lib = _lib.load()
         n = lib.gpuq_device_count()
+        if n <= 0 and os.environ.get("GPUQ_FAKE_DEVICE"):
+            n = 1  # host-plan debugging only; execute will fail at HIP calls
         if n <= 0:
             raise GpuqError(
                 "no HIP devices visible — the gpuq path requires an MI355X; "
@@ -131,6 +135,17 @@ def _parse_iso_ms(s):
     )
 
 
+class EmptyScanResult:
+    """All files pruned at planning time: the scan is an empty relation —
+    the aggregate result over zero rows (DataFusion's EmptyExec analog)."""
+
+    def __init__(self, query):
+        self.query = query
+
+    def rows(self):
+        return merge_partials([], self.query)
+
+
 class ManifestCountResult:
     """Count fast path: bare `SELECT count(*)` with no value filters is
     answered from manifest num_rows sums and never reaches the scan engine
@@ -223,6 +238,8 @@ class StandardTableProvider:
             if exact:
                 return ManifestCountResult(sum(f["num_rows"] for f in kept))
 
+        if not kept:
+            return EmptyScanResult(query)
         paths = []
         root = os.path.dirname(self.stream_dir)
         for fe in kept:
@@ -420,7 +437,7 @@ class Query:
 
     def execute(self, query: dict):
         plan = self.provider.scan(query)
-        if isinstance(plan, ManifestCountResult):
+        if isinstance(plan, (ManifestCountResult, EmptyScanResult)):
             return plan.rows(), None
         try:
             rows = plan.execute_all()

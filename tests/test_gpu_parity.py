@@ -31,13 +31,13 @@ def _all_cases():
 
 @pytest.mark.parametrize("case", _all_cases())
 def test_gpu_matches_golden(golden, session, case):
-    from parseable_amd import ManifestCountResult, Query, StandardTableProvider
+    from parseable_amd import EmptyScanResult, ManifestCountResult, Query, StandardTableProvider
 
     fx, qname = case.split("/")
     entry = golden["answers"][case]
     provider = StandardTableProvider(os.path.join(GDIR, fx), session)
     plan = provider.scan(entry["query"])
-    if isinstance(plan, ManifestCountResult):
+    if isinstance(plan, (ManifestCountResult, EmptyScanResult)):
         rows = plan.rows()
     else:
         try:
