@@ -15,6 +15,7 @@ struct DevPage {
   uint32_t row_start;    // partition-global row index of first row
   uint32_t aux;          // index into remap pool (gid decode) / dict-value pool
   uint32_t aux_lut;      // index into predicate-LUT pool (dict-mask decode)
+  uint32_t dict_n;       // dictionary entry count of this chunk (0 if none)
   uint8_t  optional;     // has def levels (max_def_level == 1)
   uint8_t  raw_copy;     // stored uncompressed: memcpy instead of LZ4
   uint8_t  encoding;     // ENC_* (meta.h)

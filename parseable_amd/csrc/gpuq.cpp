@@ -548,6 +548,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
         row_in_rg += pi.num_values;
         dp.optional = mf.meta.columns[mf.meta.col_index(c.name)].optional;
         dp.raw_copy = (t.cm->codec == CODEC_UNCOMPRESSED);
+        dp.dict_n = (uint32_t)std::max(
+            {t.remap.size(), t.dictv.size(), t.lut.size()});
         dp.encoding = (uint8_t)pi.encoding;
         dp.phys = (uint8_t)c.phys;
         int32_t page_id = (int32_t)part.pages.size();

@@ -243,6 +243,7 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
     }
     // lane-redundant run-header walk; data movement parallel per run
     const uint8_t* p = vals;
+    const uint32_t dict_n = pg.dict_n;
     uint32_t v = 0;
     while (v < nv) {
       // varint header (redundant on all lanes)
@@ -257,22 +258,29 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
         uint32_t groups = (uint32_t)(hdr >> 1);
         // group g: 8 values packed in bw bytes at p + g*bw
         for (uint32_t g = lane; g < groups; g += WAVE) {
+          // one group = 8 values packed into bw BYTES (8 values x bw bits);
+          // acc may only hold the first 8 bytes — later values re-read a
+          // shifted 8-byte window (shifts must stay < 64).
           const uint8_t* q = p + (size_t)g * bw;
           uint64_t acc = 0;
-          for (int b = 0; b < bw; b++) acc |= (uint64_t)q[b] << (8 * b);
+          for (int b = 0; b < bw && b < 8; b++) acc |= (uint64_t)q[b] << (8 * b);
           uint32_t base = v + g * 8;
           uint32_t mask_v = (bw >= 32) ? 0xffffffffu : ((1u << bw) - 1);
           for (int k = 0; k < 8; k++) {
-            uint32_t idx = (uint32_t)(acc >> (k * bw)) & mask_v;
-            if (bw > 8 && k * bw + bw > 64) {  // straddles u64 window: re-read
+            uint32_t idx;
+            if (k * bw + bw <= 64) {
+              idx = (uint32_t)(acc >> (k * bw)) & mask_v;
+            } else {  // straddles the first u64 window: re-read
               uint64_t acc2 = 0;
               const uint8_t* q2 = q + (k * bw) / 8;
               int shift = (k * bw) % 8;
-              for (int b = 0; b < 8 && q2 + b < p + (size_t)(g + 1) * bw + 8; b++)
-                acc2 |= (uint64_t)q2[b] << (8 * b);
+              for (int b = 0; b < 8; b++) acc2 |= (uint64_t)q2[b] << (8 * b);
               idx = (uint32_t)(acc2 >> shift) & mask_v;
             }
-            if (base + k < nv) emit(row0 + base + k, idx);
+            if (base + k < nv) {
+              if (idx >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); idx = 0; }
+              emit(row0 + base + k, idx);
+            }
           }
         }
         p += (size_t)groups * bw;
@@ -285,6 +293,7 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
         for (int b = 0; b < byte_w; b++) val |= (uint32_t)p[b] << (8 * b);
         p += byte_w;
         if (cnt > nv - v) cnt = nv - v;
+        if (val >= dict_n) { if (lane == 0) atomicExch(d_error, ERR_DICT_RANGE); val = 0; }
         for (uint32_t i = lane; i < cnt; i += WAVE) emit(row0 + v + i, val);
         v += cnt;
       }
@@ -296,8 +305,11 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
     int bw = *vals++;
     SerialRle idx(vals, payload + pg.uncomp_size, bw);
     for (uint32_t r = 0; r < nv; r++) {
-      if (def.next()) emit(row0 + r, bw ? idx.next() : 0);
-      else emit.null_at(row0 + r);
+      if (def.next()) {
+        uint32_t ix = bw ? idx.next() : 0;
+        if (ix >= pg.dict_n) { atomicExch(d_error, ERR_DICT_RANGE); ix = 0; }
+        emit(row0 + r, ix);
+      } else emit.null_at(row0 + r);
     }
   }
 }
