@@ -173,6 +173,7 @@ struct gpuq_plan {
   int64_t limit = -1;
   std::vector<Partition> parts;
   int32_t n_groups = 0;          // product of key sizes (incl null slots)
+  bool fused_count = false;      // single dict key + count(*)-only + no preds
   std::mutex mu;
   // metrics
   int64_t m_rows_scanned = 0, m_rows_out = 0, m_kernel_ns = 0, m_exec_ns = 0,
@@ -602,6 +603,14 @@ extern "C" gpuq_plan* gpuq_plan_build(
   for (auto& part : plan->parts)
     plan->m_rows_scanned += part.n_rows;
 
+  // fused count path: GROUP BY <one dict col>, count(*)-only, no predicates
+  {
+    bool aggs_ok = true;
+    for (auto& ap : plan->aggs) aggs_ok &= (ap.kind == AGGK_COUNT_STAR);
+    plan->fused_count = plan->group_cols.size() == 1 && aggs_ok &&
+                        plan->preds.empty() && plan->n_groups <= 8192;
+  }
+
   return plan.release();
 } catch (const std::exception& e) {
   if (ctx) ctx->set_error(e.what());
@@ -799,8 +808,9 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   HIP_TRY(hipEventCreate(&ev1));
   HIP_TRY(hipEventCreate(&ev_decomp));
 
+  const bool need_mask = !plan->preds.empty();
   HIP_TRY(hipMemsetAsync(part.d_err, 0, 4, st));
-  HIP_TRY(hipMemsetAsync(part.d_mask, 1, part.n_rows, st));
+  if (need_mask) HIP_TRY(hipMemsetAsync(part.d_mask, 1, part.n_rows, st));
   launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
                     part.d_agg_kind);
 
@@ -821,7 +831,16 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     HIP_TRY(hipFree(d_all));
   }
 
-  // 2. decode + predicate kernels
+  // 2. decode + predicate kernels (or the fused count path)
+  if (plan->fused_count) {
+    int key_col = plan->group_cols[0];
+    auto it = part.tasks.find({TK_DICT_GID, key_col});
+    if (it != part.tasks.end())
+      launch_dict_count(st, part.d_dec, part.d_pages, part.d_ids[it->first],
+                        (int)it->second.size(), part.d_remap, part.d_table,
+                        plan->n_groups, (int)plan->aggs.size(), part.d_err);
+    HIP_TRY(hipEventRecord(ev1, st));
+  } else {
   for (auto& kv : part.tasks) {
     int kind = kv.first.first, col = kv.first.second;
     int n = (int)kv.second.size();
@@ -905,10 +924,12 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       a.agg_valid[i] = itd != part.d_valid.end() ? itd->second : nullptr;
     }
   }
+  a.mask = need_mask ? part.d_mask : nullptr;
   a.table = part.d_table;
   a.n_groups = plan->n_groups;
   launch_agg(st, a);
   HIP_TRY(hipEventRecord(ev1, st));
+  }  // !fused_count
 
   // 4. D2H results
   size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size());

@@ -25,11 +25,16 @@ namespace gpuq {
 // ------------------------------------------------------------------
 // LZ4_RAW page decompression: one wave per page.
 // ------------------------------------------------------------------
+#define LZ4_RING 16384  // LDS ring: last 16 KiB of output. LZ4 offsets reach
+                        // 64 KiB, but a 64 KiB ring caps occupancy at 2
+                        // waves/CU; 16 KiB gives 10 and far matches (rare)
+                        // read the already-written global output behind an
+                        // explicit vmcnt(0) drain.
 __global__ void __launch_bounds__(WAVE)
 k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
             const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
             int n, int32_t* __restrict__ d_error) {
-  __shared__ uint8_t ring[65536];  // last 64 KiB of output (LZ4 max offset)
+  __shared__ uint8_t ring[LZ4_RING];
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
@@ -69,7 +74,7 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     for (uint32_t i = lane; i < lit; i += WAVE) {
       uint8_t v = src[s + i];
       dst[d + i] = v;
-      ring[(d + i) & 0xFFFF] = v;
+      ring[(d + i) & (LZ4_RING - 1)] = v;
     }
     s += lit; d += lit;
     if (s >= comp) break;  // last sequence: literals only
@@ -85,28 +90,45 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     }
     ml += 4;
     if (d + ml > uncomp) { bad = true; break; }
-    // ---- match copy from the LDS ring, chunked so ring writes never
-    //      overwrite the pattern region another lane still reads ----
+    // ---- match copy, chunked so ring writes never overwrite the pattern
+    //      region another lane still reads ----
     uint32_t done = 0;
-    while (done < ml) {
-      uint32_t chunk;
-      __syncthreads();  // prior writes (literals / previous chunk) visible
-      if (off < WAVE) {
-        chunk = min(ml - done, 65536u - off);
+    if (off > LZ4_RING / 2) {
+      // far match (rare): pattern no longer (fully) in the ring — read the
+      // global output we already wrote. Same-wave store->load ordering
+      // needs an explicit drain of outstanding vector stores.
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      while (done < ml) {
+        uint32_t chunk = min(ml - done, off);
         for (uint32_t i = lane; i < chunk; i += WAVE) {
-          uint8_t v = ring[(d + done - off + (i % off)) & 0xFFFF];
+          uint8_t v = dst[d + done - off + i];
           dst[d + done + i] = v;
-          ring[(d + done + i) & 0xFFFF] = v;
+          ring[(d + done + i) & (LZ4_RING - 1)] = v;
         }
-      } else {
-        chunk = min(ml - done, min(off, 65536u - off));
-        for (uint32_t i = lane; i < chunk; i += WAVE) {
-          uint8_t v = ring[(d + done - off + i) & 0xFFFF];
-          dst[d + done + i] = v;
-          ring[(d + done + i) & 0xFFFF] = v;
-        }
+        done += chunk;
+        if (done < ml) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
-      done += chunk;
+    } else {
+      while (done < ml) {
+        uint32_t chunk;
+        __syncthreads();  // prior ring writes visible
+        if (off < WAVE) {
+          chunk = min(ml - done, (uint32_t)LZ4_RING - off);
+          for (uint32_t i = lane; i < chunk; i += WAVE) {
+            uint8_t v = ring[(d + done - off + (i % off)) & (LZ4_RING - 1)];
+            dst[d + done + i] = v;
+            ring[(d + done + i) & (LZ4_RING - 1)] = v;
+          }
+        } else {
+          chunk = min(ml - done, min(off, (uint32_t)LZ4_RING - off));
+          for (uint32_t i = lane; i < chunk; i += WAVE) {
+            uint8_t v = ring[(d + done - off + i) & (LZ4_RING - 1)];
+            dst[d + done + i] = v;
+            ring[(d + done + i) & (LZ4_RING - 1)] = v;
+          }
+        }
+        done += chunk;
+      }
     }
     __syncthreads();
     d += ml;
@@ -339,6 +361,120 @@ struct EmitDictMaskP : EmitDictMask {
   const uint8_t* pool;
   __device__ void advance(const DevPage& pg) { lut = pool + pg.aux_lut; }
 };
+
+// ------------------------------------------------------------------
+// FUSED count path: GROUP BY <single dict column> + count(*) only, no
+// predicates — decode RLE/bit-packed indices straight into a per-block
+// LDS histogram over global key ids; no gid materialization, no mask,
+// no separate aggregation pass. Table layout matches k_agg
+// (slots = 1 + 2*n_aggs, every agg is COUNT_STAR).
+// ------------------------------------------------------------------
+#define FUSED_MAX_GROUPS 8192
+__global__ void __launch_bounds__(WAVE)
+k_dict_count(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+             const int32_t* __restrict__ ids, int n,
+             const int32_t* __restrict__ remap_pool,
+             uint64_t* __restrict__ table, int32_t n_groups, int n_aggs,
+             int32_t* d_error) {
+  extern __shared__ uint64_t hist[];   // n_groups counters
+  const int lane = threadIdx.x;
+  for (int g = lane; g < n_groups; g += WAVE) hist[g] = 0;
+  __syncthreads();
+
+  // several pages per block: pages n mapped gridDim-strided
+  for (int pi = blockIdx.x; pi < n; pi += gridDim.x) {
+    DevPage pg = pages[ids[pi]];
+    const int32_t* remap = remap_pool + pg.aux;
+    const uint8_t* payload = dec + pg.dst_off;
+    const uint8_t* def_start; uint32_t def_len; bool all_valid;
+    const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
+    const uint32_t nv = pg.num_values;
+    const uint32_t dict_n = pg.dict_n;
+    if (!all_valid) {
+      if (lane == 0) {
+        SerialRle def(def_start, def_start + def_len, 1);
+        int bw = *vals++;
+        SerialRle idx(vals, payload + pg.uncomp_size, bw);
+        for (uint32_t r = 0; r < nv; r++) {
+          if (def.next()) {
+            uint32_t ix = bw ? idx.next() : 0;
+            if (ix >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); ix = 0; }
+            atomicAdd((unsigned long long*)&hist[remap[ix]], 1ull);
+          } else {
+            atomicAdd((unsigned long long*)&hist[0], 1ull);  // NULL group
+          }
+        }
+      }
+      continue;
+    }
+    int bw = *vals++;
+    if (bw == 0) {
+      if (lane == 0) atomicAdd((unsigned long long*)&hist[remap[0]], (unsigned long long)nv);
+      continue;
+    }
+    const uint8_t* p = vals;
+    uint32_t v = 0;
+    while (v < nv) {
+      uint64_t hdr = 0; int sh = 0;
+      for (;;) {
+        uint8_t b = *p++;
+        hdr |= (uint64_t)(b & 0x7f) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+      }
+      if (hdr & 1) {
+        uint32_t groups = (uint32_t)(hdr >> 1);
+        for (uint32_t g = lane; g < groups; g += WAVE) {
+          const uint8_t* q = p + (size_t)g * bw;
+          uint64_t acc = 0;
+          for (int b = 0; b < bw && b < 8; b++) acc |= (uint64_t)q[b] << (8 * b);
+          uint32_t base = v + g * 8;
+          uint32_t mask_v = (bw >= 32) ? 0xffffffffu : ((1u << bw) - 1);
+          for (int k = 0; k < 8; k++) {
+            uint32_t idx;
+            if (k * bw + bw <= 64) {
+              idx = (uint32_t)(acc >> (k * bw)) & mask_v;
+            } else {
+              uint64_t acc2 = 0;
+              const uint8_t* q2 = q + (k * bw) / 8;
+              int shift = (k * bw) % 8;
+              for (int b = 0; b < 8; b++) acc2 |= (uint64_t)q2[b] << (8 * b);
+              idx = (uint32_t)(acc2 >> shift) & mask_v;
+            }
+            if (base + k < nv) {
+              if (idx >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); idx = 0; }
+              atomicAdd((unsigned long long*)&hist[remap[idx]], 1ull);
+            }
+          }
+        }
+        p += (size_t)groups * bw;
+        uint32_t add = groups * 8;
+        v += (add > nv - v) ? (nv - v) : add;
+      } else {
+        uint32_t cnt = (uint32_t)(hdr >> 1);
+        uint32_t val = 0;
+        int byte_w = (bw + 7) / 8;
+        for (int b = 0; b < byte_w; b++) val |= (uint32_t)p[b] << (8 * b);
+        p += byte_w;
+        if (cnt > nv - v) cnt = nv - v;
+        if (val >= dict_n) { if (lane == 0) atomicExch(d_error, ERR_DICT_RANGE); val = 0; }
+        if (lane == 0) atomicAdd((unsigned long long*)&hist[remap[val]], (unsigned long long)cnt);
+        v += cnt;
+      }
+    }
+  }
+  __syncthreads();
+  // flush: presence + every (COUNT_STAR) agg count slot
+  int slots = 1 + 2 * n_aggs;
+  for (int g = lane; g < n_groups; g += WAVE) {
+    uint64_t h = hist[g];
+    if (!h) continue;
+    atomicAdd((unsigned long long*)&table[(int64_t)g * slots], (unsigned long long)h);
+    for (int a = 0; a < n_aggs; a++)
+      atomicAdd((unsigned long long*)&table[(int64_t)g * slots + 2 + 2 * a],
+                (unsigned long long)h);
+  }
+}
 
 // ------------------------------------------------------------------
 // PLAIN i64 / f64 pages -> row-aligned arrays
@@ -843,6 +979,17 @@ void launch_init_table(hipStream_t st, uint64_t* table, int32_t n_groups,
   if (blocks > 4096) blocks = 4096;
   hipLaunchKernelGGL(k_init_table, dim3(blocks), dim3(256), 0, st, table, n_groups, n_aggs, d_agg_kind);
 }
+void launch_dict_count(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                       const int32_t* ids, int n, const int32_t* remap_pool,
+                       uint64_t* table, int32_t n_groups, int n_aggs,
+                       int32_t* d_err) {
+  if (!n) return;
+  int blocks = n < 16384 ? n : 16384;
+  size_t lds = (size_t)n_groups * 8;
+  hipLaunchKernelGGL(k_dict_count, dim3(blocks), dim3(WAVE), lds, st,
+                     dec, pages, ids, n, remap_pool, table, n_groups, n_aggs, d_err);
+}
+
 void launch_agg(hipStream_t st, const AggArgs& a) {
   size_t lds = (size_t)a.n_groups * (1 + 2 * a.n_aggs) * 8;
   int blocks = (int)((a.n_rows + 255) / 256);

@@ -29,5 +29,9 @@ void launch_cmp_i64(hipStream_t, const int64_t* col, const uint8_t* valid,
                     uint8_t* mask, int64_t n);
 void launch_init_table(hipStream_t, uint64_t* table, int32_t n_groups,
                        int n_aggs, const int32_t* d_agg_kind);
+void launch_dict_count(hipStream_t, const uint8_t* dec, const DevPage*,
+                       const int32_t* ids, int n, const int32_t* remap_pool,
+                       uint64_t* table, int32_t n_groups, int n_aggs,
+                       int32_t* d_err);
 void launch_agg(hipStream_t, const AggArgs&);
 }  // namespace gpuq
