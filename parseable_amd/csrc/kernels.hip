@@ -156,8 +156,11 @@ struct SerialRle {
   uint32_t rle_val = 0;
   uint64_t bit_acc = 0;
   int bit_cnt = 0;
+  uint32_t pending = 0;
+  bool has_pending = false;
   __device__ SerialRle(const uint8_t* p_, const uint8_t* end_, int bw)
       : p(p_), end(end_), bit_width(bw), byte_w((bw + 7) / 8) {}
+  __device__ void unread(uint32_t v) { pending = v; has_pending = true; }
   __device__ uint32_t varint() {
     uint64_t v = 0; int sh = 0;
     for (;;) {
@@ -168,6 +171,7 @@ struct SerialRle {
     }
   }
   __device__ uint32_t next() {
+    if (has_pending) { has_pending = false; return pending; }
     if (!run_left) {
       uint32_t hdr = varint();
       if (hdr & 1) {
@@ -685,14 +689,22 @@ k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
 // substring scan, one wave per page. lane0 walks the length-prefixed
 // values into an LDS batch of offsets; all lanes then scan one value each.
 // ------------------------------------------------------------------
-#define CBATCH 1024
+// The page's value stream (u32 length prefix + bytes) is staged through an
+// LDS window: the whole wave copies the next 16 KiB coalesced, lane0 walks
+// the length chain inside LDS (~50-cycle steps instead of dependent global
+// loads), and all lanes substring-scan their values from LDS. Values longer
+// than the window margin fall back to a global scan (rare at log sizes).
+#define CWIN 16384
+#define CVALS 2048
 __global__ void __launch_bounds__(WAVE)
 k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
                  const int32_t* __restrict__ ids, int n,
                  const uint8_t* __restrict__ needle, int nlen,
                  uint8_t* __restrict__ mask, int32_t* d_error) {
-  __shared__ uint32_t offs[CBATCH + 1];
-  __shared__ uint8_t nulls[CBATCH];
+  __shared__ uint8_t win[CWIN];
+  __shared__ uint32_t offs[CVALS + 1];   // window-relative value starts
+  __shared__ uint8_t nulls[CVALS];
+  __shared__ uint32_t ctrl[2];           // [0]=values in window, [1]=bytes consumed
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
@@ -701,33 +713,63 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
   const uint8_t* payload = dec + pg.dst_off;
   const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
   const uint32_t nv = pg.num_values, row0 = pg.row_start;
+  const uint32_t page_bytes = (uint32_t)(pg.uncomp_size - (vals - payload));
 
   SerialRle def(def_start, def_start + (def_start ? def_len : 0), 1);
-  uint32_t walk = 0;  // byte offset within vals (maintained by lane0)
-  for (uint32_t b0 = 0; b0 < nv; b0 += CBATCH) {
-    uint32_t bn = min((uint32_t)CBATCH, nv - b0);
-    if (lane == 0) {
-      uint32_t w = walk;
-      for (uint32_t i = 0; i < bn; i++) {
-        int present = all_valid ? 1 : (int)def.next();
-        nulls[i] = (uint8_t)!present;
-        offs[i] = w;
-        if (present) {
-          uint32_t l;
-          memcpy(&l, vals + w, 4);
-          w += 4 + l;
-        }
-      }
-      offs[bn] = w;
-      walk = w;
+  uint32_t done = 0;   // values consumed
+  uint32_t walk = 0;   // byte position within vals
+  while (done < nv) {
+    // stage the next window (coalesced u32 copies; over-read is padded)
+    uint32_t rem = page_bytes > walk ? page_bytes - walk : 0;
+    uint32_t wbytes = min((uint32_t)CWIN, rem + 8);
+    for (uint32_t i = lane * 4u; i < wbytes; i += WAVE * 4u) {
+      uint32_t v;
+      __builtin_memcpy(&v, vals + walk + i, 4);
+      *(uint32_t*)&win[i] = v;
     }
     __syncthreads();
+    if (lane == 0) {
+      // walk lengths inside LDS; stop when a value would cross the window
+      uint32_t w = 0, cnt = 0;
+      while (done + cnt < nv && cnt < CVALS) {
+        int present = all_valid ? 1 : (int)def.next();
+        nulls[cnt] = (uint8_t)!present;
+        offs[cnt] = w;
+        if (present) {
+          if (w + 4 > CWIN) {  // length prefix crosses the window
+            if (!all_valid) def.unread((uint32_t)present);
+            break;
+          }
+          uint32_t l;
+          __builtin_memcpy(&l, &win[w], 4);
+          if (w + 4 + l > CWIN) {
+            if (cnt == 0) {
+              // oversized single value: scan it from global below
+              offs[0] = w | 0x80000000u;
+              nulls[0] = (uint8_t)2;
+              w += 4 + l;
+              cnt = 1;
+            } else if (!all_valid) {
+              def.unread((uint32_t)present);
+            }
+            break;
+          }
+          w += 4 + l;
+        }
+        cnt++;
+      }
+      offs[cnt] = w;
+      ctrl[0] = cnt;
+      ctrl[1] = w;
+    }
+    __syncthreads();
+    uint32_t bn = ctrl[0];
     for (uint32_t i = lane; i < bn; i += WAVE) {
       uint8_t hit = 0;
-      if (!nulls[i]) {
-        uint32_t o = offs[i];
+      if (nulls[i] == 2) {  // oversized value: scan from global
+        uint32_t o = walk + (offs[i] & 0x7fffffffu);
         uint32_t vl;
-        memcpy(&vl, vals + o, 4);
+        __builtin_memcpy(&vl, vals + o, 4);
         const uint8_t* s = vals + o + 4;
         if (nlen == 0) hit = 1;
         else if (vl >= (uint32_t)nlen) {
@@ -740,11 +782,31 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
             }
           }
         }
+      } else if (!nulls[i]) {
+        uint32_t o = offs[i];
+        uint32_t vl;
+        __builtin_memcpy(&vl, &win[o], 4);
+        const uint8_t* s = &win[o + 4];
+        if (nlen == 0) hit = 1;
+        else if (vl >= (uint32_t)nlen) {
+          uint8_t c0 = needle[0];
+          for (uint32_t j = 0; j + nlen <= vl; j++) {
+            if (s[j] == c0) {
+              uint32_t k = 1;
+              while (k < (uint32_t)nlen && s[j + k] == needle[k]) k++;
+              if (k == (uint32_t)nlen) { hit = 1; break; }
+            }
+          }
+        }
       }
-      mask[row0 + b0 + i] &= hit;
+      mask[row0 + done + i] &= hit;
     }
     __syncthreads();
+    done += bn;
+    walk += ctrl[1];
+    if (bn == 0) break;  // defensive: no progress
   }
+  if (done != nv && lane == 0) atomicExch(d_error, ERR_PAGE);
 }
 
 // ------------------------------------------------------------------
@@ -995,7 +1057,9 @@ void launch_agg(hipStream_t st, const AggArgs& a) {
   int blocks = (int)((a.n_rows + 255) / 256);
   if (blocks > 2048) blocks = 2048;
   if (blocks < 1) blocks = 1;
-  if (lds <= 32 * 1024) {
+  // 64 KiB still admits 2 blocks/CU and avoids the global-atomic cliff on
+  // ~1000-group tables (c2: GROUP BY host was 100x slower via global atomics)
+  if (lds <= 64 * 1024) {
     hipLaunchKernelGGL(k_agg<true>, dim3(blocks), dim3(256), lds, st, a);
   } else {
     hipLaunchKernelGGL(k_agg<false>, dim3(blocks), dim3(256), 0, st, a);
