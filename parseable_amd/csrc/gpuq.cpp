@@ -632,7 +632,9 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   HIP_TRY(hipSetDevice(part.device));
   HIP_TRY(hipStreamCreateWithFlags(&part.stream, hipStreamNonBlocking));
 
-  HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes, 16)));
+  // +4KB+64 slack: the LZ4 kernel's input-window refill may read past the
+  // last chunk's end (see kernels.hip refill note)
+  HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes + 4160, 16)));
   HIP_TRY(hipMalloc(&part.d_dec, std::max<uint64_t>(part.dec_bytes, 16)));
   for (auto& t : part.chunks) {
     const auto& mf = *plan->files[t.file_idx];

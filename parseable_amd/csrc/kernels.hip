@@ -25,16 +25,24 @@ namespace gpuq {
 // ------------------------------------------------------------------
 // LZ4_RAW page decompression: one wave per page.
 // ------------------------------------------------------------------
-#define LZ4_RING 16384  // LDS ring: last 16 KiB of output. LZ4 offsets reach
-                        // 64 KiB, but a 64 KiB ring caps occupancy at 2
-                        // waves/CU; 16 KiB gives 10 and far matches (rare)
-                        // read the already-written global output behind an
-                        // explicit vmcnt(0) drain.
+#define LZ4_RING 16384  // LDS output ring (last 16 KiB; 64 KiB offsets fall
+                        // back to global reads behind a vmcnt drain)
+#define LZ4_IN 4096     // LDS input window for the token parse
+
+// One wave per page. v3 design notes:
+//  * the token stream is parsed lane-redundantly from an LDS-staged input
+//    window (dependent global L1/L2 loads were ~300 cycles per sequence);
+//  * a single wave's LDS operations complete in program order, so the
+//    literal-write -> match-read ordering through the ring needs NO
+//    barrier — only a compiler-ordering wave_barrier();
+//  * literals are copied from the LDS input window; matches from the LDS
+//    output ring (or global output for far offsets, behind vmcnt(0)).
 __global__ void __launch_bounds__(WAVE)
 k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
             const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
             int n, int32_t* __restrict__ d_error) {
   __shared__ uint8_t ring[LZ4_RING];
+  __shared__ uint8_t inbuf[LZ4_IN + 64];
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
@@ -42,9 +50,6 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
   uint8_t* dst = dec + pg.dst_off;
   const int lane = threadIdx.x;
 
-  // NOTE: comp_size == uncomp_size does NOT mean stored-raw — the writer
-  // always compresses v1 pages and sizes can coincide. Try LZ4 and only
-  // fall back to a raw copy if decode fails on an equal-size page.
   if (pg.raw_copy) {
     for (uint32_t i = lane * 16u; i < pg.uncomp_size; i += WAVE * 16u) {
       uint32_t rem = pg.uncomp_size - i;
@@ -57,46 +62,74 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     return;
   }
 
-  uint32_t s = 0, d = 0;
   const uint32_t comp = pg.comp_size, uncomp = pg.uncomp_size;
+  uint32_t in_base = 0;
+  bool in_valid = false;
+  // refill the input window to cover [pos, pos + LZ4_IN) (clamped to comp+pad)
+  auto refill = [&](uint32_t pos) {
+    in_base = pos & ~15u;
+    // NOTE: reads up to 64B past the page inside d_raw (chunks are packed;
+    // the partition raw buffer is sized with this slack by the host)
+    for (uint32_t i = lane * 4u; i < LZ4_IN + 64u; i += WAVE * 4u) {
+      uint32_t v;
+      __builtin_memcpy(&v, src + in_base + i, 4);  // src alignment is arbitrary
+      *(uint32_t*)&inbuf[i] = v;
+    }
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    in_valid = true;
+  };
+  auto inb = [&](uint32_t pos) -> uint8_t {
+    if (!in_valid || pos - in_base >= LZ4_IN) refill(pos);
+    return inbuf[pos - in_base];
+  };
+
+  uint32_t s = 0, d = 0;
   bool bad = false;
   while (s < comp && d < uncomp) {
-    // ---- lane-redundant parse ----
-    uint32_t token = src[s++];
+    uint32_t token = inb(s); s++;
     uint32_t lit = token >> 4;
     if (lit == 15) {
       uint32_t b;
-      do { if (s >= comp) { bad = true; break; } b = src[s++]; lit += b; } while (b == 255);
+      do { if (s >= comp) { bad = true; break; } b = inb(s); s++; lit += b; } while (b == 255);
       if (bad) break;
     }
     if (s + lit > comp || d + lit > uncomp) { bad = true; break; }
-    // ---- parallel literal copy (global->global + LDS ring) ----
-    for (uint32_t i = lane; i < lit; i += WAVE) {
-      uint8_t v = src[s + i];
-      dst[d + i] = v;
-      ring[(d + i) & (LZ4_RING - 1)] = v;
+    // literal copy: chunked through the input window
+    {
+      uint32_t doneL = 0;
+      while (doneL < lit) {
+        if (!in_valid || (s + doneL) - in_base >= LZ4_IN) refill(s + doneL);
+        uint32_t avail = LZ4_IN - ((s + doneL) - in_base);
+        uint32_t chunk = min(lit - doneL, avail);
+        const uint8_t* lsrc = &inbuf[(s + doneL) - in_base];
+        uint32_t base = d + doneL;
+        for (uint32_t i = lane; i < chunk; i += WAVE) {
+          uint8_t v = lsrc[i];
+          dst[base + i] = v;
+          ring[(base + i) & (LZ4_RING - 1)] = v;
+        }
+        doneL += chunk;
+      }
+      __builtin_amdgcn_wave_barrier();
     }
     s += lit; d += lit;
-    if (s >= comp) break;  // last sequence: literals only
+    if (s >= comp) break;  // last sequence carries only literals
     if (s + 2 > comp) { bad = true; break; }
-    uint32_t off = src[s] | ((uint32_t)src[s + 1] << 8);
+    uint32_t off = inb(s) | ((uint32_t)inb(s + 1) << 8);
     s += 2;
     if (off == 0 || off > d) { bad = true; break; }
     uint32_t ml = token & 0xf;
     if (ml == 15) {
       uint32_t b;
-      do { if (s >= comp) { bad = true; break; } b = src[s++]; ml += b; } while (b == 255);
+      do { if (s >= comp) { bad = true; break; } b = inb(s); s++; ml += b; } while (b == 255);
       if (bad) break;
     }
     ml += 4;
     if (d + ml > uncomp) { bad = true; break; }
-    // ---- match copy, chunked so ring writes never overwrite the pattern
-    //      region another lane still reads ----
     uint32_t done = 0;
     if (off > LZ4_RING / 2) {
-      // far match (rare): pattern no longer (fully) in the ring — read the
-      // global output we already wrote. Same-wave store->load ordering
-      // needs an explicit drain of outstanding vector stores.
+      // far match: read the already-written global output (rare)
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       while (done < ml) {
         uint32_t chunk = min(ml - done, off);
@@ -109,9 +142,11 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
         if (done < ml) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
     } else {
+      // near match from the ring; single-wave LDS ops are in-order, so no
+      // barrier between chunks — wave_barrier only pins compiler order
       while (done < ml) {
         uint32_t chunk;
-        __syncthreads();  // prior ring writes visible
+        __builtin_amdgcn_wave_barrier();
         if (off < WAVE) {
           chunk = min(ml - done, (uint32_t)LZ4_RING - off);
           for (uint32_t i = lane; i < chunk; i += WAVE) {
@@ -129,13 +164,12 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
         }
         done += chunk;
       }
+      __builtin_amdgcn_wave_barrier();
     }
-    __syncthreads();
     d += ml;
   }
   if (bad || d != uncomp) {
     if (pg.comp_size == pg.uncomp_size) {
-      // stored raw (equal-size page that is not valid LZ4): plain copy
       for (uint32_t i = lane; i < pg.uncomp_size; i += WAVE) dst[i] = src[i];
     } else if (lane == 0) {
       atomicExch(d_error, ERR_LZ4);
