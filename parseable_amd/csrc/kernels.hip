@@ -84,19 +84,55 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     return inbuf[pos - in_base];
   };
 
+  const uint32_t* in32 = (const uint32_t*)inbuf;
   uint32_t s = 0, d = 0;
   bool bad = false;
   while (s < comp && d < uncomp) {
-    uint32_t token = inb(s); s++;
+    // FAST SEQUENCE PARSE: match-heavy pages have a sequence every ~10
+    // bytes; byte-wise dependent LDS reads cost ~300 cycles/sequence.
+    // Pull 24 bytes into registers once and extract token/offset from
+    // register math (short-literal, short-match case: the common one).
+    if (!in_valid || s - in_base >= LZ4_IN) refill(s);
+    uint32_t rel = s - in_base;                 // < LZ4_IN; +24 fits the +64 pad
+    uint32_t w[6];
+#pragma unroll
+    for (int k = 0; k < 6; k++) w[k] = in32[(rel >> 2) + k];
+    uint32_t sub = rel & 3;
+    auto gb = [&](uint32_t j) {
+      uint32_t t = sub + j;
+      return (w[t >> 2] >> ((t & 3) * 8)) & 0xffu;
+    };
+    uint32_t token = gb(0);
     uint32_t lit = token >> 4;
-    if (lit == 15) {
+    uint32_t off, ml;
+    if (lit < 15) {
+      // literals fully inside the register/LDS window (rel+1+lit < LZ4_IN+15)
+      if (s + 1 + lit > comp || d + lit > uncomp) { bad = true; break; }
+      const uint8_t* lsrc = &inbuf[rel + 1];
+      for (uint32_t i = lane; i < lit; i += WAVE) {
+        uint8_t v = lsrc[i];
+        dst[d + i] = v;
+        ring[(d + i) & (LZ4_RING - 1)] = v;
+      }
+      __builtin_amdgcn_wave_barrier();
+      s += 1 + lit; d += lit;
+      if (s >= comp) break;                 // last sequence: literals only
+      if (s + 2 > comp) { bad = true; break; }
+      off = gb(1 + lit) | (gb(2 + lit) << 8);
+      s += 2;
+      ml = token & 0xf;
+      if (ml == 15) {
+        uint32_t b;
+        do { if (s >= comp) { bad = true; break; } b = inb(s); s++; ml += b; } while (b == 255);
+        if (bad) break;
+      }
+    } else {
+      // long-literal path (rare on match-heavy pages): byte-wise parse
+      s++;
       uint32_t b;
       do { if (s >= comp) { bad = true; break; } b = inb(s); s++; lit += b; } while (b == 255);
       if (bad) break;
-    }
-    if (s + lit > comp || d + lit > uncomp) { bad = true; break; }
-    // literal copy: chunked through the input window
-    {
+      if (s + lit > comp || d + lit > uncomp) { bad = true; break; }
       uint32_t doneL = 0;
       while (doneL < lit) {
         if (!in_valid || (s + doneL) - in_base >= LZ4_IN) refill(s + doneL);
@@ -112,19 +148,19 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
         doneL += chunk;
       }
       __builtin_amdgcn_wave_barrier();
+      s += lit; d += lit;
+      if (s >= comp) break;
+      if (s + 2 > comp) { bad = true; break; }
+      off = inb(s) | ((uint32_t)inb(s + 1) << 8);
+      s += 2;
+      ml = token & 0xf;
+      if (ml == 15) {
+        uint32_t b2;
+        do { if (s >= comp) { bad = true; break; } b2 = inb(s); s++; ml += b2; } while (b2 == 255);
+        if (bad) break;
+      }
     }
-    s += lit; d += lit;
-    if (s >= comp) break;  // last sequence carries only literals
-    if (s + 2 > comp) { bad = true; break; }
-    uint32_t off = inb(s) | ((uint32_t)inb(s + 1) << 8);
-    s += 2;
     if (off == 0 || off > d) { bad = true; break; }
-    uint32_t ml = token & 0xf;
-    if (ml == 15) {
-      uint32_t b;
-      do { if (s >= comp) { bad = true; break; } b = inb(s); s++; ml += b; } while (b == 255);
-      if (bad) break;
-    }
     ml += 4;
     if (d + ml > uncomp) { bad = true; break; }
     uint32_t done = 0;
