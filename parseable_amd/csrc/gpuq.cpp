@@ -541,7 +541,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
         DevPage dp{};
         dp.src_off = t.raw_off + (uint64_t)(pi.payload_off - t.cm->start_offset());
         dp.dst_off = part.dec_bytes;
-        part.dec_bytes += (uint64_t)pi.uncomp_size;
+        // 16-align page images so dst and LDS-ring offsets share alignment
+        part.dec_bytes += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
         dp.comp_size = pi.comp_size;
         dp.uncomp_size = pi.uncomp_size;
         dp.num_values = pi.num_values;
