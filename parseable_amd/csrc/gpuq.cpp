@@ -591,7 +591,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       if (row_in_rg != (uint32_t)mf.meta.row_groups[t.rg_idx].num_rows)
         throw std::runtime_error("page rows mismatch");
     }
-    part.dec_bytes += 64;  // over-read pad for bit unpackers
+    part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
   }
 
   // group table size
@@ -633,9 +633,9 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   HIP_TRY(hipSetDevice(part.device));
   HIP_TRY(hipStreamCreateWithFlags(&part.stream, hipStreamNonBlocking));
 
-  // +4KB+64 slack: the LZ4 kernel's input-window refill may read past the
+  // slack: the LZ4 input-window refill may read up to LZ4_IN+256 past the
   // last chunk's end (see kernels.hip refill note)
-  HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes + 4160, 16)));
+  HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes + 8192, 16)));
   HIP_TRY(hipMalloc(&part.d_dec, std::max<uint64_t>(part.dec_bytes, 16)));
   for (auto& t : part.chunks) {
     const auto& mf = *plan->files[t.file_idx];

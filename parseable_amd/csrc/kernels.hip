@@ -109,7 +109,7 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
             const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
             int n, int32_t* __restrict__ d_error) {
   __shared__ uint8_t ring[LZ4_RING];
-  __shared__ uint8_t inbuf[LZ4_IN + 64];
+  __shared__ uint8_t inbuf[LZ4_IN + 256];  // 17 x 256B per-lane copy rounds
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
@@ -135,13 +135,18 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
   // refill the input window to cover [pos, pos + LZ4_IN) (clamped to comp+pad)
   auto refill = [&](uint32_t pos) {
     in_base = pos & ~15u;
-    // NOTE: reads up to 64B past the page inside d_raw (chunks are packed;
-    // the partition raw buffer is sized with this slack by the host)
-    for (uint32_t i = lane * 4u; i < LZ4_IN + 64u; i += WAVE * 4u) {
-      uint32_t v;
-      __builtin_memcpy(&v, src + in_base + i, 4);  // src alignment is arbitrary
-      *(uint32_t*)&inbuf[i] = v;
-    }
+    // ALL loads issued before any LDS store: an interleaved load->store loop
+    // serializes on a waitcnt per iteration (~900 cycles / 256 B — the
+    // 0.6 GB/s-per-wave pathology measured by scripts/micro_lz4.cpp).
+    // Reads up to LZ4_IN+256 past the page inside d_raw (host pads the raw
+    // buffer with this slack).
+    uint32_t v[17];
+#pragma unroll
+    for (int k = 0; k < 17; k++)
+      __builtin_memcpy(&v[k], src + in_base + lane * 4u + (uint32_t)k * (WAVE * 4u), 4);
+#pragma unroll
+    for (int k = 0; k < 17; k++)
+      *(uint32_t*)&inbuf[lane * 4u + (uint32_t)k * (WAVE * 4u)] = v[k];
     __builtin_amdgcn_wave_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     in_valid = true;
@@ -861,10 +866,18 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
   while (done < nv) {
     uint32_t rem = page_bytes > walk ? page_bytes - walk : 0;
     uint32_t wbytes = min((uint32_t)CWIN, rem + 8);
-    for (uint32_t i = threadIdx.x * 4u; i < wbytes; i += CTHREADS * 4u) {
-      uint32_t v;
-      __builtin_memcpy(&v, vals + walk + i, 4);
-      *(uint32_t*)&win[i] = v;
+    // two-phase unconditional full-window copy (16 u32/thread): all global
+    // loads in flight before the LDS stores — see the refill note in
+    // k_lz4_pages. Over-reads past the page stay inside the arena (host
+    // pads the decompressed arena by CWIN).
+    {
+      uint32_t v[CWIN / (CTHREADS * 4)];
+#pragma unroll
+      for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
+        __builtin_memcpy(&v[k], vals + walk + threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u), 4);
+#pragma unroll
+      for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
+        *(uint32_t*)&win[threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u)] = v[k];
     }
     for (uint32_t i = threadIdx.x; i < CWIN / 32; i += CTHREADS) bm[i] = 0;
     __syncthreads();
