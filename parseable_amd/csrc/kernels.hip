@@ -178,9 +178,15 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     uint32_t lit = token >> 4;
     uint32_t off, ml;
     if (lit < 15) {
-      // literals fully inside the register/LDS window (rel+1+lit < LZ4_IN+15)
+      // short literal: the bytes are ALREADY in the w[] registers — write
+      // them in one divergence-masked pass, no loop, no LDS read. (Loop
+      // machinery cost ~500 cycles/sequence on a solo wave.)
       if (s + 1 + lit > comp || d + lit > uncomp) { bad = true; break; }
-      vcopy_from_lds(in32, rel + 1, dst, ring, d, lit, lane);
+      if ((uint32_t)lane < lit) {
+        uint8_t v = (uint8_t)gb(1 + lane);
+        dst[d + lane] = v;
+        ring[(d + lane) & (LZ4_RING - 1)] = v;
+      }
       __builtin_amdgcn_wave_barrier();
       s += 1 + lit; d += lit;
       if (s >= comp) break;                 // last sequence: literals only
@@ -224,6 +230,24 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
     if (off == 0 || off > d) { bad = true; break; }
     ml += 4;
     if (d + ml > uncomp) { bad = true; break; }
+    if (ml <= (uint32_t)WAVE && off <= LZ4_RING / 2) {
+      // short match, one divergence-masked pass, no loop. Source index:
+      // off >= 64 -> direct; power-of-two off (the 4/8-byte periods of
+      // PLAIN int pages) -> mask; else modulo.
+      __builtin_amdgcn_wave_barrier();
+      if ((uint32_t)lane < ml) {
+        uint32_t j;
+        if (off >= (uint32_t)WAVE) j = (uint32_t)lane;
+        else if ((off & (off - 1)) == 0) j = (uint32_t)lane & (off - 1);
+        else j = (uint32_t)lane % off;
+        uint8_t v = ring[(d - off + j) & (LZ4_RING - 1)];
+        dst[d + lane] = v;
+        ring[(d + lane) & (LZ4_RING - 1)] = v;
+      }
+      __builtin_amdgcn_wave_barrier();
+      d += ml;
+      continue;
+    }
     uint32_t done = 0;
     if (off > LZ4_RING / 2) {
       // far match: read the already-written global output (rare)

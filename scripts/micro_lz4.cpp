@@ -26,8 +26,12 @@ struct Pg { uint64_t src_off, dst_off; uint32_t comp, uncomp; };
 __global__ void __launch_bounds__(WAVE)
 k_lz4(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
       const Pg* pages, int n, uint64_t* cycles, int* err) {
+  uint64_t c_parse = 0, c_lit = 0, c_match = 0, c_refill = 0, n_seq = 0;
+  uint64_t tmark;
+#define MARK() tmark = __builtin_amdgcn_s_memtime()
+#define ACC(x) x += __builtin_amdgcn_s_memtime() - tmark
   __shared__ uint8_t ring[LZ4_RING];
-  __shared__ uint8_t inbuf[LZ4_IN + 64];
+  __shared__ uint8_t inbuf[LZ4_IN + 256];
   int pi = blockIdx.x;
   if (pi >= n) return;
   const Pg pg = pages[pi];
@@ -41,11 +45,13 @@ k_lz4(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
   bool in_valid = false;
   auto refill = [&](uint32_t pos) {
     in_base = pos & ~15u;
-    for (uint32_t i = lane * 4u; i < LZ4_IN + 64u; i += WAVE * 4u) {
-      uint32_t v;
-      __builtin_memcpy(&v, src + in_base + i, 4);
-      *(uint32_t*)&inbuf[i] = v;
-    }
+    uint32_t v[17];
+#pragma unroll
+    for (int k = 0; k < 17; k++)
+      __builtin_memcpy(&v[k], src + in_base + lane * 4u + (uint32_t)k * (WAVE * 4u), 4);
+#pragma unroll
+    for (int k = 0; k < 17; k++)
+      *(uint32_t*)&inbuf[lane * 4u + (uint32_t)k * (WAVE * 4u)] = v[k];
     __builtin_amdgcn_wave_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     in_valid = true;
@@ -58,7 +64,9 @@ k_lz4(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
   uint32_t s = 0, d = 0;
   bool bad = false;
   while (s < comp && d < uncomp) {
+    MARK();
     if (!in_valid || s - in_base >= LZ4_IN) refill(s);
+    ACC(c_refill); MARK(); n_seq++;
     uint32_t rel = s - in_base;
     uint32_t w[6];
 #pragma unroll
@@ -69,6 +77,7 @@ k_lz4(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
       return (w[t >> 2] >> ((t & 3) * 8)) & 0xffu;
     };
     uint32_t token = gb(0);
+    ACC(c_parse); MARK();
     uint32_t lit = token >> 4;
     uint32_t off, ml;
     if (lit < 15) {
@@ -125,6 +134,7 @@ k_lz4(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
       }
     }
     if (off == 0 || off > d) { bad = true; break; }
+    ACC(c_lit); MARK();
     ml += 4;
     if (d + ml > uncomp) { bad = true; break; }
     uint32_t done = 0;
@@ -164,9 +174,17 @@ k_lz4(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
       __builtin_amdgcn_wave_barrier();
     }
     d += ml;
+    ACC(c_match);
   }
   if ((bad || d != uncomp) && lane == 0) *err = 1;
-  if (lane == 0) cycles[pi] = __builtin_amdgcn_s_memtime() - t0;
+  if (lane == 0) {
+    cycles[pi] = __builtin_amdgcn_s_memtime() - t0;
+    cycles[n + pi * 5 + 0] = c_refill;
+    cycles[n + pi * 5 + 1] = c_parse;
+    cycles[n + pi * 5 + 2] = c_lit;
+    cycles[n + pi * 5 + 3] = c_match;
+    cycles[n + pi * 5 + 4] = n_seq;
+  }
 }
 
 int main(int argc, char** argv) {
@@ -196,27 +214,61 @@ int main(int argc, char** argv) {
       names.push_back(fm.columns[ci].name + (pi.type == PAGE_DICT ? "/dict" : "/data"));
     }
   }
-  rawbuf.resize(rawbuf.size() + 4160);
+  rawbuf.resize(rawbuf.size() + 8192);
+  int rep = argc > 2 ? atoi(argv[2]) : 1;
+  {
+    int base_n = (int)pages.size();
+    uint64_t base_dst = dst_off;
+    for (int r = 1; r < rep; r++)
+      for (int i = 0; i < base_n; i++) {
+        Pg p = pages[i];
+        p.dst_off += base_dst * r;
+        pages.push_back(p);
+        names.push_back(names[i]);
+      }
+    dst_off *= rep;
+  }
   int n = (int)pages.size();
-  printf("%d pages\n", n);
+  printf("%d pages (rep=%d)\n", n, rep);
   uint8_t *d_raw, *d_dec;
   Pg* d_pages; uint64_t* d_cyc; int* d_err;
   hipMalloc(&d_raw, rawbuf.size()); hipMemcpy(d_raw, rawbuf.data(), rawbuf.size(), hipMemcpyHostToDevice);
   hipMalloc(&d_dec, dst_off);
   hipMalloc(&d_pages, sizeof(Pg) * n); hipMemcpy(d_pages, pages.data(), sizeof(Pg) * n, hipMemcpyHostToDevice);
-  hipMalloc(&d_cyc, 8 * n); hipMalloc(&d_err, 4); hipMemset(d_err, 0, 4);
-  // replicate the page set to fill the chip (measure contention-free too)
-  hipLaunchKernelGGL(k_lz4, dim3(n), dim3(WAVE), 0, 0, d_raw, d_dec, d_pages, n, d_cyc, d_err);
+  hipMalloc(&d_cyc, 8 * n * 6); hipMalloc(&d_err, 4); hipMemset(d_err, 0, 4);
+  hipLaunchKernelGGL(k_lz4, dim3(n), dim3(WAVE), 0, 0, d_raw, d_dec, d_pages, n, d_cyc, d_err);  // warm
   hipDeviceSynchronize();
-  std::vector<uint64_t> cyc(n);
-  hipMemcpy(cyc.data(), d_cyc, 8 * n, hipMemcpyDeviceToHost);
+  hipEvent_t ea, eb; hipEventCreate(&ea); hipEventCreate(&eb);
+  hipEventRecord(ea);
+  hipLaunchKernelGGL(k_lz4, dim3(n), dim3(WAVE), 0, 0, d_raw, d_dec, d_pages, n, d_cyc, d_err);
+  hipEventRecord(eb); hipEventSynchronize(eb);
+  float wall; hipEventElapsedTime(&wall, ea, eb);
+  uint64_t tot_uncomp = 0; for (auto& p : pages) tot_uncomp += p.uncomp;
+  printf("WALL %.3f ms for %.1f MB uncomp -> %.1f GB/s aggregate\n", wall, tot_uncomp/1e6, tot_uncomp/(wall/1e3)/1e9);
+  std::vector<uint64_t> cyc(n * 6);
+  hipMemcpy(cyc.data(), d_cyc, 8 * n * 6, hipMemcpyDeviceToHost);
   int err = 0; hipMemcpy(&err, d_err, 4, hipMemcpyDeviceToHost);
   printf("err=%d\n", err);
-  for (int i = 0; i < n; i++) {
-    double us = cyc[i] / 2.4e3;  // shader clock ~2.4GHz
-    printf("%-14s comp=%8u uncomp=%8u cycles=%10lu (%8.1f us, %6.2f GB/s out)\n",
-           names[i].c_str(), pages[i].comp, pages[i].uncomp,
-           (unsigned long)cyc[i], us, pages[i].uncomp / (us * 1e3));
+  std::vector<uint64_t> sorted_c(cyc);
+  std::sort(sorted_c.begin(), sorted_c.end());
+  printf("page cycles p50=%lu p90=%lu p99=%lu max=%lu\n",
+         (unsigned long)sorted_c[n/2], (unsigned long)sorted_c[(int)(n*0.9)],
+         (unsigned long)sorted_c[(int)(n*0.99)], (unsigned long)sorted_c[n-1]);
+  if (rep == 1) {
+    std::vector<int> order(n);
+    for (int i = 0; i < n; i++) order[i] = i;
+    std::sort(order.begin(), order.end(), [&](int a, int b) { return cyc[a] > cyc[b]; });
+    for (int oi = 0; oi < n && oi < 20; oi++) {
+      int i = order[oi];
+      double us = cyc[i] / 2.4e3;
+      printf("%-14s comp=%8u uncomp=%8u cycles=%10lu (%8.1f us) refill=%lu parse=%lu lit=%lu match=%lu nseq=%lu cyc/seq=%lu\n",
+             names[i].c_str(), pages[i].comp, pages[i].uncomp,
+             (unsigned long)cyc[i], us,
+             (unsigned long)cyc[n + i * 5 + 0], (unsigned long)cyc[n + i * 5 + 1],
+             (unsigned long)cyc[n + i * 5 + 2], (unsigned long)cyc[n + i * 5 + 3],
+             (unsigned long)cyc[n + i * 5 + 4],
+             (unsigned long)(cyc[i] / (cyc[n + i * 5 + 4] ? cyc[n + i * 5 + 4] : 1)));
+    }
   }
   return 0;
 }
