@@ -166,15 +166,29 @@ def main():
     raw_b = rank_bytes  # compressed chunk bytes resident in HBM, read once/step
     # decompressed arena bytes: est = hbm_bytes_est - raw - dec (we stored raw+2*dec)
     dec_b = (m_after["hbm_bytes_est"] - raw_b) / 2
-    # dominant kernel: the LZ4 page-decompression sweep (algorithmic bytes =
-    # compressed read + decompressed write per launch; one launch per step)
+    # Algorithmic bytes per launch, per workload (constants documented in
+    # DESIGN.md §3/§5):
+    #  - lz4 sweep: compressed read + decompressed write
+    #  - c1 fused dict-count: decompressed pages read once (no materialization)
+    #  - c2s: dec read + ts/latency i64 writes+reads (8B x2 x2 cols) + gid
+    #    (4B x2) + valid (1x3) + mask (1x3 rw)
+    #  - c3s: dec read (values scanned from LDS windows) + mask write
     decomp_algo_bytes = raw_b + dec_b
     other_ns = max(dk - dd, 1.0)
-    decode_algo_bytes = dec_b + 9.0 * rank_rows  # dec read + gid(4)+idx reads + mask ~ coarse
+    decode_algo_bytes = {
+        "c1": dec_b,
+        "c2s": dec_b + rank_rows * (2 * 16 + 2 * 4 + 3 + 3),
+        "c3s": dec_b + rank_rows * 1,
+    }[args.workload]
+    decode_kernel = {
+        "c1": "dict_count_fused",
+        "c2s": "decode+filter+groupby",
+        "c3s": "bytes_contains(LIKE)",
+    }[args.workload]
     if dd >= other_ns:
         roof_kernel, roof_bytes, roof_ns = "lz4_page_decompress", decomp_algo_bytes, dd
     else:
-        roof_kernel, roof_bytes, roof_ns = "decode+filter+groupby", decode_algo_bytes, other_ns
+        roof_kernel, roof_bytes, roof_ns = decode_kernel, decode_algo_bytes, other_ns
     achieved_gbs = roof_bytes / max(roof_ns, 1.0)  # bytes/ns == GB/s
     roofline = {
         "bound": "hbm",
