@@ -190,6 +190,15 @@ def main():
     else:
         roof_kernel, roof_bytes, roof_ns = decode_kernel, decode_algo_bytes, other_ns
     achieved_gbs = roof_bytes / max(roof_ns, 1.0)  # bytes/ns == GB/s
+    traffic = None  # measured HBM bytes/launch from rocprofv3 --pmc passes
+    try:
+        with open(os.path.join(ROOT, "profiles", "pmc_traffic.json")) as fh:
+            t = json.load(fh)["traffic"].get(args.workload, {})
+        traffic = t.get(roof_kernel)
+        if traffic is not None:
+            traffic = round(traffic)
+    except Exception:
+        pass
     roofline = {
         "bound": "hbm",
         "kernel": roof_kernel,
@@ -197,7 +206,8 @@ def main():
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
         "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-        "traffic": None,  # filled from rocprofv3 --pmc runs (profiles/)
+        "traffic": traffic,  # rocprofv3 --pmc (profiles/r01_pmc_summary.md);
+                             # measured on the same workload at --rows 1e8/3e7
     }
 
     # --- CPU baseline: the oracle (kind=port) on a bounded sample ---
