@@ -162,7 +162,23 @@ def execute(files: list[str], query: dict) -> dict:
                 pa.types.is_dictionary(col.type)
                 and pa.types.is_string(col.type.value_type)
             ):
-                raise NotImplementedError("utf8 min/max: next rows (SURVEY §8f)")
+                # utf8 min/max: lexicographic byte order (DataFusion semantics)
+                if op not in ("min", "max"):
+                    raise NotImplementedError(f"{op} over utf8")
+                scol = col.cast(pa.string()) if pa.types.is_dictionary(col.type) else col
+                vals = scol.to_pylist()
+                out = [None] * G
+                cnt = np.zeros(G, dtype=np.int64)
+                for j, i in enumerate(sel):
+                    if not vsel[j]:
+                        continue
+                    v = vals[i]
+                    g = inv[j]
+                    cnt[g] += 1
+                    if out[g] is None or (v < out[g] if op == "min" else v > out[g]):
+                        out[g] = v
+                file_res.append((op, out, cnt))
+                continue
             vals = np.asarray(
                 col.cast(col.type.value_type) if pa.types.is_dictionary(col.type) else col
             )
@@ -205,7 +221,8 @@ def execute(files: list[str], query: dict) -> dict:
                     present = int(r[2][gidx]) > 0
                     if not present:
                         continue
-                    v = v.item()
+                    if hasattr(v, "item"):
+                        v = v.item()
                     if st[ai] is None:
                         st[ai] = v
                     elif kind == "sum":

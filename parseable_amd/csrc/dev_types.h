@@ -26,7 +26,8 @@ struct DevPage {
 enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };
 
 enum { AGGK_COUNT_STAR = 0, AGGK_COUNT, AGGK_SUM_I64, AGGK_SUM_F64,
-       AGGK_MIN_I64, AGGK_MAX_I64, AGGK_MIN_F64, AGGK_MAX_F64 };
+       AGGK_MIN_I64, AGGK_MAX_I64, AGGK_MIN_F64, AGGK_MAX_F64,
+       AGGK_MIN_RANK, AGGK_MAX_RANK };  // utf8 min/max via dict sort-ranks
 
 constexpr int MAX_KEYS = 4;
 constexpr int MAX_AGGS = 8;

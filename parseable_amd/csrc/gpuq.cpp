@@ -77,6 +77,8 @@ struct ColPlan {
   std::vector<int> lut_preds;      // preds evaluated per-dict-entry (utf8)
   std::vector<int> cmp_preds;      // preds on the decoded i64 array
   std::vector<int> contains_preds; // CONTAINS on byte_array pages
+  bool need_rank = false;   // utf8 min/max: values decoded as dict sort-ranks
+  std::vector<int32_t> rank_to_gid;  // rank -> gid (1-based), for export
   // global dictionary (need_gid): gid 1.. ; 0 = NULL
   std::vector<std::string> gdict;
   std::unordered_map<std::string, int32_t> gmap;
@@ -106,6 +108,7 @@ struct AggPlan {
 // one column chunk of one selected row group
 struct ChunkTask {
   int file_idx, rg_idx, col_idx;
+  size_t dictv_pool_base = (size_t)-1;  // for the utf8-rank post-pass
   uint64_t raw_off = 0;            // into partition d_raw
   const ColumnChunkMeta* cm = nullptr;
   std::vector<PageInfo> pages;
@@ -361,8 +364,13 @@ extern "C" gpuq_plan* gpuq_plan_build(
         throw std::runtime_error("int column needs int literal: " + pp.col);
       c.cmp_preds.push_back((int)pi);
       c.need_val = true;
+    } else if (c.phys == PT_DOUBLE) {
+      if (pp.p.lit_kind != GPUQ_LIT_F64)
+        throw std::runtime_error("double column needs float literal: " + pp.col);
+      c.cmp_preds.push_back((int)pi);
+      c.need_val = true;
     } else {
-      throw std::runtime_error("predicates on f64 columns: next row (SURVEY §8f)");
+      throw std::runtime_error("unsupported predicate column type: " + pp.col);
     }
   }
   // aggregate kinds
@@ -370,11 +378,16 @@ extern "C" gpuq_plan* gpuq_plan_build(
     if (ap.op == GPUQ_AGG_COUNT_STAR) { ap.kind = AGGK_COUNT_STAR; continue; }
     auto& c = plan->cols[ap.col_idx];
     ap.is_f64 = (c.phys == PT_DOUBLE || c.phys == PT_FLOAT);
-    if (c.phys == PT_BYTE_ARRAY && ap.op != GPUQ_AGG_COUNT)
-      throw std::runtime_error("utf8 min/max: next row (SURVEY §8f)");
-    if (c.phys == PT_BYTE_ARRAY) {          // COUNT(utf8): validity only
-      c.need_gid = true;
-      c.need_gid_valid = true;
+    if (c.phys == PT_BYTE_ARRAY && ap.op == GPUQ_AGG_SUM)
+      throw std::runtime_error("sum over utf8 column: " + ap.col);
+    if (c.phys == PT_BYTE_ARRAY) {
+      if (ap.op == GPUQ_AGG_COUNT) {        // COUNT(utf8): validity only
+        c.need_gid = true;
+        c.need_gid_valid = true;
+      } else {                              // utf8 min/max via dict sort-ranks
+        c.need_val = true;
+        c.need_rank = true;
+      }
     } else {
       c.need_val = true;
     }
@@ -382,11 +395,13 @@ extern "C" gpuq_plan* gpuq_plan_build(
       case GPUQ_AGG_COUNT: ap.kind = AGGK_COUNT; break;
       case GPUQ_AGG_SUM: ap.kind = ap.is_f64 ? AGGK_SUM_F64 : AGGK_SUM_I64; break;
       case GPUQ_AGG_MIN:
-        if (ap.is_f64) throw std::runtime_error("f64 min/max: next row");
-        ap.kind = AGGK_MIN_I64; break;
+        ap.kind = (c.phys == PT_BYTE_ARRAY) ? AGGK_MIN_RANK
+                  : ap.is_f64 ? AGGK_MIN_F64 : AGGK_MIN_I64;
+        break;
       case GPUQ_AGG_MAX:
-        if (ap.is_f64) throw std::runtime_error("f64 min/max: next row");
-        ap.kind = AGGK_MAX_I64; break;
+        ap.kind = (c.phys == PT_BYTE_ARRAY) ? AGGK_MAX_RANK
+                  : ap.is_f64 ? AGGK_MAX_F64 : AGGK_MAX_I64;
+        break;
       default: throw std::runtime_error("bad agg op");
     }
   }
@@ -485,8 +500,13 @@ extern "C" gpuq_plan* gpuq_plan_build(
             t.lut.reserve(pi.num_values);
             for (int32_t k = 0; k < pi.num_values; k++) {
               uint32_t l; memcpy(&l, q, 4); q += 4;
+              int32_t g = 0;
+              if (c.need_gid || c.need_rank)
+                g = c.gid_of(std::string((const char*)q, l));
               if (c.need_gid)
-                t.remap.push_back(c.gid_of(std::string((const char*)q, l)));
+                t.remap.push_back(g);
+              if (c.need_rank)
+                t.dictv.push_back(g);  // post-pass rewrites gid -> sort rank
               if (!c.lut_preds.empty()) {
                 uint8_t ok = 1;
                 for (int pidx : c.lut_preds) {
@@ -513,9 +533,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
         for (auto& pi : t.pages)
           if (pi.type == PAGE_DATA && pi.encoding == ENC_PLAIN)
             t.has_plain_data_pages = true;
-        if (t.has_plain_data_pages && c.need_gid)
-          throw std::runtime_error("group key with PLAIN fallback pages (high-cardinality "
-                                   "key hashing): next row (SURVEY §8f) — " + c.name);
+        if (t.has_plain_data_pages && (c.need_gid || c.need_rank))
+          throw std::runtime_error("dict-only utf8 operation with PLAIN fallback pages "
+                                   "(high-cardinality hashing): next row (SURVEY §8f) — " + c.name);
         part.chunks.push_back(std::move(t));
       }
     }
@@ -525,6 +545,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       auto& c = plan->cols[t.col_idx];
       uint32_t remap_base = (uint32_t)part.remap_pool.size();
       uint32_t dictv_base = (uint32_t)part.dictv_pool.size();
+      t.dictv_pool_base = dictv_base;
       uint32_t lut_base = (uint32_t)part.lut_pool.size();
       part.remap_pool.insert(part.remap_pool.end(), t.remap.begin(), t.remap.end());
       part.dictv_pool.insert(part.dictv_pool.end(), t.dictv.begin(), t.dictv.end());
@@ -603,6 +624,31 @@ extern "C" gpuq_plan* gpuq_plan_build(
 
   for (auto& part : plan->parts)
     plan->m_rows_scanned += part.n_rows;
+
+  // utf8 min/max post-pass: with every chunk registered, the global dict is
+  // final — compute each rank column's lexicographic sort ranks and rewrite
+  // its per-chunk dictionary-value pools from gid to rank.
+  for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+    auto& c = plan->cols[ci];
+    if (!c.need_rank) continue;
+    std::vector<int32_t> order((size_t)c.gdict.size());
+    for (size_t i = 0; i < order.size(); i++) order[i] = (int32_t)i;
+    std::sort(order.begin(), order.end(), [&](int32_t a, int32_t b) {
+      return c.gdict[a] < c.gdict[b];
+    });
+    std::vector<int64_t> rank_of_gid(c.gdict.size() + 1, 0);
+    c.rank_to_gid.assign(c.gdict.size(), 0);
+    for (size_t r = 0; r < order.size(); r++) {
+      rank_of_gid[(size_t)order[r] + 1] = (int64_t)r;
+      c.rank_to_gid[r] = order[r] + 1;
+    }
+    for (auto& part : plan->parts)
+      for (auto& t : part.chunks)
+        if (t.col_idx == (int)ci && t.dictv_pool_base != (size_t)-1)
+          for (size_t k = 0; k < t.dictv.size(); k++)
+            part.dictv_pool[t.dictv_pool_base + k] =
+                rank_of_gid[(size_t)part.dictv_pool[t.dictv_pool_base + k]];
+  }
 
   // fused count path: GROUP BY <one dict col>, count(*)-only, no predicates
   {
@@ -900,8 +946,14 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         case GPUQ_BETWEEN: mode = CMP_RANGE; break;
         default: throw std::runtime_error("bad cmp op");
       }
+      int is_f64 = (c.phys == PT_DOUBLE) ? 1 : 0;
+      int64_t lo = pp.p.i64[0], hi = pp.p.i64[1];
+      if (is_f64) {
+        memcpy(&lo, &pp.p.f64[0], 8);
+        memcpy(&hi, &pp.p.f64[1], 8);
+      }
       launch_cmp_i64(st, part.d_val[(int)ci], part.d_valid[(int)ci],
-                     pp.p.i64[0], pp.p.i64[1], mode, pp.p.hi_exclusive,
+                     lo, hi, mode, pp.p.hi_exclusive, is_f64,
                      part.d_mask, part.n_rows);
     }
   }
@@ -983,9 +1035,11 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   f++;
   for (int i = 0; i < n_aggs; i++) {
     ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
-    make_schema_field(ss->schema.children[f],
-                      plan->aggs[i].kind == AGGK_SUM_F64 ? "g" : "l",
-                      "agg" + std::to_string(i));
+    int ki = plan->aggs[i].kind;
+    const char* fmt = (ki == AGGK_SUM_F64 || ki == AGGK_MIN_F64 || ki == AGGK_MAX_F64)
+                          ? "g"
+                      : (ki == AGGK_MIN_RANK || ki == AGGK_MAX_RANK) ? "u" : "l";
+    make_schema_field(ss->schema.children[f], fmt, "agg" + std::to_string(i));
     f++;
     ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
     make_schema_field(ss->schema.children[f], "l", "agg" + std::to_string(i) + "_count");
@@ -1070,17 +1124,62 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // aggs
   for (int i = 0; i < n_aggs; i++) {
     auto* chv = make_child(f++);
-    chv->n_buffers = 2;
-    chv->buffers = (const void**)calloc(2, sizeof(void*));
-    int64_t* vv = (int64_t*)eb->grab(nr * 8);
+    int kind = plan->aggs[i].kind;
+    bool is_rank = (kind == AGGK_MIN_RANK || kind == AGGK_MAX_RANK);
     uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
     memset(validity, 0xff, (nr + 7) / 8);
     int64_t nulls = 0;
+    if (is_rank) {
+      // utf8 value column: rank -> dictionary string
+      const auto& c = plan->cols[plan->aggs[i].col_idx];
+      chv->n_buffers = 3;
+      chv->buffers = (const void**)calloc(3, sizeof(void*));
+      int32_t* offs = (int32_t*)eb->grab((nr + 1) * 4);
+      size_t total = 0;
+      std::vector<const std::string*> strs(nr, nullptr);
+      for (int64_t r = 0; r < nr; r++) {
+        size_t base = (size_t)live[r] * slots;
+        uint64_t cnt = empty_aggregate_row ? 0 : table[base + 2 + 2 * i];
+        if (cnt) {
+          int64_t rankv = (int64_t)table[base + 1 + 2 * i];
+          if (rankv >= 0 && rankv < (int64_t)c.rank_to_gid.size()) {
+            strs[r] = &c.gdict[(size_t)c.rank_to_gid[(size_t)rankv] - 1];
+            total += strs[r]->size();
+          }
+        }
+      }
+      char* sdata = (char*)eb->grab(total);
+      size_t off = 0;
+      for (int64_t r = 0; r < nr; r++) {
+        offs[r] = (int32_t)off;
+        if (strs[r]) {
+          memcpy(sdata + off, strs[r]->data(), strs[r]->size());
+          off += strs[r]->size();
+        } else {
+          validity[r / 8] &= (uint8_t)~(1 << (r % 8));
+          nulls++;
+        }
+      }
+      offs[nr] = (int32_t)off;
+      if (nulls) { chv->buffers[0] = validity; chv->null_count = nulls; }
+      chv->buffers[1] = offs;
+      chv->buffers[2] = sdata;
+      auto* chc2 = make_child(f++);
+      chc2->n_buffers = 2;
+      chc2->buffers = (const void**)calloc(2, sizeof(void*));
+      int64_t* cv2 = (int64_t*)eb->grab(nr * 8);
+      for (int64_t r = 0; r < nr; r++)
+        cv2[r] = empty_aggregate_row ? 0 : (int64_t)table[(size_t)live[r] * slots + 2 + 2 * i];
+      chc2->buffers[1] = cv2;
+      continue;
+    }
+    chv->n_buffers = 2;
+    chv->buffers = (const void**)calloc(2, sizeof(void*));
+    int64_t* vv = (int64_t*)eb->grab(nr * 8);
     for (int64_t r = 0; r < nr; r++) {
       size_t base = (size_t)live[r] * slots;
       uint64_t cnt = empty_aggregate_row ? 0 : table[base + 2 + 2 * i];
       uint64_t val = empty_aggregate_row ? 0 : table[base + 1 + 2 * i];
-      int kind = plan->aggs[i].kind;
       if (kind == AGGK_COUNT_STAR || kind == AGGK_COUNT) {
         vv[r] = (int64_t)cnt;
       } else if (cnt == 0) {
@@ -1092,8 +1191,6 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       }
     }
     if (nulls) {
-      chv->n_buffers = 2;
-      chv->buffers = (const void**)realloc(chv->buffers, 2 * sizeof(void*));
       chv->buffers[0] = validity;
       chv->null_count = nulls;
     }

@@ -1016,21 +1016,36 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
 __global__ void k_cmp_i64(const int64_t* __restrict__ col,
                           const uint8_t* __restrict__ valid,
                           int64_t lo, int64_t hi, int mode, int hi_exclusive,
-                          uint8_t* __restrict__ mask, int64_t n) {
+                          int is_f64, uint8_t* __restrict__ mask, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     uint8_t ok = valid ? valid[i] : 1;
     if (ok) {
-      int64_t v = col[i];
-      switch (mode) {
-        case CMP_EQ: ok = (v == lo); break;
-        case CMP_NE: ok = (v != lo); break;
-        case CMP_LT: ok = (v < lo); break;
-        case CMP_LE: ok = (v <= lo); break;
-        case CMP_GT: ok = (v > lo); break;
-        case CMP_GE: ok = (v >= lo); break;
-        case CMP_RANGE: ok = (v >= lo) && (hi_exclusive ? (v < hi) : (v <= hi)); break;
+      if (is_f64) {
+        double v = __longlong_as_double((long long)col[i]);
+        double flo = __longlong_as_double((long long)lo);
+        double fhi = __longlong_as_double((long long)hi);
+        switch (mode) {
+          case CMP_EQ: ok = (v == flo); break;
+          case CMP_NE: ok = (v != flo); break;
+          case CMP_LT: ok = (v < flo); break;
+          case CMP_LE: ok = (v <= flo); break;
+          case CMP_GT: ok = (v > flo); break;
+          case CMP_GE: ok = (v >= flo); break;
+          case CMP_RANGE: ok = (v >= flo) && (hi_exclusive ? (v < fhi) : (v <= fhi)); break;
+        }
+      } else {
+        int64_t v = col[i];
+        switch (mode) {
+          case CMP_EQ: ok = (v == lo); break;
+          case CMP_NE: ok = (v != lo); break;
+          case CMP_LT: ok = (v < lo); break;
+          case CMP_LE: ok = (v <= lo); break;
+          case CMP_GT: ok = (v > lo); break;
+          case CMP_GE: ok = (v >= lo); break;
+          case CMP_RANGE: ok = (v >= lo) && (hi_exclusive ? (v < hi) : (v <= hi)); break;
+        }
       }
     }
     mask[i] &= ok;
@@ -1057,6 +1072,28 @@ __device__ inline void atomic_max_i64(uint64_t* addr, int64_t val) {
                               (unsigned long long)val);
     if ((int64_t)prev == old) break;
     old = (int64_t)prev;
+  }
+}
+__device__ inline void atomic_min_f64(uint64_t* addr, double val) {
+  uint64_t old = *addr;
+  for (;;) {
+    double cur = __longlong_as_double((long long)old);
+    if (!(val < cur)) break;
+    uint64_t prev = atomicCAS((unsigned long long*)addr, (unsigned long long)old,
+                              (unsigned long long)__double_as_longlong(val));
+    if (prev == old) break;
+    old = prev;
+  }
+}
+__device__ inline void atomic_max_f64(uint64_t* addr, double val) {
+  uint64_t old = *addr;
+  for (;;) {
+    double cur = __longlong_as_double((long long)old);
+    if (!(val > cur)) break;
+    uint64_t prev = atomicCAS((unsigned long long*)addr, (unsigned long long)old,
+                              (unsigned long long)__double_as_longlong(val));
+    if (prev == old) break;
+    old = prev;
   }
 }
 __device__ inline void atomic_add_f64(uint64_t* addr, double val) {
@@ -1096,8 +1133,10 @@ k_agg(AggArgs a) {
       if (s > 0 && ((s - 1) & 1) == 0) {  // value slot
         int ai = (s - 1) / 2;
         int k = a.agg_kind[ai];
-        if (k == AGGK_MIN_I64) init = (uint64_t)INT64_MAX;
-        else if (k == AGGK_MAX_I64) init = (uint64_t)INT64_MIN;
+        if (k == AGGK_MIN_I64 || k == AGGK_MIN_RANK) init = (uint64_t)INT64_MAX;
+        else if (k == AGGK_MAX_I64 || k == AGGK_MAX_RANK) init = (uint64_t)INT64_MIN;
+        else if (k == AGGK_MIN_F64) init = (uint64_t)0x7ff0000000000000ull;   // +inf
+        else if (k == AGGK_MAX_F64) init = (uint64_t)0xfff0000000000000ull;   // -inf
       }
       tab[i] = init;
     }
@@ -1131,8 +1170,10 @@ k_agg(AggArgs a) {
       switch (k) {
         case AGGK_SUM_I64: atomicAdd((unsigned long long*)vs, (unsigned long long)v); break;
         case AGGK_SUM_F64: atomic_add_f64(vs, __longlong_as_double((long long)v)); break;
-        case AGGK_MIN_I64: atomic_min_i64(vs, v); break;
-        case AGGK_MAX_I64: atomic_max_i64(vs, v); break;
+        case AGGK_MIN_I64: case AGGK_MIN_RANK: atomic_min_i64(vs, v); break;
+        case AGGK_MAX_I64: case AGGK_MAX_RANK: atomic_max_i64(vs, v); break;
+        case AGGK_MIN_F64: atomic_min_f64(vs, __longlong_as_double((long long)v)); break;
+        case AGGK_MAX_F64: atomic_max_f64(vs, __longlong_as_double((long long)v)); break;
       }
       atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
     }
@@ -1158,10 +1199,20 @@ k_agg(AggArgs a) {
           if (d != 0.0) atomic_add_f64(g, d);
           break;
         }
-        case AGGK_MIN_I64:
+        case AGGK_MIN_I64: case AGGK_MIN_RANK:
           if ((int64_t)v != INT64_MAX) atomic_min_i64(g, (int64_t)v); break;
-        case AGGK_MAX_I64:
+        case AGGK_MAX_I64: case AGGK_MAX_RANK:
           if ((int64_t)v != INT64_MIN) atomic_max_i64(g, (int64_t)v); break;
+        case AGGK_MIN_F64: {
+          double d = __longlong_as_double((long long)v);
+          if (d != __builtin_inf()) atomic_min_f64(g, d);
+          break;
+        }
+        case AGGK_MAX_F64: {
+          double d = __longlong_as_double((long long)v);
+          if (d != -__builtin_inf()) atomic_max_f64(g, d);
+          break;
+        }
       }
     }
   }
@@ -1178,8 +1229,10 @@ __global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
     uint64_t init = 0;
     if (s > 0 && ((s - 1) & 1) == 0) {
       int k = agg_kind[(s - 1) / 2];
-      if (k == AGGK_MIN_I64) init = (uint64_t)INT64_MAX;
-      else if (k == AGGK_MAX_I64) init = (uint64_t)INT64_MIN;
+      if (k == AGGK_MIN_I64 || k == AGGK_MIN_RANK) init = (uint64_t)INT64_MAX;
+      else if (k == AGGK_MAX_I64 || k == AGGK_MAX_RANK) init = (uint64_t)INT64_MIN;
+      else if (k == AGGK_MIN_F64) init = 0x7ff0000000000000ull;
+      else if (k == AGGK_MAX_F64) init = 0xfff0000000000000ull;
     }
     table[i] = init;
   }
@@ -1229,11 +1282,11 @@ void launch_bytes_contains(hipStream_t st, const uint8_t* dec, const DevPage* pa
   if (n) hipLaunchKernelGGL(k_bytes_contains, dim3(n), dim3(CTHREADS), 0, st, dec, pages, ids, n, needle, nlen, mask, d_err);
 }
 void launch_cmp_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
-                    int64_t lo, int64_t hi, int mode, int hi_excl,
+                    int64_t lo, int64_t hi, int mode, int hi_excl, int is_f64,
                     uint8_t* mask, int64_t n) {
   int blocks = (int)((n + 255) / 256);
   if (blocks > 4096) blocks = 4096;
-  if (n) hipLaunchKernelGGL(k_cmp_i64, dim3(blocks), dim3(256), 0, st, col, valid, lo, hi, mode, hi_excl, mask, n);
+  if (n) hipLaunchKernelGGL(k_cmp_i64, dim3(blocks), dim3(256), 0, st, col, valid, lo, hi, mode, hi_excl, is_f64, mask, n);
 }
 void launch_init_table(hipStream_t st, uint64_t* table, int32_t n_groups,
                        int n_aggs, const int32_t* d_agg_kind) {

@@ -195,6 +195,7 @@ class StandardTableProvider:
     def scan(self, query: dict):
         """query: the IR of tests/golden_queries.py (select/group_by/preds/
         time_range). Returns GpuExecutionPlan or ManifestCountResult."""
+        query = {k: v for k, v in query.items() if k != "ext"}
         preds = list(query.get("preds", []))
         time_range = query.get("time_range")
         files = self._manifest_files(time_range)

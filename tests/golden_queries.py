@@ -61,6 +61,24 @@ GOLDEN_QUERIES = {
             "group_by": ["level"],
             "preds": [{"col": "level", "op": "ge", "lit": "INFO"}],
         }),
+        # extended coverage (ext: skipped by the scalar C oracle — pinned by
+        # the pyarrow oracle + Acero instead)
+        ("minmax_utf8", {
+            "ext": True,
+            "select": [{"agg": "min", "col": "host"}, {"agg": "max", "col": "host"},
+                       {"agg": "count_star"}],
+            "group_by": ["level"],
+        }),
+        ("minmax_f64", {
+            "ext": True,
+            "select": [{"agg": "min", "col": "f_f64"}, {"agg": "max", "col": "f_f64"}],
+            "group_by": ["f_str2"],
+        }),
+        ("pred_f64", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+            "preds": [{"col": "f_f64", "op": "lt", "lit": 0.25}],
+        }),
     ],
     "g_c1_pages": [
         ("count_by_level", {

@@ -12,7 +12,10 @@ from tests.golden_queries import GOLDEN_QUERIES
 
 
 def _all_cases():
-    return [f"{fx}/{q}" for fx, qs in GOLDEN_QUERIES.items() for q, _ in qs]
+    # ext queries use features the scalar C restatement does not cover
+    # (utf8/f64 min-max, f64 predicates) — pinned by the pyarrow oracle
+    return [f"{fx}/{q}" for fx, qs in GOLDEN_QUERIES.items()
+            for q, spec in qs if not spec.get("ext")]
 
 
 @pytest.fixture(scope="session", autouse=True)
