@@ -129,6 +129,23 @@ gpuq_plan* gpuq_plan_build(gpuq_ctx*,
     const gpuq_agg* aggs, int32_t n_aggs,
     int64_t limit);
 
+/* Plan straight from Parseable's catalog metadata:
+ * parses <stream_dir>/stream.json (ObjectStoreFormat snapshot,
+ * storage/mod.rs:335-380) and the daily manifest JSON
+ * (catalog/manifest.rs:143-157), selects manifests by the time window
+ * (Snapshot::manifests, catalog/snapshot.rs:42-71), prunes files by
+ * per-column min/max TypedStatistics (can_be_pruned/satisfy_constraints,
+ * stream_schema_provider.rs:1049-1137), and answers bare count(*) from
+ * manifest num_rows sums (query.rs:189-256). The injected time range is
+ * passed as a hi_exclusive BETWEEN on p_timestamp among `preds`.
+ * *fast_count: >=0 manifest-answered count (no plan); -1 plan returned
+ * (or error: check return + gpuq_last_error); -2 empty relation. */
+gpuq_plan* gpuq_plan_build_from_stream(gpuq_ctx*, const char* stream_dir,
+    const gpuq_pred* preds, int32_t n_preds,
+    const char* const* group_by, int32_t n_group_by,
+    const gpuq_agg* aggs, int32_t n_aggs, int64_t limit,
+    int64_t* fast_count);
+
 /* Independent output partitions, one per selected device (the byte-balanced
  * row-group shards of balanced_file_groups, stream_schema_provider.rs:146-165,
  * collapsed to one stream per GPU). */

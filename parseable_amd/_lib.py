@@ -73,6 +73,14 @@ def load():
     lib.gpuq_last_error.restype = C.c_char_p
     lib.gpuq_last_error.argtypes = [C.c_void_p]
     lib.gpuq_device_count.restype = C.c_int32
+    lib.gpuq_plan_build_from_stream.restype = C.c_void_p
+    lib.gpuq_plan_build_from_stream.argtypes = [
+        C.c_void_p, C.c_char_p,
+        C.POINTER(GpuqPred), C.c_int32,
+        C.POINTER(C.c_char_p), C.c_int32,
+        C.POINTER(GpuqAgg), C.c_int32,
+        C.c_int64, C.POINTER(C.c_int64),
+    ]
     lib.gpuq_plan_build.restype = C.c_void_p
     lib.gpuq_plan_build.argtypes = [
         C.c_void_p,

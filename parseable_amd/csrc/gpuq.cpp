@@ -10,6 +10,7 @@
 // (SURVEY.md §3a step 7), with the per-row work on the GPU and the
 // Partial->Final merge left to the caller (one RCCL reduce across GPUs).
 #include "../../include/gpuq.h"
+#include "catalog.h"
 #include "meta.h"
 #include "kernels_api.h"
 #include "dev_types.h"
@@ -659,6 +660,43 @@ extern "C" gpuq_plan* gpuq_plan_build(
   }
 
   return plan.release();
+} catch (const std::exception& e) {
+  if (ctx) ctx->set_error(e.what());
+  return nullptr;
+}
+
+// §8f row 1: native catalog planner — plan straight from Parseable's
+// stream.json/manifest.json (catalog.cpp). fast_count out-param:
+//   >= 0  answered from manifest num_rows sums (no plan returned)
+//   -1    a scan plan was built (or an error occurred: check return)
+//   -2    every file pruned: empty relation, no plan
+extern "C" gpuq_plan* gpuq_plan_build_from_stream(
+    gpuq_ctx* ctx, const char* stream_dir,
+    const gpuq_pred* preds, int32_t n_preds,
+    const char* const* group_by, int32_t n_group_by,
+    const gpuq_agg* aggs, int32_t n_aggs, int64_t limit,
+    int64_t* fast_count) try {
+  if (fast_count) *fast_count = -1;
+  bool bare = (n_aggs == 1 && aggs && aggs[0].op == GPUQ_AGG_COUNT_STAR &&
+               n_group_by == 0);
+  CatalogPlanInput in = catalog_plan(stream_dir, preds, n_preds, bare);
+  if (in.fast_count >= 0) {
+    if (fast_count) *fast_count = in.fast_count;
+    return nullptr;
+  }
+  if (in.files.empty()) {
+    if (fast_count) *fast_count = -2;
+    return nullptr;
+  }
+  std::vector<gpuq_file> files(in.files.size());
+  for (size_t i = 0; i < in.files.size(); i++) {
+    files[i].path = in.files[i].c_str();
+    files[i].row_groups = nullptr;
+    files[i].n_row_groups = -1;
+  }
+  return gpuq_plan_build(ctx, files.data(), (int32_t)files.size(), nullptr, 0,
+                         preds, n_preds, group_by, n_group_by, aggs, n_aggs,
+                         limit);
 } catch (const std::exception& e) {
   if (ctx) ctx->set_error(e.what());
   return nullptr;

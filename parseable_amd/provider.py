@@ -196,6 +196,17 @@ class StandardTableProvider:
         """query: the IR of tests/golden_queries.py (select/group_by/preds/
         time_range). Returns GpuExecutionPlan or ManifestCountResult."""
         query = {k: v for k, v in query.items() if k != "ext"}
+
+        if self.session is not None and not os.environ.get("GPUQ_PY_PLANNER"):
+            # native catalog planner inside libgpuq (catalog.cpp)
+            plan = GpuExecutionPlan(self.session, None, query,
+                                    stream_dir=self.stream_dir)
+            if plan.fast_count is not None:
+                return ManifestCountResult(plan.fast_count)
+            if plan.empty:
+                return EmptyScanResult(query)
+            return plan
+
         preds = list(query.get("preds", []))
         time_range = query.get("time_range")
         files = self._manifest_files(time_range)
@@ -249,14 +260,84 @@ class StandardTableProvider:
         return GpuExecutionPlan(self.session, paths, query)
 
 
+def _build_c_query(query: dict, keep):
+    """query IR -> (cpreds, n_preds, cgroup, n_group, caggs, n_aggs).
+    `keep` collects byte strings that must outlive the C call."""
+    preds = list(query.get("preds", []))
+    tr = query.get("time_range")
+    if tr:
+        preds.append({"col": "p_timestamp", "op": "between",
+                      "lo": tr[0], "hi": tr[1], "hi_exclusive": True})
+    cpreds = (GpuqPred * max(len(preds), 1))()
+    for i, p in enumerate(preds):
+        b = p["col"].encode()
+        keep.append(b)
+        cpreds[i].column = b
+        cpreds[i].op = OPS[p["op"]]
+        cpreds[i].hi_exclusive = 1 if p.get("hi_exclusive") else 0
+        if p["op"] == "between":
+            cpreds[i].lit_kind = 0
+            cpreds[i].i64[0] = p["lo"]
+            cpreds[i].i64[1] = p["hi"]
+        else:
+            lit = p["lit"]
+            if isinstance(lit, str):
+                cpreds[i].lit_kind = 2
+                lb = lit.encode()
+                keep.append(lb)
+                cpreds[i].str = lb
+            elif isinstance(lit, float):
+                cpreds[i].lit_kind = 1
+                cpreds[i].f64[0] = lit
+            else:
+                cpreds[i].lit_kind = 0
+                cpreds[i].i64[0] = lit
+    group_by = query.get("group_by", [])
+    gbb = [g.encode() for g in group_by]
+    keep.extend(gbb)
+    cgroup = (C.c_char_p * max(len(group_by), 1))(*gbb)
+    aggs = query["select"]
+    caggs = (GpuqAgg * max(len(aggs), 1))()
+    for i, a in enumerate(aggs):
+        caggs[i].op = AGGS[a["agg"]]
+        if a.get("col"):
+            ab = a["col"].encode()
+            keep.append(ab)
+            caggs[i].column = ab
+    return cpreds, len(preds), cgroup, len(group_by), caggs, len(aggs)
+
+
 class GpuExecutionPlan:
-    def __init__(self, session: GpuSession, paths: list[str], query: dict):
+    def __init__(self, session: GpuSession, paths: list[str] | None, query: dict,
+                 stream_dir: str | None = None):
         if session is None:
             raise GpuqError("a GpuSession (GPU) is required for scan execution")
         self.session = session
         self.query = query
         lib = session._lib
         self._lib = lib
+        self._keep = []
+        self.fast_count = None   # set by the native-planner path
+        self.empty = False
+
+        cpreds, np_, cgroup, ng, caggs, na = _build_c_query(query, self._keep)
+        if stream_dir is not None:
+            # native catalog planner (§8f row 1): manifest selection, pruning
+            # and the count fast path run inside libgpuq (catalog.cpp)
+            fc = C.c_int64(-1)
+            self._plan = lib.gpuq_plan_build_from_stream(
+                session._ctx, stream_dir.encode(),
+                cpreds, np_, cgroup, ng, caggs, na,
+                C.c_int64(-1), C.byref(fc))
+            if not self._plan:
+                if fc.value >= 0:
+                    self.fast_count = fc.value
+                    return
+                if fc.value == -2:
+                    self.empty = True
+                    return
+                raise GpuqError(f"plan_build_from_stream failed: {session._err()}")
+            return
 
         n = len(paths)
         self._path_bytes = [p.encode() for p in paths]
@@ -265,64 +346,13 @@ class GpuExecutionPlan:
             files[i].path = pb
             files[i].row_groups = None
             files[i].n_row_groups = -1
-
-        preds = list(query.get("preds", []))
-        tr = query.get("time_range")
-        if tr:
-            preds.append(
-                {
-                    "col": "p_timestamp",
-                    "op": "between",
-                    "lo": tr[0],
-                    "hi": tr[1],
-                    "hi_exclusive": True,
-                }
-            )
-        self._pred_strs = []
-        cpreds = (GpuqPred * max(len(preds), 1))()
-        for i, p in enumerate(preds):
-            cpreds[i].column = p["col"].encode()
-            cpreds[i].op = OPS[p["op"]]
-            cpreds[i].hi_exclusive = 1 if p.get("hi_exclusive") else 0
-            if p["op"] == "between":
-                cpreds[i].lit_kind = 0
-                cpreds[i].i64[0] = p["lo"]
-                cpreds[i].i64[1] = p["hi"]
-            else:
-                lit = p["lit"]
-                if isinstance(lit, str):
-                    cpreds[i].lit_kind = 2
-                    b = lit.encode()
-                    self._pred_strs.append(b)
-                    cpreds[i].str = b
-                elif isinstance(lit, float):
-                    cpreds[i].lit_kind = 1
-                    cpreds[i].f64[0] = lit
-                else:
-                    cpreds[i].lit_kind = 0
-                    cpreds[i].i64[0] = lit
-
-        group_by = query.get("group_by", [])
-        self._gb_bytes = [g.encode() for g in group_by]
-        cgroup = (C.c_char_p * max(len(group_by), 1))(*self._gb_bytes)
-
-        aggs = query["select"]
-        self._agg_bytes = []
-        caggs = (GpuqAgg * max(len(aggs), 1))()
-        for i, a in enumerate(aggs):
-            caggs[i].op = AGGS[a["agg"]]
-            if a.get("col"):
-                b = a["col"].encode()
-                self._agg_bytes.append(b)
-                caggs[i].column = b
-
         self._plan = lib.gpuq_plan_build(
             session._ctx,
             files, n,
             None, 0,
-            cpreds, len(preds),
-            cgroup, len(group_by),
-            caggs, len(aggs),
+            cpreds, np_,
+            cgroup, ng,
+            caggs, na,
             C.c_int64(-1),
         )
         if not self._plan:
