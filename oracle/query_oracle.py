@@ -40,7 +40,9 @@ I64_MAX = 2**63 - 1
 
 
 def _needed_columns(query: dict) -> set[str]:
-    need = set(query.get("group_by", []))
+    need = set()
+    for g in query.get("group_by", []):
+        need.add(g["bin"] if isinstance(g, dict) else g)
     for p in query.get("preds", []):
         need.add(p["col"])
     for s in query["select"]:
@@ -121,7 +123,17 @@ def execute(files: list[str], query: dict) -> dict:
         if group_by:
             codes_list, values_list = [], []
             for g in group_by:
-                c, v = _factorize(cols[g])
+                if isinstance(g, dict):
+                    # DATE_BIN: key = origin-aligned bin start (ms)
+                    ts = np.asarray(cols[g["bin"]])
+                    stride, origin = g["stride_ms"], g.get("origin", 0)
+                    bins = (ts - origin) // stride * stride + origin
+                    uniq_b = np.unique(bins)
+                    lookup = {int(b): i for i, b in enumerate(uniq_b.tolist())}
+                    c = np.array([lookup[int(b)] for b in bins], dtype=np.int64)
+                    v = [int(b) for b in uniq_b.tolist()]
+                else:
+                    c, v = _factorize(cols[g])
                 codes_list.append(c[sel] + 1)  # 0 = NULL group
                 values_list.append([None] + v)
             combined = codes_list[0]

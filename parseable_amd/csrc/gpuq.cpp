@@ -79,6 +79,10 @@ struct ColPlan {
   std::vector<int> cmp_preds;      // preds on the decoded i64 array
   std::vector<int> contains_preds; // CONTAINS on byte_array pages
   bool need_rank = false;   // utf8 min/max: values decoded as dict sort-ranks
+  bool is_bin = false;       // DATE_BIN pseudo-column (query/mod.rs:665-735)
+  int bin_src = -1;          // source column index (p_timestamp)
+  int64_t bin_stride = 0, bin_origin = 0, bin_min_idx = 0;
+  int32_t nbins = 0;
   std::vector<int32_t> rank_to_gid;  // rank -> gid (1-based), for export
   // global dictionary (need_gid): gid 1.. ; 0 = NULL
   std::vector<std::string> gdict;
@@ -323,7 +327,24 @@ extern "C" gpuq_plan* gpuq_plan_build(
     plan->preds.push_back(std::move(pp));
   }
   for (int32_t i = 0; i < n_group_by; i++) {
-    int ci = find_or_add_col(plan->cols, group_by[i]);
+    std::string g = group_by[i];
+    if (g.rfind("__bin:", 0) == 0) {
+      // "__bin:<col>:<stride_ms>:<origin>" — DATE_BIN time bins
+      size_t a = 6, b = g.find(':', a), c2 = g.find(':', b + 1);
+      std::string src = g.substr(a, b - a);
+      int ci = find_or_add_col(plan->cols, g);
+      auto& bc = plan->cols[ci];
+      bc.is_bin = true;
+      bc.bin_stride = std::stoll(g.substr(b + 1, c2 - b - 1));
+      bc.bin_origin = std::stoll(g.substr(c2 + 1));
+      if (bc.bin_stride <= 0) throw std::runtime_error("bad bin stride");
+      int si = find_or_add_col(plan->cols, src);
+      plan->cols[si].need_val = true;
+      plan->cols[ci].bin_src = si;
+      plan->group_cols.push_back(ci);
+      continue;
+    }
+    int ci = find_or_add_col(plan->cols, g);
     plan->cols[ci].need_gid = true;
     plan->group_cols.push_back(ci);
   }
@@ -338,6 +359,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
   }
   // resolve column phys types from schema + route predicates
   for (auto& c : plan->cols) {
+    if (c.is_bin) continue;
     int si = fm0.col_index(c.name);
     if (si < 0) throw std::runtime_error("no such column: " + c.name);
     c.phys = fm0.columns[si].phys_type;
@@ -466,6 +488,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       const auto& rg = mf.meta.row_groups[r.rg_idx];
       for (size_t ci = 0; ci < plan->cols.size(); ci++) {
         auto& c = plan->cols[ci];
+        if (c.is_bin) continue;
         int si = mf.meta.col_index(c.name);
         if (si < 0) throw std::runtime_error("column missing in file: " + c.name);
         const auto& cm = rg.chunks[si];
@@ -616,9 +639,38 @@ extern "C" gpuq_plan* gpuq_plan_build(
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
   }
 
+  // bin bounds from row-group footer stats of the source column
+  for (auto& c : plan->cols) {
+    if (!c.is_bin) continue;
+    const auto& src = plan->cols[c.bin_src];
+    bool any = false;
+    int64_t mn = 0, mx = 0;
+    for (auto& part : plan->parts)
+      for (auto& r : part.rgs) {
+        const auto& mf = *plan->files[r.file_idx];
+        int si = mf.meta.col_index(src.name);
+        const auto& cm = mf.meta.row_groups[r.rg_idx].chunks[si];
+        if (!cm.has_i64_stats)
+          throw std::runtime_error("DATE_BIN needs footer i64 stats on " + src.name);
+        if (!any) { mn = cm.stat_min; mx = cm.stat_max; any = true; }
+        else { mn = std::min(mn, cm.stat_min); mx = std::max(mx, cm.stat_max); }
+      }
+    auto fdiv = [](int64_t v, int64_t s) {
+      return v >= 0 ? v / s : -((-v + s - 1) / s);
+    };
+    c.bin_min_idx = any ? fdiv(mn - c.bin_origin, c.bin_stride) : 0;
+    int64_t max_idx = any ? fdiv(mx - c.bin_origin, c.bin_stride) : 0;
+    int64_t nb = max_idx - c.bin_min_idx + 1;
+    if (nb > (1 << 21)) throw std::runtime_error("too many DATE_BIN bins");
+    c.nbins = (int32_t)std::max<int64_t>(nb, 1);
+  }
+
   // group table size
   int64_t g = 1;
-  for (int ci : plan->group_cols) g *= (int64_t)plan->cols[ci].gdict.size() + 1;
+  for (int ci : plan->group_cols) {
+    const auto& c = plan->cols[ci];
+    g *= c.is_bin ? (int64_t)c.nbins + 1 : (int64_t)c.gdict.size() + 1;
+  }
   if (g > (int64_t)(1 << 22))
     throw std::runtime_error("group-key cardinality product too large: hash fallback is a next row");
   plan->n_groups = (int32_t)g;
@@ -747,6 +799,11 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   // column outputs
   for (size_t ci = 0; ci < plan->cols.size(); ci++) {
     auto& c = plan->cols[ci];
+    if (c.is_bin) {
+      int32_t* d; HIP_TRY(hipMalloc(&d, std::max<int64_t>(part.n_rows * 4, 16)));
+      part.d_gid[(int)ci] = d;
+      continue;
+    }
     if (c.need_gid) {
       int32_t* d; HIP_TRY(hipMalloc(&d, std::max<int64_t>(part.n_rows * 4, 16)));
       part.d_gid[(int)ci] = d;
@@ -996,6 +1053,15 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     }
   }
 
+  // 2b. DATE_BIN key materialization
+  for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+    auto& c = plan->cols[ci];
+    if (!c.is_bin) continue;
+    launch_bin_i64(st, part.d_val[c.bin_src], part.d_valid[c.bin_src],
+                   c.bin_origin, c.bin_stride, c.bin_min_idx, c.nbins,
+                   part.d_gid[(int)ci], part.n_rows);
+  }
+
   // 3. aggregate
   AggArgs a{};
   a.mask = part.d_mask;
@@ -1004,7 +1070,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   for (int k = 0; k < a.n_keys; k++) {
     int ci = plan->group_cols[k];
     a.key_gid[k] = part.d_gid[ci];
-    a.key_size[k] = (int32_t)plan->cols[ci].gdict.size() + 1;
+    const auto& kc = plan->cols[ci];
+    a.key_size[k] = kc.is_bin ? kc.nbins + 1 : (int32_t)kc.gdict.size() + 1;
   }
   a.n_aggs = (int)plan->aggs.size();
   for (int i = 0; i < a.n_aggs; i++) {
@@ -1065,7 +1132,9 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   int f = 0;
   for (int k = 0; k < n_keys; k++) {
     ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
-    make_schema_field(ss->schema.children[f], "u", plan->cols[plan->group_cols[k]].name);
+    const auto& kc = plan->cols[plan->group_cols[k]];
+    make_schema_field(ss->schema.children[f], kc.is_bin ? "l" : "u",
+                      kc.is_bin ? "date_bin" : kc.name.c_str());
     f++;
   }
   ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
@@ -1107,8 +1176,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   for (int64_t r = 0; r < nr; r++) {
     int64_t g = live[r];
     for (int k = n_keys - 1; k >= 0; k--) {
-      int ci = plan->group_cols[k];
-      int32_t sz = (int32_t)plan->cols[ci].gdict.size() + 1;
+      const auto& kc = plan->cols[plan->group_cols[k]];
+      int32_t sz = kc.is_bin ? kc.nbins + 1 : (int32_t)kc.gdict.size() + 1;
       key_gids[k][r] = (int32_t)(g % sz);
       g /= sz;
     }
@@ -1117,6 +1186,28 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   for (int k = 0; k < n_keys; k++) {
     auto* ch = make_child(f++);
     auto& c = plan->cols[plan->group_cols[k]];
+    if (c.is_bin) {
+      // i64 key: the bin's start timestamp (ms)
+      ch->n_buffers = 2;
+      ch->buffers = (const void**)calloc(2, sizeof(void*));
+      int64_t* v = (int64_t*)eb->grab(nr * 8);
+      uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
+      memset(validity, 0xff, (nr + 7) / 8);
+      int64_t nulls = 0;
+      for (int64_t r = 0; r < nr; r++) {
+        int32_t gid = key_gids[k][r];
+        if (gid > 0)
+          v[r] = c.bin_origin + (c.bin_min_idx + gid - 1) * c.bin_stride;
+        else {
+          v[r] = 0;
+          validity[r / 8] &= (uint8_t)~(1 << (r % 8));
+          nulls++;
+        }
+      }
+      if (nulls) { ch->buffers[0] = validity; ch->null_count = nulls; }
+      ch->buffers[1] = v;
+      continue;
+    }
     ch->n_buffers = 3;
     ch->buffers = (const void**)calloc(3, sizeof(void*));
     uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);

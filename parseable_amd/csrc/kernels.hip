@@ -1053,6 +1053,28 @@ __global__ void k_cmp_i64(const int64_t* __restrict__ col,
 }
 
 // ------------------------------------------------------------------
+// DATE_BIN group key: gid = 1 + (v - first_bin_origin)/stride for valid
+// rows, 0 for NULL (query/mod.rs:665-735 semantics: bins are
+// origin-aligned windows of stride ms)
+// ------------------------------------------------------------------
+__global__ void k_bin_i64(const int64_t* __restrict__ col,
+                          const uint8_t* __restrict__ valid,
+                          int64_t origin, int64_t stride, int64_t min_idx,
+                          int32_t nbins, int32_t* __restrict__ out, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride_t = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride_t) {
+    if (valid && !valid[i]) { out[i] = 0; continue; }
+    int64_t v = col[i] - origin;
+    int64_t idx = (v >= 0) ? v / stride : -((-v + stride - 1) / stride);
+    idx -= min_idx;
+    if (idx < 0) idx = 0;
+    if (idx >= nbins) idx = nbins - 1;
+    out[i] = (int32_t)(idx + 1);
+  }
+}
+
+// ------------------------------------------------------------------
 // aggregation: grid-stride, per-block LDS table (when it fits) flushed
 // with global atomics.
 // ------------------------------------------------------------------
@@ -1288,6 +1310,14 @@ void launch_cmp_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
   if (blocks > 4096) blocks = 4096;
   if (n) hipLaunchKernelGGL(k_cmp_i64, dim3(blocks), dim3(256), 0, st, col, valid, lo, hi, mode, hi_excl, is_f64, mask, n);
 }
+void launch_bin_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
+                    int64_t origin, int64_t stride, int64_t min_idx,
+                    int32_t nbins, int32_t* out, int64_t n) {
+  int blocks = (int)((n + 255) / 256);
+  if (blocks > 4096) blocks = 4096;
+  if (n) hipLaunchKernelGGL(k_bin_i64, dim3(blocks), dim3(256), 0, st, col, valid, origin, stride, min_idx, nbins, out, n);
+}
+
 void launch_init_table(hipStream_t st, uint64_t* table, int32_t n_groups,
                        int n_aggs, const int32_t* d_agg_kind) {
   int64_t tsz = (int64_t)n_groups * (1 + 2 * n_aggs);

@@ -27,6 +27,9 @@ void launch_bytes_contains(hipStream_t, const uint8_t* dec, const DevPage*,
 void launch_cmp_i64(hipStream_t, const int64_t* col, const uint8_t* valid,
                     int64_t lo, int64_t hi, int mode, int hi_excl, int is_f64,
                     uint8_t* mask, int64_t n);
+void launch_bin_i64(hipStream_t, const int64_t* col, const uint8_t* valid,
+                    int64_t origin, int64_t stride, int64_t min_idx,
+                    int32_t nbins, int32_t* out, int64_t n);
 void launch_init_table(hipStream_t, uint64_t* table, int32_t n_groups,
                        int n_aggs, const int32_t* d_agg_kind);
 void launch_dict_count(hipStream_t, const uint8_t* dec, const DevPage*,

@@ -61,7 +61,7 @@ class DistMerger:
             for i in range(len(self.aggs)):
                 self.is_f64[i] = pa.types.is_floating(batch.schema.field(nk + 1 + 2 * i).type)
         local = sorted(_batch_rows(batch, len(self.group_by), len(self.aggs)).keys(),
-                       key=lambda k: tuple((v is None, v or "") for v in k))
+                       key=lambda k: tuple(((1, "") if v is None else (0, v)) for v in k))
         if self.world > 1:
             gathered: list = [None] * self.world
             dist.all_gather_object(gathered, local)
@@ -70,7 +70,7 @@ class DistMerger:
                 allk.update(g)
         else:
             allk = set(local)
-        self.keyspace = sorted(allk, key=lambda k: tuple((v is None, v or "") for v in k))
+        self.keyspace = sorted(allk, key=lambda k: tuple(((1, "") if v is None else (0, v)) for v in k))
         self.key_index = {k: i for i, k in enumerate(self.keyspace)}
 
     def step(self, batch) -> list:

@@ -293,7 +293,11 @@ def _build_c_query(query: dict, keep):
                 cpreds[i].lit_kind = 0
                 cpreds[i].i64[0] = lit
     group_by = query.get("group_by", [])
-    gbb = [g.encode() for g in group_by]
+    gbb = []
+    for g in group_by:
+        if isinstance(g, dict):  # DATE_BIN pseudo-key (query/mod.rs:665-735)
+            g = f"__bin:{g['bin']}:{g['stride_ms']}:{g.get('origin', 0)}"
+        gbb.append(g.encode())
     keep.extend(gbb)
     cgroup = (C.c_char_p * max(len(group_by), 1))(*gbb)
     aggs = query["select"]
@@ -411,22 +415,23 @@ def merge_partials(batches, query):
     group_by = query.get("group_by", [])
     aggs = query["select"]
     acc = {}
+    nk = len(group_by)
     for b in batches:
         if b is None or b.num_rows == 0:
             continue
-        cols = {name: b.column(i).to_pylist() for i, name in enumerate(b.schema.names)}
+        cols = [b.column(i).to_pylist() for i in range(b.num_columns)]
         n = b.num_rows
         for r in range(n):
-            key = tuple(cols[g][r] for g in group_by)
-            presence = cols["__presence"][r]
+            key = tuple(cols[k][r] for k in range(nk))
+            presence = cols[nk][r]
             st = acc.get(key)
             if st is None:
                 st = {"presence": 0, "vals": [None] * len(aggs), "cnts": [0] * len(aggs)}
                 acc[key] = st
             st["presence"] += presence
             for i, a in enumerate(aggs):
-                v = cols[f"agg{i}"][r]
-                c = cols[f"agg{i}_count"][r]
+                v = cols[nk + 1 + 2 * i][r]
+                c = cols[nk + 2 + 2 * i][r]
                 if a["agg"] in ("count_star", "count"):
                     st["cnts"][i] += v
                     continue

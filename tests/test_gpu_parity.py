@@ -99,6 +99,32 @@ def test_gpu_like_scan_fresh_c3(session, tmp_path):
     assert_rows_equal(rows, expected, "c3 LIKE scan")
 
 
+def test_gpu_date_bin(session, tmp_path):
+    """Time-binned counts (the scanned variant of get_bin_density,
+    query/mod.rs:537-590,665-735): GPU vs oracle, plus completeness."""
+    from datagen.gen import gen_stream
+    from oracle import query_oracle as qo
+    from parseable_amd import Query, StandardTableProvider
+    from tests.golden_queries import BASE
+
+    info = gen_stream(str(tmp_path), "bins", "c1", rows=500_000,
+                      rows_per_file=100_000, seed=909, workers=4)
+    provider = StandardTableProvider(info["stream_dir"], session)
+    q = {"select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+         "group_by": [{"bin": "p_timestamp", "stride_ms": 60_000, "origin": 0}]}
+    rows, _ = Query(provider).execute(q)
+    expected = qo.execute(info["files"], q)["rows"]
+    assert_rows_equal(rows, expected, "date_bin minute counts")
+    assert sum(r[1] for r in rows) == 500_000
+    # non-aligned origin + coarser bins
+    q2 = {"select": [{"agg": "count_star"}],
+          "group_by": [{"bin": "p_timestamp", "stride_ms": 150_000,
+                        "origin": BASE + 7_000}]}
+    rows2, _ = Query(provider).execute(q2)
+    expected2 = qo.execute(info["files"], q2)["rows"]
+    assert_rows_equal(rows2, expected2, "date_bin offset origin")
+
+
 def test_gpu_metrics_shape(session):
     from parseable_amd import StandardTableProvider
 

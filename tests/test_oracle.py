@@ -54,3 +54,24 @@ def test_dialect_of_golden_files(golden):
     assert md.row_group(0).sorting_columns[0].descending
     lvl = rg.column(names.index("level"))
     assert "RLE_DICTIONARY" in lvl.encodings
+
+
+def test_oracle_date_bin_semantics(golden):
+    """DATE_BIN keys: origin-aligned windows; counts complete."""
+    from tests.golden_queries import BASE
+
+    files = golden["fixtures"]["g_c1"]["files"]
+    q = {"select": [{"agg": "count_star"}],
+         "group_by": [{"bin": "p_timestamp", "stride_ms": 60_000, "origin": 0}]}
+    r = qo.execute(files, q)
+    assert sum(row[1] for row in r["rows"]) == 120_000
+    for row in r["rows"]:
+        assert row[0] % 60_000 == 0
+        assert BASE <= row[0] < BASE + 10 * 60_000
+    # window widths partition rows: 2-minute bins merge adjacent 1-minute bins
+    q2 = {"select": [{"agg": "count_star"}],
+          "group_by": [{"bin": "p_timestamp", "stride_ms": 120_000, "origin": BASE}]}
+    r2 = qo.execute(files, q2)
+    ones = {row[0]: row[1] for row in r["rows"]}
+    for row in r2["rows"]:
+        assert row[1] == ones.get(row[0], 0) + ones.get(row[0] + 60_000, 0)
