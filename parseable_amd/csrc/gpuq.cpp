@@ -708,7 +708,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
     bool aggs_ok = true;
     for (auto& ap : plan->aggs) aggs_ok &= (ap.kind == AGGK_COUNT_STAR);
     plan->fused_count = plan->group_cols.size() == 1 && aggs_ok &&
-                        plan->preds.empty() && plan->n_groups <= 8192;
+                        plan->preds.empty() && plan->n_groups <= 8192 &&
+                        !plan->cols[plan->group_cols[0]].is_bin;
   }
 
   return plan.release();
