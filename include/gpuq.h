@@ -119,7 +119,10 @@ const char* gpuq_last_error(gpuq_ctx*);
  * predicate conjunction + group-by/aggregate spec -> executable plan.
  * Schema is taken from the parquet footers (merged; must agree).
  * projection lists output columns for non-aggregate scans (may be NULL when
- * group_by/aggs are given). limit < 0 means none. Returns NULL on error
+ * group_by/aggs are given); with no aggregates, a projection + limit runs a
+ * TOP-K scan ordered by p_timestamp DESC (the console default; ordering
+ * contract stream_schema_provider.rs:181-204) and the exported batch holds
+ * the projected rows. limit < 0 means none. Returns NULL on error
  * (see gpuq_last_error). */
 gpuq_plan* gpuq_plan_build(gpuq_ctx*,
     const gpuq_file* files, int32_t n_files,
@@ -141,6 +144,7 @@ gpuq_plan* gpuq_plan_build(gpuq_ctx*,
  * *fast_count: >=0 manifest-answered count (no plan); -1 plan returned
  * (or error: check return + gpuq_last_error); -2 empty relation. */
 gpuq_plan* gpuq_plan_build_from_stream(gpuq_ctx*, const char* stream_dir,
+    const char* const* projection, int32_t n_projection,
     const gpuq_pred* preds, int32_t n_preds,
     const char* const* group_by, int32_t n_group_by,
     const gpuq_agg* aggs, int32_t n_aggs, int64_t limit,

@@ -96,9 +96,49 @@ def _factorize(col: pa.Array):
     return codes, enc.dictionary.to_pylist()
 
 
+def execute_projection(files: list[str], query: dict) -> dict:
+    """Projection scan: SELECT cols WHERE ... ORDER BY p_timestamp DESC
+    LIMIT k (the console default; ordering contract
+    stream_schema_provider.rs:181-204). Ties at the LIMIT boundary are
+    engine-defined — compare tie-tolerantly (see tests)."""
+    cols_wanted = query["select_cols"]
+    need = sorted(set(cols_wanted)
+                  | {p["col"] for p in query.get("preds", [])}
+                  | {"p_timestamp"})
+    out = []
+    for path in files:
+        tbl = pq.read_table(path, columns=need)
+        n = tbl.num_rows
+        cols = {name: _norm_col(tbl.column(name)) for name in need}
+        mask = np.ones(n, dtype=bool)
+        tr = query.get("time_range")
+        if tr is not None:
+            ts = np.asarray(cols["p_timestamp"])
+            mask &= (ts >= tr[0]) & (ts < tr[1])
+        for p in query.get("preds", []):
+            mask &= _pred_mask(p, cols[p["col"]])
+        sel = np.nonzero(mask)[0]
+        proj = []
+        for cname in cols_wanted:
+            c = cols[cname]
+            if pa.types.is_dictionary(c.type) or pa.types.is_string(c.type):
+                vals = (c.cast(pa.string()) if pa.types.is_dictionary(c.type) else c).to_pylist()
+                proj.append([vals[i] for i in sel])
+            else:
+                arr = c.to_pylist()
+                proj.append([arr[i] for i in sel])
+        out.extend([list(r) for r in zip(*proj)])
+    ts_i = cols_wanted.index("p_timestamp")
+    out.sort(key=lambda r: -r[ts_i])
+    return {"columns": cols_wanted, "rows": out[: int(query["limit"])],
+            "all_matching": out}
+
+
 def execute(files: list[str], query: dict) -> dict:
     """Run the query. Returns {"columns": [...], "rows": [[key...,agg...]...]}
     rows sorted by key tuple, NULLs last."""
+    if query.get("select_cols"):
+        return execute_projection(files, query)
     need = sorted(_needed_columns(query))
     group_by = query.get("group_by", [])
     aggs = query["select"]

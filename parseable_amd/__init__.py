@@ -17,5 +17,6 @@ from .provider import (  # noqa: F401
     StandardTableProvider,
     merge_partials,
 )
+from .provider import merge_topk  # noqa: F401
 
 __version__ = "0.1.0"

@@ -76,6 +76,7 @@ def load():
     lib.gpuq_plan_build_from_stream.restype = C.c_void_p
     lib.gpuq_plan_build_from_stream.argtypes = [
         C.c_void_p, C.c_char_p,
+        C.POINTER(C.c_char_p), C.c_int32,
         C.POINTER(GpuqPred), C.c_int32,
         C.POINTER(C.c_char_p), C.c_int32,
         C.POINTER(GpuqAgg), C.c_int32,

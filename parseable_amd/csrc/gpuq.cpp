@@ -162,6 +162,14 @@ struct Partition {
   uint64_t* d_table = nullptr;
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;
+  // projection-scan buffers
+  int64_t* d_keys = nullptr;
+  int64_t* d_keys_sorted = nullptr;
+  uint32_t* d_rows = nullptr;
+  uint32_t* d_rows_sorted = nullptr;
+  unsigned long long* d_count = nullptr;
+  void* d_sort_temp = nullptr;
+  size_t sort_temp_bytes = 0;
   std::map<std::pair<int,int>, int32_t*> d_ids;     // (kind,col) -> ids
   std::map<int, int32_t*> d_gid;                    // col -> gid array
   std::map<int, int64_t*> d_val;                    // col -> value array
@@ -178,6 +186,9 @@ struct gpuq_plan {
   std::vector<PredPlan> preds;
   std::vector<AggPlan> aggs;
   std::vector<int> group_cols;   // indices into cols
+  std::vector<int> projection;   // projection-scan mode (ORDER BY ts DESC LIMIT k)
+  bool is_projection = false;
+  int ts_col = -1;               // sort key (p_timestamp) col index
   int64_t limit = -1;
   std::vector<Partition> parts;
   int32_t n_groups = 0;          // product of key sizes (incl null slots)
@@ -428,8 +439,43 @@ extern "C" gpuq_plan* gpuq_plan_build(
       default: throw std::runtime_error("bad agg op");
     }
   }
-  if (plan->aggs.empty())
-    throw std::runtime_error("projection-only scans: next row (SURVEY §8f #4); give aggregates");
+  if (plan->aggs.empty()) {
+    // projection scan: SELECT cols ... ORDER BY p_timestamp DESC LIMIT k
+    // (the console's default query; ordering contract
+    // stream_schema_provider.rs:181-204 — time-DESC is implied)
+    if (n_projection <= 0 || limit <= 0)
+      throw std::runtime_error(
+          "projection scans need a column list and a LIMIT (full result-set "
+          "export: next row, SURVEY §8f)");
+    if (limit > (1 << 22))
+      throw std::runtime_error("projection LIMIT too large");
+    plan->is_projection = true;
+    for (int32_t i = 0; i < n_projection; i++) {
+      int ci = find_or_add_col(plan->cols, projection[i]);
+      plan->projection.push_back(ci);
+      auto& c = plan->cols[ci];
+      if (c.phys == -1) {
+        int si = fm0.col_index(c.name);
+        if (si < 0) throw std::runtime_error("no such column: " + c.name);
+        c.phys = fm0.columns[si].phys_type;
+        c.optional = fm0.columns[si].optional;
+      }
+      if (c.phys == PT_BYTE_ARRAY) c.need_gid = true;     // export via dict
+      else c.need_val = true;                             // i64/f64 direct
+    }
+    int tci = find_or_add_col(plan->cols, "p_timestamp");
+    {
+      auto& tc = plan->cols[tci];
+      if (tc.phys == -1) {
+        int si = fm0.col_index("p_timestamp");
+        if (si < 0) throw std::runtime_error("no p_timestamp column");
+        tc.phys = fm0.columns[si].phys_type;
+        tc.optional = fm0.columns[si].optional;
+      }
+      tc.need_val = true;
+    }
+    plan->ts_col = tci;
+  }
   if (plan->group_cols.size() > (size_t)MAX_KEYS || plan->aggs.size() > (size_t)MAX_AGGS)
     throw std::runtime_error("too many keys/aggregates");
 
@@ -725,6 +771,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
 //   -2    every file pruned: empty relation, no plan
 extern "C" gpuq_plan* gpuq_plan_build_from_stream(
     gpuq_ctx* ctx, const char* stream_dir,
+    const char* const* projection, int32_t n_projection,
     const gpuq_pred* preds, int32_t n_preds,
     const char* const* group_by, int32_t n_group_by,
     const gpuq_agg* aggs, int32_t n_aggs, int64_t limit,
@@ -747,7 +794,8 @@ extern "C" gpuq_plan* gpuq_plan_build_from_stream(
     files[i].row_groups = nullptr;
     files[i].n_row_groups = -1;
   }
-  return gpuq_plan_build(ctx, files.data(), (int32_t)files.size(), nullptr, 0,
+  return gpuq_plan_build(ctx, files.data(), (int32_t)files.size(),
+                         projection, n_projection,
                          preds, n_preds, group_by, n_group_by, aggs, n_aggs,
                          limit);
 } catch (const std::exception& e) {
@@ -821,6 +869,18 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
     }
   }
   HIP_TRY(hipMalloc(&part.d_mask, std::max<int64_t>(part.n_rows, 16)));
+  if (plan->is_projection) {
+    int64_t n = std::max<int64_t>(part.n_rows, 16);
+    HIP_TRY(hipMalloc(&part.d_keys, n * 8));
+    HIP_TRY(hipMalloc(&part.d_keys_sorted, n * 8));
+    HIP_TRY(hipMalloc(&part.d_rows, n * 4));
+    HIP_TRY(hipMalloc(&part.d_rows_sorted, n * 4));
+    HIP_TRY(hipMalloc(&part.d_count, 8));
+    part.sort_temp_bytes = sort_pairs_desc(part.stream, nullptr, 0, part.d_keys,
+                                           part.d_keys_sorted, part.d_rows,
+                                           part.d_rows_sorted, part.n_rows);
+    HIP_TRY(hipMalloc(&part.d_sort_temp, std::max<size_t>(part.sort_temp_bytes, 16)));
+  }
   HIP_TRY(hipMalloc(&part.d_err, 4));
   size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size()) * 8;
   HIP_TRY(hipMalloc(&part.d_table, std::max<size_t>(tsz, 16)));
@@ -933,6 +993,167 @@ void ss_release(struct ArrowArrayStream* st) {
     delete s;
   }
   st->release = nullptr;
+}
+
+int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
+                           hipEvent_t ev0, hipEvent_t ev1, hipEvent_t ev_decomp,
+                           int64_t t0, bool need_mask,
+                           struct ArrowArrayStream* out) {
+  // compact selected (ts,row) pairs -> radix sort desc -> top-k gather
+  HIP_TRY(hipMemsetAsync(part.d_count, 0, 8, st));
+  launch_compact(st, need_mask ? part.d_mask : nullptr,
+                 part.d_val[plan->ts_col], part.n_rows, part.d_keys,
+                 part.d_rows, part.d_count);
+  unsigned long long cnt = 0;
+  HIP_TRY(hipMemcpyAsync(&cnt, part.d_count, 8, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  if (cnt)
+    sort_pairs_desc(st, part.d_sort_temp, part.sort_temp_bytes, part.d_keys,
+                    part.d_keys_sorted, part.d_rows, part.d_rows_sorted,
+                    (int64_t)cnt);
+  int64_t k = std::min<int64_t>((int64_t)cnt, plan->limit);
+
+  int np = (int)plan->projection.size();
+  std::vector<std::vector<int64_t>> vals(np);
+  std::vector<std::vector<int32_t>> gids(np);
+  std::vector<std::vector<uint8_t>> valids(np);
+  int64_t* d_g64 = nullptr;
+  int32_t* d_g32 = nullptr;
+  uint8_t* d_g8 = nullptr;
+  HIP_TRY(hipMalloc(&d_g64, std::max<int64_t>(k * 8, 16)));
+  HIP_TRY(hipMalloc(&d_g32, std::max<int64_t>(k * 4, 16)));
+  HIP_TRY(hipMalloc(&d_g8, std::max<int64_t>(k, 16)));
+  for (int p = 0; p < np; p++) {
+    int ci = plan->projection[p];
+    auto& c = plan->cols[ci];
+    if (c.phys == PT_BYTE_ARRAY) {
+      gids[p].resize(k);
+      launch_gather_i32(st, part.d_rows_sorted, k, part.d_gid[ci], d_g32);
+      HIP_TRY(hipMemcpyAsync(gids[p].data(), d_g32, k * 4,
+                             hipMemcpyDeviceToHost, st));
+      HIP_TRY(hipStreamSynchronize(st));
+    } else {
+      vals[p].resize(k);
+      valids[p].assign(k, 1);
+      launch_gather_i64(st, part.d_rows_sorted, k, part.d_val[ci], d_g64);
+      HIP_TRY(hipMemcpyAsync(vals[p].data(), d_g64, k * 8,
+                             hipMemcpyDeviceToHost, st));
+      auto itv = part.d_valid.find(ci);
+      if (itv != part.d_valid.end()) {
+        launch_gather_u8(st, part.d_rows_sorted, k, itv->second, d_g8);
+        HIP_TRY(hipMemcpyAsync(valids[p].data(), d_g8, k,
+                               hipMemcpyDeviceToHost, st));
+      }
+      HIP_TRY(hipStreamSynchronize(st));
+    }
+  }
+  HIP_TRY(hipEventRecord(ev1, st));
+  int32_t herr = 0;
+  HIP_TRY(hipMemcpyAsync(&herr, part.d_err, 4, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  HIP_TRY(hipFree(d_g64));
+  HIP_TRY(hipFree(d_g32));
+  HIP_TRY(hipFree(d_g8));
+  if (herr != 0)
+    throw std::runtime_error("kernel error code " + std::to_string(herr));
+  float ms_total = 0, ms_decomp = 0;
+  HIP_TRY(hipEventElapsedTime(&ms_total, ev0, ev1));
+  HIP_TRY(hipEventElapsedTime(&ms_decomp, ev0, ev_decomp));
+  HIP_TRY(hipEventDestroy(ev0));
+  HIP_TRY(hipEventDestroy(ev1));
+  HIP_TRY(hipEventDestroy(ev_decomp));
+
+  // assemble the row batch: projected columns in order
+  auto* ss = new StreamState();
+  memset(&ss->schema, 0, sizeof(ss->schema));
+  ss->schema.format = strdup("+s");
+  ss->schema.name = strdup("");
+  ss->schema.release = release_schema;
+  ss->schema.n_children = np;
+  ss->schema.children = (struct ArrowSchema**)calloc(np, sizeof(void*));
+  for (int p = 0; p < np; p++) {
+    auto& c = plan->cols[plan->projection[p]];
+    const char* fmt = (c.phys == PT_BYTE_ARRAY) ? "u"
+                      : (c.phys == PT_DOUBLE) ? "g" : "l";
+    ss->schema.children[p] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
+    make_schema_field(ss->schema.children[p], fmt, c.name);
+  }
+  auto* eb = new ExportedBatch();
+  memset(&ss->batch, 0, sizeof(ss->batch));
+  ss->batch.length = k;
+  ss->batch.n_buffers = 1;
+  ss->batch.buffers = (const void**)calloc(1, sizeof(void*));
+  ss->batch.n_children = np;
+  ss->batch.children = (struct ArrowArray**)calloc(np, sizeof(void*));
+  ss->batch.release = release_array;
+  ss->batch.private_data = eb;
+  for (int p = 0; p < np; p++) {
+    auto* ch = (struct ArrowArray*)calloc(1, sizeof(struct ArrowArray));
+    ss->batch.children[p] = ch;
+    ch->length = k;
+    ch->release = release_array;
+    auto& c = plan->cols[plan->projection[p]];
+    if (c.phys == PT_BYTE_ARRAY) {
+      ch->n_buffers = 3;
+      ch->buffers = (const void**)calloc(3, sizeof(void*));
+      uint8_t* validity = (uint8_t*)eb->grab((k + 7) / 8);
+      memset(validity, 0xff, std::max<int64_t>((k + 7) / 8, 1));
+      int32_t* offs = (int32_t*)eb->grab((k + 1) * 4);
+      size_t total = 0;
+      for (int64_t r = 0; r < k; r++) {
+        int32_t g = gids[p][r];
+        if (g > 0) total += c.gdict[g - 1].size();
+      }
+      char* data = (char*)eb->grab(total);
+      size_t off = 0;
+      int64_t nulls = 0;
+      for (int64_t r = 0; r < k; r++) {
+        offs[r] = (int32_t)off;
+        int32_t g = gids[p][r];
+        if (g > 0) {
+          memcpy(data + off, c.gdict[g - 1].data(), c.gdict[g - 1].size());
+          off += c.gdict[g - 1].size();
+        } else {
+          validity[r / 8] &= (uint8_t)~(1 << (r % 8));
+          nulls++;
+        }
+      }
+      offs[k] = (int32_t)off;
+      if (nulls) { ch->buffers[0] = validity; ch->null_count = nulls; }
+      ch->buffers[1] = offs;
+      ch->buffers[2] = data;
+    } else {
+      ch->n_buffers = 2;
+      ch->buffers = (const void**)calloc(2, sizeof(void*));
+      int64_t* v = (int64_t*)eb->grab(std::max<int64_t>(k * 8, 1));
+      uint8_t* validity = (uint8_t*)eb->grab(std::max<int64_t>((k + 7) / 8, 1));
+      memset(validity, 0xff, std::max<int64_t>((k + 7) / 8, 1));
+      int64_t nulls = 0;
+      for (int64_t r = 0; r < k; r++) {
+        v[r] = vals[p][r];
+        if (!valids[p][r]) {
+          validity[r / 8] &= (uint8_t)~(1 << (r % 8));
+          nulls++;
+        }
+      }
+      if (nulls) { ch->buffers[0] = validity; ch->null_count = nulls; }
+      ch->buffers[1] = v;
+    }
+  }
+  memset(out, 0, sizeof(*out));
+  out->get_schema = ss_get_schema;
+  out->get_next = ss_get_next;
+  out->get_last_error = ss_get_last_error;
+  out->release = ss_release;
+  out->private_data = ss;
+  {
+    std::lock_guard<std::mutex> g(plan->mu);
+    plan->m_kernel_ns += (int64_t)(ms_total * 1e6);
+    plan->m_decomp_ns += (int64_t)(ms_decomp * 1e6);
+    plan->m_exec_ns += now_ns() - t0;
+    plan->m_rows_out += k;
+  }
+  return 0;
 }
 
 }  // namespace
@@ -1062,6 +1283,10 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
                    c.bin_origin, c.bin_stride, c.bin_min_idx, c.nbins,
                    part.d_gid[(int)ci], part.n_rows);
   }
+
+  if (plan->is_projection)
+    return execute_projection(plan, part, st, ev0, ev1, ev_decomp, t0,
+                              need_mask, out);
 
   // 3. aggregate
   AggArgs a{};
@@ -1381,6 +1606,8 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle);
+    F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);
+    F(part.d_rows_sorted); F(part.d_count); F(part.d_sort_temp);
     for (auto& kv : part.d_ids) F(kv.second);
     for (auto& kv : part.d_gid) F(kv.second);
     for (auto& kv : part.d_val) F(kv.second);

@@ -15,6 +15,8 @@
 //    coalesced accesses; aggregation uses a per-block LDS table when the
 //    group table fits, flushed once per block with global atomics.
 #include <hip/hip_runtime.h>
+#include <cstring>
+#include <rocprim/device/device_radix_sort.hpp>
 #include "dev_types.h"
 #include "meta.h"
 
@@ -1053,6 +1055,60 @@ __global__ void k_cmp_i64(const int64_t* __restrict__ col,
 }
 
 // ------------------------------------------------------------------
+// Projection scans (ORDER BY p_timestamp DESC LIMIT k — the console's
+// default query; output-ordering contract stream_schema_provider.rs:181-204):
+// wave-ballot stream compaction of selected (sort-key, row) pairs, then a
+// device radix sort (rocPRIM) and a gather of the k winners.
+// ------------------------------------------------------------------
+__global__ void k_compact_selected(const uint8_t* __restrict__ mask,
+                                   const int64_t* __restrict__ key_col,
+                                   int64_t n_rows,
+                                   int64_t* __restrict__ out_keys,
+                                   uint32_t* __restrict__ out_rows,
+                                   unsigned long long* __restrict__ counter) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n_rows; i += stride) {
+    bool sel = mask ? (mask[i] != 0) : true;
+    // wave-level ballot + prefix: one atomicAdd per wave, lanes write at
+    // base + their popcount rank
+    unsigned long long ballot = __ballot(sel);
+    int lane = threadIdx.x & 63;
+    unsigned long long base = 0;
+    int cnt = __popcll(ballot);
+    if (cnt) {
+      if (lane == __ffsll((unsigned long long)ballot) - 1)
+        base = atomicAdd(counter, (unsigned long long)cnt);
+      base = __shfl(base, __ffsll((unsigned long long)ballot) - 1);
+      if (sel) {
+        int rank = __popcll(ballot & ((1ull << lane) - 1));
+        out_keys[base + rank] = key_col[i];
+        out_rows[base + rank] = (uint32_t)i;
+      }
+    }
+  }
+}
+
+__global__ void k_gather_i64(const uint32_t* __restrict__ rows, int64_t k,
+                             const int64_t* __restrict__ col,
+                             int64_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < k) out[i] = col[rows[i]];
+}
+__global__ void k_gather_i32(const uint32_t* __restrict__ rows, int64_t k,
+                             const int32_t* __restrict__ col,
+                             int32_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < k) out[i] = col[rows[i]];
+}
+__global__ void k_gather_u8(const uint32_t* __restrict__ rows, int64_t k,
+                            const uint8_t* __restrict__ col,
+                            uint8_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < k) out[i] = col[rows[i]];
+}
+
+// ------------------------------------------------------------------
 // DATE_BIN group key: gid = 1 + (v - first_bin_origin)/stride for valid
 // rows, 0 for NULL (query/mod.rs:665-735 semantics: bins are
 // origin-aligned windows of stride ms)
@@ -1310,6 +1366,28 @@ void launch_cmp_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
   if (blocks > 4096) blocks = 4096;
   if (n) hipLaunchKernelGGL(k_cmp_i64, dim3(blocks), dim3(256), 0, st, col, valid, lo, hi, mode, hi_excl, is_f64, mask, n);
 }
+void launch_compact(hipStream_t st, const uint8_t* mask, const int64_t* key_col,
+                    int64_t n_rows, int64_t* out_keys, uint32_t* out_rows,
+                    unsigned long long* counter) {
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 4096) blocks = 4096;
+  if (n_rows)
+    hipLaunchKernelGGL(k_compact_selected, dim3(blocks), dim3(256), 0, st,
+                       mask, key_col, n_rows, out_keys, out_rows, counter);
+}
+void launch_gather_i64(hipStream_t st, const uint32_t* rows, int64_t k,
+                       const int64_t* col, int64_t* out) {
+  if (k) hipLaunchKernelGGL(k_gather_i64, dim3((int)((k + 255) / 256)), dim3(256), 0, st, rows, k, col, out);
+}
+void launch_gather_i32(hipStream_t st, const uint32_t* rows, int64_t k,
+                       const int32_t* col, int32_t* out) {
+  if (k) hipLaunchKernelGGL(k_gather_i32, dim3((int)((k + 255) / 256)), dim3(256), 0, st, rows, k, col, out);
+}
+void launch_gather_u8(hipStream_t st, const uint32_t* rows, int64_t k,
+                      const uint8_t* col, uint8_t* out) {
+  if (k) hipLaunchKernelGGL(k_gather_u8, dim3((int)((k + 255) / 256)), dim3(256), 0, st, rows, k, col, out);
+}
+
 void launch_bin_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
                     int64_t origin, int64_t stride, int64_t min_idx,
                     int32_t nbins, int32_t* out, int64_t n) {
@@ -1348,6 +1426,22 @@ void launch_agg(hipStream_t st, const AggArgs& a) {
   } else {
     hipLaunchKernelGGL(k_agg<false>, dim3(blocks), dim3(256), 0, st, a);
   }
+}
+
+// device radix sort of (key,row) pairs, descending (AMD rocPRIM — native
+// header library, not a CUDA shim). Returns needed temp bytes when
+// d_temp == nullptr.
+size_t sort_pairs_desc(hipStream_t st, void* d_temp, size_t temp_bytes,
+                       const int64_t* keys_in, int64_t* keys_out,
+                       const uint32_t* rows_in, uint32_t* rows_out,
+                       int64_t n) {
+  size_t need = 0;
+  rocprim::radix_sort_pairs_desc(nullptr, need, keys_in, keys_out, rows_in,
+                                 rows_out, (size_t)n, 0, 64, st);
+  if (!d_temp) return need;
+  rocprim::radix_sort_pairs_desc(d_temp, temp_bytes, keys_in, keys_out,
+                                 rows_in, rows_out, (size_t)n, 0, 64, st);
+  return need;
 }
 
 }  // namespace gpuq

@@ -37,4 +37,16 @@ void launch_dict_count(hipStream_t, const uint8_t* dec, const DevPage*,
                        uint64_t* table, int32_t n_groups, int n_aggs,
                        int32_t* d_err);
 void launch_agg(hipStream_t, const AggArgs&);
+void launch_compact(hipStream_t, const uint8_t* mask, const int64_t* key_col,
+                    int64_t n_rows, int64_t* out_keys, uint32_t* out_rows,
+                    unsigned long long* counter);
+void launch_gather_i64(hipStream_t, const uint32_t* rows, int64_t k,
+                       const int64_t* col, int64_t* out);
+void launch_gather_i32(hipStream_t, const uint32_t* rows, int64_t k,
+                       const int32_t* col, int32_t* out);
+void launch_gather_u8(hipStream_t, const uint32_t* rows, int64_t k,
+                      const uint8_t* col, uint8_t* out);
+size_t sort_pairs_desc(hipStream_t, void* d_temp, size_t temp_bytes,
+                       const int64_t* keys_in, int64_t* keys_out,
+                       const uint32_t* rows_in, uint32_t* rows_out, int64_t n);
 }  // namespace gpuq

@@ -227,10 +227,11 @@ class StandardTableProvider:
 
         # count fast path (exact only when no value preds and the time range
         # either is absent or fully covers every kept file's ts bounds)
-        sel = query["select"]
+        sel = query.get("select", [])
         if (
             not preds
             and not query.get("group_by")
+            and not query.get("select_cols")
             and len(sel) == 1
             and sel[0]["agg"] == "count_star"
         ):
@@ -258,6 +259,25 @@ class StandardTableProvider:
             p = fe["file_path"]
             paths.append(p if os.path.isabs(p) else os.path.join(root, p))
         return GpuExecutionPlan(self.session, paths, query)
+
+
+def _build_c_projection(query: dict, keep):
+    cols = query.get("select_cols", [])
+    cb = [c.encode() for c in cols]
+    keep.extend(cb)
+    arr = (C.c_char_p * max(len(cols), 1))(*cb)
+    limit = int(query.get("limit", -1))
+    ob = query.get("order_by")
+    if cols:
+        if limit <= 0:
+            raise GpuqError("projection scans need a LIMIT (SURVEY §8f-4)")
+        if ob and not (ob.get("col") == "p_timestamp" and ob.get("desc", True)):
+            raise GpuqError("only ORDER BY p_timestamp DESC is supported "
+                            "(the reference's output-ordering contract, "
+                            "stream_schema_provider.rs:181-204)")
+        if "p_timestamp" not in cols:
+            raise GpuqError("projection must include p_timestamp (merge key)")
+    return arr, len(cols), limit
 
 
 def _build_c_query(query: dict, keep):
@@ -300,7 +320,7 @@ def _build_c_query(query: dict, keep):
         gbb.append(g.encode())
     keep.extend(gbb)
     cgroup = (C.c_char_p * max(len(group_by), 1))(*gbb)
-    aggs = query["select"]
+    aggs = query.get("select", [])
     caggs = (GpuqAgg * max(len(aggs), 1))()
     for i, a in enumerate(aggs):
         caggs[i].op = AGGS[a["agg"]]
@@ -325,14 +345,16 @@ class GpuExecutionPlan:
         self.empty = False
 
         cpreds, np_, cgroup, ng, caggs, na = _build_c_query(query, self._keep)
+        cproj, nproj, limit = _build_c_projection(query, self._keep)
         if stream_dir is not None:
             # native catalog planner (§8f row 1): manifest selection, pruning
             # and the count fast path run inside libgpuq (catalog.cpp)
             fc = C.c_int64(-1)
             self._plan = lib.gpuq_plan_build_from_stream(
                 session._ctx, stream_dir.encode(),
+                cproj, nproj,
                 cpreds, np_, cgroup, ng, caggs, na,
-                C.c_int64(-1), C.byref(fc))
+                C.c_int64(limit), C.byref(fc))
             if not self._plan:
                 if fc.value >= 0:
                     self.fast_count = fc.value
@@ -353,11 +375,11 @@ class GpuExecutionPlan:
         self._plan = lib.gpuq_plan_build(
             session._ctx,
             files, n,
-            None, 0,
+            cproj, nproj,
             cpreds, np_,
             cgroup, ng,
             caggs, na,
-            C.c_int64(-1),
+            C.c_int64(limit),
         )
         if not self._plan:
             raise GpuqError(f"plan_build failed: {session._err()}")
@@ -385,10 +407,13 @@ class GpuExecutionPlan:
         return batches[0] if batches else None
 
     def execute_all(self):
-        """Run every partition and apply the Final aggregation merge
-        (the reference's AggregateExec(Final))."""
+        """Run every partition and apply the Final merge: aggregation
+        (AggregateExec(Final)) or top-k row merge for projection scans."""
         batches = [self.execute(p) for p in range(self.partition_count())]
-        return merge_partials([b for b in batches if b is not None], self.query)
+        batches = [b for b in batches if b is not None]
+        if self.query.get("select_cols"):
+            return merge_topk(batches, self.query)
+        return merge_partials(batches, self.query)
 
     def metrics(self) -> dict:
         m = GpuqMetrics()
@@ -462,6 +487,21 @@ def merge_partials(batches, query):
     if not group_by and not rows:
         rows = [[0 if a["agg"] in ("count_star", "count") else None for a in aggs]]
     return rows
+
+
+def merge_topk(batches, query):
+    """Final merge of per-partition top-k row batches: re-sort by
+    p_timestamp DESC across partitions, truncate to LIMIT."""
+    cols = query["select_cols"]
+    ts_i = cols.index("p_timestamp")
+    rows = []
+    for b in batches:
+        if b is None or b.num_rows == 0:
+            continue
+        data = [b.column(i).to_pylist() for i in range(b.num_columns)]
+        rows.extend([list(r) for r in zip(*data)])
+    rows.sort(key=lambda r: -r[ts_i])
+    return rows[: int(query["limit"])]
 
 
 class Query:

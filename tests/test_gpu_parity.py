@@ -125,6 +125,39 @@ def test_gpu_date_bin(session, tmp_path):
     assert_rows_equal(rows2, expected2, "date_bin offset origin")
 
 
+def test_gpu_topk_projection(session, tmp_path):
+    """ORDER BY p_timestamp DESC LIMIT k projection scan.
+    Tie rows at the LIMIT boundary are engine-defined, so the check is:
+    (a) the returned timestamp multiset equals the oracle's top-k multiset;
+    (b) every returned row appears in the oracle's full matching set."""
+    from datagen.gen import gen_stream
+    from oracle import query_oracle as qo
+    from parseable_amd import Query, StandardTableProvider
+    from tests.golden_queries import BASE, MIN
+
+    info = gen_stream(str(tmp_path), "topk", "c1", rows=400_000,
+                      rows_per_file=100_000, seed=515, workers=4)
+    provider = StandardTableProvider(info["stream_dir"], session)
+    for q in [
+        {"select_cols": ["p_timestamp", "level", "host", "latency"],
+         "limit": 100},
+        {"select_cols": ["p_timestamp", "latency", "f_f64"],
+         "limit": 1000,
+         "preds": [{"col": "level", "op": "eq", "lit": "ERROR"}]},
+        {"select_cols": ["p_timestamp", "host"],
+         "limit": 50,
+         "time_range": [BASE, BASE + 2 * MIN]},
+    ]:
+        rows, _ = Query(provider).execute(q)
+        exp = qo.execute(info["files"], q)
+        ts_i = q["select_cols"].index("p_timestamp")
+        assert sorted(r[ts_i] for r in rows) == sorted(r[ts_i] for r in exp["rows"]), q
+        full = {tuple(r) for r in exp["all_matching"]}
+        for r in rows:
+            assert tuple(r) in full, (q, r)
+        assert len(rows) == len(exp["rows"])
+
+
 def test_gpu_metrics_shape(session):
     from parseable_amd import StandardTableProvider
 
