@@ -162,6 +162,7 @@ struct Partition {
   uint64_t* d_table = nullptr;
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;
+  int32_t* d_all_ids = nullptr;   // identity page-id list for the LZ4 sweep
   // projection-scan buffers
   int64_t* d_keys = nullptr;
   int64_t* d_keys_sorted = nullptr;
@@ -887,6 +888,11 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   std::vector<int32_t> kinds;
   for (auto& a : plan->aggs) kinds.push_back(a.kind);
   upload_pool(kinds.data(), kinds.size() * 4, (void**)&part.d_agg_kind);
+  {
+    std::vector<int32_t> all_ids(part.pages.size());
+    for (size_t i = 0; i < all_ids.size(); i++) all_ids[i] = (int32_t)i;
+    upload_pool(all_ids.data(), all_ids.size() * 4, (void**)&part.d_all_ids);
+  }
   // needle buffer (first CONTAINS pred; one per plan supported per column set)
   std::string needle;
   for (auto& pp : plan->preds)
@@ -1181,21 +1187,10 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
                     part.d_agg_kind);
 
   HIP_TRY(hipEventRecord(ev0, st));
-  // 1. decompress every data page
-  {
-    std::vector<int32_t> all_ids(part.pages.size());
-    for (size_t i = 0; i < all_ids.size(); i++) all_ids[i] = (int32_t)i;
-    int32_t* d_all = nullptr;
-    HIP_TRY(hipMalloc(&d_all, std::max<size_t>(all_ids.size() * 4, 16)));
-    HIP_TRY(hipMemcpyAsync(d_all, all_ids.data(), all_ids.size() * 4,
-                           hipMemcpyHostToDevice, st));
-    launch_lz4(st, part.d_raw, part.d_dec, part.d_pages, d_all,
-               (int)all_ids.size(), part.d_err);
-    HIP_TRY(hipEventRecord(ev_decomp, st));
-    // free after stream drains (defer: keep till end of exec and free then)
-    HIP_TRY(hipStreamSynchronize(st));
-    HIP_TRY(hipFree(d_all));
-  }
+  // 1. decompress every data page (id list staged at load)
+  launch_lz4(st, part.d_raw, part.d_dec, part.d_pages, part.d_all_ids,
+             (int)part.pages.size(), part.d_err);
+  HIP_TRY(hipEventRecord(ev_decomp, st));
 
   // 2. decode + predicate kernels (or the fused count path)
   if (plan->fused_count) {
@@ -1605,7 +1600,7 @@ gpuq_plan::~gpuq_plan() {
     auto F = [](void* p) { if (p) hipFree(p); };
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
-    F(part.d_table); F(part.d_agg_kind); F(part.d_needle);
+    F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);
     F(part.d_rows_sorted); F(part.d_count); F(part.d_sort_temp);
     for (auto& kv : part.d_ids) F(kv.second);
