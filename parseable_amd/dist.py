@@ -48,6 +48,7 @@ class DistMerger:
         self.keyspace: list[tuple] | None = None
         self.key_index: dict[tuple, int] = {}
         self.is_f64 = [False] * len(self.aggs)
+        self.is_str = [False] * len(self.aggs)
 
     @property
     def world(self):
@@ -59,7 +60,9 @@ class DistMerger:
         if batch is not None:
             nk = len(self.group_by)
             for i in range(len(self.aggs)):
-                self.is_f64[i] = pa.types.is_floating(batch.schema.field(nk + 1 + 2 * i).type)
+                t = batch.schema.field(nk + 1 + 2 * i).type
+                self.is_f64[i] = pa.types.is_floating(t)
+                self.is_str[i] = pa.types.is_string(t)
         local = sorted(_batch_rows(batch, len(self.group_by), len(self.aggs)).keys(),
                        key=lambda k: tuple(((1, "") if v is None else (0, v)) for v in k))
         if self.world > 1:
@@ -75,44 +78,80 @@ class DistMerger:
 
     def step(self, batch) -> list:
         """Merge this rank's partial with every other rank's; returns the
-        final rows (same normalized form as the oracle)."""
+        final rows (same normalized form as the oracle). Vectorized with
+        numpy — per-row Python/torch ops cost ~20ms/step at 1000 groups."""
+        import numpy as np
+
         assert self.keyspace is not None, "call setup() first"
         G = max(len(self.keyspace), 1)
         n_aggs = len(self.aggs)
-        rows = _batch_rows(batch, len(self.group_by), n_aggs)
+        for a in self.aggs:
+            if a["agg"] in ("min", "max") and self.is_str[
+                self.aggs.index(a)]:
+                raise NotImplementedError(
+                    "utf8 min/max across ranks: merge with merge_partials")
+
+        np_presence = np.zeros(G, dtype=np.int64)
+        np_counts = np.zeros((G, n_aggs), dtype=np.int64)
+        np_sums_i = np.zeros((G, n_aggs), dtype=np.int64)
+        np_sums_f = np.zeros((G, n_aggs), dtype=np.float64)
+        np_mins = np.full((G, n_aggs), I64_MAX, dtype=np.int64)
+        np_maxs = np.full((G, n_aggs), I64_MIN, dtype=np.int64)
+        np_mins_f = np.full((G, n_aggs), np.inf)
+        np_maxs_f = np.full((G, n_aggs), -np.inf)
+
+        if batch is not None and batch.num_rows:
+            nk = len(self.group_by)
+            keycols = [batch.column(k).to_pylist() for k in range(nk)]
+            keys = list(zip(*keycols)) if nk else [()] * batch.num_rows
+            try:
+                gis = np.fromiter((self.key_index[k] for k in keys),
+                                  dtype=np.int64, count=len(keys))
+            except KeyError as e:
+                raise RuntimeError(f"key {e} not in agreed key space")
+            np.add.at(np_presence, gis,
+                      batch.column(nk).to_numpy(zero_copy_only=False).astype(np.int64))
+            for i, a in enumerate(self.aggs):
+                vcol = batch.column(nk + 1 + 2 * i)
+                ccol = batch.column(nk + 2 + 2 * i).to_numpy(zero_copy_only=False).astype(np.int64)
+                if a["agg"] in ("count_star", "count"):
+                    np.add.at(np_counts[:, i], gis,
+                              vcol.to_numpy(zero_copy_only=False).astype(np.int64))
+                    continue
+                np.add.at(np_counts[:, i], gis, ccol)
+                has = ccol > 0
+                if not has.any():
+                    continue
+                g2 = gis[has]
+                if self.is_f64[i]:
+                    v = vcol.to_numpy(zero_copy_only=False).astype(np.float64)[has]
+                    if a["agg"] == "sum":
+                        np.add.at(np_sums_f[:, i], g2, v)
+                    elif a["agg"] == "min":
+                        np.minimum.at(np_mins_f[:, i], g2, v)
+                    else:
+                        np.maximum.at(np_maxs_f[:, i], g2, v)
+                else:
+                    import pyarrow as pa
+
+                    fill = {"sum": 0, "min": I64_MAX, "max": I64_MIN}[a["agg"]]
+                    v = vcol.fill_null(fill).to_numpy(zero_copy_only=False).astype(np.int64)[has]
+                    if a["agg"] == "sum":
+                        np.add.at(np_sums_i[:, i], g2, v)
+                    elif a["agg"] == "min":
+                        np.minimum.at(np_mins[:, i], g2, v)
+                    else:
+                        np.maximum.at(np_maxs[:, i], g2, v)
 
         dev = self.device
-        presence = torch.zeros(G, dtype=torch.int64, device=dev)
-        counts = torch.zeros(G, n_aggs, dtype=torch.int64, device=dev)
-        sums_i = torch.zeros(G, n_aggs, dtype=torch.int64, device=dev)
-        sums_f = torch.zeros(G, n_aggs, dtype=torch.float64, device=dev)
-        mins = torch.full((G, n_aggs), I64_MAX, dtype=torch.int64, device=dev)
-        maxs = torch.full((G, n_aggs), I64_MIN, dtype=torch.int64, device=dev)
-
-        for key, (p, aggvals) in rows.items():
-            gi = self.key_index.get(key)
-            if gi is None:
-                # a key outside the agreed space would mean setup() raced a
-                # plan change — fail loudly rather than drop rows
-                raise RuntimeError(f"key {key} not in agreed key space")
-            presence[gi] += p
-            for i, a in enumerate(self.aggs):
-                v, c = aggvals[i]
-                if a["agg"] in ("count_star", "count"):
-                    counts[gi, i] += v
-                    continue
-                if c == 0 or v is None:
-                    continue
-                counts[gi, i] += c
-                if a["agg"] == "sum":
-                    if self.is_f64[i]:
-                        sums_f[gi, i] += v
-                    else:
-                        sums_i[gi, i] += v
-                elif a["agg"] == "min":
-                    mins[gi, i] = min(mins[gi, i].item(), v)
-                elif a["agg"] == "max":
-                    maxs[gi, i] = max(maxs[gi, i].item(), v)
+        presence = torch.from_numpy(np_presence).to(dev)
+        counts = torch.from_numpy(np_counts).to(dev)
+        sums_i = torch.from_numpy(np_sums_i).to(dev)
+        sums_f = torch.from_numpy(np_sums_f).to(dev)
+        mins = torch.from_numpy(np_mins).to(dev)
+        maxs = torch.from_numpy(np_maxs).to(dev)
+        mins_f = torch.from_numpy(np_mins_f).to(dev)
+        maxs_f = torch.from_numpy(np_maxs_f).to(dev)
 
         if self.world > 1:
             dist.all_reduce(presence, op=dist.ReduceOp.SUM)
