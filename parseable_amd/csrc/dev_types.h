@@ -22,6 +22,22 @@ struct DevPage {
   uint8_t  phys;         // PT_* (meta.h)
 };
 
+// LZ4 segment decompression (host-planned; meta.h Lz4Plan)
+struct DevSeg {
+  uint64_t src_off;    // absolute into d_raw
+  uint64_t dst_off;    // absolute into d_dec
+  uint32_t comp_len;
+  uint32_t out_len;
+  uint8_t big;         // giant single sequence: stream straight to global
+  uint8_t raw;         // stored uncompressed: plain copy
+};
+struct DevBr {         // deferred match, absolute into d_dec (src < dst)
+  uint64_t dst, src;
+  uint32_t len;
+  uint32_t _pad;
+};
+struct DevPageBr { uint32_t start, count; };  // per page: range into the br array
+
 // comparison kernel ops (matches gpuq_op order where applicable)
 enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };
 

@@ -146,6 +146,10 @@ struct Partition {
   std::vector<int32_t> remap_pool;
   std::vector<int64_t> dictv_pool;
   std::vector<uint8_t> lut_pool;
+  // segment-parallel LZ4 (host structure walk, meta.cpp lz4_walk)
+  std::vector<DevSeg> segs;
+  std::vector<DevBr> brs;
+  std::vector<DevPageBr> pagebrs;
   uint64_t raw_bytes = 0, dec_bytes = 0;
   int64_t bytes_scanned = 0, rowgroup_bytes_total = 0;
 
@@ -163,6 +167,9 @@ struct Partition {
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;
   int32_t* d_all_ids = nullptr;   // identity page-id list for the LZ4 sweep
+  DevSeg* d_segs = nullptr;
+  DevBr* d_brs = nullptr;
+  DevPageBr* d_pagebrs = nullptr;
   // projection-scan buffers
   int64_t* d_keys = nullptr;
   int64_t* d_keys_sorted = nullptr;
@@ -647,6 +654,46 @@ extern "C" gpuq_plan* gpuq_plan_build(
         dp.encoding = (uint8_t)pi.encoding;
         dp.phys = (uint8_t)c.phys;
         int32_t page_id = (int32_t)part.pages.size();
+        // host LZ4 structure walk -> parallel segments + backref records
+        {
+          const uint8_t* praw = mf.data + pi.payload_off;
+          bool raw_page = dp.raw_copy != 0;
+          Lz4Plan lp;
+          if (!raw_page) {
+            try {
+              lp = lz4_walk(praw, pi.comp_size, pi.uncomp_size, 16384);
+            } catch (const std::exception&) {
+              if (pi.comp_size == pi.uncomp_size) raw_page = true;  // stored raw
+              else throw;
+            }
+          }
+          if (raw_page) {
+            DevSeg sgl{};
+            sgl.src_off = dp.src_off;
+            sgl.dst_off = dp.dst_off;
+            sgl.comp_len = pi.comp_size;
+            sgl.out_len = pi.uncomp_size;
+            sgl.raw = 1;
+            part.segs.push_back(sgl);
+          } else {
+            for (const auto& sg : lp.segs) {
+              DevSeg d2{};
+              d2.src_off = dp.src_off + sg.s_off;
+              d2.dst_off = dp.dst_off + sg.d_off;
+              d2.comp_len = sg.comp_len;
+              d2.out_len = sg.out_len;
+              d2.big = sg.big;
+              part.segs.push_back(d2);
+            }
+            if (!lp.backrefs.empty()) {
+              DevPageBr pb{(uint32_t)part.brs.size(), (uint32_t)lp.backrefs.size()};
+              part.pagebrs.push_back(pb);
+              for (const auto& br : lp.backrefs)
+                part.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
+                                    br.len, 0});
+            }
+          }
+        }
 
         bool dict_enc = (pi.encoding == ENC_RLE_DICT || pi.encoding == ENC_PLAIN_DICT);
         // aux = remap pool (gid decode) XOR dict-value pool (value decode):
@@ -888,11 +935,12 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   std::vector<int32_t> kinds;
   for (auto& a : plan->aggs) kinds.push_back(a.kind);
   upload_pool(kinds.data(), kinds.size() * 4, (void**)&part.d_agg_kind);
-  {
-    std::vector<int32_t> all_ids(part.pages.size());
-    for (size_t i = 0; i < all_ids.size(); i++) all_ids[i] = (int32_t)i;
-    upload_pool(all_ids.data(), all_ids.size() * 4, (void**)&part.d_all_ids);
-  }
+  upload_pool(part.segs.data(), part.segs.size() * sizeof(DevSeg),
+              (void**)&part.d_segs);
+  upload_pool(part.brs.data(), part.brs.size() * sizeof(DevBr),
+              (void**)&part.d_brs);
+  upload_pool(part.pagebrs.data(), part.pagebrs.size() * sizeof(DevPageBr),
+              (void**)&part.d_pagebrs);
   // needle buffer (first CONTAINS pred; one per plan supported per column set)
   std::string needle;
   for (auto& pp : plan->preds)
@@ -1187,9 +1235,11 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
                     part.d_agg_kind);
 
   HIP_TRY(hipEventRecord(ev0, st));
-  // 1. decompress every data page (id list staged at load)
-  launch_lz4(st, part.d_raw, part.d_dec, part.d_pages, part.d_all_ids,
-             (int)part.pages.size(), part.d_err);
+  // 1. decompress: parallel segments, then ordered backref resolution
+  launch_lz4_seg(st, part.d_raw, part.d_dec, part.d_segs,
+                 (int)part.segs.size(), part.d_err);
+  launch_lz4_backrefs(st, part.d_dec, part.d_brs, part.d_pagebrs,
+                      (int)part.pagebrs.size());
   HIP_TRY(hipEventRecord(ev_decomp, st));
 
   // 2. decode + predicate kernels (or the fused count path)
@@ -1601,6 +1651,7 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
+    F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);
     F(part.d_rows_sorted); F(part.d_count); F(part.d_sort_temp);
     for (auto& kv : part.d_ids) F(kv.second);

@@ -309,6 +309,208 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
 }
 
 // ------------------------------------------------------------------
+// Segment-parallel LZ4 (v7): the host's load-time structure walk
+// (meta.cpp lz4_walk) splits each page into <=16KB-output segments at
+// sequence boundaries; segments decompress IN PARALLEL, each wave
+// assembling its segment in LDS (in-segment matches read the linear LDS
+// image — no ring, no modulo) and flushing coalesced. Matches reaching
+// before the segment start were deferred by the host as backref records;
+// any in-segment match influenced by such a gap was deferred
+// transitively, so phase-1 bytes under a backref dst are simply
+// overwritten by phase 2 (k_lz4_backrefs), which resolves each page's
+// records in dst order.
+#define SEG_MAX 16384
+__global__ void __launch_bounds__(WAVE)
+k_lz4_seg(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
+          const DevSeg* __restrict__ segs, int n, int32_t* __restrict__ d_error) {
+  __shared__ uint8_t seg[SEG_MAX];
+  __shared__ uint8_t inbuf[LZ4_IN + 256];
+  int si = blockIdx.x;
+  if (si >= n) return;
+  const DevSeg sg = segs[si];
+  const uint8_t* src = raw + sg.src_off;
+  uint8_t* dst = dec + sg.dst_off;
+  const int lane = threadIdx.x;
+
+  if (sg.raw) {
+    for (uint32_t i = lane * 16u; i < sg.out_len; i += WAVE * 16u) {
+      uint32_t rem = sg.out_len - i;
+      if (rem >= 16 && (((uintptr_t)(src + i)) & 15) == 0 && (((uintptr_t)(dst + i)) & 15) == 0)
+        *(uint4*)(dst + i) = *(const uint4*)(src + i);
+      else
+        for (uint32_t b = 0; b < 16 && i + b < sg.out_len; b++) dst[i + b] = src[i + b];
+    }
+    return;
+  }
+
+  const uint32_t comp = sg.comp_len, uncomp = sg.out_len;
+  uint32_t in_base = 0;
+  bool in_valid = false;
+  auto refill = [&](uint32_t pos) {
+    in_base = pos & ~15u;
+    uint32_t v[17];
+#pragma unroll
+    for (int k = 0; k < 17; k++)
+      __builtin_memcpy(&v[k], src + in_base + lane * 4u + (uint32_t)k * (WAVE * 4u), 4);
+#pragma unroll
+    for (int k = 0; k < 17; k++)
+      *(uint32_t*)&inbuf[lane * 4u + (uint32_t)k * (WAVE * 4u)] = v[k];
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    in_valid = true;
+  };
+  auto inb = [&](uint32_t pos) -> uint8_t {
+    if (!in_valid || pos - in_base >= LZ4_IN) refill(pos);
+    return inbuf[pos - in_base];
+  };
+  const uint32_t* in32 = (const uint32_t*)inbuf;
+  uint32_t s = 0, d = 0;   // d: OUTPUT offset relative to segment start
+  bool bad = false;
+  const bool big = sg.big != 0;
+  while (s < comp && d < uncomp) {
+    if (!in_valid || s - in_base >= LZ4_IN) refill(s);
+    uint32_t rel = s - in_base;
+    uint32_t w[6];
+#pragma unroll
+    for (int k = 0; k < 6; k++) w[k] = in32[(rel >> 2) + k];
+    uint32_t sub = rel & 3;
+    auto gb = [&](uint32_t j) {
+      uint32_t t = sub + j;
+      return (w[t >> 2] >> ((t & 3) * 8)) & 0xffu;
+    };
+    uint32_t token = gb(0);
+    uint32_t lit = token >> 4;
+    uint32_t off, ml;
+    if (lit < 15 && !big) {
+      if (s + 1 + lit > comp || d + lit > uncomp) { bad = true; break; }
+      if ((uint32_t)lane < lit)
+        seg[d + lane] = (uint8_t)gb(1 + lane);
+      __builtin_amdgcn_wave_barrier();
+      s += 1 + lit; d += lit;
+      if (s >= comp) break;
+      if (s + 2 > comp) { bad = true; break; }
+      off = gb(1 + lit) | (gb(2 + lit) << 8);
+      s += 2;
+      ml = token & 0xf;
+      if (ml == 15) {
+        uint32_t b;
+        do { if (s >= comp) { bad = true; break; } b = inb(s); s++; ml += b; } while (b == 255);
+        if (bad) break;
+      }
+    } else {
+      // long literal (or BIG segment): chunked copy through the window
+      s++;
+      if (lit == 15) {
+        uint32_t b;
+        do { if (s >= comp) { bad = true; break; } b = inb(s); s++; lit += b; } while (b == 255);
+        if (bad) break;
+      }
+      if (s + lit > comp || d + lit > uncomp) { bad = true; break; }
+      uint32_t doneL = 0;
+      while (doneL < lit) {
+        if (!in_valid || (s + doneL) - in_base >= LZ4_IN) refill(s + doneL);
+        uint32_t avail = LZ4_IN - ((s + doneL) - in_base);
+        uint32_t chunk = min(lit - doneL, avail);
+        if (big) {
+          // stream to global (no LDS image for giant sequences)
+          const uint8_t* lsrc = &inbuf[(s + doneL) - in_base];
+          uint32_t base = d + doneL;
+          for (uint32_t i = lane; i < chunk; i += WAVE) dst[base + i] = lsrc[i];
+        } else {
+          const uint8_t* lsrc = &inbuf[(s + doneL) - in_base];
+          uint32_t base = d + doneL;
+          for (uint32_t i = lane; i < chunk; i += WAVE) seg[base + i] = lsrc[i];
+        }
+        doneL += chunk;
+      }
+      __builtin_amdgcn_wave_barrier();
+      s += lit; d += lit;
+      if (s >= comp) break;
+      if (s + 2 > comp) { bad = true; break; }
+      off = inb(s) | ((uint32_t)inb(s + 1) << 8);
+      s += 2;
+      ml = token & 0xf;
+      if (ml == 15) {
+        uint32_t b2;
+        do { if (s >= comp) { bad = true; break; } b2 = inb(s); s++; ml += b2; } while (b2 == 255);
+        if (bad) break;
+      }
+    }
+    ml += 4;
+    if (d + ml > uncomp) { bad = true; break; }
+    if (big || off > d) {
+      // deferred by the host (backref) — leave the gap for phase 2
+      d += ml;
+      continue;
+    }
+    // in-segment match: linear LDS image, no wrap
+    {
+      uint32_t done = 0;
+      __builtin_amdgcn_wave_barrier();
+      while (done < ml) {
+        uint32_t chunk = min(ml - done, off);
+        uint32_t sbase = d + done - off;
+        uint32_t dbase = d + done;
+        for (uint32_t i = lane; i < chunk; i += WAVE)
+          seg[dbase + i] = seg[sbase + i];
+        __builtin_amdgcn_wave_barrier();
+        done += chunk;
+      }
+      d += ml;
+    }
+  }
+  if (bad || d != uncomp) {
+    if (lane == 0) atomicExch(d_error, ERR_LZ4);
+    return;
+  }
+  if (!big) {
+    // coalesced flush LDS segment -> global (dst-aligned u32 body; segment
+    // boundaries within a page are arbitrary byte offsets)
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    uint32_t head = (uint32_t)((4 - ((uintptr_t)dst & 3)) & 3);
+    if (head > uncomp) head = uncomp;
+    for (uint32_t i = lane; i < head; i += WAVE) dst[i] = seg[i];
+    uint32_t body = (uncomp - head) & ~3u;
+    for (uint32_t i = lane * 4u; i < body; i += WAVE * 4u) {
+      uint32_t t = head + i;
+      uint32_t sh = (t & 3) * 8;
+      const uint32_t* s32 = (const uint32_t*)seg;
+      uint32_t v = s32[t >> 2] >> sh;
+      if (sh) v |= s32[(t >> 2) + 1] << (32 - sh);
+      *(uint32_t*)(dst + t) = v;
+    }
+    for (uint32_t i = head + body + lane; i < uncomp; i += WAVE) dst[i] = seg[i];
+  }
+}
+
+// phase 2: resolve each page's deferred matches in dst order. One wave per
+// page; records are few (segment heads + giant matches). The pattern copy
+// is chunked by the offset with a vmcnt drain so later chunks see earlier
+// writes of the same wave.
+__global__ void __launch_bounds__(WAVE)
+k_lz4_backrefs(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs,
+               const DevPageBr* __restrict__ pages, int n) {
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPageBr pb = pages[pi];
+  const int lane = threadIdx.x;
+  for (uint32_t r = 0; r < pb.count; r++) {
+    const DevBr br = brs[pb.start + r];
+    uint64_t off = br.dst - br.src;
+    uint32_t done = 0;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    while (done < br.len) {
+      uint32_t chunk = (uint32_t)min((uint64_t)(br.len - done), off);
+      for (uint32_t i = lane; i < chunk; i += WAVE)
+        dec[br.dst + done + i] = dec[br.src + done + i];
+      done += chunk;
+      if (done < br.len) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+  }
+}
+
+// ------------------------------------------------------------------
 // serial readers (lane-redundant or lane0) for slow paths
 // ------------------------------------------------------------------
 struct SerialRle {
@@ -1322,6 +1524,14 @@ __global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
 void launch_lz4(hipStream_t st, const uint8_t* raw, uint8_t* dec,
                 const DevPage* pages, const int32_t* ids, int n, int32_t* d_err) {
   if (n) hipLaunchKernelGGL(k_lz4_pages, dim3(n), dim3(WAVE), 0, st, raw, dec, pages, ids, n, d_err);
+}
+void launch_lz4_seg(hipStream_t st, const uint8_t* raw, uint8_t* dec,
+                    const DevSeg* segs, int n, int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_lz4_seg, dim3(n), dim3(WAVE), 0, st, raw, dec, segs, n, d_err);
+}
+void launch_lz4_backrefs(hipStream_t st, uint8_t* dec, const DevBr* brs,
+                         const DevPageBr* pages, int n) {
+  if (n) hipLaunchKernelGGL(k_lz4_backrefs, dim3(n), dim3(WAVE), 0, st, dec, brs, pages, n);
 }
 void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int32_t* remap_pool,

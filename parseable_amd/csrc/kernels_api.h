@@ -4,6 +4,10 @@
 #include "dev_types.h"
 
 namespace gpuq {
+void launch_lz4_seg(hipStream_t, const uint8_t* raw, uint8_t* dec,
+                    const DevSeg*, int n, int32_t* d_err);
+void launch_lz4_backrefs(hipStream_t, uint8_t* dec, const DevBr*,
+                         const DevPageBr*, int n);
 void launch_lz4(hipStream_t, const uint8_t* raw, uint8_t* dec,
                 const DevPage*, const int32_t* ids, int n, int32_t* d_err);
 void launch_dict_gid(hipStream_t, const uint8_t* dec, const DevPage*,

@@ -277,4 +277,90 @@ int lz4_decompress_host(const uint8_t* src, size_t src_len,
   return (int)(dp - dst);
 }
 
+Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
+                 uint32_t seg_max) {
+  Lz4Plan plan;
+  size_t s = 0, d = 0;
+  uint32_t seg_s = 0, seg_d = 0;        // current segment start
+  // gaps: decompressed intervals the CURRENT segment has deferred to
+  // phase 2 (few; matches overlapping a gap defer transitively)
+  std::vector<std::pair<size_t, size_t>> gaps;
+  auto close_segment = [&](size_t end_s, size_t end_d, bool big) {
+    if (end_d > seg_d || end_s > seg_s) {
+      plan.segs.push_back({seg_s, seg_d, (uint32_t)(end_s - seg_s),
+                           (uint32_t)(end_d - seg_d), (uint8_t)big});
+    }
+    seg_s = (uint32_t)end_s;
+    seg_d = (uint32_t)end_d;
+    gaps.clear();
+  };
+  while (s < comp && d < uncomp) {
+    size_t seq_s = s, seq_d = d;
+    uint8_t token = src[s++];
+    size_t lit = token >> 4;
+    if (lit == 15) {
+      uint8_t b;
+      do { if (s >= comp) throw std::runtime_error("lz4 walk: eof"); b = src[s++]; lit += b; } while (b == 255);
+    }
+    if (s + lit > comp || d + lit > uncomp) throw std::runtime_error("lz4 walk: overrun");
+    size_t ml = 0, off = 0;
+    bool has_match = false;
+    size_t after_lit_s = s + lit;
+    if (after_lit_s < comp) {
+      if (after_lit_s + 2 > comp) throw std::runtime_error("lz4 walk: bad tail");
+      off = src[after_lit_s] | ((size_t)src[after_lit_s + 1] << 8);
+      size_t p2 = after_lit_s + 2;
+      ml = token & 0xf;
+      if (ml == 15) {
+        uint8_t b;
+        do { if (p2 >= comp) throw std::runtime_error("lz4 walk: eof"); b = src[p2++]; ml += b; } while (b == 255);
+      }
+      ml += 4;
+      has_match = true;
+      if (off == 0 || off > d + lit) throw std::runtime_error("lz4 walk: bad offset");
+      s = p2;
+    } else {
+      s = after_lit_s;
+    }
+    size_t seq_out = lit + ml;
+    if (d + seq_out > uncomp && has_match) throw std::runtime_error("lz4 walk: output overrun");
+
+    if (seq_out > seg_max) {
+      // giant sequence: its own BIG segment; the match part (if any) is
+      // always deferred (phase 2 resolves pattern fills generically)
+      close_segment(seq_s, seq_d, false);
+      if (has_match && ml) {
+        plan.backrefs.push_back({(uint32_t)(d + lit), (uint32_t)(d + lit - off),
+                                 (uint32_t)ml});
+      }
+      d += seq_out;
+      close_segment(s, d, true);
+      continue;
+    }
+    if ((d + seq_out) - seg_d > seg_max) {
+      // would overflow the segment buffer: close before this sequence
+      close_segment(seq_s, seq_d, false);
+    }
+    d += lit;
+    if (has_match && ml) {
+      size_t m_src = d - off;
+      bool defer = m_src < seg_d;
+      if (!defer) {
+        // in-segment source: defer if it overlaps a deferred gap
+        size_t read_lo = m_src, read_hi = std::min(d, m_src + ml);
+        for (auto& g : gaps)
+          if (read_lo < g.second && g.first < read_hi) { defer = true; break; }
+      }
+      if (defer) {
+        plan.backrefs.push_back({(uint32_t)d, (uint32_t)(d - off), (uint32_t)ml});
+        gaps.emplace_back(d, d + ml);
+      }
+      d += ml;
+    }
+  }
+  if (d != uncomp) throw std::runtime_error("lz4 walk: size mismatch");
+  close_segment(s, d, false);
+  return plan;
+}
+
 }  // namespace gpuq

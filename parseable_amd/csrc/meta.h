@@ -79,4 +79,31 @@ std::vector<PageInfo> walk_pages(const uint8_t* buf, const ColumnChunkMeta& cm,
 int lz4_decompress_host(const uint8_t* src, size_t src_len,
                         uint8_t* dst, size_t dst_cap);
 
+// ---- segment plan for parallel GPU decompression -------------------
+// The host walks the LZ4 sequence structure once (load-time metadata, the
+// page-index analog) and splits each page into segments of <= seg_max
+// output bytes at sequence boundaries. Matches whose source lies before
+// the segment start — or overlaps a byte such a match should have written
+// (transitive gaps) — become explicit backref records resolved by a second
+// kernel in order. All parsing and byte movement still happens on the GPU
+// every execution; the walk only yields offsets.
+struct Lz4Segment {
+  uint32_t s_off;      // into the page's compressed bytes
+  uint32_t d_off;      // into the page's decompressed image
+  uint32_t comp_len;
+  uint32_t out_len;
+  uint8_t big;         // single giant sequence: decompress straight to global
+};
+struct Lz4Backref {
+  uint32_t dst, src;   // page-relative decompressed offsets (src < dst)
+  uint32_t len;
+};
+struct Lz4Plan {
+  std::vector<Lz4Segment> segs;
+  std::vector<Lz4Backref> backrefs;   // ordered by dst
+};
+// Throws on malformed streams. seg_max must match the kernel's LDS buffer.
+Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
+                 uint32_t seg_max);
+
 }  // namespace gpuq
