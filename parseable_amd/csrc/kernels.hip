@@ -484,29 +484,81 @@ k_lz4_seg(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
   }
 }
 
-// phase 2: resolve each page's deferred matches in dst order. One wave per
-// page; records are few (segment heads + giant matches). The pattern copy
-// is chunked by the offset with a vmcnt drain so later chunks see earlier
-// writes of the same wave.
+// phase 2: resolve each page's deferred matches in dst order, one wave per
+// page. On pattern-heavy pages (PLAIN int columns) the transitive gap rule
+// defers nearly every match, so this must NOT pay a memory drain per
+// record: the wave keeps a sliding 16 KiB LDS image of the output window
+// (invariant: slot b & MASK holds byte b for b in [wend-16K, wend)).
+// Sources are always window-resident when off + len <= 16K (periodicity:
+// out[dst+i] == window[src + i mod off]); record writes go through the
+// window AND global, and single-wave LDS ordering replaces all drains.
+// Window advances load only bytes this wave never wrote (later dst gaps
+// get overwritten in LDS before any later record reads them).
+#define BR_WIN 16384
 __global__ void __launch_bounds__(WAVE)
 k_lz4_backrefs(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs,
                const DevPageBr* __restrict__ pages, int n) {
+  __shared__ uint8_t win[BR_WIN];
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPageBr pb = pages[pi];
   const int lane = threadIdx.x;
+  uint64_t page0 = pb.count ? brs[pb.start].dst : 0;  // absolute anchor
+  // window covers [wend-16K, wend); start it just below the first record's
+  // dst so the first advance pulls in the preceding (phase-1) bytes
+  uint64_t wend = page0 > BR_WIN ? page0 - BR_WIN : 0;
   for (uint32_t r = 0; r < pb.count; r++) {
     const DevBr br = brs[pb.start + r];
     uint64_t off = br.dst - br.src;
-    uint32_t done = 0;
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    while (done < br.len) {
-      uint32_t chunk = (uint32_t)min((uint64_t)(br.len - done), off);
-      for (uint32_t i = lane; i < chunk; i += WAVE)
-        dec[br.dst + done + i] = dec[br.src + done + i];
-      done += chunk;
-      if (done < br.len) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (off + br.len > BR_WIN) {
+      // rare far/huge record: chunked global copy with drains
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      uint32_t done = 0;
+      while (done < br.len) {
+        uint32_t chunk = (uint32_t)min((uint64_t)(br.len - done), off);
+        for (uint32_t i = lane; i < chunk; i += WAVE)
+          dec[br.dst + done + i] = dec[br.src + done + i];
+        done += chunk;
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      // the global writes were not mirrored in the window: reload any
+      // overlap between [dst, dst+len) and the current window
+      {
+        uint64_t wlo = wend > BR_WIN ? wend - BR_WIN : 0;
+        uint64_t a = br.dst > wlo ? br.dst : wlo;
+        uint64_t b = (br.dst + br.len) < wend ? (br.dst + br.len) : wend;
+        if (a < b) {
+          for (uint64_t x = a + lane; x < b; x += WAVE)
+            win[x & (BR_WIN - 1)] = dec[x];
+          __builtin_amdgcn_wave_barrier();
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        }
+      }
+      continue;
     }
+    // advance the window to cover [.., dst+len)
+    uint64_t need_end = br.dst + br.len;
+    if (need_end > wend) {
+      uint64_t new_w0 = need_end > BR_WIN ? need_end - BR_WIN : 0;
+      uint64_t load_from = wend > new_w0 ? wend : new_w0;
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // cheap, per ~16KB
+      for (uint64_t b = load_from + lane; b < need_end; b += WAVE)
+        win[b & (BR_WIN - 1)] = dec[b];
+      __builtin_amdgcn_wave_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      wend = need_end;
+    }
+    // sources resident: src + (i mod off) in [wend-16K, dst) always;
+    // dst slots never alias source slots (the whole span fits the window)
+    uint32_t off32 = (uint32_t)off;
+    uint32_t pow2 = (off32 & (off32 - 1)) == 0;
+    for (uint32_t i = lane; i < br.len; i += WAVE) {
+      uint32_t j = pow2 ? (i & (off32 - 1)) : (i % off32);
+      uint8_t v = win[(br.src + j) & (BR_WIN - 1)];
+      dec[br.dst + i] = v;
+      win[(br.dst + i) & (BR_WIN - 1)] = v;
+    }
+    __builtin_amdgcn_wave_barrier();
   }
 }
 

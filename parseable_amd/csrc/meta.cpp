@@ -353,7 +353,12 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
       }
       if (defer) {
         plan.backrefs.push_back({(uint32_t)d, (uint32_t)(d - off), (uint32_t)ml});
-        gaps.emplace_back(d, d + ml);
+        // coalesce with the last gap when adjacent/overlapping (cascades on
+        // pattern-heavy pages would otherwise grow this list per match)
+        if (!gaps.empty() && d <= gaps.back().second)
+          gaps.back().second = std::max(gaps.back().second, d + ml);
+        else
+          gaps.emplace_back(d, d + ml);
       }
       d += ml;
     }
