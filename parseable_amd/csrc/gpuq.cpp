@@ -661,7 +661,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
           Lz4Plan lp;
           if (!raw_page) {
             try {
-              lp = lz4_walk(praw, pi.comp_size, pi.uncomp_size, 16384);
+              lp = lz4_walk(praw, pi.comp_size, pi.uncomp_size, 8192);
             } catch (const std::exception&) {
               if (pi.comp_size == pi.uncomp_size) raw_page = true;  // stored raw
               else throw;

@@ -319,7 +319,7 @@ k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
 // transitively, so phase-1 bytes under a backref dst are simply
 // overwritten by phase 2 (k_lz4_backrefs), which resolves each page's
 // records in dst order.
-#define SEG_MAX 16384
+#define SEG_MAX 8192   // 8K output/segment: 12 blocks/CU in phase 1
 __global__ void __launch_bounds__(WAVE)
 k_lz4_seg(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
           const DevSeg* __restrict__ segs, int n, int32_t* __restrict__ d_error) {
@@ -510,7 +510,7 @@ k_lz4_backrefs(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs,
   for (uint32_t r = 0; r < pb.count; r++) {
     const DevBr br = brs[pb.start + r];
     uint64_t off = br.dst - br.src;
-    if (off + br.len > BR_WIN) {
+    if (off + br.len > BR_WIN / 2) {  // window lookahead halves coverage
       // rare far/huge record: chunked global copy with drains
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       uint32_t done = 0;
@@ -536,17 +536,20 @@ k_lz4_backrefs(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs,
       }
       continue;
     }
-    // advance the window to cover [.., dst+len)
+    // advance the window in half-window steps (a per-record advance would
+    // reintroduce a drain per record); lookahead bytes may be unresolved
+    // gaps — later records overwrite their LDS slots before any read
     uint64_t need_end = br.dst + br.len;
     if (need_end > wend) {
-      uint64_t new_w0 = need_end > BR_WIN ? need_end - BR_WIN : 0;
+      uint64_t target = need_end + BR_WIN / 2;
+      uint64_t new_w0 = target > BR_WIN ? target - BR_WIN : 0;
       uint64_t load_from = wend > new_w0 ? wend : new_w0;
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // cheap, per ~16KB
-      for (uint64_t b = load_from + lane; b < need_end; b += WAVE)
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      for (uint64_t b = load_from + lane; b < target; b += WAVE)
         win[b & (BR_WIN - 1)] = dec[b];
       __builtin_amdgcn_wave_barrier();
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      wend = need_end;
+      wend = target;
     }
     // sources resident: src + (i mod off) in [wend-16K, dst) always;
     // dst slots never alias source slots (the whole span fits the window)
