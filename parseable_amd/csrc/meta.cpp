@@ -368,4 +368,25 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
   return plan;
 }
 
+std::vector<int> backref_depths(const std::vector<Lz4Backref>& brs) {
+  std::vector<int> depth(brs.size(), 0);
+  // records sorted by dst, disjoint dst ranges
+  std::vector<uint32_t> dsts(brs.size());
+  for (size_t i = 0; i < brs.size(); i++) dsts[i] = brs[i].dst;
+  for (size_t i = 0; i < brs.size(); i++) {
+    uint32_t off = brs[i].dst - brs[i].src;
+    uint32_t rlo = brs[i].src;
+    uint32_t rhi = brs[i].src + std::min(brs[i].len, off);
+    // find first record with dst+len > rlo
+    size_t j = std::upper_bound(dsts.begin(), dsts.end(), rlo) - dsts.begin();
+    if (j > 0 && brs[j - 1].dst + brs[j - 1].len > rlo) j--;
+    int d = 0;
+    for (; j < i && brs[j].dst < rhi; j++)
+      if (brs[j].dst + brs[j].len > rlo)
+        d = std::max(d, depth[j] + 1);
+    depth[i] = d;
+  }
+  return depth;
+}
+
 }  // namespace gpuq

@@ -98,6 +98,11 @@ struct Lz4Backref {
   uint32_t dst, src;   // page-relative decompressed offsets (src < dst)
   uint32_t len;
 };
+
+// dependency depth per record: 0 = source reads only phase-1 bytes;
+// d = 1 + max depth of records whose dst range overlaps the source region
+// [src, src + min(len, off)). Records' dst ranges are disjoint and sorted.
+std::vector<int> backref_depths(const std::vector<Lz4Backref>& brs);
 struct Lz4Plan {
   std::vector<Lz4Segment> segs;
   std::vector<Lz4Backref> backrefs;   // ordered by dst
