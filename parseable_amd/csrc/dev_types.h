@@ -37,6 +37,12 @@ struct DevBr {         // deferred match, absolute into d_dec (src < dst)
   uint32_t _pad;
 };
 struct DevPageBr { uint32_t start, count; };  // per page: range into the br array
+struct DevBrRes {      // literal-resolved record (absolute dec offsets)
+  uint64_t dst;
+  uint32_t len, off;
+  uint32_t piece_start, piece_n;
+};
+struct DevPiece { uint64_t src; uint32_t len; uint32_t _pad; };
 
 // comparison kernel ops (matches gpuq_op order where applicable)
 enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };

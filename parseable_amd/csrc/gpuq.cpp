@@ -150,10 +150,9 @@ struct Partition {
   std::vector<DevSeg> segs;
   std::vector<DevBr> brs;            // deep/serial pages only (windowed wave)
   std::vector<DevPageBr> pagebrs;
-  // depth-scheduled records: launch d resolves depth-d records in parallel
-  struct DepthBatch { int64_t lane_off, lane_n, wave_off, wave_n; };
-  std::vector<DevBr> brs_lane, brs_wave;   // grouped by depth
-  std::vector<DepthBatch> depth_batches;
+  // literal-resolved records (single launch; meta.cpp lz4_walk)
+  std::vector<DevBrRes> res_lane, res_wave;
+  std::vector<DevPiece> piece_pool;
   uint64_t raw_bytes = 0, dec_bytes = 0;
   int64_t bytes_scanned = 0, rowgroup_bytes_total = 0;
 
@@ -174,8 +173,9 @@ struct Partition {
   DevSeg* d_segs = nullptr;
   DevBr* d_brs = nullptr;
   DevPageBr* d_pagebrs = nullptr;
-  DevBr* d_brs_lane = nullptr;
-  DevBr* d_brs_wave = nullptr;
+  DevBrRes* d_res_lane = nullptr;
+  DevBrRes* d_res_wave = nullptr;
+  DevPiece* d_piece_pool = nullptr;
   // projection-scan buffers
   int64_t* d_keys = nullptr;
   int64_t* d_keys_sorted = nullptr;
@@ -691,27 +691,22 @@ extern "C" gpuq_plan* gpuq_plan_build(
               d2.big = sg.big;
               part.segs.push_back(d2);
             }
-            if (!lp.backrefs.empty()) {
-              std::vector<int> depths = backref_depths(lp.backrefs);
-              int maxd = *std::max_element(depths.begin(), depths.end());
-              if (maxd >= 32) {
-                // pathologically deep chain: serial windowed wave per page
-                DevPageBr pb{(uint32_t)part.brs.size(), (uint32_t)lp.backrefs.size()};
-                part.pagebrs.push_back(pb);
-                for (const auto& br : lp.backrefs)
-                  part.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
-                                      br.len, 0});
-              } else {
-                if ((int)part.depth_batches.size() < maxd + 1)
-                  part.depth_batches.resize(maxd + 1);
-                // stage into temporary per-depth buckets via the pad field
-                for (size_t k = 0; k < lp.backrefs.size(); k++) {
-                  const auto& br = lp.backrefs[k];
-                  DevBr rec{dp.dst_off + br.dst, dp.dst_off + br.src, br.len,
-                            (uint32_t)depths[k]};
-                  if (br.len <= 512) part.brs_lane.push_back(rec);
-                  else part.brs_wave.push_back(rec);
-                }
+            if (lp.fallback) {
+              // piece explosion: serial windowed wave per page
+              DevPageBr pb{(uint32_t)part.brs.size(), (uint32_t)lp.backrefs.size()};
+              part.pagebrs.push_back(pb);
+              for (const auto& br : lp.backrefs)
+                part.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
+                                    br.len, 0});
+            } else if (!lp.resolved.empty()) {
+              uint32_t pbase = (uint32_t)part.piece_pool.size();
+              for (const auto& pc : lp.pieces)
+                part.piece_pool.push_back({dp.dst_off + pc.src, pc.len, 0});
+              for (const auto& rr : lp.resolved) {
+                DevBrRes rec{dp.dst_off + rr.dst, rr.len, rr.off,
+                             pbase + rr.piece_start, rr.piece_n};
+                if (rr.len <= 256) part.res_lane.push_back(rec);
+                else part.res_wave.push_back(rec);
               }
             }
           }
@@ -753,23 +748,6 @@ extern "C" gpuq_plan* gpuq_plan_build(
         throw std::runtime_error("page rows mismatch");
     }
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
-
-    // group depth-scheduled records into per-depth launch ranges
-    {
-      auto by_depth = [](const DevBr& a, const DevBr& b) { return a._pad < b._pad; };
-      std::stable_sort(part.brs_lane.begin(), part.brs_lane.end(), by_depth);
-      std::stable_sort(part.brs_wave.begin(), part.brs_wave.end(), by_depth);
-      size_t li = 0, wi = 0;
-      for (size_t d = 0; d < part.depth_batches.size(); d++) {
-        auto& b = part.depth_batches[d];
-        b.lane_off = (int64_t)li;
-        while (li < part.brs_lane.size() && part.brs_lane[li]._pad == d) li++;
-        b.lane_n = (int64_t)li - b.lane_off;
-        b.wave_off = (int64_t)wi;
-        while (wi < part.brs_wave.size() && part.brs_wave[wi]._pad == d) wi++;
-        b.wave_n = (int64_t)wi - b.wave_off;
-      }
-    }
   }
 
   // bin bounds from row-group footer stats of the source column
@@ -980,10 +958,12 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
               (void**)&part.d_brs);
   upload_pool(part.pagebrs.data(), part.pagebrs.size() * sizeof(DevPageBr),
               (void**)&part.d_pagebrs);
-  upload_pool(part.brs_lane.data(), part.brs_lane.size() * sizeof(DevBr),
-              (void**)&part.d_brs_lane);
-  upload_pool(part.brs_wave.data(), part.brs_wave.size() * sizeof(DevBr),
-              (void**)&part.d_brs_wave);
+  upload_pool(part.res_lane.data(), part.res_lane.size() * sizeof(DevBrRes),
+              (void**)&part.d_res_lane);
+  upload_pool(part.res_wave.data(), part.res_wave.size() * sizeof(DevBrRes),
+              (void**)&part.d_res_wave);
+  upload_pool(part.piece_pool.data(), part.piece_pool.size() * sizeof(DevPiece),
+              (void**)&part.d_piece_pool);
   // needle buffer (first CONTAINS pred; one per plan supported per column set)
   std::string needle;
   for (auto& pp : plan->preds)
@@ -1281,13 +1261,13 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // 1. decompress: parallel segments, then ordered backref resolution
   launch_lz4_seg(st, part.d_raw, part.d_dec, part.d_segs,
                  (int)part.segs.size(), part.d_err);
-  // deferred-match resolution: one launch per dependency depth (records of
-  // one depth are mutually independent; kernel boundaries give coherence),
-  // plus the serial windowed wave for pathologically deep pages
-  for (const auto& b : plan->parts[pi].depth_batches) {
-    launch_br_lane(st, part.d_dec, part.d_brs_lane + b.lane_off, b.lane_n);
-    launch_br_wave(st, part.d_dec, part.d_brs_wave + b.wave_off, (int)b.wave_n);
-  }
+  // deferred-match resolution: host-resolved records are independent —
+  // one launch each (kernel boundary after phase 1 gives coherence);
+  // piece-explosion pages fall back to the serial windowed wave
+  launch_brres_lane(st, part.d_dec, part.d_res_lane, part.d_piece_pool,
+                    (int64_t)part.res_lane.size());
+  launch_brres_wave(st, part.d_dec, part.d_res_wave, part.d_piece_pool,
+                    (int)part.res_wave.size());
   launch_lz4_backrefs(st, part.d_dec, part.d_brs, part.d_pagebrs,
                       (int)part.pagebrs.size());
   HIP_TRY(hipEventRecord(ev_decomp, st));
@@ -1702,7 +1682,7 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);
-    F(part.d_brs_lane); F(part.d_brs_wave);
+    F(part.d_res_lane); F(part.d_res_wave); F(part.d_piece_pool);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);
     F(part.d_rows_sorted); F(part.d_count); F(part.d_sort_temp);
     for (auto& kv : part.d_ids) F(kv.second);

@@ -494,32 +494,42 @@ k_lz4_seg(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
 // window AND global, and single-wave LDS ordering replaces all drains.
 // Window advances load only bytes this wave never wrote (later dst gaps
 // get overwritten in LDS before any later record reads them).
-// depth-batched resolvers: all records of one launch are mutually
-// independent (their sources only overlap SHALLOWER records, resolved by
-// earlier launches — kernel-boundary coherence). Lane-per-record for
-// short records; wave-per-record (parallel periodic fill) for long ones.
-__global__ void k_br_lane(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs,
-                          int64_t n) {
+// literal-resolved record resolvers: the host composed every deferred
+// match's pattern down to PHASE-1 source pieces, so all records are
+// mutually independent — one launch, no ordering. out[dst + rep*off +
+// pat_off + j] = dec[piece.src + j].
+__global__ void k_brres_lane(uint8_t* __restrict__ dec,
+                             const DevBrRes* __restrict__ recs,
+                             const DevPiece* __restrict__ pieces, int64_t n) {
   int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (r >= n) return;
-  const DevBr br = brs[r];
-  uint32_t off = (uint32_t)(br.dst - br.src);
-  uint32_t pow2 = (off & (off - 1)) == 0;
-  for (uint32_t i = 0; i < br.len; i++) {
-    uint32_t j = pow2 ? (i & (off - 1)) : (i % off);
-    dec[br.dst + i] = dec[br.src + j];
+  const DevBrRes rec = recs[r];
+  uint32_t pat_off = 0;
+  for (uint32_t k = 0; k < rec.piece_n; k++) {
+    const DevPiece pc = pieces[rec.piece_start + k];
+    for (uint32_t base = pat_off; base < rec.len; base += rec.off) {
+      uint32_t m = min(pc.len, rec.len - base);
+      for (uint32_t j = 0; j < m; j++)
+        dec[rec.dst + base + j] = dec[pc.src + j];
+    }
+    pat_off += pc.len;
   }
 }
 __global__ void __launch_bounds__(WAVE)
-k_br_wave(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs, int n) {
+k_brres_wave(uint8_t* __restrict__ dec, const DevBrRes* __restrict__ recs,
+             const DevPiece* __restrict__ pieces, int n) {
   int r = blockIdx.x;
   if (r >= n) return;
-  const DevBr br = brs[r];
-  uint32_t off = (uint32_t)(br.dst - br.src);
-  uint32_t pow2 = (off & (off - 1)) == 0;
-  for (uint32_t i = threadIdx.x; i < br.len; i += WAVE) {
-    uint32_t j = pow2 ? (i & (off - 1)) : (i % off);
-    dec[br.dst + i] = dec[br.src + j];
+  const DevBrRes rec = recs[r];
+  uint32_t pat_off = 0;
+  for (uint32_t k = 0; k < rec.piece_n; k++) {
+    const DevPiece pc = pieces[rec.piece_start + k];
+    for (uint32_t base = pat_off; base < rec.len; base += rec.off) {
+      uint32_t m = min(pc.len, rec.len - base);
+      for (uint32_t j = threadIdx.x; j < m; j += WAVE)
+        dec[rec.dst + base + j] = dec[pc.src + j];
+    }
+    pat_off += pc.len;
   }
 }
 
@@ -1617,11 +1627,13 @@ void launch_lz4_backrefs(hipStream_t st, uint8_t* dec, const DevBr* brs,
                          const DevPageBr* pages, int n) {
   if (n) hipLaunchKernelGGL(k_lz4_backrefs, dim3(n), dim3(WAVE), 0, st, dec, brs, pages, n);
 }
-void launch_br_lane(hipStream_t st, uint8_t* dec, const DevBr* brs, int64_t n) {
-  if (n) hipLaunchKernelGGL(k_br_lane, dim3((int)((n + 255) / 256)), dim3(256), 0, st, dec, brs, n);
+void launch_brres_lane(hipStream_t st, uint8_t* dec, const DevBrRes* recs,
+                       const DevPiece* pieces, int64_t n) {
+  if (n) hipLaunchKernelGGL(k_brres_lane, dim3((int)((n + 255) / 256)), dim3(256), 0, st, dec, recs, pieces, n);
 }
-void launch_br_wave(hipStream_t st, uint8_t* dec, const DevBr* brs, int n) {
-  if (n) hipLaunchKernelGGL(k_br_wave, dim3(n), dim3(WAVE), 0, st, dec, brs, n);
+void launch_brres_wave(hipStream_t st, uint8_t* dec, const DevBrRes* recs,
+                       const DevPiece* pieces, int n) {
+  if (n) hipLaunchKernelGGL(k_brres_wave, dim3(n), dim3(WAVE), 0, st, dec, recs, pieces, n);
 }
 void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int32_t* remap_pool,

@@ -8,8 +8,10 @@ void launch_lz4_seg(hipStream_t, const uint8_t* raw, uint8_t* dec,
                     const DevSeg*, int n, int32_t* d_err);
 void launch_lz4_backrefs(hipStream_t, uint8_t* dec, const DevBr*,
                          const DevPageBr*, int n);
-void launch_br_lane(hipStream_t, uint8_t* dec, const DevBr*, int64_t n);
-void launch_br_wave(hipStream_t, uint8_t* dec, const DevBr*, int n);
+void launch_brres_lane(hipStream_t, uint8_t* dec, const DevBrRes*,
+                       const DevPiece*, int64_t n);
+void launch_brres_wave(hipStream_t, uint8_t* dec, const DevBrRes*,
+                       const DevPiece*, int n);
 void launch_lz4(hipStream_t, const uint8_t* raw, uint8_t* dec,
                 const DevPage*, const int32_t* ids, int n, int32_t* d_err);
 void launch_dict_gid(hipStream_t, const uint8_t* dec, const DevPage*,

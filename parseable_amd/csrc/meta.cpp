@@ -281,10 +281,92 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
                  uint32_t seg_max) {
   Lz4Plan plan;
   size_t s = 0, d = 0;
-  uint32_t seg_s = 0, seg_d = 0;        // current segment start
-  // gaps: decompressed intervals the CURRENT segment has deferred to
-  // phase 2 (few; matches overlapping a gap defer transitively)
-  std::vector<std::pair<size_t, size_t>> gaps;
+  uint32_t seg_s = 0, seg_d = 0;
+
+  // interval map of deferred (gap) regions, sorted and disjoint. Each entry
+  // mirrors a resolved record: bytes [start, start+len) have value
+  // pattern[(x-start) % off], pattern described by literal pieces.
+  struct MapEntry { uint32_t start, len, off, piece_start, piece_n; };
+  std::vector<MapEntry> M;
+  constexpr int MAX_PIECES_PER_RECORD = 64;
+
+  auto entry_at = [&](uint32_t x) -> size_t {
+    // first entry with start+len > x
+    size_t lo = 0, hi = M.size();
+    while (lo < hi) {
+      size_t mid = (lo + hi) / 2;
+      if (M[mid].start + M[mid].len > x) hi = mid;
+      else lo = mid + 1;
+    }
+    return lo;
+  };
+  auto overlaps_gap = [&](uint32_t a, uint32_t b) {
+    size_t i = entry_at(a);
+    return i < M.size() && M[i].start < b;
+  };
+  // decompose [a, b) into literal-source pieces; false on piece explosion
+  auto resolve = [&](uint32_t a, uint32_t b, std::vector<Lz4Piece>& out) -> bool {
+    size_t i = entry_at(a);
+    uint32_t x = a;
+    while (x < b) {
+      if ((int)out.size() > MAX_PIECES_PER_RECORD) return false;
+      if (i >= M.size() || M[i].start >= b) {
+        out.push_back({x, b - x});           // literal region
+        x = b;
+        break;
+      }
+      const MapEntry& e = M[i];
+      if (x < e.start) {
+        out.push_back({x, e.start - x});     // literal prefix
+        x = e.start;
+        continue;
+      }
+      uint32_t y = std::min<uint32_t>(b, e.start + e.len);
+      while (x < y) {
+        if ((int)out.size() > MAX_PIECES_PER_RECORD) return false;
+        uint32_t p = (x - e.start) % e.off;  // pattern position
+        uint32_t chunk = std::min(y - x, e.off - p);
+        // map [p, p+chunk) through e's pieces (pattern offsets = prefix sums)
+        uint32_t po = 0, q = p, left = chunk;
+        for (uint32_t k = 0; k < e.piece_n && left; k++) {
+          const Lz4Piece& pc = plan.pieces[e.piece_start + k];
+          if (q < po + pc.len) {
+            uint32_t within = q - po;
+            uint32_t take = std::min(pc.len - within, left);
+            out.push_back({pc.src + within, take});
+            q += take;
+            left -= take;
+          }
+          po += pc.len;
+        }
+        if (left) return false;              // pattern pieces inconsistent
+        x += chunk;
+      }
+      i++;
+    }
+    return true;
+  };
+  auto defer_match = [&](size_t dst, size_t off, size_t ml) {
+    plan.backrefs.push_back({(uint32_t)dst, (uint32_t)(dst - off), (uint32_t)ml});
+    if (!plan.fallback) {
+      uint32_t pat = (uint32_t)std::min(off, ml);
+      std::vector<Lz4Piece> pieces;
+      if (resolve((uint32_t)(dst - off), (uint32_t)(dst - off + pat), pieces)) {
+        uint32_t ps = (uint32_t)plan.pieces.size();
+        plan.pieces.insert(plan.pieces.end(), pieces.begin(), pieces.end());
+        plan.resolved.push_back({(uint32_t)dst, (uint32_t)ml, (uint32_t)off,
+                                 ps, (uint32_t)pieces.size()});
+        // insert map entry (records arrive in ascending dst order)
+        M.push_back({(uint32_t)dst, (uint32_t)ml, (uint32_t)off, ps,
+                     (uint32_t)pieces.size()});
+      } else {
+        plan.fallback = true;
+        plan.resolved.clear();
+        plan.pieces.clear();
+      }
+    }
+  };
+
   auto close_segment = [&](size_t end_s, size_t end_d, bool big) {
     if (end_d > seg_d || end_s > seg_s) {
       plan.segs.push_back({seg_s, seg_d, (uint32_t)(end_s - seg_s),
@@ -292,7 +374,6 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
     }
     seg_s = (uint32_t)end_s;
     seg_d = (uint32_t)end_d;
-    gaps.clear();
   };
   while (s < comp && d < uncomp) {
     size_t seq_s = s, seq_d = d;
@@ -326,67 +407,28 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
     if (d + seq_out > uncomp && has_match) throw std::runtime_error("lz4 walk: output overrun");
 
     if (seq_out > seg_max) {
-      // giant sequence: its own BIG segment; the match part (if any) is
-      // always deferred (phase 2 resolves pattern fills generically)
       close_segment(seq_s, seq_d, false);
-      if (has_match && ml) {
-        plan.backrefs.push_back({(uint32_t)(d + lit), (uint32_t)(d + lit - off),
-                                 (uint32_t)ml});
-      }
+      if (has_match && ml) defer_match(d + lit, off, ml);
       d += seq_out;
       close_segment(s, d, true);
       continue;
     }
     if ((d + seq_out) - seg_d > seg_max) {
-      // would overflow the segment buffer: close before this sequence
       close_segment(seq_s, seq_d, false);
     }
     d += lit;
     if (has_match && ml) {
       size_t m_src = d - off;
-      bool defer = m_src < seg_d;
-      if (!defer) {
-        // in-segment source: defer if it overlaps a deferred gap
-        size_t read_lo = m_src, read_hi = std::min(d, m_src + ml);
-        for (auto& g : gaps)
-          if (read_lo < g.second && g.first < read_hi) { defer = true; break; }
-      }
-      if (defer) {
-        plan.backrefs.push_back({(uint32_t)d, (uint32_t)(d - off), (uint32_t)ml});
-        // coalesce with the last gap when adjacent/overlapping (cascades on
-        // pattern-heavy pages would otherwise grow this list per match)
-        if (!gaps.empty() && d <= gaps.back().second)
-          gaps.back().second = std::max(gaps.back().second, d + ml);
-        else
-          gaps.emplace_back(d, d + ml);
-      }
+      bool defer = m_src < seg_d ||
+                   overlaps_gap((uint32_t)m_src,
+                                (uint32_t)std::min(d, m_src + ml));
+      if (defer) defer_match(d, off, ml);
       d += ml;
     }
   }
   if (d != uncomp) throw std::runtime_error("lz4 walk: size mismatch");
   close_segment(s, d, false);
   return plan;
-}
-
-std::vector<int> backref_depths(const std::vector<Lz4Backref>& brs) {
-  std::vector<int> depth(brs.size(), 0);
-  // records sorted by dst, disjoint dst ranges
-  std::vector<uint32_t> dsts(brs.size());
-  for (size_t i = 0; i < brs.size(); i++) dsts[i] = brs[i].dst;
-  for (size_t i = 0; i < brs.size(); i++) {
-    uint32_t off = brs[i].dst - brs[i].src;
-    uint32_t rlo = brs[i].src;
-    uint32_t rhi = brs[i].src + std::min(brs[i].len, off);
-    // find first record with dst+len > rlo
-    size_t j = std::upper_bound(dsts.begin(), dsts.end(), rlo) - dsts.begin();
-    if (j > 0 && brs[j - 1].dst + brs[j - 1].len > rlo) j--;
-    int d = 0;
-    for (; j < i && brs[j].dst < rhi; j++)
-      if (brs[j].dst + brs[j].len > rlo)
-        d = std::max(d, depth[j] + 1);
-    depth[i] = d;
-  }
-  return depth;
 }
 
 }  // namespace gpuq

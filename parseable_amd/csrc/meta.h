@@ -99,13 +99,24 @@ struct Lz4Backref {
   uint32_t len;
 };
 
-// dependency depth per record: 0 = source reads only phase-1 bytes;
-// d = 1 + max depth of records whose dst range overlaps the source region
-// [src, src + min(len, off)). Records' dst ranges are disjoint and sorted.
-std::vector<int> backref_depths(const std::vector<Lz4Backref>& brs);
+// literal-resolved deferred match: out[dst + i] = pattern[i % off] where the
+// pattern (min(off, len) bytes) is described by pieces whose sources are all
+// PHASE-1 bytes (literals / in-segment matches) — so every record resolves
+// in one parallel launch with no ordering. Built by composing the interval
+// maps of earlier records during the host walk.
+struct Lz4Resolved {
+  uint32_t dst, len, off;
+  uint32_t piece_start, piece_n;   // into Lz4Plan::pieces
+};
+struct Lz4Piece {
+  uint32_t src, len;               // page-relative literal-region source
+};
 struct Lz4Plan {
   std::vector<Lz4Segment> segs;
-  std::vector<Lz4Backref> backrefs;   // ordered by dst
+  std::vector<Lz4Backref> backrefs;    // fallback (serial window) records
+  std::vector<Lz4Resolved> resolved;   // literal-resolved records
+  std::vector<Lz4Piece> pieces;
+  bool fallback = false;               // use `backrefs` for the whole page
 };
 // Throws on malformed streams. seg_max must match the kernel's LDS buffer.
 Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
