@@ -1171,7 +1171,7 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
   __shared__ uint8_t win[CWIN];
   __shared__ uint32_t offs[CVALS + 1];
   __shared__ uint8_t nulls[CVALS];
-  __shared__ uint32_t ctrl[2];
+  __shared__ uint32_t ctrl[3];   // [2]: first value oversized (global scan)
   __shared__ uint32_t bm[CWIN / 32];
   int pi = blockIdx.x;
   if (pi >= n) return;
@@ -1204,35 +1204,58 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
     for (uint32_t i = threadIdx.x; i < CWIN / 32; i += CTHREADS) bm[i] = 0;
     __syncthreads();
     if (threadIdx.x == 0) {
-      const uint32_t* win32 = (const uint32_t*)win;
       uint32_t w = 0, cnt = 0;
-      while (done + cnt < nv && cnt < CVALS) {
-        int present = all_valid ? 1 : (int)def.next();
-        nulls[cnt] = (uint8_t)!present;
-        offs[cnt] = w;
-        if (present) {
-          if (w + 4 > CWIN) {
-            if (!all_valid) def.unread((uint32_t)present);
-            break;
-          }
-          // unaligned u32 length via two aligned LDS reads
-          uint32_t sh = (w & 3) * 8;
-          uint32_t l = win32[w >> 2] >> sh;
-          if (sh) l |= win32[(w >> 2) + 1] << (32 - sh);
+      ctrl[2] = 0;
+      if (all_valid) {
+        // streamlined walk: no def levels, no nulls[] writes, branchless
+        // 8-byte LDS length read (the walk is the block's serial spine —
+        // every instruction here is wall time)
+        while (done + cnt < nv && cnt < CVALS && w + 4 <= CWIN) {
+          uint64_t ww;
+          __builtin_memcpy(&ww, &win[w & ~3u], 8);
+          uint32_t l = (uint32_t)(ww >> ((w & 3) * 8));
           if (w + 4 + l > CWIN) {
             if (cnt == 0) {                 // oversized single value
               offs[0] = w | 0x80000000u;
-              nulls[0] = (uint8_t)2;
+              ctrl[2] = 1;
               w += 4 + l;
               cnt = 1;
-            } else if (!all_valid) {
-              def.unread((uint32_t)present);
             }
             break;
           }
+          offs[cnt] = w;
           w += 4 + l;
+          cnt++;
         }
-        cnt++;
+      } else {
+        while (done + cnt < nv && cnt < CVALS) {
+          int present = (int)def.next();
+          nulls[cnt] = (uint8_t)!present;
+          offs[cnt] = w;
+          if (present) {
+            if (w + 4 > CWIN) {
+              def.unread((uint32_t)present);
+              break;
+            }
+            uint64_t ww;
+            __builtin_memcpy(&ww, &win[w & ~3u], 8);
+            uint32_t l = (uint32_t)(ww >> ((w & 3) * 8));
+            if (w + 4 + l > CWIN) {
+              if (cnt == 0) {
+                offs[0] = w | 0x80000000u;
+                nulls[0] = (uint8_t)2;
+                ctrl[2] = 1;
+                w += 4 + l;
+                cnt = 1;
+              } else {
+                def.unread((uint32_t)present);
+              }
+              break;
+            }
+            w += 4 + l;
+          }
+          cnt++;
+        }
       }
       ctrl[0] = cnt;
       ctrl[1] = w;
@@ -1258,9 +1281,12 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
     }
     __syncthreads();
     uint32_t bn = ctrl[0];
+    uint32_t oversized0 = ctrl[2];
     for (uint32_t i = threadIdx.x; i < bn; i += CTHREADS) {
+      uint8_t nl = all_valid ? (uint8_t)((i == 0 && oversized0) ? 2 : 0)
+                             : nulls[i];
       uint8_t hit = 0;
-      if (nulls[i] == 2) {                  // oversized value: global scan
+      if (nl == 2) {                        // oversized value: global scan
         uint32_t o = walk + (offs[i] & 0x7fffffffu);
         uint32_t vl;
         __builtin_memcpy(&vl, vals + o, 4);
@@ -1273,7 +1299,7 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
             hit = (k == nlen);
           }
         }
-      } else if (!nulls[i]) {
+      } else if (!nl) {
         if (nlen == 0) hit = 1;
         else {
           uint32_t o = offs[i];
