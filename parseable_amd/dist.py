@@ -83,6 +83,12 @@ class DistMerger:
         import numpy as np
 
         assert self.keyspace is not None, "call setup() first"
+        if self.world == 1:
+            # single rank: the partial IS the final table — skip the dense
+            # keyspace/tensor machinery entirely
+            from .provider import merge_partials
+
+            return merge_partials([batch], self.query)
         G = max(len(self.keyspace), 1)
         n_aggs = len(self.aggs)
         for a in self.aggs:
