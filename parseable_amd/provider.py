@@ -439,8 +439,20 @@ def merge_partials(batches, query):
     [key..., agg...] sorted by key tuple, NULL keys last."""
     group_by = query.get("group_by", [])
     aggs = query["select"]
-    acc = {}
     nk = len(group_by)
+    if len(batches) == 1 and batches[0] is not None and batches[0].num_rows:
+        # a single partial IS the final result: the exported value columns
+        # already hold final values (counts in the value slot for count
+        # aggregates, NULL validity where count==0) — pure column zips,
+        # no per-row Python
+        b = batches[0]
+        cols = [b.column(i).to_pylist() for i in range(b.num_columns)]
+        picked = cols[:nk] + [cols[nk + 1 + 2 * i] for i in range(len(aggs))]
+        rows = [list(t) for t in zip(*picked)]
+        rows.sort(key=lambda r: tuple(((1, "") if v is None else (0, v))
+                                      for v in r[:nk]))
+        return rows
+    acc = {}
     for b in batches:
         if b is None or b.num_rows == 0:
             continue
