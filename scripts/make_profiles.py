@@ -39,7 +39,7 @@ def main():
           "Units: FETCH_SIZE/WRITE_SIZE counters are KB; fetch doubled per the",
           "gfx950 half-reporting correction (MI355X_MICROARCH.md §HBM).",
           "Per-launch values = total / dispatch count.\n"]
-    for wl in ["c1", "c3s"]:
+    for wl in ["c1", "c2s", "c3s"]:
         for f in ["kernel_stats", "kernel_trace"]:
             s = os.path.join(SRC, f"{wl}_stats_{f}.csv")
             if os.path.exists(s) and f == "kernel_stats":
@@ -63,7 +63,8 @@ def main():
             if fpl + wpl < 1e6:
                 continue
             md.append(f"| {k} | {n} | {fpl / 1e9:.3f} | {wpl / 1e9:.3f} |")
-            short = ("lz4_page_decompress" if "lz4" in k else
+            short = ("lz4_page_decompress" if "lz4_seg" in k or "lz4_pages" in k else
+                     "lz4_backrefs" if "backref" in k or "brres" in k else
                      "dict_count_fused" if "dict_count" in k else
                      "bytes_contains(LIKE)" if "contains" in k else
                      "decode+filter+groupby" if "agg" in k or "dict_pages" in k or "delta" in k or "plain" in k else k)
