@@ -24,289 +24,8 @@ namespace gpuq {
 
 #define WAVE 64
 
-// ------------------------------------------------------------------
-// LZ4_RAW page decompression: one wave per page.
-// ------------------------------------------------------------------
-#define LZ4_RING 16384  // LDS output ring (last 16 KiB; 64 KiB offsets fall
-                        // back to global reads behind a vmcnt drain)
+#define LZ4_RING 16384  // LDS window for the serial fallback resolver
 #define LZ4_IN 4096     // LDS input window for the token parse
-
-// One wave per page. v3 design notes:
-//  * the token stream is parsed lane-redundantly from an LDS-staged input
-//    window (dependent global L1/L2 loads were ~300 cycles per sequence);
-//  * a single wave's LDS operations complete in program order, so the
-//    literal-write -> match-read ordering through the ring needs NO
-//    barrier — only a compiler-ordering wave_barrier();
-//  * literals are copied from the LDS input window; matches from the LDS
-//    output ring (or global output for far offsets, behind vmcnt(0)).
-// Vectorized wave copy: n bytes from an LDS source (arbitrary byte offset,
-// read via aligned u32 pairs + shifts) to global dst + the LDS ring, with
-// the body as dst-aligned u32 stores. Page images are 16-aligned in the
-// arena so (dst + d) and ring index d share alignment. ~6x the byte loop
-// (iterations pipeline; 4x fewer memory ops).
-__device__ inline void vcopy_from_lds(const uint32_t* __restrict__ src32, uint32_t so,
-                                      uint8_t* __restrict__ dst,
-                                      uint8_t* __restrict__ ring,
-                                      uint32_t d, uint32_t n, int lane) {
-  const uint8_t* src8 = (const uint8_t*)src32;
-  uint32_t head = (4 - (d & 3)) & 3;
-  if (head > n) head = n;
-  for (uint32_t i = lane; i < head; i += WAVE) {
-    uint8_t v = src8[so + i];
-    dst[d + i] = v;
-    ring[(d + i) & (LZ4_RING - 1)] = v;
-  }
-  uint32_t body = (n - head) & ~3u;
-  uint32_t sbase = so + head;
-  for (uint32_t i = lane * 4u; i < body; i += WAVE * 4u) {
-    uint32_t t = sbase + i;
-    uint32_t sh = (t & 3) * 8;
-    uint32_t v = src32[t >> 2] >> sh;
-    if (sh) v |= src32[(t >> 2) + 1] << (32 - sh);
-    uint32_t dd = d + head + i;
-    *(uint32_t*)(dst + dd) = v;
-    *(uint32_t*)&ring[dd & (LZ4_RING - 1)] = v;
-  }
-  for (uint32_t i = head + body + lane; i < n; i += WAVE) {
-    uint8_t v = src8[so + i];
-    dst[d + i] = v;
-    ring[(d + i) & (LZ4_RING - 1)] = v;
-  }
-}
-
-// Same, with the RING as source (match copies, off >= 64). The caller bounds
-// the chunk so the source region does not wrap the ring.
-__device__ inline void vcopy_from_ring(uint8_t* __restrict__ ring,
-                                       uint32_t src_idx /* ring index, unwrapped base & masked by caller */,
-                                       uint8_t* __restrict__ dst,
-                                       uint32_t d, uint32_t n, int lane) {
-  const uint32_t* ring32 = (const uint32_t*)ring;
-  uint32_t head = (4 - (d & 3)) & 3;
-  if (head > n) head = n;
-  for (uint32_t i = lane; i < head; i += WAVE) {
-    uint8_t v = ring[(src_idx + i) & (LZ4_RING - 1)];
-    dst[d + i] = v;
-    ring[(d + i) & (LZ4_RING - 1)] = v;
-  }
-  uint32_t body = (n - head) & ~3u;
-  uint32_t sbase = src_idx + head;
-  for (uint32_t i = lane * 4u; i < body; i += WAVE * 4u) {
-    uint32_t t = sbase + i;  // caller guarantees no wrap within [sbase, sbase+body+4)
-    uint32_t sh = (t & 3) * 8;
-    uint32_t v = ring32[(t & (LZ4_RING - 1)) >> 2] >> sh;
-    if (sh) v |= ring32[(((t + 4) & (LZ4_RING - 1)) >> 2)] << (32 - sh);
-    uint32_t dd = d + head + i;
-    *(uint32_t*)(dst + dd) = v;
-    *(uint32_t*)&ring[dd & (LZ4_RING - 1)] = v;
-  }
-  for (uint32_t i = head + body + lane; i < n; i += WAVE) {
-    uint8_t v = ring[(src_idx + i) & (LZ4_RING - 1)];
-    dst[d + i] = v;
-    ring[(d + i) & (LZ4_RING - 1)] = v;
-  }
-}
-
-__global__ void __launch_bounds__(WAVE)
-k_lz4_pages(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
-            const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
-            int n, int32_t* __restrict__ d_error) {
-  __shared__ uint8_t ring[LZ4_RING];
-  __shared__ uint8_t inbuf[LZ4_IN + 256];  // 17 x 256B per-lane copy rounds
-  int pi = blockIdx.x;
-  if (pi >= n) return;
-  const DevPage pg = pages[ids[pi]];
-  const uint8_t* src = raw + pg.src_off;
-  uint8_t* dst = dec + pg.dst_off;
-  const int lane = threadIdx.x;
-
-  if (pg.raw_copy) {
-    for (uint32_t i = lane * 16u; i < pg.uncomp_size; i += WAVE * 16u) {
-      uint32_t rem = pg.uncomp_size - i;
-      if (rem >= 16 && (((uintptr_t)(src + i)) & 15) == 0 && (((uintptr_t)(dst + i)) & 15) == 0) {
-        *(uint4*)(dst + i) = *(const uint4*)(src + i);
-      } else {
-        for (uint32_t b = 0; b < 16 && i + b < pg.uncomp_size; b++) dst[i + b] = src[i + b];
-      }
-    }
-    return;
-  }
-
-  const uint32_t comp = pg.comp_size, uncomp = pg.uncomp_size;
-  uint32_t in_base = 0;
-  bool in_valid = false;
-  // refill the input window to cover [pos, pos + LZ4_IN) (clamped to comp+pad)
-  auto refill = [&](uint32_t pos) {
-    in_base = pos & ~15u;
-    // ALL loads issued before any LDS store: an interleaved load->store loop
-    // serializes on a waitcnt per iteration (~900 cycles / 256 B — the
-    // 0.6 GB/s-per-wave pathology measured by scripts/micro_lz4.cpp).
-    // Reads up to LZ4_IN+256 past the page inside d_raw (host pads the raw
-    // buffer with this slack).
-    uint32_t v[17];
-#pragma unroll
-    for (int k = 0; k < 17; k++)
-      __builtin_memcpy(&v[k], src + in_base + lane * 4u + (uint32_t)k * (WAVE * 4u), 4);
-#pragma unroll
-    for (int k = 0; k < 17; k++)
-      *(uint32_t*)&inbuf[lane * 4u + (uint32_t)k * (WAVE * 4u)] = v[k];
-    __builtin_amdgcn_wave_barrier();
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    in_valid = true;
-  };
-  auto inb = [&](uint32_t pos) -> uint8_t {
-    if (!in_valid || pos - in_base >= LZ4_IN) refill(pos);
-    return inbuf[pos - in_base];
-  };
-
-  const uint32_t* in32 = (const uint32_t*)inbuf;
-  uint32_t s = 0, d = 0;
-  bool bad = false;
-  while (s < comp && d < uncomp) {
-    // FAST SEQUENCE PARSE: match-heavy pages have a sequence every ~10
-    // bytes; byte-wise dependent LDS reads cost ~300 cycles/sequence.
-    // Pull 24 bytes into registers once and extract token/offset from
-    // register math (short-literal, short-match case: the common one).
-    if (!in_valid || s - in_base >= LZ4_IN) refill(s);
-    uint32_t rel = s - in_base;                 // < LZ4_IN; +24 fits the +64 pad
-    uint32_t w[6];
-#pragma unroll
-    for (int k = 0; k < 6; k++) w[k] = in32[(rel >> 2) + k];
-    uint32_t sub = rel & 3;
-    auto gb = [&](uint32_t j) {
-      uint32_t t = sub + j;
-      return (w[t >> 2] >> ((t & 3) * 8)) & 0xffu;
-    };
-    uint32_t token = gb(0);
-    uint32_t lit = token >> 4;
-    uint32_t off, ml;
-    if (lit < 15) {
-      // short literal: the bytes are ALREADY in the w[] registers — write
-      // them in one divergence-masked pass, no loop, no LDS read. (Loop
-      // machinery cost ~500 cycles/sequence on a solo wave.)
-      if (s + 1 + lit > comp || d + lit > uncomp) { bad = true; break; }
-      if ((uint32_t)lane < lit) {
-        uint8_t v = (uint8_t)gb(1 + lane);
-        dst[d + lane] = v;
-        ring[(d + lane) & (LZ4_RING - 1)] = v;
-      }
-      __builtin_amdgcn_wave_barrier();
-      s += 1 + lit; d += lit;
-      if (s >= comp) break;                 // last sequence: literals only
-      if (s + 2 > comp) { bad = true; break; }
-      off = gb(1 + lit) | (gb(2 + lit) << 8);
-      s += 2;
-      ml = token & 0xf;
-      if (ml == 15) {
-        uint32_t b;
-        do { if (s >= comp) { bad = true; break; } b = inb(s); s++; ml += b; } while (b == 255);
-        if (bad) break;
-      }
-    } else {
-      // long-literal path (rare on match-heavy pages): byte-wise parse
-      s++;
-      uint32_t b;
-      do { if (s >= comp) { bad = true; break; } b = inb(s); s++; lit += b; } while (b == 255);
-      if (bad) break;
-      if (s + lit > comp || d + lit > uncomp) { bad = true; break; }
-      uint32_t doneL = 0;
-      while (doneL < lit) {
-        if (!in_valid || (s + doneL) - in_base >= LZ4_IN) refill(s + doneL);
-        uint32_t avail = LZ4_IN - ((s + doneL) - in_base);
-        uint32_t chunk = min(lit - doneL, avail);
-        vcopy_from_lds(in32, (s + doneL) - in_base, dst, ring, d + doneL, chunk, lane);
-        doneL += chunk;
-      }
-      __builtin_amdgcn_wave_barrier();
-      s += lit; d += lit;
-      if (s >= comp) break;
-      if (s + 2 > comp) { bad = true; break; }
-      off = inb(s) | ((uint32_t)inb(s + 1) << 8);
-      s += 2;
-      ml = token & 0xf;
-      if (ml == 15) {
-        uint32_t b2;
-        do { if (s >= comp) { bad = true; break; } b2 = inb(s); s++; ml += b2; } while (b2 == 255);
-        if (bad) break;
-      }
-    }
-    if (off == 0 || off > d) { bad = true; break; }
-    ml += 4;
-    if (d + ml > uncomp) { bad = true; break; }
-    if (ml <= (uint32_t)WAVE && off <= LZ4_RING / 2) {
-      // short match, one divergence-masked pass, no loop. Source index:
-      // off >= 64 -> direct; power-of-two off (the 4/8-byte periods of
-      // PLAIN int pages) -> mask; else modulo.
-      __builtin_amdgcn_wave_barrier();
-      if ((uint32_t)lane < ml) {
-        uint32_t j;
-        if (off >= (uint32_t)WAVE) j = (uint32_t)lane;
-        else if ((off & (off - 1)) == 0) j = (uint32_t)lane & (off - 1);
-        else j = (uint32_t)lane % off;
-        uint8_t v = ring[(d - off + j) & (LZ4_RING - 1)];
-        dst[d + lane] = v;
-        ring[(d + lane) & (LZ4_RING - 1)] = v;
-      }
-      __builtin_amdgcn_wave_barrier();
-      d += ml;
-      continue;
-    }
-    uint32_t done = 0;
-    if (off > LZ4_RING / 2) {
-      // far match: read the already-written global output (rare)
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      while (done < ml) {
-        uint32_t chunk = min(ml - done, off);
-        for (uint32_t i = lane; i < chunk; i += WAVE) {
-          uint8_t v = dst[d + done - off + i];
-          dst[d + done + i] = v;
-          ring[(d + done + i) & (LZ4_RING - 1)] = v;
-        }
-        done += chunk;
-        if (done < ml) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-    } else {
-      // near match from the ring; single-wave LDS ops are in-order, so no
-      // barrier between chunks — wave_barrier only pins compiler order
-      while (done < ml) {
-        uint32_t chunk;
-        __builtin_amdgcn_wave_barrier();
-        if (off < WAVE) {
-          chunk = min(ml - done, (uint32_t)LZ4_RING - off);
-          for (uint32_t i = lane; i < chunk; i += WAVE) {
-            uint8_t v = ring[(d + done - off + (i % off)) & (LZ4_RING - 1)];
-            dst[d + done + i] = v;
-            ring[(d + done + i) & (LZ4_RING - 1)] = v;
-          }
-        } else {
-          chunk = min(ml - done, min(off, (uint32_t)LZ4_RING - off));
-          uint32_t sidx = (d + done - off) & (LZ4_RING - 1);
-          // keep the source region from wrapping (vcopy reads t and t+4)
-          chunk = min(chunk, (uint32_t)LZ4_RING - 4 - sidx);
-          if (chunk == 0) {
-            for (uint32_t i = lane; i < min(ml - done, 4u); i += WAVE) {
-              uint8_t v = ring[(d + done - off + i) & (LZ4_RING - 1)];
-              dst[d + done + i] = v;
-              ring[(d + done + i) & (LZ4_RING - 1)] = v;
-            }
-            chunk = min(ml - done, 4u);
-          } else {
-            vcopy_from_ring(ring, sidx, dst, d + done, chunk, lane);
-          }
-        }
-        done += chunk;
-      }
-      __builtin_amdgcn_wave_barrier();
-    }
-    d += ml;
-  }
-  if (bad || d != uncomp) {
-    if (pg.comp_size == pg.uncomp_size) {
-      for (uint32_t i = lane; i < pg.uncomp_size; i += WAVE) dst[i] = src[i];
-    } else if (lane == 0) {
-      atomicExch(d_error, ERR_LZ4);
-    }
-  }
-}
 
 // ------------------------------------------------------------------
 // Segment-parallel LZ4 (v7): the host's load-time structure walk
@@ -1641,10 +1360,6 @@ __global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
 // ------------------------------------------------------------------
 // host-side launchers (called from gpuq.cpp, same TU set)
 // ------------------------------------------------------------------
-void launch_lz4(hipStream_t st, const uint8_t* raw, uint8_t* dec,
-                const DevPage* pages, const int32_t* ids, int n, int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_lz4_pages, dim3(n), dim3(WAVE), 0, st, raw, dec, pages, ids, n, d_err);
-}
 void launch_lz4_seg(hipStream_t st, const uint8_t* raw, uint8_t* dec,
                     const DevSeg* segs, int n, int32_t* d_err) {
   if (n) hipLaunchKernelGGL(k_lz4_seg, dim3(n), dim3(WAVE), 0, st, raw, dec, segs, n, d_err);

@@ -12,8 +12,6 @@ void launch_brres_lane(hipStream_t, uint8_t* dec, const DevBrRes*,
                        const DevPiece*, int64_t n);
 void launch_brres_wave(hipStream_t, uint8_t* dec, const DevBrRes*,
                        const DevPiece*, int n);
-void launch_lz4(hipStream_t, const uint8_t* raw, uint8_t* dec,
-                const DevPage*, const int32_t* ids, int n, int32_t* d_err);
 void launch_dict_gid(hipStream_t, const uint8_t* dec, const DevPage*,
                      const int32_t* ids, int n, const int32_t* remap_pool,
                      int32_t* out, uint8_t* valid, int32_t* d_err);
