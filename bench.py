@@ -45,6 +45,12 @@ WORKLOADS = {
         "select": [{"agg": "count_star"}],
         "preds": [{"col": "message", "op": "contains", "lit": "error"}],
     }),
+    "c4s": ("c4", lambda a: {   # c4-shaped: OTel 64 sparse cols, 3-key group-by
+        "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"},
+                   {"agg": "sum", "col": "attr_i0"}],
+        "group_by": ["service", "span_kind", "status"],
+        "preds": [{"col": "attr_i1", "op": "ge", "lit": 200_000}],
+    }),
 }
 
 
@@ -179,11 +185,13 @@ def main():
         "c1": dec_b,
         "c2s": dec_b + rank_rows * (2 * 16 + 2 * 4 + 3 + 3),
         "c3s": dec_b + rank_rows * 1,
+        "c4s": dec_b + rank_rows * (3 * 16 + 3 * 4 + 4 + 3),
     }[args.workload]
     decode_kernel = {
         "c1": "dict_count_fused",
         "c2s": "decode+filter+groupby",
         "c3s": "bytes_contains(LIKE)",
+        "c4s": "decode+filter+groupby",
     }[args.workload]
     if dd >= other_ns:
         roof_kernel, roof_bytes, roof_ns = "lz4_page_decompress", decomp_algo_bytes, dd
@@ -256,6 +264,7 @@ def main():
                     "c1": "SELECT level,count(*) GROUP BY level",
                     "c2s": "SELECT host,count(*),max(latency) WHERE ts BETWEEN ... GROUP BY host",
                     "c3s": "SELECT count(*) WHERE message LIKE '%error%'",
+                    "c4s": "SELECT service,span_kind,status,count(*),max(latency),sum(attr_i0) WHERE attr_i1>=200000 GROUP BY 1,2,3",
                 }[args.workload],
                 "rows_per_gpu": args.rows,
                 "files_per_gpu": n_files,

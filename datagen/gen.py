@@ -119,10 +119,13 @@ def _minute_batch(config: str, rng: np.random.Generator, n: int, minute: int):
         for j in range(30):
             null_frac = 0.5 + 0.4 * (j % 5) / 4
             mask = rng.random(n) < null_frac
-            vals = _dict_col(rng, n, [f"attr{j}-v{k}" for k in range(12)])
-            cols[f"attr_s{j}"] = pa.array(
-                [None if mask[i] else vals[i].as_py() for i in range(n)], type=pa.string()
-            )
+            values = [f"attr{j}-v{k}" for k in range(12)]
+            cum = np.cumsum(_zipf_probs(len(values), 1.2))
+            idx = np.searchsorted(cum, rng.random(n)).astype(np.int32)
+            np.clip(idx, 0, len(values) - 1, out=idx)
+            ind = pa.array(idx, type=pa.int32(), mask=mask)
+            cols[f"attr_s{j}"] = pa.DictionaryArray.from_arrays(
+                ind, pa.array(values)).cast(pa.string())
         for j in range(31):
             null_frac = 0.5 + 0.4 * (j % 5) / 4
             mask = rng.random(n) < null_frac

@@ -1676,8 +1676,8 @@ extern "C" int32_t gpuq_plan_metrics(gpuq_plan* p, gpuq_metrics* out) {
 gpuq_plan::~gpuq_plan() {
   for (auto& part : parts) {
     if (!part.loaded) continue;
-    hipSetDevice(part.device);
-    auto F = [](void* p) { if (p) hipFree(p); };
+    (void)hipSetDevice(part.device);
+    auto F = [](void* p) { if (p) (void)hipFree(p); };
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
@@ -1689,7 +1689,7 @@ gpuq_plan::~gpuq_plan() {
     for (auto& kv : part.d_gid) F(kv.second);
     for (auto& kv : part.d_val) F(kv.second);
     for (auto& kv : part.d_valid) F(kv.second);
-    if (part.stream) hipStreamDestroy(part.stream);
+    if (part.stream) (void)hipStreamDestroy(part.stream);
   }
 }
 

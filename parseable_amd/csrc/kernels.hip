@@ -1489,11 +1489,11 @@ size_t sort_pairs_desc(hipStream_t st, void* d_temp, size_t temp_bytes,
                        const uint32_t* rows_in, uint32_t* rows_out,
                        int64_t n) {
   size_t need = 0;
-  rocprim::radix_sort_pairs_desc(nullptr, need, keys_in, keys_out, rows_in,
-                                 rows_out, (size_t)n, 0, 64, st);
+  (void)rocprim::radix_sort_pairs_desc(nullptr, need, keys_in, keys_out,
+                                       rows_in, rows_out, (size_t)n, 0, 64, st);
   if (!d_temp) return need;
-  rocprim::radix_sort_pairs_desc(d_temp, temp_bytes, keys_in, keys_out,
-                                 rows_in, rows_out, (size_t)n, 0, 64, st);
+  (void)rocprim::radix_sort_pairs_desc(d_temp, temp_bytes, keys_in, keys_out,
+                                       rows_in, rows_out, (size_t)n, 0, 64, st);
   return need;
 }
 
