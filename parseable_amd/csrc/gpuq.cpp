@@ -169,6 +169,9 @@ struct Partition {
   uint64_t* d_table = nullptr;
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;
+  uint32_t* d_rowof = nullptr;     // dense->row map for null-bearing pages
+  uint32_t* d_present = nullptr;   // per-page dense counts
+  uint8_t* d_tmpvalid = nullptr;   // scratch validity for gid/mask-only cols
   int32_t* d_all_ids = nullptr;   // identity page-id list for the LZ4 sweep
   DevSeg* d_segs = nullptr;
   DevBr* d_brs = nullptr;
@@ -934,6 +937,9 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
     }
   }
   HIP_TRY(hipMalloc(&part.d_mask, std::max<int64_t>(part.n_rows, 16)));
+  HIP_TRY(hipMalloc(&part.d_rowof, std::max<int64_t>(part.n_rows * 4, 16)));
+  HIP_TRY(hipMalloc(&part.d_present, std::max<size_t>(part.pages.size() * 4, 16)));
+  HIP_TRY(hipMalloc(&part.d_tmpvalid, std::max<int64_t>(part.n_rows, 16)));
   if (plan->is_projection) {
     int64_t n = std::max<int64_t>(part.n_rows, 16);
     HIP_TRY(hipMalloc(&part.d_keys, n * 8));
@@ -1289,27 +1295,42 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     switch (kind) {
       case TK_DICT_GID: {
         auto it = part.d_valid.find(col);
+        uint8_t* v = it != part.d_valid.end() ? it->second : part.d_tmpvalid;
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n, v,
+                          part.d_rowof, part.d_present, part.d_err);
         launch_dict_gid(st, part.d_dec, part.d_pages, ids, n, part.d_remap,
                         part.d_gid[col],
                         it != part.d_valid.end() ? it->second : nullptr,
-                        part.d_err);
+                        part.d_rowof, part.d_present, v, part.d_err);
         break;
       }
       case TK_DICT_VAL:
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
+                          part.d_valid[col], part.d_rowof, part.d_present,
+                          part.d_err);
         launch_dict_i64(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
-                        part.d_val[col], part.d_valid[col], part.d_err);
+                        part.d_val[col], part.d_valid[col], part.d_rowof,
+                        part.d_present, part.d_err);
         break;
       case TK_PLAIN_VAL:
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
+                          part.d_valid[col], part.d_rowof, part.d_present,
+                          part.d_err);
         launch_plain_fixed(st, part.d_dec, part.d_pages, ids, n,
-                           part.d_val[col], part.d_valid[col], part.d_err);
+                           part.d_val[col], part.d_valid[col], part.d_rowof,
+                           part.d_present, part.d_err);
         break;
       case TK_DELTA_VAL:
         launch_delta_i64(st, part.d_dec, part.d_pages, ids, n,
                          part.d_val[col], part.d_valid[col], part.d_err);
         break;
       case TK_DICT_MASK:
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
+                          part.d_tmpvalid, part.d_rowof, part.d_present,
+                          part.d_err);
         launch_dict_mask(st, part.d_dec, part.d_pages, ids, n, part.d_lut,
-                         part.d_mask, part.d_err);
+                         part.d_mask, part.d_rowof, part.d_present,
+                         part.d_tmpvalid, part.d_err);
         break;
       case TK_BYTES_CONTAINS: {
         std::string needle;
@@ -1681,6 +1702,7 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
+    F(part.d_rowof); F(part.d_present); F(part.d_tmpvalid);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);
     F(part.d_res_lane); F(part.d_res_wave); F(part.d_piece_pool);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);

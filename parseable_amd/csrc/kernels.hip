@@ -405,6 +405,137 @@ __device__ inline const uint8_t* def_levels(const DevPage& pg, const uint8_t* pa
 }
 
 // ------------------------------------------------------------------
+// Wave-parallel definition-level decode for null-bearing pages: decodes
+// the RLE/bit-packed def stream into (a) valid bytes, (b) a dense->row
+// mapping rowof[row0 + k] = row of the k-th present value, and (c) the
+// page's present count. Value decoders then run their PARALLEL paths over
+// the dense stream and scatter through rowof — replacing the serial lane0
+// fallback that cost ~8ms on a 262k-row 60%-null page.
+// LDS: row bitmap 32KB (262,144 rows max) + 512 tile counts.
+#define DEF_MAX_ROWS 262144
+__global__ void __launch_bounds__(WAVE)
+k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+             const int32_t* __restrict__ ids, int n,
+             uint8_t* __restrict__ valid, uint32_t* __restrict__ rowof,
+             uint32_t* __restrict__ present, int32_t* d_error) {
+  __shared__ uint64_t bits[DEF_MAX_ROWS / 64];
+  __shared__ uint32_t tile_base[DEF_MAX_ROWS / 512];
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const int lane = threadIdx.x;
+  if (pg.num_values > DEF_MAX_ROWS) {
+    if (lane == 0) atomicExch(d_error, ERR_PAGE);
+    return;
+  }
+  const uint8_t* payload = dec + pg.dst_off;
+  {
+    const uint8_t* ds; uint32_t dl_probe; bool av;
+    def_levels(pg, payload, &ds, &dl_probe, &av);
+    if (av) return;  // all-valid page: consumers take the direct fast path
+  }
+  uint32_t dl = 0;
+  const uint8_t* p = payload;
+  if (pg.optional) {
+    memcpy(&dl, payload, 4);
+    p = payload + 4;
+  }
+  const uint32_t nv = pg.num_values, row0 = pg.row_start;
+  uint32_t nwords = (nv + 63) / 64;
+  for (uint32_t w = lane; w < nwords; w += WAVE) bits[w] = 0;
+  __builtin_amdgcn_wave_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+  if (!pg.optional) {
+    for (uint32_t w = lane; w < nwords; w += WAVE) bits[w] = ~0ull;
+  } else {
+    // lane-redundant run walk; parallel fills (def bit_width == 1)
+    const uint8_t* end = p + dl;
+    uint32_t v = 0;
+    while (v < nv && p < end) {
+      uint64_t hdr = 0; int sh = 0;
+      for (;;) {
+        uint8_t b = *p++;
+        hdr |= (uint64_t)(b & 0x7f) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+      }
+      if (hdr & 1) {
+        uint32_t groups = (uint32_t)(hdr >> 1);   // 8 values per byte
+        for (uint32_t g = lane; g < groups; g += WAVE) {
+          uint8_t byte = p[g];
+          uint32_t base = v + g * 8;
+          if (byte && base < nv)
+            atomicOr((unsigned long long*)&bits[base >> 6],
+                     ((unsigned long long)byte) << (base & 63));
+        }
+        p += groups;
+        uint32_t add = groups * 8;
+        v += (add > nv - v) ? (nv - v) : add;
+      } else {
+        uint32_t cnt = (uint32_t)(hdr >> 1);
+        uint8_t val = *p++;
+        if (cnt > nv - v) cnt = nv - v;
+        if (val) {
+          // set bits [v, v+cnt)
+          for (uint32_t w = lane; w * 64 < v + cnt; w += WAVE) {
+            uint32_t lo = w * 64, hi = lo + 64;
+            if (hi <= v || lo >= v + cnt) continue;
+            uint64_t m = ~0ull;
+            if (v > lo) m &= ~0ull << (v - lo);
+            if (v + cnt < hi) m &= ~0ull >> (hi - (v + cnt));
+            atomicOr((unsigned long long*)&bits[w], (unsigned long long)m);
+          }
+        }
+        v += cnt;
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+    // mask tail bits beyond nv
+    if (lane == 0 && (nv & 63)) bits[nv >> 6] &= (~0ull >> (64 - (nv & 63)));
+  }
+  __builtin_amdgcn_wave_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+  // tile counts (8 words = 512 rows per tile) + serial scan
+  uint32_t ntiles = (nv + 511) / 512;
+  for (uint32_t t = lane; t < ntiles; t += WAVE) {
+    uint32_t c = 0;
+    for (uint32_t w = t * 8; w < (t + 1) * 8 && w < nwords; w++)
+      c += (uint32_t)__popcll(bits[w]);
+    tile_base[t] = c;
+  }
+  __builtin_amdgcn_wave_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  if (lane == 0) {
+    uint32_t run = 0;
+    for (uint32_t t = 0; t < ntiles; t++) {
+      uint32_t c = tile_base[t];
+      tile_base[t] = run;
+      run += c;
+    }
+    present[ids[pi]] = run;
+  }
+  __builtin_amdgcn_wave_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+  // emit valid bytes + rowof (each lane owns tiles round-robin)
+  for (uint32_t t = lane; t < ntiles; t += WAVE) {
+    uint32_t rank = tile_base[t];
+    for (uint32_t w = t * 8; w < (t + 1) * 8 && w < nwords; w++) {
+      uint64_t b = bits[w];
+      for (uint32_t j = 0; j < 64; j++) {
+        uint32_t r = w * 64 + j;
+        if (r >= nv) break;
+        uint8_t ok = (uint8_t)((b >> j) & 1);
+        valid[row0 + r] = ok;
+        if (ok) rowof[row0 + rank++] = row0 + r;
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------
 // dictionary-index decode (RLE/bit-packed hybrid), one wave per page.
 // Emit policy via template: remap-to-gid, gather-i64, gather-f64, LUT-mask.
 // ------------------------------------------------------------------
@@ -434,17 +565,33 @@ struct EmitDictMask {
 
 template <class Emit>
 __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit emit,
-                                 int32_t* d_error) {
+                                 const uint32_t* __restrict__ rowof,
+                                 const uint32_t* __restrict__ present,
+                                 const uint8_t* __restrict__ valid_in,
+                                 int32_t page_id, int32_t* d_error) {
   const int lane = threadIdx.x;
   const uint8_t* def_start; uint32_t def_len; bool all_valid;
   const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
-  const uint32_t nv = pg.num_values;
+  uint32_t nv = pg.num_values;
   const uint32_t row0 = pg.row_start;
+  const uint32_t* ind = nullptr;
+  if (!all_valid) {
+    // dense mode: k_def_levels (launched just before) produced the
+    // dense->row map and the present count; run the SAME parallel run
+    // decode over the dense stream and scatter through it, then sweep
+    // null rows once
+    ind = rowof;
+    nv = present[page_id];
+    for (uint32_t r = lane; r < pg.num_values; r += WAVE)
+      if (!valid_in[row0 + r]) emit.null_at(row0 + r);
+    __builtin_amdgcn_wave_barrier();
+  }
 
-  if (all_valid) {
+  {
     int bw = *vals++;
+    auto target = [&](uint32_t k) { return ind ? ind[row0 + k] : row0 + k; };
     if (bw == 0) {  // all values are dict id 0
-      for (uint32_t i = lane; i < nv; i += WAVE) emit(row0 + i, 0);
+      for (uint32_t i = lane; i < nv; i += WAVE) emit(target(i), 0);
       return;
     }
     // lane-redundant run-header walk; data movement parallel per run
@@ -485,7 +632,7 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
             }
             if (base + k < nv) {
               if (idx >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); idx = 0; }
-              emit(row0 + base + k, idx);
+              emit(target(base + k), idx);
             }
           }
         }
@@ -500,22 +647,9 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
         p += byte_w;
         if (cnt > nv - v) cnt = nv - v;
         if (val >= dict_n) { if (lane == 0) atomicExch(d_error, ERR_DICT_RANGE); val = 0; }
-        for (uint32_t i = lane; i < cnt; i += WAVE) emit(row0 + v + i, val);
+        for (uint32_t i = lane; i < cnt; i += WAVE) emit(target(v + i), val);
         v += cnt;
       }
-    }
-  } else {
-    // nulls present: lane0 serial interleave of def levels and indices
-    if (lane != 0) return;
-    SerialRle def(def_start, def_start + def_len, 1);
-    int bw = *vals++;
-    SerialRle idx(vals, payload + pg.uncomp_size, bw);
-    for (uint32_t r = 0; r < nv; r++) {
-      if (def.next()) {
-        uint32_t ix = bw ? idx.next() : 0;
-        if (ix >= pg.dict_n) { atomicExch(d_error, ERR_DICT_RANGE); ix = 0; }
-        emit(row0 + r, ix);
-      } else emit.null_at(row0 + r);
     }
   }
 }
@@ -523,13 +657,16 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
 template <class Emit>
 __global__ void __launch_bounds__(WAVE)
 k_dict_pages(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
-             const int32_t* __restrict__ ids, int n, Emit emit, int32_t* d_error) {
+             const int32_t* __restrict__ ids, int n, Emit emit,
+             const uint32_t* rowof, const uint32_t* present,
+             const uint8_t* valid_in, int32_t* d_error) {
   int pi = blockIdx.x;
   if (pi >= n) return;
   DevPage pg = pages[ids[pi]];
   Emit e = emit;
   e.advance(pg);  // per-page aux pool offsets
-  dict_page_decode(pg, dec + pg.dst_off, e, d_error);
+  dict_page_decode(pg, dec + pg.dst_off, e, rowof, present, valid_in,
+                   ids[pi], d_error);
 }
 
 // wrappers adding per-page aux advance
@@ -667,15 +804,17 @@ __global__ void __launch_bounds__(WAVE)
 k_plain_fixed(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
               const int32_t* __restrict__ ids, int n,
               int64_t* __restrict__ out, uint8_t* __restrict__ valid,
-              int32_t* d_error) {
+              const uint32_t* __restrict__ rowof,
+              const uint32_t* __restrict__ present, int32_t* d_error) {
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
   const int lane = threadIdx.x;
   const uint8_t* def_start; uint32_t def_len; bool all_valid;
   const uint8_t* vals = def_levels(pg, dec + pg.dst_off, &def_start, &def_len, &all_valid);
-  const uint32_t nv = pg.num_values, row0 = pg.row_start;
+  const uint32_t row0 = pg.row_start;
   if (all_valid) {
+    uint32_t nv = pg.num_values;
     for (uint32_t i = lane; i < nv; i += WAVE) {
       int64_t v;
       memcpy(&v, vals + (size_t)i * 8, 8);
@@ -683,15 +822,12 @@ k_plain_fixed(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages
       if (valid) valid[row0 + i] = 1;
     }
   } else {
-    if (lane != 0) return;
-    SerialRle def(def_start, def_start + def_len, 1);
-    const uint8_t* q = vals;
-    for (uint32_t r = 0; r < nv; r++) {
-      if (def.next()) {
-        int64_t v; memcpy(&v, q, 8); q += 8;
-        out[row0 + r] = v;
-        if (valid) valid[row0 + r] = 1;
-      } else if (valid) valid[row0 + r] = 0;
+    // dense scatter through the k_def_levels mapping (valid already set)
+    uint32_t nd = present[ids[pi]];
+    for (uint32_t i = lane; i < nd; i += WAVE) {
+      int64_t v;
+      memcpy(&v, vals + (size_t)i * 8, 8);
+      out[rowof[row0 + i]] = v;
     }
   }
 }
@@ -1376,31 +1512,42 @@ void launch_brres_wave(hipStream_t st, uint8_t* dec, const DevBrRes* recs,
                        const DevPiece* pieces, int n) {
   if (n) hipLaunchKernelGGL(k_brres_wave, dim3(n), dim3(WAVE), 0, st, dec, recs, pieces, n);
 }
+void launch_def_levels(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                       const int32_t* ids, int n, uint8_t* valid,
+                       uint32_t* rowof, uint32_t* present, int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_def_levels, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, valid, rowof, present, d_err);
+}
 void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int32_t* remap_pool,
-                     int32_t* out, uint8_t* valid, int32_t* d_err) {
+                     int32_t* out, uint8_t* valid, const uint32_t* rowof,
+                     const uint32_t* present, const uint8_t* valid_in,
+                     int32_t* d_err) {
   if (!n) return;
   EmitGidP e{}; e.pool = remap_pool; e.out = out; e.valid = valid;
-  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, rowof, present, valid_in, d_err);
 }
 void launch_dict_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int64_t* dictv_pool,
-                     int64_t* out, uint8_t* valid, int32_t* d_err) {
+                     int64_t* out, uint8_t* valid, const uint32_t* rowof,
+                     const uint32_t* present, int32_t* d_err) {
   if (!n) return;
   EmitDictI64P e{}; e.pool = dictv_pool; e.out = out; e.valid = valid;
-  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, rowof, present, valid, d_err);
 }
 void launch_dict_mask(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, const uint8_t* lut_pool,
-                      uint8_t* mask, int32_t* d_err) {
+                      uint8_t* mask, const uint32_t* rowof,
+                      const uint32_t* present, const uint8_t* valid_in,
+                      int32_t* d_err) {
   if (!n) return;
   EmitDictMaskP e{}; e.pool = lut_pool; e.mask = mask;
-  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, rowof, present, valid_in, d_err);
 }
 void launch_plain_fixed(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                         const int32_t* ids, int n, int64_t* out, uint8_t* valid,
+                        const uint32_t* rowof, const uint32_t* present,
                         int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_plain_fixed, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, d_err);
+  if (n) hipLaunchKernelGGL(k_plain_fixed, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, rowof, present, d_err);
 }
 void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
