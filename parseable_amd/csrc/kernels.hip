@@ -465,9 +465,19 @@ k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
         for (uint32_t g = lane; g < groups; g += WAVE) {
           uint8_t byte = p[g];
           uint32_t base = v + g * 8;
-          if (byte && base < nv)
-            atomicOr((unsigned long long*)&bits[base >> 6],
-                     ((unsigned long long)byte) << (base & 63));
+          if (byte && base < nv) {
+            // base need not be 8-aligned (an RLE run of arbitrary length may
+            // precede) — clip to nv and spill across the word boundary
+            uint64_t bb = byte;
+            if (base + 8 > nv) bb &= (1ull << (nv - base)) - 1;
+            uint32_t w0 = base >> 6, sh = base & 63;
+            if (bb << sh)
+              atomicOr((unsigned long long*)&bits[w0],
+                       (unsigned long long)(bb << sh));
+            if (sh && (bb >> (64 - sh)))
+              atomicOr((unsigned long long*)&bits[w0 + 1],
+                       (unsigned long long)(bb >> (64 - sh)));
+          }
         }
         p += groups;
         uint32_t add = groups * 8;
