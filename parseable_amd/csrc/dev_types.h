@@ -43,6 +43,8 @@ struct DevBrRes {      // literal-resolved record (absolute dec offsets)
   uint32_t piece_start, piece_n;
 };
 struct DevPiece { uint64_t src; uint32_t len; uint32_t _pad; };
+// litpar literal copy: raw[src..src+len) -> dec[dst..dst+len)  (absolute)
+struct DevLit { uint64_t src, dst; uint32_t len; uint32_t _pad; };
 
 // comparison kernel ops (matches gpuq_op order where applicable)
 enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };

@@ -153,6 +153,7 @@ struct Partition {
   // literal-resolved records (single launch; meta.cpp lz4_walk)
   std::vector<DevBrRes> res_lane, res_wave;
   std::vector<DevPiece> piece_pool;
+  std::vector<DevLit> lits_lane, lits_wave;   // litpar pages (meta.h)
   uint64_t raw_bytes = 0, dec_bytes = 0;
   int64_t bytes_scanned = 0, rowgroup_bytes_total = 0;
 
@@ -179,6 +180,8 @@ struct Partition {
   DevBrRes* d_res_lane = nullptr;
   DevBrRes* d_res_wave = nullptr;
   DevPiece* d_piece_pool = nullptr;
+  DevLit* d_lits_lane = nullptr;
+  DevLit* d_lits_wave = nullptr;
   // projection-scan buffers
   int64_t* d_keys = nullptr;
   int64_t* d_keys_sorted = nullptr;
@@ -671,6 +674,15 @@ extern "C" gpuq_plan* gpuq_plan_build(
           if (!raw_page) {
             try {
               lp = lz4_walk(praw, pi.comp_size, pi.uncomp_size, 8192);
+              // dense short-sequence pages (LZ4 over near-random dict
+              // indices) degenerate the segment kernel into a serial token
+              // parse — switch those to the all-literal/all-resolved plan
+              if (!lp.fallback && lp.n_seq >= 256 &&
+                  (uint32_t)pi.uncomp_size / lp.n_seq < 96) {
+                Lz4Plan lp2 = lz4_walk(praw, pi.comp_size, pi.uncomp_size,
+                                       8192, /*litpar=*/true);
+                if (!lp2.fallback) lp = std::move(lp2);
+              }
             } catch (const std::exception&) {
               if (pi.comp_size == pi.uncomp_size) raw_page = true;  // stored raw
               else throw;
@@ -693,6 +705,10 @@ extern "C" gpuq_plan* gpuq_plan_build(
               d2.out_len = sg.out_len;
               d2.big = sg.big;
               part.segs.push_back(d2);
+            }
+            for (const auto& lt : lp.lits) {
+              DevLit dl{dp.src_off + lt.src, dp.dst_off + lt.dst, lt.len, 0};
+              (lt.len <= 256 ? part.lits_lane : part.lits_wave).push_back(dl);
             }
             if (lp.fallback) {
               // piece explosion: serial windowed wave per page
@@ -970,6 +986,10 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
               (void**)&part.d_res_wave);
   upload_pool(part.piece_pool.data(), part.piece_pool.size() * sizeof(DevPiece),
               (void**)&part.d_piece_pool);
+  upload_pool(part.lits_lane.data(), part.lits_lane.size() * sizeof(DevLit),
+              (void**)&part.d_lits_lane);
+  upload_pool(part.lits_wave.data(), part.lits_wave.size() * sizeof(DevLit),
+              (void**)&part.d_lits_wave);
   // needle buffer (first CONTAINS pred; one per plan supported per column set)
   std::string needle;
   for (auto& pp : plan->preds)
@@ -1267,6 +1287,11 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // 1. decompress: parallel segments, then ordered backref resolution
   launch_lz4_seg(st, part.d_raw, part.d_dec, part.d_segs,
                  (int)part.segs.size(), part.d_err);
+  // litpar pages: flat literal copies (independent of the segment pass)
+  launch_lit_lane(st, part.d_raw, part.d_dec, part.d_lits_lane,
+                  (int64_t)part.lits_lane.size());
+  launch_lit_wave(st, part.d_raw, part.d_dec, part.d_lits_wave,
+                  (int)part.lits_wave.size());
   // deferred-match resolution: host-resolved records are independent —
   // one launch each (kernel boundary after phase 1 gives coherence);
   // piece-explosion pages fall back to the serial windowed wave
@@ -1703,6 +1728,7 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_present); F(part.d_tmpvalid);
+    F(part.d_lits_lane); F(part.d_lits_wave);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);
     F(part.d_res_lane); F(part.d_res_wave); F(part.d_piece_pool);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);

@@ -252,6 +252,43 @@ k_brres_wave(uint8_t* __restrict__ dec, const DevBrRes* __restrict__ recs,
   }
 }
 
+// litpar literal copies: one record per sequence's literal run (host walk,
+// meta.cpp lz4_walk litpar mode). Records are independent; short runs get a
+// lane each (records are dst-sorted, so adjacent lanes touch adjacent
+// memory), long runs a wave.
+__global__ void k_lit_lane(const uint8_t* __restrict__ raw,
+                           uint8_t* __restrict__ dec,
+                           const DevLit* __restrict__ lits, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const DevLit L = lits[i];
+  const uint8_t* s = raw + L.src;
+  uint8_t* o = dec + L.dst;
+  uint32_t k = 0;
+  for (; k + 8 <= L.len; k += 8) {
+    uint64_t w;
+    __builtin_memcpy(&w, s + k, 8);
+    __builtin_memcpy(o + k, &w, 8);
+  }
+  for (; k < L.len; k++) o[k] = s[k];
+}
+__global__ void __launch_bounds__(WAVE)
+k_lit_wave(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
+           const DevLit* __restrict__ lits, int n) {
+  if (blockIdx.x >= (unsigned)n) return;
+  const DevLit L = lits[blockIdx.x];
+  const uint8_t* s = raw + L.src;
+  uint8_t* o = dec + L.dst;
+  uint32_t nw = L.len / 8;
+  for (uint32_t w = threadIdx.x; w < nw; w += WAVE) {
+    uint64_t v;
+    __builtin_memcpy(&v, s + w * 8, 8);
+    __builtin_memcpy(o + w * 8, &v, 8);
+  }
+  uint32_t t = nw * 8 + threadIdx.x;
+  if (t < L.len) o[t] = s[t];
+}
+
 #define BR_WIN 16384
 __global__ void __launch_bounds__(WAVE)
 k_lz4_backrefs(uint8_t* __restrict__ dec, const DevBr* __restrict__ brs,
@@ -1509,6 +1546,14 @@ __global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
 void launch_lz4_seg(hipStream_t st, const uint8_t* raw, uint8_t* dec,
                     const DevSeg* segs, int n, int32_t* d_err) {
   if (n) hipLaunchKernelGGL(k_lz4_seg, dim3(n), dim3(WAVE), 0, st, raw, dec, segs, n, d_err);
+}
+void launch_lit_lane(hipStream_t st, const uint8_t* raw, uint8_t* dec,
+                     const DevLit* lits, int64_t n) {
+  if (n) hipLaunchKernelGGL(k_lit_lane, dim3((int)((n + 255) / 256)), dim3(256), 0, st, raw, dec, lits, n);
+}
+void launch_lit_wave(hipStream_t st, const uint8_t* raw, uint8_t* dec,
+                     const DevLit* lits, int n) {
+  if (n) hipLaunchKernelGGL(k_lit_wave, dim3(n), dim3(WAVE), 0, st, raw, dec, lits, n);
 }
 void launch_lz4_backrefs(hipStream_t st, uint8_t* dec, const DevBr* brs,
                          const DevPageBr* pages, int n) {

@@ -6,6 +6,10 @@
 namespace gpuq {
 void launch_lz4_seg(hipStream_t, const uint8_t* raw, uint8_t* dec,
                     const DevSeg*, int n, int32_t* d_err);
+void launch_lit_lane(hipStream_t, const uint8_t* raw, uint8_t* dec,
+                     const DevLit* lits, int64_t n);
+void launch_lit_wave(hipStream_t, const uint8_t* raw, uint8_t* dec,
+                     const DevLit* lits, int n);
 void launch_lz4_backrefs(hipStream_t, uint8_t* dec, const DevBr*,
                          const DevPageBr*, int n);
 void launch_brres_lane(hipStream_t, uint8_t* dec, const DevBrRes*,

@@ -278,8 +278,9 @@ int lz4_decompress_host(const uint8_t* src, size_t src_len,
 }
 
 Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
-                 uint32_t seg_max) {
+                 uint32_t seg_max, bool litpar) {
   Lz4Plan plan;
+  plan.litpar = litpar;
   size_t s = 0, d = 0;
   uint32_t seg_s = 0, seg_d = 0;
 
@@ -377,6 +378,7 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
   };
   while (s < comp && d < uncomp) {
     size_t seq_s = s, seq_d = d;
+    plan.n_seq++;
     uint8_t token = src[s++];
     size_t lit = token >> 4;
     if (lit == 15) {
@@ -384,6 +386,7 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
       do { if (s >= comp) throw std::runtime_error("lz4 walk: eof"); b = src[s++]; lit += b; } while (b == 255);
     }
     if (s + lit > comp || d + lit > uncomp) throw std::runtime_error("lz4 walk: overrun");
+    size_t lit_src = s;
     size_t ml = 0, off = 0;
     bool has_match = false;
     size_t after_lit_s = s + lit;
@@ -406,6 +409,16 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
     size_t seq_out = lit + ml;
     if (d + seq_out > uncomp && has_match) throw std::runtime_error("lz4 walk: output overrun");
 
+    if (litpar) {
+      if (lit) plan.lits.push_back({(uint32_t)d, (uint32_t)lit_src, (uint32_t)lit});
+      d += lit;
+      if (has_match && ml) {
+        defer_match(d, off, ml);
+        if (plan.fallback) return plan;  // caller keeps the segment plan
+        d += ml;
+      }
+      continue;
+    }
     if (seq_out > seg_max) {
       close_segment(seq_s, seq_d, false);
       if (has_match && ml) defer_match(d + lit, off, ml);
@@ -427,7 +440,7 @@ Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
     }
   }
   if (d != uncomp) throw std::runtime_error("lz4 walk: size mismatch");
-  close_segment(s, d, false);
+  if (!litpar) close_segment(s, d, false);
   return plan;
 }
 

@@ -111,15 +111,29 @@ struct Lz4Resolved {
 struct Lz4Piece {
   uint32_t src, len;               // page-relative literal-region source
 };
+// litpar mode: pages whose content is a dense stream of short sequences
+// (e.g. LZ4 over near-random dictionary indices) degenerate the segment
+// kernel into a serial token parse. For those pages the walk instead emits
+// one literal-copy record per sequence (compressed src -> decompressed dst,
+// both page-relative) and defers EVERY match to the resolved-record pass —
+// no segments, no serial parse on the GPU at all.
+struct Lz4Lit {
+  uint32_t dst, src, len;
+};
 struct Lz4Plan {
   std::vector<Lz4Segment> segs;
   std::vector<Lz4Backref> backrefs;    // fallback (serial window) records
   std::vector<Lz4Resolved> resolved;   // literal-resolved records
   std::vector<Lz4Piece> pieces;
+  std::vector<Lz4Lit> lits;            // litpar mode only
   bool fallback = false;               // use `backrefs` for the whole page
+  bool litpar = false;
+  uint32_t n_seq = 0;                  // sequence count (mode decision)
 };
 // Throws on malformed streams. seg_max must match the kernel's LDS buffer.
+// litpar=true produces the all-literal/all-resolved plan (fallback=true if
+// a record's piece decomposition explodes — caller keeps the segment plan).
 Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
-                 uint32_t seg_max);
+                 uint32_t seg_max, bool litpar = false);
 
 }  // namespace gpuq

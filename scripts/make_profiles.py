@@ -39,7 +39,7 @@ def main():
           "Units: FETCH_SIZE/WRITE_SIZE counters are KB; fetch doubled per the",
           "gfx950 half-reporting correction (MI355X_MICROARCH.md §HBM).",
           "Per-launch values = total / dispatch count.\n"]
-    for wl in ["c1", "c2s", "c3s"]:
+    for wl in ["c1", "c2s", "c3s", "c4s"]:
         for f in ["kernel_stats", "kernel_trace"]:
             s = os.path.join(SRC, f"{wl}_stats_{f}.csv")
             if os.path.exists(s) and f == "kernel_stats":
