@@ -63,7 +63,7 @@ def main():
             if fpl + wpl < 1e6:
                 continue
             md.append(f"| {k} | {n} | {fpl / 1e9:.3f} | {wpl / 1e9:.3f} |")
-            short = ("lz4_page_decompress" if "lz4_seg" in k or "lz4_pages" in k else
+            short = ("lz4_page_decompress" if "lz4_seg" in k or "lz4_pages" in k or "lit_" in k else
                      "lz4_backrefs" if "backref" in k or "brres" in k else
                      "dict_count_fused" if "dict_count" in k else
                      "bytes_contains(LIKE)" if "contains" in k else
