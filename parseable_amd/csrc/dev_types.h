@@ -45,6 +45,18 @@ struct DevBrRes {      // literal-resolved record (absolute dec offsets)
 struct DevPiece { uint64_t src; uint32_t len; uint32_t _pad; };
 // litpar literal copy: raw[src..src+len) -> dec[dst..dst+len)  (absolute)
 struct DevLit { uint64_t src, dst; uint32_t len; uint32_t _pad; };
+// contains window: a value-aligned run of consecutive non-null values of one
+// PLAIN byte-array page, <= CWIN bytes total (or a single oversized value,
+// nbytes > CWIN). Host builds these at load time by decompressing the page
+// and walking the length chain once — the kernel has NO serial spine.
+struct DevCWin {
+  uint64_t src;       // absolute dec-arena offset of the first length field
+  uint64_t starts;    // index into the u16 window-relative starts pool
+  uint32_t nbytes;    // total bytes ([len][bytes] records) in this window
+  uint32_t n_values;
+  uint32_t dense0;    // non-null value index of value 0 within its page
+  int32_t page_id;
+};
 
 // comparison kernel ops (matches gpuq_op order where applicable)
 enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };

@@ -22,7 +22,9 @@
 #include <unistd.h>
 
 #include <algorithm>
+#include <atomic>
 #include <cstring>
+#include <thread>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -154,6 +156,10 @@ struct Partition {
   std::vector<DevBrRes> res_lane, res_wave;
   std::vector<DevPiece> piece_pool;
   std::vector<DevLit> lits_lane, lits_wave;   // litpar pages (meta.h)
+  // contains windows (host length-chain walk at load time; dev_types.h)
+  std::vector<DevCWin> cwins;
+  std::vector<uint16_t> cstarts;
+  std::map<int, std::pair<uint32_t, uint32_t>> cwin_ranges;  // col -> (off, n)
   uint64_t raw_bytes = 0, dec_bytes = 0;
   int64_t bytes_scanned = 0, rowgroup_bytes_total = 0;
 
@@ -182,6 +188,8 @@ struct Partition {
   DevPiece* d_piece_pool = nullptr;
   DevLit* d_lits_lane = nullptr;
   DevLit* d_lits_wave = nullptr;
+  DevCWin* d_cwins = nullptr;
+  uint16_t* d_cstarts = nullptr;
   // projection-scan buffers
   int64_t* d_keys = nullptr;
   int64_t* d_keys_sorted = nullptr;
@@ -631,6 +639,15 @@ extern "C" gpuq_plan* gpuq_plan_build(
     }
 
     // device page descriptors + aux pools + task lists
+    struct CItem {
+      const uint8_t* praw;
+      int32_t comp, uncomp, page_id;
+      uint64_t dst_off;
+      uint32_t num_values;
+      bool optional, raw_codec;
+      int col;
+    };
+    std::vector<CItem> citems;
     for (auto& t : part.chunks) {
       auto& c = plan->cols[t.col_idx];
       uint32_t remap_base = (uint32_t)part.remap_pool.size();
@@ -759,6 +776,10 @@ extern "C" gpuq_plan* gpuq_plan_build(
             if ((int)c.contains_preds.size() != (int)c.lut_preds.size())
               throw std::runtime_error("non-LIKE predicate on PLAIN utf8 pages: next row");
             part.tasks[{TK_BYTES_CONTAINS, t.col_idx}].push_back(page_id);
+            citems.push_back({mf.data + pi.payload_off, pi.comp_size,
+                              pi.uncomp_size, page_id, dp.dst_off,
+                              (uint32_t)pi.num_values, dp.optional != 0,
+                              dp.raw_copy != 0, t.col_idx});
           } else throw std::runtime_error("unsupported encoding for string predicate");
         }
         part.pages.push_back(dp);
@@ -767,6 +788,138 @@ extern "C" gpuq_plan* gpuq_plan_build(
         throw std::runtime_error("page rows mismatch");
     }
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
+
+    // CONTAINS window build: decompress each PLAIN byte-array page once on
+    // the host (load-time, parallel over pages), walk the [u32 len][bytes]
+    // chain, and emit value-aligned <=16KB windows + a u16 start offset per
+    // value — the kernel-side serial length walk disappears entirely.
+    if (!citems.empty()) {
+      struct CLocal {
+        std::vector<DevCWin> wins;
+        std::vector<uint16_t> starts;
+      };
+      std::vector<CLocal> locals(citems.size());
+      std::atomic<size_t> next{0};
+      std::atomic<bool> failed{false};
+      auto worker = [&]() {
+        std::vector<uint8_t> img, defs;
+        for (;;) {
+          size_t i = next.fetch_add(1);
+          if (i >= citems.size() || failed.load()) return;
+          const auto& it = citems[i];
+          auto& L = locals[i];
+          const uint8_t* data = it.praw;
+          if (!it.raw_codec) {
+            img.resize(it.uncomp);
+            if (lz4_decompress_host(it.praw, it.comp, img.data(), img.size())
+                != it.uncomp) { failed.store(true); return; }
+            data = img.data();
+          }
+          uint32_t pos = 0;
+          const uint32_t nv = it.num_values;
+          bool has_def = false;
+          if (it.optional) {
+            uint32_t dl;
+            std::memcpy(&dl, data, 4);
+            defs.assign(nv, 1);
+            // def bit-width 1 RLE/bit-packed hybrid (parquet-format RLE)
+            const uint8_t* p = data + 4;
+            const uint8_t* end = p + dl;
+            uint32_t v = 0;
+            while (v < nv && p < end) {
+              uint64_t hdr = 0;
+              int sh = 0;
+              for (;;) {
+                uint8_t b = *p++;
+                hdr |= (uint64_t)(b & 0x7f) << sh;
+                if (!(b & 0x80)) break;
+                sh += 7;
+              }
+              if (hdr & 1) {
+                uint32_t groups = (uint32_t)(hdr >> 1);
+                for (uint32_t g = 0; g < groups; g++) {
+                  uint8_t byte = p[g];
+                  for (int j = 0; j < 8; j++) {
+                    uint32_t idx = v + g * 8 + j;
+                    if (idx < nv) defs[idx] = (byte >> j) & 1;
+                  }
+                }
+                p += groups;
+                uint32_t add = groups * 8;
+                v += add > nv - v ? nv - v : add;
+              } else {
+                uint32_t cnt = (uint32_t)(hdr >> 1);
+                uint8_t val = *p++;
+                if (cnt > nv - v) cnt = nv - v;
+                std::fill(defs.begin() + v, defs.begin() + v + cnt, val);
+                v += cnt;
+              }
+            }
+            pos = 4 + dl;
+            has_def = true;
+          }
+          uint32_t dense = 0;
+          int64_t wfirst = -1;
+          uint32_t wbytes = 0, wvals = 0, wdense0 = 0;
+          auto flush = [&]() {
+            if (wfirst < 0) return;
+            L.wins.push_back({it.dst_off + (uint64_t)wfirst,
+                              (uint64_t)(L.starts.size() - wvals), wbytes,
+                              wvals, wdense0, it.page_id});
+            wfirst = -1;
+            wbytes = wvals = 0;
+          };
+          for (uint32_t vI = 0; vI < nv; vI++) {
+            if (has_def && !defs[vI]) continue;
+            if (pos + 4 > (uint32_t)it.uncomp) { failed.store(true); return; }
+            uint32_t len;
+            std::memcpy(&len, data + pos, 4);
+            uint64_t rec = 4ull + len;
+            if (pos + rec > (uint64_t)it.uncomp) { failed.store(true); return; }
+            if (rec > 16384) {  // oversized value: its own window
+              flush();
+              L.starts.push_back(0);
+              L.wins.push_back({it.dst_off + pos, L.starts.size() - 1,
+                                (uint32_t)rec, 1, dense, it.page_id});
+            } else {
+              if (wfirst >= 0 && wbytes + rec > 16384) flush();
+              if (wfirst < 0) { wfirst = pos; wdense0 = dense; }
+              L.starts.push_back((uint16_t)(pos - wfirst));
+              wbytes += (uint32_t)rec;
+              wvals++;
+            }
+            pos += (uint32_t)rec;
+            dense++;
+          }
+          flush();
+        }
+      };
+      size_t nthreads = std::min<size_t>(
+          std::max(1u, std::thread::hardware_concurrency()), citems.size());
+      std::vector<std::thread> pool;
+      for (size_t i = 0; i < nthreads; i++) pool.emplace_back(worker);
+      for (auto& th : pool) th.join();
+      if (failed.load())
+        throw std::runtime_error("contains window walk failed (corrupt page)");
+      // append in task order, per column
+      std::map<int32_t, size_t> by_page;
+      for (size_t i = 0; i < citems.size(); i++) by_page[citems[i].page_id] = i;
+      for (auto& [key, ids] : part.tasks) {
+        if (key.first != TK_BYTES_CONTAINS) continue;
+        uint32_t off = (uint32_t)part.cwins.size();
+        for (int32_t pid : ids) {
+          const auto& L = locals[by_page.at(pid)];
+          uint64_t sbase = part.cstarts.size();
+          part.cstarts.insert(part.cstarts.end(), L.starts.begin(),
+                              L.starts.end());
+          for (DevCWin w : L.wins) {
+            w.starts += sbase;
+            part.cwins.push_back(w);
+          }
+        }
+        part.cwin_ranges[key.second] = {off, (uint32_t)part.cwins.size() - off};
+      }
+    }
   }
 
   // bin bounds from row-group footer stats of the source column
@@ -990,6 +1143,10 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
               (void**)&part.d_lits_lane);
   upload_pool(part.lits_wave.data(), part.lits_wave.size() * sizeof(DevLit),
               (void**)&part.d_lits_wave);
+  upload_pool(part.cwins.data(), part.cwins.size() * sizeof(DevCWin),
+              (void**)&part.d_cwins);
+  upload_pool(part.cstarts.data(), part.cstarts.size() * sizeof(uint16_t),
+              (void**)&part.d_cstarts);
   // needle buffer (first CONTAINS pred; one per plan supported per column set)
   std::string needle;
   for (auto& pp : plan->preds)
@@ -1321,7 +1478,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       case TK_DICT_GID: {
         auto it = part.d_valid.find(col);
         uint8_t* v = it != part.d_valid.end() ? it->second : part.d_tmpvalid;
-        launch_def_levels(st, part.d_dec, part.d_pages, ids, n, v,
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n, v, nullptr,
                           part.d_rowof, part.d_present, part.d_err);
         launch_dict_gid(st, part.d_dec, part.d_pages, ids, n, part.d_remap,
                         part.d_gid[col],
@@ -1331,16 +1488,16 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       }
       case TK_DICT_VAL:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
-                          part.d_valid[col], part.d_rowof, part.d_present,
-                          part.d_err);
+                          part.d_valid[col], nullptr, part.d_rowof,
+                          part.d_present, part.d_err);
         launch_dict_i64(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
                         part.d_val[col], part.d_valid[col], part.d_rowof,
                         part.d_present, part.d_err);
         break;
       case TK_PLAIN_VAL:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
-                          part.d_valid[col], part.d_rowof, part.d_present,
-                          part.d_err);
+                          part.d_valid[col], nullptr, part.d_rowof,
+                          part.d_present, part.d_err);
         launch_plain_fixed(st, part.d_dec, part.d_pages, ids, n,
                            part.d_val[col], part.d_valid[col], part.d_rowof,
                            part.d_present, part.d_err);
@@ -1351,8 +1508,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         break;
       case TK_DICT_MASK:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
-                          part.d_tmpvalid, part.d_rowof, part.d_present,
-                          part.d_err);
+                          part.d_tmpvalid, nullptr, part.d_rowof,
+                          part.d_present, part.d_err);
         launch_dict_mask(st, part.d_dec, part.d_pages, ids, n, part.d_lut,
                          part.d_mask, part.d_rowof, part.d_present,
                          part.d_tmpvalid, part.d_err);
@@ -1361,9 +1518,16 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         std::string needle;
         for (int pidx : plan->cols[col].contains_preds)
           needle = plan->preds[pidx].str_lit;  // single CONTAINS per col
-        launch_bytes_contains(st, part.d_dec, part.d_pages, ids, n,
-                              part.d_needle, (int)needle.size(),
-                              part.d_mask, part.d_err);
+        // def levels zero null rows in the mask and build the dense->row
+        // map; the window kernel then has no serial work at all
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
+                          part.d_tmpvalid, part.d_mask, part.d_rowof,
+                          part.d_present, part.d_err);
+        auto rng = part.cwin_ranges.at(col);
+        launch_contains_win(st, part.d_dec, part.d_cwins + rng.first,
+                            (int)rng.second, part.d_pages, part.d_cstarts,
+                            part.d_needle, (int)needle.size(), part.d_rowof,
+                            part.d_mask);
         break;
       }
     }
@@ -1729,6 +1893,7 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_present); F(part.d_tmpvalid);
     F(part.d_lits_lane); F(part.d_lits_wave);
+    F(part.d_cwins); F(part.d_cstarts);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);
     F(part.d_res_lane); F(part.d_res_wave); F(part.d_piece_pool);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);

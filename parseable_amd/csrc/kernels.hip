@@ -453,7 +453,8 @@ __device__ inline const uint8_t* def_levels(const DevPage& pg, const uint8_t* pa
 __global__ void __launch_bounds__(WAVE)
 k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
              const int32_t* __restrict__ ids, int n,
-             uint8_t* __restrict__ valid, uint32_t* __restrict__ rowof,
+             uint8_t* __restrict__ valid, uint8_t* __restrict__ null_mask,
+             uint32_t* __restrict__ rowof,
              uint32_t* __restrict__ present, int32_t* d_error) {
   __shared__ uint64_t bits[DEF_MAX_ROWS / 64];
   __shared__ uint32_t tile_base[DEF_MAX_ROWS / 512];
@@ -576,6 +577,7 @@ k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
         if (r >= nv) break;
         uint8_t ok = (uint8_t)((b >> j) & 1);
         valid[row0 + r] = ok;
+        if (null_mask && !ok) null_mask[row0 + r] = 0;  // NULL never matches
         if (ok) rowof[row0 + rank++] = row0 + r;
       }
     }
@@ -1239,6 +1241,114 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
 // ------------------------------------------------------------------
 // row-parallel comparison on i64 arrays
 // ------------------------------------------------------------------
+// ------------------------------------------------------------------
+// window-parallel CONTAINS over PLAIN byte-array pages: one block per
+// DevCWin. The host decompressed each page ONCE at load time, walked the
+// [u32 len][bytes] chain, and emitted value-aligned <=16KB windows plus a
+// u16 start offset per value — so the kernel has NO serial spine at all:
+// stage window -> all 256 threads sweep needle candidates (zero-byte trick,
+// verified from byte 0) into a match-start bitmap -> thread-parallel
+// per-value bitmap range checks. Null rows are zeroed by the preceding
+// k_def_levels launch (null_mask).
+__global__ void __launch_bounds__(CTHREADS)
+k_contains_win(const uint8_t* __restrict__ dec,
+               const DevCWin* __restrict__ wins, int n,
+               const DevPage* __restrict__ pages,
+               const uint16_t* __restrict__ starts_pool,
+               const uint8_t* __restrict__ needle, int nlen,
+               const uint32_t* __restrict__ rowof, uint8_t* __restrict__ mask) {
+  __shared__ uint8_t win[CWIN];
+  __shared__ uint32_t bm[CWIN / 32];
+  __shared__ int shit;
+  if (blockIdx.x >= (unsigned)n) return;
+  const DevCWin W = wins[blockIdx.x];
+  const DevPage pg = pages[W.page_id];
+  const uint32_t row0 = pg.row_start;
+  bool direct = true;  // row mapping: probe mirrors the value decoders
+  if (pg.optional) {
+    const uint8_t* ds; uint32_t dl; bool av;
+    def_levels(pg, dec + pg.dst_off, &ds, &dl, &av);
+    direct = av;
+  }
+  const uint8_t* src = dec + W.src;
+  if (W.nbytes > CWIN) {  // single oversized value: strided global scan
+    uint32_t vl;
+    __builtin_memcpy(&vl, src, 4);
+    if (threadIdx.x == 0) shit = (nlen == 0);
+    __syncthreads();
+    if (nlen && vl >= (uint32_t)nlen) {
+      for (uint32_t j = threadIdx.x; j + (uint32_t)nlen <= vl && !shit;
+           j += CTHREADS) {
+        int k = 0;
+        while (k < nlen && src[4 + j + k] == needle[k]) k++;
+        if (k == nlen) shit = 1;
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint32_t di = W.dense0;
+      mask[direct ? row0 + di : rowof[row0 + di]] &= (uint8_t)(shit != 0);
+    }
+    return;
+  }
+  {  // stage the full window unconditionally (arena is padded by CWIN)
+    uint32_t v[CWIN / (CTHREADS * 4)];
+#pragma unroll
+    for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
+      __builtin_memcpy(&v[k], src + threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u), 4);
+#pragma unroll
+    for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
+      *(uint32_t*)&win[threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u)] = v[k];
+  }
+  for (uint32_t i = threadIdx.x; i < CWIN / 32; i += CTHREADS) bm[i] = 0;
+  __syncthreads();
+  if (nlen) {
+    const uint32_t pat = 0x01010101u * needle[0];
+    for (uint32_t p = threadIdx.x * 4u; p < W.nbytes; p += CTHREADS * 4u) {
+      uint32_t w = *(const uint32_t*)&win[p];
+      uint32_t x = w ^ pat;
+      uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
+      while (cand) {
+        int b = (__builtin_ctz(cand)) >> 3;
+        cand &= cand - 1;
+        uint32_t pos = p + b;
+        if (pos + nlen <= CWIN) {
+          int k = 0;  // verify from 0: the borrow trick has false positives
+          while (k < nlen && win[pos + k] == needle[k]) k++;
+          if (k == nlen) atomicOr(&bm[pos >> 5], 1u << (pos & 31));
+        }
+      }
+    }
+  }
+  __syncthreads();
+  const uint32_t* win32 = (const uint32_t*)win;
+  for (uint32_t i = threadIdx.x; i < W.n_values; i += CTHREADS) {
+    uint32_t o = starts_pool[W.starts + i];
+    uint32_t sh = (o & 3) * 8;
+    uint32_t vl = win32[o >> 2] >> sh;
+    if (sh) vl |= win32[(o >> 2) + 1] << (32 - sh);
+    uint8_t hit = 0;
+    if (nlen == 0) {
+      hit = 1;
+    } else if (vl >= (uint32_t)nlen) {
+      uint32_t lo = o + 4, hi = o + 4 + vl - nlen;  // inclusive starts
+      uint32_t w0 = lo >> 5, w1 = hi >> 5;
+      if (w0 == w1) {
+        uint32_t m = (hi - lo == 31) ? ~0u
+                                     : (((1u << (hi - lo + 1)) - 1) << (lo & 31));
+        hit = (bm[w0] & m) != 0;
+      } else {
+        uint32_t m0 = ~0u << (lo & 31);
+        uint32_t m1 = ((hi & 31) == 31) ? ~0u : ((1u << ((hi & 31) + 1)) - 1);
+        hit = ((bm[w0] & m0) != 0) | ((bm[w1] & m1) != 0);
+        for (uint32_t w = w0 + 1; w < w1 && !hit; w++) hit |= (bm[w] != 0);
+      }
+    }
+    uint32_t di = W.dense0 + i;
+    mask[direct ? row0 + di : rowof[row0 + di]] &= hit;
+  }
+}
+
 __global__ void k_cmp_i64(const int64_t* __restrict__ col,
                           const uint8_t* __restrict__ valid,
                           int64_t lo, int64_t hi, int mode, int hi_exclusive,
@@ -1567,10 +1677,17 @@ void launch_brres_wave(hipStream_t st, uint8_t* dec, const DevBrRes* recs,
                        const DevPiece* pieces, int n) {
   if (n) hipLaunchKernelGGL(k_brres_wave, dim3(n), dim3(WAVE), 0, st, dec, recs, pieces, n);
 }
+void launch_contains_win(hipStream_t st, const uint8_t* dec,
+                         const DevCWin* wins, int n, const DevPage* pages,
+                         const uint16_t* starts_pool, const uint8_t* needle,
+                         int nlen, const uint32_t* rowof, uint8_t* mask) {
+  if (n) hipLaunchKernelGGL(k_contains_win, dim3(n), dim3(CTHREADS), 0, st, dec, wins, n, pages, starts_pool, needle, nlen, rowof, mask);
+}
 void launch_def_levels(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                        const int32_t* ids, int n, uint8_t* valid,
-                       uint32_t* rowof, uint32_t* present, int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_def_levels, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, valid, rowof, present, d_err);
+                       uint8_t* null_mask, uint32_t* rowof, uint32_t* present,
+                       int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_def_levels, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, valid, null_mask, rowof, present, d_err);
 }
 void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int32_t* remap_pool,

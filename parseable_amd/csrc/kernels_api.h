@@ -18,7 +18,12 @@ void launch_brres_wave(hipStream_t, uint8_t* dec, const DevBrRes*,
                        const DevPiece*, int n);
 void launch_def_levels(hipStream_t, const uint8_t* dec, const DevPage*,
                        const int32_t* ids, int n, uint8_t* valid,
-                       uint32_t* rowof, uint32_t* present, int32_t* d_err);
+                       uint8_t* null_mask, uint32_t* rowof, uint32_t* present,
+                       int32_t* d_err);
+void launch_contains_win(hipStream_t, const uint8_t* dec, const DevCWin* wins,
+                         int n, const DevPage* pages,
+                         const uint16_t* starts_pool, const uint8_t* needle,
+                         int nlen, const uint32_t* rowof, uint8_t* mask);
 void launch_dict_gid(hipStream_t, const uint8_t* dec, const DevPage*,
                      const int32_t* ids, int n, const int32_t* remap_pool,
                      int32_t* out, uint8_t* valid, const uint32_t* rowof,
