@@ -45,6 +45,9 @@ struct DevBrRes {      // literal-resolved record (absolute dec offsets)
 struct DevPiece { uint64_t src; uint32_t len; uint32_t _pad; };
 // litpar literal copy: raw[src..src+len) -> dec[dst..dst+len)  (absolute)
 struct DevLit { uint64_t src, dst; uint32_t len; uint32_t _pad; };
+// resolved match whose pattern (<= 8 bytes) the host inlined from the
+// compressed literal bytes: write-only, no scattered pattern reads.
+struct DevBrInl { uint64_t dst, pat; uint32_t len, period; };
 // contains window: a value-aligned run of consecutive non-null values of one
 // PLAIN byte-array page, <= CWIN bytes total (or a single oversized value,
 // nbytes > CWIN). Host builds these at load time by decompressing the page

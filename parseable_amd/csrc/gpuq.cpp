@@ -155,6 +155,7 @@ struct Partition {
   // literal-resolved records (single launch; meta.cpp lz4_walk)
   std::vector<DevBrRes> res_lane, res_wave;
   std::vector<DevPiece> piece_pool;
+  std::vector<DevBrInl> brinl;       // host-inlined <=8B patterns (litpar)
   std::vector<DevLit> lits_lane, lits_wave;   // litpar pages (meta.h)
   // contains windows (host length-chain walk at load time; dev_types.h)
   std::vector<DevCWin> cwins;
@@ -185,6 +186,7 @@ struct Partition {
   DevPageBr* d_pagebrs = nullptr;
   DevBrRes* d_res_lane = nullptr;
   DevBrRes* d_res_wave = nullptr;
+  DevBrInl* d_brinl = nullptr;
   DevPiece* d_piece_pool = nullptr;
   DevLit* d_lits_lane = nullptr;
   DevLit* d_lits_wave = nullptr;
@@ -735,12 +737,53 @@ extern "C" gpuq_plan* gpuq_plan_build(
                 part.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
                                     br.len, 0});
             } else if (!lp.resolved.empty()) {
-              uint32_t pbase = (uint32_t)part.piece_pool.size();
-              for (const auto& pc : lp.pieces)
-                part.piece_pool.push_back({dp.dst_off + pc.src, pc.len, 0});
+              // litpar pieces are literal-backed: the host can read any
+              // pattern of <= 8 bytes straight from the compressed stream
+              // and inline it — the resolver then never touches dec sources
+              auto lit_bytes = [&](uint32_t src, uint32_t len,
+                                   uint8_t* out) -> bool {
+                const auto& Ls = lp.lits;   // dst-ascending by construction
+                size_t lo = 0, hi = Ls.size();
+                while (lo < hi) {
+                  size_t mid = (lo + hi) / 2;
+                  if (Ls[mid].dst <= src) lo = mid + 1;
+                  else hi = mid;
+                }
+                if (lo == 0) return false;
+                const Lz4Lit& L = Ls[lo - 1];
+                if (src < L.dst || src + len > L.dst + L.len) return false;
+                std::memcpy(out, praw + L.src + (src - L.dst), len);
+                return true;
+              };
               for (const auto& rr : lp.resolved) {
-                DevBrRes rec{dp.dst_off + rr.dst, rr.len, rr.off,
-                             pbase + rr.piece_start, rr.piece_n};
+                uint32_t pat_len = 0;
+                for (uint32_t k = 0; k < rr.piece_n; k++)
+                  pat_len += lp.pieces[rr.piece_start + k].len;
+                if (lp.litpar && pat_len > 0 && pat_len <= 8 &&
+                    rr.len <= 256) {
+                  uint8_t buf[8] = {0};
+                  uint32_t o = 0;
+                  bool ok = true;
+                  for (uint32_t k = 0; k < rr.piece_n && ok; k++) {
+                    const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
+                    ok = lit_bytes(pc.src, pc.len, buf + o);
+                    o += pc.len;
+                  }
+                  if (ok) {
+                    uint64_t pat;
+                    std::memcpy(&pat, buf, 8);
+                    part.brinl.push_back(
+                        {dp.dst_off + rr.dst, pat, rr.len, pat_len});
+                    continue;
+                  }
+                }
+                uint32_t ps = (uint32_t)part.piece_pool.size();
+                for (uint32_t k = 0; k < rr.piece_n; k++) {
+                  const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
+                  part.piece_pool.push_back({dp.dst_off + pc.src, pc.len, 0});
+                }
+                DevBrRes rec{dp.dst_off + rr.dst, rr.len, rr.off, ps,
+                             rr.piece_n};
                 if (rr.len <= 256) part.res_lane.push_back(rec);
                 else part.res_wave.push_back(rec);
               }
@@ -1143,6 +1186,8 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
               (void**)&part.d_lits_lane);
   upload_pool(part.lits_wave.data(), part.lits_wave.size() * sizeof(DevLit),
               (void**)&part.d_lits_wave);
+  upload_pool(part.brinl.data(), part.brinl.size() * sizeof(DevBrInl),
+              (void**)&part.d_brinl);
   upload_pool(part.cwins.data(), part.cwins.size() * sizeof(DevCWin),
               (void**)&part.d_cwins);
   upload_pool(part.cstarts.data(), part.cstarts.size() * sizeof(uint16_t),
@@ -1452,6 +1497,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // deferred-match resolution: host-resolved records are independent —
   // one launch each (kernel boundary after phase 1 gives coherence);
   // piece-explosion pages fall back to the serial windowed wave
+  launch_brres_inl(st, part.d_dec, part.d_brinl, (int64_t)part.brinl.size());
   launch_brres_lane(st, part.d_dec, part.d_res_lane, part.d_piece_pool,
                     (int64_t)part.res_lane.size());
   launch_brres_wave(st, part.d_dec, part.d_res_wave, part.d_piece_pool,
@@ -1893,7 +1939,7 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_present); F(part.d_tmpvalid);
     F(part.d_lits_lane); F(part.d_lits_wave);
-    F(part.d_cwins); F(part.d_cstarts);
+    F(part.d_cwins); F(part.d_cstarts); F(part.d_brinl);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);
     F(part.d_res_lane); F(part.d_res_wave); F(part.d_piece_pool);
     F(part.d_keys); F(part.d_keys_sorted); F(part.d_rows);

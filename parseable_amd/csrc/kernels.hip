@@ -234,6 +234,17 @@ __global__ void k_brres_lane(uint8_t* __restrict__ dec,
     pat_off += pc.len;
   }
 }
+// inlined-pattern resolved matches: pure writes (records are dst-sorted,
+// adjacent lanes write adjacent regions)
+__global__ void k_brres_inl(uint8_t* __restrict__ dec,
+                            const DevBrInl* __restrict__ recs, int64_t n) {
+  int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= n) return;
+  const DevBrInl R = recs[r];
+  uint8_t* o = dec + R.dst;
+  for (uint32_t j = 0; j < R.len; j++)
+    o[j] = (uint8_t)(R.pat >> ((j % R.period) * 8));
+}
 __global__ void __launch_bounds__(WAVE)
 k_brres_wave(uint8_t* __restrict__ dec, const DevBrRes* __restrict__ recs,
              const DevPiece* __restrict__ pieces, int n) {
@@ -1672,6 +1683,10 @@ void launch_lz4_backrefs(hipStream_t st, uint8_t* dec, const DevBr* brs,
 void launch_brres_lane(hipStream_t st, uint8_t* dec, const DevBrRes* recs,
                        const DevPiece* pieces, int64_t n) {
   if (n) hipLaunchKernelGGL(k_brres_lane, dim3((int)((n + 255) / 256)), dim3(256), 0, st, dec, recs, pieces, n);
+}
+void launch_brres_inl(hipStream_t st, uint8_t* dec, const DevBrInl* recs,
+                      int64_t n) {
+  if (n) hipLaunchKernelGGL(k_brres_inl, dim3((int)((n + 255) / 256)), dim3(256), 0, st, dec, recs, n);
 }
 void launch_brres_wave(hipStream_t st, uint8_t* dec, const DevBrRes* recs,
                        const DevPiece* pieces, int n) {

@@ -14,6 +14,8 @@ void launch_lz4_backrefs(hipStream_t, uint8_t* dec, const DevBr*,
                          const DevPageBr*, int n);
 void launch_brres_lane(hipStream_t, uint8_t* dec, const DevBrRes*,
                        const DevPiece*, int64_t n);
+void launch_brres_inl(hipStream_t, uint8_t* dec, const DevBrInl* recs,
+                      int64_t n);
 void launch_brres_wave(hipStream_t, uint8_t* dec, const DevBrRes*,
                        const DevPiece*, int n);
 void launch_def_levels(hipStream_t, const uint8_t* dec, const DevPage*,

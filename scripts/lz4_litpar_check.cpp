@@ -55,6 +55,34 @@ int main(int argc, char** argv) {
           // phase 1: literal copies (any order)
           std::vector<uint8_t> got(pi.uncomp_size, 0xcd);
           for (auto& L : lp.lits) memcpy(&got[L.dst], praw + L.src, L.len);
+          // cross-check the host pattern-inline path (gpuq.cpp lit_bytes):
+          // every piece's bytes must be reachable from the COMPRESSED stream
+          // through the lits map and equal the decompressed literal bytes
+          long inl_mismatch = 0;
+          for (auto& r : lp.resolved) {
+            for (uint32_t k = 0; k < r.piece_n; k++) {
+              const Lz4Piece& pc = lp.pieces[r.piece_start + k];
+              size_t lo = 0, hi = lp.lits.size();
+              while (lo < hi) {
+                size_t mid = (lo + hi) / 2;
+                if (lp.lits[mid].dst <= pc.src) lo = mid + 1;
+                else hi = mid;
+              }
+              if (lo == 0) { inl_mismatch++; continue; }
+              const Lz4Lit& L = lp.lits[lo - 1];
+              if (pc.src < L.dst || pc.src + pc.len > L.dst + L.len) {
+                inl_mismatch++;
+                continue;
+              }
+              if (memcmp(praw + L.src + (pc.src - L.dst), &want[pc.src],
+                         pc.len) != 0)
+                inl_mismatch++;
+            }
+          }
+          if (inl_mismatch) {
+            printf("INLINE SOURCE MISMATCH x%ld\n", inl_mismatch);
+            bad++;
+          }
           // phase 2: resolved records (any order) — pattern from pieces,
           // applied periodically, exactly as k_brres does
           for (auto& r : lp.resolved) {
