@@ -178,6 +178,8 @@ struct Partition {
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;
   uint32_t* d_rowof = nullptr;     // dense->row map for null-bearing pages
+  uint32_t* d_rank = nullptr;      // row->dense rank (expansion pass)
+  uint8_t* d_scr = nullptr;        // dense decode scratch (8B/row)
   uint32_t* d_present = nullptr;   // per-page dense counts
   uint8_t* d_tmpvalid = nullptr;   // scratch validity for gid/mask-only cols
   int32_t* d_all_ids = nullptr;   // identity page-id list for the LZ4 sweep
@@ -1150,6 +1152,8 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   }
   HIP_TRY(hipMalloc(&part.d_mask, std::max<int64_t>(part.n_rows, 16)));
   HIP_TRY(hipMalloc(&part.d_rowof, std::max<int64_t>(part.n_rows * 4, 16)));
+  HIP_TRY(hipMalloc(&part.d_rank, std::max<int64_t>(part.n_rows * 4, 16)));
+  HIP_TRY(hipMalloc(&part.d_scr, std::max<int64_t>(part.n_rows * 8, 16)));
   HIP_TRY(hipMalloc(&part.d_present, std::max<size_t>(part.pages.size() * 4, 16)));
   HIP_TRY(hipMalloc(&part.d_tmpvalid, std::max<int64_t>(part.n_rows, 16)));
   if (plan->is_projection) {
@@ -1525,28 +1529,46 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         auto it = part.d_valid.find(col);
         uint8_t* v = it != part.d_valid.end() ? it->second : part.d_tmpvalid;
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n, v, nullptr,
-                          part.d_rowof, part.d_present, part.d_err);
+                          part.d_rowof, part.d_rank, part.d_present,
+                          part.d_err);
         launch_dict_gid(st, part.d_dec, part.d_pages, ids, n, part.d_remap,
                         part.d_gid[col],
                         it != part.d_valid.end() ? it->second : nullptr,
-                        part.d_rowof, part.d_present, v, part.d_err);
+                        part.d_present, 0, part.d_err);
+        launch_dict_gid(st, part.d_dec, part.d_pages, ids, n, part.d_remap,
+                        (int32_t*)part.d_scr, nullptr, part.d_present, 1,
+                        part.d_err);
+        launch_expand(st, part.d_dec, part.d_pages, ids, n, part.d_scr,
+                      part.d_rank, v, (uint8_t*)part.d_gid[col], 1);
         break;
       }
       case TK_DICT_VAL:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
                           part.d_valid[col], nullptr, part.d_rowof,
-                          part.d_present, part.d_err);
+                          part.d_rank, part.d_present, part.d_err);
         launch_dict_i64(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
-                        part.d_val[col], part.d_valid[col], part.d_rowof,
-                        part.d_present, part.d_err);
+                        part.d_val[col], part.d_valid[col], part.d_present, 0,
+                        part.d_err);
+        launch_dict_i64(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
+                        (int64_t*)part.d_scr, nullptr, part.d_present, 1,
+                        part.d_err);
+        launch_expand(st, part.d_dec, part.d_pages, ids, n, part.d_scr,
+                      part.d_rank, part.d_valid[col],
+                      (uint8_t*)part.d_val[col], 0);
         break;
       case TK_PLAIN_VAL:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
                           part.d_valid[col], nullptr, part.d_rowof,
-                          part.d_present, part.d_err);
+                          part.d_rank, part.d_present, part.d_err);
         launch_plain_fixed(st, part.d_dec, part.d_pages, ids, n,
-                           part.d_val[col], part.d_valid[col], part.d_rowof,
-                           part.d_present, part.d_err);
+                           part.d_val[col], part.d_valid[col], part.d_present,
+                           0, part.d_err);
+        launch_plain_fixed(st, part.d_dec, part.d_pages, ids, n,
+                           (int64_t*)part.d_scr, nullptr, part.d_present, 1,
+                           part.d_err);
+        launch_expand(st, part.d_dec, part.d_pages, ids, n, part.d_scr,
+                      part.d_rank, part.d_valid[col],
+                      (uint8_t*)part.d_val[col], 0);
         break;
       case TK_DELTA_VAL:
         launch_delta_i64(st, part.d_dec, part.d_pages, ids, n,
@@ -1555,10 +1577,13 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       case TK_DICT_MASK:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
                           part.d_tmpvalid, nullptr, part.d_rowof,
-                          part.d_present, part.d_err);
+                          part.d_rank, part.d_present, part.d_err);
         launch_dict_mask(st, part.d_dec, part.d_pages, ids, n, part.d_lut,
-                         part.d_mask, part.d_rowof, part.d_present,
-                         part.d_tmpvalid, part.d_err);
+                         part.d_mask, part.d_err);
+        launch_dict_lut_scr(st, part.d_dec, part.d_pages, ids, n, part.d_lut,
+                            part.d_scr, part.d_present, part.d_err);
+        launch_expand(st, part.d_dec, part.d_pages, ids, n, part.d_scr,
+                      part.d_rank, part.d_tmpvalid, part.d_mask, 2);
         break;
       case TK_BYTES_CONTAINS: {
         std::string needle;
@@ -1568,7 +1593,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         // map; the window kernel then has no serial work at all
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
                           part.d_tmpvalid, part.d_mask, part.d_rowof,
-                          part.d_present, part.d_err);
+                          part.d_rank, part.d_present, part.d_err);
         auto rng = part.cwin_ranges.at(col);
         launch_contains_win(st, part.d_dec, part.d_cwins + rng.first,
                             (int)rng.second, part.d_pages, part.d_cstarts,
@@ -1937,7 +1962,8 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
-    F(part.d_rowof); F(part.d_present); F(part.d_tmpvalid);
+    F(part.d_rowof); F(part.d_rank); F(part.d_scr);
+    F(part.d_present); F(part.d_tmpvalid);
     F(part.d_lits_lane); F(part.d_lits_wave);
     F(part.d_cwins); F(part.d_cstarts); F(part.d_brinl);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);

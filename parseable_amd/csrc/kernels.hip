@@ -465,7 +465,7 @@ __global__ void __launch_bounds__(WAVE)
 k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
              const int32_t* __restrict__ ids, int n,
              uint8_t* __restrict__ valid, uint8_t* __restrict__ null_mask,
-             uint32_t* __restrict__ rowof,
+             uint32_t* __restrict__ rowof, uint32_t* __restrict__ rankout,
              uint32_t* __restrict__ present, int32_t* d_error) {
   __shared__ uint64_t bits[DEF_MAX_ROWS / 64];
   __shared__ uint32_t tile_base[DEF_MAX_ROWS / 512];
@@ -589,6 +589,7 @@ k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
         uint8_t ok = (uint8_t)((b >> j) & 1);
         valid[row0 + r] = ok;
         if (null_mask && !ok) null_mask[row0 + r] = 0;  // NULL never matches
+        if (rankout) rankout[row0 + r] = rank;  // dense index (valid rows)
         if (ok) rowof[row0 + rank++] = row0 + r;
       }
     }
@@ -613,8 +614,8 @@ struct EmitDictI64 {
   const int64_t* dictv;
   int64_t* out;
   uint8_t* valid;
-  __device__ void operator()(uint32_t row, uint32_t idx) const { out[row] = dictv[idx]; valid[row] = 1; }
-  __device__ void null_at(uint32_t row) const { valid[row] = 0; }
+  __device__ void operator()(uint32_t row, uint32_t idx) const { out[row] = dictv[idx]; if (valid) valid[row] = 1; }
+  __device__ void null_at(uint32_t row) const { if (valid) valid[row] = 0; }
 };
 struct EmitDictMask {
   const uint8_t* lut;  // per-local-dict-id pred result
@@ -623,33 +624,30 @@ struct EmitDictMask {
   __device__ void null_at(uint32_t row) const { mask[row] = 0; }  // NULL never matches
 };
 
+// mode 0 (DIRECT): all-valid pages only — decode straight to the final
+// row-aligned arrays. mode 1 (SCRATCH): null-bearing pages only — decode
+// the dense stream to scratch at [row0+k]; a separate expansion kernel
+// then writes the final arrays COALESCED through the k_def_levels rank map
+// (a scatter here would cost a read-modify-write line fetch per value).
 template <class Emit>
 __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit emit,
-                                 const uint32_t* __restrict__ rowof,
                                  const uint32_t* __restrict__ present,
-                                 const uint8_t* __restrict__ valid_in,
-                                 int32_t page_id, int32_t* d_error) {
+                                 int32_t page_id, int mode, int32_t* d_error) {
   const int lane = threadIdx.x;
   const uint8_t* def_start; uint32_t def_len; bool all_valid;
   const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
   uint32_t nv = pg.num_values;
   const uint32_t row0 = pg.row_start;
-  const uint32_t* ind = nullptr;
-  if (!all_valid) {
-    // dense mode: k_def_levels (launched just before) produced the
-    // dense->row map and the present count; run the SAME parallel run
-    // decode over the dense stream and scatter through it, then sweep
-    // null rows once
-    ind = rowof;
+  if (mode == 0) {
+    if (!all_valid) return;
+  } else {
+    if (all_valid) return;
     nv = present[page_id];
-    for (uint32_t r = lane; r < pg.num_values; r += WAVE)
-      if (!valid_in[row0 + r]) emit.null_at(row0 + r);
-    __builtin_amdgcn_wave_barrier();
   }
 
   {
     int bw = *vals++;
-    auto target = [&](uint32_t k) { return ind ? ind[row0 + k] : row0 + k; };
+    auto target = [&](uint32_t k) { return row0 + k; };
     if (bw == 0) {  // all values are dict id 0
       for (uint32_t i = lane; i < nv; i += WAVE) emit(target(i), 0);
       return;
@@ -718,15 +716,13 @@ template <class Emit>
 __global__ void __launch_bounds__(WAVE)
 k_dict_pages(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
              const int32_t* __restrict__ ids, int n, Emit emit,
-             const uint32_t* rowof, const uint32_t* present,
-             const uint8_t* valid_in, int32_t* d_error) {
+             const uint32_t* present, int mode, int32_t* d_error) {
   int pi = blockIdx.x;
   if (pi >= n) return;
   DevPage pg = pages[ids[pi]];
   Emit e = emit;
   e.advance(pg);  // per-page aux pool offsets
-  dict_page_decode(pg, dec + pg.dst_off, e, rowof, present, valid_in,
-                   ids[pi], d_error);
+  dict_page_decode(pg, dec + pg.dst_off, e, present, ids[pi], mode, d_error);
 }
 
 // wrappers adding per-page aux advance
@@ -740,6 +736,14 @@ struct EmitDictI64P : EmitDictI64 {
 };
 struct EmitDictMaskP : EmitDictMask {
   const uint8_t* pool;
+  __device__ void advance(const DevPage& pg) { lut = pool + pg.aux_lut; }
+};
+struct EmitLutD {  // scratch mode: STORE the per-value LUT byte densely
+  const uint8_t* pool;
+  const uint8_t* lut;
+  uint8_t* scr;
+  __device__ void operator()(uint32_t row, uint32_t idx) const { scr[row] = lut[idx]; }
+  __device__ void null_at(uint32_t) const {}
   __device__ void advance(const DevPage& pg) { lut = pool + pg.aux_lut; }
 };
 
@@ -864,8 +868,8 @@ __global__ void __launch_bounds__(WAVE)
 k_plain_fixed(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
               const int32_t* __restrict__ ids, int n,
               int64_t* __restrict__ out, uint8_t* __restrict__ valid,
-              const uint32_t* __restrict__ rowof,
-              const uint32_t* __restrict__ present, int32_t* d_error) {
+              const uint32_t* __restrict__ present, int mode,
+              int32_t* d_error) {
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
@@ -873,7 +877,8 @@ k_plain_fixed(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages
   const uint8_t* def_start; uint32_t def_len; bool all_valid;
   const uint8_t* vals = def_levels(pg, dec + pg.dst_off, &def_start, &def_len, &all_valid);
   const uint32_t row0 = pg.row_start;
-  if (all_valid) {
+  if (mode == 0) {
+    if (!all_valid) return;
     uint32_t nv = pg.num_values;
     for (uint32_t i = lane; i < nv; i += WAVE) {
       int64_t v;
@@ -882,13 +887,48 @@ k_plain_fixed(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages
       if (valid) valid[row0 + i] = 1;
     }
   } else {
-    // dense scatter through the k_def_levels mapping (valid already set)
-    uint32_t nd = present[ids[pi]];
+    if (all_valid) return;
+    uint32_t nd = present[ids[pi]];   // dense to scratch; expansion follows
     for (uint32_t i = lane; i < nd; i += WAVE) {
       int64_t v;
       memcpy(&v, vals + (size_t)i * 8, 8);
-      out[rowof[row0 + i]] = v;
+      out[row0 + i] = v;
     }
+  }
+}
+
+// ------------------------------------------------------------------
+// expansion: null-bearing pages only — coalesced row-side pass writing the
+// final arrays from the dense scratch through the k_def_levels rank map.
+// mode 0: i64 value (valid written by k_def_levels); mode 1: gid i32
+// (NULL -> 0); mode 2: predicate mask AND (NULL never matches).
+// ------------------------------------------------------------------
+__global__ void __launch_bounds__(WAVE)
+k_expand(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+         const int32_t* __restrict__ ids, int n,
+         const uint8_t* __restrict__ scr, const uint32_t* __restrict__ rank,
+         const uint8_t* __restrict__ valid, uint8_t* __restrict__ out,
+         int mode) {
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const uint8_t* def_start; uint32_t def_len; bool all_valid;
+  def_levels(pg, dec + pg.dst_off, &def_start, &def_len, &all_valid);
+  if (all_valid) return;
+  const uint32_t row0 = pg.row_start, nv = pg.num_values;
+  if (mode == 0) {
+    const int64_t* s = (const int64_t*)scr;
+    int64_t* o = (int64_t*)out;
+    for (uint32_t r = threadIdx.x; r < nv; r += WAVE)
+      if (valid[row0 + r]) o[row0 + r] = s[row0 + rank[row0 + r]];
+  } else if (mode == 1) {
+    const int32_t* s = (const int32_t*)scr;
+    int32_t* o = (int32_t*)out;
+    for (uint32_t r = threadIdx.x; r < nv; r += WAVE)
+      o[row0 + r] = valid[row0 + r] ? s[row0 + rank[row0 + r]] : 0;
+  } else {
+    for (uint32_t r = threadIdx.x; r < nv; r += WAVE)
+      out[row0 + r] &= valid[row0 + r] ? scr[row0 + rank[row0 + r]] : 0;
   }
 }
 
@@ -1700,41 +1740,51 @@ void launch_contains_win(hipStream_t st, const uint8_t* dec,
 }
 void launch_def_levels(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                        const int32_t* ids, int n, uint8_t* valid,
-                       uint8_t* null_mask, uint32_t* rowof, uint32_t* present,
-                       int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_def_levels, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, valid, null_mask, rowof, present, d_err);
+                       uint8_t* null_mask, uint32_t* rowof, uint32_t* rank,
+                       uint32_t* present, int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_def_levels, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, valid, null_mask, rowof, rank, present, d_err);
 }
 void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int32_t* remap_pool,
-                     int32_t* out, uint8_t* valid, const uint32_t* rowof,
-                     const uint32_t* present, const uint8_t* valid_in,
-                     int32_t* d_err) {
+                     int32_t* out, uint8_t* valid, const uint32_t* present,
+                     int mode, int32_t* d_err) {
   if (!n) return;
   EmitGidP e{}; e.pool = remap_pool; e.out = out; e.valid = valid;
-  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, rowof, present, valid_in, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, present, mode, d_err);
 }
 void launch_dict_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int64_t* dictv_pool,
-                     int64_t* out, uint8_t* valid, const uint32_t* rowof,
-                     const uint32_t* present, int32_t* d_err) {
+                     int64_t* out, uint8_t* valid, const uint32_t* present,
+                     int mode, int32_t* d_err) {
   if (!n) return;
   EmitDictI64P e{}; e.pool = dictv_pool; e.out = out; e.valid = valid;
-  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, rowof, present, valid, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, present, mode, d_err);
 }
 void launch_dict_mask(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, const uint8_t* lut_pool,
-                      uint8_t* mask, const uint32_t* rowof,
-                      const uint32_t* present, const uint8_t* valid_in,
-                      int32_t* d_err) {
+                      uint8_t* mask, int32_t* d_err) {
   if (!n) return;
   EmitDictMaskP e{}; e.pool = lut_pool; e.mask = mask;
-  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, rowof, present, valid_in, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, (const uint32_t*)nullptr, 0, d_err);
+}
+void launch_dict_lut_scr(hipStream_t st, const uint8_t* dec,
+                         const DevPage* pages, const int32_t* ids, int n,
+                         const uint8_t* lut_pool, uint8_t* scr,
+                         const uint32_t* present, int32_t* d_err) {
+  if (!n) return;
+  EmitLutD e{}; e.pool = lut_pool; e.scr = scr;
+  hipLaunchKernelGGL(k_dict_pages<EmitLutD>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, present, 1, d_err);
 }
 void launch_plain_fixed(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                         const int32_t* ids, int n, int64_t* out, uint8_t* valid,
-                        const uint32_t* rowof, const uint32_t* present,
-                        int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_plain_fixed, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, rowof, present, d_err);
+                        const uint32_t* present, int mode, int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_plain_fixed, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, present, mode, d_err);
+}
+void launch_expand(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                   const int32_t* ids, int n, const uint8_t* scr,
+                   const uint32_t* rank, const uint8_t* valid, uint8_t* out,
+                   int mode) {
+  if (n) hipLaunchKernelGGL(k_expand, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, scr, rank, valid, out, mode);
 }
 void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,

@@ -20,30 +20,33 @@ void launch_brres_wave(hipStream_t, uint8_t* dec, const DevBrRes*,
                        const DevPiece*, int n);
 void launch_def_levels(hipStream_t, const uint8_t* dec, const DevPage*,
                        const int32_t* ids, int n, uint8_t* valid,
-                       uint8_t* null_mask, uint32_t* rowof, uint32_t* present,
-                       int32_t* d_err);
+                       uint8_t* null_mask, uint32_t* rowof, uint32_t* rank,
+                       uint32_t* present, int32_t* d_err);
 void launch_contains_win(hipStream_t, const uint8_t* dec, const DevCWin* wins,
                          int n, const DevPage* pages,
                          const uint16_t* starts_pool, const uint8_t* needle,
                          int nlen, const uint32_t* rowof, uint8_t* mask);
 void launch_dict_gid(hipStream_t, const uint8_t* dec, const DevPage*,
                      const int32_t* ids, int n, const int32_t* remap_pool,
-                     int32_t* out, uint8_t* valid, const uint32_t* rowof,
-                     const uint32_t* present, const uint8_t* valid_in,
-                     int32_t* d_err);
+                     int32_t* out, uint8_t* valid, const uint32_t* present,
+                     int mode, int32_t* d_err);
 void launch_dict_i64(hipStream_t, const uint8_t* dec, const DevPage*,
                      const int32_t* ids, int n, const int64_t* dictv_pool,
-                     int64_t* out, uint8_t* valid, const uint32_t* rowof,
-                     const uint32_t* present, int32_t* d_err);
+                     int64_t* out, uint8_t* valid, const uint32_t* present,
+                     int mode, int32_t* d_err);
 void launch_dict_mask(hipStream_t, const uint8_t* dec, const DevPage*,
                       const int32_t* ids, int n, const uint8_t* lut_pool,
-                      uint8_t* mask, const uint32_t* rowof,
-                      const uint32_t* present, const uint8_t* valid_in,
-                      int32_t* d_err);
+                      uint8_t* mask, int32_t* d_err);
+void launch_dict_lut_scr(hipStream_t, const uint8_t* dec, const DevPage*,
+                         const int32_t* ids, int n, const uint8_t* lut_pool,
+                         uint8_t* scr, const uint32_t* present, int32_t* d_err);
 void launch_plain_fixed(hipStream_t, const uint8_t* dec, const DevPage*,
                         const int32_t* ids, int n, int64_t* out, uint8_t* valid,
-                        const uint32_t* rowof, const uint32_t* present,
-                        int32_t* d_err);
+                        const uint32_t* present, int mode, int32_t* d_err);
+void launch_expand(hipStream_t, const uint8_t* dec, const DevPage*,
+                   const int32_t* ids, int n, const uint8_t* scr,
+                   const uint32_t* rank, const uint8_t* valid, uint8_t* out,
+                   int mode);
 void launch_delta_i64(hipStream_t, const uint8_t* dec, const DevPage*,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
                       int32_t* d_err);
