@@ -1529,8 +1529,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         auto it = part.d_valid.find(col);
         uint8_t* v = it != part.d_valid.end() ? it->second : part.d_tmpvalid;
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n, v, nullptr,
-                          part.d_rowof, part.d_rank, part.d_present,
-                          part.d_err);
+                          nullptr, part.d_rank, part.d_present, part.d_err);
         launch_dict_gid(st, part.d_dec, part.d_pages, ids, n, part.d_remap,
                         part.d_gid[col],
                         it != part.d_valid.end() ? it->second : nullptr,
@@ -1544,7 +1543,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       }
       case TK_DICT_VAL:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
-                          part.d_valid[col], nullptr, part.d_rowof,
+                          part.d_valid[col], nullptr, nullptr,
                           part.d_rank, part.d_present, part.d_err);
         launch_dict_i64(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
                         part.d_val[col], part.d_valid[col], part.d_present, 0,
@@ -1558,7 +1557,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         break;
       case TK_PLAIN_VAL:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
-                          part.d_valid[col], nullptr, part.d_rowof,
+                          part.d_valid[col], nullptr, nullptr,
                           part.d_rank, part.d_present, part.d_err);
         launch_plain_fixed(st, part.d_dec, part.d_pages, ids, n,
                            part.d_val[col], part.d_valid[col], part.d_present,
@@ -1576,7 +1575,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         break;
       case TK_DICT_MASK:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
-                          part.d_tmpvalid, nullptr, part.d_rowof,
+                          part.d_tmpvalid, nullptr, nullptr,
                           part.d_rank, part.d_present, part.d_err);
         launch_dict_mask(st, part.d_dec, part.d_pages, ids, n, part.d_lut,
                          part.d_mask, part.d_err);

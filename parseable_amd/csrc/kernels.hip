@@ -578,21 +578,25 @@ k_def_levels(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
   __builtin_amdgcn_wave_barrier();
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-  // emit valid bytes + rowof (each lane owns tiles round-robin)
-  for (uint32_t t = lane; t < ntiles; t += WAVE) {
-    uint32_t rank = tile_base[t];
-    for (uint32_t w = t * 8; w < (t + 1) * 8 && w < nwords; w++) {
-      uint64_t b = bits[w];
-      for (uint32_t j = 0; j < 64; j++) {
-        uint32_t r = w * 64 + j;
-        if (r >= nv) break;
-        uint8_t ok = (uint8_t)((b >> j) & 1);
-        valid[row0 + r] = ok;
-        if (null_mask && !ok) null_mask[row0 + r] = 0;  // NULL never matches
-        if (rankout) rankout[row0 + r] = rank;  // dense index (valid rows)
-        if (ok) rowof[row0 + rank++] = row0 + r;
-      }
-    }
+  // emit: one WORD per wave iteration, one ROW per lane — valid/rank stores
+  // are 64/256 consecutive bytes per instruction (the per-lane tile walk
+  // this replaces scattered byte stores 512 rows apart); rank within the
+  // word from a lane-masked popcount, word prefix from tile_base + at most
+  // 7 sibling-word popcounts.
+  const uint64_t lmask = (1ull << lane) - 1;
+  for (uint32_t w = 0; w < nwords; w++) {
+    uint64_t b = bits[w];
+    uint32_t base = tile_base[w >> 3];
+    for (uint32_t ww = w & ~7u; ww < w; ww++)
+      base += (uint32_t)__popcll(bits[ww]);
+    uint32_t r = w * 64 + lane;
+    if (r >= nv) break;
+    uint8_t ok = (uint8_t)((b >> lane) & 1);
+    valid[row0 + r] = ok;
+    if (null_mask && !ok) null_mask[row0 + r] = 0;  // NULL never matches
+    uint32_t rk = base + (uint32_t)__popcll(b & lmask);
+    if (rankout) rankout[row0 + r] = rk;  // dense index (valid rows)
+    if (rowof && ok) rowof[row0 + rk] = row0 + r;
   }
 }
 
