@@ -946,7 +946,8 @@ k_expand(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
 // ------------------------------------------------------------------
 #define MAX_MB 8192
 #define MAX_BLK 2048
-__global__ void __launch_bounds__(WAVE)
+#define DELTA_T 256
+__global__ void __launch_bounds__(DELTA_T)
 k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
             const int32_t* __restrict__ ids, int n,
             int64_t* __restrict__ out, uint8_t* __restrict__ valid,
@@ -1049,7 +1050,7 @@ k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
   __syncthreads();
 
   // Phase B: per-miniblock delta sums (parallel over miniblocks)
-  for (uint32_t m = lane; m < n_mb; m += WAVE) {
+  for (uint32_t m = lane; m < n_mb; m += DELTA_T) {
     const uint8_t* p = vals + mb_off[m];
     int bw = mb_bw[m];
     int64_t md = blk_md[m / (uint32_t)mpb];
@@ -1081,7 +1082,7 @@ k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
   if (total) {
     if (lane == 0) { out[pg.row_start] = first; if (valid) valid[pg.row_start] = 1; }
   }
-  for (uint32_t m = lane; m < n_mb; m += WAVE) {
+  for (uint32_t m = lane; m < n_mb; m += DELTA_T) {
     const uint8_t* p = vals + mb_off[m];
     int bw = mb_bw[m];
     int64_t md = blk_md[m / (uint32_t)mpb];
@@ -1793,7 +1794,7 @@ void launch_expand(hipStream_t st, const uint8_t* dec, const DevPage* pages,
 void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
                       int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, out, valid, d_err);
+  if (n) hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(DELTA_T), 0, st, dec, pages, ids, n, out, valid, d_err);
 }
 void launch_bytes_contains(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                            const int32_t* ids, int n, const uint8_t* needle, int nlen,
