@@ -351,6 +351,51 @@ def _finish_stream(root, stream, stream_dir, manifest_files, file_paths, rows,
     }
 
 
+def gen_staging(
+    staging_dir: str,
+    config: str = "c1",
+    rows: int = 30_000,
+    seed: int = 77,
+    n_arrows: int = 2,
+    n_parquet: int = 1,
+    start_minute: int = 10_000,
+):
+    """Staging-window data, as the reference's staging directory holds it
+    between ingest and upload (src/parseable/staging/): per-minute `.arrows`
+    Arrow-IPC stream files plus `.parquet` files already converted
+    (parseable/streams.rs:922-1001) but not yet uploaded/manifested.
+    Timestamps start at BASE_TS_MS + start_minute minutes; callers set the
+    provider's now_ms inside [min_ts, max_ts + window)."""
+    os.makedirs(staging_dir, exist_ok=True)
+    total = n_arrows + n_parquet
+    per = max(1, rows // max(total, 1))
+    out = {"arrows": [], "parquet": []}
+    for i in range(total):
+        rng = np.random.default_rng([seed, 9000 + i])
+        tbl = _minute_batch(config, rng, per, start_minute + i)
+        if i < n_parquet:
+            p = os.path.join(staging_dir, f"staged.{i:04d}.parquet")
+            ts_idx = tbl.schema.get_field_index("p_timestamp")
+            pq.write_table(
+                tbl, p, row_group_size=ROW_GROUP_SIZE, compression="lz4",
+                use_dictionary=[c for c in tbl.column_names if c != "p_timestamp"],
+                column_encoding={"p_timestamp": "DELTA_BINARY_PACKED"},
+                data_page_version="1.0", write_statistics=True,
+                sorting_columns=[pq.SortingColumn(ts_idx, descending=True,
+                                                  nulls_first=False)],
+            )
+            out["parquet"].append(p)
+        else:
+            p = os.path.join(staging_dir, f"{start_minute + i}.data.arrows")
+            with pa.ipc.new_stream(p, tbl.schema) as w:
+                for b in tbl.to_batches(max_chunksize=8192):
+                    w.write_batch(b)
+            out["arrows"].append(p)
+    out["min_ts"] = BASE_TS_MS + start_minute * MINUTE_MS
+    out["max_ts"] = BASE_TS_MS + (start_minute + total) * MINUTE_MS
+    return out
+
+
 if __name__ == "__main__":
     import argparse
 

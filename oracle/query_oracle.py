@@ -96,7 +96,17 @@ def _factorize(col: pa.Array):
     return codes, enc.dictionary.to_pylist()
 
 
-def execute_projection(files: list[str], query: dict) -> dict:
+def _sources(files, extra_tables, need):
+    """Yield pyarrow Tables: parquet files (column-pruned reads) + in-memory
+    tables (the staging .arrows slice of SURVEY §8f-2)."""
+    for path in files:
+        yield pq.read_table(path, columns=need or None)
+    for t in extra_tables or []:
+        yield t.select(need) if need else t
+
+
+def execute_projection(files: list[str], query: dict,
+                       extra_tables=None) -> dict:
     """Projection scan: SELECT cols WHERE ... ORDER BY p_timestamp DESC
     LIMIT k (the console default; ordering contract
     stream_schema_provider.rs:181-204). Ties at the LIMIT boundary are
@@ -106,8 +116,7 @@ def execute_projection(files: list[str], query: dict) -> dict:
                   | {p["col"] for p in query.get("preds", [])}
                   | {"p_timestamp"})
     out = []
-    for path in files:
-        tbl = pq.read_table(path, columns=need)
+    for tbl in _sources(files, extra_tables, need):
         n = tbl.num_rows
         cols = {name: _norm_col(tbl.column(name)) for name in need}
         mask = np.ones(n, dtype=bool)
@@ -134,18 +143,18 @@ def execute_projection(files: list[str], query: dict) -> dict:
             "all_matching": out}
 
 
-def execute(files: list[str], query: dict) -> dict:
+def execute(files: list[str], query: dict, extra_tables=None) -> dict:
     """Run the query. Returns {"columns": [...], "rows": [[key...,agg...]...]}
-    rows sorted by key tuple, NULLs last."""
+    rows sorted by key tuple, NULLs last. extra_tables: in-memory pyarrow
+    Tables included as additional sources (staging .arrows)."""
     if query.get("select_cols"):
-        return execute_projection(files, query)
+        return execute_projection(files, query, extra_tables)
     need = sorted(_needed_columns(query))
     group_by = query.get("group_by", [])
     aggs = query["select"]
     acc: dict[tuple, list] = {}
 
-    for path in files:
-        tbl = pq.read_table(path, columns=need or None)
+    for tbl in _sources(files, extra_tables, need):
         n = tbl.num_rows
         cols = {name: _norm_col(tbl.column(name)) for name in need}
         mask = np.ones(n, dtype=bool)

@@ -158,14 +158,234 @@ class ManifestCountResult:
         return [[self.count]]
 
 
+class StagingScanner:
+    """The staging-window leg: `.arrows` Arrow-IPC files the
+    ingest path has staged but not yet converted (read newest-first, the
+    reverse reader of parseable/staging/reader.rs:316+). CPU evaluation of
+    this in-memory slice is acceptable per §8f-2 and mirrors the reference's
+    MemTable leg (stream_schema_provider.rs:292-350); it is NOT a fallback
+    for the parquet hot path — staging .parquet files join the GPU plan."""
+
+    def __init__(self, staging_dir: str):
+        self.staging_dir = staging_dir
+
+    def parquet_paths(self):
+        import glob as _glob
+
+        return sorted(_glob.glob(os.path.join(self.staging_dir, "*.parquet")))
+
+    def _tables(self):
+        import glob as _glob
+        import pyarrow as pa
+
+        for p in sorted(_glob.glob(os.path.join(self.staging_dir, "*.arrows")),
+                        reverse=True):
+            with pa.ipc.open_stream(p) as r:
+                yield r.read_all()
+
+    @staticmethod
+    def _mask(tbl, query):
+        import pyarrow as pa
+        import pyarrow.compute as pc
+
+        mask = pa.array([True] * tbl.num_rows)
+        tr = query.get("time_range")
+        preds = list(query.get("preds", []))
+        if tr is not None:
+            preds.append({"col": "p_timestamp", "op": "between",
+                          "lo": tr[0], "hi": tr[1], "hi_exclusive": True})
+        for p in preds:
+            col = tbl.column(p["col"])
+            if pa.types.is_dictionary(col.type):
+                col = col.cast(col.type.value_type)
+            if pa.types.is_timestamp(col.type):
+                col = col.cast(pa.int64())
+            op = p["op"]
+            if op == "between":
+                m = pc.and_(pc.greater_equal(col, p["lo"]),
+                            pc.less(col, p["hi"]) if p.get("hi_exclusive")
+                            else pc.less_equal(col, p["hi"]))
+            elif op == "contains":
+                m = pc.match_substring(col, p["lit"])
+            else:
+                f = {"eq": pc.equal, "ne": pc.not_equal, "lt": pc.less,
+                     "le": pc.less_equal, "gt": pc.greater,
+                     "ge": pc.greater_equal}[op]
+                m = f(col, p["lit"])
+            mask = pc.and_(mask, pc.fill_null(m, False))
+        return mask
+
+    def partial_batch(self, query):
+        """-> pyarrow.RecordBatch in the partial-aggregate schema
+        [key..., __presence, agg_i, agg_i_count, ...] or None."""
+        import pyarrow as pa
+        import pyarrow.compute as pc
+
+        group_by = query.get("group_by", [])
+        aggs = query["select"]
+        acc = {}
+        for tbl in self._tables():
+            sel = tbl.filter(self._mask(tbl, query))
+            if sel.num_rows == 0:
+                continue
+            keys = []
+            for g in group_by:
+                if isinstance(g, dict):  # DATE_BIN pseudo-key
+                    ts = sel.column(g["bin"]).cast(pa.int64())
+                    stride, origin = g["stride_ms"], g.get("origin", 0)
+                    b = pc.add(pc.multiply(pc.floor(pc.divide(
+                        pc.subtract(ts, origin), stride)).cast(pa.int64()),
+                        stride), origin)
+                    keys.append(b.to_pylist())
+                else:
+                    c = sel.column(g)
+                    if pa.types.is_dictionary(c.type):
+                        c = c.cast(c.type.value_type)
+                    keys.append(c.to_pylist())
+            vals = []
+            for a in aggs:
+                if a["agg"] == "count_star":
+                    vals.append(None)
+                else:
+                    c = sel.column(a["col"])
+                    if pa.types.is_dictionary(c.type):
+                        c = c.cast(c.type.value_type)
+                    if pa.types.is_timestamp(c.type):
+                        c = c.cast(pa.int64())
+                    vals.append(c.to_pylist())
+            for r in range(sel.num_rows):
+                key = tuple(k[r] for k in keys)
+                st = acc.get(key)
+                if st is None:
+                    st = [0, [None] * len(aggs), [0] * len(aggs)]
+                    acc[key] = st
+                st[0] += 1
+                for i, a in enumerate(aggs):
+                    if a["agg"] == "count_star":
+                        st[2][i] += 1
+                        continue
+                    v = vals[i][r]
+                    if v is None:
+                        continue
+                    st[2][i] += 1
+                    if a["agg"] == "count":
+                        continue
+                    cur = st[1][i]
+                    if cur is None:
+                        st[1][i] = v
+                    elif a["agg"] == "sum":
+                        st[1][i] = cur + v
+                    elif a["agg"] == "min":
+                        st[1][i] = min(cur, v)
+                    elif a["agg"] == "max":
+                        st[1][i] = max(cur, v)
+        if not acc:
+            return None
+        nk = len(group_by)
+        ks = sorted(acc.keys(), key=lambda k: tuple(
+            ((1, "") if v is None else (0, v)) for v in k))
+        cols, names = [], []
+        for i in range(nk):
+            names.append(f"k{i}")
+            cols.append(pa.array([k[i] for k in ks]))
+        names.append("__presence")
+        cols.append(pa.array([acc[k][0] for k in ks], type=pa.int64()))
+        for i, a in enumerate(aggs):
+            names.append(f"a{i}")
+            if a["agg"] in ("count_star", "count"):
+                cols.append(pa.array([acc[k][2][i] for k in ks], type=pa.int64()))
+            else:
+                cols.append(pa.array([acc[k][1][i] for k in ks]))
+            names.append(f"a{i}_count")
+            cols.append(pa.array([acc[k][2][i] for k in ks], type=pa.int64()))
+        return pa.record_batch(cols, names=names)
+
+    def matching_rows(self, query):
+        """Projection leg: matching staging rows [select_cols...]."""
+        import pyarrow as pa
+
+        out = []
+        cols_wanted = query["select_cols"]
+        for tbl in self._tables():
+            sel = tbl.filter(self._mask(tbl, query))
+            if sel.num_rows == 0:
+                continue
+            proj = []
+            for cname in cols_wanted:
+                c = sel.column(cname)
+                if pa.types.is_dictionary(c.type):
+                    c = c.cast(c.type.value_type)
+                if pa.types.is_timestamp(c.type):
+                    c = c.cast(pa.int64())
+                proj.append(c.to_pylist())
+            out.extend([list(r) for r in zip(*proj)])
+        return out
+
+
+class StagedPlan:
+    """UnionExec analog over the GPU plan (manifest + staging parquet) and
+    the CPU staging-arrows leg (stream_schema_provider.rs:384-401,637-647)."""
+
+    def __init__(self, gpu_plan, scanner: StagingScanner, query: dict):
+        self.gpu_plan = gpu_plan   # None when every parquet file was pruned
+        self.scanner = scanner
+        self.query = query
+
+    def partition_count(self):
+        return self.gpu_plan.partition_count() if self.gpu_plan else 0
+
+    def load(self, partition=None):
+        if self.gpu_plan:
+            self.gpu_plan.load(partition)
+
+    def execute(self, partition):
+        return self.gpu_plan.execute(partition)
+
+    def execute_all(self):
+        batches = []
+        if self.gpu_plan:
+            batches = [self.gpu_plan.execute(p)
+                       for p in range(self.gpu_plan.partition_count())]
+            batches = [b for b in batches if b is not None]
+        if self.query.get("select_cols"):
+            return merge_topk(batches, self.query,
+                              extra_rows=self.scanner.matching_rows(self.query))
+        sb = self.scanner.partial_batch(self.query)
+        if sb is not None:
+            batches.append(sb)
+        return merge_partials(batches, self.query)
+
+    def metrics(self):
+        return self.gpu_plan.metrics() if self.gpu_plan else {}
+
+    def close(self):
+        if self.gpu_plan:
+            self.gpu_plan.close()
+
+
 class StandardTableProvider:
-    def __init__(self, stream_dir: str, session: GpuSession | None = None):
+    def __init__(self, stream_dir: str, session: GpuSession | None = None,
+                 staging_dir: str | None = None, now_ms: int | None = None,
+                 staging_window_ms: int = 5 * 60_000):
         self.stream_dir = stream_dir
         self.session = session
+        self.staging_dir = staging_dir
+        self.now_ms = now_ms
+        self.staging_window_ms = staging_window_ms
         snap_path = os.path.join(stream_dir, "stream.json")
         with open(snap_path) as fh:
             self.stream_json = json.load(fh)
         self.snapshot = self.stream_json["snapshot"]
+
+    # is_within_staging_window (stream_schema_provider.rs:936-958): the query
+    # range touches staging when its end reaches past now - window (or there
+    # is no range / no clock).
+    def _staging_touches(self, time_range):
+        if self.staging_dir is None:
+            return False
+        if time_range is None or self.now_ms is None:
+            return True
+        return time_range[1] > self.now_ms - self.staging_window_ms
 
     # Snapshot::manifests (catalog/snapshot.rs:42-71): retain manifests whose
     # [lower,upper] overlaps the time predicates.
@@ -194,8 +414,13 @@ class StandardTableProvider:
 
     def scan(self, query: dict):
         """query: the IR of tests/golden_queries.py (select/group_by/preds/
-        time_range). Returns GpuExecutionPlan or ManifestCountResult."""
+        time_range). Returns GpuExecutionPlan, StagedPlan (staging window),
+        ManifestCountResult or EmptyScanResult."""
         query = {k: v for k, v in query.items() if k != "ext"}
+
+        staging_hit = self._staging_touches(query.get("time_range"))
+        if staging_hit:
+            return self._scan_with_staging(query)
 
         if self.session is not None and not os.environ.get("GPUQ_PY_PLANNER"):
             # native catalog planner inside libgpuq (catalog.cpp)
@@ -259,6 +484,30 @@ class StandardTableProvider:
             p = fe["file_path"]
             paths.append(p if os.path.isabs(p) else os.path.join(root, p))
         return GpuExecutionPlan(self.session, paths, query)
+
+    def _scan_with_staging(self, query):
+        """Staging branch (stream_schema_provider.rs:637-647): manifest
+        parquet + staging parquet execute on the GPU; staging .arrows run on
+        the CPU leg; the manifest count fast path is SKIPPED because staging
+        rows are not in any manifest yet."""
+        scanner = StagingScanner(self.staging_dir)
+        preds = list(query.get("preds", []))
+        time_range = query.get("time_range")
+        files = self._manifest_files(time_range)
+        prune_preds = preds.copy()
+        if time_range:
+            prune_preds.append({"col": "p_timestamp", "op": "between",
+                                "lo": time_range[0], "hi": time_range[1],
+                                "hi_exclusive": True})
+        kept = [f for f in files if not _file_pruned(f, prune_preds)]
+        root = os.path.dirname(self.stream_dir)
+        paths = []
+        for fe in kept:
+            p = fe["file_path"]
+            paths.append(p if os.path.isabs(p) else os.path.join(root, p))
+        paths.extend(scanner.parquet_paths())  # scanned unconditionally
+        gpu_plan = GpuExecutionPlan(self.session, paths, query) if paths else None
+        return StagedPlan(gpu_plan, scanner, query)
 
 
 def _build_c_projection(query: dict, keep):
@@ -501,12 +750,13 @@ def merge_partials(batches, query):
     return rows
 
 
-def merge_topk(batches, query):
+def merge_topk(batches, query, extra_rows=None):
     """Final merge of per-partition top-k row batches: re-sort by
-    p_timestamp DESC across partitions, truncate to LIMIT."""
+    p_timestamp DESC across partitions, truncate to LIMIT. extra_rows:
+    already-projected rows from the staging CPU leg."""
     cols = query["select_cols"]
     ts_i = cols.index("p_timestamp")
-    rows = []
+    rows = list(extra_rows) if extra_rows else []
     for b in batches:
         if b is None or b.num_rows == 0:
             continue

@@ -1,0 +1,170 @@
+"""Staging-window branch (stream_schema_provider.rs:292-350,
+:637-647, :936-958): queries whose time range touches the last ~5 minutes
+must include data still sitting in the staging directory — `.arrows`
+Arrow-IPC files (evaluated on the CPU leg, mirroring the reference's
+MemTable) and already-converted staging `.parquet` (joins the GPU plan).
+
+CPU tests use staging that lies OUTSIDE the manifested stream's time range,
+so every manifest file is pruned and the plan is the CPU leg alone — no GPU
+required. The GPU test (marked) runs the combined union."""
+
+import os
+import sys
+
+import pyarrow as pa
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from datagen.gen import BASE_TS_MS, MINUTE_MS, gen_staging, gen_stream  # noqa: E402
+from oracle import query_oracle as qo  # noqa: E402
+from oracle.compare import assert_rows_equal  # noqa: E402
+from parseable_amd.provider import (  # noqa: E402
+    ManifestCountResult,
+    StagedPlan,
+    StandardTableProvider,
+)
+
+STAGE_MIN = 10_000  # staging minutes start here; stream occupies [0, ~2)
+NOW_MS = BASE_TS_MS + (STAGE_MIN + 3) * MINUTE_MS
+STAGING_RANGE = (BASE_TS_MS + STAGE_MIN * MINUTE_MS, NOW_MS)
+
+
+@pytest.fixture(scope="module")
+def staged_stream(tmp_path_factory):
+    root = tmp_path_factory.mktemp("staging")
+    info = gen_stream(str(root), "s", "c1", rows=40_000, seed=11)
+    sdir = str(root / "s_staging")
+    st = gen_staging(sdir, "c1", rows=24_000, seed=12, n_arrows=2,
+                     n_parquet=1, start_minute=STAGE_MIN)
+    return {"stream_dir": info["stream_dir"], "files": info["files"],
+            "staging_dir": sdir, "st": st}
+
+
+def _staging_tables(st, arrows_only=False):
+    tables = []
+    for p in st["arrows"]:
+        with pa.ipc.open_stream(p) as r:
+            tables.append(r.read_all())
+    files = [] if arrows_only else list(st["parquet"])
+    return files, tables
+
+
+def _provider(s, session=None):
+    return StandardTableProvider(s["stream_dir"], session,
+                                 staging_dir=s["staging_dir"], now_ms=NOW_MS)
+
+
+CPU_QUERIES = [
+    {"select": [{"agg": "count_star"}], "group_by": ["level"],
+     "time_range": STAGING_RANGE},
+    {"select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"},
+                {"agg": "sum", "col": "f_i64"}],
+     "group_by": ["host"], "preds": [{"col": "level", "op": "eq", "lit": "INFO"}],
+     "time_range": STAGING_RANGE},
+    {"select": [{"agg": "min", "col": "f_f64"}, {"agg": "count", "col": "level"}],
+     "time_range": STAGING_RANGE},
+    {"select": [{"agg": "count_star"}],
+     "group_by": [{"bin": "p_timestamp", "stride_ms": 60_000}],
+     "time_range": STAGING_RANGE},
+]
+
+
+@pytest.mark.parametrize("qi", range(len(CPU_QUERIES)))
+def test_staging_only_cpu(staged_stream, qi):
+    """Range entirely inside the staging window: every manifest file prunes
+    away, staging parquet is excluded (arrows_only fixture side), and the
+    CPU leg alone must match the oracle over the same .arrows tables."""
+    s = dict(staged_stream)
+    # staging dir with only the .arrows files (parquet leg needs a GPU)
+    import shutil
+
+    sdir2 = s["staging_dir"] + f"_arrows{qi}"
+    if not os.path.exists(sdir2):
+        os.makedirs(sdir2)
+        for p in s["st"]["arrows"]:
+            shutil.copy(p, sdir2)
+    prov = StandardTableProvider(s["stream_dir"], None, staging_dir=sdir2,
+                                 now_ms=NOW_MS)
+    q = CPU_QUERIES[qi]
+    plan = prov.scan(dict(q))
+    assert isinstance(plan, StagedPlan) and plan.gpu_plan is None
+    got = plan.execute_all()
+    _, tables = _staging_tables(s["st"], arrows_only=True)
+    want = qo.execute([], dict(q), extra_tables=tables)["rows"]
+    assert_rows_equal(got, want, q)
+
+
+def test_staging_projection_cpu(staged_stream):
+    s = staged_stream
+    import shutil
+
+    sdir2 = s["staging_dir"] + "_arrowsP"
+    if not os.path.exists(sdir2):
+        os.makedirs(sdir2)
+        for p in s["st"]["arrows"]:
+            shutil.copy(p, sdir2)
+    prov = StandardTableProvider(s["stream_dir"], None, staging_dir=sdir2,
+                                 now_ms=NOW_MS)
+    q = {"select_cols": ["p_timestamp", "level", "latency"], "limit": 25,
+         "order_by": {"col": "p_timestamp", "desc": True},
+         "time_range": STAGING_RANGE}
+    plan = prov.scan(dict(q))
+    got = plan.execute_all()
+    _, tables = _staging_tables(s["st"], arrows_only=True)
+    want = qo.execute([], dict(q), extra_tables=tables)
+    assert [r[0] for r in got] == [r[0] for r in want["rows"]]
+
+
+def test_fast_count_skipped_when_staging_touched(staged_stream):
+    """Bare count(*) normally answers from manifest sums; staging rows are
+    in no manifest, so the fast path must NOT fire for ranges touching the
+    window (and must still fire for ranges that do not)."""
+    s = staged_stream
+    sdir2 = s["staging_dir"] + "_arrows0"   # arrows-only copy from the fixture
+    prov = StandardTableProvider(s["stream_dir"], None, staging_dir=sdir2,
+                                 now_ms=NOW_MS)
+    q = {"select": [{"agg": "count_star"}], "time_range": STAGING_RANGE}
+    plan = prov.scan(dict(q))
+    assert isinstance(plan, StagedPlan)
+    # outside the window: fast path intact
+    old = {"select": [{"agg": "count_star"}],
+           "time_range": (BASE_TS_MS, BASE_TS_MS + 2 * MINUTE_MS)}
+    plan2 = prov.scan(dict(old))
+    assert isinstance(plan2, ManifestCountResult)
+    assert plan2.rows() == [[40_000]]
+
+
+def test_no_staging_dir_unchanged(staged_stream):
+    s = staged_stream
+    prov = StandardTableProvider(s["stream_dir"], None)
+    q = {"select": [{"agg": "count_star"}]}
+    assert isinstance(prov.scan(dict(q)), ManifestCountResult)
+
+
+@pytest.mark.gpu
+def test_staging_union_gpu(staged_stream):
+    """Combined plan: manifested parquet + staging parquet on the GPU,
+    .arrows on the CPU leg — whole-range result must match the oracle over
+    all three sources."""
+    from parseable_amd import GpuSession
+
+    s = staged_stream
+    sess = GpuSession(device_mask=1)
+    prov = _provider(s, sess)
+    for q in [
+        {"select": [{"agg": "count_star"}], "group_by": ["level"]},
+        {"select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+         "group_by": ["host"],
+         "preds": [{"col": "level", "op": "eq", "lit": "ERROR"}]},
+        {"select": [{"agg": "sum", "col": "f_i64"},
+                    {"agg": "min", "col": "latency"}]},
+    ]:
+        plan = prov.scan(dict(q))
+        assert isinstance(plan, StagedPlan) and plan.gpu_plan is not None
+        plan.load()
+        got = plan.execute_all()
+        files, tables = _staging_tables(s["st"])
+        want = qo.execute(s["files"] + files, dict(q), extra_tables=tables)["rows"]
+        assert_rows_equal(got, want, q)
+        plan.close()
