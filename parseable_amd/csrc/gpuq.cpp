@@ -1888,7 +1888,10 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     int64_t* vv = (int64_t*)eb->grab(nr * 8);
     for (int64_t r = 0; r < nr; r++) {
       size_t base = (size_t)live[r] * slots;
-      uint64_t cnt = empty_aggregate_row ? 0 : table[base + 2 + 2 * i];
+      // count(*) == presence: k_agg no longer spends atomics on its slots
+      uint64_t cnt = empty_aggregate_row ? 0
+                     : (kind == AGGK_COUNT_STAR ? table[base]
+                                                : table[base + 2 + 2 * i]);
       uint64_t val = empty_aggregate_row ? 0 : table[base + 1 + 2 * i];
       if (kind == AGGK_COUNT_STAR || kind == AGGK_COUNT) {
         vv[r] = (int64_t)cnt;
@@ -1910,8 +1913,12 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     chc->n_buffers = 2;
     chc->buffers = (const void**)calloc(2, sizeof(void*));
     int64_t* cv = (int64_t*)eb->grab(nr * 8);
-    for (int64_t r = 0; r < nr; r++)
-      cv[r] = empty_aggregate_row ? 0 : (int64_t)table[(size_t)live[r] * slots + 2 + 2 * i];
+    for (int64_t r = 0; r < nr; r++) {
+      size_t base = (size_t)live[r] * slots;
+      cv[r] = empty_aggregate_row ? 0
+              : (int64_t)(kind == AGGK_COUNT_STAR ? table[base]
+                                                  : table[base + 2 + 2 * i]);
+    }
     chc->buffers[1] = cv;
   }
 

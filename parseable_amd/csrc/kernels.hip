@@ -1624,10 +1624,7 @@ k_agg(AggArgs a) {
     atomicAdd((unsigned long long*)&row[0], 1ull);
     for (int ai = 0; ai < a.n_aggs; ai++) {
       int k = a.agg_kind[ai];
-      if (k == AGGK_COUNT_STAR) {
-        atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
-        continue;
-      }
+      if (k == AGGK_COUNT_STAR) continue;  // == presence (export reads slot 0)
       if (a.agg_valid[ai] && !a.agg_valid[ai][i]) continue;
       if (k == AGGK_COUNT) {  // validity only; no value array needed
         atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
