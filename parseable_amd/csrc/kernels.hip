@@ -1298,6 +1298,8 @@ k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pa
 // row-parallel comparison on i64 arrays
 // ------------------------------------------------------------------
 // ------------------------------------------------------------------
+#define CQMAX 4096
+__device__ inline uint32_t pat_full(uint8_t c0) { return 0x01010101u * c0; }
 // window-parallel CONTAINS over PLAIN byte-array pages: one block per
 // DevCWin. The host decompressed each page ONCE at load time, walked the
 // [u32 len][bytes] chain, and emitted value-aligned <=16KB windows plus a
@@ -1315,6 +1317,8 @@ k_contains_win(const uint8_t* __restrict__ dec,
                const uint32_t* __restrict__ rowof, uint8_t* __restrict__ mask) {
   __shared__ uint8_t win[CWIN];
   __shared__ uint32_t bm[CWIN / 32];
+  __shared__ uint16_t q[CQMAX];   // candidate queue (register presweep)
+  __shared__ uint32_t qn;
   __shared__ int shit;
   if (blockIdx.x >= (unsigned)n) return;
   const DevCWin W = wins[blockIdx.x];
@@ -1347,7 +1351,12 @@ k_contains_win(const uint8_t* __restrict__ dec,
     }
     return;
   }
-  {  // stage the full window unconditionally (arena is padded by CWIN)
+  if (threadIdx.x == 0) qn = 0;
+  __syncthreads();
+  {  // stage the full window unconditionally (arena is padded by CWIN) and
+     // presweep needle candidates while the words are STILL IN REGISTERS —
+     // the verify pass then touches only queued positions instead of
+     // re-reading the whole window from LDS
     uint32_t v[CWIN / (CTHREADS * 4)];
 #pragma unroll
     for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
@@ -1355,23 +1364,49 @@ k_contains_win(const uint8_t* __restrict__ dec,
 #pragma unroll
     for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
       *(uint32_t*)&win[threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u)] = v[k];
+    if (nlen) {
+      const uint32_t pat = 0x01010101u * needle[0];
+#pragma unroll
+      for (int k = 0; k < CWIN / (CTHREADS * 4); k++) {
+        uint32_t p = threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u);
+        if (p >= W.nbytes) continue;
+        uint32_t x = v[k] ^ pat;
+        uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
+        while (cand) {
+          int b = (__builtin_ctz(cand)) >> 3;
+          cand &= cand - 1;
+          uint32_t qi = atomicAdd(&qn, 1u);
+          if (qi < CQMAX) q[qi] = (uint16_t)(p + b);
+        }
+      }
+    }
   }
   for (uint32_t i = threadIdx.x; i < CWIN / 32; i += CTHREADS) bm[i] = 0;
   __syncthreads();
   if (nlen) {
-    const uint32_t pat = 0x01010101u * needle[0];
-    for (uint32_t p = threadIdx.x * 4u; p < W.nbytes; p += CTHREADS * 4u) {
-      uint32_t w = *(const uint32_t*)&win[p];
-      uint32_t x = w ^ pat;
-      uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
-      while (cand) {
-        int b = (__builtin_ctz(cand)) >> 3;
-        cand &= cand - 1;
-        uint32_t pos = p + b;
+    if (qn <= CQMAX) {
+      for (uint32_t i = threadIdx.x; i < qn; i += CTHREADS) {
+        uint32_t pos = q[i];
         if (pos + nlen <= CWIN) {
           int k = 0;  // verify from 0: the borrow trick has false positives
           while (k < nlen && win[pos + k] == needle[k]) k++;
           if (k == nlen) atomicOr(&bm[pos >> 5], 1u << (pos & 31));
+        }
+      }
+    } else {  // queue overflow (pathological needle[0] density): full sweep
+      for (uint32_t p = threadIdx.x * 4u; p < W.nbytes; p += CTHREADS * 4u) {
+        uint32_t w = *(const uint32_t*)&win[p];
+        uint32_t x = w ^ pat_full(needle[0]);
+        uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
+        while (cand) {
+          int b = (__builtin_ctz(cand)) >> 3;
+          cand &= cand - 1;
+          uint32_t pos = p + b;
+          if (pos + nlen <= CWIN) {
+            int k = 0;
+            while (k < nlen && win[pos + k] == needle[k]) k++;
+            if (k == nlen) atomicOr(&bm[pos >> 5], 1u << (pos & 31));
+          }
         }
       }
     }
