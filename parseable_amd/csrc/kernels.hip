@@ -1106,198 +1106,8 @@ k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
 }
 
 // ------------------------------------------------------------------
-// PLAIN byte_array CONTAINS (LIKE '%needle%'): fused offsets walk +
-// substring scan, one wave per page. lane0 walks the length-prefixed
-// values into an LDS batch of offsets; all lanes then scan one value each.
-// ------------------------------------------------------------------
-// LIKE '%needle%' over PLAIN byte_array pages — bitmap-sweep design,
-// validated by scripts/micro_contains.hip (3.5x the per-value scan):
-//  * 256-thread blocks (4 waves) per page; the value stream is staged
-//    through a 16 KiB LDS window with coalesced u32 copies;
-//  * thread 0 walks the length chain (serial, LDS-latency) while threads
-//    1..255 sweep the window word-parallel for needle candidates
-//    (zero-byte trick; false positives verified INCLUDING position 0 —
-//    the subtract borrow fabricates candidates) into a match-start bitmap;
-//  * each value then checks its in-value match-start range against the
-//    bitmap with 1-3 word reads — no per-byte rescan.
 #define CWIN 16384
-#define CVALS 2048
 #define CTHREADS 256
-__global__ void __launch_bounds__(CTHREADS)
-k_bytes_contains(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
-                 const int32_t* __restrict__ ids, int n,
-                 const uint8_t* __restrict__ needle, int nlen,
-                 uint8_t* __restrict__ mask, int32_t* d_error) {
-  __shared__ uint8_t win[CWIN];
-  __shared__ uint32_t offs[CVALS + 1];
-  __shared__ uint8_t nulls[CVALS];
-  __shared__ uint32_t ctrl[3];   // [2]: first value oversized (global scan)
-  __shared__ uint32_t bm[CWIN / 32];
-  int pi = blockIdx.x;
-  if (pi >= n) return;
-  const DevPage pg = pages[ids[pi]];
-  const uint8_t* def_start; uint32_t def_len; bool all_valid;
-  const uint8_t* payload = dec + pg.dst_off;
-  const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
-  const uint32_t nv = pg.num_values, row0 = pg.row_start;
-  const uint32_t page_bytes = (uint32_t)(pg.uncomp_size - (vals - payload));
-  const uint8_t c0 = nlen ? needle[0] : 0;
-
-  SerialRle def(def_start, def_start + (def_start ? def_len : 0), 1);
-  uint32_t done = 0, walk = 0;
-  while (done < nv) {
-    uint32_t rem = page_bytes > walk ? page_bytes - walk : 0;
-    uint32_t wbytes = min((uint32_t)CWIN, rem + 8);
-    // two-phase unconditional full-window copy (16 u32/thread): all global
-    // loads in flight before the LDS stores — see the refill note in
-    // k_lz4_pages. Over-reads past the page stay inside the arena (host
-    // pads the decompressed arena by CWIN).
-    {
-      uint32_t v[CWIN / (CTHREADS * 4)];
-#pragma unroll
-      for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
-        __builtin_memcpy(&v[k], vals + walk + threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u), 4);
-#pragma unroll
-      for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
-        *(uint32_t*)&win[threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u)] = v[k];
-    }
-    for (uint32_t i = threadIdx.x; i < CWIN / 32; i += CTHREADS) bm[i] = 0;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      uint32_t w = 0, cnt = 0;
-      ctrl[2] = 0;
-      if (all_valid) {
-        // streamlined walk: no def levels, no nulls[] writes, branchless
-        // 8-byte LDS length read (the walk is the block's serial spine —
-        // every instruction here is wall time)
-        while (done + cnt < nv && cnt < CVALS && w + 4 <= CWIN) {
-          uint64_t ww;
-          __builtin_memcpy(&ww, &win[w & ~3u], 8);
-          uint32_t l = (uint32_t)(ww >> ((w & 3) * 8));
-          if (w + 4 + l > CWIN) {
-            if (cnt == 0) {                 // oversized single value
-              offs[0] = w | 0x80000000u;
-              ctrl[2] = 1;
-              w += 4 + l;
-              cnt = 1;
-            }
-            break;
-          }
-          offs[cnt] = w;
-          w += 4 + l;
-          cnt++;
-        }
-      } else {
-        while (done + cnt < nv && cnt < CVALS) {
-          int present = (int)def.next();
-          nulls[cnt] = (uint8_t)!present;
-          offs[cnt] = w;
-          if (present) {
-            if (w + 4 > CWIN) {
-              def.unread((uint32_t)present);
-              break;
-            }
-            uint64_t ww;
-            __builtin_memcpy(&ww, &win[w & ~3u], 8);
-            uint32_t l = (uint32_t)(ww >> ((w & 3) * 8));
-            if (w + 4 + l > CWIN) {
-              if (cnt == 0) {
-                offs[0] = w | 0x80000000u;
-                nulls[0] = (uint8_t)2;
-                ctrl[2] = 1;
-                w += 4 + l;
-                cnt = 1;
-              } else {
-                def.unread((uint32_t)present);
-              }
-              break;
-            }
-            w += 4 + l;
-          }
-          cnt++;
-        }
-      }
-      ctrl[0] = cnt;
-      ctrl[1] = w;
-    } else if (nlen) {
-      // word-parallel candidate sweep (threads 1..255)
-      uint32_t lane = threadIdx.x - 1;
-      const uint32_t pat = 0x01010101u * c0;
-      for (uint32_t p = lane * 4u; p + 4 <= wbytes; p += (CTHREADS - 1) * 4u) {
-        uint32_t w = *(const uint32_t*)&win[p];
-        uint32_t x = w ^ pat;
-        uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
-        while (cand) {
-          int b = (__builtin_ctz(cand)) >> 3;
-          cand &= cand - 1;
-          uint32_t pos = p + b;
-          if (pos + nlen <= CWIN) {
-            int k = 0;  // verify from 0: the borrow trick has false positives
-            while (k < nlen && win[pos + k] == needle[k]) k++;
-            if (k == nlen) atomicOr(&bm[pos >> 5], 1u << (pos & 31));
-          }
-        }
-      }
-    }
-    __syncthreads();
-    uint32_t bn = ctrl[0];
-    uint32_t oversized0 = ctrl[2];
-    for (uint32_t i = threadIdx.x; i < bn; i += CTHREADS) {
-      uint8_t nl = all_valid ? (uint8_t)((i == 0 && oversized0) ? 2 : 0)
-                             : nulls[i];
-      uint8_t hit = 0;
-      if (nl == 2) {                        // oversized value: global scan
-        uint32_t o = walk + (offs[i] & 0x7fffffffu);
-        uint32_t vl;
-        __builtin_memcpy(&vl, vals + o, 4);
-        const uint8_t* sp = vals + o + 4;
-        if (nlen == 0) hit = 1;
-        else if (vl >= (uint32_t)nlen) {
-          for (uint32_t j = 0; j + nlen <= vl && !hit; j++) {
-            int k = 0;
-            while (k < nlen && sp[j + k] == needle[k]) k++;
-            hit = (k == nlen);
-          }
-        }
-      } else if (!nl) {
-        if (nlen == 0) hit = 1;
-        else {
-          uint32_t o = offs[i];
-          uint32_t vl;
-          uint32_t sh = (o & 3) * 8;
-          const uint32_t* win32 = (const uint32_t*)win;
-          vl = win32[o >> 2] >> sh;
-          if (sh) vl |= win32[(o >> 2) + 1] << (32 - sh);
-          if (vl >= (uint32_t)nlen) {
-            uint32_t lo = o + 4, hi = o + 4 + vl - nlen;  // inclusive starts
-            uint32_t w0 = lo >> 5, w1 = hi >> 5;
-            if (w0 == w1) {
-              uint32_t m = (hi - lo == 31) ? ~0u
-                                           : (((1u << (hi - lo + 1)) - 1) << (lo & 31));
-              hit = (bm[w0] & m) != 0;
-            } else {
-              uint32_t m0 = ~0u << (lo & 31);
-              uint32_t m1 = ((hi & 31) == 31) ? ~0u : ((1u << ((hi & 31) + 1)) - 1);
-              hit = ((bm[w0] & m0) != 0) | ((bm[w1] & m1) != 0);
-              for (uint32_t w = w0 + 1; w < w1 && !hit; w++) hit |= (bm[w] != 0);
-            }
-          }
-        }
-      }
-      mask[row0 + done + i] &= hit;
-    }
-    __syncthreads();
-    done += bn;
-    walk += ctrl[1];
-    if (bn == 0) break;  // defensive: no progress
-  }
-  if (done != nv && threadIdx.x == 0) atomicExch(d_error, ERR_PAGE);
-}
-
-// ------------------------------------------------------------------
-// row-parallel comparison on i64 arrays
-// ------------------------------------------------------------------
-// ------------------------------------------------------------------
 #define CQMAX 4096
 __device__ inline uint32_t pat_full(uint8_t c0) { return 0x01010101u * c0; }
 // window-parallel CONTAINS over PLAIN byte-array pages: one block per
@@ -1827,11 +1637,6 @@ void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
                       int32_t* d_err) {
   if (n) hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(DELTA_T), 0, st, dec, pages, ids, n, out, valid, d_err);
-}
-void launch_bytes_contains(hipStream_t st, const uint8_t* dec, const DevPage* pages,
-                           const int32_t* ids, int n, const uint8_t* needle, int nlen,
-                           uint8_t* mask, int32_t* d_err) {
-  if (n) hipLaunchKernelGGL(k_bytes_contains, dim3(n), dim3(CTHREADS), 0, st, dec, pages, ids, n, needle, nlen, mask, d_err);
 }
 void launch_cmp_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
                     int64_t lo, int64_t hi, int mode, int hi_excl, int is_f64,

@@ -50,9 +50,6 @@ void launch_expand(hipStream_t, const uint8_t* dec, const DevPage*,
 void launch_delta_i64(hipStream_t, const uint8_t* dec, const DevPage*,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
                       int32_t* d_err);
-void launch_bytes_contains(hipStream_t, const uint8_t* dec, const DevPage*,
-                           const int32_t* ids, int n, const uint8_t* needle,
-                           int nlen, uint8_t* mask, int32_t* d_err);
 void launch_cmp_i64(hipStream_t, const int64_t* col, const uint8_t* valid,
                     int64_t lo, int64_t hi, int mode, int hi_excl, int is_f64,
                     uint8_t* mask, int64_t n);
