@@ -35,7 +35,7 @@ def agg_counters(path):
 def main():
     os.makedirs(DST, exist_ok=True)
     traffic = {}
-    md = ["# Round 1 PMC summary (rocprofv3, gfx950)\n",
+    md = ["# PMC summary (rocprofv3, gfx950)\n",
           "Units: FETCH_SIZE/WRITE_SIZE counters are KB; fetch doubled per the",
           "gfx950 half-reporting correction (MI355X_MICROARCH.md §HBM).",
           "Per-launch values = total / dispatch count.\n"]
@@ -67,7 +67,7 @@ def main():
                      "lz4_backrefs" if "backref" in k or "brres" in k else
                      "dict_count_fused" if "dict_count" in k else
                      "bytes_contains(LIKE)" if "contains" in k else
-                     "decode+filter+groupby" if "agg" in k or "dict_pages" in k or "delta" in k or "plain" in k else k)
+                     "decode+filter+groupby" if "agg" in k or "dict_pages" in k or "delta" in k or "plain" in k or "expand" in k or "def_levels" in k else k)
             traffic[wl].setdefault(short, 0.0)
             traffic[wl][short] += fpl + wpl
     with open(os.path.join(DST, "r01_pmc_summary.md"), "w") as fh:
