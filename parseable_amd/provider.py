@@ -693,14 +693,28 @@ def merge_partials(batches, query):
         # a single partial IS the final result: the exported value columns
         # already hold final values (counts in the value slot for count
         # aggregates, NULL validity where count==0) — pure column zips,
-        # no per-row Python
+        # no per-row Python. The partial arrives key-sorted from the C
+        # export (live-group order == key order only for single-key
+        # plans), so sort via numpy lexsort on key codes instead of a
+        # python tuple sort.
+        import numpy as np
+
         b = batches[0]
-        cols = [b.column(i).to_pylist() for i in range(b.num_columns)]
-        picked = cols[:nk] + [cols[nk + 1 + 2 * i] for i in range(len(aggs))]
-        rows = [list(t) for t in zip(*picked)]
-        rows.sort(key=lambda r: tuple(((1, "") if v is None else (0, v))
-                                      for v in r[:nk]))
-        return rows
+        picked_idx = list(range(nk)) + [nk + 1 + 2 * i
+                                        for i in range(len(aggs))]
+        cols = [b.column(i).to_pylist() for i in picked_idx]
+        if nk:
+            keycodes = []
+            for k in range(nk - 1, -1, -1):
+                vals = cols[k]
+                # NULLs last: encode as a code past every real value
+                uniq = sorted({v for v in vals if v is not None})
+                code = {v: i for i, v in enumerate(uniq)}
+                keycodes.append(np.array(
+                    [len(uniq) if v is None else code[v] for v in vals]))
+            order = np.lexsort(keycodes)
+            return [[c[i] for c in cols] for i in order.tolist()]
+        return [list(t) for t in zip(*cols)]
     acc = {}
     for b in batches:
         if b is None or b.num_rows == 0:
