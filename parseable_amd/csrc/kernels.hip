@@ -1056,28 +1056,61 @@ __global__ void k_delta_sum(const uint8_t* __restrict__ dec,
   mb_sum[(size_t)pi * MAX_MB + m] = s;
 }
 
-__global__ void __launch_bounds__(WAVE)
+#define DSCAN_T 256
+__global__ void __launch_bounds__(DSCAN_T)
 k_delta_scan(const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
              int n, const DeltaHdr* __restrict__ hdr,
              int64_t* __restrict__ mb_sum, int64_t* __restrict__ out,
              uint8_t* __restrict__ valid) {
+  // exclusive scan of up to 8192 miniblock sums, staged through LDS (a
+  // serial scan over GLOBAL memory costs ~8192 dependent round trips):
+  // chunk-serial per lane -> serial scan of 256 partials -> rewrite.
+  __shared__ int64_t s_sum[MAX_MB];
+  __shared__ int64_t s_part[DSCAN_T];
   int pi = blockIdx.x;
   if (pi >= n) return;
-  if (threadIdx.x != 0) return;
   const DeltaHdr h = hdr[pi];
-  if (h.nulls) return;
+  if (h.nulls || !h.n_mb) {
+    if (threadIdx.x == 0 && !h.nulls && h.total) {
+      const DevPage pg = pages[ids[pi]];
+      out[pg.row_start] = h.first;
+      if (valid) valid[pg.row_start] = 1;
+    }
+    return;
+  }
   const DevPage pg = pages[ids[pi]];
-  if (h.total) {
+  if (threadIdx.x == 0 && h.total) {
     out[pg.row_start] = h.first;
     if (valid) valid[pg.row_start] = 1;
   }
-  int64_t run = h.first;
+  const uint32_t nmb = h.n_mb;
   int64_t* ms = mb_sum + (size_t)pi * MAX_MB;
-  for (uint32_t m = 0; m < h.n_mb; m++) {
-    int64_t s = ms[m];
-    ms[m] = run;
-    run += s;
+  for (uint32_t m = threadIdx.x; m < nmb; m += DSCAN_T) s_sum[m] = ms[m];
+  __syncthreads();
+  const uint32_t chunk = (nmb + DSCAN_T - 1) / DSCAN_T;
+  const uint32_t lo = threadIdx.x * chunk;
+  const uint32_t hi = lo + chunk < nmb ? lo + chunk : nmb;
+  int64_t s = 0;
+  for (uint32_t m = lo; m < hi; m++) s += s_sum[m];
+  s_part[threadIdx.x] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int64_t run = h.first;
+    for (int t = 0; t < DSCAN_T; t++) {
+      int64_t x = s_part[t];
+      s_part[t] = run;
+      run += x;
+    }
   }
+  __syncthreads();
+  int64_t run = s_part[threadIdx.x];
+  for (uint32_t m = lo; m < hi; m++) {
+    int64_t x = s_sum[m];
+    s_sum[m] = run;
+    run += x;
+  }
+  __syncthreads();
+  for (uint32_t m = threadIdx.x; m < nmb; m += DSCAN_T) ms[m] = s_sum[m];
 }
 
 __global__ void k_delta_rec(const uint8_t* __restrict__ dec,
@@ -1717,7 +1750,7 @@ void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
   int64_t units = (int64_t)n * MAX_MB;
   int blocks = (int)((units + 255) / 256);
   hipLaunchKernelGGL(k_delta_sum, dim3(blocks), dim3(256), 0, st, dec, pages, ids, n, h, mb_off, mb_bw, mb_md, mb_sum);
-  hipLaunchKernelGGL(k_delta_scan, dim3(n), dim3(WAVE), 0, st, pages, ids, n, h, mb_sum, out, valid);
+  hipLaunchKernelGGL(k_delta_scan, dim3(n), dim3(DSCAN_T), 0, st, pages, ids, n, h, mb_sum, out, valid);
   hipLaunchKernelGGL(k_delta_rec, dim3(blocks), dim3(256), 0, st, dec, pages, ids, n, h, mb_off, mb_bw, mb_md, mb_sum, out, valid);
   hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(DELTA_T), 0, st, dec, pages, ids, n, out, valid, d_err);  // null-page fallback
 }
