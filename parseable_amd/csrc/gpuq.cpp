@@ -182,12 +182,6 @@ struct Partition {
   uint8_t* d_scr = nullptr;        // dense decode scratch (8B/row)
   uint32_t* d_present = nullptr;   // per-page dense counts
   uint8_t* d_tmpvalid = nullptr;   // scratch validity for gid/mask-only cols
-  // delta split-kernel scratch (sized for the largest TK_DELTA_VAL task)
-  uint8_t* d_delta_hdr = nullptr;
-  uint32_t* d_dmb_off = nullptr;
-  uint8_t* d_dmb_bw = nullptr;
-  int64_t* d_dmb_md = nullptr;
-  int64_t* d_dmb_sum = nullptr;
   int32_t* d_all_ids = nullptr;   // identity page-id list for the LZ4 sweep
   DevSeg* d_segs = nullptr;
   DevBr* d_brs = nullptr;
@@ -1162,18 +1156,6 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   HIP_TRY(hipMalloc(&part.d_scr, std::max<int64_t>(part.n_rows * 8, 16)));
   HIP_TRY(hipMalloc(&part.d_present, std::max<size_t>(part.pages.size() * 4, 16)));
   HIP_TRY(hipMalloc(&part.d_tmpvalid, std::max<int64_t>(part.n_rows, 16)));
-  {
-    size_t max_delta = 0;
-    for (auto& [key, ids2] : part.tasks)
-      if (key.first == TK_DELTA_VAL) max_delta = std::max(max_delta, ids2.size());
-    if (max_delta) {
-      HIP_TRY(hipMalloc(&part.d_delta_hdr, max_delta * 40));  // sizeof(DeltaHdr)
-      HIP_TRY(hipMalloc(&part.d_dmb_off, max_delta * 8192 * 4));
-      HIP_TRY(hipMalloc(&part.d_dmb_bw, max_delta * 8192));
-      HIP_TRY(hipMalloc(&part.d_dmb_md, max_delta * 8192 * 8));
-      HIP_TRY(hipMalloc(&part.d_dmb_sum, max_delta * 8192 * 8));
-    }
-  }
   if (plan->is_projection) {
     int64_t n = std::max<int64_t>(part.n_rows, 16);
     HIP_TRY(hipMalloc(&part.d_keys, n * 8));
@@ -1589,9 +1571,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         break;
       case TK_DELTA_VAL:
         launch_delta_i64(st, part.d_dec, part.d_pages, ids, n,
-                         part.d_val[col], part.d_valid[col],
-                         part.d_delta_hdr, part.d_dmb_off, part.d_dmb_bw,
-                         part.d_dmb_md, part.d_dmb_sum, part.d_err);
+                         part.d_val[col], part.d_valid[col], part.d_err);
         break;
       case TK_DICT_MASK:
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
@@ -1990,8 +1970,6 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_rank); F(part.d_scr);
     F(part.d_present); F(part.d_tmpvalid);
-    F(part.d_delta_hdr); F(part.d_dmb_off); F(part.d_dmb_bw);
-    F(part.d_dmb_md); F(part.d_dmb_sum);
     F(part.d_lits_lane); F(part.d_lits_wave);
     F(part.d_cwins); F(part.d_cstarts); F(part.d_brinl);
     F(part.d_segs); F(part.d_brs); F(part.d_pagebrs);

@@ -937,228 +937,29 @@ k_expand(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
 }
 
 // ------------------------------------------------------------------
-// DELTA_BINARY_PACKED i64 -> row-aligned, split into four GLOBALLY
-// parallel phases so one-page-per-wave occupancy (382 waves for a 100M-row
-// shard — half the chip idle) stops being the bound:
-//   k_delta_hdr  (wave/page): header+block walk -> per-miniblock
-//                (offset, bit width, min_delta) in global scratch
-//   k_delta_sum  (thread/miniblock over ALL pages): delta sums
-//   k_delta_scan (wave/page): serial exclusive scan -> start values
-//   k_delta_rec  (thread/miniblock over ALL pages): reconstruct + write
-// k_delta_i64 below remains ONLY for null-bearing time pages (never hit by
-// Parseable streams) as a serial fallback.
+// DELTA_BINARY_PACKED i64 -> row-aligned (one wave per page).
+// Phase A: lane-redundant block-header walk storing per-miniblock
+//   (data offset, bit width, block min_delta ref); Phase B: parallel
+//   per-miniblock delta sums; Phase C: serial scan of miniblock sums;
+//   Phase D: parallel value reconstruction.
+// LDS budget: 8192 miniblocks (262,144 values @ 32/miniblock).
 // ------------------------------------------------------------------
 #define MAX_MB 8192
 #define MAX_BLK 2048
 #define DELTA_T 256
-
-struct DeltaHdr {
-  uint32_t n_mb, per_mini, vals_off, total;
-  int64_t first;
-  uint32_t nulls, mpb;
-};
-
-__global__ void __launch_bounds__(WAVE)
-k_delta_hdr(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
-            const int32_t* __restrict__ ids, int n, DeltaHdr* __restrict__ hdr,
-            uint32_t* __restrict__ mb_off, uint8_t* __restrict__ mb_bw,
-            int64_t* __restrict__ mb_md, int32_t* d_error) {
-  int pi = blockIdx.x;
-  if (pi >= n) return;
-  if (threadIdx.x != 0) return;
-  const DevPage pg = pages[ids[pi]];
-  const uint8_t* def_start; uint32_t def_len; bool all_valid;
-  const uint8_t* payload = dec + pg.dst_off;
-  const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
-  DeltaHdr h{};
-  if (!all_valid) {
-    h.nulls = 1;  // serial fallback kernel owns this page
-    hdr[pi] = h;
-    return;
-  }
-  const uint8_t* q = vals;
-  auto rv = [&]() { uint64_t v = 0; int sh = 0; for (;;) { uint8_t b = *q++; v |= (uint64_t)(b & 0x7f) << sh; if (!(b & 0x80)) return v; sh += 7; } };
-  auto rz = [&]() { uint64_t v = rv(); return (int64_t)(v >> 1) ^ -(int64_t)(v & 1); };
-  uint64_t blk_size = rv(), mpb = rv(), total = rv();
-  int64_t first = rz();
-  uint64_t per_mini = mpb ? blk_size / mpb : 0;
-  if (!per_mini || per_mini % 8 || total > pg.num_values || mpb > 256) {
-    atomicExch(d_error, ERR_DELTA);
-    hdr[pi] = h;  // n_mb 0: later phases skip
-    return;
-  }
-  uint64_t n_deltas = total ? total - 1 : 0;
-  uint32_t n_mb = (uint32_t)((n_deltas + per_mini - 1) / per_mini);
-  uint32_t n_blk = (uint32_t)((n_mb + mpb - 1) / mpb);
-  if (n_mb > MAX_MB || n_blk > MAX_BLK) {
-    atomicExch(d_error, ERR_DELTA);
-    hdr[pi] = h;
-    return;
-  }
-  uint32_t mb = 0;
-  uint32_t* moff = mb_off + (size_t)pi * MAX_MB;
-  uint8_t* mbw = mb_bw + (size_t)pi * MAX_MB;
-  int64_t* mmd = mb_md + (size_t)pi * MAX_MB;
-  for (uint32_t b = 0; b < n_blk; b++) {
-    int64_t md = rz();
-    const uint8_t* bws = q;
-    q += mpb;
-    for (uint64_t m = 0; m < mpb && mb < n_mb; m++, mb++) {
-      moff[mb] = (uint32_t)(q - vals);
-      mbw[mb] = bws[m];
-      mmd[mb] = md;
-      q += (per_mini * bws[m]) / 8;
-    }
-  }
-  h.n_mb = n_mb;
-  h.per_mini = (uint32_t)per_mini;
-  h.vals_off = (uint32_t)(vals - payload);
-  h.total = (uint32_t)total;
-  h.first = first;
-  h.mpb = (uint32_t)mpb;
-  hdr[pi] = h;
-}
-
-__global__ void k_delta_sum(const uint8_t* __restrict__ dec,
-                            const DevPage* __restrict__ pages,
-                            const int32_t* __restrict__ ids, int n,
-                            const DeltaHdr* __restrict__ hdr,
-                            const uint32_t* __restrict__ mb_off,
-                            const uint8_t* __restrict__ mb_bw,
-                            const int64_t* __restrict__ mb_md,
-                            int64_t* __restrict__ mb_sum) {
-  int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int pi = (int)(g / MAX_MB);
-  uint32_t m = (uint32_t)(g % MAX_MB);
-  if (pi >= n) return;
-  const DeltaHdr h = hdr[pi];
-  if (h.nulls || m >= h.n_mb) return;
-  const DevPage pg = pages[ids[pi]];
-  const uint8_t* vals = dec + pg.dst_off + h.vals_off;
-  const uint8_t* p = vals + mb_off[(size_t)pi * MAX_MB + m];
-  int bw = mb_bw[(size_t)pi * MAX_MB + m];
-  int64_t md = mb_md[(size_t)pi * MAX_MB + m];
-  uint64_t n_deltas = h.total ? h.total - 1 : 0;
-  uint64_t cnt = h.per_mini;
-  if ((uint64_t)(m + 1) * h.per_mini > n_deltas)
-    cnt = n_deltas - (uint64_t)m * h.per_mini;
-  int64_t s = 0;
-  uint64_t acc = 0; int nbits = 0;
-  for (uint64_t i = 0; i < cnt; i++) {
-    uint64_t dv = 0;
-    if (bw) {
-      while (nbits < bw) { acc |= (uint64_t)(*p++) << nbits; nbits += 8; }
-      dv = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
-      acc >>= bw; nbits -= bw;
-    }
-    s += md + (int64_t)dv;
-  }
-  mb_sum[(size_t)pi * MAX_MB + m] = s;
-}
-
-#define DSCAN_T 256
-__global__ void __launch_bounds__(DSCAN_T)
-k_delta_scan(const DevPage* __restrict__ pages, const int32_t* __restrict__ ids,
-             int n, const DeltaHdr* __restrict__ hdr,
-             int64_t* __restrict__ mb_sum, int64_t* __restrict__ out,
-             uint8_t* __restrict__ valid) {
-  // exclusive scan of up to 8192 miniblock sums, staged through LDS (a
-  // serial scan over GLOBAL memory costs ~8192 dependent round trips):
-  // chunk-serial per lane -> serial scan of 256 partials -> rewrite.
-  __shared__ int64_t s_sum[MAX_MB];
-  __shared__ int64_t s_part[DSCAN_T];
-  int pi = blockIdx.x;
-  if (pi >= n) return;
-  const DeltaHdr h = hdr[pi];
-  if (h.nulls || !h.n_mb) {
-    if (threadIdx.x == 0 && !h.nulls && h.total) {
-      const DevPage pg = pages[ids[pi]];
-      out[pg.row_start] = h.first;
-      if (valid) valid[pg.row_start] = 1;
-    }
-    return;
-  }
-  const DevPage pg = pages[ids[pi]];
-  if (threadIdx.x == 0 && h.total) {
-    out[pg.row_start] = h.first;
-    if (valid) valid[pg.row_start] = 1;
-  }
-  const uint32_t nmb = h.n_mb;
-  int64_t* ms = mb_sum + (size_t)pi * MAX_MB;
-  for (uint32_t m = threadIdx.x; m < nmb; m += DSCAN_T) s_sum[m] = ms[m];
-  __syncthreads();
-  const uint32_t chunk = (nmb + DSCAN_T - 1) / DSCAN_T;
-  const uint32_t lo = threadIdx.x * chunk;
-  const uint32_t hi = lo + chunk < nmb ? lo + chunk : nmb;
-  int64_t s = 0;
-  for (uint32_t m = lo; m < hi; m++) s += s_sum[m];
-  s_part[threadIdx.x] = s;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    int64_t run = h.first;
-    for (int t = 0; t < DSCAN_T; t++) {
-      int64_t x = s_part[t];
-      s_part[t] = run;
-      run += x;
-    }
-  }
-  __syncthreads();
-  int64_t run = s_part[threadIdx.x];
-  for (uint32_t m = lo; m < hi; m++) {
-    int64_t x = s_sum[m];
-    s_sum[m] = run;
-    run += x;
-  }
-  __syncthreads();
-  for (uint32_t m = threadIdx.x; m < nmb; m += DSCAN_T) ms[m] = s_sum[m];
-}
-
-__global__ void k_delta_rec(const uint8_t* __restrict__ dec,
-                            const DevPage* __restrict__ pages,
-                            const int32_t* __restrict__ ids, int n,
-                            const DeltaHdr* __restrict__ hdr,
-                            const uint32_t* __restrict__ mb_off,
-                            const uint8_t* __restrict__ mb_bw,
-                            const int64_t* __restrict__ mb_md,
-                            const int64_t* __restrict__ mb_sum,
-                            int64_t* __restrict__ out,
-                            uint8_t* __restrict__ valid) {
-  int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int pi = (int)(g / MAX_MB);
-  uint32_t m = (uint32_t)(g % MAX_MB);
-  if (pi >= n) return;
-  const DeltaHdr h = hdr[pi];
-  if (h.nulls || m >= h.n_mb) return;
-  const DevPage pg = pages[ids[pi]];
-  const uint8_t* vals = dec + pg.dst_off + h.vals_off;
-  const uint8_t* p = vals + mb_off[(size_t)pi * MAX_MB + m];
-  int bw = mb_bw[(size_t)pi * MAX_MB + m];
-  int64_t md = mb_md[(size_t)pi * MAX_MB + m];
-  uint64_t n_deltas = h.total ? h.total - 1 : 0;
-  uint64_t cnt = h.per_mini;
-  if ((uint64_t)(m + 1) * h.per_mini > n_deltas)
-    cnt = n_deltas - (uint64_t)m * h.per_mini;
-  int64_t v = mb_sum[(size_t)pi * MAX_MB + m];
-  uint64_t acc = 0; int nbits = 0;
-  uint64_t base = pg.row_start + 1 + (uint64_t)m * h.per_mini;
-  for (uint64_t i = 0; i < cnt; i++) {
-    uint64_t dv = 0;
-    if (bw) {
-      while (nbits < bw) { acc |= (uint64_t)(*p++) << nbits; nbits += 8; }
-      dv = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
-      acc >>= bw; nbits -= bw;
-    }
-    v += md + (int64_t)dv;
-    out[base + i] = v;
-    if (valid) valid[base + i] = 1;
-  }
-}
-
 __global__ void __launch_bounds__(DELTA_T)
 k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
             const int32_t* __restrict__ ids, int n,
             int64_t* __restrict__ out, uint8_t* __restrict__ valid,
             int32_t* d_error) {
+  // LDS budget (fits 160 KiB/CU): 32K off + 8K bw + 16K md + 64K sum = 120K.
+  // mb_sum is re-used in place as the post-scan starting value; the block id
+  // of miniblock m is m / mpb (uniform miniblocks per block).
+  __shared__ uint32_t mb_off[MAX_MB];     // payload-relative offset of miniblock data
+  __shared__ uint8_t mb_bw[MAX_MB];
+  __shared__ int64_t blk_md[MAX_BLK];     // min_delta per block
+  __shared__ int64_t mb_sum[MAX_MB];      // phase B: delta sums; phase C: start values
+
   int pi = blockIdx.x;
   if (pi >= n) return;
   const DevPage pg = pages[ids[pi]];
@@ -1167,8 +968,7 @@ k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
   const uint8_t* payload = dec + pg.dst_off;
   const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
 
-  if (all_valid) return;  // split kernels below own the dense path
-  {
+  if (!all_valid) {
     // nulls in the time column: lane0 fully serial (never hit by Parseable
     // streams — p_timestamp is always set by ingest, event/format/mod.rs:167)
     if (lane != 0) return;
@@ -1211,8 +1011,101 @@ k_delta_i64(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
     return;
   }
 
+  // ---- fast path: no nulls ----
+  // Phase A (lane-redundant): header + block walk
+  const uint8_t* q = vals;
+  auto rv = [&]() { uint64_t v = 0; int sh = 0; for (;;) { uint8_t b = *q++; v |= (uint64_t)(b & 0x7f) << sh; if (!(b & 0x80)) return v; sh += 7; } };
+  auto rz = [&]() { uint64_t v = rv(); return (int64_t)(v >> 1) ^ -(int64_t)(v & 1); };
+  uint64_t blk_size = rv(), mpb = rv(), total = rv();
+  int64_t first = rz();
+  uint64_t per_mini = blk_size / mpb;
+  if (per_mini % 8 || total > pg.num_values || mpb > 256) {
+    if (lane == 0) atomicExch(d_error, ERR_DELTA);
+    return;
+  }
+  uint64_t n_deltas = total ? total - 1 : 0;
+  uint32_t n_mb = (uint32_t)((n_deltas + per_mini - 1) / per_mini);
+  uint32_t n_blk = (uint32_t)((n_mb + mpb - 1) / mpb);
+  if (n_mb > MAX_MB || n_blk > MAX_BLK) {
+    if (lane == 0) atomicExch(d_error, ERR_DELTA);
+    return;
+  }
+  // walk blocks redundantly; every lane records into LDS identically
+  {
+    uint32_t mb = 0;
+    for (uint32_t b = 0; b < n_blk; b++) {
+      int64_t md = rz();
+      if (lane == 0) blk_md[b] = md;
+      const uint8_t* bws = q;
+      q += mpb;
+      for (uint64_t m = 0; m < mpb && mb < n_mb; m++, mb++) {
+        if (lane == 0) {
+          mb_off[mb] = (uint32_t)(q - vals);
+          mb_bw[mb] = bws[m];
+        }
+        q += (per_mini * bws[m]) / 8;
+      }
+    }
+  }
+  __syncthreads();
+
+  // Phase B: per-miniblock delta sums (parallel over miniblocks)
+  for (uint32_t m = lane; m < n_mb; m += DELTA_T) {
+    const uint8_t* p = vals + mb_off[m];
+    int bw = mb_bw[m];
+    int64_t md = blk_md[m / (uint32_t)mpb];
+    uint64_t cnt = per_mini;
+    if ((uint64_t)(m + 1) * per_mini > n_deltas) cnt = n_deltas - (uint64_t)m * per_mini;
+    int64_t s = 0;
+    uint64_t acc = 0; int nbits = 0;
+    for (uint64_t i = 0; i < cnt; i++) {
+      uint64_t dv = 0;
+      if (bw) {
+        while (nbits < bw) { acc |= (uint64_t)(*p++) << nbits; nbits += 8; }
+        dv = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
+        acc >>= bw; nbits -= bw;
+      }
+      s += md + (int64_t)dv;
+    }
+    mb_sum[m] = s;
+  }
+  __syncthreads();
+
+  // Phase C: serial exclusive scan of miniblock sums -> starting value
+  if (lane == 0) {
+    int64_t run = first;
+    for (uint32_t m = 0; m < n_mb; m++) { int64_t s = mb_sum[m]; mb_sum[m] = run; run += s; }
+  }
+  __syncthreads();
+
+  // Phase D: reconstruct values
+  if (total) {
+    if (lane == 0) { out[pg.row_start] = first; if (valid) valid[pg.row_start] = 1; }
+  }
+  for (uint32_t m = lane; m < n_mb; m += DELTA_T) {
+    const uint8_t* p = vals + mb_off[m];
+    int bw = mb_bw[m];
+    int64_t md = blk_md[m / (uint32_t)mpb];
+    uint64_t cnt = per_mini;
+    if ((uint64_t)(m + 1) * per_mini > n_deltas) cnt = n_deltas - (uint64_t)m * per_mini;
+    int64_t v = mb_sum[m];
+    uint64_t acc = 0; int nbits = 0;
+    uint64_t base = 1 + (uint64_t)m * per_mini;  // value index of first delta output
+    for (uint64_t i = 0; i < cnt; i++) {
+      uint64_t dv = 0;
+      if (bw) {
+        while (nbits < bw) { acc |= (uint64_t)(*p++) << nbits; nbits += 8; }
+        dv = (bw >= 64) ? acc : (acc & ((1ull << bw) - 1));
+        acc >>= bw; nbits -= bw;
+      }
+      v += md + (int64_t)dv;
+      out[pg.row_start + base + i] = v;
+      if (valid) valid[pg.row_start + base + i] = 1;
+    }
+  }
 }
 
+// ------------------------------------------------------------------
 #define CWIN 16384
 #define CTHREADS 256
 #define CQMAX 4096
@@ -1742,20 +1635,8 @@ void launch_expand(hipStream_t st, const uint8_t* dec, const DevPage* pages,
 }
 void launch_delta_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
-                      void* hdr, uint32_t* mb_off, uint8_t* mb_bw,
-                      int64_t* mb_md, int64_t* mb_sum, int32_t* d_err) {
-  if (!n) return;
-  DeltaHdr* h = (DeltaHdr*)hdr;
-  hipLaunchKernelGGL(k_delta_hdr, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, h, mb_off, mb_bw, mb_md, d_err);
-  int64_t units = (int64_t)n * MAX_MB;
-  int blocks = (int)((units + 255) / 256);
-  hipLaunchKernelGGL(k_delta_sum, dim3(blocks), dim3(256), 0, st, dec, pages, ids, n, h, mb_off, mb_bw, mb_md, mb_sum);
-  hipLaunchKernelGGL(k_delta_scan, dim3(n), dim3(DSCAN_T), 0, st, pages, ids, n, h, mb_sum, out, valid);
-  hipLaunchKernelGGL(k_delta_rec, dim3(blocks), dim3(256), 0, st, dec, pages, ids, n, h, mb_off, mb_bw, mb_md, mb_sum, out, valid);
-  hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(DELTA_T), 0, st, dec, pages, ids, n, out, valid, d_err);  // null-page fallback
-}
-size_t delta_scratch_bytes_per_page() {
-  return sizeof(DeltaHdr) + (size_t)MAX_MB * (4 + 1 + 8 + 8);
+                      int32_t* d_err) {
+  if (n) hipLaunchKernelGGL(k_delta_i64, dim3(n), dim3(DELTA_T), 0, st, dec, pages, ids, n, out, valid, d_err);
 }
 void launch_cmp_i64(hipStream_t st, const int64_t* col, const uint8_t* valid,
                     int64_t lo, int64_t hi, int mode, int hi_excl, int is_f64,

@@ -49,9 +49,7 @@ void launch_expand(hipStream_t, const uint8_t* dec, const DevPage*,
                    int mode);
 void launch_delta_i64(hipStream_t, const uint8_t* dec, const DevPage*,
                       const int32_t* ids, int n, int64_t* out, uint8_t* valid,
-                      void* hdr, uint32_t* mb_off, uint8_t* mb_bw,
-                      int64_t* mb_md, int64_t* mb_sum, int32_t* d_err);
-size_t delta_scratch_bytes_per_page();
+                      int32_t* d_err);
 void launch_cmp_i64(hipStream_t, const int64_t* col, const uint8_t* valid,
                     int64_t lo, int64_t hi, int mode, int hi_excl, int is_f64,
                     uint8_t* mask, int64_t n);
