@@ -697,23 +697,20 @@ def merge_partials(batches, query):
         # export (live-group order == key order only for single-key
         # plans), so sort via numpy lexsort on key codes instead of a
         # python tuple sort.
-        import numpy as np
+        import pyarrow as pa
+        import pyarrow.compute as pc
 
         b = batches[0]
         picked_idx = list(range(nk)) + [nk + 1 + 2 * i
                                         for i in range(len(aggs))]
         cols = [b.column(i).to_pylist() for i in picked_idx]
         if nk:
-            keycodes = []
-            for k in range(nk - 1, -1, -1):
-                vals = cols[k]
-                # NULLs last: encode as a code past every real value
-                uniq = sorted({v for v in vals if v is not None})
-                code = {v: i for i, v in enumerate(uniq)}
-                keycodes.append(np.array(
-                    [len(uniq) if v is None else code[v] for v in vals]))
-            order = np.lexsort(keycodes)
-            return [[c[i] for c in cols] for i in order.tolist()]
+            kt = pa.table({f"k{i}": b.column(i) for i in range(nk)})
+            order = pc.sort_indices(
+                kt, sort_keys=[pc.SortKey(f"k{i}", "ascending",
+                                          null_placement="at_end")
+                               for i in range(nk)]).to_pylist()
+            return [[c[i] for c in cols] for i in order]
         return [list(t) for t in zip(*cols)]
     acc = {}
     for b in batches:
