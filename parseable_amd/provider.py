@@ -706,9 +706,10 @@ def merge_partials(batches, query):
         cols = [b.column(i).to_pylist() for i in picked_idx]
         if nk:
             kt = pa.table({f"k{i}": b.column(i) for i in range(nk)})
+            # default null_placement is at_end — matches the oracle's
+            # NULLs-last normalized ordering
             order = pc.sort_indices(
-                kt, sort_keys=[pc.SortKey(f"k{i}", "ascending",
-                                          null_placement="at_end")
+                kt, sort_keys=[(f"k{i}", "ascending")
                                for i in range(nk)]).to_pylist()
             return [[c[i] for c in cols] for i in order]
         return [list(t) for t in zip(*cols)]
