@@ -103,7 +103,35 @@ def main():
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--cpu-check", action="store_true")
     ap.add_argument("--data-dir", default="/tmp/gpuq_fuzz")
+    ap.add_argument("--only", type=int, default=-1,
+                    help="run only case #N (rng stream still advances)")
+    ap.add_argument("--isolate", action="store_true",
+                    help="one subprocess per case: a GPU fault kills only "
+                         "that case and is reported with its query")
     args = ap.parse_args()
+
+    if args.isolate:
+        import subprocess
+
+        bad = 0
+        for t in range(args.n):
+            r = subprocess.run(
+                [sys.executable, "-u", os.path.abspath(__file__),
+                 "--n", str(args.n), "--seed", str(args.seed),
+                 "--only", str(t), "--data-dir", args.data_dir],
+                capture_output=True, text=True, timeout=600)
+            out = (r.stdout + r.stderr).strip().splitlines()
+            line = next((l for l in out if l.startswith(("ok", "MISMATCH"))),
+                        None)
+            if r.returncode != 0 or line is None or "MISMATCH" in (line or ""):
+                bad += 1
+                print(f"FAIL #{t} rc={r.returncode}")
+                for l in out[-12:]:
+                    print("   ", l)
+            else:
+                print(line, flush=True)
+        print(f"fuzz done (isolated): {args.n - bad}/{args.n} clean")
+        return 1 if bad else 0
 
     streams = {}
     for cfg_name, cfg in STREAMS.items():
@@ -131,6 +159,9 @@ def main():
         cfg_name = rng.choice(list(STREAMS.keys()))
         stream_dir, files, cfg = streams[cfg_name]
         q = gen_query(rng, cfg, cfg["rows"])
+        if args.only >= 0 and t != args.only:
+            continue
+        print(f"case #{t} [{cfg_name}] {q}", flush=True)
         want = qo.execute(files, dict(q))["rows"]
         if args.cpu_check:
             if any(isinstance(g, dict) for g in q["group_by"]):
