@@ -13,7 +13,10 @@ struct DevPage {
   uint32_t uncomp_size;
   uint32_t num_values;   // rows incl. nulls (data pages)
   uint32_t row_start;    // partition-global row index of first row
-  uint32_t aux;          // index into remap pool (gid decode) / dict-value pool
+  uint32_t aux;          // index into remap pool (gid decode)
+  uint32_t aux_val;      // index into dict-value pool (value/rank decode) —
+                         // separate from aux: a utf8 column can need BOTH
+                         // (group key or count -> gid; min/max -> ranks)
   uint32_t aux_lut;      // index into predicate-LUT pool (dict-mask decode)
   uint32_t dict_n;       // dictionary entry count of this chunk (0 if none)
   uint8_t  optional;     // has def levels (max_def_level == 1)

@@ -794,17 +794,18 @@ extern "C" gpuq_plan* gpuq_plan_build(
         }
 
         bool dict_enc = (pi.encoding == ENC_RLE_DICT || pi.encoding == ENC_PLAIN_DICT);
-        // aux = remap pool (gid decode) XOR dict-value pool (value decode):
-        // a column never needs both (utf8 aggregates beyond COUNT are
-        // rejected above); aux_lut is independent so a group key may also
-        // carry a string predicate.
+        // aux (remap / gid), aux_val (dict values or utf8 sort-ranks) and
+        // aux_lut (predicate LUT) are independent slots: one utf8 column
+        // can be group key + min/max agg + predicate at once (found by the
+        // query fuzzer — aliasing aux produced garbage gids and OOB table
+        // writes).
         if (c.need_gid && dict_enc) {
           dp.aux = remap_base;
           part.tasks[{TK_DICT_GID, t.col_idx}].push_back(page_id);
         }
         if (c.need_val) {
           if (dict_enc) {
-            dp.aux = dictv_base;
+            dp.aux_val = dictv_base;
             part.tasks[{TK_DICT_VAL, t.col_idx}].push_back(page_id);
           } else if (pi.encoding == ENC_PLAIN) {
             part.tasks[{TK_PLAIN_VAL, t.col_idx}].push_back(page_id);

@@ -736,7 +736,7 @@ struct EmitGidP : EmitGid {
 };
 struct EmitDictI64P : EmitDictI64 {
   const int64_t* pool;
-  __device__ void advance(const DevPage& pg) { dictv = pool + pg.aux; }
+  __device__ void advance(const DevPage& pg) { dictv = pool + pg.aux_val; }
 };
 struct EmitDictMaskP : EmitDictMask {
   const uint8_t* pool;

@@ -79,6 +79,26 @@ GOLDEN_QUERIES = {
             "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
             "preds": [{"col": "f_f64", "op": "lt", "lit": 0.25}],
         }),
+        # fuzz-found regressions: a utf8 column serving several roles at
+        # once (group key / min-max rank agg / count / predicate) — aux-slot
+        # aliasing produced garbage gids and OOB group-table writes
+        ("key_and_minmax_same_col", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "f_str2"}],
+            "group_by": ["host", "f_str2"],
+        }),
+        ("rank_count_other_col", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "f_str1"},
+                       {"agg": "count", "col": "f_str1"}],
+            "group_by": ["f_str2", "level"],
+        }),
+        ("key_rank_pred_same_col", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "min", "col": "level"}],
+            "group_by": ["host", "level"],
+            "preds": [{"col": "level", "op": "ne", "lit": "DEBUG"}],
+        }),
     ],
     "g_c1_pages": [
         ("count_by_level", {
