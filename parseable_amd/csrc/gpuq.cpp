@@ -693,8 +693,18 @@ extern "C" gpuq_plan* gpuq_plan_build(
           bool raw_page = dp.raw_copy != 0;
           Lz4Plan lp;
           if (!raw_page) {
+            // test knob: exercise the serial windowed fallback kernel on
+            // arbitrary content (no organic fixture produces a
+            // piece-explosion page)
+            static const bool force_fb =
+                std::getenv("GPUQ_FORCE_LZ4_FALLBACK") != nullptr;
             try {
               lp = lz4_walk(praw, pi.comp_size, pi.uncomp_size, 8192);
+              if (force_fb) {
+                lp.fallback = true;
+                lp.resolved.clear();
+                lp.pieces.clear();
+              }
               // dense short-sequence pages (LZ4 over near-random dict
               // indices) degenerate the segment kernel into a serial token
               // parse — switch those to the all-literal/all-resolved plan

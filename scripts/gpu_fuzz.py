@@ -87,6 +87,11 @@ def gen_query(rng: random.Random, cfg: dict, n_rows: int):
         elif cfg["contains"]:
             preds.append({"col": cfg["contains"], "op": "contains",
                           "lit": rng.choice(["error", "qx", "ab", "zzz"])})
+    if rng.random() < 0.2 and cfg["i64"]:
+        col = rng.choice(cfg["i64"])
+        lo = rng.randint(0, 800_000)
+        preds.append({"col": col, "op": "between", "lo": lo,
+                      "hi": lo + rng.randint(1, 400_000)})
     q["preds"] = preds
     if rng.random() < 0.35:
         n_files = (n_rows + 262_143) // 262_144
@@ -94,6 +99,14 @@ def gen_query(rng: random.Random, cfg: dict, n_rows: int):
         lo = BASE_TS_MS + rng.randint(0, span // 2)
         hi = lo + rng.randint(MINUTE_MS // 4, span)
         q["time_range"] = (lo, hi)
+    if rng.random() < 0.12:  # top-k projection scan instead of aggregation
+        q.pop("select", None)
+        q["group_by"] = []
+        cols = ["p_timestamp"] + rng.sample(cfg["keys"] + cfg["i64"],
+                                            rng.randint(1, 3))
+        q["select_cols"] = cols
+        q["limit"] = rng.choice([5, 50, 500])
+        q["order_by"] = {"col": "p_timestamp", "desc": True}
     return q
 
 
@@ -164,15 +177,21 @@ def main():
         print(f"case #{t} [{cfg_name}] {q}", flush=True)
         want = qo.execute(files, dict(q))["rows"]
         if args.cpu_check:
-            if any(isinstance(g, dict) for g in q["group_by"]):
-                print(f"skip #{t} [{cfg_name}] (acero leg lacks DATE_BIN)")
+            if any(isinstance(g, dict) for g in q["group_by"]) or \
+                    q.get("select_cols"):
+                print(f"skip #{t} [{cfg_name}] (acero leg lacks "
+                      f"DATE_BIN/projection)")
                 continue
             got = qo.execute_acero(files, dict(q))["rows"]
             label = "acero"
         else:
             got, _ = Query(StandardTableProvider(stream_dir, sess)).execute(dict(q))
             label = "gpu"
-        ok = rows_equal(got, want)
+        if q.get("select_cols"):
+            ts_i = q["select_cols"].index("p_timestamp")
+            ok = [r[ts_i] for r in got] == [r[ts_i] for r in want]
+        else:
+            ok = rows_equal(got, want)
         if not ok:
             bad += 1
             print(f"MISMATCH #{t} [{cfg_name}] {q}")
