@@ -199,8 +199,8 @@ def main():
             print(f"  oracle: {want[:4]}")
         else:
             print(f"ok #{t} [{cfg_name}] keys={len(q['group_by'])} "
-                  f"aggs={len(q['select'])} preds={len(q['preds'])} "
-                  f"rows={len(want)}")
+                  f"aggs={len(q.get('select', []))} "
+                  f"preds={len(q['preds'])} rows={len(want)}")
     print(f"fuzz done: {args.n - bad}/{args.n} matched")
     return 1 if bad else 0
 

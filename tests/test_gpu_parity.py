@@ -198,7 +198,7 @@ import json, os, sys
 sys.path.insert(0, %r)
 from parseable_amd import GpuSession, Query, StandardTableProvider
 gdir = %r
-entry = json.load(open(os.path.join(os.path.dirname(gdir), "answers.json")))
+entry = json.load(open(os.path.join(os.path.dirname(gdir), "answers.json")))["answers"]
 case = "g_c1/count_max_by_host_between"
 q = entry[case]["query"]
 provider = StandardTableProvider(os.path.join(gdir, "g_c1"), GpuSession())
