@@ -176,7 +176,8 @@ struct Partition {
   int32_t* d_err = nullptr;
   uint64_t* d_table = nullptr;
   int32_t* d_agg_kind = nullptr;
-  uint8_t* d_needle = nullptr;
+  uint8_t* d_needle = nullptr;        // concatenated CONTAINS needles
+  std::vector<uint32_t> needle_off;   // per plan-pred offset into the pool
   uint32_t* d_rowof = nullptr;     // dense->row map for null-bearing pages
   uint32_t* d_rank = nullptr;      // row->dense rank (expansion pass)
   uint8_t* d_scr = nullptr;        // dense decode scratch (8B/row)
@@ -1207,10 +1208,22 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
               (void**)&part.d_cwins);
   upload_pool(part.cstarts.data(), part.cstarts.size() * sizeof(uint16_t),
               (void**)&part.d_cstarts);
-  // needle buffer (first CONTAINS pred; one per plan supported per column set)
+  // needle pool: every CONTAINS pred's literal, addressed per pred index
+  // (two LIKE predicates — same or different columns — each launch with
+  // their own needle)
   std::string needle;
-  for (auto& pp : plan->preds)
-    if (pp.p.op == GPUQ_CONTAINS) { needle = pp.str_lit; break; }
+  part.needle_off.assign(plan->preds.size(), 0);
+  {
+    std::string pool;
+    for (size_t i = 0; i < plan->preds.size(); i++) {
+      if (plan->preds[i].p.op == GPUQ_CONTAINS) {
+        part.needle_off[i] = (uint32_t)pool.size();
+        pool += plan->preds[i].str_lit;
+        pool.push_back('\0');
+      }
+    }
+    needle = pool;
+  }
   upload_pool(needle.data(), needle.size() + 1, (void**)&part.d_needle);
 
   HIP_TRY(hipStreamSynchronize(part.stream));
@@ -1596,19 +1609,20 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
                       part.d_rank, part.d_tmpvalid, part.d_mask, 2);
         break;
       case TK_BYTES_CONTAINS: {
-        std::string needle;
-        for (int pidx : plan->cols[col].contains_preds)
-          needle = plan->preds[pidx].str_lit;  // single CONTAINS per col
         // def levels zero null rows in the mask and build the dense->row
         // map; the window kernel then has no serial work at all
         launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
                           part.d_tmpvalid, part.d_mask, part.d_rowof,
                           part.d_rank, part.d_present, part.d_err);
         auto rng = part.cwin_ranges.at(col);
-        launch_contains_win(st, part.d_dec, part.d_cwins + rng.first,
-                            (int)rng.second, part.d_pages, part.d_cstarts,
-                            part.d_needle, (int)needle.size(), part.d_rowof,
-                            part.d_mask);
+        // one launch per CONTAINS pred, each with its own needle (mask &=)
+        for (int pidx : plan->cols[col].contains_preds) {
+          const std::string& lit = plan->preds[pidx].str_lit;
+          launch_contains_win(st, part.d_dec, part.d_cwins + rng.first,
+                              (int)rng.second, part.d_pages, part.d_cstarts,
+                              part.d_needle + part.needle_off[pidx],
+                              (int)lit.size(), part.d_rowof, part.d_mask);
+        }
         break;
       }
     }

@@ -87,6 +87,11 @@ def gen_query(rng: random.Random, cfg: dict, n_rows: int):
         elif cfg["contains"]:
             preds.append({"col": cfg["contains"], "op": "contains",
                           "lit": rng.choice(["error", "qx", "ab", "zzz"])})
+    if cfg["contains"] and rng.random() < 0.15:
+        for lit in rng.sample(["error", "qx", "ab", "ror", "err"],
+                              rng.randint(1, 2)):
+            preds.append({"col": cfg["contains"], "op": "contains",
+                          "lit": lit})
     if rng.random() < 0.2 and cfg["i64"]:
         col = rng.choice(cfg["i64"])
         lo = rng.randint(0, 800_000)

@@ -126,6 +126,12 @@ GOLDEN_QUERIES = {
             "select": [{"agg": "count_star"}],
             "preds": [{"col": "message", "op": "contains", "lit": "zzzzzzzzzzzz"}],
         }),
+        ("like_two_needles", {
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["level"],
+            "preds": [{"col": "message", "op": "contains", "lit": "err"},
+                      {"col": "message", "op": "contains", "lit": "ror"}],
+        }),
     ],
     "g_c4": [
         ("three_key_groupby", {
