@@ -671,32 +671,33 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
       }
       if (hdr & 1) {
         uint32_t groups = (uint32_t)(hdr >> 1);
-        // group g: 8 values packed in bw bytes at p + g*bw
-        for (uint32_t g = lane; g < groups; g += WAVE) {
-          // one group = 8 values packed into bw BYTES (8 values x bw bits);
-          // acc may only hold the first 8 bytes — later values re-read a
-          // shifted 8-byte window (shifts must stay < 64).
+        // one VALUE per lane (adjacent lanes -> adjacent rows): the emit
+        // stores coalesce into full lines. The group-of-8-per-lane layout
+        // this replaces made every store instruction span 64 half-used
+        // lines (PMC: 2.6x write amplification); the redundant group-byte
+        // reads (8 lanes share a group) stay in L1/L2.
+        uint32_t run_vals = groups * 8;
+        if (run_vals > nv - v) run_vals = nv - v;
+        for (uint32_t i = lane; i < run_vals; i += WAVE) {
+          uint32_t g = i >> 3;
+          int k = (int)(i & 7);
           const uint8_t* q = p + (size_t)g * bw;
-          uint64_t acc = 0;
-          for (int b = 0; b < bw && b < 8; b++) acc |= (uint64_t)q[b] << (8 * b);
-          uint32_t base = v + g * 8;
           uint32_t mask_v = (bw >= 32) ? 0xffffffffu : ((1u << bw) - 1);
-          for (int k = 0; k < 8; k++) {
-            uint32_t idx;
-            if (k * bw + bw <= 64) {
-              idx = (uint32_t)(acc >> (k * bw)) & mask_v;
-            } else {  // straddles the first u64 window: re-read
-              uint64_t acc2 = 0;
-              const uint8_t* q2 = q + (k * bw) / 8;
-              int shift = (k * bw) % 8;
-              for (int b = 0; b < 8; b++) acc2 |= (uint64_t)q2[b] << (8 * b);
-              idx = (uint32_t)(acc2 >> shift) & mask_v;
-            }
-            if (base + k < nv) {
-              if (idx >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); idx = 0; }
-              emit(target(base + k), idx);
-            }
+          uint32_t idx;
+          if (k * bw + bw <= 64) {
+            uint64_t acc = 0;
+            for (int b = 0; b < bw && b < 8; b++)
+              acc |= (uint64_t)q[b] << (8 * b);
+            idx = (uint32_t)(acc >> (k * bw)) & mask_v;
+          } else {  // value straddles the first u64 window: shifted re-read
+            uint64_t acc2 = 0;
+            const uint8_t* q2 = q + (k * bw) / 8;
+            int shift = (k * bw) % 8;
+            for (int b = 0; b < 8; b++) acc2 |= (uint64_t)q2[b] << (8 * b);
+            idx = (uint32_t)(acc2 >> shift) & mask_v;
           }
+          if (idx >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); idx = 0; }
+          emit(target(v + i), idx);
         }
         p += (size_t)groups * bw;
         uint32_t add = groups * 8;
