@@ -683,18 +683,18 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
           int k = (int)(i & 7);
           const uint8_t* q = p + (size_t)g * bw;
           uint32_t mask_v = (bw >= 32) ? 0xffffffffu : ((1u << bw) - 1);
+          // one unaligned 8B load replaces the byte loop: bits beyond the
+          // group's bw bytes are never selected (k*bw+bw <= 64 => the used
+          // window lies inside q[0..8)); arena padding covers the over-read
           uint32_t idx;
           if (k * bw + bw <= 64) {
-            uint64_t acc = 0;
-            for (int b = 0; b < bw && b < 8; b++)
-              acc |= (uint64_t)q[b] << (8 * b);
+            uint64_t acc;
+            __builtin_memcpy(&acc, q, 8);
             idx = (uint32_t)(acc >> (k * bw)) & mask_v;
           } else {  // value straddles the first u64 window: shifted re-read
-            uint64_t acc2 = 0;
-            const uint8_t* q2 = q + (k * bw) / 8;
-            int shift = (k * bw) % 8;
-            for (int b = 0; b < 8; b++) acc2 |= (uint64_t)q2[b] << (8 * b);
-            idx = (uint32_t)(acc2 >> shift) & mask_v;
+            uint64_t acc2;
+            __builtin_memcpy(&acc2, q + (k * bw) / 8, 8);
+            idx = (uint32_t)(acc2 >> ((k * bw) % 8)) & mask_v;
           }
           if (idx >= dict_n) { atomicExch(d_error, ERR_DICT_RANGE); idx = 0; }
           emit(target(v + i), idx);
