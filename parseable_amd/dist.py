@@ -66,11 +66,17 @@ class DistMerger:
         local = sorted(_batch_rows(batch, len(self.group_by), len(self.aggs)).keys(),
                        key=lambda k: tuple(((1, "") if v is None else (0, v)) for v in k))
         if self.world > 1:
+            # ship the per-agg types with the key space: a rank whose shard
+            # produced no rows (batch None) must still agree on f64/str slots,
+            # or it would emit int-slot values and skip the string-merge path
+            # while data-bearing ranks take it.
             gathered: list = [None] * self.world
-            dist.all_gather_object(gathered, local)
+            dist.all_gather_object(gathered, (local, self.is_f64, self.is_str))
             allk = set()
-            for g in gathered:
-                allk.update(g)
+            for keys, f64s, strs in gathered:
+                allk.update(keys)
+                self.is_f64 = [a or b for a, b in zip(self.is_f64, f64s)]
+                self.is_str = [a or b for a, b in zip(self.is_str, strs)]
         else:
             allk = set(local)
         self.keyspace = sorted(allk, key=lambda k: tuple(((1, "") if v is None else (0, v)) for v in k))
@@ -91,11 +97,12 @@ class DistMerger:
             return merge_partials([batch], self.query)
         G = max(len(self.keyspace), 1)
         n_aggs = len(self.aggs)
-        for a in self.aggs:
-            if a["agg"] in ("min", "max") and self.is_str[
-                self.aggs.index(a)]:
-                raise NotImplementedError(
-                    "utf8 min/max across ranks: merge with merge_partials")
+        # utf8 min/max: rare, tiny payload (one string per live group) —
+        # merged via an object gather alongside the dense tensor reduces.
+        str_minmax: list[dict | None] = [None] * n_aggs
+        for i, a in enumerate(self.aggs):
+            if a["agg"] in ("min", "max") and self.is_str[i]:
+                str_minmax[i] = {}
 
         np_presence = np.zeros(G, dtype=np.int64)
         np_counts = np.zeros((G, n_aggs), dtype=np.int64)
@@ -129,6 +136,21 @@ class DistMerger:
                 if not has.any():
                     continue
                 g2 = gis[has]
+                if str_minmax[i] is not None:
+                    d = str_minmax[i]
+                    vals = vcol.to_pylist()
+                    for gi_, v_ in zip(g2.tolist(),
+                                       (vals[j] for j in np.nonzero(has)[0])):
+                        if v_ is None:
+                            continue
+                        cur = d.get(gi_)
+                        if cur is None:
+                            d[gi_] = v_
+                        elif a["agg"] == "min":
+                            d[gi_] = min(cur, v_)
+                        else:
+                            d[gi_] = max(cur, v_)
+                    continue
                 if self.is_f64[i]:
                     v = vcol.to_numpy(zero_copy_only=False).astype(np.float64)[has]
                     if a["agg"] == "sum":
@@ -167,13 +189,29 @@ class DistMerger:
                 dist.all_reduce(sums_f, op=dist.ReduceOp.SUM)
             if any(a["agg"] == "min" for a in self.aggs):
                 dist.all_reduce(mins, op=dist.ReduceOp.MIN)
+                dist.all_reduce(mins_f, op=dist.ReduceOp.MIN)
             if any(a["agg"] == "max" for a in self.aggs):
                 dist.all_reduce(maxs, op=dist.ReduceOp.MAX)
+                dist.all_reduce(maxs_f, op=dist.ReduceOp.MAX)
+            if any(d is not None for d in str_minmax):
+                gathered: list = [None] * self.world
+                dist.all_gather_object(gathered, str_minmax)
+                for i, a in enumerate(self.aggs):
+                    if str_minmax[i] is None:
+                        continue
+                    merged = {}
+                    pick = min if a["agg"] == "min" else max
+                    for per_rank in gathered:
+                        for gi, v in per_rank[i].items():
+                            cur = merged.get(gi)
+                            merged[gi] = v if cur is None else pick(cur, v)
+                    str_minmax[i] = merged
 
         presence = presence.cpu()
         counts = counts.cpu()
         sums_i, sums_f = sums_i.cpu(), sums_f.cpu()
         mins, maxs = mins.cpu(), maxs.cpu()
+        mins_f, maxs_f = mins_f.cpu(), maxs_f.cpu()
         out = []
         for gi, key in enumerate(self.keyspace):
             if self.group_by and presence[gi].item() == 0:
@@ -188,10 +226,14 @@ class DistMerger:
                 elif a["agg"] == "sum":
                     row.append(sums_f[gi, i].item() if self.is_f64[i]
                                else sums_i[gi, i].item())
+                elif str_minmax[i] is not None:
+                    row.append(str_minmax[i].get(gi))
                 elif a["agg"] == "min":
-                    row.append(mins[gi, i].item())
+                    row.append(mins_f[gi, i].item() if self.is_f64[i]
+                               else mins[gi, i].item())
                 elif a["agg"] == "max":
-                    row.append(maxs[gi, i].item())
+                    row.append(maxs_f[gi, i].item() if self.is_f64[i]
+                               else maxs[gi, i].item())
             out.append(row)
         if not self.group_by and not out:
             out = [[0 if a["agg"] in ("count_star", "count") else None for a in self.aggs]]

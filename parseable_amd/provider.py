@@ -125,14 +125,15 @@ def _file_pruned(file_entry, preds):
 
 
 def _parse_iso_ms(s):
+    """chrono serde emits both '...%S.%fZ' and (when the fractional part is
+    zero) '...%SZ' — accept both, matching catalog.cpp parse_iso_ms."""
     from datetime import datetime, timezone
 
-    return int(
-        datetime.strptime(s, "%Y-%m-%dT%H:%M:%S.%fZ")
-        .replace(tzinfo=timezone.utc)
-        .timestamp()
-        * 1000
-    )
+    try:
+        dt = datetime.strptime(s, "%Y-%m-%dT%H:%M:%S.%fZ")
+    except ValueError:
+        dt = datetime.strptime(s, "%Y-%m-%dT%H:%M:%SZ")
+    return int(dt.replace(tzinfo=timezone.utc).timestamp() * 1000)
 
 
 class EmptyScanResult:
@@ -378,14 +379,16 @@ class StandardTableProvider:
         self.snapshot = self.stream_json["snapshot"]
 
     # is_within_staging_window (stream_schema_provider.rs:936-958): the query
-    # range touches staging when its end reaches past now - window (or there
-    # is no range / no clock).
+    # range touches staging when its upper bound reaches now - window (or
+    # there is no range / no clock). The reference truncates the boundary to
+    # the minute (with_second(0).with_nanosecond(0)) and compares with >=.
     def _staging_touches(self, time_range):
         if self.staging_dir is None:
             return False
         if time_range is None or self.now_ms is None:
             return True
-        return time_range[1] > self.now_ms - self.staging_window_ms
+        boundary = (self.now_ms - self.staging_window_ms) // 60_000 * 60_000
+        return time_range[1] >= boundary
 
     # Snapshot::manifests (catalog/snapshot.rs:42-71): retain manifests whose
     # [lower,upper] overlaps the time predicates.

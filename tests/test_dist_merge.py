@@ -22,12 +22,38 @@ def _partial(keys, presence, vals, cnts):
     )
 
 
+def _partial2(keys, presence, aggs):
+    """aggs: list of (values, counts, pa type) per aggregate slot."""
+    cols = {"level": pa.array(keys, type=pa.string()),
+            "__presence": pa.array(presence, type=pa.int64())}
+    for i, (vals, cnts, typ) in enumerate(aggs):
+        cols[f"agg{i}"] = pa.array(vals, type=typ)
+        cols[f"agg{i}_count"] = pa.array(cnts, type=pa.int64())
+    return pa.record_batch(cols)
+
+
 QUERY = {
     "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
     "group_by": ["level"],
 }
 
 EXPECTED = [["ERROR", 5, 400], ["INFO", 5, 300], ["WARN", 4, 77]]
+
+# float64 min/max across ranks, with rank 1 contributing NOTHING (batch
+# None): the agreed per-agg types must come from the gather, not the local
+# batch (ADVICE round 1, high + medium findings).
+QUERY_F = {
+    "select": [{"agg": "min", "col": "f_f64"}, {"agg": "max", "col": "f_f64"}],
+    "group_by": ["level"],
+}
+EXPECTED_F = [["INFO", 1.5, 9.0], ["WARN", 2.5, 7.25]]
+
+# utf8 min/max across ranks (was NotImplementedError)
+QUERY_S = {
+    "select": [{"agg": "min", "col": "host"}, {"agg": "max", "col": "host"}],
+    "group_by": ["level"],
+}
+EXPECTED_S = [["INFO", "a-host", "z-host"], ["WARN", "b", "b"]]
 
 
 def _rank_main(rank, world, port, q):
@@ -39,14 +65,37 @@ def _rank_main(rank, world, port, q):
         "gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank, world_size=world
     )
     try:
+        out = {}
         if rank == 0:
             batch = _partial(["INFO", "WARN"], [3, 4], [300, 77], [3, 4])
         else:
             batch = _partial(["INFO", "ERROR"], [2, 5], [250, 400], [2, 5])
         m = DistMerger(QUERY, device="cpu")
         m.setup(batch)
-        rows = m.step(batch)
-        q.put((rank, rows))
+        out["int"] = m.step(batch)
+
+        if rank == 0:
+            fb = _partial2(["INFO", "WARN"], [3, 4],
+                           [([1.5, 2.5], [3, 4], pa.float64()),
+                            ([9.0, 7.25], [3, 4], pa.float64())])
+        else:
+            fb = None
+        m = DistMerger(QUERY_F, device="cpu")
+        m.setup(fb)
+        out["f64"] = m.step(fb)
+
+        if rank == 0:
+            sb = _partial2(["INFO"], [3],
+                           [(["a-host"], [3], pa.string()),
+                            (["m-host"], [3], pa.string())])
+        else:
+            sb = _partial2(["INFO", "WARN"], [2, 1],
+                           [(["c", "b"], [2, 1], pa.string()),
+                            (["z-host", "b"], [2, 1], pa.string())])
+        m = DistMerger(QUERY_S, device="cpu")
+        m.setup(sb)
+        out["str"] = m.step(sb)
+        q.put((rank, out))
     finally:
         dist.destroy_process_group()
 
@@ -71,5 +120,7 @@ def test_gloo_world2_merge():
     for p in procs:
         p.join(60)
         assert p.exitcode == 0
-    assert results[0] == EXPECTED
-    assert results[1] == EXPECTED  # every rank sees the same final table
+    for r in (0, 1):   # every rank sees the same final table
+        assert results[r]["int"] == EXPECTED
+        assert results[r]["f64"] == EXPECTED_F
+        assert results[r]["str"] == EXPECTED_S

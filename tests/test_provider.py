@@ -149,3 +149,34 @@ def test_merge_partials_null_key_sorts_last():
 def test_merge_partials_empty():
     q = {"select": [{"agg": "count_star"}, {"agg": "sum", "col": "x"}]}
     assert merge_partials([], q) == [[0, None]]
+
+
+def test_parse_iso_ms_accepts_both_chrono_forms():
+    # chrono serde emits '...%S.%fZ' normally but '...%SZ' when the
+    # fractional part is zero: both must parse, matching
+    # the native planner's catalog.cpp parse_iso_ms.
+    from parseable_amd.provider import _parse_iso_ms
+
+    assert _parse_iso_ms("2025-09-01T00:01:02.500Z") == 1756684862500
+    assert _parse_iso_ms("2025-09-01T00:01:02Z") == 1756684862000
+
+
+def test_staging_window_boundary_minute_truncated(tmp_path):
+    # is_within_staging_window (stream_schema_provider.rs:936-958): boundary
+    # = (now - 5min) truncated to the minute, compared with >= on the upper
+    # bound.
+    import json as _json
+
+    sdir = tmp_path / "stream"
+    sdir.mkdir()
+    (sdir / "stream.json").write_text(_json.dumps(
+        {"snapshot": {"manifest_list": []}}))
+    staging = tmp_path / "staging"
+    staging.mkdir()
+    now_ms = 1_700_000_000_000 + 37_123  # mid-minute "now"
+    p = StandardTableProvider(str(sdir), None, staging_dir=str(staging),
+                              now_ms=now_ms)
+    boundary = (now_ms - 5 * 60_000) // 60_000 * 60_000
+    assert p._staging_touches((0, boundary))          # exactly on: touches
+    assert p._staging_touches((0, boundary + 1))
+    assert not p._staging_touches((0, boundary - 1))  # within-minute end: not
