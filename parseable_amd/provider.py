@@ -136,6 +136,36 @@ def _parse_iso_ms(s):
     return int(dt.replace(tzinfo=timezone.utc).timestamp() * 1000)
 
 
+def supports_filters_pushdown(preds):
+    """Mirror of TableProvider::supports_filters_pushdown
+    (stream_schema_provider.rs:759-777) + expr_in_boundary (:960-976):
+    a filter is Exact only when it is a minute-aligned comparison
+    (>, >=, <, <=) on p_timestamp — those are fully answered by the
+    minute-long partition prefixes; everything else is Inexact (row-group
+    pruning + in-scan evaluation). A BETWEEN on p_timestamp is the injected
+    two-bound conjunction (query/mod.rs:829-888): Exact iff both bounds are
+    minute-aligned. NOTE: the gpuq engine evaluates EVERY pushed predicate
+    on the GPU, so results are exact either way — this surface exists for a
+    Rust ExecutionPlan shim that must answer DataFusion's planner
+    (INTEGRATION.md)."""
+    out = []
+    for p in preds:
+        op = p["op"]
+        if p.get("col") == "p_timestamp":
+            lit = p.get("lit")
+            if (op in ("gt", "ge", "lt", "le")
+                    and isinstance(lit, int) and not isinstance(lit, bool)
+                    and lit % 60_000 == 0):
+                out.append("exact")
+                continue
+            if (op == "between" and p["lo"] % 60_000 == 0
+                    and p["hi"] % 60_000 == 0):
+                out.append("exact")
+                continue
+        out.append("inexact")
+    return out
+
+
 class EmptyScanResult:
     """All files pruned at planning time: the scan is an empty relation —
     the aggregate result over zero rows (DataFusion's EmptyExec analog)."""

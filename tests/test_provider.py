@@ -180,3 +180,26 @@ def test_staging_window_boundary_minute_truncated(tmp_path):
     assert p._staging_touches((0, boundary))          # exactly on: touches
     assert p._staging_touches((0, boundary + 1))
     assert not p._staging_touches((0, boundary - 1))  # within-minute end: not
+
+
+def test_supports_filters_pushdown_boundary():
+    # stream_schema_provider.rs:759-777 + expr_in_boundary :960-976:
+    # Exact only for minute-aligned >,>=,<,<= (or the injected BETWEEN)
+    # on p_timestamp; everything else Inexact.
+    from parseable_amd.provider import supports_filters_pushdown as sfp
+
+    m = 60_000
+    assert sfp([{"col": "p_timestamp", "op": "ge", "lit": 5 * m}]) == ["exact"]
+    assert sfp([{"col": "p_timestamp", "op": "lt", "lit": 5 * m}]) == ["exact"]
+    # not minute-aligned -> Inexact (second()/nanosecond() != 0)
+    assert sfp([{"col": "p_timestamp", "op": "ge", "lit": 5 * m + 1}]) == ["inexact"]
+    # eq is not in the boundary op set
+    assert sfp([{"col": "p_timestamp", "op": "eq", "lit": 5 * m}]) == ["inexact"]
+    # non-time columns always Inexact
+    assert sfp([{"col": "level", "op": "eq", "lit": "INFO"},
+                {"col": "latency", "op": "ge", "lit": 0}]) == ["inexact", "inexact"]
+    # injected time range: exact iff both bounds minute-aligned
+    assert sfp([{"col": "p_timestamp", "op": "between", "lo": 0, "hi": 2 * m,
+                 "hi_exclusive": True}]) == ["exact"]
+    assert sfp([{"col": "p_timestamp", "op": "between", "lo": 0, "hi": 2 * m + 5,
+                 "hi_exclusive": True}]) == ["inexact"]
