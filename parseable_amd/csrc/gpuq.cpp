@@ -328,6 +328,45 @@ int64_t now_ns() {
   return (int64_t)ts.tv_sec * 1000000000 + ts.tv_nsec;
 }
 
+// Simple parallel-for over [0, n): the 1 B-row configs have thousands of
+// files/chunks whose footer parses, dict decompresses and LZ4 structure
+// walks are independent — serial host planning was the wall at that scale.
+// Rethrows the first worker exception on the calling thread.
+template <class F>
+void parallel_for(size_t n, F&& fn, size_t max_threads = 0) {
+  if (n == 0) return;
+  size_t hw = std::thread::hardware_concurrency();
+  if (hw == 0) hw = 8;
+  if (max_threads) hw = std::min(hw, max_threads);
+  size_t nthreads = std::min(n, hw);
+  if (nthreads <= 1) {
+    for (size_t i = 0; i < n; i++) fn(i);
+    return;
+  }
+  std::atomic<size_t> next{0};
+  std::mutex emu;
+  std::exception_ptr eptr;
+  auto worker = [&]() {
+    for (;;) {
+      size_t i = next.fetch_add(1);
+      if (i >= n) return;
+      try {
+        fn(i);
+      } catch (...) {
+        std::lock_guard<std::mutex> g(emu);
+        if (!eptr) eptr = std::current_exception();
+        next.store(n);  // drain
+        return;
+      }
+    }
+  };
+  std::vector<std::thread> pool;
+  pool.reserve(nthreads);
+  for (size_t t = 0; t < nthreads; t++) pool.emplace_back(worker);
+  for (auto& th : pool) th.join();
+  if (eptr) std::rethrow_exception(eptr);
+}
+
 }  // namespace
 
 extern "C" gpuq_plan* gpuq_plan_build(
@@ -346,8 +385,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
     throw std::runtime_error("empty file list: caller should use the empty-relation path "
                              "(provider returns the empty aggregate without a scan)");
 
-  // --- map files + parse footers ---
-  for (int32_t i = 0; i < n_files; i++) {
+  // --- map files + parse footers (parallel: 1 B rows ≈ 3,815 files) ---
+  plan->files.resize(n_files);
+  parallel_for((size_t)n_files, [&](size_t i) {
     auto mf = std::make_unique<MappedFile>();
     mf->path = files[i].path;
     mf->fd = open(files[i].path, O_RDONLY);
@@ -358,8 +398,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
     mf->data = (const uint8_t*)mmap(nullptr, mf->size, PROT_READ, MAP_PRIVATE, mf->fd, 0);
     if (mf->data == MAP_FAILED) throw std::runtime_error("mmap failed: " + mf->path);
     mf->meta = parse_footer(mf->data, mf->size);
-    plan->files.push_back(std::move(mf));
-  }
+    plan->files[i] = std::move(mf);
+  });
   const FileMeta& fm0 = plan->files[0]->meta;
 
   // --- query spec -> column roles ---
@@ -552,7 +592,12 @@ extern "C" gpuq_plan* gpuq_plan_build(
     plan->parts[best].rgs.push_back(r);
   }
 
-  // --- per-partition host build: page walks, dict processing, device images ---
+  // --- per-partition host build: page walks, dict processing, device
+  // images. The 1 B-row configs have thousands of (row-group x column)
+  // chunks: every per-chunk pass below (page-header walks, dict-page
+  // decompress, LZ4 structure walks) runs parallel across host cores, with
+  // serial phases only for global-dict id assignment and the
+  // order-preserving merges into the device pools. ---
   for (auto& part : plan->parts) {
     uint32_t row_cursor = 0;
     for (auto& r : part.rgs) {
@@ -562,6 +607,14 @@ extern "C" gpuq_plan* gpuq_plan_build(
     }
     part.n_rows = row_cursor;
 
+    // chunk stubs (serial, cheap): raw-arena offsets + byte accounting
+    struct Stub {
+      int file_idx, rg_idx, col_idx;
+      const ColumnChunkMeta* cm;
+      uint64_t raw_off;
+      uint32_t rstart;
+    };
+    std::vector<Stub> stubs;
     for (auto& r : part.rgs) {
       const auto& mf = *plan->files[r.file_idx];
       const auto& rg = mf.meta.row_groups[r.rg_idx];
@@ -571,79 +624,128 @@ extern "C" gpuq_plan* gpuq_plan_build(
         int si = mf.meta.col_index(c.name);
         if (si < 0) throw std::runtime_error("column missing in file: " + c.name);
         const auto& cm = rg.chunks[si];
-        ChunkTask t;
-        t.file_idx = r.file_idx; t.rg_idx = r.rg_idx; t.col_idx = (int)ci;
-        t.cm = &cm;
-        t.raw_off = part.raw_bytes;
+        stubs.push_back({r.file_idx, r.rg_idx, (int)ci, &cm, part.raw_bytes,
+                         r.row_start});
         part.raw_bytes += cm.total_compressed_size;
         part.bytes_scanned += cm.total_compressed_size;
-        t.pages = walk_pages(mf.data, cm, rg.num_rows);
-
-        // host-side dictionary processing
-        for (auto& pi : t.pages) {
-          if (pi.type != PAGE_DICT) continue;
-          std::vector<uint8_t> dbuf(pi.uncomp_size);
-          const uint8_t* d;
-          if (cm.codec == CODEC_UNCOMPRESSED) {
-            d = mf.data + pi.payload_off;
-          } else if (cm.codec == CODEC_LZ4_RAW) {
-            // ALWAYS try LZ4 first: the writer compresses every v1 page, and
-            // comp_size may coincidentally equal uncomp_size (seen in golden
-            // data); only a failed decode of an equal-size page means the
-            // page was stored raw (robustness for other writers).
-            int n = lz4_decompress_host(mf.data + pi.payload_off, pi.comp_size,
-                                        dbuf.data(), dbuf.size());
-            if (n == pi.uncomp_size) d = dbuf.data();
-            else if (pi.comp_size == pi.uncomp_size) d = mf.data + pi.payload_off;
-            else throw std::runtime_error("dict page lz4 failure");
-          } else throw std::runtime_error("unsupported codec");
-          if (c.phys == PT_BYTE_ARRAY) {
-            const uint8_t* q = d;
-            t.remap.reserve(pi.num_values);
-            t.lut.reserve(pi.num_values);
-            for (int32_t k = 0; k < pi.num_values; k++) {
-              uint32_t l; memcpy(&l, q, 4); q += 4;
-              int32_t g = 0;
-              if (c.need_gid || c.need_rank)
-                g = c.gid_of(std::string((const char*)q, l));
-              if (c.need_gid)
-                t.remap.push_back(g);
-              if (c.need_rank)
-                t.dictv.push_back(g);  // post-pass rewrites gid -> sort rank
-              if (!c.lut_preds.empty()) {
-                uint8_t ok = 1;
-                for (int pidx : c.lut_preds) {
-                  const auto& pp = plan->preds[pidx];
-                  ok &= (uint8_t)eval_str_pred(pp.p, pp.str_lit, q, l);
-                }
-                t.lut.push_back(ok);
-              }
-              q += l;
-            }
-          } else if (c.phys == PT_INT64) {
-            t.dictv.resize(pi.num_values);
-            memcpy(t.dictv.data(), d, 8 * (size_t)pi.num_values);
-          } else if (c.phys == PT_INT32) {
-            t.dictv.resize(pi.num_values);
-            for (int32_t k = 0; k < pi.num_values; k++) {
-              int32_t v; memcpy(&v, d + 4 * (size_t)k, 4); t.dictv[k] = v;
-            }
-          } else if (c.phys == PT_DOUBLE) {
-            t.dictv.resize(pi.num_values);
-            memcpy(t.dictv.data(), d, 8 * (size_t)pi.num_values);
-          }
-        }
-        for (auto& pi : t.pages)
-          if (pi.type == PAGE_DATA && pi.encoding == ENC_PLAIN)
-            t.has_plain_data_pages = true;
-        if (t.has_plain_data_pages && (c.need_gid || c.need_rank))
-          throw std::runtime_error("dict-only utf8 operation with PLAIN fallback pages "
-                                   "(high-cardinality hashing): next row (SURVEY §8f) — " + c.name);
-        part.chunks.push_back(std::move(t));
       }
     }
 
-    // device page descriptors + aux pools + task lists
+    // phase 1 (parallel): page walks + dict-page decompress/processing —
+    // everything except global-dictionary id assignment (shared state)
+    part.chunks.resize(stubs.size());
+    std::vector<std::vector<std::string>> dict_strs(stubs.size());
+    parallel_for(stubs.size(), [&](size_t i) {
+      const Stub& s = stubs[i];
+      const auto& mf = *plan->files[s.file_idx];
+      const auto& cm = *s.cm;
+      auto& c = plan->cols[s.col_idx];
+      ChunkTask t;
+      t.file_idx = s.file_idx; t.rg_idx = s.rg_idx; t.col_idx = s.col_idx;
+      t.cm = s.cm;
+      t.raw_off = s.raw_off;
+      t.pages = walk_pages(mf.data, cm, mf.meta.row_groups[s.rg_idx].num_rows);
+
+      // host-side dictionary processing
+      for (auto& pi : t.pages) {
+        if (pi.type != PAGE_DICT) continue;
+        std::vector<uint8_t> dbuf(pi.uncomp_size);
+        const uint8_t* d;
+        if (cm.codec == CODEC_UNCOMPRESSED) {
+          d = mf.data + pi.payload_off;
+        } else if (cm.codec == CODEC_LZ4_RAW) {
+          // ALWAYS try LZ4 first: the writer compresses every v1 page, and
+          // comp_size may coincidentally equal uncomp_size (seen in golden
+          // data); only a failed decode of an equal-size page means the
+          // page was stored raw (robustness for other writers).
+          int n = lz4_decompress_host(mf.data + pi.payload_off, pi.comp_size,
+                                      dbuf.data(), dbuf.size());
+          if (n == pi.uncomp_size) d = dbuf.data();
+          else if (pi.comp_size == pi.uncomp_size) d = mf.data + pi.payload_off;
+          else throw std::runtime_error("dict page lz4 failure");
+        } else throw std::runtime_error("unsupported codec");
+        if (c.phys == PT_BYTE_ARRAY) {
+          const uint8_t* q = d;
+          if (c.need_gid || c.need_rank) dict_strs[i].reserve(pi.num_values);
+          for (int32_t k = 0; k < pi.num_values; k++) {
+            uint32_t l; memcpy(&l, q, 4); q += 4;
+            if (c.need_gid || c.need_rank)
+              dict_strs[i].emplace_back((const char*)q, l);
+            if (!c.lut_preds.empty()) {
+              uint8_t ok = 1;
+              for (int pidx : c.lut_preds) {
+                const auto& pp = plan->preds[pidx];
+                ok &= (uint8_t)eval_str_pred(pp.p, pp.str_lit, q, l);
+              }
+              t.lut.push_back(ok);
+            }
+            q += l;
+          }
+        } else if (c.phys == PT_INT64) {
+          t.dictv.resize(pi.num_values);
+          memcpy(t.dictv.data(), d, 8 * (size_t)pi.num_values);
+        } else if (c.phys == PT_INT32) {
+          t.dictv.resize(pi.num_values);
+          for (int32_t k = 0; k < pi.num_values; k++) {
+            int32_t v; memcpy(&v, d + 4 * (size_t)k, 4); t.dictv[k] = v;
+          }
+        } else if (c.phys == PT_DOUBLE) {
+          t.dictv.resize(pi.num_values);
+          memcpy(t.dictv.data(), d, 8 * (size_t)pi.num_values);
+        }
+      }
+      for (auto& pi : t.pages)
+        if (pi.type == PAGE_DATA && pi.encoding == ENC_PLAIN)
+          t.has_plain_data_pages = true;
+      if (t.has_plain_data_pages && (c.need_gid || c.need_rank))
+        throw std::runtime_error("dict-only utf8 operation with PLAIN fallback pages "
+                                 "(high-cardinality hashing): next row (SURVEY §8f) — " + c.name);
+      part.chunks[i] = std::move(t);
+    });
+
+    // phase 2 (serial): global-dictionary gid assignment per chunk, in order
+    for (size_t i = 0; i < part.chunks.size(); i++) {
+      auto& t = part.chunks[i];
+      auto& c = plan->cols[t.col_idx];
+      if (c.phys != PT_BYTE_ARRAY || dict_strs[i].empty()) continue;
+      if (c.need_gid) t.remap.reserve(dict_strs[i].size());
+      if (c.need_rank) t.dictv.reserve(dict_strs[i].size());
+      for (auto& sv : dict_strs[i]) {
+        int32_t g = c.gid_of(sv);
+        if (c.need_gid) t.remap.push_back(g);
+        if (c.need_rank) t.dictv.push_back(g);  // post-pass rewrites gid->rank
+      }
+      dict_strs[i].clear();
+      dict_strs[i].shrink_to_fit();
+    }
+
+    // phase 3 (serial): pool / page-id / dec-arena bases per chunk
+    struct Bases {
+      uint32_t remap, dictv, lut;
+      uint64_t dec;
+      int32_t pid;
+    };
+    std::vector<Bases> bases(part.chunks.size());
+    int32_t pid_cursor = 0;
+    for (size_t i = 0; i < part.chunks.size(); i++) {
+      auto& t = part.chunks[i];
+      bases[i] = {(uint32_t)part.remap_pool.size(),
+                  (uint32_t)part.dictv_pool.size(),
+                  (uint32_t)part.lut_pool.size(), part.dec_bytes, pid_cursor};
+      t.dictv_pool_base = bases[i].dictv;
+      part.remap_pool.insert(part.remap_pool.end(), t.remap.begin(), t.remap.end());
+      part.dictv_pool.insert(part.dictv_pool.end(), t.dictv.begin(), t.dictv.end());
+      part.lut_pool.insert(part.lut_pool.end(), t.lut.begin(), t.lut.end());
+      for (auto& pi : t.pages) {
+        if (pi.type != PAGE_DATA) continue;
+        pid_cursor++;
+        // 16-align page images so dst and LDS-ring offsets share alignment
+        part.dec_bytes += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
+      }
+    }
+
+    // phase 4 (parallel): per-chunk device images — DevPage descriptors,
+    // LZ4 structure walks, seg/lit/backref records, task-list routing
     struct CItem {
       const uint8_t* praw;
       int32_t comp, uncomp, page_id;
@@ -653,29 +755,38 @@ extern "C" gpuq_plan* gpuq_plan_build(
       int col;
     };
     std::vector<CItem> citems;
-    for (auto& t : part.chunks) {
+    struct CB {
+      std::vector<DevPage> pages;
+      std::map<std::pair<int,int>, std::vector<int32_t>> tasks;
+      std::vector<DevSeg> segs;
+      std::vector<DevBr> brs;
+      std::vector<DevPageBr> pagebrs;
+      std::vector<DevBrRes> res_lane, res_wave;
+      std::vector<DevPiece> piece_pool;
+      std::vector<DevBrInl> brinl;
+      std::vector<DevLit> lits_lane, lits_wave;
+      std::vector<CItem> citems;
+    };
+    std::vector<CB> cbs(part.chunks.size());
+    parallel_for(part.chunks.size(), [&](size_t ti) {
+      auto& t = part.chunks[ti];
       auto& c = plan->cols[t.col_idx];
-      uint32_t remap_base = (uint32_t)part.remap_pool.size();
-      uint32_t dictv_base = (uint32_t)part.dictv_pool.size();
-      t.dictv_pool_base = dictv_base;
-      uint32_t lut_base = (uint32_t)part.lut_pool.size();
-      part.remap_pool.insert(part.remap_pool.end(), t.remap.begin(), t.remap.end());
-      part.dictv_pool.insert(part.dictv_pool.end(), t.dictv.begin(), t.dictv.end());
-      part.lut_pool.insert(part.lut_pool.end(), t.lut.begin(), t.lut.end());
-
+      auto& cb = cbs[ti];
       const auto& mf = *plan->files[t.file_idx];
-      uint32_t rstart = 0;
-      for (auto& r : part.rgs)
-        if (r.file_idx == t.file_idx && r.rg_idx == t.rg_idx) { rstart = r.row_start; break; }
+      const uint32_t remap_base = bases[ti].remap;
+      const uint32_t dictv_base = bases[ti].dictv;
+      const uint32_t lut_base = bases[ti].lut;
+      uint64_t dec_off = bases[ti].dec;
+      int32_t page_id = bases[ti].pid;
+      const uint32_t rstart = stubs[ti].rstart;
 
       uint32_t row_in_rg = 0;
       for (auto& pi : t.pages) {
         if (pi.type != PAGE_DATA) continue;
         DevPage dp{};
         dp.src_off = t.raw_off + (uint64_t)(pi.payload_off - t.cm->start_offset());
-        dp.dst_off = part.dec_bytes;
-        // 16-align page images so dst and LDS-ring offsets share alignment
-        part.dec_bytes += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
+        dp.dst_off = dec_off;
+        dec_off += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
         dp.comp_size = pi.comp_size;
         dp.uncomp_size = pi.uncomp_size;
         dp.num_values = pi.num_values;
@@ -687,7 +798,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
             {t.remap.size(), t.dictv.size(), t.lut.size()});
         dp.encoding = (uint8_t)pi.encoding;
         dp.phys = (uint8_t)c.phys;
-        int32_t page_id = (int32_t)part.pages.size();
+        const int32_t this_pid = page_id++;
         // host LZ4 structure walk -> parallel segments + backref records
         {
           const uint8_t* praw = mf.data + pi.payload_off;
@@ -727,7 +838,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
             sgl.comp_len = pi.comp_size;
             sgl.out_len = pi.uncomp_size;
             sgl.raw = 1;
-            part.segs.push_back(sgl);
+            cb.segs.push_back(sgl);
           } else {
             for (const auto& sg : lp.segs) {
               DevSeg d2{};
@@ -736,19 +847,19 @@ extern "C" gpuq_plan* gpuq_plan_build(
               d2.comp_len = sg.comp_len;
               d2.out_len = sg.out_len;
               d2.big = sg.big;
-              part.segs.push_back(d2);
+              cb.segs.push_back(d2);
             }
             for (const auto& lt : lp.lits) {
               DevLit dl{dp.src_off + lt.src, dp.dst_off + lt.dst, lt.len, 0};
-              (lt.len <= 256 ? part.lits_lane : part.lits_wave).push_back(dl);
+              (lt.len <= 256 ? cb.lits_lane : cb.lits_wave).push_back(dl);
             }
             if (lp.fallback) {
               // piece explosion: serial windowed wave per page
-              DevPageBr pb{(uint32_t)part.brs.size(), (uint32_t)lp.backrefs.size()};
-              part.pagebrs.push_back(pb);
+              DevPageBr pb{(uint32_t)cb.brs.size(), (uint32_t)lp.backrefs.size()};
+              cb.pagebrs.push_back(pb);
               for (const auto& br : lp.backrefs)
-                part.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
-                                    br.len, 0});
+                cb.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
+                                  br.len, 0});
             } else if (!lp.resolved.empty()) {
               // litpar pieces are literal-backed: the host can read any
               // pattern of <= 8 bytes straight from the compressed stream
@@ -785,20 +896,20 @@ extern "C" gpuq_plan* gpuq_plan_build(
                   if (ok) {
                     uint64_t pat;
                     std::memcpy(&pat, buf, 8);
-                    part.brinl.push_back(
+                    cb.brinl.push_back(
                         {dp.dst_off + rr.dst, pat, rr.len, pat_len});
                     continue;
                   }
                 }
-                uint32_t ps = (uint32_t)part.piece_pool.size();
+                uint32_t ps = (uint32_t)cb.piece_pool.size();
                 for (uint32_t k = 0; k < rr.piece_n; k++) {
                   const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
-                  part.piece_pool.push_back({dp.dst_off + pc.src, pc.len, 0});
+                  cb.piece_pool.push_back({dp.dst_off + pc.src, pc.len, 0});
                 }
                 DevBrRes rec{dp.dst_off + rr.dst, rr.len, rr.off, ps,
                              rr.piece_n};
-                if (rr.len <= 256) part.res_lane.push_back(rec);
-                else part.res_wave.push_back(rec);
+                if (rr.len <= 256) cb.res_lane.push_back(rec);
+                else cb.res_wave.push_back(rec);
               }
             }
           }
@@ -812,37 +923,71 @@ extern "C" gpuq_plan* gpuq_plan_build(
         // writes).
         if (c.need_gid && dict_enc) {
           dp.aux = remap_base;
-          part.tasks[{TK_DICT_GID, t.col_idx}].push_back(page_id);
+          cb.tasks[{TK_DICT_GID, t.col_idx}].push_back(this_pid);
         }
         if (c.need_val) {
           if (dict_enc) {
             dp.aux_val = dictv_base;
-            part.tasks[{TK_DICT_VAL, t.col_idx}].push_back(page_id);
+            cb.tasks[{TK_DICT_VAL, t.col_idx}].push_back(this_pid);
           } else if (pi.encoding == ENC_PLAIN) {
-            part.tasks[{TK_PLAIN_VAL, t.col_idx}].push_back(page_id);
+            cb.tasks[{TK_PLAIN_VAL, t.col_idx}].push_back(this_pid);
           } else if (pi.encoding == ENC_DELTA_BP) {
-            part.tasks[{TK_DELTA_VAL, t.col_idx}].push_back(page_id);
+            cb.tasks[{TK_DELTA_VAL, t.col_idx}].push_back(this_pid);
           } else throw std::runtime_error("unsupported encoding for values");
         }
         if (!c.lut_preds.empty()) {
           if (dict_enc) {
             dp.aux_lut = lut_base;
-            part.tasks[{TK_DICT_MASK, t.col_idx}].push_back(page_id);
+            cb.tasks[{TK_DICT_MASK, t.col_idx}].push_back(this_pid);
           } else if (pi.encoding == ENC_PLAIN) {
             // PLAIN fallback page of a string column: only CONTAINS supported
             if ((int)c.contains_preds.size() != (int)c.lut_preds.size())
               throw std::runtime_error("non-LIKE predicate on PLAIN utf8 pages: next row");
-            part.tasks[{TK_BYTES_CONTAINS, t.col_idx}].push_back(page_id);
-            citems.push_back({mf.data + pi.payload_off, pi.comp_size,
-                              pi.uncomp_size, page_id, dp.dst_off,
-                              (uint32_t)pi.num_values, dp.optional != 0,
-                              dp.raw_copy != 0, t.col_idx});
+            cb.tasks[{TK_BYTES_CONTAINS, t.col_idx}].push_back(this_pid);
+            cb.citems.push_back({mf.data + pi.payload_off, pi.comp_size,
+                                 pi.uncomp_size, this_pid, dp.dst_off,
+                                 (uint32_t)pi.num_values, dp.optional != 0,
+                                 dp.raw_copy != 0, t.col_idx});
           } else throw std::runtime_error("unsupported encoding for string predicate");
         }
-        part.pages.push_back(dp);
+        cb.pages.push_back(dp);
       }
       if (row_in_rg != (uint32_t)mf.meta.row_groups[t.rg_idx].num_rows)
         throw std::runtime_error("page rows mismatch");
+    });
+
+    // phase 5 (serial): order-preserving merge of the per-chunk images
+    for (size_t i = 0; i < cbs.size(); i++) {
+      auto& cb = cbs[i];
+      part.pages.insert(part.pages.end(), cb.pages.begin(), cb.pages.end());
+      for (auto& [key, ids] : cb.tasks) {
+        auto& dstv = part.tasks[key];
+        dstv.insert(dstv.end(), ids.begin(), ids.end());
+      }
+      part.segs.insert(part.segs.end(), cb.segs.begin(), cb.segs.end());
+      uint32_t br_base = (uint32_t)part.brs.size();
+      part.brs.insert(part.brs.end(), cb.brs.begin(), cb.brs.end());
+      for (DevPageBr pb : cb.pagebrs) {
+        pb.start += br_base;
+        part.pagebrs.push_back(pb);
+      }
+      uint32_t piece_base = (uint32_t)part.piece_pool.size();
+      part.piece_pool.insert(part.piece_pool.end(), cb.piece_pool.begin(),
+                             cb.piece_pool.end());
+      for (DevBrRes r2 : cb.res_lane) {
+        r2.piece_start += piece_base;
+        part.res_lane.push_back(r2);
+      }
+      for (DevBrRes r2 : cb.res_wave) {
+        r2.piece_start += piece_base;
+        part.res_wave.push_back(r2);
+      }
+      part.brinl.insert(part.brinl.end(), cb.brinl.begin(), cb.brinl.end());
+      part.lits_lane.insert(part.lits_lane.end(), cb.lits_lane.begin(),
+                            cb.lits_lane.end());
+      part.lits_wave.insert(part.lits_wave.end(), cb.lits_wave.begin(),
+                            cb.lits_wave.end());
+      citems.insert(citems.end(), cb.citems.begin(), cb.citems.end());
     }
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
 
@@ -1116,11 +1261,64 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   // last chunk's end (see kernels.hip refill note)
   HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes + 8192, 16)));
   HIP_TRY(hipMalloc(&part.d_dec, std::max<uint64_t>(part.dec_bytes, 16)));
-  for (auto& t : part.chunks) {
-    const auto& mf = *plan->files[t.file_idx];
-    HIP_TRY(hipMemcpyAsync(part.d_raw + t.raw_off, mf.data + t.cm->start_offset(),
-                           t.cm->total_compressed_size, hipMemcpyHostToDevice,
-                           part.stream));
+  {
+    // H2D staging through a ring of pinned buffers, each filled from the
+    // mmap'd files by a parallel host memcpy. Per-chunk hipMemcpyAsync from
+    // pageable mmap pages was fault + staging bound (~1.3 GB/s measured);
+    // this sustains PCIe-class rates on the 1 B-row (tens of GB) shards.
+    // Chunks are packed contiguously into d_raw (raw_off is cumulative), so
+    // the destination cursor just advances by the bytes staged.
+    constexpr size_t BUFSZ = 256ull << 20;
+    const int NBUF = part.raw_bytes > BUFSZ ? 4 : 1;
+    std::vector<void*> bufs(NBUF);
+    std::vector<hipEvent_t> evts(NBUF);
+    for (int b = 0; b < NBUF; b++) {
+      HIP_TRY(hipHostMalloc(&bufs[b], std::min<uint64_t>(BUFSZ, std::max<uint64_t>(part.raw_bytes, 16))));
+      HIP_TRY(hipEventCreate(&evts[b]));
+      HIP_TRY(hipEventRecord(evts[b], part.stream));
+    }
+    struct Span { const uint8_t* src; uint64_t len; };
+    std::vector<Span> spans;
+    spans.reserve(part.chunks.size());
+    for (auto& t : part.chunks) {
+      const auto& mf = *plan->files[t.file_idx];
+      spans.push_back({mf.data + t.cm->start_offset(),
+                       (uint64_t)t.cm->total_compressed_size});
+    }
+    size_t si = 0;
+    uint64_t soff = 0, dst = 0;
+    int b = 0;
+    while (si < spans.size()) {
+      HIP_TRY(hipEventSynchronize(evts[b]));
+      struct Fill { const uint8_t* src; uint8_t* dstp; size_t len; };
+      std::vector<Fill> fills;
+      size_t filled = 0;
+      while (si < spans.size() && filled < BUFSZ) {
+        uint64_t take = std::min<uint64_t>(spans[si].len - soff, BUFSZ - filled);
+        // slice large chunks so the parallel fill balances
+        while (take > 0) {
+          uint64_t piece = std::min<uint64_t>(take, 4ull << 20);
+          fills.push_back({spans[si].src + soff, (uint8_t*)bufs[b] + filled, piece});
+          filled += piece;
+          soff += piece;
+          take -= piece;
+        }
+        if (soff == spans[si].len) { si++; soff = 0; }
+      }
+      parallel_for(fills.size(), [&](size_t f) {
+        memcpy(fills[f].dstp, fills[f].src, fills[f].len);
+      }, 32);
+      HIP_TRY(hipMemcpyAsync(part.d_raw + dst, bufs[b], filled,
+                             hipMemcpyHostToDevice, part.stream));
+      HIP_TRY(hipEventRecord(evts[b], part.stream));
+      dst += filled;
+      b = (b + 1) % NBUF;
+    }
+    HIP_TRY(hipStreamSynchronize(part.stream));
+    for (int i2 = 0; i2 < NBUF; i2++) {
+      HIP_TRY(hipHostFree(bufs[i2]));
+      HIP_TRY(hipEventDestroy(evts[i2]));
+    }
   }
   HIP_TRY(hipMalloc(&part.d_pages, std::max<size_t>(part.pages.size() * sizeof(DevPage), 16)));
   HIP_TRY(hipMemcpyAsync(part.d_pages, part.pages.data(),
