@@ -7,16 +7,17 @@ for N>1 via torch.distributed.run with one rank per GPU (RCCL). A "step" is
 one pass of the hot path (LZ4_RAW decompress -> decode -> filter -> hash
 group-by -> cross-rank RCCL merge) over this rank's shard, with raw column
 chunks already resident in HBM when the timed region starts (plan.load() is
-untimed; the PCIe-inclusive rate is reported as load_gbps and discussed in
-DESIGN.md). Weak scaling: each rank scans its own full-size HBM-resident
-shard; shard CONTENT is replicated across ranks (rank 0 generates once at
-full core count) so the N=8 run does not pay 8x the datagen wall-clock —
-per-GPU work, memory and the RCCL merge are identical to distinct shards.
+untimed; the first-touch cost is reported as config.load_s / plan_build_s
+and the PCIe-inclusive rate as load_gbps_pcie — see DESIGN.md §5). Weak
+scaling: each rank generates and scans its own DISTINCT full-size shard
+(seed = 42 + rank) so cross-rank key spaces diverge like real per-minute
+shards do; the RCCL merge reduces the union key space.
 
-Default workload: BASELINE.json configs[1] ("c1": 100M rows x 8 cols,
-SELECT level,count(*) GROUP BY level) — the quoted single-GPU config.
-`--workload c2s|c3s` run the c2/c3-shaped queries at --rows scale for
-profiling (not the driver's line).
+Default workload: the BASELINE metric's own shape at its stated scale —
+1 B rows, SELECT host,count(*),max(latency) WHERE ts BETWEEN ... GROUP BY
+host (BASELINE.json configs[2], run per-GPU; the BETWEEN selects ~50% of
+row groups). `--workload c1|c3s|c4s` run the other BASELINE shapes at
+--rows scale.
 
 Rank 0 prints ONE JSON line."""
 
@@ -38,11 +39,13 @@ WORKLOADS = {
         "select": [{"agg": "count_star"}],
         "group_by": ["level"],
     }),
-    "c2s": ("c1", lambda a: {   # c2-shaped at reduced scale, single node
+    "c2s": ("c1", lambda a: {   # the BASELINE c2 shape, per-GPU scale
         "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
         "group_by": ["host"],
         "preds": [{"col": "p_timestamp", "op": "between",
-                   "lo": 1756684800000, "hi": 1756684800000 + a.between_ms}],
+                   "lo": 1756684800000, "hi": 1756684800000 + (
+                       a.between_ms or  # default: ~50% of row groups
+                       ((args_n_files(a) + 1) // 2) * 60_000)}],
     }),
     "c3s": ("c3", lambda a: {   # c3-shaped LIKE byte scan
         "select": [{"agg": "count_star"}],
@@ -57,6 +60,10 @@ WORKLOADS = {
 }
 
 
+def args_n_files(a):
+    return (a.rows + 262_143) // 262_144
+
+
 def log(msg):
     if int(os.environ.get("RANK", "0")) == 0:
         print(msg, file=sys.stderr, flush=True)
@@ -67,14 +74,18 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--workload", default="c1")
-    ap.add_argument("--rows", type=int, default=100_000_000,
-                    help="rows per GPU (weak scaling)")
-    ap.add_argument("--between-ms", type=int, default=150 * 60_000,
-                    help="c2s: BETWEEN window width (~50%% of range)")
+    ap.add_argument("--workload", default="c2s")
+    ap.add_argument("--rows", type=int, default=1_000_000_000,
+                    help="rows per GPU (weak scaling); BASELINE quotes 1B")
+    ap.add_argument("--between-ms", type=int, default=0,
+                    help="c2s: BETWEEN window width (default ~50%% of range)")
     ap.add_argument("--data-dir", default=os.environ.get("GPUQ_DATA", "/tmp/gpuq_bench"))
     ap.add_argument("--gen-workers", type=int, default=0)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--cpu-baseline-full", action="store_true",
+                    help="time the oracle over the FULL stream (all files, "
+                    "all cores) instead of the bounded sample — the "
+                    "once-per-round measured baseline (BASELINE.md)")
     args = ap.parse_args()
 
     import torch
@@ -101,28 +112,31 @@ def main():
     cfg, qbuild = WORKLOADS[args.workload]
     query = qbuild(args)
 
-    # --- shard (rank 0 generates once; content replicated across ranks) ---
-    n_files = (args.rows + 262_143) // 262_144
-    shard = os.path.join(args.data_dir, f"{args.workload}_{args.rows}")
+    # --- shard (DISTINCT content per rank: seed = 42 + rank) ---
+    n_files = args_n_files(args)
+    shard = os.path.join(args.data_dir, f"{args.workload}_{args.rows}_r{rank}")
     stream_dir = os.path.join(shard, "stream")
-    workers = args.gen_workers or (os.cpu_count() or 8)
-    if rank == 0 and not os.path.exists(os.path.join(stream_dir, "stream.json")):
-        log(f"[bench] generating {args.rows} rows ({n_files} files, "
-            f"{workers} workers) under {shard} ...")
+    workers = args.gen_workers or max(8, (os.cpu_count() or 8) // world)
+    if not os.path.exists(os.path.join(stream_dir, "stream.json")):
+        log(f"[bench] generating {args.rows} rows/rank ({n_files} files, "
+            f"{workers} workers/rank) under {shard} ...")
         t0 = time.time()
-        gen_stream(shard, "stream", cfg, rows=args.rows, seed=42,
+        gen_stream(shard, "stream", cfg, rows=args.rows, seed=42 + rank,
                    workers=workers)
         log(f"[bench] datagen took {time.time() - t0:.1f}s")
     if distributed:
         dist.barrier()
 
-    # --- plan build + HBM residency (untimed) ---
+    # --- plan build + HBM residency (untimed; first-touch cost reported) ---
     session = GpuSession(device_mask=1 << local_rank)
     provider = StandardTableProvider(stream_dir, session)
+    t0 = time.time()
     plan = provider.scan(query)
+    plan_s = time.time() - t0
     t0 = time.time()
     plan.load()
-    log(f"[bench] plan built + loaded to HBM in {time.time() - t0:.1f}s")
+    load_s = time.time() - t0
+    log(f"[bench] plan built in {plan_s:.1f}s, loaded to HBM in {load_s:.1f}s")
 
     merger = DistMerger(query, device=merge_dev)
     first = plan.execute(0)
@@ -230,11 +244,18 @@ def main():
         from oracle import query_oracle as qo
         import pyarrow as pa
 
-        sample_files = sorted(
+        all_files = sorted(
             os.path.join(dp, f)
             for dp, _, fs in os.walk(stream_dir)
             for f in fs if f.endswith(".parquet")
-        )[: max(2, min(12, n_files // 8))]
+        )
+        # bounded sample (~10-30s of CPU work) by default; --cpu-baseline-full
+        # runs the whole stream once per round (the measured, not
+        # extrapolated, figure for BASELINE.md)
+        if args.cpu_baseline_full:
+            sample_files = all_files
+        else:
+            sample_files = all_files[: max(2, min(256, n_files // 8))]
         import pyarrow.parquet as pq
 
         sample_rows = sum(pq.read_metadata(f).num_rows for f in sample_files)
@@ -246,7 +267,8 @@ def main():
             "unit": "rows/s",
             "cores": pa.cpu_count(),
             "kind": "port",
-            "sample": f"oracle (pyarrow-decode + numpy agg) on {len(sample_files)} "
+            "sample": ("FULL stream: " if args.cpu_baseline_full else "") +
+                      f"oracle (pyarrow-decode + numpy agg) on {len(sample_files)} "
                       f"files = {sample_rows} rows, {tcb:.1f}s",
         }
 
@@ -274,12 +296,14 @@ def main():
                 }[args.workload],
                 "rows_per_gpu": args.rows,
                 "files_per_gpu": n_files,
-                "rank_shards": "replicated content (each rank scans its own HBM copy)",
+                "rank_shards": "distinct content per rank (seed = 42 + rank)",
                 "parallelism": f"dp{world}",
                 "gb_per_sec_scanned": round(gbps_scanned, 2),
                 "gb_per_sec_rowgroup_bytes": round(gbps_rowgroup, 2),
                 "kernel_ms_per_step": round(dk / 1e6, 3),
                 "decomp_ms_per_step": round(dd / 1e6, 3),
+                "plan_build_s": round(plan_s, 2),
+                "load_s": round(load_s, 2),
                 "load_gbps_pcie": round(
                     rank_bytes / max(m_after["load_ns"], 1), 2),
             },
