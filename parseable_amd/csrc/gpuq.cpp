@@ -210,6 +210,27 @@ struct Partition {
   int64_t load_ns = 0;
 };
 
+// H2D staging ring: pinned buffers filled from host memory (mmap'd files /
+// pool vectors) by a parallel memcpy, async-copied to the device.
+// hipMemcpyAsync straight from pageable mmap pages was fault + staging
+// bound (~1.3 GB/s measured); the 1 B-row shards move tens of GB of raw
+// chunks plus multi-GB aux pools (literal records, dictionary values), so
+// every bulk upload in gpuq_plan_load goes through this ring at PCIe-class
+// rates.
+struct Ring {
+  static constexpr size_t BUFSZ = 256ull << 20;
+  static constexpr int NBUF = 4;
+  void* bufs[NBUF] = {};
+  hipEvent_t evts[NBUF] = {};
+  hipStream_t stream;
+  int b = 0;
+  explicit Ring(hipStream_t st);
+  ~Ring();
+  struct Span { const uint8_t* src; uint64_t len; };
+  void copy_spans(uint8_t* d_dst, const std::vector<Span>& spans);
+  void copy(void* d_dst, const void* src, size_t len);
+};
+
 }  // namespace
 
 struct gpuq_plan {
@@ -365,6 +386,56 @@ void parallel_for(size_t n, F&& fn, size_t max_threads = 0) {
   for (size_t t = 0; t < nthreads; t++) pool.emplace_back(worker);
   for (auto& th : pool) th.join();
   if (eptr) std::rethrow_exception(eptr);
+}
+
+Ring::Ring(hipStream_t st) : stream(st) {
+  for (int i = 0; i < NBUF; i++) {
+    HIP_TRY(hipHostMalloc(&bufs[i], BUFSZ));
+    HIP_TRY(hipEventCreate(&evts[i]));
+    HIP_TRY(hipEventRecord(evts[i], stream));
+  }
+}
+Ring::~Ring() {
+  for (int i = 0; i < NBUF; i++) {
+    if (bufs[i]) (void)hipHostFree(bufs[i]);
+    if (evts[i]) (void)hipEventDestroy(evts[i]);
+  }
+}
+// copy the concatenation of spans to d_dst (device), pipelined: fill of
+// buffer b overlaps the in-flight H2D copies of the other ring slots
+void Ring::copy_spans(uint8_t* d_dst, const std::vector<Span>& spans) {
+  size_t si = 0;
+  uint64_t soff = 0, dst = 0;
+  while (si < spans.size()) {
+    if (spans[si].len == 0) { si++; continue; }
+    HIP_TRY(hipEventSynchronize(evts[b]));
+    struct Fill { const uint8_t* src; uint8_t* dstp; size_t len; };
+    std::vector<Fill> fills;
+    size_t filled = 0;
+    while (si < spans.size() && filled < BUFSZ) {
+      uint64_t take = std::min<uint64_t>(spans[si].len - soff, BUFSZ - filled);
+      while (take > 0) {  // slice so the parallel fill balances
+        uint64_t piece = std::min<uint64_t>(take, 4ull << 20);
+        fills.push_back({spans[si].src + soff, (uint8_t*)bufs[b] + filled,
+                         piece});
+        filled += piece;
+        soff += piece;
+        take -= piece;
+      }
+      if (soff == spans[si].len) { si++; soff = 0; }
+    }
+    parallel_for(fills.size(), [&](size_t f) {
+      memcpy(fills[f].dstp, fills[f].src, fills[f].len);
+    }, 32);
+    HIP_TRY(hipMemcpyAsync(d_dst + dst, bufs[b], filled,
+                           hipMemcpyHostToDevice, stream));
+    HIP_TRY(hipEventRecord(evts[b], stream));
+    dst += filled;
+    b = (b + 1) % NBUF;
+  }
+}
+void Ring::copy(void* d_dst, const void* src, size_t len) {
+  if (len) copy_spans((uint8_t*)d_dst, {{(const uint8_t*)src, len}});
 }
 
 }  // namespace
@@ -1197,6 +1268,20 @@ extern "C" gpuq_plan* gpuq_plan_build(
                         !plan->cols[plan->group_cols[0]].is_bin;
   }
 
+  if (getenv("GPUQ_PLAN_DEBUG")) {
+    for (size_t p = 0; p < plan->parts.size(); p++) {
+      const auto& pt = plan->parts[p];
+      fprintf(stderr,
+              "[gpuq plan] part %zu: rows=%lld chunks=%zu pages=%zu segs=%zu "
+              "lits=%zu/%zu brinl=%zu res=%zu/%zu pieces=%zu raw=%.2fGB "
+              "dec=%.2fGB cwins=%zu\n",
+              p, (long long)pt.n_rows, pt.chunks.size(), pt.pages.size(),
+              pt.segs.size(), pt.lits_lane.size(), pt.lits_wave.size(),
+              pt.brinl.size(), pt.res_lane.size(), pt.res_wave.size(),
+              pt.piece_pool.size(), pt.raw_bytes / 1e9, pt.dec_bytes / 1e9,
+              pt.cwins.size());
+    }
+  }
   return plan.release();
 } catch (const std::exception& e) {
   if (ctx) ctx->set_error(e.what());
@@ -1261,64 +1346,19 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   // last chunk's end (see kernels.hip refill note)
   HIP_TRY(hipMalloc(&part.d_raw, std::max<uint64_t>(part.raw_bytes + 8192, 16)));
   HIP_TRY(hipMalloc(&part.d_dec, std::max<uint64_t>(part.dec_bytes, 16)));
+  // H2D staging through a ring of pinned buffers (Ring, above): raw chunks
+  // are packed contiguously into d_raw (raw_off cumulative), and every
+  // multi-GB aux pool below rides the same ring.
+  Ring ring(part.stream);
   {
-    // H2D staging through a ring of pinned buffers, each filled from the
-    // mmap'd files by a parallel host memcpy. Per-chunk hipMemcpyAsync from
-    // pageable mmap pages was fault + staging bound (~1.3 GB/s measured);
-    // this sustains PCIe-class rates on the 1 B-row (tens of GB) shards.
-    // Chunks are packed contiguously into d_raw (raw_off is cumulative), so
-    // the destination cursor just advances by the bytes staged.
-    constexpr size_t BUFSZ = 256ull << 20;
-    const int NBUF = part.raw_bytes > BUFSZ ? 4 : 1;
-    std::vector<void*> bufs(NBUF);
-    std::vector<hipEvent_t> evts(NBUF);
-    for (int b = 0; b < NBUF; b++) {
-      HIP_TRY(hipHostMalloc(&bufs[b], std::min<uint64_t>(BUFSZ, std::max<uint64_t>(part.raw_bytes, 16))));
-      HIP_TRY(hipEventCreate(&evts[b]));
-      HIP_TRY(hipEventRecord(evts[b], part.stream));
-    }
-    struct Span { const uint8_t* src; uint64_t len; };
-    std::vector<Span> spans;
+    std::vector<Ring::Span> spans;
     spans.reserve(part.chunks.size());
     for (auto& t : part.chunks) {
       const auto& mf = *plan->files[t.file_idx];
       spans.push_back({mf.data + t.cm->start_offset(),
                        (uint64_t)t.cm->total_compressed_size});
     }
-    size_t si = 0;
-    uint64_t soff = 0, dst = 0;
-    int b = 0;
-    while (si < spans.size()) {
-      HIP_TRY(hipEventSynchronize(evts[b]));
-      struct Fill { const uint8_t* src; uint8_t* dstp; size_t len; };
-      std::vector<Fill> fills;
-      size_t filled = 0;
-      while (si < spans.size() && filled < BUFSZ) {
-        uint64_t take = std::min<uint64_t>(spans[si].len - soff, BUFSZ - filled);
-        // slice large chunks so the parallel fill balances
-        while (take > 0) {
-          uint64_t piece = std::min<uint64_t>(take, 4ull << 20);
-          fills.push_back({spans[si].src + soff, (uint8_t*)bufs[b] + filled, piece});
-          filled += piece;
-          soff += piece;
-          take -= piece;
-        }
-        if (soff == spans[si].len) { si++; soff = 0; }
-      }
-      parallel_for(fills.size(), [&](size_t f) {
-        memcpy(fills[f].dstp, fills[f].src, fills[f].len);
-      }, 32);
-      HIP_TRY(hipMemcpyAsync(part.d_raw + dst, bufs[b], filled,
-                             hipMemcpyHostToDevice, part.stream));
-      HIP_TRY(hipEventRecord(evts[b], part.stream));
-      dst += filled;
-      b = (b + 1) % NBUF;
-    }
-    HIP_TRY(hipStreamSynchronize(part.stream));
-    for (int i2 = 0; i2 < NBUF; i2++) {
-      HIP_TRY(hipHostFree(bufs[i2]));
-      HIP_TRY(hipEventDestroy(evts[i2]));
-    }
+    ring.copy_spans(part.d_raw, spans);
   }
   HIP_TRY(hipMalloc(&part.d_pages, std::max<size_t>(part.pages.size() * sizeof(DevPage), 16)));
   HIP_TRY(hipMemcpyAsync(part.d_pages, part.pages.data(),
@@ -1326,8 +1366,7 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
                          part.stream));
   auto upload_pool = [&](const void* src, size_t bytes, void** dst) {
     HIP_TRY(hipMalloc(dst, std::max<size_t>(bytes, 16)));
-    if (bytes)
-      HIP_TRY(hipMemcpyAsync(*dst, src, bytes, hipMemcpyHostToDevice, part.stream));
+    if (bytes) ring.copy(*dst, src, bytes);
   };
   upload_pool(part.remap_pool.data(), part.remap_pool.size() * 4, (void**)&part.d_remap);
   upload_pool(part.dictv_pool.data(), part.dictv_pool.size() * 8, (void**)&part.d_dictv);
