@@ -196,14 +196,16 @@ def main():
     # DESIGN.md §3/§5):
     #  - lz4 sweep: compressed read + decompressed write
     #  - c1 fused dict-count: decompressed pages read once (no materialization)
-    #  - c2s: dec read + ts/latency i64 writes+reads (8B x2 x2 cols) + gid
-    #    (4B x2) + valid (1x3) + mask (1x3 rw)
+    #  - c2s: dec read + latency val write+read (8B x2) + valid w+r (1x2) +
+    #    gid w+r (4B x2) + mask read (1); the ts BETWEEN is chunk-stats
+    #    elided for fully-covered row groups (pred_all_true), so ts decode
+    #    and cmp touch only boundary row groups (~0 B/row at 1B scale)
     #  - c3s: dec read (values scanned from LDS windows) + mask write
     decomp_algo_bytes = raw_b + dec_b
     other_ns = max(dk - dd, 1.0)
     decode_algo_bytes = {
         "c1": dec_b,
-        "c2s": dec_b + rank_rows * (2 * 16 + 2 * 4 + 3 + 3),
+        "c2s": dec_b + rank_rows * (2 * 8 + 2 * 1 + 2 * 4 + 1),
         "c3s": dec_b + rank_rows * 1,
         "c4s": dec_b + rank_rows * (3 * 16 + 3 * 4 + 4 + 3),
     }[args.workload]

@@ -27,6 +27,7 @@
 #include <string.h>
 #include <stdint.h>
 #include <inttypes.h>
+#include <math.h>
 
 static void die(const char *msg) { fprintf(stderr, "cpu_ref: %s\n", msg); exit(1); }
 
@@ -593,6 +594,7 @@ typedef struct {
     int64_t cnt[16];             /* per-agg count of accumulated values */
     int64_t i64v[16];
     double f64v[16];
+    double f64c[16];      /* Neumaier compensation for f64 sums */
 } Group;
 
 typedef struct { Group *slots; uint32_t cap, n; } HashTab;
@@ -752,8 +754,17 @@ static void run_query(Query *q) {
                     if (c->f64) {
                         g_agg_is_f64[i] = 1;
                         double v = c->f64[r];
-                        if (!gr->cnt[i]) { gr->f64v[i] = v; }
-                        else if (a->op == AGG_SUM) gr->f64v[i] += v;
+                        if (!gr->cnt[i]) { gr->f64v[i] = v; gr->f64c[i] = 0.0; }
+                        else if (a->op == AGG_SUM) {
+                            /* Neumaier compensated sum (order-dependent but
+                               ~exact; final gate vs the oracle is rtol) */
+                            double s2 = gr->f64v[i] + v;
+                            if (fabs(gr->f64v[i]) >= fabs(v))
+                                gr->f64c[i] += (gr->f64v[i] - s2) + v;
+                            else
+                                gr->f64c[i] += (v - s2) + gr->f64v[i];
+                            gr->f64v[i] = s2;
+                        }
                         else if (a->op == AGG_MIN) { if (v < gr->f64v[i]) gr->f64v[i] = v; }
                         else if (a->op == AGG_MAX) { if (v > gr->f64v[i]) gr->f64v[i] = v; }
                         gr->cnt[i]++;
@@ -823,7 +834,9 @@ static void run_query(Query *q) {
             Agg *a = &q->aggs[i];
             if (a->op == AGG_COUNT_STAR || a->op == AGG_COUNT) printf("%" PRId64, gr->cnt[i]);
             else if (!gr->cnt[i]) printf("\\N");
-            else if (g_agg_is_f64[i]) printf("%.17g", gr->f64v[i]);
+            else if (g_agg_is_f64[i])
+                printf("%.17g", q->aggs[i].op == AGG_SUM
+                                    ? gr->f64v[i] + gr->f64c[i] : gr->f64v[i]);
             else printf("%" PRId64, gr->i64v[i]);
         }
         putchar('\n');

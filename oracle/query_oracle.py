@@ -247,7 +247,18 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
             gi = inv[vsel]
             if op == "sum":
                 is_f = np.issubdtype(fv.dtype, np.floating)
-                out = np.zeros(G, dtype=np.float64 if is_f else np.int64)
+                if is_f:
+                    # f64 sums: keep the per-group VALUES so the final sum is
+                    # math.fsum — correctly rounded, matching the GPU's exact
+                    # 256-bit superaccumulator within 1 ULP (BASELINE gate)
+                    order = np.argsort(gi, kind="stable")
+                    sgi, sfv = gi[order], fv[order]
+                    bounds = np.searchsorted(sgi, np.arange(G + 1))
+                    parts = [sfv[bounds[g]:bounds[g + 1]] for g in range(G)]
+                    cnt = np.bincount(gi, minlength=G)
+                    file_res.append(("fsum", parts, cnt))
+                    continue
+                out = np.zeros(G, dtype=np.int64)
                 np.add.at(out, gi, fv)
                 cnt = np.bincount(gi, minlength=G)
                 file_res.append(("sum", out, cnt))
@@ -277,6 +288,12 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
                 if kind == "count":
                     v = int(r[1][gidx])
                     st[ai] = v if st[ai] is None else st[ai] + v
+                elif kind == "fsum":
+                    part = r[1][gidx]
+                    if len(part):
+                        if st[ai] is None:
+                            st[ai] = []
+                        st[ai].append(part)
                 else:
                     v = r[1][gidx]
                     present = int(r[2][gidx]) > 0
@@ -300,6 +317,10 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
         for a, s in zip(aggs, st):
             if s is None and a["agg"] in ("count", "count_star"):
                 s = 0
+            elif isinstance(s, list):  # f64 sum: exact, rounded once
+                import math
+
+                s = math.fsum(np.concatenate(s))
             row.append(s)
         rows.append(row)
     rows.sort(key=lambda r: tuple(_sort_key(v) for v in r[: len(group_by)]))

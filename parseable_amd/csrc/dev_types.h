@@ -77,6 +77,13 @@ constexpr int MAX_AGGS = 8;
 // Aggregate table layout: G groups x (1 + 2*n_aggs) u64 slots:
 //   slot 0: presence (selected-row count)
 //   per agg a: slot 1+2a = value (i64 / f64 bits), slot 2+2a = non-null count
+//
+// f64 SUM is exact and order-independent: each value is decomposed into a
+// 256-bit fixed-point two's-complement contribution (lsb weight 2^-160) and
+// accumulated limb-wise with carry-propagating u64 atomics into `fsum`
+// ([fsum_n][n_groups][4] limbs); the export rounds the exact 256-bit total
+// to the nearest double once. Covers |x| in [2^-160, 2^86) — far beyond log
+// analytics ranges; out-of-window values set *err = ERR_FSUM_RANGE (loud).
 struct AggArgs {
   const uint8_t* mask;          // selection mask (1 byte/row), may be null (=all)
   int64_t n_rows;               // partition rows
@@ -87,12 +94,18 @@ struct AggArgs {
   int32_t agg_kind[MAX_AGGS];        // AGGK_*
   const int64_t* agg_val[MAX_AGGS];  // i64 array or f64 bits (same width)
   const uint8_t* agg_valid[MAX_AGGS];// may be null (= all valid)
+  uint8_t cnt_skip[MAX_AGGS];        // footer null_count==0 for every chunk:
+                                     // count == presence, skip count atomics
+  int32_t fsum_idx[MAX_AGGS];        // agg -> fsum table index (-1 = none)
+  int32_t fsum_n;                    // number of f64-sum aggregates
+  uint64_t* fsum;                    // superaccumulators (zeroed before launch)
   uint64_t* table;
   int32_t n_groups;
+  int32_t* err;
 };
 
 // error codes written to *d_error by kernels
 enum { ERR_NONE = 0, ERR_LZ4 = 1, ERR_RLE = 2, ERR_DELTA = 3, ERR_DICT_RANGE = 4,
-       ERR_PAGE = 5 };
+       ERR_PAGE = 5, ERR_FSUM_RANGE = 6 };
 
 }  // namespace gpuq

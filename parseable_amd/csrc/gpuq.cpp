@@ -75,6 +75,10 @@ struct ColPlan {
   // required representations
   bool need_gid = false;    // group key (utf8 dict) or COUNT(utf8 col)
   bool need_val = false;    // i64/f64 row-aligned values + valid
+  bool val_always = false;  // values feed aggs/bins/projection — decode every
+                            // chunk (false = predicate-only: chunks whose
+                            // footer stats prove the predicate for all rows
+                            // skip decode entirely)
   bool need_gid_valid = false;  // validity bytes alongside gid (COUNT(utf8))
   // predicate routing
   std::vector<int> lut_preds;      // preds evaluated per-dict-entry (utf8)
@@ -110,6 +114,9 @@ struct AggPlan {
   int col_idx = -1;   // into plan.cols
   int kind = 0;       // AGGK_*
   bool is_f64 = false;
+  bool cnt_is_presence = false;  // every chunk's footer null_count == 0:
+                                 // the non-null count IS the presence count
+  int fsum_idx = -1;             // AGGK_SUM_F64 -> superaccumulator table idx
 };
 
 // one column chunk of one selected row group
@@ -163,6 +170,9 @@ struct Partition {
   std::map<int, std::pair<uint32_t, uint32_t>> cwin_ranges;  // col -> (off, n)
   uint64_t raw_bytes = 0, dec_bytes = 0;
   int64_t bytes_scanned = 0, rowgroup_bytes_total = 0;
+  // per-pred row ranges still needing per-row evaluation (chunk-stats
+  // elision, pred_all_true): merged-adjacent [start, start+len) pairs
+  std::map<int, std::vector<std::pair<int64_t, int64_t>>> pred_ranges;
 
   // device state
   bool loaded = false;
@@ -175,6 +185,7 @@ struct Partition {
   uint8_t* d_mask = nullptr;
   int32_t* d_err = nullptr;
   uint64_t* d_table = nullptr;
+  uint64_t* d_fsum = nullptr;    // [n_fsum][n_groups][4] superacc limbs
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;        // concatenated CONTAINS needles
   std::vector<uint32_t> needle_off;   // per plan-pred offset into the pool
@@ -246,6 +257,7 @@ struct gpuq_plan {
   int64_t limit = -1;
   std::vector<Partition> parts;
   int32_t n_groups = 0;          // product of key sizes (incl null slots)
+  int32_t n_fsum = 0;            // number of exact-f64-sum side tables
   bool fused_count = false;      // single dict key + count(*)-only + no preds
   std::mutex mu;
   // metrics
@@ -320,6 +332,33 @@ bool rg_pruned_by_stats(const gpuq_plan& plan, const FileMeta& fm,
     if (!can_match) return true;
   }
   return false;
+}
+
+// Chunk-stats predicate elision: TRUE when the footer stats prove EVERY row
+// of the chunk satisfies the predicate — the exactness the reference's
+// engine gets from its minute-aligned pruning predicates
+// (build_parquet_scan_components, stream_schema_provider.rs:127-144:
+// files fully inside the window are scanned with the Exact filter already
+// satisfied, so DataFusion never evaluates it per row). Such chunks keep
+// their memset-1 selection mask and, when the column is predicate-only,
+// are never decoded at all. Requires null_count == 0: a NULL row must not
+// satisfy any predicate.
+bool pred_all_true(const gpuq_pred& p, const ColumnChunkMeta& cm) {
+  if (p.lit_kind != GPUQ_LIT_I64 || !cm.has_i64_stats) return false;
+  if (cm.null_count != 0) return false;
+  int64_t mn = cm.stat_min, mx = cm.stat_max;
+  switch (p.op) {
+    case GPUQ_EQ: return mn == p.i64[0] && mx == p.i64[0];
+    case GPUQ_NE: return p.i64[0] < mn || p.i64[0] > mx;
+    case GPUQ_LT: return mx < p.i64[0];
+    case GPUQ_LE: return mx <= p.i64[0];
+    case GPUQ_GT: return mn > p.i64[0];
+    case GPUQ_GE: return mn >= p.i64[0];
+    case GPUQ_BETWEEN:
+      return mn >= p.i64[0] &&
+             (p.hi_exclusive ? mx < p.i64[1] : mx <= p.i64[1]);
+    default: return false;
+  }
 }
 
 // evaluate a string predicate against one dict entry (host; LUT build)
@@ -495,6 +534,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       if (bc.bin_stride <= 0) throw std::runtime_error("bad bin stride");
       int si = find_or_add_col(plan->cols, src);
       plan->cols[si].need_val = true;
+      plan->cols[si].val_always = true;
       plan->cols[ci].bin_src = si;
       plan->group_cols.push_back(ci);
       continue;
@@ -564,14 +604,19 @@ extern "C" gpuq_plan* gpuq_plan_build(
         c.need_gid_valid = true;
       } else {                              // utf8 min/max via dict sort-ranks
         c.need_val = true;
+        c.val_always = true;
         c.need_rank = true;
       }
     } else {
       c.need_val = true;
+      c.val_always = true;
     }
     switch (ap.op) {
       case GPUQ_AGG_COUNT: ap.kind = AGGK_COUNT; break;
-      case GPUQ_AGG_SUM: ap.kind = ap.is_f64 ? AGGK_SUM_F64 : AGGK_SUM_I64; break;
+      case GPUQ_AGG_SUM:
+        ap.kind = ap.is_f64 ? AGGK_SUM_F64 : AGGK_SUM_I64;
+        if (ap.is_f64) ap.fsum_idx = plan->n_fsum++;
+        break;
       case GPUQ_AGG_MIN:
         ap.kind = (c.phys == PT_BYTE_ARRAY) ? AGGK_MIN_RANK
                   : ap.is_f64 ? AGGK_MIN_F64 : AGGK_MIN_I64;
@@ -605,7 +650,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
         c.optional = fm0.columns[si].optional;
       }
       if (c.phys == PT_BYTE_ARRAY) c.need_gid = true;     // export via dict
-      else c.need_val = true;                             // i64/f64 direct
+      else { c.need_val = true; c.val_always = true; }    // i64/f64 direct
     }
     int tci = find_or_add_col(plan->cols, "p_timestamp");
     {
@@ -617,6 +662,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
         tc.optional = fm0.columns[si].optional;
       }
       tc.need_val = true;
+      tc.val_always = true;
     }
     plan->ts_col = tci;
   }
@@ -678,25 +724,48 @@ extern "C" gpuq_plan* gpuq_plan_build(
     }
     part.n_rows = row_cursor;
 
-    // chunk stubs (serial, cheap): raw-arena offsets + byte accounting
+    // chunk stubs (serial, cheap): raw-arena offsets + byte accounting +
+    // chunk-stats predicate elision (pred_all_true): a predicate proven for
+    // every row of a row group keeps the memset-1 mask there — no per-row
+    // evaluation, and predicate-only columns skip decode for that chunk.
     struct Stub {
       int file_idx, rg_idx, col_idx;
       const ColumnChunkMeta* cm;
       uint64_t raw_off;
       uint32_t rstart;
+      bool need_val_decode;
     };
     std::vector<Stub> stubs;
     for (auto& r : part.rgs) {
       const auto& mf = *plan->files[r.file_idx];
       const auto& rg = mf.meta.row_groups[r.rg_idx];
+      // which predicates still need per-row evaluation in this row group
+      std::vector<bool> pred_eval(plan->preds.size(), false);
+      for (size_t pi2 = 0; pi2 < plan->preds.size(); pi2++) {
+        const auto& pp = plan->preds[pi2];
+        int si = mf.meta.col_index(pp.col);
+        bool elided = false;
+        if (si >= 0) elided = pred_all_true(pp.p, rg.chunks[si]);
+        pred_eval[pi2] = !elided;
+        if (!elided) {
+          auto& rv = part.pred_ranges[(int)pi2];
+          int64_t lo2 = r.row_start, n2 = r.rows;
+          if (!rv.empty() && rv.back().first + rv.back().second == lo2)
+            rv.back().second += n2;  // merge adjacent row groups
+          else
+            rv.emplace_back(lo2, n2);
+        }
+      }
       for (size_t ci = 0; ci < plan->cols.size(); ci++) {
         auto& c = plan->cols[ci];
         if (c.is_bin) continue;
         int si = mf.meta.col_index(c.name);
         if (si < 0) throw std::runtime_error("column missing in file: " + c.name);
         const auto& cm = rg.chunks[si];
+        bool needs_val = c.val_always;
+        for (int pidx : c.cmp_preds) needs_val |= pred_eval[pidx];
         stubs.push_back({r.file_idx, r.rg_idx, (int)ci, &cm, part.raw_bytes,
-                         r.row_start});
+                         r.row_start, needs_val});
         part.raw_bytes += cm.total_compressed_size;
         part.bytes_scanned += cm.total_compressed_size;
       }
@@ -996,7 +1065,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
           dp.aux = remap_base;
           cb.tasks[{TK_DICT_GID, t.col_idx}].push_back(this_pid);
         }
-        if (c.need_val) {
+        if (c.need_val && stubs[ti].need_val_decode) {
           if (dict_enc) {
             dp.aux_val = dictv_base;
             cb.tasks[{TK_DICT_VAL, t.col_idx}].push_back(this_pid);
@@ -1234,6 +1303,20 @@ extern "C" gpuq_plan* gpuq_plan_build(
   for (auto& part : plan->parts)
     plan->m_rows_scanned += part.n_rows;
 
+  // count-slot elision (footer stats): when every chunk of an aggregate's
+  // column has null_count == 0, its non-null count equals the presence
+  // count — k_agg skips the per-row count atomics and the validity read,
+  // and the export reads slot 0 instead (the same trick the reference's
+  // engine gets from its statistics-based null handling).
+  for (auto& ap : plan->aggs) {
+    if (ap.op == GPUQ_AGG_COUNT_STAR || ap.col_idx < 0) continue;
+    bool all0 = true;
+    for (auto& part : plan->parts)
+      for (auto& t : part.chunks)
+        if (t.col_idx == ap.col_idx && t.cm->null_count != 0) all0 = false;
+    ap.cnt_is_presence = all0;
+  }
+
   // utf8 min/max post-pass: with every chunk registered, the global dict is
   // final — compute each rank column's lexicographic sort ranks and rewrite
   // its per-chunk dictionary-value pools from gid to rank.
@@ -1420,6 +1503,8 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   HIP_TRY(hipMalloc(&part.d_err, 4));
   size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size()) * 8;
   HIP_TRY(hipMalloc(&part.d_table, std::max<size_t>(tsz, 16)));
+  size_t fsz = (size_t)plan->n_fsum * plan->n_groups * 4 * 8;
+  HIP_TRY(hipMalloc(&part.d_fsum, std::max<size_t>(fsz, 16)));
   std::vector<int32_t> kinds;
   for (auto& a : plan->aggs) kinds.push_back(a.kind);
   upload_pool(kinds.data(), kinds.size() * 4, (void**)&part.d_agg_kind);
@@ -1563,6 +1648,45 @@ void ss_release(struct ArrowArrayStream* st) {
     delete s;
   }
   st->release = nullptr;
+}
+
+// Round the exact 256-bit two's-complement fixed-point sum (lsb weight
+// 2^-160; kernels.hip acc256_*) to the nearest double, ties to even —
+// the once-per-query rounding that makes f64 SUM order-independent and
+// within 1 ULP of the correctly rounded exact sum (BASELINE parity gate).
+double acc256_to_double(const uint64_t li[4]) {
+  uint64_t l[4] = {li[0], li[1], li[2], li[3]};
+  bool neg = (l[3] >> 63) != 0;
+  if (neg) {
+    uint64_t c = 1;
+    for (int i = 0; i < 4; i++) { l[i] = ~l[i] + c; c = (c && l[i] == 0) ? 1 : 0; }
+  }
+  int t = -1;
+  for (int i = 3; i >= 0; i--)
+    if (l[i]) { t = i * 64 + 63 - __builtin_clzll(l[i]); break; }
+  if (t < 0) return 0.0;
+  double d;
+  if (t <= 52) {  // every bit fits the mantissa: exact
+    d = ldexp((double)l[0], -160);
+  } else {
+    int sh = t - 52;  // keep bits [sh, t]; bits below are round/sticky
+    auto word = [&](int i) -> uint64_t { return (i >= 0 && i < 4) ? l[i] : 0; };
+    int wi = sh >> 6, sb = sh & 63;
+    uint64_t mant = sb ? ((word(wi) >> sb) | (word(wi + 1) << (64 - sb)))
+                       : word(wi);
+    int rb = (int)((word((sh - 1) >> 6) >> ((sh - 1) & 63)) & 1);
+    bool sticky = false;
+    for (int i = 0; i < 4 && !sticky; i++) {
+      int lo = i * 64;
+      int top = sh - 2;  // highest sticky bit
+      if (top < lo) break;
+      uint64_t m = (top - lo >= 63) ? ~0ull : ((1ull << (top - lo + 1)) - 1);
+      if (l[i] & m) sticky = true;
+    }
+    if (rb && (sticky || (mant & 1))) mant++;
+    d = ldexp((double)mant, sh - 160);
+  }
+  return neg ? -d : d;
 }
 
 int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
@@ -1749,6 +1873,9 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   if (need_mask) HIP_TRY(hipMemsetAsync(part.d_mask, 1, part.n_rows, st));
   launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
                     part.d_agg_kind);
+  if (plan->n_fsum)
+    HIP_TRY(hipMemsetAsync(part.d_fsum, 0,
+                           (size_t)plan->n_fsum * plan->n_groups * 4 * 8, st));
 
   HIP_TRY(hipEventRecord(ev0, st));
   // 1. decompress: parallel segments, then ordered backref resolution
@@ -1886,9 +2013,15 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         memcpy(&lo, &pp.p.f64[0], 8);
         memcpy(&hi, &pp.p.f64[1], 8);
       }
-      launch_cmp_i64(st, part.d_val[(int)ci], part.d_valid[(int)ci],
-                     lo, hi, mode, pp.p.hi_exclusive, is_f64,
-                     part.d_mask, part.n_rows);
+      // evaluate only over the row ranges the chunk stats could not prove
+      // (pred_all_true); proven rows keep their memset-1 mask
+      auto rit = part.pred_ranges.find(pidx);
+      if (rit == part.pred_ranges.end()) continue;
+      for (const auto& [row0, nrows] : rit->second)
+        launch_cmp_i64(st, part.d_val[(int)ci] + row0,
+                       part.d_valid[(int)ci] + row0,
+                       lo, hi, mode, pp.p.hi_exclusive, is_f64,
+                       part.d_mask + row0, nrows);
     }
   }
 
@@ -1920,14 +2053,21 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   for (int i = 0; i < a.n_aggs; i++) {
     const auto& ap = plan->aggs[i];
     a.agg_kind[i] = ap.kind;
+    a.cnt_skip[i] = ap.cnt_is_presence ? 1 : 0;
+    a.fsum_idx[i] = ap.fsum_idx;
     if (ap.col_idx >= 0) {
       auto itv = part.d_val.find(ap.col_idx);
       a.agg_val[i] = itv != part.d_val.end() ? itv->second : nullptr;
+      // null-free column (footer-proven): skip the validity read entirely
       auto itd = part.d_valid.find(ap.col_idx);
-      a.agg_valid[i] = itd != part.d_valid.end() ? itd->second : nullptr;
+      a.agg_valid[i] = (ap.cnt_is_presence || itd == part.d_valid.end())
+                           ? nullptr : itd->second;
     }
   }
   a.mask = need_mask ? part.d_mask : nullptr;
+  a.fsum_n = plan->n_fsum;
+  a.fsum = part.d_fsum;
+  a.err = part.d_err;
   a.table = part.d_table;
   a.n_groups = plan->n_groups;
   launch_agg(st, a);
@@ -1939,6 +2079,10 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   std::vector<uint64_t> table(tsz);
   HIP_TRY(hipMemcpyAsync(table.data(), part.d_table, tsz * 8,
                          hipMemcpyDeviceToHost, st));
+  std::vector<uint64_t> fsums((size_t)plan->n_fsum * plan->n_groups * 4);
+  if (!fsums.empty())
+    HIP_TRY(hipMemcpyAsync(fsums.data(), part.d_fsum, fsums.size() * 8,
+                           hipMemcpyDeviceToHost, st));
   int32_t herr = 0;
   HIP_TRY(hipMemcpyAsync(&herr, part.d_err, 4, hipMemcpyDeviceToHost, st));
   HIP_TRY(hipStreamSynchronize(st));
@@ -2111,7 +2255,9 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       std::vector<const std::string*> strs(nr, nullptr);
       for (int64_t r = 0; r < nr; r++) {
         size_t base = (size_t)live[r] * slots;
-        uint64_t cnt = empty_aggregate_row ? 0 : table[base + 2 + 2 * i];
+        uint64_t cnt = empty_aggregate_row ? 0
+                       : (plan->aggs[i].cnt_is_presence ? table[base]
+                                                        : table[base + 2 + 2 * i]);
         if (cnt) {
           int64_t rankv = (int64_t)table[base + 1 + 2 * i];
           if (rankv >= 0 && rankv < (int64_t)c.rank_to_gid.size()) {
@@ -2140,20 +2286,27 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       chc2->n_buffers = 2;
       chc2->buffers = (const void**)calloc(2, sizeof(void*));
       int64_t* cv2 = (int64_t*)eb->grab(nr * 8);
-      for (int64_t r = 0; r < nr; r++)
-        cv2[r] = empty_aggregate_row ? 0 : (int64_t)table[(size_t)live[r] * slots + 2 + 2 * i];
+      for (int64_t r = 0; r < nr; r++) {
+        size_t base = (size_t)live[r] * slots;
+        cv2[r] = empty_aggregate_row ? 0
+                 : (int64_t)(plan->aggs[i].cnt_is_presence
+                                 ? table[base] : table[base + 2 + 2 * i]);
+      }
       chc2->buffers[1] = cv2;
       continue;
     }
     chv->n_buffers = 2;
     chv->buffers = (const void**)calloc(2, sizeof(void*));
     int64_t* vv = (int64_t*)eb->grab(nr * 8);
+    bool cip = plan->aggs[i].cnt_is_presence;
+    int fsi = plan->aggs[i].fsum_idx;
     for (int64_t r = 0; r < nr; r++) {
       size_t base = (size_t)live[r] * slots;
-      // count(*) == presence: k_agg no longer spends atomics on its slots
+      // count(*) == presence: k_agg no longer spends atomics on its slots;
+      // same for any agg whose column is footer-proven null-free
       uint64_t cnt = empty_aggregate_row ? 0
-                     : (kind == AGGK_COUNT_STAR ? table[base]
-                                                : table[base + 2 + 2 * i]);
+                     : ((kind == AGGK_COUNT_STAR || cip)
+                            ? table[base] : table[base + 2 + 2 * i]);
       uint64_t val = empty_aggregate_row ? 0 : table[base + 1 + 2 * i];
       if (kind == AGGK_COUNT_STAR || kind == AGGK_COUNT) {
         vv[r] = (int64_t)cnt;
@@ -2161,6 +2314,11 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         vv[r] = 0;
         validity[r / 8] &= (uint8_t)~(1 << (r % 8));
         nulls++;
+      } else if (kind == AGGK_SUM_F64) {
+        // exact 256-bit superaccumulator, rounded once (acc256_to_double)
+        double d = acc256_to_double(
+            &fsums[((size_t)fsi * plan->n_groups + (size_t)live[r]) * 4]);
+        memcpy(&vv[r], &d, 8);
       } else {
         vv[r] = (int64_t)val;
       }
@@ -2178,8 +2336,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     for (int64_t r = 0; r < nr; r++) {
       size_t base = (size_t)live[r] * slots;
       cv[r] = empty_aggregate_row ? 0
-              : (int64_t)(kind == AGGK_COUNT_STAR ? table[base]
-                                                  : table[base + 2 + 2 * i]);
+              : (int64_t)((kind == AGGK_COUNT_STAR || cip)
+                              ? table[base] : table[base + 2 + 2 * i]);
     }
     chc->buffers[1] = cv;
   }
@@ -2229,7 +2387,7 @@ gpuq_plan::~gpuq_plan() {
     auto F = [](void* p) { if (p) (void)hipFree(p); };
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
-    F(part.d_table); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
+    F(part.d_table); F(part.d_fsum); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_rank); F(part.d_scr);
     F(part.d_present); F(part.d_tmpvalid);
     F(part.d_lits_lane); F(part.d_lits_wave);

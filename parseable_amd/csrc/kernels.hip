@@ -1432,31 +1432,84 @@ __device__ inline void atomic_min_i64_s(uint64_t* addr, int64_t val) {
   }
 }
 
+// ---- exact f64 SUM: 256-bit fixed-point superaccumulator (dev_types.h) ----
+// Decompose d into 4 sign-extended two's-complement limbs, lsb = 2^-160.
+// Window: mantissa lsb >= 2^-160 and |x| < 2^61 (p <= 168 keeps 35 bits of
+// headroom above the largest mantissa bit for 2^34-row partitions).
+// Returns false when d is outside the window (inf/nan/denormal/huge).
+__device__ inline bool acc256_decompose(double d, uint64_t a[4]) {
+  uint64_t bits = (uint64_t)__double_as_longlong(d);
+  uint64_t mant = bits & ((1ull << 52) - 1);
+  int exp = (int)((bits >> 52) & 0x7ff);
+  if (exp == 0x7ff) return false;  // inf / nan
+  if (exp) mant |= 1ull << 52;
+  int e = (exp ? exp : 1) - 1075;
+  int p = e + 160;  // bit position of the mantissa lsb in the 256-bit field
+  if (p < 0 || p > 168) return false;
+  a[0] = a[1] = a[2] = a[3] = 0;
+  int idx = p >> 6, sh = p & 63;
+  a[idx] = mant << sh;
+  if (sh) a[idx + 1] |= mant >> (64 - sh);
+  if (bits >> 63) {  // negative: two's complement of the 256-bit value
+    uint64_t c = 1;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      a[i] = ~a[i] + c;
+      c = (c && a[i] == 0) ? 1 : 0;
+    }
+  }
+  return true;
+}
+
+// limb-wise atomic add with carry propagation (exact mod 2^256 — two's
+// complement keeps mixed-sign accumulation correct; the window bound keeps
+// the true sum inside +-2^255 so the dropped final carry never matters)
+__device__ inline void acc256_add(uint64_t* l, const uint64_t a[4]) {
+  uint64_t carry = 0;
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    uint64_t add = a[i] + carry;
+    uint64_t c2 = (add < carry) ? 1ull : 0ull;  // a[i]+carry wrapped (add==0)
+    if (add) {
+      uint64_t old = atomicAdd((unsigned long long*)&l[i],
+                               (unsigned long long)add);
+      if (old + add < old) c2 = 1;
+    }
+    carry = c2;
+  }
+}
+
 template <bool USE_LDS>
 __global__ void __launch_bounds__(256)
 k_agg(AggArgs a) {
   extern __shared__ uint64_t lt[];
   const int slots = 1 + 2 * a.n_aggs;
   const int64_t tsz = (int64_t)a.n_groups * slots;
+  const int64_t fsz = (int64_t)a.fsum_n * a.n_groups * 4;
   uint64_t* tab;
+  uint64_t* fs;
   if (USE_LDS) {
     tab = lt;
-    for (int64_t i = threadIdx.x; i < tsz; i += blockDim.x) {
-      int s = (int)(i % slots);
+    fs = lt + tsz;
+    for (int64_t i = threadIdx.x; i < tsz + fsz; i += blockDim.x) {
       uint64_t init = 0;
-      if (s > 0 && ((s - 1) & 1) == 0) {  // value slot
-        int ai = (s - 1) / 2;
-        int k = a.agg_kind[ai];
-        if (k == AGGK_MIN_I64 || k == AGGK_MIN_RANK) init = (uint64_t)INT64_MAX;
-        else if (k == AGGK_MAX_I64 || k == AGGK_MAX_RANK) init = (uint64_t)INT64_MIN;
-        else if (k == AGGK_MIN_F64) init = (uint64_t)0x7ff0000000000000ull;   // +inf
-        else if (k == AGGK_MAX_F64) init = (uint64_t)0xfff0000000000000ull;   // -inf
+      if (i < tsz) {
+        int s = (int)(i % slots);
+        if (s > 0 && ((s - 1) & 1) == 0) {  // value slot
+          int ai = (s - 1) / 2;
+          int k = a.agg_kind[ai];
+          if (k == AGGK_MIN_I64 || k == AGGK_MIN_RANK) init = (uint64_t)INT64_MAX;
+          else if (k == AGGK_MAX_I64 || k == AGGK_MAX_RANK) init = (uint64_t)INT64_MIN;
+          else if (k == AGGK_MIN_F64) init = (uint64_t)0x7ff0000000000000ull;   // +inf
+          else if (k == AGGK_MAX_F64) init = (uint64_t)0xfff0000000000000ull;   // -inf
+        }
       }
-      tab[i] = init;
+      lt[i] = init;
     }
     __syncthreads();
   } else {
     tab = a.table;  // pre-initialized by k_init_table
+    fs = a.fsum;    // zeroed before launch
   }
 
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1473,20 +1526,34 @@ k_agg(AggArgs a) {
       if (k == AGGK_COUNT_STAR) continue;  // == presence (export reads slot 0)
       if (a.agg_valid[ai] && !a.agg_valid[ai][i]) continue;
       if (k == AGGK_COUNT) {  // validity only; no value array needed
-        atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+        if (!a.cnt_skip[ai])
+          atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
         continue;
       }
       int64_t v = a.agg_val[ai][i];
       uint64_t* vs = &row[1 + 2 * ai];
       switch (k) {
         case AGGK_SUM_I64: atomicAdd((unsigned long long*)vs, (unsigned long long)v); break;
-        case AGGK_SUM_F64: atomic_add_f64(vs, __longlong_as_double((long long)v)); break;
+        case AGGK_SUM_F64: {
+          double d = __longlong_as_double((long long)v);
+          if (d != 0.0) {
+            uint64_t acc[4];
+            if (!acc256_decompose(d, acc)) {
+              if (a.err) atomicExch(a.err, ERR_FSUM_RANGE);
+            } else {
+              acc256_add(fs + ((int64_t)a.fsum_idx[ai] * a.n_groups + g) * 4,
+                         acc);
+            }
+          }
+          break;
+        }
         case AGGK_MIN_I64: case AGGK_MIN_RANK: atomic_min_i64(vs, v); break;
         case AGGK_MAX_I64: case AGGK_MAX_RANK: atomic_max_i64(vs, v); break;
         case AGGK_MIN_F64: atomic_min_f64(vs, __longlong_as_double((long long)v)); break;
         case AGGK_MAX_F64: atomic_max_f64(vs, __longlong_as_double((long long)v)); break;
       }
-      atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+      if (!a.cnt_skip[ai])
+        atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
     }
   }
 
@@ -1505,11 +1572,7 @@ k_agg(AggArgs a) {
           if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v); break;
         case AGGK_SUM_I64:
           if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v); break;
-        case AGGK_SUM_F64: {
-          double d = __longlong_as_double((long long)v);
-          if (d != 0.0) atomic_add_f64(g, d);
-          break;
-        }
+        case AGGK_SUM_F64: break;  // carried exactly in the fsum flush below
         case AGGK_MIN_I64: case AGGK_MIN_RANK:
           if ((int64_t)v != INT64_MAX) atomic_min_i64(g, (int64_t)v); break;
         case AGGK_MAX_I64: case AGGK_MAX_RANK:
@@ -1524,6 +1587,15 @@ k_agg(AggArgs a) {
           if (d != -__builtin_inf()) atomic_max_f64(g, d);
           break;
         }
+      }
+    }
+    // flush the block's exact superaccumulators (limb-wise with carries)
+    for (int64_t t = threadIdx.x; t < (int64_t)a.fsum_n * a.n_groups;
+         t += blockDim.x) {
+      uint64_t* src = fs + t * 4;
+      if (src[0] | src[1] | src[2] | src[3]) {
+        uint64_t v4[4] = {src[0], src[1], src[2], src[3]};
+        acc256_add(a.fsum + t * 4, v4);
       }
     }
   }
@@ -1695,7 +1767,8 @@ void launch_dict_count(hipStream_t st, const uint8_t* dec, const DevPage* pages,
 }
 
 void launch_agg(hipStream_t st, const AggArgs& a) {
-  size_t lds = (size_t)a.n_groups * (1 + 2 * a.n_aggs) * 8;
+  size_t lds = ((size_t)a.n_groups * (1 + 2 * a.n_aggs) +
+                (size_t)a.fsum_n * a.n_groups * 4) * 8;
   int blocks = (int)((a.n_rows + 255) / 256);
   if (blocks > 2048) blocks = 2048;
   if (blocks < 1) blocks = 1;

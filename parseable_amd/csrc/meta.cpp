@@ -69,7 +69,8 @@ struct Reader {
 
 void parse_stats(Reader& r, ColumnChunkMeta& cm, int phys_type) {
   // parquet Statistics struct: 5=max_value, 6=min_value (new);
-  // 1=max, 2=min (deprecated) — all binary, little-endian for numerics.
+  // 1=max, 2=min (deprecated) — all binary, little-endian for numerics;
+  // 3=null_count (i64).
   int16_t fid = 0;
   std::string mn, mx;
   for (;;) {
@@ -77,6 +78,7 @@ void parse_stats(Reader& r, ColumnChunkMeta& cm, int phys_type) {
     if (!t) break;
     if ((fid == 5 || fid == 1) && t == 8) mx = r.binary();
     else if ((fid == 6 || fid == 2) && t == 8) mn = r.binary();
+    else if (fid == 3 && (t == 5 || t == 6)) cm.null_count = r.zigzag();
     else r.skip(t);
   }
   if ((phys_type == PT_INT64) && mn.size() == 8 && mx.size() == 8) {

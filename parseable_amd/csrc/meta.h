@@ -32,9 +32,12 @@ struct ColumnChunkMeta {
   int64_t total_compressed_size = 0;
   int64_t num_values = 0;
   int codec = CODEC_UNCOMPRESSED;
-  // footer statistics (row-group level min/max), for pruning
+  // footer statistics (row-group level min/max), for pruning and for
+  // chunk-level predicate elision (all-rows-satisfy proofs need
+  // null_count == 0; -1 = not present in the footer)
   bool has_i64_stats = false;
   int64_t stat_min = 0, stat_max = 0;
+  int64_t null_count = -1;
   int64_t start_offset() const {
     return (dict_page_offset >= 0 && dict_page_offset < data_page_offset)
                ? dict_page_offset : data_page_offset;

@@ -17,7 +17,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from datagen.gen import gen_stream  # noqa: E402
 from oracle import query_oracle as qo  # noqa: E402
-from oracle.compare import rows_equal  # noqa: E402
+from oracle.compare import FLOAT_RTOL, rows_equal  # noqa: E402
 from tests.golden_queries import GOLDEN_FIXTURES, GOLDEN_QUERIES  # noqa: E402
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
@@ -37,7 +37,7 @@ def main():
         for qname, query in GOLDEN_QUERIES.get(name, []):
             r1 = qo.execute(info["files"], query)
             r2 = qo.execute_acero(info["files"], query)
-            assert rows_equal(r1["rows"], r2["rows"]), (
+            assert rows_equal(r1["rows"], r2["rows"], float_rtol=FLOAT_RTOL), (
                 f"oracle disagreement on {name}/{qname}:\n{r1['rows']}\nvs\n{r2['rows']}"
             )
             answers[f"{name}/{qname}"] = {"query": query, "result": r1}

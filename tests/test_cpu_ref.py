@@ -7,7 +7,7 @@ decoders implement — is correct, without pyarrow in the loop."""
 import pytest
 
 from oracle import cpu_ref_runner
-from oracle.compare import assert_rows_equal
+from oracle.compare import FLOAT_RTOL, assert_rows_equal
 from tests.golden_queries import GOLDEN_QUERIES
 
 
@@ -29,4 +29,7 @@ def test_cpu_ref_matches_golden(golden, case):
     entry = golden["answers"][case]
     files = golden["fixtures"][fx]["files"]
     r = cpu_ref_runner.execute(files, entry["query"])
-    assert_rows_equal(r["rows"], entry["result"]["rows"], case)
+    # the scalar C restatement's sums are compensated but order-
+    # dependent: rtol gate (the 1-ULP gate is GPU vs oracle)
+    assert_rows_equal(r["rows"], entry["result"]["rows"], case,
+                      float_rtol=FLOAT_RTOL)
