@@ -567,7 +567,7 @@ static void decode_chunk(const uint8_t *file, ChunkMeta *cm, SchemaCol *sc,
 /* query model                                                        */
 /* ----------------------------------------------------------------- */
 enum { OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_BETWEEN, OP_CONTAINS };
-enum { AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_MIN, AGG_MAX };
+enum { AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_MIN, AGG_MAX, AGG_AVG };
 
 typedef struct {
     char col[256]; int op;
@@ -755,7 +755,7 @@ static void run_query(Query *q) {
                         g_agg_is_f64[i] = 1;
                         double v = c->f64[r];
                         if (!gr->cnt[i]) { gr->f64v[i] = v; gr->f64c[i] = 0.0; }
-                        else if (a->op == AGG_SUM) {
+                        else if (a->op == AGG_SUM || a->op == AGG_AVG) {
                             /* Neumaier compensated sum (order-dependent but
                                ~exact; final gate vs the oracle is rtol) */
                             double s2 = gr->f64v[i] + v;
@@ -771,7 +771,7 @@ static void run_query(Query *q) {
                     } else {
                         int64_t v = c->i64[r];
                         if (!gr->cnt[i]) { gr->i64v[i] = v; }
-                        else if (a->op == AGG_SUM) gr->i64v[i] += v;
+                        else if (a->op == AGG_SUM || a->op == AGG_AVG) gr->i64v[i] += v;
                         else if (a->op == AGG_MIN) { if (v < gr->i64v[i]) gr->i64v[i] = v; }
                         else if (a->op == AGG_MAX) { if (v > gr->i64v[i]) gr->i64v[i] = v; }
                         gr->cnt[i]++;
@@ -834,6 +834,10 @@ static void run_query(Query *q) {
             Agg *a = &q->aggs[i];
             if (a->op == AGG_COUNT_STAR || a->op == AGG_COUNT) printf("%" PRId64, gr->cnt[i]);
             else if (!gr->cnt[i]) printf("\\N");
+            else if (a->op == AGG_AVG)
+                printf("%.17g", (g_agg_is_f64[i] ? gr->f64v[i] + gr->f64c[i]
+                                                 : (double)gr->i64v[i]) /
+                                    (double)gr->cnt[i]);
             else if (g_agg_is_f64[i])
                 printf("%.17g", q->aggs[i].op == AGG_SUM
                                     ? gr->f64v[i] + gr->f64c[i] : gr->f64v[i]);
@@ -865,6 +869,7 @@ int main(int argc, char **argv) {
                 *colon = 0;
                 if (!strcmp(s, "count")) a->op = AGG_COUNT;
                 else if (!strcmp(s, "sum")) a->op = AGG_SUM;
+                else if (!strcmp(s, "avg")) a->op = AGG_AVG;
                 else if (!strcmp(s, "min")) a->op = AGG_MIN;
                 else if (!strcmp(s, "max")) a->op = AGG_MAX;
                 else die("bad agg");

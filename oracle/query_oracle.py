@@ -245,7 +245,7 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
             )
             fv = vals[sel][vsel]
             gi = inv[vsel]
-            if op == "sum":
+            if op in ("sum", "avg"):
                 is_f = np.issubdtype(fv.dtype, np.floating)
                 if is_f:
                     # f64 sums: keep the per-group VALUES so the final sum is
@@ -261,7 +261,7 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
                 out = np.zeros(G, dtype=np.int64)
                 np.add.at(out, gi, fv)
                 cnt = np.bincount(gi, minlength=G)
-                file_res.append(("sum", out, cnt))
+                file_res.append((op, out, cnt))  # "sum" or "avg" (i64)
             elif op == "min":
                 out = np.full(G, np.inf if np.issubdtype(fv.dtype, np.floating) else I64_MAX,
                               dtype=fv.dtype if np.issubdtype(fv.dtype, np.floating) else np.int64)
@@ -294,6 +294,12 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
                         if st[ai] is None:
                             st[ai] = []
                         st[ai].append(part)
+                elif kind == "avg":  # i64 avg: carry (exact int sum, count)
+                    c2 = int(r[2][gidx])
+                    if c2:
+                        v = int(r[1][gidx])
+                        st[ai] = ((v, c2) if st[ai] is None
+                                  else (st[ai][0] + v, st[ai][1] + c2))
                 else:
                     v = r[1][gidx]
                     present = int(r[2][gidx]) > 0
@@ -317,10 +323,15 @@ def execute(files: list[str], query: dict, extra_tables=None) -> dict:
         for a, s in zip(aggs, st):
             if s is None and a["agg"] in ("count", "count_star"):
                 s = 0
-            elif isinstance(s, list):  # f64 sum: exact, rounded once
+            elif isinstance(s, list):  # f64 sum/avg: exact, rounded once
                 import math
 
-                s = math.fsum(np.concatenate(s))
+                allv = np.concatenate(s)
+                s = math.fsum(allv)
+                if a["agg"] == "avg":
+                    s = s / len(allv)
+            elif isinstance(s, tuple):  # i64 avg: exact sum / count
+                s = s[0] / s[1]
             row.append(s)
         rows.append(row)
     rows.sort(key=lambda r: tuple(_sort_key(v) for v in r[: len(group_by)]))
@@ -408,8 +419,9 @@ def execute_acero(files: list[str], query: dict) -> dict:
             agglist.append(([], "count_all"))
             names.append("count_all")
         else:
-            agglist.append((a["col"], op if op != "count" else "count"))
-            names.append(f"{a['col']}_{op}")
+            aop = {"avg": "mean"}.get(op, op)
+            agglist.append((a["col"], aop))
+            names.append(f"{a['col']}_{aop}")
 
     if group_by:
         res = pa.TableGroupBy(tbl, group_by).aggregate(agglist)

@@ -1868,7 +1868,10 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   HIP_TRY(hipEventCreate(&ev1));
   HIP_TRY(hipEventCreate(&ev_decomp));
 
-  const bool need_mask = !plan->preds.empty();
+  // a query whose every predicate is chunk-stats-proven everywhere needs no
+  // selection mask at all (string/contains preds are never elided, so their
+  // presence keeps pred_ranges non-empty)
+  const bool need_mask = !plan->preds.empty() && !part.pred_ranges.empty();
   HIP_TRY(hipMemsetAsync(part.d_err, 0, 4, st));
   if (need_mask) HIP_TRY(hipMemsetAsync(part.d_mask, 1, part.n_rows, st));
   launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),

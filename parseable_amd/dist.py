@@ -153,7 +153,7 @@ class DistMerger:
                     continue
                 if self.is_f64[i]:
                     v = vcol.to_numpy(zero_copy_only=False).astype(np.float64)[has]
-                    if a["agg"] == "sum":
+                    if a["agg"] in ("sum", "avg"):
                         np.add.at(np_sums_f[:, i], g2, v)
                     elif a["agg"] == "min":
                         np.minimum.at(np_mins_f[:, i], g2, v)
@@ -162,9 +162,10 @@ class DistMerger:
                 else:
                     import pyarrow as pa
 
-                    fill = {"sum": 0, "min": I64_MAX, "max": I64_MIN}[a["agg"]]
+                    fill = {"sum": 0, "avg": 0, "min": I64_MAX,
+                            "max": I64_MIN}[a["agg"]]
                     v = vcol.fill_null(fill).to_numpy(zero_copy_only=False).astype(np.int64)[has]
-                    if a["agg"] == "sum":
+                    if a["agg"] in ("sum", "avg"):
                         np.add.at(np_sums_i[:, i], g2, v)
                     elif a["agg"] == "min":
                         np.minimum.at(np_mins[:, i], g2, v)
@@ -184,7 +185,7 @@ class DistMerger:
         if self.world > 1:
             dist.all_reduce(presence, op=dist.ReduceOp.SUM)
             dist.all_reduce(counts, op=dist.ReduceOp.SUM)
-            if any(a["agg"] == "sum" for a in self.aggs):
+            if any(a["agg"] in ("sum", "avg") for a in self.aggs):
                 dist.all_reduce(sums_i, op=dist.ReduceOp.SUM)
                 dist.all_reduce(sums_f, op=dist.ReduceOp.SUM)
             if any(a["agg"] == "min" for a in self.aggs):
@@ -226,6 +227,10 @@ class DistMerger:
                 elif a["agg"] == "sum":
                     row.append(sums_f[gi, i].item() if self.is_f64[i]
                                else sums_i[gi, i].item())
+                elif a["agg"] == "avg":
+                    sv = (sums_f[gi, i].item() if self.is_f64[i]
+                          else sums_i[gi, i].item())
+                    row.append(sv / c)
                 elif str_minmax[i] is not None:
                     row.append(str_minmax[i].get(gi))
                 elif a["agg"] == "min":

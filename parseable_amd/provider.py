@@ -304,7 +304,7 @@ class StagingScanner:
                     cur = st[1][i]
                     if cur is None:
                         st[1][i] = v
-                    elif a["agg"] == "sum":
+                    elif a["agg"] in ("sum", "avg"):
                         st[1][i] = cur + v
                     elif a["agg"] == "min":
                         st[1][i] = min(cur, v)
@@ -605,7 +605,9 @@ def _build_c_query(query: dict, keep):
     aggs = query.get("select", [])
     caggs = (GpuqAgg * max(len(aggs), 1))()
     for i, a in enumerate(aggs):
-        caggs[i].op = AGGS[a["agg"]]
+        # avg = sum/count at the Final merge (DataFusion's Avg partial state
+        # is (sum, count) too); the engine only ever computes the exact sum
+        caggs[i].op = AGGS["sum" if a["agg"] == "avg" else a["agg"]]
         if a.get("col"):
             ab = a["col"].encode()
             keep.append(ab)
@@ -737,6 +739,14 @@ def merge_partials(batches, query):
         picked_idx = list(range(nk)) + [nk + 1 + 2 * i
                                         for i in range(len(aggs))]
         cols = [b.column(i).to_pylist() for i in picked_idx]
+        for i, a in enumerate(aggs):
+            if a["agg"] != "avg":
+                continue
+            cnts = b.column(nk + 2 + 2 * i).to_pylist()
+            cols[nk + i] = [
+                (None if not c else s / c)
+                for s, c in zip(cols[nk + i], cnts)
+            ]
         if nk:
             kt = pa.table({f"k{i}": b.column(i) for i in range(nk)})
             # default null_placement is at_end — matches the oracle's
@@ -771,7 +781,7 @@ def merge_partials(batches, query):
                 st["cnts"][i] += c
                 if st["vals"][i] is None:
                     st["vals"][i] = v
-                elif a["agg"] == "sum":
+                elif a["agg"] in ("sum", "avg"):
                     st["vals"][i] += v
                 elif a["agg"] == "min":
                     st["vals"][i] = min(st["vals"][i], v)
@@ -786,6 +796,9 @@ def merge_partials(batches, query):
         for i, a in enumerate(aggs):
             if a["agg"] in ("count_star", "count"):
                 row.append(st["cnts"][i])
+            elif a["agg"] == "avg":
+                row.append(st["vals"][i] / st["cnts"][i]
+                           if st["cnts"][i] > 0 else None)
             else:
                 row.append(st["vals"][i] if st["cnts"][i] > 0 else None)
         rows.append(row)

@@ -51,7 +51,7 @@ def gen_query(rng: random.Random, cfg: dict, n_rows: int):
     numcols = cfg["i64"] + cfg["f64"]
     for _ in range(rng.randint(0, 3)):
         col = rng.choice(numcols)
-        op = rng.choice(["sum", "min", "max", "count"])
+        op = rng.choice(["sum", "avg", "min", "max", "count"])
         aggs.append({"agg": op, "col": col})
     if rng.random() < 0.25 and cfg["keys"]:
         aggs.append({"agg": rng.choice(["min", "max"]),

@@ -61,6 +61,16 @@ GOLDEN_QUERIES = {
             "group_by": ["level"],
             "preds": [{"col": "level", "op": "ge", "lit": "INFO"}],
         }),
+        ("avg_latency_by_level", {
+            "select": [{"agg": "avg", "col": "latency"}, {"agg": "count_star"}],
+            "group_by": ["level"],
+        }),
+        ("avg_f64_by_key", {
+            "select": [{"agg": "avg", "col": "f_f64"},
+                       {"agg": "sum", "col": "f_f64"}],
+            "group_by": ["f_str2"],
+            "preds": [{"col": "latency", "op": "ge", "lit": 250_000}],
+        }),
         # extended coverage (ext: skipped by the scalar C oracle — pinned by
         # the pyarrow oracle + Acero instead)
         ("minmax_utf8", {
