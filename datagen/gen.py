@@ -110,6 +110,23 @@ def _minute_batch(config: str, rng: np.random.Generator, n: int, minute: int):
                 for i, m in enumerate(msgs)
             ]
             cols["message"] = pa.array(msgs)
+    elif config == "c5":
+        # raw-byte utf8 stress: moderate/high-cardinality string columns that
+        # overflow the dictionary page (written with a small
+        # dictionary_pagesize_limit) into PLAIN-fallback data pages — the
+        # writer behavior of parseable/streams.rs:705-780 under dict
+        # overflow, which forces the engine's row-hash group-by path.
+        cols["latency"] = pa.array(
+            rng.integers(0, 10**6, n, dtype=np.int64), type=pa.int64()
+        )
+        cols["level"] = _dict_col(rng, n, LEVELS)
+        cols["host"] = _dict_col(rng, n, [f"host-{i:04d}" for i in range(1000)])
+        tr = rng.integers(0, 1500, n)
+        cols["trace"] = pa.array([f"tr-{v:08d}" for v in tr])
+        mask = rng.random(n) < 0.3
+        tagv = rng.integers(0, 800, n)
+        cols["opt_tag"] = pa.array(
+            [None if m else f"tag-{v:06d}" for m, v in zip(mask, tagv)])
     elif config == "c4":
         # OTel-shaped: 3 group keys + 61 sparse attribute columns, 50-90% null
         cols["service"] = _dict_col(rng, n, [f"svc-{i}" for i in range(30)])
@@ -249,6 +266,9 @@ def _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
         kw = {}
         if data_page_size:
             kw["data_page_size"] = data_page_size
+        if config == "c5":
+            # force dict-page overflow -> PLAIN fallback mid-chunk
+            kw["dictionary_pagesize_limit"] = 4096
         ts_idx = tbl.schema.get_field_index("p_timestamp")
         pq.write_table(
             tbl,

@@ -591,6 +591,7 @@ typedef struct {
 /* ----------------------------------------------------------------- */
 typedef struct {
     char *key;                   /* encoded key (NUL-joined, \1 marks NULL) */
+    uint32_t klen;               /* full key length (keys embed NULs) */
     int64_t cnt[16];             /* per-agg count of accumulated values */
     int64_t i64v[16];
     double f64v[16];
@@ -610,7 +611,10 @@ static Group *ht_get(HashTab *ht, const char *key, size_t klen) {
         Group *ns = calloc(ncap, sizeof(Group));
         for (uint32_t i = 0; i < ht->cap; i++) {
             if (!ht->slots[i].key) continue;
-            uint64_t h = fnv1a(ht->slots[i].key, strlen(ht->slots[i].key) + 1);
+            /* hash the FULL stored length: multi-key buffers embed NUL
+               separators, so strlen() truncated and split groups after the
+               first rehash (caught by the g_c5 7k-group fixture) */
+            uint64_t h = fnv1a(ht->slots[i].key, ht->slots[i].klen);
             uint32_t j = h & (ncap - 1);
             while (ns[j].key) j = (j + 1) & (ncap - 1);
             ns[j] = ht->slots[i];
@@ -623,10 +627,12 @@ static Group *ht_get(HashTab *ht, const char *key, size_t klen) {
         if (!ht->slots[j].key) {
             ht->slots[j].key = malloc(klen);
             memcpy(ht->slots[j].key, key, klen);
+            ht->slots[j].klen = (uint32_t)klen;
             ht->n++;
             return &ht->slots[j];
         }
-        if (!memcmp(ht->slots[j].key, key, klen) && ht->slots[j].key[klen - 1] == 0)
+        if (ht->slots[j].klen == (uint32_t)klen &&
+            !memcmp(ht->slots[j].key, key, klen))
             return &ht->slots[j];
         j = (j + 1) & (ht->cap - 1);
     }

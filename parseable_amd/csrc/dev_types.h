@@ -69,7 +69,8 @@ enum CmpMode { CMP_EQ = 0, CMP_NE, CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_RANGE };
 
 enum { AGGK_COUNT_STAR = 0, AGGK_COUNT, AGGK_SUM_I64, AGGK_SUM_F64,
        AGGK_MIN_I64, AGGK_MAX_I64, AGGK_MIN_F64, AGGK_MAX_F64,
-       AGGK_MIN_RANK, AGGK_MAX_RANK };  // utf8 min/max via dict sort-ranks
+       AGGK_MIN_RANK, AGGK_MAX_RANK,   // utf8 min/max via dict sort-ranks
+       AGGK_MIN_STR, AGGK_MAX_STR };   // utf8 min/max via strrefs (hash cols)
 
 constexpr int MAX_KEYS = 4;
 constexpr int MAX_AGGS = 8;
@@ -99,6 +100,7 @@ struct AggArgs {
   int32_t fsum_idx[MAX_AGGS];        // agg -> fsum table index (-1 = none)
   int32_t fsum_n;                    // number of f64-sum aggregates
   uint64_t* fsum;                    // superaccumulators (zeroed before launch)
+  const uint8_t* dec;                // arena base (strref compares/hashes)
   uint64_t* table;
   int32_t n_groups;
   int32_t* err;

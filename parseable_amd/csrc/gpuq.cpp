@@ -79,6 +79,15 @@ struct ColPlan {
                             // chunk (false = predicate-only: chunks whose
                             // footer stats prove the predicate for all rows
                             // skip decode entirely)
+  // raw-byte utf8 mode: the column has PLAIN-fallback data pages (dict
+  // overflow, parseable/streams.rs writer defaults) but is used as a group
+  // key / min-max / non-LIKE predicate — the dict-only gid path cannot
+  // cover it. Rows carry STRREFs (dec-arena offsets) in d_val; group ids
+  // come from the device hash build (k_hash_build/lookup); min/max and
+  // predicates compare bytes in the arena. Mirrors DataFusion's row-hash
+  // aggregation over arbitrary keys (stream_schema_provider.rs:219-225
+  // hands the plan to that engine).
+  bool hash_mode = false;
   bool need_gid_valid = false;  // validity bytes alongside gid (COUNT(utf8))
   // predicate routing
   std::vector<int> lut_preds;      // preds evaluated per-dict-entry (utf8)
@@ -131,6 +140,12 @@ struct ChunkTask {
   std::vector<int64_t> dictv;      // local dict values (i64 / f64 bits)
   std::vector<uint8_t> lut;        // combined pred LUT (AND of lut_preds)
   bool has_plain_data_pages = false;
+  // hash-mode (raw-byte utf8) extras:
+  std::vector<int64_t> strof;      // dict entries: page-relative [len] offsets
+  std::vector<int64_t> pvals;      // PLAIN pages: page-relative value offsets
+  std::vector<uint32_t> pval_page_n;  // per PLAIN data page: #values in pvals
+  std::vector<uint32_t> pval_base;    // per PLAIN data page: dictv-pool base
+  uint64_t dict_dst = (uint64_t)-1;   // dec-arena dst of the dict page image
 };
 
 struct RgRef {
@@ -141,7 +156,7 @@ struct RgRef {
 
 // decode-task lists per (kind,col); ids index into the partition's page array
 enum TaskKind { TK_DICT_GID, TK_DICT_VAL, TK_PLAIN_VAL, TK_DELTA_VAL,
-                TK_DICT_MASK, TK_BYTES_CONTAINS, TK_N };
+                TK_DICT_MASK, TK_BYTES_CONTAINS, TK_POOL_VAL, TK_N };
 
 struct Partition {
   int device = -1;
@@ -186,6 +201,12 @@ struct Partition {
   int32_t* d_err = nullptr;
   uint64_t* d_table = nullptr;
   uint64_t* d_fsum = nullptr;    // [n_fsum][n_groups][4] superacc limbs
+  // raw-byte utf8 hash state (shared table, rebuilt per hash column)
+  uint64_t* d_hkeys = nullptr;   // open-addressing slots (strref keys)
+  int32_t* d_hgids = nullptr;
+  uint32_t* d_hcount = nullptr;
+  std::map<int, uint64_t*> d_gid2ref;   // hash col -> gid -> strref
+  std::map<int, uint32_t> hash_claimed; // hash col -> claimed count (last exec)
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;        // concatenated CONTAINS needles
   std::vector<uint32_t> needle_off;   // per plan-pred offset into the pool
@@ -244,6 +265,11 @@ struct Ring {
 
 }  // namespace
 
+namespace {
+constexpr int HASH_LOG2 = 24;          // shared open-addressing table slots
+constexpr int32_t GID_CAP = 1 << 22;   // max distinct groups (per col & total)
+}  // namespace
+
 struct gpuq_plan {
   gpuq_ctx* ctx = nullptr;
   std::vector<std::unique_ptr<MappedFile>> files;
@@ -258,6 +284,7 @@ struct gpuq_plan {
   std::vector<Partition> parts;
   int32_t n_groups = 0;          // product of key sizes (incl null slots)
   int32_t n_fsum = 0;            // number of exact-f64-sum side tables
+  bool has_hash = false;         // some column runs raw-byte utf8 hash mode
   bool fused_count = false;      // single dict key + count(*)-only + no preds
   std::mutex mu;
   // metrics
@@ -380,6 +407,48 @@ bool eval_str_pred(const gpuq_pred& p, const std::string& lit,
     case GPUQ_GE: return c >= 0;
   }
   return false;
+}
+
+// parse a v1 data page's def levels (bit-width-1 RLE/bit-packed hybrid,
+// parquet-format RLE); returns the payload offset after the levels and
+// fills defs (1 = value present)
+uint32_t parse_def1(const uint8_t* data, uint32_t nv, std::vector<uint8_t>& defs) {
+  uint32_t dl;
+  memcpy(&dl, data, 4);
+  defs.assign(nv, 1);
+  const uint8_t* p = data + 4;
+  const uint8_t* end = p + dl;
+  uint32_t v = 0;
+  while (v < nv && p < end) {
+    uint64_t hdr = 0;
+    int sh = 0;
+    for (;;) {
+      uint8_t b = *p++;
+      hdr |= (uint64_t)(b & 0x7f) << sh;
+      if (!(b & 0x80)) break;
+      sh += 7;
+    }
+    if (hdr & 1) {
+      uint32_t groups = (uint32_t)(hdr >> 1);
+      for (uint32_t g = 0; g < groups; g++) {
+        uint8_t byte = p[g];
+        for (int j = 0; j < 8; j++) {
+          uint32_t idx = v + g * 8 + j;
+          if (idx < nv) defs[idx] = (byte >> j) & 1;
+        }
+      }
+      p += groups;
+      uint32_t add = groups * 8;
+      v += add > nv - v ? nv - v : add;
+    } else {
+      uint32_t cnt = (uint32_t)(hdr >> 1);
+      uint8_t val = *p++;
+      if (cnt > nv - v) cnt = nv - v;
+      std::fill(defs.begin() + v, defs.begin() + v + cnt, val);
+      v += cnt;
+    }
+  }
+  return 4 + dl;
 }
 
 int64_t now_ns() {
@@ -709,6 +778,59 @@ extern "C" gpuq_plan* gpuq_plan_build(
     plan->parts[best].rgs.push_back(r);
   }
 
+  // --- raw-byte utf8 (hash-mode) detection: a utf8 column used as group
+  // key / min-max / non-LIKE predicate whose selected chunks contain
+  // PLAIN-fallback data pages cannot ride the dict-only gid path. Decided
+  // BEFORE the per-partition build so every partition routes consistently.
+  // (LIKE alone stays on the contains-window path: it handles PLAIN pages
+  // already and is much faster than per-row byte scans.) ---
+  {
+    struct Probe { int col; const MappedFile* mf; const ColumnChunkMeta* cm; int64_t rows; };
+    std::vector<Probe> probes;
+    for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+      auto& c = plan->cols[ci];
+      if (c.is_bin || c.phys != PT_BYTE_ARRAY) continue;
+      bool non_like = false;
+      for (int pidx : c.lut_preds)
+        non_like |= (plan->preds[pidx].p.op != GPUQ_CONTAINS);
+      if (!(c.need_gid || c.need_rank || non_like)) continue;
+      for (auto& r : selected) {
+        const auto& mf = *plan->files[r.file_idx];
+        int si = mf.meta.col_index(c.name);
+        if (si < 0) continue;
+        probes.push_back({(int)ci, &mf,
+                          &mf.meta.row_groups[r.rg_idx].chunks[si],
+                          mf.meta.row_groups[r.rg_idx].num_rows});
+      }
+    }
+    std::vector<uint8_t> plain(probes.size(), 0);
+    parallel_for(probes.size(), [&](size_t i) {
+      auto pages = walk_pages(probes[i].mf->data, *probes[i].cm, probes[i].rows);
+      for (auto& pi : pages)
+        if (pi.type == PAGE_DATA && pi.encoding == ENC_PLAIN) plain[i] = 1;
+    });
+    for (size_t i = 0; i < probes.size(); i++)
+      if (plain[i]) plan->cols[probes[i].col].hash_mode = true;
+    for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+      auto& c = plan->cols[ci];
+      if (!c.hash_mode) continue;
+      plan->has_hash = true;
+      c.need_val = true;       // d_val carries row strrefs
+      c.val_always = true;
+      if (c.need_rank) {       // min/max via strref byte compares
+        c.need_rank = false;
+        for (auto& ap : plan->aggs)
+          if (ap.col_idx == (int)ci) {
+            if (ap.kind == AGGK_MIN_RANK) ap.kind = AGGK_MIN_STR;
+            if (ap.kind == AGGK_MAX_RANK) ap.kind = AGGK_MAX_STR;
+          }
+      }
+      bool is_key = false;
+      for (int gci : plan->group_cols) is_key |= (gci == (int)ci);
+      if (!is_key) { c.need_gid = false; c.need_gid_valid = false; }
+    }
+  }
+
   // --- per-partition host build: page walks, dict processing, device
   // images. The 1 B-row configs have thousands of (row-group x column)
   // chunks: every per-chunk pass below (page-header walks, dict-page
@@ -806,12 +928,15 @@ extern "C" gpuq_plan* gpuq_plan_build(
         } else throw std::runtime_error("unsupported codec");
         if (c.phys == PT_BYTE_ARRAY) {
           const uint8_t* q = d;
-          if (c.need_gid || c.need_rank) dict_strs[i].reserve(pi.num_values);
+          if (!c.hash_mode && (c.need_gid || c.need_rank))
+            dict_strs[i].reserve(pi.num_values);
           for (int32_t k = 0; k < pi.num_values; k++) {
             uint32_t l; memcpy(&l, q, 4); q += 4;
-            if (c.need_gid || c.need_rank)
+            if (c.hash_mode)
+              t.strof.push_back((int64_t)(q - d) - 4);  // page-relative [len]
+            else if (c.need_gid || c.need_rank)
               dict_strs[i].emplace_back((const char*)q, l);
-            if (!c.lut_preds.empty()) {
+            if (!c.hash_mode && !c.lut_preds.empty()) {
               uint8_t ok = 1;
               for (int pidx : c.lut_preds) {
                 const auto& pp = plan->preds[pidx];
@@ -837,9 +962,55 @@ extern "C" gpuq_plan* gpuq_plan_build(
       for (auto& pi : t.pages)
         if (pi.type == PAGE_DATA && pi.encoding == ENC_PLAIN)
           t.has_plain_data_pages = true;
-      if (t.has_plain_data_pages && (c.need_gid || c.need_rank))
+      if (c.hash_mode) {
+        // PLAIN pages: walk the [u32 len][bytes] chain once (def-level
+        // aware) and record each non-null value's page-relative offset —
+        // made absolute in phase 3 once page dst slots are known
+        std::vector<uint8_t> img, defs;
+        bool optional = false;
+        {
+          int si2 = mf.meta.col_index(c.name);
+          if (si2 >= 0) optional = mf.meta.columns[si2].optional;
+        }
+        for (auto& pi : t.pages) {
+          if (pi.type != PAGE_DATA || pi.encoding != ENC_PLAIN) continue;
+          const uint8_t* data = mf.data + pi.payload_off;
+          if (cm.codec == CODEC_LZ4_RAW) {
+            img.resize(pi.uncomp_size);
+            int n2 = lz4_decompress_host(data, pi.comp_size, img.data(),
+                                         img.size());
+            if (n2 == pi.uncomp_size) data = img.data();
+            else if (pi.comp_size != pi.uncomp_size)
+              throw std::runtime_error("plain page lz4 failure (hash col)");
+          } else if (cm.codec != CODEC_UNCOMPRESSED) {
+            throw std::runtime_error("unsupported codec");
+          }
+          uint32_t pos = 0;
+          const uint32_t nv = pi.num_values;
+          bool has_def = false;
+          if (optional) {
+            pos = parse_def1(data, nv, defs);
+            has_def = true;
+          }
+          uint32_t cnt0 = (uint32_t)t.pvals.size();
+          for (uint32_t vI = 0; vI < nv; vI++) {
+            if (has_def && !defs[vI]) continue;
+            if (pos + 4 > (uint32_t)pi.uncomp_size)
+              throw std::runtime_error("plain page overrun (hash col)");
+            uint32_t len;
+            memcpy(&len, data + pos, 4);
+            t.pvals.push_back((int64_t)pos);
+            pos += 4 + len;
+            if (pos > (uint32_t)pi.uncomp_size)
+              throw std::runtime_error("plain page overrun (hash col)");
+          }
+          t.pval_page_n.push_back((uint32_t)t.pvals.size() - cnt0);
+        }
+      }
+      if (t.has_plain_data_pages && !c.hash_mode &&
+          (c.need_gid || c.need_rank))
         throw std::runtime_error("dict-only utf8 operation with PLAIN fallback pages "
-                                 "(high-cardinality hashing): next row (SURVEY §8f) — " + c.name);
+                                 "outside hash mode — " + c.name);
       part.chunks[i] = std::move(t);
     });
 
@@ -869,18 +1040,40 @@ extern "C" gpuq_plan* gpuq_plan_build(
     int32_t pid_cursor = 0;
     for (size_t i = 0; i < part.chunks.size(); i++) {
       auto& t = part.chunks[i];
+      auto& c = plan->cols[t.col_idx];
       bases[i] = {(uint32_t)part.remap_pool.size(),
                   (uint32_t)part.dictv_pool.size(),
                   (uint32_t)part.lut_pool.size(), part.dec_bytes, pid_cursor};
       t.dictv_pool_base = bases[i].dictv;
       part.remap_pool.insert(part.remap_pool.end(), t.remap.begin(), t.remap.end());
-      part.dictv_pool.insert(part.dictv_pool.end(), t.dictv.begin(), t.dictv.end());
+      if (!c.hash_mode)
+        part.dictv_pool.insert(part.dictv_pool.end(), t.dictv.begin(), t.dictv.end());
       part.lut_pool.insert(part.lut_pool.end(), t.lut.begin(), t.lut.end());
+      if (c.hash_mode) {
+        // the dict page image joins the dec arena (device-side strings);
+        // its entry offsets become absolute strrefs in the dictv pool
+        for (auto& pi : t.pages)
+          if (pi.type == PAGE_DICT) {
+            t.dict_dst = part.dec_bytes;
+            part.dec_bytes += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
+          }
+        for (int64_t rel : t.strof)
+          part.dictv_pool.push_back((int64_t)(t.dict_dst + (uint64_t)rel));
+      }
+      size_t plain_i = 0, pv_off = 0;
       for (auto& pi : t.pages) {
         if (pi.type != PAGE_DATA) continue;
+        uint64_t pg_dst = part.dec_bytes;
         pid_cursor++;
         // 16-align page images so dst and LDS-ring offsets share alignment
         part.dec_bytes += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
+        if (c.hash_mode && pi.encoding == ENC_PLAIN) {
+          t.pval_base.push_back((uint32_t)part.dictv_pool.size());
+          uint32_t nvp = t.pval_page_n[plain_i++];
+          for (uint32_t k = 0; k < nvp; k++)
+            part.dictv_pool.push_back((int64_t)(pg_dst + (uint64_t)t.pvals[pv_off + k]));
+          pv_off += nvp;
+        }
       }
     }
 
@@ -920,7 +1113,131 @@ extern "C" gpuq_plan* gpuq_plan_build(
       int32_t page_id = bases[ti].pid;
       const uint32_t rstart = stubs[ti].rstart;
 
+      // decompress planning for one page image (data or hash-col dict):
+      // segment walk + litpar/backref records, exactly the round-1 logic
+      auto plan_decomp = [&](const uint8_t* praw, uint64_t src_abs,
+                             uint64_t dst_abs, uint32_t comp, uint32_t uncomp,
+                             bool raw_page) {
+        Lz4Plan lp;
+        if (!raw_page) {
+          // test knob: exercise the serial windowed fallback kernel on
+          // arbitrary content (no organic fixture produces a
+          // piece-explosion page)
+          static const bool force_fb =
+              std::getenv("GPUQ_FORCE_LZ4_FALLBACK") != nullptr;
+          try {
+            lp = lz4_walk(praw, comp, uncomp, 8192);
+            if (force_fb) {
+              lp.fallback = true;
+              lp.resolved.clear();
+              lp.pieces.clear();
+            }
+            // dense short-sequence pages (LZ4 over near-random dict
+            // indices) degenerate the segment kernel into a serial token
+            // parse — switch those to the all-literal/all-resolved plan
+            if (!lp.fallback && lp.n_seq >= 256 && uncomp / lp.n_seq < 96) {
+              Lz4Plan lp2 = lz4_walk(praw, comp, uncomp, 8192, /*litpar=*/true);
+              if (!lp2.fallback) lp = std::move(lp2);
+            }
+          } catch (const std::exception&) {
+            if (comp == uncomp) raw_page = true;  // stored raw
+            else throw;
+          }
+        }
+        if (raw_page) {
+          DevSeg sgl{};
+          sgl.src_off = src_abs;
+          sgl.dst_off = dst_abs;
+          sgl.comp_len = comp;
+          sgl.out_len = uncomp;
+          sgl.raw = 1;
+          cb.segs.push_back(sgl);
+          return;
+        }
+        for (const auto& sg : lp.segs) {
+          DevSeg d2{};
+          d2.src_off = src_abs + sg.s_off;
+          d2.dst_off = dst_abs + sg.d_off;
+          d2.comp_len = sg.comp_len;
+          d2.out_len = sg.out_len;
+          d2.big = sg.big;
+          cb.segs.push_back(d2);
+        }
+        for (const auto& lt : lp.lits) {
+          DevLit dl{src_abs + lt.src, dst_abs + lt.dst, lt.len, 0};
+          (lt.len <= 256 ? cb.lits_lane : cb.lits_wave).push_back(dl);
+        }
+        if (lp.fallback) {
+          // piece explosion: serial windowed wave per page
+          DevPageBr pb{(uint32_t)cb.brs.size(), (uint32_t)lp.backrefs.size()};
+          cb.pagebrs.push_back(pb);
+          for (const auto& br : lp.backrefs)
+            cb.brs.push_back({dst_abs + br.dst, dst_abs + br.src, br.len, 0});
+        } else if (!lp.resolved.empty()) {
+          // litpar pieces are literal-backed: the host can read any
+          // pattern of <= 8 bytes straight from the compressed stream
+          // and inline it — the resolver then never touches dec sources
+          auto lit_bytes = [&](uint32_t src, uint32_t len,
+                               uint8_t* out) -> bool {
+            const auto& Ls = lp.lits;   // dst-ascending by construction
+            size_t lo = 0, hi = Ls.size();
+            while (lo < hi) {
+              size_t mid = (lo + hi) / 2;
+              if (Ls[mid].dst <= src) lo = mid + 1;
+              else hi = mid;
+            }
+            if (lo == 0) return false;
+            const Lz4Lit& L = Ls[lo - 1];
+            if (src < L.dst || src + len > L.dst + L.len) return false;
+            std::memcpy(out, praw + L.src + (src - L.dst), len);
+            return true;
+          };
+          for (const auto& rr : lp.resolved) {
+            uint32_t pat_len = 0;
+            for (uint32_t k = 0; k < rr.piece_n; k++)
+              pat_len += lp.pieces[rr.piece_start + k].len;
+            if (lp.litpar && pat_len > 0 && pat_len <= 8 && rr.len <= 256) {
+              uint8_t buf[8] = {0};
+              uint32_t o = 0;
+              bool ok = true;
+              for (uint32_t k = 0; k < rr.piece_n && ok; k++) {
+                const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
+                ok = lit_bytes(pc.src, pc.len, buf + o);
+                o += pc.len;
+              }
+              if (ok) {
+                uint64_t pat;
+                std::memcpy(&pat, buf, 8);
+                cb.brinl.push_back({dst_abs + rr.dst, pat, rr.len, pat_len});
+                continue;
+              }
+            }
+            uint32_t ps = (uint32_t)cb.piece_pool.size();
+            for (uint32_t k = 0; k < rr.piece_n; k++) {
+              const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
+              cb.piece_pool.push_back({dst_abs + pc.src, pc.len, 0});
+            }
+            DevBrRes rec{dst_abs + rr.dst, rr.len, rr.off, ps, rr.piece_n};
+            if (rr.len <= 256) cb.res_lane.push_back(rec);
+            else cb.res_wave.push_back(rec);
+          }
+        }
+      };
+
+      // hash-mode dict page: decompress its image into the reserved slot
+      if (c.hash_mode) {
+        for (auto& pi : t.pages) {
+          if (pi.type != PAGE_DICT) continue;
+          plan_decomp(mf.data + pi.payload_off,
+                      t.raw_off + (uint64_t)(pi.payload_off - t.cm->start_offset()),
+                      t.dict_dst, pi.comp_size, pi.uncomp_size,
+                      t.cm->codec == CODEC_UNCOMPRESSED);
+          dec_off += ((uint64_t)pi.uncomp_size + 15) & ~15ull;
+        }
+      }
+
       uint32_t row_in_rg = 0;
+      size_t plain_i4 = 0;
       for (auto& pi : t.pages) {
         if (pi.type != PAGE_DATA) continue;
         DevPage dp{};
@@ -935,138 +1252,35 @@ extern "C" gpuq_plan* gpuq_plan_build(
         dp.optional = mf.meta.columns[mf.meta.col_index(c.name)].optional;
         dp.raw_copy = (t.cm->codec == CODEC_UNCOMPRESSED);
         dp.dict_n = (uint32_t)std::max(
-            {t.remap.size(), t.dictv.size(), t.lut.size()});
+            {t.remap.size(), t.dictv.size(), t.lut.size(), t.strof.size()});
         dp.encoding = (uint8_t)pi.encoding;
         dp.phys = (uint8_t)c.phys;
         const int32_t this_pid = page_id++;
         // host LZ4 structure walk -> parallel segments + backref records
-        {
-          const uint8_t* praw = mf.data + pi.payload_off;
-          bool raw_page = dp.raw_copy != 0;
-          Lz4Plan lp;
-          if (!raw_page) {
-            // test knob: exercise the serial windowed fallback kernel on
-            // arbitrary content (no organic fixture produces a
-            // piece-explosion page)
-            static const bool force_fb =
-                std::getenv("GPUQ_FORCE_LZ4_FALLBACK") != nullptr;
-            try {
-              lp = lz4_walk(praw, pi.comp_size, pi.uncomp_size, 8192);
-              if (force_fb) {
-                lp.fallback = true;
-                lp.resolved.clear();
-                lp.pieces.clear();
-              }
-              // dense short-sequence pages (LZ4 over near-random dict
-              // indices) degenerate the segment kernel into a serial token
-              // parse — switch those to the all-literal/all-resolved plan
-              if (!lp.fallback && lp.n_seq >= 256 &&
-                  (uint32_t)pi.uncomp_size / lp.n_seq < 96) {
-                Lz4Plan lp2 = lz4_walk(praw, pi.comp_size, pi.uncomp_size,
-                                       8192, /*litpar=*/true);
-                if (!lp2.fallback) lp = std::move(lp2);
-              }
-            } catch (const std::exception&) {
-              if (pi.comp_size == pi.uncomp_size) raw_page = true;  // stored raw
-              else throw;
-            }
-          }
-          if (raw_page) {
-            DevSeg sgl{};
-            sgl.src_off = dp.src_off;
-            sgl.dst_off = dp.dst_off;
-            sgl.comp_len = pi.comp_size;
-            sgl.out_len = pi.uncomp_size;
-            sgl.raw = 1;
-            cb.segs.push_back(sgl);
-          } else {
-            for (const auto& sg : lp.segs) {
-              DevSeg d2{};
-              d2.src_off = dp.src_off + sg.s_off;
-              d2.dst_off = dp.dst_off + sg.d_off;
-              d2.comp_len = sg.comp_len;
-              d2.out_len = sg.out_len;
-              d2.big = sg.big;
-              cb.segs.push_back(d2);
-            }
-            for (const auto& lt : lp.lits) {
-              DevLit dl{dp.src_off + lt.src, dp.dst_off + lt.dst, lt.len, 0};
-              (lt.len <= 256 ? cb.lits_lane : cb.lits_wave).push_back(dl);
-            }
-            if (lp.fallback) {
-              // piece explosion: serial windowed wave per page
-              DevPageBr pb{(uint32_t)cb.brs.size(), (uint32_t)lp.backrefs.size()};
-              cb.pagebrs.push_back(pb);
-              for (const auto& br : lp.backrefs)
-                cb.brs.push_back({dp.dst_off + br.dst, dp.dst_off + br.src,
-                                  br.len, 0});
-            } else if (!lp.resolved.empty()) {
-              // litpar pieces are literal-backed: the host can read any
-              // pattern of <= 8 bytes straight from the compressed stream
-              // and inline it — the resolver then never touches dec sources
-              auto lit_bytes = [&](uint32_t src, uint32_t len,
-                                   uint8_t* out) -> bool {
-                const auto& Ls = lp.lits;   // dst-ascending by construction
-                size_t lo = 0, hi = Ls.size();
-                while (lo < hi) {
-                  size_t mid = (lo + hi) / 2;
-                  if (Ls[mid].dst <= src) lo = mid + 1;
-                  else hi = mid;
-                }
-                if (lo == 0) return false;
-                const Lz4Lit& L = Ls[lo - 1];
-                if (src < L.dst || src + len > L.dst + L.len) return false;
-                std::memcpy(out, praw + L.src + (src - L.dst), len);
-                return true;
-              };
-              for (const auto& rr : lp.resolved) {
-                uint32_t pat_len = 0;
-                for (uint32_t k = 0; k < rr.piece_n; k++)
-                  pat_len += lp.pieces[rr.piece_start + k].len;
-                if (lp.litpar && pat_len > 0 && pat_len <= 8 &&
-                    rr.len <= 256) {
-                  uint8_t buf[8] = {0};
-                  uint32_t o = 0;
-                  bool ok = true;
-                  for (uint32_t k = 0; k < rr.piece_n && ok; k++) {
-                    const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
-                    ok = lit_bytes(pc.src, pc.len, buf + o);
-                    o += pc.len;
-                  }
-                  if (ok) {
-                    uint64_t pat;
-                    std::memcpy(&pat, buf, 8);
-                    cb.brinl.push_back(
-                        {dp.dst_off + rr.dst, pat, rr.len, pat_len});
-                    continue;
-                  }
-                }
-                uint32_t ps = (uint32_t)cb.piece_pool.size();
-                for (uint32_t k = 0; k < rr.piece_n; k++) {
-                  const Lz4Piece& pc = lp.pieces[rr.piece_start + k];
-                  cb.piece_pool.push_back({dp.dst_off + pc.src, pc.len, 0});
-                }
-                DevBrRes rec{dp.dst_off + rr.dst, rr.len, rr.off, ps,
-                             rr.piece_n};
-                if (rr.len <= 256) cb.res_lane.push_back(rec);
-                else cb.res_wave.push_back(rec);
-              }
-            }
-          }
-        }
-
+        plan_decomp(mf.data + pi.payload_off, dp.src_off, dp.dst_off,
+                    pi.comp_size, pi.uncomp_size, dp.raw_copy != 0);
         bool dict_enc = (pi.encoding == ENC_RLE_DICT || pi.encoding == ENC_PLAIN_DICT);
         // aux (remap / gid), aux_val (dict values or utf8 sort-ranks) and
         // aux_lut (predicate LUT) are independent slots: one utf8 column
         // can be group key + min/max agg + predicate at once (found by the
         // query fuzzer — aliasing aux produced garbage gids and OOB table
         // writes).
-        if (c.need_gid && dict_enc) {
+        if (c.need_gid && dict_enc && !c.hash_mode) {
           dp.aux = remap_base;
           cb.tasks[{TK_DICT_GID, t.col_idx}].push_back(this_pid);
         }
         if (c.need_val && stubs[ti].need_val_decode) {
-          if (dict_enc) {
+          if (c.hash_mode) {
+            // d_val carries strrefs: dict pages gather the chunk's absolute
+            // entry offsets; PLAIN pages copy the host-walked value offsets
+            if (dict_enc) {
+              dp.aux_val = dictv_base;
+              cb.tasks[{TK_DICT_VAL, t.col_idx}].push_back(this_pid);
+            } else if (pi.encoding == ENC_PLAIN) {
+              dp.aux_val = t.pval_base[plain_i4++];
+              cb.tasks[{TK_POOL_VAL, t.col_idx}].push_back(this_pid);
+            } else throw std::runtime_error("unsupported encoding for hash col");
+          } else if (dict_enc) {
             dp.aux_val = dictv_base;
             cb.tasks[{TK_DICT_VAL, t.col_idx}].push_back(this_pid);
           } else if (pi.encoding == ENC_PLAIN) {
@@ -1075,7 +1289,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
             cb.tasks[{TK_DELTA_VAL, t.col_idx}].push_back(this_pid);
           } else throw std::runtime_error("unsupported encoding for values");
         }
-        if (!c.lut_preds.empty()) {
+        if (!c.lut_preds.empty() && !c.hash_mode) {
+          // (hash-mode columns evaluate every predicate over row strrefs
+          // with k_cmp_str at execute — no LUT/window tasks)
           if (dict_enc) {
             dp.aux_lut = lut_base;
             cb.tasks[{TK_DICT_MASK, t.col_idx}].push_back(this_pid);
@@ -1348,7 +1564,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
     for (auto& ap : plan->aggs) aggs_ok &= (ap.kind == AGGK_COUNT_STAR);
     plan->fused_count = plan->group_cols.size() == 1 && aggs_ok &&
                         plan->preds.empty() && plan->n_groups <= 8192 &&
-                        !plan->cols[plan->group_cols[0]].is_bin;
+                        !plan->cols[plan->group_cols[0]].is_bin &&
+                        !plan->cols[plan->group_cols[0]].hash_mode;
   }
 
   if (getenv("GPUQ_PLAN_DEBUG")) {
@@ -1501,10 +1718,24 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
     HIP_TRY(hipMalloc(&part.d_sort_temp, std::max<size_t>(part.sort_temp_bytes, 16)));
   }
   HIP_TRY(hipMalloc(&part.d_err, 4));
-  size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size()) * 8;
+  // with hash-mode group keys the cardinality is only known at execute:
+  // size the tables for the cap
+  size_t groups_cap = plan->has_hash ? (size_t)GID_CAP : (size_t)plan->n_groups;
+  size_t tsz = groups_cap * (1 + 2 * plan->aggs.size()) * 8;
   HIP_TRY(hipMalloc(&part.d_table, std::max<size_t>(tsz, 16)));
-  size_t fsz = (size_t)plan->n_fsum * plan->n_groups * 4 * 8;
+  size_t fsz = (size_t)plan->n_fsum * groups_cap * 4 * 8;
   HIP_TRY(hipMalloc(&part.d_fsum, std::max<size_t>(fsz, 16)));
+  if (plan->has_hash) {
+    HIP_TRY(hipMalloc(&part.d_hkeys, (1ull << HASH_LOG2) * 8));
+    HIP_TRY(hipMalloc(&part.d_hgids, (1ull << HASH_LOG2) * 4));
+    HIP_TRY(hipMalloc(&part.d_hcount, 4));
+    for (int ci : plan->group_cols)
+      if (plan->cols[ci].hash_mode) {
+        uint64_t* g2r = nullptr;
+        HIP_TRY(hipMalloc(&g2r, (size_t)GID_CAP * 8));
+        part.d_gid2ref[ci] = g2r;
+      }
+  }
   std::vector<int32_t> kinds;
   for (auto& a : plan->aggs) kinds.push_back(a.kind);
   upload_pool(kinds.data(), kinds.size() * 4, (void**)&part.d_agg_kind);
@@ -1538,7 +1769,7 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   {
     std::string pool;
     for (size_t i = 0; i < plan->preds.size(); i++) {
-      if (plan->preds[i].p.op == GPUQ_CONTAINS) {
+      if (plan->preds[i].p.lit_kind == GPUQ_LIT_STR) {
         part.needle_off[i] = (uint32_t)pool.size();
         pool += plan->preds[i].str_lit;
         pool.push_back('\0');
@@ -1650,6 +1881,43 @@ void ss_release(struct ArrowArrayStream* st) {
   st->release = nullptr;
 }
 
+// Fetch the [len][bytes] strings behind a list of strrefs from the device
+// dec arena (export side of the raw-byte utf8 path): lens kernel -> host
+// prefix -> gather kernel -> one packed D2H.
+std::vector<std::string> fetch_ref_strings(Partition& part, hipStream_t st,
+                                           const std::vector<uint64_t>& refs) {
+  int64_t n = (int64_t)refs.size();
+  std::vector<std::string> out(n);
+  if (!n) return out;
+  uint64_t* d_refs = nullptr;
+  uint32_t* d_lens = nullptr;
+  HIP_TRY(hipMalloc(&d_refs, n * 8));
+  HIP_TRY(hipMalloc(&d_lens, n * 4));
+  HIP_TRY(hipMemcpyAsync(d_refs, refs.data(), n * 8, hipMemcpyHostToDevice, st));
+  launch_ref_lens(st, part.d_dec, d_refs, n, d_lens);
+  std::vector<uint32_t> lens(n);
+  HIP_TRY(hipMemcpyAsync(lens.data(), d_lens, n * 4, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  std::vector<uint64_t> offs(n);
+  uint64_t total = 0;
+  for (int64_t i = 0; i < n; i++) { offs[i] = total; total += lens[i]; }
+  uint64_t* d_offs = nullptr;
+  uint8_t* d_out = nullptr;
+  HIP_TRY(hipMalloc(&d_offs, n * 8));
+  HIP_TRY(hipMalloc(&d_out, std::max<uint64_t>(total, 16)));
+  HIP_TRY(hipMemcpyAsync(d_offs, offs.data(), n * 8, hipMemcpyHostToDevice, st));
+  launch_ref_gather(st, part.d_dec, d_refs, d_offs, n, d_out);
+  std::vector<uint8_t> bytes(total);
+  if (total)
+    HIP_TRY(hipMemcpyAsync(bytes.data(), d_out, total, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  for (int64_t i = 0; i < n; i++)
+    out[i].assign((const char*)bytes.data() + offs[i], lens[i]);
+  (void)hipFree(d_refs); (void)hipFree(d_lens);
+  (void)hipFree(d_offs); (void)hipFree(d_out);
+  return out;
+}
+
 // Round the exact 256-bit two's-complement fixed-point sum (lsb weight
 // 2^-160; kernels.hip acc256_*) to the nearest double, ties to even —
 // the once-per-query rounding that makes f64 SUM order-independent and
@@ -1711,6 +1979,7 @@ int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
   std::vector<std::vector<int64_t>> vals(np);
   std::vector<std::vector<int32_t>> gids(np);
   std::vector<std::vector<uint8_t>> valids(np);
+  std::vector<std::vector<std::string>> pstrs(np);  // hash-mode utf8 cols
   int64_t* d_g64 = nullptr;
   int32_t* d_g32 = nullptr;
   uint8_t* d_g8 = nullptr;
@@ -1720,7 +1989,30 @@ int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
   for (int p = 0; p < np; p++) {
     int ci = plan->projection[p];
     auto& c = plan->cols[ci];
-    if (c.phys == PT_BYTE_ARRAY) {
+    if (c.phys == PT_BYTE_ARRAY && c.hash_mode) {
+      // raw-byte utf8: gather the winners' strrefs + validity, then fetch
+      // the bytes from the arena
+      std::vector<int64_t> refs(k);
+      valids[p].assign(k, 1);
+      launch_gather_i64(st, part.d_rows_sorted, k, part.d_val[ci], d_g64);
+      HIP_TRY(hipMemcpyAsync(refs.data(), d_g64, k * 8,
+                             hipMemcpyDeviceToHost, st));
+      auto itv = part.d_valid.find(ci);
+      if (itv != part.d_valid.end()) {
+        launch_gather_u8(st, part.d_rows_sorted, k, itv->second, d_g8);
+        HIP_TRY(hipMemcpyAsync(valids[p].data(), d_g8, k,
+                               hipMemcpyDeviceToHost, st));
+      }
+      HIP_TRY(hipStreamSynchronize(st));
+      std::vector<uint64_t> vr;
+      std::vector<int64_t> vrow;
+      for (int64_t r = 0; r < k; r++)
+        if (valids[p][r]) { vr.push_back((uint64_t)refs[r]); vrow.push_back(r); }
+      auto strs = fetch_ref_strings(part, st, vr);
+      pstrs[p].assign(k, std::string());
+      for (size_t j = 0; j < vrow.size(); j++)
+        pstrs[p][vrow[j]] = std::move(strs[j]);
+    } else if (c.phys == PT_BYTE_ARRAY) {
       gids[p].resize(k);
       launch_gather_i32(st, part.d_rows_sorted, k, part.d_gid[ci], d_g32);
       HIP_TRY(hipMemcpyAsync(gids[p].data(), d_g32, k * 4,
@@ -1788,25 +2080,28 @@ int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
     ch->release = release_array;
     auto& c = plan->cols[plan->projection[p]];
     if (c.phys == PT_BYTE_ARRAY) {
+      bool via_ref = c.hash_mode;
+      auto str_at = [&](int64_t r) -> const std::string* {
+        if (via_ref) return valids[p][r] ? &pstrs[p][r] : nullptr;
+        int32_t g = gids[p][r];
+        return g > 0 ? &c.gdict[g - 1] : nullptr;
+      };
       ch->n_buffers = 3;
       ch->buffers = (const void**)calloc(3, sizeof(void*));
       uint8_t* validity = (uint8_t*)eb->grab((k + 7) / 8);
       memset(validity, 0xff, std::max<int64_t>((k + 7) / 8, 1));
       int32_t* offs = (int32_t*)eb->grab((k + 1) * 4);
       size_t total = 0;
-      for (int64_t r = 0; r < k; r++) {
-        int32_t g = gids[p][r];
-        if (g > 0) total += c.gdict[g - 1].size();
-      }
+      for (int64_t r = 0; r < k; r++)
+        if (const std::string* sp = str_at(r)) total += sp->size();
       char* data = (char*)eb->grab(total);
       size_t off = 0;
       int64_t nulls = 0;
       for (int64_t r = 0; r < k; r++) {
         offs[r] = (int32_t)off;
-        int32_t g = gids[p][r];
-        if (g > 0) {
-          memcpy(data + off, c.gdict[g - 1].data(), c.gdict[g - 1].size());
-          off += c.gdict[g - 1].size();
+        if (const std::string* sp = str_at(r)) {
+          memcpy(data + off, sp->data(), sp->size());
+          off += sp->size();
         } else {
           validity[r / 8] &= (uint8_t)~(1 << (r % 8));
           nulls++;
@@ -1874,11 +2169,12 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   const bool need_mask = !plan->preds.empty() && !part.pred_ranges.empty();
   HIP_TRY(hipMemsetAsync(part.d_err, 0, 4, st));
   if (need_mask) HIP_TRY(hipMemsetAsync(part.d_mask, 1, part.n_rows, st));
-  launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
-                    part.d_agg_kind);
-  if (plan->n_fsum)
-    HIP_TRY(hipMemsetAsync(part.d_fsum, 0,
-                           (size_t)plan->n_fsum * plan->n_groups * 4 * 8, st));
+  // group count with hash keys is only known after the hash build, so the
+  // table init moves to just before aggregation; n_groups_exec tracks it
+  int64_t n_groups_exec = plan->n_groups;
+  if (plan->fused_count)
+    launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
+                      part.d_agg_kind);
 
   HIP_TRY(hipEventRecord(ev0, st));
   // 1. decompress: parallel segments, then ordered backref resolution
@@ -1960,6 +2256,18 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
                       part.d_rank, part.d_valid[col],
                       (uint8_t*)part.d_val[col], 0);
         break;
+      case TK_POOL_VAL:  // PLAIN byte-array pages of hash cols -> strrefs
+        launch_def_levels(st, part.d_dec, part.d_pages, ids, n,
+                          part.d_valid[col], nullptr, nullptr,
+                          part.d_rank, part.d_present, part.d_err);
+        launch_pool_vals(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
+                         part.d_val[col], part.d_valid[col], part.d_present, 0);
+        launch_pool_vals(st, part.d_dec, part.d_pages, ids, n, part.d_dictv,
+                         (int64_t*)part.d_scr, nullptr, part.d_present, 1);
+        launch_expand(st, part.d_dec, part.d_pages, ids, n, part.d_scr,
+                      part.d_rank, part.d_valid[col],
+                      (uint8_t*)part.d_val[col], 0);
+        break;
       case TK_DELTA_VAL:
         launch_delta_i64(st, part.d_dec, part.d_pages, ids, n,
                          part.d_val[col], part.d_valid[col], part.d_err);
@@ -1994,6 +2302,66 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       }
     }
   }
+  // raw-byte utf8: device hash build assigns dense gids per hash-mode
+  // group key (one shared table, rebuilt per column); string predicates
+  // evaluate over the row strrefs
+  if (plan->has_hash) {
+    for (int ci : plan->group_cols) {
+      auto& c = plan->cols[ci];
+      if (!c.hash_mode) continue;
+      HIP_TRY(hipMemsetAsync(part.d_hkeys, 0xFF, (1ull << HASH_LOG2) * 8, st));
+      HIP_TRY(hipMemsetAsync(part.d_hgids, 0xFF, (1ull << HASH_LOG2) * 4, st));
+      HIP_TRY(hipMemsetAsync(part.d_hcount, 0, 4, st));
+      auto itv = part.d_valid.find(ci);
+      uint8_t* v = itv != part.d_valid.end() ? itv->second : nullptr;
+      launch_hash_build(st, part.d_dec, part.d_val[ci], v, part.n_rows,
+                        part.d_hkeys, part.d_hgids, HASH_LOG2, part.d_hcount,
+                        part.d_gid2ref[ci], GID_CAP, part.d_err);
+      launch_hash_lookup(st, part.d_dec, part.d_val[ci], v, part.n_rows,
+                         part.d_hkeys, part.d_hgids, HASH_LOG2,
+                         part.d_gid[ci]);
+      uint32_t claimed = 0;
+      HIP_TRY(hipMemcpyAsync(&claimed, part.d_hcount, 4,
+                             hipMemcpyDeviceToHost, st));
+      HIP_TRY(hipStreamSynchronize(st));
+      part.hash_claimed[ci] = claimed;
+    }
+    int64_t g2 = 1;
+    for (int ci : plan->group_cols) {
+      auto& c = plan->cols[ci];
+      g2 *= c.is_bin ? (int64_t)c.nbins + 1
+            : c.hash_mode ? (int64_t)part.hash_claimed[ci] + 1
+                          : (int64_t)c.gdict.size() + 1;
+    }
+    if (g2 > (int64_t)GID_CAP)
+      throw std::runtime_error(
+          "group-key cardinality product too large (hash keys)");
+    n_groups_exec = g2;
+    for (size_t ci = 0; ci < plan->cols.size(); ci++) {
+      auto& c = plan->cols[ci];
+      if (!c.hash_mode || c.lut_preds.empty()) continue;
+      for (int pidx : c.lut_preds) {
+        const auto& pp = plan->preds[pidx];
+        int op;
+        switch (pp.p.op) {
+          case GPUQ_CONTAINS: op = -1; break;
+          case GPUQ_EQ: op = CMP_EQ; break;
+          case GPUQ_NE: op = CMP_NE; break;
+          case GPUQ_LT: op = CMP_LT; break;
+          case GPUQ_LE: op = CMP_LE; break;
+          case GPUQ_GT: op = CMP_GT; break;
+          default: op = CMP_GE; break;
+        }
+        auto itv = part.d_valid.find((int)ci);
+        launch_cmp_str(st, part.d_dec, part.d_val[(int)ci],
+                       itv != part.d_valid.end() ? itv->second : nullptr,
+                       part.d_needle + part.needle_off[pidx],
+                       (uint32_t)pp.str_lit.size(), op, part.d_mask,
+                       part.n_rows);
+      }
+    }
+  }
+
   // i64 comparisons on decoded arrays
   for (size_t ci = 0; ci < plan->cols.size(); ci++) {
     auto& c = plan->cols[ci];
@@ -2050,7 +2418,9 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     int ci = plan->group_cols[k];
     a.key_gid[k] = part.d_gid[ci];
     const auto& kc = plan->cols[ci];
-    a.key_size[k] = kc.is_bin ? kc.nbins + 1 : (int32_t)kc.gdict.size() + 1;
+    a.key_size[k] = kc.is_bin ? kc.nbins + 1
+                    : kc.hash_mode ? (int32_t)part.hash_claimed[ci] + 1
+                                   : (int32_t)kc.gdict.size() + 1;
   }
   a.n_aggs = (int)plan->aggs.size();
   for (int i = 0; i < a.n_aggs; i++) {
@@ -2071,18 +2441,24 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   a.fsum_n = plan->n_fsum;
   a.fsum = part.d_fsum;
   a.err = part.d_err;
+  a.dec = part.d_dec;
   a.table = part.d_table;
-  a.n_groups = plan->n_groups;
+  a.n_groups = (int32_t)n_groups_exec;
+  launch_init_table(st, part.d_table, (int32_t)n_groups_exec,
+                    (int)plan->aggs.size(), part.d_agg_kind);
+  if (plan->n_fsum)
+    HIP_TRY(hipMemsetAsync(part.d_fsum, 0,
+                           (size_t)plan->n_fsum * n_groups_exec * 4 * 8, st));
   launch_agg(st, a);
   HIP_TRY(hipEventRecord(ev1, st));
   }  // !fused_count
 
   // 4. D2H results
-  size_t tsz = (size_t)plan->n_groups * (1 + 2 * plan->aggs.size());
+  size_t tsz = (size_t)n_groups_exec * (1 + 2 * plan->aggs.size());
   std::vector<uint64_t> table(tsz);
   HIP_TRY(hipMemcpyAsync(table.data(), part.d_table, tsz * 8,
                          hipMemcpyDeviceToHost, st));
-  std::vector<uint64_t> fsums((size_t)plan->n_fsum * plan->n_groups * 4);
+  std::vector<uint64_t> fsums((size_t)plan->n_fsum * n_groups_exec * 4);
   if (!fsums.empty())
     HIP_TRY(hipMemcpyAsync(fsums.data(), part.d_fsum, fsums.size() * 8,
                            hipMemcpyDeviceToHost, st));
@@ -2104,7 +2480,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   int n_aggs = (int)plan->aggs.size();
   int slots = 1 + 2 * n_aggs;
   std::vector<int64_t> live;  // group ids with presence > 0
-  for (int64_t g = 0; g < plan->n_groups; g++)
+  for (int64_t g = 0; g < n_groups_exec; g++)
     if (table[(size_t)g * slots]) live.push_back(g);
   int64_t nr = (int64_t)live.size();
   // no-group aggregate over empty selection: emit the single empty row
@@ -2135,7 +2511,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     int ki = plan->aggs[i].kind;
     const char* fmt = (ki == AGGK_SUM_F64 || ki == AGGK_MIN_F64 || ki == AGGK_MAX_F64)
                           ? "g"
-                      : (ki == AGGK_MIN_RANK || ki == AGGK_MAX_RANK) ? "u" : "l";
+                      : (ki == AGGK_MIN_RANK || ki == AGGK_MAX_RANK ||
+                         ki == AGGK_MIN_STR || ki == AGGK_MAX_STR) ? "u" : "l";
     make_schema_field(ss->schema.children[f], fmt, "agg" + std::to_string(i));
     f++;
     ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
@@ -2167,10 +2544,39 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     int64_t g = live[r];
     for (int k = n_keys - 1; k >= 0; k--) {
       const auto& kc = plan->cols[plan->group_cols[k]];
-      int32_t sz = kc.is_bin ? kc.nbins + 1 : (int32_t)kc.gdict.size() + 1;
+      int32_t sz = kc.is_bin ? kc.nbins + 1
+                   : kc.hash_mode ? (int32_t)part.hash_claimed[plan->group_cols[k]] + 1
+                                  : (int32_t)kc.gdict.size() + 1;
       key_gids[k][r] = (int32_t)(g % sz);
       g /= sz;
     }
+  }
+  // hash-mode key columns: fetch the strings for the local gids that
+  // actually appear (gid -> gid2ref -> dec-arena bytes)
+  std::map<int, std::unordered_map<int32_t, std::string>> hash_names;
+  for (int k = 0; k < n_keys; k++) {
+    int ci = plan->group_cols[k];
+    const auto& kc = plan->cols[ci];
+    if (!kc.hash_mode) continue;
+    std::vector<int32_t> need;
+    {
+      std::unordered_map<int32_t, char> seen;
+      for (int64_t r = 0; r < nr; r++) {
+        int32_t gid = key_gids[k][r];
+        if (gid > 0 && !seen.count(gid)) { seen[gid] = 1; need.push_back(gid); }
+      }
+    }
+    uint32_t claimed = part.hash_claimed[ci];
+    std::vector<uint64_t> g2r(claimed);
+    if (claimed)
+      HIP_TRY(hipMemcpy(g2r.data(), part.d_gid2ref[ci], (size_t)claimed * 8,
+                        hipMemcpyDeviceToHost));
+    std::vector<uint64_t> refs;
+    refs.reserve(need.size());
+    for (int32_t gid : need) refs.push_back(g2r[(size_t)gid - 1]);
+    auto strs = fetch_ref_strings(part, st, refs);
+    auto& m = hash_names[ci];
+    for (size_t i2 = 0; i2 < need.size(); i2++) m[need[i2]] = std::move(strs[i2]);
   }
   f = 0;
   for (int k = 0; k < n_keys; k++) {
@@ -2203,10 +2609,14 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
     memset(validity, 0, (nr + 7) / 8);
     int32_t* offs = (int32_t*)eb->grab((nr + 1) * 4);
+    auto key_str = [&](int32_t gid) -> const std::string& {
+      return c.hash_mode ? hash_names[plan->group_cols[k]][gid]
+                         : c.gdict[gid - 1];
+    };
     size_t total = 0;
     for (int64_t r = 0; r < nr; r++) {
       int32_t gid = key_gids[k][r];
-      if (gid > 0) total += c.gdict[gid - 1].size();
+      if (gid > 0) total += key_str(gid).size();
     }
     char* data = (char*)eb->grab(total);
     size_t off = 0;
@@ -2216,7 +2626,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       int32_t gid = key_gids[k][r];
       if (gid > 0) {
         validity[r / 8] |= (uint8_t)(1 << (r % 8));
-        const auto& s = c.gdict[gid - 1];
+        const auto& s = key_str(gid);
         memcpy(data + off, s.data(), s.size());
         off += s.size();
       } else nulls++;
@@ -2244,18 +2654,41 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   for (int i = 0; i < n_aggs; i++) {
     auto* chv = make_child(f++);
     int kind = plan->aggs[i].kind;
-    bool is_rank = (kind == AGGK_MIN_RANK || kind == AGGK_MAX_RANK);
+    bool is_rank = (kind == AGGK_MIN_RANK || kind == AGGK_MAX_RANK ||
+                    kind == AGGK_MIN_STR || kind == AGGK_MAX_STR);
     uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
     memset(validity, 0xff, (nr + 7) / 8);
     int64_t nulls = 0;
     if (is_rank) {
-      // utf8 value column: rank -> dictionary string
+      // utf8 value column: rank -> dictionary string, or (hash mode)
+      // strref -> dec-arena bytes
       const auto& c = plan->cols[plan->aggs[i].col_idx];
+      bool via_ref = (kind == AGGK_MIN_STR || kind == AGGK_MAX_STR);
+      std::vector<std::string> ref_strs;
+      std::vector<int64_t> ref_rows;
+      if (via_ref) {
+        std::vector<uint64_t> refs;
+        for (int64_t r = 0; r < nr; r++) {
+          size_t base = (size_t)live[r] * slots;
+          uint64_t cnt = empty_aggregate_row ? 0
+                         : (plan->aggs[i].cnt_is_presence ? table[base]
+                                                          : table[base + 2 + 2 * i]);
+          uint64_t ref = table[base + 1 + 2 * i];
+          if (cnt && ref != ~0ull) { refs.push_back(ref); ref_rows.push_back(r); }
+        }
+        ref_strs = fetch_ref_strings(part, st, refs);
+      }
       chv->n_buffers = 3;
       chv->buffers = (const void**)calloc(3, sizeof(void*));
       int32_t* offs = (int32_t*)eb->grab((nr + 1) * 4);
       size_t total = 0;
       std::vector<const std::string*> strs(nr, nullptr);
+      if (via_ref) {
+        for (size_t j = 0; j < ref_rows.size(); j++) {
+          strs[ref_rows[j]] = &ref_strs[j];
+          total += ref_strs[j].size();
+        }
+      } else
       for (int64_t r = 0; r < nr; r++) {
         size_t base = (size_t)live[r] * slots;
         uint64_t cnt = empty_aggregate_row ? 0
@@ -2320,7 +2753,7 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       } else if (kind == AGGK_SUM_F64) {
         // exact 256-bit superaccumulator, rounded once (acc256_to_double)
         double d = acc256_to_double(
-            &fsums[((size_t)fsi * plan->n_groups + (size_t)live[r]) * 4]);
+            &fsums[((size_t)fsi * n_groups_exec + (size_t)live[r]) * 4]);
         memcpy(&vv[r], &d, 8);
       } else {
         vv[r] = (int64_t)val;
@@ -2390,7 +2823,9 @@ gpuq_plan::~gpuq_plan() {
     auto F = [](void* p) { if (p) (void)hipFree(p); };
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
-    F(part.d_table); F(part.d_fsum); F(part.d_agg_kind); F(part.d_needle); F(part.d_all_ids);
+    F(part.d_table); F(part.d_fsum); F(part.d_agg_kind);
+    F(part.d_hkeys); F(part.d_hgids); F(part.d_hcount);
+    for (auto& kv : part.d_gid2ref) F(kv.second); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_rank); F(part.d_scr);
     F(part.d_present); F(part.d_tmpvalid);
     F(part.d_lits_lane); F(part.d_lits_wave);

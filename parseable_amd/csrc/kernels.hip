@@ -938,6 +938,227 @@ k_expand(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
 }
 
 // ------------------------------------------------------------------
+// Raw-byte utf8 machinery (hash group-by / predicates / min-max over
+// columns with PLAIN-fallback pages — the dict-only gid path cannot cover
+// them). A row's string is a STRREF: the absolute dec-arena offset of its
+// [u32 len][bytes] record — dict entries and PLAIN values both live in the
+// arena (dict pages of hash columns are decompressed on device too).
+// ------------------------------------------------------------------
+
+// PLAIN byte-array pages: the host walked the [len][bytes] chain at plan
+// time and left one absolute strref per (non-null) value in the pool;
+// emit them row-aligned (mode 0 direct / mode 1 dense-scratch + k_expand,
+// exactly the k_plain_fixed contract).
+__global__ void __launch_bounds__(WAVE)
+k_pool_vals(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
+            const int32_t* __restrict__ ids, int n,
+            const int64_t* __restrict__ pool,
+            int64_t* __restrict__ out, uint8_t* __restrict__ valid,
+            const uint32_t* __restrict__ present, int mode) {
+  int pi = blockIdx.x;
+  if (pi >= n) return;
+  const DevPage pg = pages[ids[pi]];
+  const int lane = threadIdx.x;
+  const uint8_t* def_start; uint32_t def_len; bool all_valid;
+  def_levels(pg, dec + pg.dst_off, &def_start, &def_len, &all_valid);
+  const uint32_t row0 = pg.row_start;
+  const int64_t* src = pool + pg.aux_val;
+  if (mode == 0) {
+    if (!all_valid) return;
+    for (uint32_t i = lane; i < pg.num_values; i += WAVE) {
+      out[row0 + i] = src[i];
+      if (valid) valid[row0 + i] = 1;
+    }
+  } else {
+    if (all_valid) return;
+    uint32_t nd = present[ids[pi]];
+    for (uint32_t i = lane; i < nd; i += WAVE) out[row0 + i] = src[i];
+  }
+}
+
+__device__ inline uint32_t ref_len(const uint8_t* dec, uint64_t ref) {
+  uint32_t l;
+  __builtin_memcpy(&l, dec + ref, 4);
+  return l;
+}
+__device__ inline bool ref_eq(const uint8_t* dec, uint64_t a, uint64_t b) {
+  if (a == b) return true;
+  uint32_t la = ref_len(dec, a), lb = ref_len(dec, b);
+  if (la != lb) return false;
+  const uint8_t* pa = dec + a + 4;
+  const uint8_t* pb = dec + b + 4;
+  uint32_t i = 0;
+  for (; i + 8 <= la; i += 8) {
+    uint64_t wa, wb;
+    __builtin_memcpy(&wa, pa + i, 8);
+    __builtin_memcpy(&wb, pb + i, 8);
+    if (wa != wb) return false;
+  }
+  for (; i < la; i++)
+    if (pa[i] != pb[i]) return false;
+  return true;
+}
+// lexicographic byte order (utf8 min/max semantics): a < b
+__device__ inline int ref_cmp(const uint8_t* dec, uint64_t a, uint64_t b) {
+  if (a == b) return 0;
+  uint32_t la = ref_len(dec, a), lb = ref_len(dec, b);
+  const uint8_t* pa = dec + a + 4;
+  const uint8_t* pb = dec + b + 4;
+  uint32_t n = la < lb ? la : lb;
+  for (uint32_t i = 0; i < n; i++) {
+    if (pa[i] != pb[i]) return pa[i] < pb[i] ? -1 : 1;
+  }
+  return la == lb ? 0 : (la < lb ? -1 : 1);
+}
+
+__device__ inline uint64_t ref_hash(const uint8_t* dec, uint64_t ref) {
+  uint32_t len = ref_len(dec, ref);
+  const uint8_t* p = dec + ref + 4;
+  uint64_t h = 0xcbf29ce484222325ull ^ len;
+  uint32_t i = 0;
+  for (; i + 8 <= len; i += 8) {
+    uint64_t w;
+    __builtin_memcpy(&w, p + i, 8);
+    h = (h ^ w) * 0x100000001b3ull;
+    h ^= h >> 29;
+  }
+  uint64_t tail = 0;
+  for (uint32_t k = 0; i < len; i++, k += 8) tail |= (uint64_t)p[i] << k;
+  h = (h ^ tail) * 0x100000001b3ull;
+  h ^= h >> 32;
+  return h;
+}
+
+#define HREF_EMPTY (~0ull)
+
+// pass 1: claim distinct strings. Open addressing, linear probe; the slot
+// key is the FIRST strref seen for the string (byte-equality dedups dict
+// entries vs PLAIN occurrences of the same value). Claimers draw a dense
+// gid from *counter and record gid2ref; hgids stores are completed before
+// pass 2 runs (kernel boundary), so lookup never spins.
+__global__ void k_hash_build(const uint8_t* __restrict__ dec,
+                             const int64_t* __restrict__ refs,
+                             const uint8_t* __restrict__ valid, int64_t n_rows,
+                             uint64_t* __restrict__ hkeys,
+                             int32_t* __restrict__ hgids, int clog2,
+                             uint32_t* counter, uint64_t* __restrict__ gid2ref,
+                             int32_t gid_cap, int32_t* d_error) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const uint64_t mask = (1ull << clog2) - 1;
+  for (; i < n_rows; i += stride) {
+    if (valid && !valid[i]) continue;
+    uint64_t ref = (uint64_t)refs[i];
+    uint64_t slot = ref_hash(dec, ref) & mask;
+    for (uint32_t probe = 0; ; probe++) {
+      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+      uint64_t old = atomicCAS((unsigned long long*)&hkeys[slot],
+                               (unsigned long long)HREF_EMPTY,
+                               (unsigned long long)ref);
+      if (old == HREF_EMPTY) {
+        uint32_t g = atomicAdd(counter, 1u);
+        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+        gid2ref[g] = ref;
+        hgids[slot] = (int32_t)g;
+        break;
+      }
+      if (ref_eq(dec, old, ref)) break;
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+
+// pass 2: resolve every row's gid (1-based; 0 = NULL group)
+__global__ void k_hash_lookup(const uint8_t* __restrict__ dec,
+                              const int64_t* __restrict__ refs,
+                              const uint8_t* __restrict__ valid, int64_t n_rows,
+                              const uint64_t* __restrict__ hkeys,
+                              const int32_t* __restrict__ hgids, int clog2,
+                              int32_t* __restrict__ out_gid) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const uint64_t mask = (1ull << clog2) - 1;
+  for (; i < n_rows; i += stride) {
+    if (valid && !valid[i]) { out_gid[i] = 0; continue; }
+    uint64_t ref = (uint64_t)refs[i];
+    uint64_t slot = ref_hash(dec, ref) & mask;
+    for (;;) {
+      uint64_t k = hkeys[slot];
+      if (k == HREF_EMPTY) { out_gid[i] = 0; break; }  // unreachable
+      if (ref_eq(dec, k, ref)) { out_gid[i] = hgids[slot] + 1; break; }
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+
+// string predicates over row strrefs (mixed dict/PLAIN chunks where the
+// per-dict-entry LUT cannot cover the PLAIN pages). op: CmpMode, or -1 for
+// CONTAINS (byte substring).
+__global__ void k_cmp_str(const uint8_t* __restrict__ dec,
+                          const int64_t* __restrict__ refs,
+                          const uint8_t* __restrict__ valid,
+                          const uint8_t* __restrict__ lit, uint32_t lit_len,
+                          int op, uint8_t* __restrict__ mask, int64_t n_rows) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n_rows; i += stride) {
+    if (valid && !valid[i]) { mask[i] = 0; continue; }
+    uint64_t ref = (uint64_t)refs[i];
+    uint32_t len = ref_len(dec, ref);
+    const uint8_t* s = dec + ref + 4;
+    bool ok;
+    if (op < 0) {  // CONTAINS
+      ok = false;
+      if (lit_len == 0) ok = true;
+      else if (len >= lit_len) {
+        uint8_t c0 = lit[0];
+        for (uint32_t j = 0; j + lit_len <= len && !ok; j++) {
+          if (s[j] != c0) continue;
+          uint32_t k = 1;
+          while (k < lit_len && s[j + k] == lit[k]) k++;
+          ok = (k == lit_len);
+        }
+      }
+    } else {
+      uint32_t n = len < lit_len ? len : lit_len;
+      int c = 0;
+      for (uint32_t j = 0; j < n && !c; j++)
+        c = s[j] < lit[j] ? -1 : (s[j] > lit[j] ? 1 : 0);
+      if (!c) c = len == lit_len ? 0 : (len < lit_len ? -1 : 1);
+      switch (op) {
+        case CMP_EQ: ok = c == 0; break;
+        case CMP_NE: ok = c != 0; break;
+        case CMP_LT: ok = c < 0; break;
+        case CMP_LE: ok = c <= 0; break;
+        case CMP_GT: ok = c > 0; break;
+        default: ok = c >= 0; break;
+      }
+    }
+    if (!ok) mask[i] = 0;
+  }
+}
+
+// export-side string fetch: lengths then packed bytes for a strref list
+__global__ void k_ref_lens(const uint8_t* __restrict__ dec,
+                           const uint64_t* __restrict__ refs, int64_t n,
+                           uint32_t* __restrict__ lens) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) lens[i] = ref_len(dec, refs[i]);
+}
+__global__ void k_ref_gather(const uint8_t* __restrict__ dec,
+                             const uint64_t* __restrict__ refs,
+                             const uint64_t* __restrict__ offs, int64_t n,
+                             uint8_t* __restrict__ out) {
+  int64_t i = blockIdx.x;
+  if (i >= n) return;
+  uint64_t ref = refs[i];
+  uint32_t len = ref_len(dec, ref);
+  const uint8_t* s = dec + ref + 4;
+  uint8_t* d = out + offs[i];
+  for (uint32_t j = threadIdx.x; j < len; j += blockDim.x) d[j] = s[j];
+}
+
+// ------------------------------------------------------------------
 // DELTA_BINARY_PACKED i64 -> row-aligned (one wave per page).
 // Phase A: lane-redundant block-header walk storing per-miniblock
 //   (data offset, bit width, block min_delta ref); Phase B: parallel
@@ -1479,6 +1700,26 @@ __device__ inline void acc256_add(uint64_t* l, const uint64_t a[4]) {
   }
 }
 
+// utf8 min/max over strrefs (hash-mode columns): CAS keep-the-winner loop
+// with lexicographic byte compare against the dec arena. Converges like the
+// numeric CAS min/max — after the winner settles, the compare short-circuits
+// and no atomic issues.
+__device__ inline void atomic_minmax_str(const uint8_t* dec, uint64_t* addr,
+                                         uint64_t ref, bool want_min) {
+  uint64_t old = *addr;
+  for (;;) {
+    if (old != HREF_EMPTY) {
+      int c = ref_cmp(dec, ref, old);
+      if (want_min ? (c >= 0) : (c <= 0)) break;
+    }
+    uint64_t prev = atomicCAS((unsigned long long*)addr,
+                              (unsigned long long)old,
+                              (unsigned long long)ref);
+    if (prev == old) break;
+    old = prev;
+  }
+}
+
 template <bool USE_LDS>
 __global__ void __launch_bounds__(256)
 k_agg(AggArgs a) {
@@ -1502,6 +1743,7 @@ k_agg(AggArgs a) {
           else if (k == AGGK_MAX_I64 || k == AGGK_MAX_RANK) init = (uint64_t)INT64_MIN;
           else if (k == AGGK_MIN_F64) init = (uint64_t)0x7ff0000000000000ull;   // +inf
           else if (k == AGGK_MAX_F64) init = (uint64_t)0xfff0000000000000ull;   // -inf
+          else if (k == AGGK_MIN_STR || k == AGGK_MAX_STR) init = HREF_EMPTY;
         }
       }
       lt[i] = init;
@@ -1551,6 +1793,8 @@ k_agg(AggArgs a) {
         case AGGK_MAX_I64: case AGGK_MAX_RANK: atomic_max_i64(vs, v); break;
         case AGGK_MIN_F64: atomic_min_f64(vs, __longlong_as_double((long long)v)); break;
         case AGGK_MAX_F64: atomic_max_f64(vs, __longlong_as_double((long long)v)); break;
+        case AGGK_MIN_STR: atomic_minmax_str(a.dec, vs, (uint64_t)v, true); break;
+        case AGGK_MAX_STR: atomic_minmax_str(a.dec, vs, (uint64_t)v, false); break;
       }
       if (!a.cnt_skip[ai])
         atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
@@ -1587,6 +1831,12 @@ k_agg(AggArgs a) {
           if (d != -__builtin_inf()) atomic_max_f64(g, d);
           break;
         }
+        case AGGK_MIN_STR:
+          if (v != HREF_EMPTY) atomic_minmax_str(a.dec, g, v, true);
+          break;
+        case AGGK_MAX_STR:
+          if (v != HREF_EMPTY) atomic_minmax_str(a.dec, g, v, false);
+          break;
       }
     }
     // flush the block's exact superaccumulators (limb-wise with carries)
@@ -1616,6 +1866,7 @@ __global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
       else if (k == AGGK_MAX_I64 || k == AGGK_MAX_RANK) init = (uint64_t)INT64_MIN;
       else if (k == AGGK_MIN_F64) init = 0x7ff0000000000000ull;
       else if (k == AGGK_MAX_F64) init = 0xfff0000000000000ull;
+      else if (k == AGGK_MIN_STR || k == AGGK_MAX_STR) init = ~0ull;
     }
     table[i] = init;
   }
@@ -1779,6 +2030,54 @@ void launch_agg(hipStream_t st, const AggArgs& a) {
   } else {
     hipLaunchKernelGGL(k_agg<false>, dim3(blocks), dim3(256), 0, st, a);
   }
+}
+
+void launch_pool_vals(hipStream_t st, const uint8_t* dec, const DevPage* pages,
+                      const int32_t* ids, int n, const int64_t* pool,
+                      int64_t* out, uint8_t* valid, const uint32_t* present,
+                      int mode) {
+  if (n) hipLaunchKernelGGL(k_pool_vals, dim3(n), dim3(WAVE), 0, st, dec,
+                            pages, ids, n, pool, out, valid, present, mode);
+}
+void launch_hash_build(hipStream_t st, const uint8_t* dec, const int64_t* refs,
+                       const uint8_t* valid, int64_t n_rows, uint64_t* hkeys,
+                       int32_t* hgids, int clog2, uint32_t* counter,
+                       uint64_t* gid2ref, int32_t gid_cap, int32_t* d_err) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_hash_build, dim3(blocks), dim3(256), 0, st, dec, refs,
+                     valid, n_rows, hkeys, hgids, clog2, counter, gid2ref,
+                     gid_cap, d_err);
+}
+void launch_hash_lookup(hipStream_t st, const uint8_t* dec, const int64_t* refs,
+                        const uint8_t* valid, int64_t n_rows,
+                        const uint64_t* hkeys, const int32_t* hgids, int clog2,
+                        int32_t* out_gid) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_hash_lookup, dim3(blocks), dim3(256), 0, st, dec, refs,
+                     valid, n_rows, hkeys, hgids, clog2, out_gid);
+}
+void launch_cmp_str(hipStream_t st, const uint8_t* dec, const int64_t* refs,
+                    const uint8_t* valid, const uint8_t* lit, uint32_t lit_len,
+                    int op, uint8_t* mask, int64_t n_rows) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_cmp_str, dim3(blocks), dim3(256), 0, st, dec, refs,
+                     valid, lit, lit_len, op, mask, n_rows);
+}
+void launch_ref_lens(hipStream_t st, const uint8_t* dec, const uint64_t* refs,
+                     int64_t n, uint32_t* lens) {
+  if (n) hipLaunchKernelGGL(k_ref_lens, dim3((int)((n + 255) / 256)), dim3(256),
+                            0, st, dec, refs, n, lens);
+}
+void launch_ref_gather(hipStream_t st, const uint8_t* dec, const uint64_t* refs,
+                       const uint64_t* offs, int64_t n, uint8_t* out) {
+  if (n) hipLaunchKernelGGL(k_ref_gather, dim3((int)n), dim3(64), 0, st, dec,
+                            refs, offs, n, out);
 }
 
 // device radix sort of (key,row) pairs, descending (AMD rocPRIM — native

@@ -63,6 +63,25 @@ void launch_dict_count(hipStream_t, const uint8_t* dec, const DevPage*,
                        uint64_t* table, int32_t n_groups, int n_aggs,
                        int32_t* d_err);
 void launch_agg(hipStream_t, const AggArgs&);
+void launch_pool_vals(hipStream_t, const uint8_t* dec, const DevPage*,
+                      const int32_t* ids, int n, const int64_t* pool,
+                      int64_t* out, uint8_t* valid, const uint32_t* present,
+                      int mode);
+void launch_hash_build(hipStream_t, const uint8_t* dec, const int64_t* refs,
+                       const uint8_t* valid, int64_t n_rows, uint64_t* hkeys,
+                       int32_t* hgids, int clog2, uint32_t* counter,
+                       uint64_t* gid2ref, int32_t gid_cap, int32_t* d_err);
+void launch_hash_lookup(hipStream_t, const uint8_t* dec, const int64_t* refs,
+                        const uint8_t* valid, int64_t n_rows,
+                        const uint64_t* hkeys, const int32_t* hgids, int clog2,
+                        int32_t* out_gid);
+void launch_cmp_str(hipStream_t, const uint8_t* dec, const int64_t* refs,
+                    const uint8_t* valid, const uint8_t* lit, uint32_t lit_len,
+                    int op, uint8_t* mask, int64_t n_rows);
+void launch_ref_lens(hipStream_t, const uint8_t* dec, const uint64_t* refs,
+                     int64_t n, uint32_t* lens);
+void launch_ref_gather(hipStream_t, const uint8_t* dec, const uint64_t* refs,
+                       const uint64_t* offs, int64_t n, uint8_t* out);
 void launch_compact(hipStream_t, const uint8_t* mask, const int64_t* key_col,
                     int64_t n_rows, int64_t* out_keys, uint32_t* out_rows,
                     unsigned long long* counter);

@@ -26,6 +26,9 @@ STREAMS = {
            "keys": ["service", "span_kind", "status", "attr_s1", "attr_s7"],
            "i64": ["latency", "attr_i0", "attr_i3", "attr_i11"], "f64": [],
            "contains": None},
+    # dict-overflow PLAIN-fallback utf8 (raw-byte hash group-by path)
+    "c5": {"rows": 300_000, "keys": ["level", "host", "trace", "opt_tag"],
+           "i64": ["latency"], "f64": [], "contains": "trace"},
 }
 KEY_VALUES = {
     "level": ["TRACE", "DEBUG", "INFO", "WARN", "ERROR"],
@@ -37,6 +40,8 @@ KEY_VALUES = {
     "status": ["OK", "ERROR", "UNSET"],
     "attr_s1": [f"attr1-v{k}" for k in range(12)],
     "attr_s7": [f"attr7-v{k}" for k in range(12)],
+    "trace": [f"tr-{v:08d}" for v in range(1500)],
+    "opt_tag": [f"tag-{v:06d}" for v in range(800)],
 }
 
 

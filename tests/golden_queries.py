@@ -176,6 +176,54 @@ GOLDEN_QUERIES = {
 }
 
 # fixture name -> datagen parameters
+GOLDEN_QUERIES["g_c5"] = [
+    ("hash_group_count", {
+        "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+        "group_by": ["trace"],
+    }),
+    ("hash_two_keys_mixed", {
+        # hash key (trace, PLAIN fallback) x dict key (level, pure dict)
+        "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+        "group_by": ["level", "trace"],
+        "preds": [{"col": "latency", "op": "ge", "lit": 100_000}],
+    }),
+    ("hash_minmax_str", {
+        "ext": True,  # utf8 min/max: pinned by pyarrow oracle + Acero
+        "select": [{"agg": "min", "col": "trace"}, {"agg": "max", "col": "trace"},
+                   {"agg": "count_star"}],
+        "group_by": ["level"],
+    }),
+    ("hash_eq_pred_plain", {
+        # non-LIKE predicate on a PLAIN-fallback utf8 column
+        "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+        "preds": [{"col": "trace", "op": "eq", "lit": "tr-00000042"}],
+    }),
+    ("hash_nullable_key", {
+        # nullable hash key: NULL group + def-level routing on PLAIN pages
+        "select": [{"agg": "count_star"}],
+        "group_by": ["opt_tag"],
+    }),
+    ("hash_count_nullable", {
+        "select": [{"agg": "count", "col": "opt_tag"}, {"agg": "count_star"}],
+        "group_by": ["level"],
+    }),
+    ("hash_contains_on_key", {
+        # contains predicate riding the strref path (hash mode via group key)
+        "select": [{"agg": "count_star"}],
+        "group_by": ["trace"],
+        "preds": [{"col": "trace", "op": "contains", "lit": "42"}],
+    }),
+    ("hash_ge_pred_str", {
+        "select": [{"agg": "count_star"}],
+        "preds": [{"col": "opt_tag", "op": "ge", "lit": "tag-000600"}],
+    }),
+    ("hash_avg_by_tag", {
+        "select": [{"agg": "avg", "col": "latency"}, {"agg": "count_star"}],
+        "group_by": ["opt_tag"],
+        "time_range": [BASE, BASE + 2 * MIN],
+    }),
+]
+
 GOLDEN_FIXTURES = {
     "g_c1":       dict(config="c1", rows=120_000, rows_per_file=40_000, seed=1001),
     "g_c1_pages": dict(config="c1", rows=40_000, rows_per_file=40_000, seed=1002,
@@ -183,4 +231,8 @@ GOLDEN_FIXTURES = {
     "g_c3":       dict(config="c3", rows=25_000, rows_per_file=25_000, seed=1003),
     "g_c4":       dict(config="c4", rows=20_000, rows_per_file=20_000, seed=1004),
     "g_edge":     dict(config="c1", rows=9, rows_per_file=4, seed=1005),
+    # raw-byte utf8: dict-overflow -> PLAIN-fallback pages (written with a
+    # tiny dictionary_pagesize_limit); exercises the device hash group-by,
+    # strref predicates and strref min/max
+    "g_c5":       dict(config="c5", rows=60_000, rows_per_file=20_000, seed=1006),
 }
