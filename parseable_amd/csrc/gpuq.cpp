@@ -207,6 +207,8 @@ struct Partition {
   uint32_t* d_hcount = nullptr;
   std::map<int, uint64_t*> d_gid2ref;   // hash col -> gid -> strref
   std::map<int, uint32_t> hash_claimed; // hash col -> claimed count (last exec)
+  int32_t* d_cgid = nullptr;            // cascaded combined gid per row
+  std::vector<uint64_t*> d_gid2pair;    // per cascade level: gid -> packed pair
   int32_t* d_agg_kind = nullptr;
   uint8_t* d_needle = nullptr;        // concatenated CONTAINS needles
   std::vector<uint32_t> needle_off;   // per plan-pred offset into the pool
@@ -285,6 +287,7 @@ struct gpuq_plan {
   int32_t n_groups = 0;          // product of key sizes (incl null slots)
   int32_t n_fsum = 0;            // number of exact-f64-sum side tables
   bool has_hash = false;         // some column runs raw-byte utf8 hash mode
+  bool needs_cascade = false;    // dense key product exceeds GID_CAP
   bool fused_count = false;      // single dict key + count(*)-only + no preds
   std::mutex mu;
   // metrics
@@ -1506,14 +1509,19 @@ extern "C" gpuq_plan* gpuq_plan_build(
     c.nbins = (int32_t)std::max<int64_t>(nb, 1);
   }
 
-  // group table size
+  // group table size (saturating product; a product beyond GID_CAP takes
+  // the exact pair-cascade path at execute, like very-high-cardinality
+  // hash keys)
   int64_t g = 1;
   for (int ci : plan->group_cols) {
     const auto& c = plan->cols[ci];
-    g *= c.is_bin ? (int64_t)c.nbins + 1 : (int64_t)c.gdict.size() + 1;
+    int64_t sz = c.is_bin ? (int64_t)c.nbins + 1 : (int64_t)c.gdict.size() + 1;
+    g = (g > (int64_t)GID_CAP) ? g : g * sz;
   }
-  if (g > (int64_t)(1 << 22))
-    throw std::runtime_error("group-key cardinality product too large: hash fallback is a next row");
+  if (g > (int64_t)GID_CAP) {
+    plan->needs_cascade = true;
+    g = GID_CAP;
+  }
   plan->n_groups = (int32_t)g;
 
   for (auto& part : plan->parts)
@@ -1720,12 +1728,13 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   HIP_TRY(hipMalloc(&part.d_err, 4));
   // with hash-mode group keys the cardinality is only known at execute:
   // size the tables for the cap
-  size_t groups_cap = plan->has_hash ? (size_t)GID_CAP : (size_t)plan->n_groups;
+  bool hash_machinery = plan->has_hash || plan->needs_cascade;
+  size_t groups_cap = hash_machinery ? (size_t)GID_CAP : (size_t)plan->n_groups;
   size_t tsz = groups_cap * (1 + 2 * plan->aggs.size()) * 8;
   HIP_TRY(hipMalloc(&part.d_table, std::max<size_t>(tsz, 16)));
   size_t fsz = (size_t)plan->n_fsum * groups_cap * 4 * 8;
   HIP_TRY(hipMalloc(&part.d_fsum, std::max<size_t>(fsz, 16)));
-  if (plan->has_hash) {
+  if (hash_machinery) {
     HIP_TRY(hipMalloc(&part.d_hkeys, (1ull << HASH_LOG2) * 8));
     HIP_TRY(hipMalloc(&part.d_hgids, (1ull << HASH_LOG2) * 4));
     HIP_TRY(hipMalloc(&part.d_hcount, 4));
@@ -1735,6 +1744,12 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
         HIP_TRY(hipMalloc(&g2r, (size_t)GID_CAP * 8));
         part.d_gid2ref[ci] = g2r;
       }
+    if (plan->group_cols.size() >= 2) {
+      HIP_TRY(hipMalloc(&part.d_cgid, std::max<int64_t>(part.n_rows * 4, 16)));
+      part.d_gid2pair.resize(plan->group_cols.size() - 1, nullptr);
+      for (auto& ptr : part.d_gid2pair)
+        HIP_TRY(hipMalloc(&ptr, (size_t)GID_CAP * 8));
+    }
   }
   std::vector<int32_t> kinds;
   for (auto& a : plan->aggs) kinds.push_back(a.kind);
@@ -2172,6 +2187,14 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // group count with hash keys is only known after the hash build, so the
   // table init moves to just before aggregation; n_groups_exec tracks it
   int64_t n_groups_exec = plan->n_groups;
+  bool cascaded = false;
+  std::vector<uint32_t> level_claimed;
+  auto key_card = [&](int ci) -> int64_t {
+    const auto& kc = plan->cols[ci];
+    return kc.is_bin ? (int64_t)kc.nbins + 1
+           : kc.hash_mode ? (int64_t)part.hash_claimed[ci] + 1
+                          : (int64_t)kc.gdict.size() + 1;
+  };
   if (plan->fused_count)
     launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
                       part.d_agg_kind);
@@ -2326,17 +2349,6 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       HIP_TRY(hipStreamSynchronize(st));
       part.hash_claimed[ci] = claimed;
     }
-    int64_t g2 = 1;
-    for (int ci : plan->group_cols) {
-      auto& c = plan->cols[ci];
-      g2 *= c.is_bin ? (int64_t)c.nbins + 1
-            : c.hash_mode ? (int64_t)part.hash_claimed[ci] + 1
-                          : (int64_t)c.gdict.size() + 1;
-    }
-    if (g2 > (int64_t)GID_CAP)
-      throw std::runtime_error(
-          "group-key cardinality product too large (hash keys)");
-    n_groups_exec = g2;
     for (size_t ci = 0; ci < plan->cols.size(); ci++) {
       auto& c = plan->cols[ci];
       if (!c.hash_mode || c.lut_preds.empty()) continue;
@@ -2405,6 +2417,44 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
                    part.d_gid[(int)ci], part.n_rows);
   }
 
+  // 2c. group sizing: with hash keys the cardinality is per-execute; if
+  // the dense product space exceeds GID_CAP, combine keys exactly with the
+  // pair cascade (two at a time, single-u64 CAS claims) and aggregate on
+  // the dense combined gid — DataFusion's row-hash over arbitrary tuples.
+  if (!plan->group_cols.empty()) {
+    int64_t g2 = 1;
+    for (int ci : plan->group_cols) {
+      g2 *= key_card(ci);
+      if (g2 > (int64_t)GID_CAP * 2) break;  // saturate
+    }
+    if (g2 > (int64_t)GID_CAP) {
+      if (plan->group_cols.size() < 2 || !part.d_cgid)
+        throw std::runtime_error("group-key cardinality exceeds the 2^22 cap");
+      const int32_t* cur = part.d_gid[plan->group_cols[0]];
+      for (size_t k = 1; k < plan->group_cols.size(); k++) {
+        HIP_TRY(hipMemsetAsync(part.d_hkeys, 0xFF, (1ull << HASH_LOG2) * 8, st));
+        HIP_TRY(hipMemsetAsync(part.d_hgids, 0xFF, (1ull << HASH_LOG2) * 4, st));
+        HIP_TRY(hipMemsetAsync(part.d_hcount, 0, 4, st));
+        const int32_t* nxt = part.d_gid[plan->group_cols[k]];
+        launch_pair_build(st, cur, nxt, part.n_rows, part.d_hkeys,
+                          part.d_hgids, HASH_LOG2, part.d_hcount,
+                          part.d_gid2pair[k - 1], GID_CAP, part.d_err);
+        launch_pair_lookup(st, cur, nxt, part.n_rows, part.d_hkeys,
+                           part.d_hgids, HASH_LOG2, part.d_cgid);
+        uint32_t claimed = 0;
+        HIP_TRY(hipMemcpyAsync(&claimed, part.d_hcount, 4,
+                               hipMemcpyDeviceToHost, st));
+        HIP_TRY(hipStreamSynchronize(st));
+        level_claimed.push_back(claimed);
+        cur = part.d_cgid;
+      }
+      cascaded = true;
+      n_groups_exec = level_claimed.back();
+    } else {
+      n_groups_exec = g2;
+    }
+  }
+
   if (plan->is_projection)
     return execute_projection(plan, part, st, ev0, ev1, ev_decomp, t0,
                               need_mask, out);
@@ -2413,14 +2463,18 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   AggArgs a{};
   a.mask = part.d_mask;
   a.n_rows = part.n_rows;
-  a.n_keys = (int)plan->group_cols.size();
-  for (int k = 0; k < a.n_keys; k++) {
-    int ci = plan->group_cols[k];
-    a.key_gid[k] = part.d_gid[ci];
-    const auto& kc = plan->cols[ci];
-    a.key_size[k] = kc.is_bin ? kc.nbins + 1
-                    : kc.hash_mode ? (int32_t)part.hash_claimed[ci] + 1
-                                   : (int32_t)kc.gdict.size() + 1;
+  if (cascaded) {
+    // the cascade already produced a dense combined gid per row
+    a.n_keys = 1;
+    a.key_gid[0] = part.d_cgid;
+    a.key_size[0] = (int32_t)n_groups_exec;
+  } else {
+    a.n_keys = (int)plan->group_cols.size();
+    for (int k = 0; k < a.n_keys; k++) {
+      int ci = plan->group_cols[k];
+      a.key_gid[k] = part.d_gid[ci];
+      a.key_size[k] = (int32_t)key_card(ci);
+    }
   }
   a.n_aggs = (int)plan->aggs.size();
   for (int i = 0; i < a.n_aggs; i++) {
@@ -2540,13 +2594,32 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
 
   // decode combined gid -> per-key local gids
   std::vector<std::vector<int32_t>> key_gids(n_keys, std::vector<int32_t>(nr));
+  if (cascaded) {
+    // unwind the pair cascade: level maps are (prev_combined << 32) | gid_k
+    std::vector<std::vector<uint64_t>> pair_maps(level_claimed.size());
+    for (size_t L = 0; L < level_claimed.size(); L++) {
+      pair_maps[L].resize(level_claimed[L]);
+      if (level_claimed[L])
+        HIP_TRY(hipMemcpy(pair_maps[L].data(), part.d_gid2pair[L],
+                          (size_t)level_claimed[L] * 8,
+                          hipMemcpyDeviceToHost));
+    }
+    for (int64_t r = 0; r < nr; r++) {
+      int64_t g = live[r];
+      for (int k = n_keys - 1; k >= 1; k--) {
+        uint64_t pr = pair_maps[(size_t)k - 1][(size_t)g];
+        key_gids[k][r] = (int32_t)(uint32_t)pr;
+        g = (int64_t)(pr >> 32);
+      }
+      key_gids[0][r] = (int32_t)g;
+    }
+  } else
   for (int64_t r = 0; r < nr; r++) {
     int64_t g = live[r];
     for (int k = n_keys - 1; k >= 0; k--) {
       const auto& kc = plan->cols[plan->group_cols[k]];
-      int32_t sz = kc.is_bin ? kc.nbins + 1
-                   : kc.hash_mode ? (int32_t)part.hash_claimed[plan->group_cols[k]] + 1
-                                  : (int32_t)kc.gdict.size() + 1;
+      int32_t sz = (int32_t)key_card(plan->group_cols[k]);
+      (void)kc;
       key_gids[k][r] = (int32_t)(g % sz);
       g /= sz;
     }
@@ -2824,8 +2897,9 @@ gpuq_plan::~gpuq_plan() {
     F(part.d_raw); F(part.d_dec); F(part.d_pages); F(part.d_remap);
     F(part.d_dictv); F(part.d_lut); F(part.d_mask); F(part.d_err);
     F(part.d_table); F(part.d_fsum); F(part.d_agg_kind);
-    F(part.d_hkeys); F(part.d_hgids); F(part.d_hcount);
-    for (auto& kv : part.d_gid2ref) F(kv.second); F(part.d_needle); F(part.d_all_ids);
+    F(part.d_hkeys); F(part.d_hgids); F(part.d_hcount); F(part.d_cgid);
+    for (auto& kv : part.d_gid2ref) F(kv.second);
+    for (auto* ptr : part.d_gid2pair) F(ptr); F(part.d_needle); F(part.d_all_ids);
     F(part.d_rowof); F(part.d_rank); F(part.d_scr);
     F(part.d_present); F(part.d_tmpvalid);
     F(part.d_lits_lane); F(part.d_lits_wave);

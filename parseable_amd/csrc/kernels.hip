@@ -1091,6 +1091,68 @@ __global__ void k_hash_lookup(const uint8_t* __restrict__ dec,
   }
 }
 
+// pair cascade: when the dense per-key product space exceeds the group cap,
+// keys combine two at a time — (combined_so_far, next_gid) packs into one
+// u64 (each side < 2^32), claimed with a single-word CAS into the shared
+// table. Exact (key compare is ==), no spinning (two-pass like the string
+// hash). Mirrors DataFusion's row-hash over arbitrary key tuples.
+__device__ inline uint64_t mix64(uint64_t x) {
+  x ^= x >> 33;
+  x *= 0xff51afd7ed558ccdull;
+  x ^= x >> 33;
+  x *= 0xc4ceb9fe1a85ec53ull;
+  x ^= x >> 33;
+  return x;
+}
+__global__ void k_pair_build(const int32_t* __restrict__ a,
+                             const int32_t* __restrict__ b, int64_t n_rows,
+                             uint64_t* __restrict__ hkeys,
+                             int32_t* __restrict__ hgids, int clog2,
+                             uint32_t* counter, uint64_t* __restrict__ gid2pair,
+                             int32_t gid_cap, int32_t* d_error) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const uint64_t mask = (1ull << clog2) - 1;
+  for (; i < n_rows; i += stride) {
+    uint64_t key = ((uint64_t)(uint32_t)a[i] << 32) | (uint32_t)b[i];
+    uint64_t slot = mix64(key) & mask;
+    for (uint32_t probe = 0;; probe++) {
+      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+      uint64_t old = atomicCAS((unsigned long long*)&hkeys[slot],
+                               (unsigned long long)HREF_EMPTY,
+                               (unsigned long long)key);
+      if (old == HREF_EMPTY) {
+        uint32_t g = atomicAdd(counter, 1u);
+        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+        gid2pair[g] = key;
+        hgids[slot] = (int32_t)g;
+        break;
+      }
+      if (old == key) break;
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+__global__ void k_pair_lookup(const int32_t* __restrict__ a,
+                              const int32_t* __restrict__ b, int64_t n_rows,
+                              const uint64_t* __restrict__ hkeys,
+                              const int32_t* __restrict__ hgids, int clog2,
+                              int32_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const uint64_t mask = (1ull << clog2) - 1;
+  for (; i < n_rows; i += stride) {
+    uint64_t key = ((uint64_t)(uint32_t)a[i] << 32) | (uint32_t)b[i];
+    uint64_t slot = mix64(key) & mask;
+    for (;;) {
+      uint64_t k = hkeys[slot];
+      if (k == key) { out[i] = hgids[slot]; break; }
+      if (k == HREF_EMPTY) { out[i] = 0; break; }  // unreachable
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+
 // string predicates over row strrefs (mixed dict/PLAIN chunks where the
 // per-dict-entry LUT cannot cover the PLAIN pages). op: CmpMode, or -1 for
 // CONTAINS (byte substring).
@@ -2059,6 +2121,26 @@ void launch_hash_lookup(hipStream_t st, const uint8_t* dec, const int64_t* refs,
   if (blocks > 8192) blocks = 8192;
   hipLaunchKernelGGL(k_hash_lookup, dim3(blocks), dim3(256), 0, st, dec, refs,
                      valid, n_rows, hkeys, hgids, clog2, out_gid);
+}
+void launch_pair_build(hipStream_t st, const int32_t* a, const int32_t* b,
+                       int64_t n_rows, uint64_t* hkeys, int32_t* hgids,
+                       int clog2, uint32_t* counter, uint64_t* gid2pair,
+                       int32_t gid_cap, int32_t* d_err) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_pair_build, dim3(blocks), dim3(256), 0, st, a, b,
+                     n_rows, hkeys, hgids, clog2, counter, gid2pair, gid_cap,
+                     d_err);
+}
+void launch_pair_lookup(hipStream_t st, const int32_t* a, const int32_t* b,
+                        int64_t n_rows, const uint64_t* hkeys,
+                        const int32_t* hgids, int clog2, int32_t* out) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_pair_lookup, dim3(blocks), dim3(256), 0, st, a, b,
+                     n_rows, hkeys, hgids, clog2, out);
 }
 void launch_cmp_str(hipStream_t st, const uint8_t* dec, const int64_t* refs,
                     const uint8_t* valid, const uint8_t* lit, uint32_t lit_len,

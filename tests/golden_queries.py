@@ -217,6 +217,13 @@ GOLDEN_QUERIES["g_c5"] = [
         "select": [{"agg": "count_star"}],
         "preds": [{"col": "opt_tag", "op": "ge", "lit": "tag-000600"}],
     }),
+    ("hash_cascade_three_keys", {
+        # dense key-product space 1501x1001x801 >> the 2^22 cap: the exact
+        # pair cascade combines keys two at a time on device
+        "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+        "group_by": ["trace", "host", "opt_tag"],
+        "preds": [{"col": "latency", "op": "lt", "lit": 15_000}],
+    }),
     ("hash_avg_by_tag", {
         "select": [{"agg": "avg", "col": "latency"}, {"agg": "count_star"}],
         "group_by": ["opt_tag"],
