@@ -184,6 +184,10 @@ typedef struct {
   int64_t exec_ns;
   int64_t load_ns;
   int64_t decomp_ns;          /* decompression kernels only */
+  int64_t cache_hit_bytes;    /* compressed bytes served from the GPU-resident
+                                 hot tier (decompressed chunk images cached
+                                 across plans/queries, SURVEY §8f-3;
+                                 keyed like hottier.rs:1405-1417) */
 } gpuq_metrics;
 int32_t gpuq_plan_metrics(gpuq_plan*, gpuq_metrics* out);
 

@@ -48,6 +48,7 @@ class GpuqMetrics(C.Structure):
         ("exec_ns", C.c_int64),
         ("load_ns", C.c_int64),
         ("decomp_ns", C.c_int64),
+        ("cache_hit_bytes", C.c_int64),
     ]
 
 

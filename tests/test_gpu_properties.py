@@ -138,3 +138,33 @@ def test_hash_projection_topk(hash_stream):
     # ties at equal timestamps are engine-defined: compare the (ts, trace)
     # multiset instead of exact order
     assert sorted(map(tuple, rows)) == sorted(map(tuple, want["rows"]))
+
+
+def test_hot_tier_repeat_query(stream):
+    """Cross-query GPU hot tier: a SECOND plan over the same
+    chunks is served from the session cache — no raw re-upload, no LZ4
+    structure walk, decompression kernels gone (decomp_ns ~ 0)."""
+    from parseable_amd import GpuSession, StandardTableProvider
+
+    sess = GpuSession()
+    prov = StandardTableProvider(stream["stream_dir"], sess)
+    q1 = {"select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+          "group_by": ["level"]}
+    p1 = prov.scan(dict(q1))
+    p1.load()
+    r1 = p1.execute_all()
+    m1 = p1.metrics()
+    assert m1["cache_hit_bytes"] == 0
+    p1.close()
+
+    q2 = {"select": [{"agg": "count_star"}, {"agg": "min", "col": "latency"}],
+          "group_by": ["level"]}
+    p2 = prov.scan(dict(q2))
+    p2.load()
+    r2 = p2.execute_all()
+    m2 = p2.metrics()
+    assert m2["cache_hit_bytes"] == m2["bytes_scanned"]
+    assert m2["decomp_ns"] < 1_000_000  # decompress kernels gone
+    assert [row[0] for row in r2] == [row[0] for row in r1]
+    assert [row[1] for row in r2] == [row[1] for row in r1]  # same counts
+    p2.close()
