@@ -139,7 +139,9 @@ def execute_projection(files: list[str], query: dict,
         out.extend([list(r) for r in zip(*proj)])
     ts_i = cols_wanted.index("p_timestamp")
     out.sort(key=lambda r: -r[ts_i])
-    return {"columns": cols_wanted, "rows": out[: int(query["limit"])],
+    lim = query.get("limit")
+    return {"columns": cols_wanted,
+            "rows": out[: int(lim)] if lim else out,
             "all_matching": out}
 
 

@@ -791,12 +791,11 @@ extern "C" gpuq_plan* gpuq_plan_build(
     // projection scan: SELECT cols ... ORDER BY p_timestamp DESC LIMIT k
     // (the console's default query; ordering contract
     // stream_schema_provider.rs:181-204 — time-DESC is implied)
-    if (n_projection <= 0 || limit <= 0)
-      throw std::runtime_error(
-          "projection scans need a column list and a LIMIT (full result-set "
-          "export: next row, SURVEY §8f)");
-    if (limit > (1 << 22))
-      throw std::runtime_error("projection LIMIT too large");
+    if (n_projection <= 0)
+      throw std::runtime_error("projection scans need a column list");
+    // limit < 0 = unlimited: the result is exported as a true multi-batch
+    // ArrowArrayStream in 20k-row batches (P_EXECUTION_BATCH_SIZE,
+    // cli.rs:476-482) — see execute_projection
     plan->is_projection = true;
     for (int32_t i = 0; i < n_projection; i++) {
       int ci = find_or_add_col(plan->cols, projection[i]);
@@ -2116,50 +2115,85 @@ double acc256_to_double(const uint64_t li[4]) {
   return neg ? -d : d;
 }
 
-int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
-                           hipEvent_t ev0, hipEvent_t ev1, hipEvent_t ev_decomp,
-                           int64_t t0, bool need_mask,
-                           struct ArrowArrayStream* out) {
-  // compact selected (ts,row) pairs -> radix sort desc -> top-k gather
-  HIP_TRY(hipMemsetAsync(part.d_count, 0, 8, st));
-  launch_compact(st, need_mask ? part.d_mask : nullptr,
-                 part.d_val[plan->ts_col], part.n_rows, part.d_keys,
-                 part.d_rows, part.d_count);
-  unsigned long long cnt = 0;
-  HIP_TRY(hipMemcpyAsync(&cnt, part.d_count, 8, hipMemcpyDeviceToHost, st));
-  HIP_TRY(hipStreamSynchronize(st));
-  if (cnt)
-    sort_pairs_desc(st, part.d_sort_temp, part.sort_temp_bytes, part.d_keys,
-                    part.d_keys_sorted, part.d_rows, part.d_rows_sorted,
-                    (int64_t)cnt);
-  int64_t k = std::min<int64_t>((int64_t)cnt, plan->limit);
+// ---- projection scans --------------------------------------------------
+// Selected rows sorted by p_timestamp DESC (the console ordering contract,
+// stream_schema_provider.rs:181-204). With a LIMIT the stream holds the
+// top-k; without one it is a TRUE multi-batch ArrowArrayStream: each
+// get_next gathers the next P_EXECUTION_BATCH_SIZE rows (20,000 — the
+// reference batch size, cli.rs:476-482) from the device-resident sorted row
+// list, so host and device memory stay bounded regardless of result size.
+// The stream borrows the plan's device state: drain it before
+// gpuq_plan_destroy.
+namespace {
+constexpr int64_t PROJ_BATCH = 20000;
 
-  int np = (int)plan->projection.size();
-  std::vector<std::vector<int64_t>> vals(np);
-  std::vector<std::vector<int32_t>> gids(np);
-  std::vector<std::vector<uint8_t>> valids(np);
-  std::vector<std::vector<std::string>> pstrs(np);  // hash-mode utf8 cols
+struct ProjState {
+  gpuq_plan* plan = nullptr;
+  Partition* part = nullptr;
+  int64_t total = 0, cursor = 0;
+  struct ArrowSchema schema;
+  bool schema_moved = false;
+  std::string err;
   int64_t* d_g64 = nullptr;
   int32_t* d_g32 = nullptr;
   uint8_t* d_g8 = nullptr;
-  HIP_TRY(hipMalloc(&d_g64, std::max<int64_t>(k * 8, 16)));
-  HIP_TRY(hipMalloc(&d_g32, std::max<int64_t>(k * 4, 16)));
-  HIP_TRY(hipMalloc(&d_g8, std::max<int64_t>(k, 16)));
+};
+
+int proj_get_schema(struct ArrowArrayStream* st0, struct ArrowSchema* out) {
+  auto* s = (ProjState*)st0->private_data;
+  *out = s->schema;
+  s->schema_moved = true;
+  s->schema.release = nullptr;
+  return 0;
+}
+const char* proj_get_last_error(struct ArrowArrayStream* st0) {
+  auto* s = (ProjState*)st0->private_data;
+  return s->err.empty() ? nullptr : s->err.c_str();
+}
+void proj_release(struct ArrowArrayStream* st0) {
+  auto* s = (ProjState*)st0->private_data;
+  if (s) {
+    if (s->schema.release) release_schema(&s->schema);
+    if (s->d_g64) (void)hipFree(s->d_g64);
+    if (s->d_g32) (void)hipFree(s->d_g32);
+    if (s->d_g8) (void)hipFree(s->d_g8);
+    delete s;
+  }
+  st0->release = nullptr;
+}
+
+int proj_get_next(struct ArrowArrayStream* st0, struct ArrowArray* out) try {
+  auto* s = (ProjState*)st0->private_data;
+  gpuq_plan* plan = s->plan;
+  Partition& part = *s->part;
+  if (s->cursor >= s->total) {
+    memset(out, 0, sizeof(*out));
+    out->release = nullptr;
+    return 0;
+  }
+  HIP_TRY(hipSetDevice(part.device));
+  hipStream_t st = part.stream;
+  const int64_t k = std::min<int64_t>(PROJ_BATCH, s->total - s->cursor);
+  const uint32_t* rows = part.d_rows_sorted + s->cursor;
+  int np = (int)plan->projection.size();
+
+  std::vector<std::vector<int64_t>> vals(np);
+  std::vector<std::vector<int32_t>> gids(np);
+  std::vector<std::vector<uint8_t>> valids(np);
+  std::vector<std::vector<std::string>> pstrs(np);
   for (int p = 0; p < np; p++) {
     int ci = plan->projection[p];
     auto& c = plan->cols[ci];
     if (c.phys == PT_BYTE_ARRAY && c.hash_mode) {
-      // raw-byte utf8: gather the winners' strrefs + validity, then fetch
-      // the bytes from the arena
       std::vector<int64_t> refs(k);
       valids[p].assign(k, 1);
-      launch_gather_i64(st, part.d_rows_sorted, k, part.d_val[ci], d_g64);
-      HIP_TRY(hipMemcpyAsync(refs.data(), d_g64, k * 8,
+      launch_gather_i64(st, rows, k, part.d_val[ci], s->d_g64);
+      HIP_TRY(hipMemcpyAsync(refs.data(), s->d_g64, k * 8,
                              hipMemcpyDeviceToHost, st));
       auto itv = part.d_valid.find(ci);
       if (itv != part.d_valid.end()) {
-        launch_gather_u8(st, part.d_rows_sorted, k, itv->second, d_g8);
-        HIP_TRY(hipMemcpyAsync(valids[p].data(), d_g8, k,
+        launch_gather_u8(st, rows, k, itv->second, s->d_g8);
+        HIP_TRY(hipMemcpyAsync(valids[p].data(), s->d_g8, k,
                                hipMemcpyDeviceToHost, st));
       }
       HIP_TRY(hipStreamSynchronize(st));
@@ -2173,69 +2207,38 @@ int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
         pstrs[p][vrow[j]] = std::move(strs[j]);
     } else if (c.phys == PT_BYTE_ARRAY) {
       gids[p].resize(k);
-      launch_gather_i32(st, part.d_rows_sorted, k, part.d_gid[ci], d_g32);
-      HIP_TRY(hipMemcpyAsync(gids[p].data(), d_g32, k * 4,
+      launch_gather_i32(st, rows, k, part.d_gid[ci], s->d_g32);
+      HIP_TRY(hipMemcpyAsync(gids[p].data(), s->d_g32, k * 4,
                              hipMemcpyDeviceToHost, st));
       HIP_TRY(hipStreamSynchronize(st));
     } else {
       vals[p].resize(k);
       valids[p].assign(k, 1);
-      launch_gather_i64(st, part.d_rows_sorted, k, part.d_val[ci], d_g64);
-      HIP_TRY(hipMemcpyAsync(vals[p].data(), d_g64, k * 8,
+      launch_gather_i64(st, rows, k, part.d_val[ci], s->d_g64);
+      HIP_TRY(hipMemcpyAsync(vals[p].data(), s->d_g64, k * 8,
                              hipMemcpyDeviceToHost, st));
       auto itv = part.d_valid.find(ci);
       if (itv != part.d_valid.end()) {
-        launch_gather_u8(st, part.d_rows_sorted, k, itv->second, d_g8);
-        HIP_TRY(hipMemcpyAsync(valids[p].data(), d_g8, k,
+        launch_gather_u8(st, rows, k, itv->second, s->d_g8);
+        HIP_TRY(hipMemcpyAsync(valids[p].data(), s->d_g8, k,
                                hipMemcpyDeviceToHost, st));
       }
       HIP_TRY(hipStreamSynchronize(st));
     }
   }
-  HIP_TRY(hipEventRecord(ev1, st));
-  int32_t herr = 0;
-  HIP_TRY(hipMemcpyAsync(&herr, part.d_err, 4, hipMemcpyDeviceToHost, st));
-  HIP_TRY(hipStreamSynchronize(st));
-  HIP_TRY(hipFree(d_g64));
-  HIP_TRY(hipFree(d_g32));
-  HIP_TRY(hipFree(d_g8));
-  if (herr != 0)
-    throw std::runtime_error("kernel error code " + std::to_string(herr));
-  hot_tier_populate(plan, part, st);
-  float ms_total = 0, ms_decomp = 0;
-  HIP_TRY(hipEventElapsedTime(&ms_total, ev0, ev1));
-  HIP_TRY(hipEventElapsedTime(&ms_decomp, ev0, ev_decomp));
-  HIP_TRY(hipEventDestroy(ev0));
-  HIP_TRY(hipEventDestroy(ev1));
-  HIP_TRY(hipEventDestroy(ev_decomp));
 
-  // assemble the row batch: projected columns in order
-  auto* ss = new StreamState();
-  memset(&ss->schema, 0, sizeof(ss->schema));
-  ss->schema.format = strdup("+s");
-  ss->schema.name = strdup("");
-  ss->schema.release = release_schema;
-  ss->schema.n_children = np;
-  ss->schema.children = (struct ArrowSchema**)calloc(np, sizeof(void*));
-  for (int p = 0; p < np; p++) {
-    auto& c = plan->cols[plan->projection[p]];
-    const char* fmt = (c.phys == PT_BYTE_ARRAY) ? "u"
-                      : (c.phys == PT_DOUBLE) ? "g" : "l";
-    ss->schema.children[p] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
-    make_schema_field(ss->schema.children[p], fmt, c.name);
-  }
   auto* eb = new ExportedBatch();
-  memset(&ss->batch, 0, sizeof(ss->batch));
-  ss->batch.length = k;
-  ss->batch.n_buffers = 1;
-  ss->batch.buffers = (const void**)calloc(1, sizeof(void*));
-  ss->batch.n_children = np;
-  ss->batch.children = (struct ArrowArray**)calloc(np, sizeof(void*));
-  ss->batch.release = release_array;
-  ss->batch.private_data = eb;
+  memset(out, 0, sizeof(*out));
+  out->length = k;
+  out->n_buffers = 1;
+  out->buffers = (const void**)calloc(1, sizeof(void*));
+  out->n_children = np;
+  out->children = (struct ArrowArray**)calloc(np, sizeof(void*));
+  out->release = release_array;
+  out->private_data = eb;
   for (int p = 0; p < np; p++) {
     auto* ch = (struct ArrowArray*)calloc(1, sizeof(struct ArrowArray));
-    ss->batch.children[p] = ch;
+    out->children[p] = ch;
     ch->length = k;
     ch->release = release_array;
     auto& c = plan->cols[plan->projection[p]];
@@ -2289,12 +2292,77 @@ int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
       ch->buffers[1] = v;
     }
   }
+  s->cursor += k;
+  return 0;
+} catch (const std::exception& e) {
+  auto* s = (ProjState*)st0->private_data;
+  s->err = e.what();
+  return 1;
+}
+
+}  // namespace
+
+int32_t execute_projection(gpuq_plan* plan, Partition& part, hipStream_t st,
+                           hipEvent_t ev0, hipEvent_t ev1, hipEvent_t ev_decomp,
+                           int64_t t0, bool need_mask,
+                           struct ArrowArrayStream* out) {
+  // compact selected (ts,row) pairs -> radix sort desc -> batched gather
+  HIP_TRY(hipMemsetAsync(part.d_count, 0, 8, st));
+  launch_compact(st, need_mask ? part.d_mask : nullptr,
+                 part.d_val[plan->ts_col], part.n_rows, part.d_keys,
+                 part.d_rows, part.d_count);
+  unsigned long long cnt = 0;
+  HIP_TRY(hipMemcpyAsync(&cnt, part.d_count, 8, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  if (cnt)
+    sort_pairs_desc(st, part.d_sort_temp, part.sort_temp_bytes, part.d_keys,
+                    part.d_keys_sorted, part.d_rows, part.d_rows_sorted,
+                    (int64_t)cnt);
+  int64_t k = plan->limit > 0 ? std::min<int64_t>((int64_t)cnt, plan->limit)
+                              : (int64_t)cnt;
+
+  HIP_TRY(hipEventRecord(ev1, st));
+  int32_t herr = 0;
+  HIP_TRY(hipMemcpyAsync(&herr, part.d_err, 4, hipMemcpyDeviceToHost, st));
+  HIP_TRY(hipStreamSynchronize(st));
+  if (herr != 0)
+    throw std::runtime_error("kernel error code " + std::to_string(herr));
+  hot_tier_populate(plan, part, st);
+  float ms_total = 0, ms_decomp = 0;
+  HIP_TRY(hipEventElapsedTime(&ms_total, ev0, ev1));
+  HIP_TRY(hipEventElapsedTime(&ms_decomp, ev0, ev_decomp));
+  HIP_TRY(hipEventDestroy(ev0));
+  HIP_TRY(hipEventDestroy(ev1));
+  HIP_TRY(hipEventDestroy(ev_decomp));
+
+  int np = (int)plan->projection.size();
+  auto* ps = new ProjState();
+  ps->plan = plan;
+  ps->part = &part;
+  ps->total = k;
+  HIP_TRY(hipMalloc(&ps->d_g64, PROJ_BATCH * 8));
+  HIP_TRY(hipMalloc(&ps->d_g32, PROJ_BATCH * 4));
+  HIP_TRY(hipMalloc(&ps->d_g8, PROJ_BATCH));
+  memset(&ps->schema, 0, sizeof(ps->schema));
+  ps->schema.format = strdup("+s");
+  ps->schema.name = strdup("");
+  ps->schema.release = release_schema;
+  ps->schema.n_children = np;
+  ps->schema.children = (struct ArrowSchema**)calloc(np, sizeof(void*));
+  for (int p = 0; p < np; p++) {
+    auto& c = plan->cols[plan->projection[p]];
+    const char* fmt = (c.phys == PT_BYTE_ARRAY) ? "u"
+                      : (c.phys == PT_DOUBLE) ? "g" : "l";
+    ps->schema.children[p] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
+    make_schema_field(ps->schema.children[p], fmt, c.name);
+  }
   memset(out, 0, sizeof(*out));
-  out->get_schema = ss_get_schema;
-  out->get_next = ss_get_next;
-  out->get_last_error = ss_get_last_error;
-  out->release = ss_release;
-  out->private_data = ss;
+  out->get_schema = proj_get_schema;
+  out->get_next = proj_get_next;
+  out->get_last_error = proj_get_last_error;
+  out->release = proj_release;
+  out->private_data = ps;
+
   {
     std::lock_guard<std::mutex> g(plan->mu);
     plan->m_kernel_ns += (int64_t)(ms_total * 1e6);

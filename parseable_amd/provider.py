@@ -548,11 +548,9 @@ def _build_c_projection(query: dict, keep):
     cb = [c.encode() for c in cols]
     keep.extend(cb)
     arr = (C.c_char_p * max(len(cols), 1))(*cb)
-    limit = int(query.get("limit", -1))
+    limit = int(query.get("limit", -1))  # < 0: unlimited (streamed batches)
     ob = query.get("order_by")
     if cols:
-        if limit <= 0:
-            raise GpuqError("projection scans need a LIMIT (SURVEY §8f-4)")
         if ob and not (ob.get("col") == "p_timestamp" and ob.get("desc", True)):
             raise GpuqError("only ORDER BY p_timestamp DESC is supported "
                             "(the reference's output-ordering contract, "
@@ -677,8 +675,11 @@ class GpuExecutionPlan:
             if self._lib.gpuq_plan_load(self._plan, p) != 0:
                 raise GpuqError(f"plan_load failed: {self.session._err()}")
 
-    def execute(self, partition: int):
-        """-> pyarrow.RecordBatch of the PARTIAL aggregate for this shard."""
+    def execute_reader(self, partition: int):
+        """-> pyarrow.RecordBatchReader over this partition's result: one
+        partial-aggregate batch for aggregations, or a true multi-batch
+        stream of 20k-row batches for (unlimited) projection scans —
+        ExecutionPlan::execute's pull-based stream (query/mod.rs:341-368)."""
         import pyarrow as pa
 
         # allocate an ArrowArrayStream struct (5 ptr fields + private)
@@ -686,9 +687,17 @@ class GpuExecutionPlan:
         rc = self._lib.gpuq_plan_execute(self._plan, partition, C.cast(buf, C.c_void_p))
         if rc != 0:
             raise GpuqError(f"plan_execute failed: {self.session._err()}")
-        reader = pa.RecordBatchReader._import_from_c(C.addressof(buf))
-        batches = list(reader)
-        return batches[0] if batches else None
+        return pa.RecordBatchReader._import_from_c(C.addressof(buf))
+
+    def execute(self, partition: int):
+        """Drained form of execute_reader: one combined RecordBatch."""
+        import pyarrow as pa
+
+        batches = list(self.execute_reader(partition))
+        if len(batches) <= 1:
+            return batches[0] if batches else None
+        t = pa.Table.from_batches(batches).combine_chunks()
+        return t.to_batches()[0] if t.num_rows else None
 
     def execute_all(self):
         """Run every partition and apply the Final merge: aggregation
@@ -809,8 +818,9 @@ def merge_partials(batches, query):
 
 
 def merge_topk(batches, query, extra_rows=None):
-    """Final merge of per-partition top-k row batches: re-sort by
-    p_timestamp DESC across partitions, truncate to LIMIT. extra_rows:
+    """Final merge of per-partition projection batches: re-sort by
+    p_timestamp DESC across partitions, truncate to LIMIT when one is set
+    (unlimited scans return every matching row). extra_rows:
     already-projected rows from the staging CPU leg."""
     cols = query["select_cols"]
     ts_i = cols.index("p_timestamp")
@@ -821,7 +831,8 @@ def merge_topk(batches, query, extra_rows=None):
         data = [b.column(i).to_pylist() for i in range(b.num_columns)]
         rows.extend([list(r) for r in zip(*data)])
     rows.sort(key=lambda r: -r[ts_i])
-    return rows[: int(query["limit"])]
+    lim = query.get("limit")
+    return rows[: int(lim)] if lim else rows
 
 
 class Query:

@@ -168,3 +168,27 @@ def test_staging_union_gpu(staged_stream):
         want = qo.execute(s["files"] + files, dict(q), extra_tables=tables)["rows"]
         assert_rows_equal(got, want, q)
         plan.close()
+
+
+def test_staging_projection_unlimited_cpu(staged_stream):
+    """Unlimited projection (no LIMIT): every matching row, ts-DESC order
+    (merge_topk without truncation)."""
+    s = staged_stream
+    sdir2 = s["staging_dir"] + "_arrowsP"
+    import shutil
+
+    if not os.path.exists(sdir2):
+        os.makedirs(sdir2)
+        for p in s["st"]["arrows"]:
+            shutil.copy(p, sdir2)
+    prov = StandardTableProvider(s["stream_dir"], None, staging_dir=sdir2,
+                                 now_ms=NOW_MS)
+    q = {"select_cols": ["p_timestamp", "level"],
+         "preds": [{"col": "level", "op": "eq", "lit": "ERROR"}],
+         "time_range": STAGING_RANGE}
+    plan = prov.scan(dict(q))
+    got = plan.execute_all()
+    _, tables = _staging_tables(s["st"], arrows_only=True)
+    want = qo.execute([], dict(q), extra_tables=tables)["rows"]
+    assert len(got) == len(want)
+    assert sorted(map(tuple, got)) == sorted(map(tuple, want))
