@@ -193,9 +193,9 @@ def _column_stats_entry(col_meta, name):
 
 def _gen_one_file(args):
     (root, stream, config, rows, seed, rows_per_file, data_page_size, m,
-     minute_offset) = args
+     minute_offset, compression) = args
     return _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
-                           data_page_size, m, minute_offset)
+                           data_page_size, m, minute_offset, compression)
 
 
 def gen_stream(
@@ -209,6 +209,8 @@ def gen_stream(
     quiet: bool = True,
     workers: int = 1,
     minute_offset: int = 0,
+    compression: str = "lz4",
+    manifest_codec: str | None = None,
 ):
     """Write a synthetic stream. Returns dict with file list + manifest paths.
     workers > 1 parallelizes per-file generation (deterministic: per-file rng
@@ -224,7 +226,7 @@ def gen_stream(
 
         argl = [
             (root, stream, config, rows, seed, rows_per_file, data_page_size,
-             m, minute_offset)
+             m, minute_offset, compression)
             for m in range(n_files)
         ]
         with ProcessPoolExecutor(max_workers=workers) as ex:
@@ -234,22 +236,22 @@ def gen_stream(
                 if not quiet and (i % 50 == 0):
                     print(f"  wrote {i + 1}/{n_files} files", flush=True)
         return _finish_stream(root, stream, stream_dir, manifest_files,
-                              file_paths, rows, config)
+                              file_paths, rows, config, manifest_codec)
 
     for m in range(n_files):
         abs_path, rel_path, entry = _gen_file_inner(
             root, stream, config, rows, seed, rows_per_file, data_page_size,
-            m, minute_offset)
+            m, minute_offset, compression)
         manifest_files.append(entry)
         file_paths.append(abs_path)
         if not quiet and (m % 50 == 0):
             print(f"  wrote {m + 1}/{n_files} files", flush=True)
     return _finish_stream(root, stream, stream_dir, manifest_files, file_paths,
-                          rows, config)
+                          rows, config, manifest_codec)
 
 
 def _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
-                    data_page_size, m, minute_offset):
+                    data_page_size, m, minute_offset, compression="lz4"):
     if True:
         n = min(rows_per_file, rows - m * rows_per_file)
         rng = np.random.default_rng([seed, m])  # per-file stream: parallel-safe
@@ -274,7 +276,9 @@ def _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
             tbl,
             abs_path,
             row_group_size=ROW_GROUP_SIZE,
-            compression="lz4",            # codec 7 LZ4_RAW, verified
+            compression=compression,      # "lz4" = codec 7 LZ4_RAW (default,
+                                          # cli.rs:484-491); "snappy" = the
+                                          # test-compose codec
             use_dictionary=[c for c in tbl.column_names if c != "p_timestamp"],
             column_encoding={"p_timestamp": "DELTA_BINARY_PACKED"},
             data_page_version="1.0",
@@ -325,8 +329,25 @@ def _gen_file_inner(root, stream, config, rows, seed, rows_per_file,
         return abs_path, rel_path, entry
 
 
+def _zstd_compress(data: bytes, level: int = 3) -> bytes:
+    # same library + level as the reference's manifest codec
+    # (catalog/manifest.rs ZSTD_LEVEL = 3, via the zstd crate -> libzstd)
+    import ctypes
+
+    z = ctypes.CDLL("libzstd.so.1")
+    z.ZSTD_compressBound.restype = ctypes.c_size_t
+    z.ZSTD_compress.restype = ctypes.c_size_t
+    z.ZSTD_isError.restype = ctypes.c_uint
+    bound = z.ZSTD_compressBound(len(data))
+    buf = ctypes.create_string_buffer(bound)
+    n = z.ZSTD_compress(buf, bound, data, len(data), level)
+    if z.ZSTD_isError(n):
+        raise RuntimeError("zstd compress failed")
+    return buf.raw[:n]
+
+
 def _finish_stream(root, stream, stream_dir, manifest_files, file_paths, rows,
-                   config):
+                   config, manifest_codec=None):
     # daily manifests (partition bounds = whole UTC day, src/catalog/mod.rs:176-187)
     by_day = {}
     for f in manifest_files:
@@ -336,8 +357,11 @@ def _finish_stream(root, stream, stream_dir, manifest_files, file_paths, rows,
     for day, files in sorted(by_day.items()):
         rel_mpath = f"{stream}/date={day}/manifest.json"
         mpath = os.path.join(root, rel_mpath)
-        with open(mpath, "w") as fh:
-            json.dump({"version": "v2", "files": files}, fh)
+        mbytes = json.dumps({"version": "v2", "files": files}).encode()
+        if manifest_codec == "zstd":
+            mbytes = _zstd_compress(mbytes)
+        with open(mpath, "wb") as fh:
+            fh.write(mbytes)
         d0 = datetime.strptime(day, "%Y-%m-%d").replace(tzinfo=timezone.utc)
         lo = int(d0.timestamp() * 1000)
         manifest_list.append(

@@ -310,6 +310,61 @@ static int lz4_decompress(const uint8_t *src, size_t src_len,
 }
 
 /* decompress a page; returns pointer (into dst or src for uncompressed) */
+static int snappy_decompress(const uint8_t *src, size_t comp, uint8_t *dst,
+                             size_t cap) {
+    size_t s = 0, d = 0;
+    uint64_t ulen = 0; int sh = 0;
+    for (;;) {
+        if (s >= comp) return -1;
+        uint8_t b = src[s++];
+        ulen |= (uint64_t)(b & 0x7f) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+    }
+    if (ulen > cap) return -1;
+    while (s < comp) {
+        uint8_t tag = src[s++];
+        int type = tag & 3;
+        if (type == 0) {
+            size_t len = (size_t)(tag >> 2) + 1;
+            if (len > 60) {
+                int extra = (int)len - 60;
+                if (s + extra > comp) return -1;
+                len = 0;
+                for (int i = 0; i < extra; i++) len |= (size_t)src[s + i] << (8 * i);
+                len += 1;
+                s += extra;
+            }
+            if (s + len > comp || d + len > cap) return -1;
+            memcpy(dst + d, src + s, len);
+            s += len; d += len;
+        } else {
+            size_t ml, off;
+            if (type == 1) {
+                ml = ((tag >> 2) & 7) + 4;
+                if (s >= comp) return -1;
+                off = ((size_t)(tag >> 5) << 8) | src[s++];
+            } else if (type == 2) {
+                ml = (size_t)(tag >> 2) + 1;
+                if (s + 2 > comp) return -1;
+                off = src[s] | ((size_t)src[s + 1] << 8);
+                s += 2;
+            } else {
+                ml = (size_t)(tag >> 2) + 1;
+                if (s + 4 > comp) return -1;
+                off = src[s] | ((size_t)src[s + 1] << 8) |
+                      ((size_t)src[s + 2] << 16) | ((size_t)src[s + 3] << 24);
+                s += 4;
+            }
+            if (off == 0 || off > d || d + ml > cap) return -1;
+            const uint8_t *mp = dst + d - off;
+            for (size_t i = 0; i < ml; i++) dst[d + i] = mp[i];
+            d += ml;
+        }
+    }
+    return (int)d;
+}
+
 static const uint8_t *page_payload(int codec, const uint8_t *src, int32_t comp,
                                    uint8_t *dst, int32_t uncomp) {
     if (codec == CODEC_UNCOMPRESSED || comp == uncomp) {
@@ -323,6 +378,11 @@ static const uint8_t *page_payload(int codec, const uint8_t *src, int32_t comp,
         if (n == (int)uncomp) return dst;
         if (comp == uncomp) return src;     /* stored raw */
         die("lz4 decode failed");
+    }
+    if (codec == CODEC_SNAPPY) {
+        int n = snappy_decompress(src, (size_t)comp, dst, (size_t)uncomp);
+        if (n == (int)uncomp) return dst;
+        die("snappy decode failed");
     }
     die("unsupported codec");
     return NULL;

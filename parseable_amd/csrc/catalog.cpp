@@ -4,10 +4,51 @@
 #include <cstring>
 #include <ctime>
 #include <fstream>
+#include <dlfcn.h>
 #include <sstream>
 
 namespace gpuq {
 namespace {
+
+// zstd frame decode via the system libzstd (the reference binds the same
+// library through the zstd crate, catalog/manifest.rs ZSTD_LEVEL=3).
+// dlopen'd lazily so plain-JSON-only deployments never need it.
+std::string zstd_decompress_str(const std::string& in) {
+  static void* h = dlopen("libzstd.so.1", RTLD_NOW);
+  if (!h) throw std::runtime_error("libzstd.so.1 not available for compressed manifest");
+  using FBound = unsigned long long (*)(const void*, size_t);
+  using FDec = size_t (*)(void*, size_t, const void*, size_t);
+  using FErr = unsigned (*)(size_t);
+  static auto f_size = (FBound)dlsym(h, "ZSTD_getFrameContentSize");
+  static auto f_dec = (FDec)dlsym(h, "ZSTD_decompress");
+  static auto f_err = (FErr)dlsym(h, "ZSTD_isError");
+  if (!f_size || !f_dec || !f_err)
+    throw std::runtime_error("libzstd symbols missing");
+  unsigned long long need = f_size(in.data(), in.size());
+  size_t cap = (need + 1 < 2) ? (in.size() * 32 + (1 << 20))  // unknown size
+                              : (size_t)need;
+  for (;;) {
+    std::string out(cap, '\0');
+    size_t n = f_dec(out.data(), out.size(), in.data(), in.size());
+    if (!f_err(n)) {
+      out.resize(n);
+      return out;
+    }
+    if (need + 1 >= 2 || cap > (1ull << 32))
+      throw std::runtime_error("zstd manifest decompression failed");
+    cap *= 4;  // content size unknown: grow and retry
+  }
+}
+
+// Manifest bytes -> JSON text, sniffing the zstd magic exactly as
+// decode_manifest does (catalog/manifest.rs:53-110: ZSTD_MAGIC
+// 28 B5 2F FD; plain pre-compression manifests stay readable as-is).
+std::string decode_manifest_text(std::string raw) {
+  if (raw.size() >= 4 && (uint8_t)raw[0] == 0x28 && (uint8_t)raw[1] == 0xB5 &&
+      (uint8_t)raw[2] == 0x2F && (uint8_t)raw[3] == 0xFD)
+    return zstd_decompress_str(raw);
+  return raw;
+}
 
 std::string read_file(const std::string& path) {
   std::ifstream in(path, std::ios::binary);
@@ -173,7 +214,7 @@ CatalogPlanInput catalog_plan(const std::string& stream_dir,
     }
     std::string mpath = item->at("manifest_path").s;
     if (!mpath.empty() && mpath[0] != '/') mpath = root + "/" + mpath;
-    JPtr man = JsonParser(read_file(mpath)).parse();
+    JPtr man = JsonParser(decode_manifest_text(read_file(mpath))).parse();
     for (const auto& fe : man->at("files").arr) {
       if (file_pruned(*fe, preds, n_preds)) continue;
       kept_files.push_back(fe);

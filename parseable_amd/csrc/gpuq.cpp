@@ -272,6 +272,8 @@ struct Partition {
   // per-pred row ranges still needing per-row evaluation (chunk-stats
   // elision, pred_all_true): merged-adjacent [start, start+len) pairs
   std::map<int, std::vector<std::pair<int64_t, int64_t>>> pred_ranges;
+  // host-decompressed page images -> dec arena at load (snappy fallback)
+  std::vector<std::pair<uint64_t, std::vector<uint8_t>>> himgs;
 
   // device state
   bool loaded = false;
@@ -496,6 +498,29 @@ bool eval_str_pred(const gpuq_pred& p, const std::string& lit,
     case GPUQ_GE: return c >= 0;
   }
   return false;
+}
+
+// decompress one page payload on the host (dict pages / plan-time walks):
+// LZ4_RAW, snappy, or stored raw
+const uint8_t* host_page_payload(int codec, const uint8_t* src, uint32_t comp,
+                                 uint32_t uncomp, std::vector<uint8_t>& buf) {
+  if (codec == CODEC_UNCOMPRESSED) return src;
+  buf.resize(uncomp);
+  if (codec == CODEC_LZ4_RAW) {
+    // ALWAYS try LZ4 first: the writer compresses every v1 page, and
+    // comp_size may coincidentally equal uncomp_size; only a failed decode
+    // of an equal-size page means the page was stored raw
+    int n = lz4_decompress_host(src, comp, buf.data(), buf.size());
+    if (n == (int)uncomp) return buf.data();
+    if (comp == uncomp) return src;
+    throw std::runtime_error("page lz4 failure");
+  }
+  if (codec == CODEC_SNAPPY) {
+    int n = snappy_decompress_host(src, comp, buf.data(), buf.size());
+    if (n == (int)uncomp) return buf.data();
+    throw std::runtime_error("page snappy failure");
+  }
+  throw std::runtime_error("unsupported codec");
 }
 
 // parse a v1 data page's def levels (bit-width-1 RLE/bit-packed hybrid,
@@ -1000,21 +1025,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
       // host-side dictionary processing
       for (auto& pi : t.pages) {
         if (pi.type != PAGE_DICT) continue;
-        std::vector<uint8_t> dbuf(pi.uncomp_size);
-        const uint8_t* d;
-        if (cm.codec == CODEC_UNCOMPRESSED) {
-          d = mf.data + pi.payload_off;
-        } else if (cm.codec == CODEC_LZ4_RAW) {
-          // ALWAYS try LZ4 first: the writer compresses every v1 page, and
-          // comp_size may coincidentally equal uncomp_size (seen in golden
-          // data); only a failed decode of an equal-size page means the
-          // page was stored raw (robustness for other writers).
-          int n = lz4_decompress_host(mf.data + pi.payload_off, pi.comp_size,
-                                      dbuf.data(), dbuf.size());
-          if (n == pi.uncomp_size) d = dbuf.data();
-          else if (pi.comp_size == pi.uncomp_size) d = mf.data + pi.payload_off;
-          else throw std::runtime_error("dict page lz4 failure");
-        } else throw std::runtime_error("unsupported codec");
+        std::vector<uint8_t> dbuf;
+        const uint8_t* d = host_page_payload(cm.codec, mf.data + pi.payload_off,
+                                             pi.comp_size, pi.uncomp_size, dbuf);
         if (c.phys == PT_BYTE_ARRAY) {
           const uint8_t* q = d;
           if (!c.hash_mode && (c.need_gid || c.need_rank))
@@ -1063,17 +1076,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
         }
         for (auto& pi : t.pages) {
           if (pi.type != PAGE_DATA || pi.encoding != ENC_PLAIN) continue;
-          const uint8_t* data = mf.data + pi.payload_off;
-          if (cm.codec == CODEC_LZ4_RAW) {
-            img.resize(pi.uncomp_size);
-            int n2 = lz4_decompress_host(data, pi.comp_size, img.data(),
-                                         img.size());
-            if (n2 == pi.uncomp_size) data = img.data();
-            else if (pi.comp_size != pi.uncomp_size)
-              throw std::runtime_error("plain page lz4 failure (hash col)");
-          } else if (cm.codec != CODEC_UNCOMPRESSED) {
-            throw std::runtime_error("unsupported codec");
-          }
+          const uint8_t* data = host_page_payload(
+              cm.codec, mf.data + pi.payload_off, pi.comp_size,
+              pi.uncomp_size, img);
           uint32_t pos = 0;
           const uint32_t nv = pi.num_values;
           bool has_def = false;
@@ -1187,7 +1192,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
       int32_t comp, uncomp, page_id;
       uint64_t dst_off;
       uint32_t num_values;
-      bool optional, raw_codec;
+      bool optional;
+      int codec;
       int col;
     };
     std::vector<CItem> citems;
@@ -1202,6 +1208,9 @@ extern "C" gpuq_plan* gpuq_plan_build(
       std::vector<DevBrInl> brinl;
       std::vector<DevLit> lits_lane, lits_wave;
       std::vector<CItem> citems;
+      // host-decompressed page images staged straight into the dec arena
+      // (snappy piece-explosion fallback — no serial snappy device kernel)
+      std::vector<std::pair<uint64_t, std::vector<uint8_t>>> himgs;
     };
     std::vector<CB> cbs(part.chunks.size());
     parallel_for(part.chunks.size(), [&](size_t ti) {
@@ -1218,11 +1227,24 @@ extern "C" gpuq_plan* gpuq_plan_build(
 
       // decompress planning for one page image (data or hash-col dict):
       // segment walk + litpar/backref records, exactly the round-1 logic
+      const int chunk_codec = t.cm->codec;
       auto plan_decomp = [&](const uint8_t* praw, uint64_t src_abs,
                              uint64_t dst_abs, uint32_t comp, uint32_t uncomp,
                              bool raw_page) {
         Lz4Plan lp;
-        if (!raw_page) {
+        if (!raw_page && chunk_codec == CODEC_SNAPPY) {
+          lp = snappy_walk(praw, comp, uncomp);
+          if (lp.fallback) {
+            // piece explosion (rare): host-decompress the page and stage
+            // its image straight into the dec arena at load
+            std::vector<uint8_t> img(uncomp);
+            if (snappy_decompress_host(praw, comp, img.data(), uncomp) !=
+                (int)uncomp)
+              throw std::runtime_error("snappy page decode failure");
+            cb.himgs.emplace_back(dst_abs, std::move(img));
+            return;
+          }
+        } else if (!raw_page) {
           // test knob: exercise the serial windowed fallback kernel on
           // arbitrary content (no organic fixture produces a
           // piece-explosion page)
@@ -1409,7 +1431,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
             cb.citems.push_back({mf.data + pi.payload_off, pi.comp_size,
                                  pi.uncomp_size, this_pid, dp.dst_off,
                                  (uint32_t)pi.num_values, dp.optional != 0,
-                                 dp.raw_copy != 0, t.col_idx});
+                                 t.cm->codec, t.col_idx});
           } else throw std::runtime_error("unsupported encoding for string predicate");
         }
         cb.pages.push_back(dp);
@@ -1450,6 +1472,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       part.lits_wave.insert(part.lits_wave.end(), cb.lits_wave.begin(),
                             cb.lits_wave.end());
       citems.insert(citems.end(), cb.citems.begin(), cb.citems.end());
+      for (auto& h : cb.himgs) part.himgs.emplace_back(h.first, std::move(h.second));
     }
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
 
@@ -1472,12 +1495,13 @@ extern "C" gpuq_plan* gpuq_plan_build(
           if (i >= citems.size() || failed.load()) return;
           const auto& it = citems[i];
           auto& L = locals[i];
-          const uint8_t* data = it.praw;
-          if (!it.raw_codec) {
-            img.resize(it.uncomp);
-            if (lz4_decompress_host(it.praw, it.comp, img.data(), img.size())
-                != it.uncomp) { failed.store(true); return; }
-            data = img.data();
+          const uint8_t* data;
+          try {
+            data = host_page_payload(it.codec, it.praw, it.comp, it.uncomp,
+                                     img);
+          } catch (const std::exception&) {
+            failed.store(true);
+            return;
           }
           uint32_t pos = 0;
           const uint32_t nv = it.num_values;
@@ -1791,6 +1815,9 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
       HIP_TRY(hipMemcpyAsync(part.d_dec + t.dec_base, src, t.dec_len,
                              hipMemcpyDeviceToDevice, part.stream));
     }
+    // host-decompressed page images (snappy fallback pages)
+    for (auto& h : part.himgs)
+      ring.copy(part.d_dec + h.first, h.second.data(), h.second.size());
   }
   HIP_TRY(hipMalloc(&part.d_pages, std::max<size_t>(part.pages.size() * sizeof(DevPage), 16)));
   HIP_TRY(hipMemcpyAsync(part.d_pages, part.pages.data(),

@@ -279,6 +279,203 @@ int lz4_decompress_host(const uint8_t* src, size_t src_len,
   return (int)(dp - dst);
 }
 
+// ---- snappy (parquet codec 1) ----------------------------------------
+// Raw snappy block format: uvarint uncompressed length, then tagged
+// elements (literals; copies with 1/2/4-byte offsets, overlap = pattern
+// repeat exactly like LZ4). The reference's test deployments write snappy
+// (docker-compose-test.yaml:45; the codec is a knob, cli.rs:484-491).
+
+int snappy_decompress_host(const uint8_t* src, size_t comp, uint8_t* dst,
+                           size_t dst_cap) {
+  size_t s = 0, d = 0;
+  uint64_t ulen = 0;
+  int sh = 0;
+  for (;;) {
+    if (s >= comp) return -1;
+    uint8_t b = src[s++];
+    ulen |= (uint64_t)(b & 0x7f) << sh;
+    if (!(b & 0x80)) break;
+    sh += 7;
+  }
+  if (ulen > dst_cap) return -1;
+  while (s < comp) {
+    uint8_t tag = src[s++];
+    int type = tag & 3;
+    if (type == 0) {  // literal
+      size_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        int extra = (int)len - 60;
+        if (s + extra > comp) return -1;
+        len = 0;
+        for (int i = 0; i < extra; i++) len |= (size_t)src[s + i] << (8 * i);
+        len += 1;
+        s += extra;
+      }
+      if (s + len > comp || d + len > dst_cap) return -1;
+      memcpy(dst + d, src + s, len);
+      s += len;
+      d += len;
+    } else {
+      size_t ml, off;
+      if (type == 1) {
+        ml = ((tag >> 2) & 7) + 4;
+        if (s >= comp) return -1;
+        off = ((size_t)(tag >> 5) << 8) | src[s++];
+      } else if (type == 2) {
+        ml = (tag >> 2) + 1;
+        if (s + 2 > comp) return -1;
+        off = src[s] | ((size_t)src[s + 1] << 8);
+        s += 2;
+      } else {
+        ml = (tag >> 2) + 1;
+        if (s + 4 > comp) return -1;
+        off = src[s] | ((size_t)src[s + 1] << 8) | ((size_t)src[s + 2] << 16) |
+              ((size_t)src[s + 3] << 24);
+        s += 4;
+      }
+      if (off == 0 || off > d || d + ml > dst_cap) return -1;
+      const uint8_t* mp = dst + d - off;
+      for (size_t i = 0; i < ml; i++) dst[d + i] = mp[i];  // overlap repeats
+      d += ml;
+    }
+  }
+  return (int)d;
+}
+
+// Snappy pages ride the litpar machinery unchanged: every literal becomes a
+// raw->dec copy record, every copy a host-resolved match (same interval-map
+// composition as lz4_walk's litpar mode) — the GPU never parses snappy at
+// all. fallback=true on piece explosion (caller host-decompresses the page
+// and stages the image directly).
+Lz4Plan snappy_walk(const uint8_t* src, size_t comp, size_t uncomp) {
+  Lz4Plan plan;
+  plan.litpar = true;
+  size_t s = 0, d = 0;
+
+  struct MapEntry { uint32_t start, len, off, piece_start, piece_n; };
+  std::vector<MapEntry> M;
+  constexpr int MAX_PIECES_PER_RECORD = 64;
+  auto entry_at = [&](uint32_t x) -> size_t {
+    size_t lo = 0, hi = M.size();
+    while (lo < hi) {
+      size_t mid = (lo + hi) / 2;
+      if (M[mid].start + M[mid].len > x) hi = mid;
+      else lo = mid + 1;
+    }
+    return lo;
+  };
+  auto resolve = [&](uint32_t a, uint32_t b, std::vector<Lz4Piece>& out) -> bool {
+    size_t i = entry_at(a);
+    uint32_t x = a;
+    while (x < b) {
+      if ((int)out.size() > MAX_PIECES_PER_RECORD) return false;
+      if (i >= M.size() || M[i].start >= b) {
+        out.push_back({x, b - x});
+        break;
+      }
+      const MapEntry& e = M[i];
+      if (x < e.start) {
+        out.push_back({x, e.start - x});
+        x = e.start;
+        continue;
+      }
+      uint32_t y = std::min<uint32_t>(b, e.start + e.len);
+      while (x < y) {
+        if ((int)out.size() > MAX_PIECES_PER_RECORD) return false;
+        uint32_t p = (x - e.start) % e.off;
+        uint32_t chunk = std::min(y - x, e.off - p);
+        uint32_t po = 0, q = p, left = chunk;
+        for (uint32_t k = 0; k < e.piece_n && left; k++) {
+          const Lz4Piece& pc = plan.pieces[e.piece_start + k];
+          if (q < po + pc.len) {
+            uint32_t within = q - po;
+            uint32_t take = std::min(pc.len - within, left);
+            out.push_back({pc.src + within, take});
+            q += take;
+            left -= take;
+          }
+          po += pc.len;
+        }
+        if (left) return false;
+        x += chunk;
+      }
+      i++;
+    }
+    return true;
+  };
+  auto defer_match = [&](size_t dst, size_t off, size_t ml) {
+    uint32_t pat = (uint32_t)std::min(off, ml);
+    std::vector<Lz4Piece> pieces;
+    if (resolve((uint32_t)(dst - off), (uint32_t)(dst - off + pat), pieces)) {
+      uint32_t ps = (uint32_t)plan.pieces.size();
+      plan.pieces.insert(plan.pieces.end(), pieces.begin(), pieces.end());
+      plan.resolved.push_back({(uint32_t)dst, (uint32_t)ml, (uint32_t)off,
+                               ps, (uint32_t)pieces.size()});
+      M.push_back({(uint32_t)dst, (uint32_t)ml, (uint32_t)off, ps,
+                   (uint32_t)pieces.size()});
+    } else {
+      plan.fallback = true;
+    }
+  };
+
+  uint64_t ulen = 0;
+  int sh = 0;
+  for (;;) {
+    if (s >= comp) throw std::runtime_error("snappy walk: eof in preamble");
+    uint8_t b = src[s++];
+    ulen |= (uint64_t)(b & 0x7f) << sh;
+    if (!(b & 0x80)) break;
+    sh += 7;
+  }
+  if (ulen != uncomp) throw std::runtime_error("snappy walk: size mismatch");
+  while (s < comp) {
+    plan.n_seq++;
+    uint8_t tag = src[s++];
+    int type = tag & 3;
+    if (type == 0) {
+      size_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        int extra = (int)len - 60;
+        if (s + extra > comp) throw std::runtime_error("snappy walk: eof");
+        len = 0;
+        for (int i = 0; i < extra; i++) len |= (size_t)src[s + i] << (8 * i);
+        len += 1;
+        s += extra;
+      }
+      if (s + len > comp || d + len > uncomp)
+        throw std::runtime_error("snappy walk: overrun");
+      plan.lits.push_back({(uint32_t)d, (uint32_t)s, (uint32_t)len});
+      s += len;
+      d += len;
+    } else {
+      size_t ml, off;
+      if (type == 1) {
+        ml = ((tag >> 2) & 7) + 4;
+        if (s >= comp) throw std::runtime_error("snappy walk: eof");
+        off = ((size_t)(tag >> 5) << 8) | src[s++];
+      } else if (type == 2) {
+        ml = (tag >> 2) + 1;
+        if (s + 2 > comp) throw std::runtime_error("snappy walk: eof");
+        off = src[s] | ((size_t)src[s + 1] << 8);
+        s += 2;
+      } else {
+        ml = (tag >> 2) + 1;
+        if (s + 4 > comp) throw std::runtime_error("snappy walk: eof");
+        off = src[s] | ((size_t)src[s + 1] << 8) | ((size_t)src[s + 2] << 16) |
+              ((size_t)src[s + 3] << 24);
+        s += 4;
+      }
+      if (off == 0 || off > d || d + ml > uncomp)
+        throw std::runtime_error("snappy walk: bad copy");
+      defer_match(d, off, ml);
+      if (plan.fallback) return plan;
+      d += ml;
+    }
+  }
+  if (d != uncomp) throw std::runtime_error("snappy walk: size mismatch");
+  return plan;
+}
+
 Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
                  uint32_t seg_max, bool litpar) {
   Lz4Plan plan;

@@ -139,4 +139,10 @@ struct Lz4Plan {
 Lz4Plan lz4_walk(const uint8_t* src, size_t comp, size_t uncomp,
                  uint32_t seg_max, bool litpar = false);
 
+// Snappy (parquet codec 1): scalar host decode, and the litpar-only walk
+// whose records ride the SAME device kernels as LZ4's litpar mode.
+int snappy_decompress_host(const uint8_t* src, size_t comp, uint8_t* dst,
+                           size_t dst_cap);
+Lz4Plan snappy_walk(const uint8_t* src, size_t comp, size_t uncomp);
+
 }  // namespace gpuq

@@ -124,6 +124,42 @@ def _file_pruned(file_entry, preds):
     return False
 
 
+ZSTD_MAGIC = b"\x28\xb5\x2f\xfd"
+
+
+def _zstd_decompress(raw: bytes) -> bytes:
+    """zstd frame decode via the system libzstd (the reference binds the
+    same library through the zstd crate)."""
+    import ctypes
+
+    z = ctypes.CDLL("libzstd.so.1")
+    z.ZSTD_getFrameContentSize.restype = ctypes.c_ulonglong
+    z.ZSTD_decompress.restype = ctypes.c_size_t
+    z.ZSTD_isError.restype = ctypes.c_uint
+    need = z.ZSTD_getFrameContentSize(raw, len(raw))
+    cap = len(raw) * 32 + (1 << 20) if need + 1 < 2 else int(need)
+    while True:
+        buf = ctypes.create_string_buffer(cap)
+        n = z.ZSTD_decompress(buf, cap, raw, len(raw))
+        if not z.ZSTD_isError(n):
+            return buf.raw[:n]
+        if need + 1 >= 2 or cap > 1 << 32:
+            raise RuntimeError("zstd manifest decompression failed")
+        cap *= 4
+
+
+def _load_manifest(path):
+    """Manifest bytes -> dict, sniffing the zstd magic exactly as the
+    reference's decode_manifest (catalog/manifest.rs:53-110): compressed and
+    plain JSON manifests share names; the leading frame magic is the only
+    distinction."""
+    with open(path, "rb") as fh:
+        raw = fh.read()
+    if raw[:4] == ZSTD_MAGIC:
+        raw = _zstd_decompress(raw)
+    return json.loads(raw)
+
+
 def _parse_iso_ms(s):
     """chrono serde emits both '...%S.%fZ' and (when the fractional part is
     zero) '...%SZ' — accept both, matching catalog.cpp parse_iso_ms."""
@@ -440,9 +476,7 @@ class StandardTableProvider:
             mpath = it["manifest_path"]
             if not os.path.isabs(mpath):
                 mpath = os.path.join(os.path.dirname(self.stream_dir), mpath)
-            with open(mpath) as fh:
-                m = json.load(fh)
-            files.extend(m["files"])
+            files.extend(_load_manifest(mpath)["files"])
         return files
 
     def scan(self, query: dict):

@@ -231,6 +231,24 @@ GOLDEN_QUERIES["g_c5"] = [
     }),
 ]
 
+GOLDEN_QUERIES["g_c1snap"] = [
+    ("snap_count_by_level", {
+        "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+        "group_by": ["level"],
+    }),
+    ("snap_between_pred", {
+        "select": [{"agg": "count_star"}, {"agg": "sum", "col": "latency"}],
+        "group_by": ["host"],
+        "preds": [{"col": "p_timestamp", "op": "between",
+                   "lo": BASE + MIN // 2, "hi": BASE + 2 * MIN}],
+    }),
+    ("snap_sum_f64", {
+        "select": [{"agg": "sum", "col": "f_f64"}, {"agg": "count_star"}],
+        "group_by": ["f_str2"],
+        "preds": [{"col": "level", "op": "eq", "lit": "INFO"}],
+    }),
+]
+
 GOLDEN_FIXTURES = {
     "g_c1":       dict(config="c1", rows=120_000, rows_per_file=40_000, seed=1001),
     "g_c1_pages": dict(config="c1", rows=40_000, rows_per_file=40_000, seed=1002,
@@ -242,4 +260,9 @@ GOLDEN_FIXTURES = {
     # tiny dictionary_pagesize_limit); exercises the device hash group-by,
     # strref predicates and strref min/max
     "g_c5":       dict(config="c5", rows=60_000, rows_per_file=20_000, seed=1006),
+    # snappy page codec (the reference's test-compose codec,
+    # docker-compose-test.yaml:45) + zstd-compressed manifests
+    # (catalog/manifest.rs:53-110)
+    "g_c1snap":   dict(config="c1", rows=60_000, rows_per_file=20_000,
+                       seed=1007, compression="snappy", manifest_codec="zstd"),
 }
