@@ -456,6 +456,51 @@ class StandardTableProvider:
         boundary = (self.now_ms - self.staging_window_ms) // 60_000 * 60_000
         return time_range[1] >= boundary
 
+    # Legacy listing branch (SURVEY §2 (★); listing_table_builder.rs:46-118
+    # + is_overlapping_query / return_listing_time_filters,
+    # stream_schema_provider.rs:672-689,845-930): data older than the first
+    # manifest is discovered by minute-granularity prefix listing
+    # (date=/hour=/minute=) and joins the same scan plan — per-row time
+    # filters stay exact on the engine side.
+    def _first_manifest_lower(self):
+        items = self.snapshot["manifest_list"]
+        if not items:
+            return None
+        return min(_parse_iso_ms(it["time_lower_bound"]) for it in items)
+
+    def _legacy_listing_files(self, lo_ms, hi_ms):
+        from datetime import datetime, timezone
+
+        files = []
+        minute = lo_ms // 60_000 * 60_000
+        while minute < hi_ms:
+            dt = datetime.fromtimestamp(minute / 1000, tz=timezone.utc)
+            prefix = os.path.join(self.stream_dir, f"date={dt:%Y-%m-%d}",
+                                  f"hour={dt:%H}", f"minute={dt:%M}")
+            if os.path.isdir(prefix):
+                files.extend(
+                    os.path.join(prefix, f) for f in os.listdir(prefix)
+                    if f.endswith(".parquet"))
+            minute += 60_000
+        files.sort(reverse=True)  # listing.sorted().rev()
+        return files
+
+    def _legacy_files(self, time_range):
+        """Files for the pre-manifest slice of the query range, or []."""
+        first_lower = self._first_manifest_lower()
+        if first_lower is None:
+            # no manifests at all: everything rides the listing — an explicit
+            # range is required (listing_table_builder.rs:79-84)
+            if time_range is None:
+                raise GpuqError(
+                    "time predicate required to query pre-manifest data "
+                    "(listing_table_builder.rs:79-84)")
+            return self._legacy_listing_files(time_range[0], time_range[1])
+        if time_range is None or time_range[0] >= first_lower:
+            return []  # is_overlapping_query == false
+        return self._legacy_listing_files(time_range[0],
+                                          min(time_range[1], first_lower))
+
     # Snapshot::manifests (catalog/snapshot.rs:42-71): retain manifests whose
     # [lower,upper] overlaps the time predicates.
     def _select_manifests(self, time_range):
@@ -488,6 +533,11 @@ class StandardTableProvider:
         staging_hit = self._staging_touches(query.get("time_range"))
         if staging_hit:
             return self._scan_with_staging(query)
+
+        legacy_paths = self._legacy_files(query.get("time_range"))
+
+        if legacy_paths:
+            return self._scan_with_legacy(query, legacy_paths)
 
         if self.session is not None and not os.environ.get("GPUQ_PY_PLANNER"):
             # native catalog planner inside libgpuq (catalog.cpp)
@@ -575,6 +625,29 @@ class StandardTableProvider:
         paths.extend(scanner.parquet_paths())  # scanned unconditionally
         gpu_plan = GpuExecutionPlan(self.session, paths, query) if paths else None
         return StagedPlan(gpu_plan, scanner, query)
+
+    def _scan_with_legacy(self, query, legacy_paths):
+        """Pre-manifest files join the manifested scan: legacy files carry
+        no manifest stats (only footer stats prune them), and the count fast
+        path is skipped — their row counts are in no manifest
+        (stream_schema_provider.rs:672-689)."""
+        preds = list(query.get("preds", []))
+        time_range = query.get("time_range")
+        files = self._manifest_files(time_range)
+        prune_preds = preds.copy()
+        if time_range:
+            prune_preds.append({"col": "p_timestamp", "op": "between",
+                                "lo": time_range[0], "hi": time_range[1],
+                                "hi_exclusive": True})
+        kept = [f for f in files if not _file_pruned(f, prune_preds)]
+        root = os.path.dirname(self.stream_dir)
+        paths = list(legacy_paths)
+        for fe in kept:
+            p = fe["file_path"]
+            paths.append(p if os.path.isabs(p) else os.path.join(root, p))
+        if not paths:
+            return EmptyScanResult(query)
+        return GpuExecutionPlan(self.session, paths, query)
 
 
 def _build_c_projection(query: dict, keep):

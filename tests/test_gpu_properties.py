@@ -194,3 +194,23 @@ def test_unlimited_projection_stream(stream):
     # ties at equal ts are engine-defined: compare as multisets
     assert sorted(map(tuple, got)) == sorted(map(tuple, want))
     plan.close()
+
+
+def test_legacy_listing_gpu_parity(tmp_path_factory):
+    """Pre-manifest files (prefix listing) + manifested files answer as one
+    scan, matching the oracle over all four files."""
+    from oracle import query_oracle as qo
+    from oracle.compare import assert_rows_equal
+    from parseable_amd import GpuSession, Query, StandardTableProvider
+    from tests.test_provider import _make_legacy_stream
+
+    td = tmp_path_factory.mktemp("legacy")
+    info, base, minute = _make_legacy_stream(td)
+    prov = StandardTableProvider(info["stream_dir"], GpuSession())
+    q = {"select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+         "group_by": ["level"],
+         "time_range": [base, base + 4 * minute]}
+    rows, _ = Query(prov).execute(dict(q))
+    want = qo.execute(info["files"], dict(q))["rows"]
+    assert_rows_equal(rows, want, "legacy listing union")
+    assert sum(r[1] for r in rows) == 16_000

@@ -241,3 +241,72 @@ def test_zstd_manifest_roundtrip(tmp_path):
         assert plan.fast_count == 12_000
     finally:
         del os.environ["GPUQ_FAKE_DEVICE"]
+
+
+def _make_legacy_stream(tmp_path):
+    """A stream whose FIRST day-of-files predates every manifest: generate 4
+    minutes, then rewrite the manifest/snapshot to cover only minutes 2-3 —
+    minutes 0-1 become pre-manifest data reachable only via prefix listing
+    (listing_table_builder.rs:46-118)."""
+    import json as _json
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from datagen.gen import BASE_TS_MS, MINUTE_MS, gen_stream
+
+    info = gen_stream(str(tmp_path), "s", "c1", rows=16_000, rows_per_file=4_000,
+                      seed=21)
+    mdir = os.path.join(str(tmp_path), "s", "date=2025-09-01")
+    with open(os.path.join(mdir, "manifest.json")) as fh:
+        man = _json.load(fh)
+    kept = [f for f in man["files"]
+            if "minute=02" in f["file_path"] or "minute=03" in f["file_path"]]
+    assert len(kept) == 2
+    with open(os.path.join(mdir, "manifest.json"), "w") as fh:
+        _json.dump({"version": "v2", "files": kept}, fh)
+    snap_path = os.path.join(info["stream_dir"], "stream.json")
+    with open(snap_path) as fh:
+        sj = _json.load(fh)
+    item = sj["snapshot"]["manifest_list"][0]
+    lo = BASE_TS_MS + 2 * MINUTE_MS
+    from datetime import datetime, timezone
+
+    item["time_lower_bound"] = datetime.fromtimestamp(
+        lo / 1000, tz=timezone.utc).strftime("%Y-%m-%dT%H:%M:%S.%f") + "Z"
+    with open(snap_path, "w") as fh:
+        _json.dump(sj, fh)
+    return info, BASE_TS_MS, MINUTE_MS
+
+
+def test_legacy_listing_discovers_premanifest_files(tmp_path):
+    from parseable_amd.provider import EmptyScanResult
+
+    info, base, minute = _make_legacy_stream(tmp_path)
+    prov = StandardTableProvider(info["stream_dir"], None)
+    # range covering everything: 2 legacy files (listed) + 2 manifested
+    q = {"select": [{"agg": "count_star"}],
+         "time_range": [base, base + 4 * minute]}
+    legacy = prov._legacy_files(q["time_range"])
+    assert len(legacy) == 2
+    assert all("minute=00" in p or "minute=01" in p for p in legacy)
+    # range entirely inside the manifested window: no listing
+    assert prov._legacy_files([base + 2 * minute, base + 4 * minute]) == []
+    # range entirely pre-manifest: listing only, manifests all pruned
+    assert len(prov._legacy_files([base, base + 2 * minute])) == 2
+
+
+def test_legacy_listing_requires_time_range_when_no_manifests(tmp_path):
+    import json as _json
+
+    sdir = tmp_path / "stream"
+    sdir.mkdir()
+    (sdir / "stream.json").write_text(_json.dumps(
+        {"snapshot": {"manifest_list": []}}))
+    prov = StandardTableProvider(str(sdir), None)
+    import pytest as _pytest
+
+    from parseable_amd.provider import GpuqError
+
+    with _pytest.raises(GpuqError):
+        prov.scan({"select": [{"agg": "count_star"}]})
