@@ -108,6 +108,7 @@ struct AggArgs {
 
 // error codes written to *d_error by kernels
 enum { ERR_NONE = 0, ERR_LZ4 = 1, ERR_RLE = 2, ERR_DELTA = 3, ERR_DICT_RANGE = 4,
-       ERR_PAGE = 5, ERR_FSUM_RANGE = 6 };
+       ERR_PAGE = 5, ERR_FSUM_RANGE = 6, ERR_HASH_CAP = 7, ERR_HASH_PROBE = 8,
+       ERR_PAIR_CAP = 9, ERR_PAIR_PROBE = 10 };
 
 }  // namespace gpuq

@@ -83,11 +83,14 @@ struct gpuq_ctx {
         (void)hipFree(kv.second.dev);
       }
   }
-  // returns true and pins the entry when present
-  bool cache_pin(const std::string& key) {
+  // returns true and pins the entry when present AND its image length
+  // matches the requesting plan's span (defense in depth: the key already
+  // encodes the layout mode, see the ":h" suffix for hash-mode chunks
+  // whose span includes the dict-page image)
+  bool cache_pin(const std::string& key, size_t expect_bytes) {
     std::lock_guard<std::mutex> g(cache_mu);
     auto it = cache.find(key);
-    if (it == cache.end()) return false;
+    if (it == cache.end() || it->second.bytes != expect_bytes) return false;
     it->second.refs++;
     it->second.last_use = ++cache_clock;
     return true;
@@ -1174,10 +1177,14 @@ extern "C" gpuq_plan* gpuq_plan_build(
       // hot tier: the decompressed chunk image may already be resident
       {
         const auto& mf2 = *plan->files[t.file_idx];
+        // layout-qualified key: a hash-mode chunk's image includes the
+        // decompressed dict page, a dict/LUT-mode chunk's does not — the
+        // same (file, rg, col) caches separately per layout
         t.cache_key = mf2.path + "\x01" + std::to_string(mf2.size) + ":" +
                       std::to_string(mf2.mtime_ns) + ":" +
-                      std::to_string(t.rg_idx) + ":" + c.name;
-        if (ctx->cache_pin(t.cache_key)) {
+                      std::to_string(t.rg_idx) + ":" + c.name +
+                      (c.hash_mode ? ":h" : "");
+        if (ctx->cache_pin(t.cache_key, (size_t)t.dec_len)) {
           t.cached = true;
           plan->pinned_keys.push_back(t.cache_key);
           part.bytes_cache_hit += (int64_t)t.cm->total_compressed_size;

@@ -1051,13 +1051,13 @@ __global__ void k_hash_build(const uint8_t* __restrict__ dec,
     uint64_t ref = (uint64_t)refs[i];
     uint64_t slot = ref_hash(dec, ref) & mask;
     for (uint32_t probe = 0; ; probe++) {
-      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_HASH_PROBE); return; }
       uint64_t old = atomicCAS((unsigned long long*)&hkeys[slot],
                                (unsigned long long)HREF_EMPTY,
                                (unsigned long long)ref);
       if (old == HREF_EMPTY) {
         uint32_t g = atomicAdd(counter, 1u);
-        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_HASH_CAP); return; }
         gid2ref[g] = ref;
         hgids[slot] = (int32_t)g;
         break;
@@ -1117,13 +1117,13 @@ __global__ void k_pair_build(const int32_t* __restrict__ a,
     uint64_t key = ((uint64_t)(uint32_t)a[i] << 32) | (uint32_t)b[i];
     uint64_t slot = mix64(key) & mask;
     for (uint32_t probe = 0;; probe++) {
-      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_PAIR_PROBE); return; }
       uint64_t old = atomicCAS((unsigned long long*)&hkeys[slot],
                                (unsigned long long)HREF_EMPTY,
                                (unsigned long long)key);
       if (old == HREF_EMPTY) {
         uint32_t g = atomicAdd(counter, 1u);
-        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_DICT_RANGE); return; }
+        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_PAIR_CAP); return; }
         gid2pair[g] = key;
         hgids[slot] = (int32_t)g;
         break;

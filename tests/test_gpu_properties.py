@@ -214,3 +214,29 @@ def test_legacy_listing_gpu_parity(tmp_path_factory):
     want = qo.execute(info["files"], dict(q))["rows"]
     assert_rows_equal(rows, want, "legacy listing union")
     assert sum(r[1] for r in rows) == 16_000
+
+
+def test_hot_tier_layout_modes_do_not_collide(hash_stream):
+    """Regression (fuzz seed 44): a chunk cached under HASH-mode layout
+    (dict-page image inside its arena span) must not serve a later
+    dict/LUT-mode plan for the same (file, rg, col) — the spans differ and
+    a prefix copy corrupts the dict indices. Keys are layout-qualified."""
+    from oracle import query_oracle as qo
+    from oracle.compare import assert_rows_equal
+    from parseable_amd import GpuSession, Query, StandardTableProvider
+
+    sess = GpuSession()
+    prov = StandardTableProvider(hash_stream["stream_dir"], sess)
+    q1 = {"select": [{"agg": "count_star"}], "group_by": ["trace"]}  # hash
+    rows1, _ = Query(prov).execute(dict(q1))
+    assert sum(r[1] for r in rows1) == 1_500_000
+    # same column, NON-hash layout (contains-only -> window/LUT path)
+    q2 = {"select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"},
+                     {"agg": "count", "col": "opt_tag"}],
+          "preds": [{"col": "trace", "op": "contains", "lit": "42"}]}
+    rows2, _ = Query(prov).execute(dict(q2))
+    want = qo.execute(hash_stream["files"], dict(q2))["rows"]
+    assert_rows_equal(rows2, want, "layout-mode cache collision")
+    # and back to hash layout again (its own cache entry, now warm)
+    rows3, _ = Query(prov).execute(dict(q1))
+    assert rows3 == rows1
