@@ -75,6 +75,29 @@ def _dict_col(rng, n, values, s=1.2):
     )
 
 
+def _msg_col(rng, n):
+    """Vectorized c3-shape message column (20-120 B, ~1% contain "error"):
+    numpy byte assembly + Arrow buffers — the per-row Python string path is
+    ~30x slower at the 1 B-row bench scale."""
+    alphabet = (string.ascii_lowercase + string.digits).encode()
+    lut = np.frombuffer(alphabet, dtype=np.uint8)
+    lens = rng.integers(20, 121, n, dtype=np.int64)
+    offsets = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(lens, out=offsets[1:])
+    total = int(offsets[-1])
+    chars = lut[rng.integers(0, len(alphabet), total, dtype=np.int64)]
+    hit = rng.random(n) < 0.01
+    pos_in = rng.integers(0, 15, n, dtype=np.int64)
+    starts = offsets[:-1][hit] + (pos_in[hit] % np.maximum(lens[hit] - 5, 1))
+    if len(starts):
+        idx = (starts[:, None] + np.arange(5)).ravel()
+        chars[idx] = np.tile(np.frombuffer(b"error", dtype=np.uint8),
+                             len(starts))
+    return pa.StringArray.from_buffers(
+        n, pa.py_buffer(offsets.astype(np.int32).tobytes()),
+        pa.py_buffer(chars.tobytes()))
+
+
 def _minute_batch(config: str, rng: np.random.Generator, n: int, minute: int):
     """One file's rows. Timestamps descending within the minute
     (src/parseable/streams.rs:756-760: SortingColumn time DESC)."""
@@ -110,6 +133,11 @@ def _minute_batch(config: str, rng: np.random.Generator, n: int, minute: int):
                 for i, m in enumerate(msgs)
             ]
             cols["message"] = pa.array(msgs)
+    elif config == "c3b":
+        # the BASELINE c3 bench shape, slim: 1 B rows of (p_timestamp,
+        # message) fit the GPU box's local disk; the LIKE scan reads only
+        # the message column either way (bytes_scanned is identical)
+        cols["message"] = _msg_col(rng, n)
     elif config == "c5":
         # raw-byte utf8 stress: moderate/high-cardinality string columns that
         # overflow the dictionary page (written with a small
