@@ -81,7 +81,9 @@ def main():
                     help="rows per GPU (weak scaling); BASELINE quotes 1B")
     ap.add_argument("--between-ms", type=int, default=0,
                     help="c2s: BETWEEN window width (default ~50%% of range)")
-    ap.add_argument("--data-dir", default=os.environ.get("GPUQ_DATA", "/tmp/gpuq_bench"))
+    ap.add_argument("--data-dir", default=None,
+                    help="default: $GPUQ_DATA, else /dev/shm when large "
+                    "(the 8-rank 1B shards exceed small /tmp overlays)")
     ap.add_argument("--gen-workers", type=int, default=0)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--cpu-baseline-full", action="store_true",
@@ -89,6 +91,17 @@ def main():
                     "all cores) instead of the bounded sample — the "
                     "once-per-round measured baseline (BASELINE.md)")
     args = ap.parse_args()
+
+    if args.data_dir is None:
+        args.data_dir = os.environ.get("GPUQ_DATA")
+    if args.data_dir is None:
+        import shutil
+
+        try:
+            big_shm = shutil.disk_usage("/dev/shm").free > 200e9
+        except OSError:
+            big_shm = False
+        args.data_dir = "/dev/shm/gpuq_bench" if big_shm else "/tmp/gpuq_bench"
 
     import torch
     import torch.distributed as dist
@@ -119,6 +132,8 @@ def main():
     shard = os.path.join(args.data_dir, f"{args.workload}_{args.rows}_r{rank}")
     stream_dir = os.path.join(shard, "stream")
     workers = args.gen_workers or max(8, (os.cpu_count() or 8) // world)
+    # cap per-rank host planning threads too (8 ranks x 256 threads thrash)
+    os.environ.setdefault("GPUQ_HOST_THREADS", str(workers))
     if not os.path.exists(os.path.join(stream_dir, "stream.json")):
         log(f"[bench] generating {args.rows} rows/rank ({n_files} files, "
             f"{workers} workers/rank) under {shard} ...")

@@ -583,6 +583,13 @@ void parallel_for(size_t n, F&& fn, size_t max_threads = 0) {
   if (n == 0) return;
   size_t hw = std::thread::hardware_concurrency();
   if (hw == 0) hw = 8;
+  // one-process-per-GPU launches cap their host planning threads so eight
+  // ranks don't spawn 8 x cores workers (GPUQ_HOST_THREADS; bench.py sets it)
+  static const size_t env_cap = [] {
+    const char* e = getenv("GPUQ_HOST_THREADS");
+    return e ? (size_t)strtoul(e, nullptr, 10) : (size_t)0;
+  }();
+  if (env_cap) hw = std::min(hw, env_cap);
   if (max_threads) hw = std::min(hw, max_threads);
   size_t nthreads = std::min(n, hw);
   if (nthreads <= 1) {
