@@ -46,11 +46,16 @@ struct DevBrRes {      // literal-resolved record (absolute dec offsets)
   uint32_t piece_start, piece_n;
 };
 struct DevPiece { uint64_t src; uint32_t len; uint32_t _pad; };
-// litpar literal copy: raw[src..src+len) -> dec[dst..dst+len)  (absolute)
-struct DevLit { uint64_t src, dst; uint32_t len; uint32_t _pad; };
+// litpar literal copy: raw[src..src+len) -> dec[dst..dst+len), PACKED to
+// 16 B — at 1 B-row scale these record streams are the largest kernel
+// fetch (PMC: 6-8 GB/launch at 24 B each), so the layout is the traffic.
+// a = src | len << 40 (arenas < 1 TB so offsets fit 40 bits; literal runs
+// < 2^24); b = dst.
+struct DevLit { uint64_t a, b; };
 // resolved match whose pattern (<= 8 bytes) the host inlined from the
 // compressed literal bytes: write-only, no scattered pattern reads.
-struct DevBrInl { uint64_t dst, pat; uint32_t len, period; };
+// meta = dst | len << 40 | period << 52 (len <= 256, period <= 8).
+struct DevBrInl { uint64_t meta, pat; };
 // contains window: a value-aligned run of consecutive non-null values of one
 // PLAIN byte-array page, <= CWIN bytes total (or a single oversized value,
 // nbytes > CWIN). Host builds these at load time by decompressing the page

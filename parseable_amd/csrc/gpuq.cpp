@@ -1303,7 +1303,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
           cb.segs.push_back(d2);
         }
         for (const auto& lt : lp.lits) {
-          DevLit dl{src_abs + lt.src, dst_abs + lt.dst, lt.len, 0};
+          DevLit dl{(src_abs + lt.src) | ((uint64_t)lt.len << 40),
+                    dst_abs + lt.dst};
           (lt.len <= 256 ? cb.lits_lane : cb.lits_wave).push_back(dl);
         }
         if (lp.fallback) {
@@ -1347,7 +1348,10 @@ extern "C" gpuq_plan* gpuq_plan_build(
               if (ok) {
                 uint64_t pat;
                 std::memcpy(&pat, buf, 8);
-                cb.brinl.push_back({dst_abs + rr.dst, pat, rr.len, pat_len});
+                cb.brinl.push_back({(dst_abs + rr.dst) |
+                                        ((uint64_t)rr.len << 40) |
+                                        ((uint64_t)pat_len << 52),
+                                    pat});
                 continue;
               }
             }

@@ -241,9 +241,12 @@ __global__ void k_brres_inl(uint8_t* __restrict__ dec,
   int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (r >= n) return;
   const DevBrInl R = recs[r];
-  uint8_t* o = dec + R.dst;
-  for (uint32_t j = 0; j < R.len; j++)
-    o[j] = (uint8_t)(R.pat >> ((j % R.period) * 8));
+  const uint64_t dst = R.meta & ((1ull << 40) - 1);
+  const uint32_t len = (uint32_t)(R.meta >> 40) & 0xfff;
+  const uint32_t period = (uint32_t)(R.meta >> 52);
+  uint8_t* o = dec + dst;
+  for (uint32_t j = 0; j < len; j++)
+    o[j] = (uint8_t)(R.pat >> ((j % period) * 8));
 }
 __global__ void __launch_bounds__(WAVE)
 k_brres_wave(uint8_t* __restrict__ dec, const DevBrRes* __restrict__ recs,
@@ -273,31 +276,33 @@ __global__ void k_lit_lane(const uint8_t* __restrict__ raw,
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const DevLit L = lits[i];
-  const uint8_t* s = raw + L.src;
-  uint8_t* o = dec + L.dst;
+  const uint8_t* s = raw + (L.a & ((1ull << 40) - 1));
+  const uint32_t len = (uint32_t)(L.a >> 40);
+  uint8_t* o = dec + L.b;
   uint32_t k = 0;
-  for (; k + 8 <= L.len; k += 8) {
+  for (; k + 8 <= len; k += 8) {
     uint64_t w;
     __builtin_memcpy(&w, s + k, 8);
     __builtin_memcpy(o + k, &w, 8);
   }
-  for (; k < L.len; k++) o[k] = s[k];
+  for (; k < len; k++) o[k] = s[k];
 }
 __global__ void __launch_bounds__(WAVE)
 k_lit_wave(const uint8_t* __restrict__ raw, uint8_t* __restrict__ dec,
            const DevLit* __restrict__ lits, int n) {
   if (blockIdx.x >= (unsigned)n) return;
   const DevLit L = lits[blockIdx.x];
-  const uint8_t* s = raw + L.src;
-  uint8_t* o = dec + L.dst;
-  uint32_t nw = L.len / 8;
+  const uint8_t* s = raw + (L.a & ((1ull << 40) - 1));
+  const uint32_t len = (uint32_t)(L.a >> 40);
+  uint8_t* o = dec + L.b;
+  uint32_t nw = len / 8;
   for (uint32_t w = threadIdx.x; w < nw; w += WAVE) {
     uint64_t v;
     __builtin_memcpy(&v, s + w * 8, 8);
     __builtin_memcpy(o + w * 8, &v, 8);
   }
   uint32_t t = nw * 8 + threadIdx.x;
-  if (t < L.len) o[t] = s[t];
+  if (t < len) o[t] = s[t];
 }
 
 #define BR_WIN 16384
@@ -1816,8 +1821,71 @@ k_agg(AggArgs a) {
     fs = a.fsum;    // zeroed before launch
   }
 
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  // four rows per thread per iteration: the per-row chain (mask -> gid ->
+  // val loads -> LDS atomics) is latency-bound (PMC: fetch exactly
+  // algorithmic at ~1.6 TB/s); unrolling forces four independent load
+  // chains in flight
+  const int64_t tid0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride1 = (int64_t)gridDim.x * blockDim.x;
+  const int64_t n4 = a.n_rows & ~3ll;
+  for (int64_t i4 = tid0 * 4; i4 < n4; i4 += stride1 * 4) {
+    uint32_t m4 = 0x01010101u;
+    if (a.mask) __builtin_memcpy(&m4, a.mask + i4, 4);
+    int32_t gs[4];
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      gs[r] = 0;
+      if (!((m4 >> (r * 8)) & 0xff)) { gs[r] = -1; continue; }
+      for (int k = 0; k < a.n_keys; k++)
+        gs[r] = gs[r] * a.key_size[k] + a.key_gid[k][i4 + r];
+    }
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      if (gs[r] < 0) continue;
+      const int64_t i = i4 + r;
+      uint64_t* row = tab + (int64_t)gs[r] * slots;
+      atomicAdd((unsigned long long*)&row[0], 1ull);
+      for (int ai = 0; ai < a.n_aggs; ai++) {
+        int k = a.agg_kind[ai];
+        if (k == AGGK_COUNT_STAR) continue;
+        if (a.agg_valid[ai] && !a.agg_valid[ai][i]) continue;
+        if (k == AGGK_COUNT) {
+          if (!a.cnt_skip[ai])
+            atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+          continue;
+        }
+        int64_t v = a.agg_val[ai][i];
+        uint64_t* vs = &row[1 + 2 * ai];
+        switch (k) {
+          case AGGK_SUM_I64: atomicAdd((unsigned long long*)vs, (unsigned long long)v); break;
+          case AGGK_SUM_F64: {
+            double d = __longlong_as_double((long long)v);
+            if (d != 0.0) {
+              uint64_t acc[4];
+              if (!acc256_decompose(d, acc)) {
+                if (a.err) atomicExch(a.err, ERR_FSUM_RANGE);
+              } else {
+                acc256_add(fs + ((int64_t)a.fsum_idx[ai] * a.n_groups + gs[r]) * 4,
+                           acc);
+              }
+            }
+            break;
+          }
+          case AGGK_MIN_I64: case AGGK_MIN_RANK: atomic_min_i64(vs, v); break;
+          case AGGK_MAX_I64: case AGGK_MAX_RANK: atomic_max_i64(vs, v); break;
+          case AGGK_MIN_F64: atomic_min_f64(vs, __longlong_as_double((long long)v)); break;
+          case AGGK_MAX_F64: atomic_max_f64(vs, __longlong_as_double((long long)v)); break;
+          case AGGK_MIN_STR: atomic_minmax_str(a.dec, vs, (uint64_t)v, true); break;
+          case AGGK_MAX_STR: atomic_minmax_str(a.dec, vs, (uint64_t)v, false); break;
+        }
+        if (!a.cnt_skip[ai])
+          atomicAdd((unsigned long long*)&row[1 + 2 * ai + 1], 1ull);
+      }
+    }
+  }
+  // tail rows (n_rows % 4)
+  int64_t i = n4 + tid0;
+  int64_t stride = stride1;
   for (; i < a.n_rows; i += stride) {
     if (a.mask && !a.mask[i]) continue;
     int32_t g = 0;
