@@ -1455,48 +1455,27 @@ k_contains_win(const uint8_t* __restrict__ dec,
   {  // stage the full window unconditionally (arena is padded by CWIN) and
      // presweep needle candidates while the words are STILL IN REGISTERS —
      // the verify pass then touches only queued positions instead of
-     // re-reading the whole window from LDS. 16 B vector loads; candidate
-     // slots reserved with ONE LDS atomic per thread (per-candidate
-     // atomicAdd on the single queue counter serialized ~hundreds deep on
-     // common needle bytes).
-    constexpr int KW = CWIN / (CTHREADS * 16);  // uint4 chunks per thread
-    uint4 v4[KW];
+     // re-reading the whole window from LDS
+    uint32_t v[CWIN / (CTHREADS * 4)];
 #pragma unroll
-    for (int k = 0; k < KW; k++)
-      __builtin_memcpy(&v4[k], src + threadIdx.x * 16u + (uint32_t)k * (CTHREADS * 16u), 16);
+    for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
+      __builtin_memcpy(&v[k], src + threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u), 4);
 #pragma unroll
-    for (int k = 0; k < KW; k++)
-      *(uint4*)&win[threadIdx.x * 16u + (uint32_t)k * (CTHREADS * 16u)] = v4[k];
+    for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
+      *(uint32_t*)&win[threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u)] = v[k];
     if (nlen) {
       const uint32_t pat = 0x01010101u * needle[0];
-      uint32_t cm[KW * 4];
-      uint32_t mine = 0;
 #pragma unroll
-      for (int k = 0; k < KW; k++) {
-        const uint32_t* w4 = (const uint32_t*)&v4[k];
-#pragma unroll
-        for (int j = 0; j < 4; j++) {
-          uint32_t p = threadIdx.x * 16u + (uint32_t)k * (CTHREADS * 16u) + j * 4u;
-          uint32_t x = w4[j] ^ pat;
-          uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
-          if (p >= W.nbytes) cand = 0;
-          cm[k * 4 + j] = cand;
-          mine += __builtin_popcount(cand);
-        }
-      }
-      uint32_t base = mine ? atomicAdd(&qn, mine) : 0;
-#pragma unroll
-      for (int k = 0; k < KW; k++) {
-#pragma unroll
-        for (int j = 0; j < 4; j++) {
-          uint32_t p = threadIdx.x * 16u + (uint32_t)k * (CTHREADS * 16u) + j * 4u;
-          uint32_t cand = cm[k * 4 + j];
-          while (cand) {
-            int b = (__builtin_ctz(cand)) >> 3;
-            cand &= cand - 1;
-            if (base < CQMAX) q[base] = (uint16_t)(p + b);
-            base++;
-          }
+      for (int k = 0; k < CWIN / (CTHREADS * 4); k++) {
+        uint32_t p = threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u);
+        if (p >= W.nbytes) continue;
+        uint32_t x = v[k] ^ pat;
+        uint32_t cand = (x - 0x01010101u) & ~x & 0x80808080u;
+        while (cand) {
+          int b = (__builtin_ctz(cand)) >> 3;
+          cand &= cand - 1;
+          uint32_t qi = atomicAdd(&qn, 1u);
+          if (qi < CQMAX) q[qi] = (uint16_t)(p + b);
         }
       }
     }
