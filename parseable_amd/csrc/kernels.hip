@@ -638,11 +638,17 @@ struct EmitDictMask {
 // the dense stream to scratch at [row0+k]; a separate expansion kernel
 // then writes the final arrays COALESCED through the k_def_levels rank map
 // (a scatter here would cost a read-modify-write line fetch per value).
+// Workgroups carry DP_WAVES waves per page: every wave walks the (cheap,
+// lane-redundant) run headers, but runs are unpacked round-robin by wave —
+// at 1 B-row scale a page-per-wave launch was ~1.9k waves, far too few to
+// hide latency on the unpack loads.
+#define DP_WAVES 4
 template <class Emit>
 __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit emit,
                                  const uint32_t* __restrict__ present,
                                  int32_t page_id, int mode, int32_t* d_error) {
-  const int lane = threadIdx.x;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
   const uint8_t* def_start; uint32_t def_len; bool all_valid;
   const uint8_t* vals = def_levels(pg, payload, &def_start, &def_len, &all_valid);
   uint32_t nv = pg.num_values;
@@ -658,13 +664,16 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
     int bw = *vals++;
     auto target = [&](uint32_t k) { return row0 + k; };
     if (bw == 0) {  // all values are dict id 0
-      for (uint32_t i = lane; i < nv; i += WAVE) emit(target(i), 0);
+      for (uint32_t i = threadIdx.x; i < nv; i += WAVE * DP_WAVES)
+        emit(target(i), 0);
       return;
     }
-    // lane-redundant run-header walk; data movement parallel per run
+    // lane-redundant run-header walk; data movement parallel per run,
+    // runs striped across the block's waves
     const uint8_t* p = vals;
     const uint32_t dict_n = pg.dict_n;
     uint32_t v = 0;
+    uint32_t rc = 0;
     while (v < nv) {
       // varint header (redundant on all lanes)
       uint64_t hdr = 0; int sh = 0;
@@ -683,6 +692,7 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
         // reads (8 lanes share a group) stay in L1/L2.
         uint32_t run_vals = groups * 8;
         if (run_vals > nv - v) run_vals = nv - v;
+        if ((rc++ % DP_WAVES) == (uint32_t)wave)
         for (uint32_t i = lane; i < run_vals; i += WAVE) {
           uint32_t g = i >> 3;
           int k = (int)(i & 7);
@@ -714,7 +724,8 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
         for (int b = 0; b < byte_w; b++) val |= (uint32_t)p[b] << (8 * b);
         p += byte_w;
         if (cnt > nv - v) cnt = nv - v;
-        if (val >= dict_n) { if (lane == 0) atomicExch(d_error, ERR_DICT_RANGE); val = 0; }
+        if (val >= dict_n) { if (threadIdx.x == 0) atomicExch(d_error, ERR_DICT_RANGE); val = 0; }
+        if ((rc++ % DP_WAVES) == (uint32_t)wave)
         for (uint32_t i = lane; i < cnt; i += WAVE) emit(target(v + i), val);
         v += cnt;
       }
@@ -723,7 +734,7 @@ __device__ void dict_page_decode(const DevPage& pg, const uint8_t* payload, Emit
 }
 
 template <class Emit>
-__global__ void __launch_bounds__(WAVE)
+__global__ void __launch_bounds__(WAVE * DP_WAVES)
 k_dict_pages(const uint8_t* __restrict__ dec, const DevPage* __restrict__ pages,
              const int32_t* __restrict__ ids, int n, Emit emit,
              const uint32_t* present, int mode, int32_t* d_error) {
@@ -2051,7 +2062,7 @@ void launch_dict_gid(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      int mode, int32_t* d_err) {
   if (!n) return;
   EmitGidP e{}; e.pool = remap_pool; e.out = out; e.valid = valid;
-  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, present, mode, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitGidP>, dim3(n), dim3(WAVE * DP_WAVES), 0, st, dec, pages, ids, n, e, present, mode, d_err);
 }
 void launch_dict_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      const int32_t* ids, int n, const int64_t* dictv_pool,
@@ -2059,14 +2070,14 @@ void launch_dict_i64(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                      int mode, int32_t* d_err) {
   if (!n) return;
   EmitDictI64P e{}; e.pool = dictv_pool; e.out = out; e.valid = valid;
-  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, present, mode, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitDictI64P>, dim3(n), dim3(WAVE * DP_WAVES), 0, st, dec, pages, ids, n, e, present, mode, d_err);
 }
 void launch_dict_mask(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                       const int32_t* ids, int n, const uint8_t* lut_pool,
                       uint8_t* mask, int32_t* d_err) {
   if (!n) return;
   EmitDictMaskP e{}; e.pool = lut_pool; e.mask = mask;
-  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, (const uint32_t*)nullptr, 0, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitDictMaskP>, dim3(n), dim3(WAVE * DP_WAVES), 0, st, dec, pages, ids, n, e, (const uint32_t*)nullptr, 0, d_err);
 }
 void launch_dict_lut_scr(hipStream_t st, const uint8_t* dec,
                          const DevPage* pages, const int32_t* ids, int n,
@@ -2074,7 +2085,7 @@ void launch_dict_lut_scr(hipStream_t st, const uint8_t* dec,
                          const uint32_t* present, int32_t* d_err) {
   if (!n) return;
   EmitLutD e{}; e.pool = lut_pool; e.scr = scr;
-  hipLaunchKernelGGL(k_dict_pages<EmitLutD>, dim3(n), dim3(WAVE), 0, st, dec, pages, ids, n, e, present, 1, d_err);
+  hipLaunchKernelGGL(k_dict_pages<EmitLutD>, dim3(n), dim3(WAVE * DP_WAVES), 0, st, dec, pages, ids, n, e, present, 1, d_err);
 }
 void launch_plain_fixed(hipStream_t st, const uint8_t* dec, const DevPage* pages,
                         const int32_t* ids, int n, int64_t* out, uint8_t* valid,

@@ -29,6 +29,12 @@ STREAMS = {
     # dict-overflow PLAIN-fallback utf8 (raw-byte hash group-by path)
     "c5": {"rows": 300_000, "keys": ["level", "host", "trace", "opt_tag"],
            "i64": ["latency"], "f64": [], "contains": "trace"},
+    # the c1 shape written with SNAPPY pages + zstd manifests (the
+    # reference's test-compose codec / compressed-manifest combination)
+    "c6": {"rows": 300_000, "keys": ["level", "host", "f_str1", "f_str2"],
+           "i64": ["latency", "f_i64"], "f64": ["f_f64"], "contains": None,
+           "gen": {"config": "c1", "compression": "snappy",
+                   "manifest_codec": "zstd"}},
 }
 KEY_VALUES = {
     "level": ["TRACE", "DEBUG", "INFO", "WARN", "ERROR"],
@@ -161,8 +167,10 @@ def main():
         root = os.path.join(args.data_dir, cfg_name)
         info_path = os.path.join(root, "stream", "stream.json")
         if not os.path.exists(info_path):
-            gen_stream(root, "stream", cfg_name, rows=cfg["rows"], seed=5,
-                       workers=4)
+            g = cfg.get("gen", {})
+            gen_stream(root, "stream", g.get("config", cfg_name),
+                       rows=cfg["rows"], seed=5, workers=4,
+                       **{k: v for k, v in g.items() if k != "config"})
         import glob
 
         files = sorted(glob.glob(os.path.join(root, "stream", "**",
