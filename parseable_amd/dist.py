@@ -25,7 +25,9 @@ def _batch_rows(batch, n_keys, n_aggs):
     out = {}
     if batch is None or batch.num_rows == 0:
         return out
-    cols = [batch.column(i).to_pylist() for i in range(batch.num_columns)]
+    from .provider import _col_list
+
+    cols = [_col_list(batch.column(i)) for i in range(batch.num_columns)]
     for r in range(batch.num_rows):
         key = tuple(cols[k][r] for k in range(n_keys))
         presence = cols[n_keys][r]
@@ -115,7 +117,9 @@ class DistMerger:
 
         if batch is not None and batch.num_rows:
             nk = len(self.group_by)
-            keycols = [batch.column(k).to_pylist() for k in range(nk)]
+            from .provider import _col_list
+
+            keycols = [_col_list(batch.column(k)) for k in range(nk)]
             keys = list(zip(*keycols)) if nk else [()] * batch.num_rows
             try:
                 gis = np.fromiter((self.key_index[k] for k in keys),

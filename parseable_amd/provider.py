@@ -833,6 +833,16 @@ class GpuExecutionPlan:
             pass
 
 
+def _col_list(col):
+    """Fast column -> python list: to_pylist() costs ~1 us/element (it
+    dominated the per-step Final merge at 1000 groups); the numpy route is
+    20-50x faster. Null-bearing columns keep to_pylist (to_numpy would
+    coerce nulled ints to floats)."""
+    if col.null_count == 0:
+        return col.to_numpy(zero_copy_only=False).tolist()
+    return col.to_pylist()
+
+
 def merge_partials(batches, query):
     """Final merge of partial aggregate tables (cross-partition and
     cross-rank). Output rows match the oracle's normalized form:
@@ -854,7 +864,7 @@ def merge_partials(batches, query):
         b = batches[0]
         picked_idx = list(range(nk)) + [nk + 1 + 2 * i
                                         for i in range(len(aggs))]
-        cols = [b.column(i).to_pylist() for i in picked_idx]
+        cols = [_col_list(b.column(i)) for i in picked_idx]
         for i, a in enumerate(aggs):
             if a["agg"] != "avg":
                 continue
@@ -876,7 +886,7 @@ def merge_partials(batches, query):
     for b in batches:
         if b is None or b.num_rows == 0:
             continue
-        cols = [b.column(i).to_pylist() for i in range(b.num_columns)]
+        cols = [_col_list(b.column(i)) for i in range(b.num_columns)]
         n = b.num_rows
         for r in range(n):
             key = tuple(cols[k][r] for k in range(nk))
@@ -935,7 +945,7 @@ def merge_topk(batches, query, extra_rows=None):
     for b in batches:
         if b is None or b.num_rows == 0:
             continue
-        data = [b.column(i).to_pylist() for i in range(b.num_columns)]
+        data = [_col_list(b.column(i)) for i in range(b.num_columns)]
         rows.extend([list(r) for r in zip(*data)])
     rows.sort(key=lambda r: -r[ts_i])
     lim = query.get("limit")
