@@ -407,11 +407,13 @@ def execute_acero(files: list[str], query: dict) -> dict:
     group_by = query.get("group_by", [])
     need = sorted(_needed_columns(query))
     tbl = dataset.to_table(filter=filt, columns=need or None)
-    # decode dictionary key columns for grouping
+    # decode dictionary key columns for grouping; timestamps compare as ms
     for g in group_by:
         i = tbl.schema.get_field_index(g)
         if pa.types.is_dictionary(tbl.schema.field(g).type):
             tbl = tbl.set_column(i, g, tbl.column(g).cast(pa.string()))
+        elif pa.types.is_timestamp(tbl.schema.field(g).type):
+            tbl = tbl.set_column(i, g, tbl.column(g).cast(pa.int64()))
 
     agglist = []
     names = []

@@ -169,6 +169,9 @@ struct ColPlan {
   // aggregation over arbitrary keys (stream_schema_provider.rs:219-225
   // hands the plan to that engine).
   bool hash_mode = false;
+  // numeric (i64/i32/timestamp) group key: dense gids from the value hash
+  // (k_numhash_*); d_val carries the key values, d_gid the assigned gids
+  bool numeric_key = false;
   bool need_gid_valid = false;  // validity bytes alongside gid (COUNT(utf8))
   // predicate routing
   std::vector<int> lut_preds;      // preds evaluated per-dict-entry (utf8)
@@ -377,6 +380,7 @@ struct gpuq_plan {
   int32_t n_fsum = 0;            // number of exact-f64-sum side tables
   bool has_hash = false;         // some column runs raw-byte utf8 hash mode
   bool needs_cascade = false;    // dense key product exceeds GID_CAP
+  bool has_numkey = false;       // some group key is numeric (value-hashed)
   std::vector<std::string> pinned_keys;  // hot-tier entries pinned by this plan
   int64_t m_cache_hit_bytes = 0;
   bool fused_count = false;      // single dict key + count(*)-only + no preds
@@ -753,8 +757,21 @@ extern "C" gpuq_plan* gpuq_plan_build(
     if (si < 0) throw std::runtime_error("no such column: " + c.name);
     c.phys = fm0.columns[si].phys_type;
     c.optional = fm0.columns[si].optional;
-    if (c.need_gid && c.phys != PT_BYTE_ARRAY)
-      throw std::runtime_error("group key must be utf8: " + c.name);
+    if (c.need_gid && c.phys != PT_BYTE_ARRAY) {
+      if (c.phys == PT_INT64 || c.phys == PT_INT32) {
+        // numeric group key: DataFusion's row-hash groups by any column —
+        // dense gids come from hashing the decoded values (k_numhash_*)
+        c.need_gid = false;
+        c.numeric_key = true;
+        c.need_val = true;
+        c.val_always = true;
+        plan->has_numkey = true;
+      } else {
+        throw std::runtime_error(
+            "unsupported group-key type (utf8/i64/i32/timestamp only): " +
+            c.name);
+      }
+    }
   }
   for (size_t pi = 0; pi < plan->preds.size(); pi++) {
     auto& pp = plan->preds[pi];
@@ -1718,7 +1735,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
     plan->fused_count = plan->group_cols.size() == 1 && aggs_ok &&
                         plan->preds.empty() && plan->n_groups <= 8192 &&
                         !plan->cols[plan->group_cols[0]].is_bin &&
-                        !plan->cols[plan->group_cols[0]].hash_mode;
+                        !plan->cols[plan->group_cols[0]].hash_mode &&
+                        !plan->cols[plan->group_cols[0]].numeric_key;
   }
 
   if (getenv("GPUQ_PLAN_DEBUG")) {
@@ -1856,10 +1874,10 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   // column outputs
   for (size_t ci = 0; ci < plan->cols.size(); ci++) {
     auto& c = plan->cols[ci];
-    if (c.is_bin) {
+    if (c.is_bin || c.numeric_key) {
       int32_t* d; HIP_TRY(hipMalloc(&d, std::max<int64_t>(part.n_rows * 4, 16)));
       part.d_gid[(int)ci] = d;
-      continue;
+      if (c.is_bin) continue;
     }
     if (c.need_gid) {
       int32_t* d; HIP_TRY(hipMalloc(&d, std::max<int64_t>(part.n_rows * 4, 16)));
@@ -1897,18 +1915,21 @@ extern "C" int32_t gpuq_plan_load(gpuq_plan* plan, int32_t pi) try {
   HIP_TRY(hipMalloc(&part.d_err, 4));
   // with hash-mode group keys the cardinality is only known at execute:
   // size the tables for the cap
-  bool hash_machinery = plan->has_hash || plan->needs_cascade;
+  bool hash_machinery = plan->has_hash || plan->needs_cascade ||
+                        plan->has_numkey;
   size_t groups_cap = hash_machinery ? (size_t)GID_CAP : (size_t)plan->n_groups;
   size_t tsz = groups_cap * (1 + 2 * plan->aggs.size()) * 8;
   HIP_TRY(hipMalloc(&part.d_table, std::max<size_t>(tsz, 16)));
   size_t fsz = (size_t)plan->n_fsum * groups_cap * 4 * 8;
   HIP_TRY(hipMalloc(&part.d_fsum, std::max<size_t>(fsz, 16)));
   if (hash_machinery) {
-    HIP_TRY(hipMalloc(&part.d_hkeys, (1ull << HASH_LOG2) * 8));
-    HIP_TRY(hipMalloc(&part.d_hgids, (1ull << HASH_LOG2) * 4));
+    // +1 entry: the numeric-key overflow slot for the value that equals
+    // the EMPTY sentinel (k_numhash_build)
+    HIP_TRY(hipMalloc(&part.d_hkeys, ((1ull << HASH_LOG2) + 1) * 8));
+    HIP_TRY(hipMalloc(&part.d_hgids, ((1ull << HASH_LOG2) + 1) * 4));
     HIP_TRY(hipMalloc(&part.d_hcount, 4));
     for (int ci : plan->group_cols)
-      if (plan->cols[ci].hash_mode) {
+      if (plan->cols[ci].hash_mode || plan->cols[ci].numeric_key) {
         uint64_t* g2r = nullptr;
         HIP_TRY(hipMalloc(&g2r, (size_t)GID_CAP * 8));
         part.d_gid2ref[ci] = g2r;
@@ -2450,8 +2471,9 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   auto key_card = [&](int ci) -> int64_t {
     const auto& kc = plan->cols[ci];
     return kc.is_bin ? (int64_t)kc.nbins + 1
-           : kc.hash_mode ? (int64_t)part.hash_claimed[ci] + 1
-                          : (int64_t)kc.gdict.size() + 1;
+           : (kc.hash_mode || kc.numeric_key)
+               ? (int64_t)part.hash_claimed[ci] + 1
+               : (int64_t)kc.gdict.size() + 1;
   };
   if (plan->fused_count)
     launch_init_table(st, part.d_table, plan->n_groups, (int)plan->aggs.size(),
@@ -2586,21 +2608,33 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // raw-byte utf8: device hash build assigns dense gids per hash-mode
   // group key (one shared table, rebuilt per column); string predicates
   // evaluate over the row strrefs
-  if (plan->has_hash) {
+  if (plan->has_hash || plan->has_numkey) {
     for (int ci : plan->group_cols) {
       auto& c = plan->cols[ci];
-      if (!c.hash_mode) continue;
-      HIP_TRY(hipMemsetAsync(part.d_hkeys, 0xFF, (1ull << HASH_LOG2) * 8, st));
-      HIP_TRY(hipMemsetAsync(part.d_hgids, 0xFF, (1ull << HASH_LOG2) * 4, st));
+      if (!c.hash_mode && !c.numeric_key) continue;
+      HIP_TRY(hipMemsetAsync(part.d_hkeys, 0xFF,
+                             ((1ull << HASH_LOG2) + 1) * 8, st));
+      HIP_TRY(hipMemsetAsync(part.d_hgids, 0xFF,
+                             ((1ull << HASH_LOG2) + 1) * 4, st));
       HIP_TRY(hipMemsetAsync(part.d_hcount, 0, 4, st));
       auto itv = part.d_valid.find(ci);
       uint8_t* v = itv != part.d_valid.end() ? itv->second : nullptr;
-      launch_hash_build(st, part.d_dec, part.d_val[ci], v, part.n_rows,
-                        part.d_hkeys, part.d_hgids, HASH_LOG2, part.d_hcount,
-                        part.d_gid2ref[ci], GID_CAP, part.d_err);
-      launch_hash_lookup(st, part.d_dec, part.d_val[ci], v, part.n_rows,
-                         part.d_hkeys, part.d_hgids, HASH_LOG2,
-                         part.d_gid[ci]);
+      if (c.numeric_key) {
+        launch_numhash_build(st, part.d_val[ci], v, part.n_rows,
+                             part.d_hkeys, part.d_hgids, HASH_LOG2,
+                             part.d_hcount, part.d_gid2ref[ci], GID_CAP,
+                             part.d_err);
+        launch_numhash_lookup(st, part.d_val[ci], v, part.n_rows,
+                              part.d_hkeys, part.d_hgids, HASH_LOG2,
+                              part.d_gid[ci]);
+      } else {
+        launch_hash_build(st, part.d_dec, part.d_val[ci], v, part.n_rows,
+                          part.d_hkeys, part.d_hgids, HASH_LOG2, part.d_hcount,
+                          part.d_gid2ref[ci], GID_CAP, part.d_err);
+        launch_hash_lookup(st, part.d_dec, part.d_val[ci], v, part.n_rows,
+                           part.d_hkeys, part.d_hgids, HASH_LOG2,
+                           part.d_gid[ci]);
+      }
       uint32_t claimed = 0;
       HIP_TRY(hipMemcpyAsync(&claimed, part.d_hcount, 4,
                              hipMemcpyDeviceToHost, st));
@@ -2812,7 +2846,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   for (int k = 0; k < n_keys; k++) {
     ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
     const auto& kc = plan->cols[plan->group_cols[k]];
-    make_schema_field(ss->schema.children[f], kc.is_bin ? "l" : "u",
+    make_schema_field(ss->schema.children[f],
+                      (kc.is_bin || kc.numeric_key) ? "l" : "u",
                       kc.is_bin ? "date_bin" : kc.name.c_str());
     f++;
   }
@@ -2886,9 +2921,19 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
   // hash-mode key columns: fetch the strings for the local gids that
   // actually appear (gid -> gid2ref -> dec-arena bytes)
   std::map<int, std::unordered_map<int32_t, std::string>> hash_names;
+  std::map<int, std::vector<uint64_t>> num_keys;  // numeric key: gid -> value
   for (int k = 0; k < n_keys; k++) {
     int ci = plan->group_cols[k];
     const auto& kc = plan->cols[ci];
+    if (kc.numeric_key) {
+      uint32_t claimed = part.hash_claimed[ci];
+      auto& g2k = num_keys[ci];
+      g2k.resize(claimed);
+      if (claimed)
+        HIP_TRY(hipMemcpy(g2k.data(), part.d_gid2ref[ci],
+                          (size_t)claimed * 8, hipMemcpyDeviceToHost));
+      continue;
+    }
     if (!kc.hash_mode) continue;
     std::vector<int32_t> need;
     {
@@ -2927,6 +2972,29 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
         if (gid > 0)
           v[r] = c.bin_origin + (c.bin_min_idx + gid - 1) * c.bin_stride;
         else {
+          v[r] = 0;
+          validity[r / 8] &= (uint8_t)~(1 << (r % 8));
+          nulls++;
+        }
+      }
+      if (nulls) { ch->buffers[0] = validity; ch->null_count = nulls; }
+      ch->buffers[1] = v;
+      continue;
+    }
+    if (c.numeric_key) {
+      // i64 key values from the claim table (NULL group at gid 0)
+      ch->n_buffers = 2;
+      ch->buffers = (const void**)calloc(2, sizeof(void*));
+      int64_t* v = (int64_t*)eb->grab(nr * 8);
+      uint8_t* validity = (uint8_t*)eb->grab((nr + 7) / 8);
+      memset(validity, 0xff, (nr + 7) / 8);
+      const auto& g2k = num_keys[plan->group_cols[k]];
+      int64_t nulls = 0;
+      for (int64_t r = 0; r < nr; r++) {
+        int32_t gid = key_gids[k][r];
+        if (gid > 0) {
+          v[r] = (int64_t)g2k[(size_t)gid - 1];
+        } else {
           v[r] = 0;
           validity[r / 8] &= (uint8_t)~(1 << (r % 8));
           nulls++;

@@ -1169,6 +1169,78 @@ __global__ void k_pair_lookup(const int32_t* __restrict__ a,
   }
 }
 
+// numeric (i64) group keys: dense gids via the same two-pass claim/lookup
+// structure (the reference's engine groups by any column; utf8 keys get
+// strref hashing, numeric keys hash the 64-bit value itself). The all-ones
+// bit pattern is both a valid value (-1) and the EMPTY sentinel, so rows
+// with value -1 claim a dedicated slot at index 2^clog2 instead of probing.
+__global__ void k_numhash_build(const int64_t* __restrict__ vals,
+                                const uint8_t* __restrict__ valid,
+                                int64_t n_rows, uint64_t* __restrict__ hkeys,
+                                int32_t* __restrict__ hgids, int clog2,
+                                uint32_t* counter,
+                                uint64_t* __restrict__ gid2key,
+                                int32_t gid_cap, int32_t* d_error) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const uint64_t mask = (1ull << clog2) - 1;
+  const uint64_t cap_slot = 1ull << clog2;
+  for (; i < n_rows; i += stride) {
+    if (valid && !valid[i]) continue;
+    uint64_t key = (uint64_t)vals[i];
+    if (key == HREF_EMPTY) {  // value -1: the dedicated overflow slot
+      uint64_t old = atomicCAS((unsigned long long*)&hkeys[cap_slot],
+                               (unsigned long long)HREF_EMPTY, 0ull);
+      if (old == HREF_EMPTY) {
+        uint32_t g = atomicAdd(counter, 1u);
+        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_HASH_CAP); return; }
+        gid2key[g] = key;
+        hgids[cap_slot] = (int32_t)g;
+      }
+      continue;
+    }
+    uint64_t slot = mix64(key) & mask;
+    for (uint32_t probe = 0;; probe++) {
+      if (probe > (1u << clog2)) { atomicExch(d_error, ERR_HASH_PROBE); return; }
+      uint64_t old = atomicCAS((unsigned long long*)&hkeys[slot],
+                               (unsigned long long)HREF_EMPTY,
+                               (unsigned long long)key);
+      if (old == HREF_EMPTY) {
+        uint32_t g = atomicAdd(counter, 1u);
+        if ((int32_t)g >= gid_cap) { atomicExch(d_error, ERR_HASH_CAP); return; }
+        gid2key[g] = key;
+        hgids[slot] = (int32_t)g;
+        break;
+      }
+      if (old == key) break;
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+__global__ void k_numhash_lookup(const int64_t* __restrict__ vals,
+                                 const uint8_t* __restrict__ valid,
+                                 int64_t n_rows,
+                                 const uint64_t* __restrict__ hkeys,
+                                 const int32_t* __restrict__ hgids, int clog2,
+                                 int32_t* __restrict__ out_gid) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const uint64_t mask = (1ull << clog2) - 1;
+  const uint64_t cap_slot = 1ull << clog2;
+  for (; i < n_rows; i += stride) {
+    if (valid && !valid[i]) { out_gid[i] = 0; continue; }
+    uint64_t key = (uint64_t)vals[i];
+    if (key == HREF_EMPTY) { out_gid[i] = hgids[cap_slot] + 1; continue; }
+    uint64_t slot = mix64(key) & mask;
+    for (;;) {
+      uint64_t k = hkeys[slot];
+      if (k == key) { out_gid[i] = hgids[slot] + 1; break; }
+      if (k == HREF_EMPTY) { out_gid[i] = 0; break; }  // unreachable
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+
 // string predicates over row strrefs (mixed dict/PLAIN chunks where the
 // per-dict-entry LUT cannot cover the PLAIN pages). op: CmpMode, or -1 for
 // CONTAINS (byte substring).
@@ -2220,6 +2292,28 @@ void launch_pair_lookup(hipStream_t st, const int32_t* a, const int32_t* b,
   if (blocks > 8192) blocks = 8192;
   hipLaunchKernelGGL(k_pair_lookup, dim3(blocks), dim3(256), 0, st, a, b,
                      n_rows, hkeys, hgids, clog2, out);
+}
+void launch_numhash_build(hipStream_t st, const int64_t* vals,
+                          const uint8_t* valid, int64_t n_rows,
+                          uint64_t* hkeys, int32_t* hgids, int clog2,
+                          uint32_t* counter, uint64_t* gid2key,
+                          int32_t gid_cap, int32_t* d_err) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_numhash_build, dim3(blocks), dim3(256), 0, st, vals,
+                     valid, n_rows, hkeys, hgids, clog2, counter, gid2key,
+                     gid_cap, d_err);
+}
+void launch_numhash_lookup(hipStream_t st, const int64_t* vals,
+                           const uint8_t* valid, int64_t n_rows,
+                           const uint64_t* hkeys, const int32_t* hgids,
+                           int clog2, int32_t* out_gid) {
+  if (!n_rows) return;
+  int blocks = (int)((n_rows + 255) / 256);
+  if (blocks > 8192) blocks = 8192;
+  hipLaunchKernelGGL(k_numhash_lookup, dim3(blocks), dim3(256), 0, st, vals,
+                     valid, n_rows, hkeys, hgids, clog2, out_gid);
 }
 void launch_cmp_str(hipStream_t st, const uint8_t* dec, const int64_t* refs,
                     const uint8_t* valid, const uint8_t* lit, uint32_t lit_len,

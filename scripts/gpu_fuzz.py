@@ -18,14 +18,15 @@ from oracle.compare import rows_equal  # noqa: E402
 
 STREAMS = {
     "c1": {"rows": 500_000, "keys": ["level", "host", "f_str1", "f_str2"],
-           "i64": ["latency", "f_i64"], "f64": ["f_f64"], "contains": None},
+           "i64": ["latency", "f_i64"], "f64": ["f_f64"], "contains": None,
+           "ikeys": ["latency", "f_i64"]},
     "c3": {"rows": 300_000, "keys": ["level", "host", "f_str1"],
            "i64": ["latency", "f_i64"], "f64": ["f_f64"],
            "contains": "message"},
     "c4": {"rows": 300_000,
            "keys": ["service", "span_kind", "status", "attr_s1", "attr_s7"],
            "i64": ["latency", "attr_i0", "attr_i3", "attr_i11"], "f64": [],
-           "contains": None},
+           "contains": None, "ikeys": ["attr_i5", "attr_i11"]},
     # dict-overflow PLAIN-fallback utf8 (raw-byte hash group-by path)
     "c5": {"rows": 300_000, "keys": ["level", "host", "trace", "opt_tag"],
            "i64": ["latency"], "f64": [], "contains": "trace"},
@@ -57,6 +58,8 @@ def gen_query(rng: random.Random, cfg: dict, n_rows: int):
     if keys and rng.random() < 0.15:
         keys[0] = {"bin": "p_timestamp",
                    "stride_ms": rng.choice([60_000, 300_000])}
+    if rng.random() < 0.2 and cfg.get("ikeys") and len(keys) < 3:
+        keys.append(rng.choice(cfg["ikeys"]))  # numeric group key
     q["group_by"] = keys
     aggs = [{"agg": "count_star"}]
     numcols = cfg["i64"] + cfg["f64"]

@@ -103,6 +103,27 @@ GOLDEN_QUERIES = {
                        {"agg": "count", "col": "f_str1"}],
             "group_by": ["f_str2", "level"],
         }),
+        # numeric group keys (DataFusion groups by any column; gids from
+        # the value hash — pinned by pyarrow oracle + Acero, the scalar C
+        # restatement keys strings only)
+        ("int_key_group", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "sum", "col": "f_i64"}],
+            "group_by": ["latency"],
+            "preds": [{"col": "latency", "op": "lt", "lit": 2000}],
+        }),
+        ("int_key_mixed_pair", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "min", "col": "f_f64"}],
+            "group_by": ["level", "latency"],
+            "preds": [{"col": "latency", "op": "lt", "lit": 800}],
+        }),
+        ("ts_key_group", {
+            "ext": True,
+            "select": [{"agg": "count_star"}],
+            "group_by": ["p_timestamp"],
+            "time_range": [BASE, BASE + 2_000],
+        }),
         ("key_rank_pred_same_col", {
             "ext": True,
             "select": [{"agg": "count_star"}, {"agg": "min", "col": "level"}],
@@ -155,6 +176,12 @@ GOLDEN_QUERIES = {
         ("group_by_nullable", {
             "select": [{"agg": "count_star"}, {"agg": "sum", "col": "attr_i0"}],
             "group_by": ["attr_s1"],
+        }),
+        ("int_key_nullable", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["attr_i5"],
+            "preds": [{"col": "attr_i0", "op": "lt", "lit": 30000}],
         }),
         ("pred_on_nullable", {
             "select": [{"agg": "count_star"}],
