@@ -758,7 +758,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
     c.phys = fm0.columns[si].phys_type;
     c.optional = fm0.columns[si].optional;
     if (c.need_gid && c.phys != PT_BYTE_ARRAY) {
-      if (c.phys == PT_INT64 || c.phys == PT_INT32) {
+      if (c.phys == PT_INT64 || c.phys == PT_INT32 || c.phys == PT_DOUBLE) {
         // numeric group key: DataFusion's row-hash groups by any column —
         // dense gids come from hashing the decoded values (k_numhash_*)
         c.need_gid = false;
@@ -768,7 +768,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
         plan->has_numkey = true;
       } else {
         throw std::runtime_error(
-            "unsupported group-key type (utf8/i64/i32/timestamp only): " +
+            "unsupported group-key type (utf8/i64/i32/f64/timestamp): " +
             c.name);
       }
     }
@@ -2620,12 +2620,13 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
       auto itv = part.d_valid.find(ci);
       uint8_t* v = itv != part.d_valid.end() ? itv->second : nullptr;
       if (c.numeric_key) {
+        int kf64 = (c.phys == PT_DOUBLE) ? 1 : 0;
         launch_numhash_build(st, part.d_val[ci], v, part.n_rows,
                              part.d_hkeys, part.d_hgids, HASH_LOG2,
                              part.d_hcount, part.d_gid2ref[ci], GID_CAP,
-                             part.d_err);
+                             kf64, part.d_err);
         launch_numhash_lookup(st, part.d_val[ci], v, part.n_rows,
-                              part.d_hkeys, part.d_hgids, HASH_LOG2,
+                              part.d_hkeys, part.d_hgids, HASH_LOG2, kf64,
                               part.d_gid[ci]);
       } else {
         launch_hash_build(st, part.d_dec, part.d_val[ci], v, part.n_rows,
@@ -2847,7 +2848,8 @@ extern "C" int32_t gpuq_plan_execute(gpuq_plan* plan, int32_t pi,
     ss->schema.children[f] = (struct ArrowSchema*)malloc(sizeof(struct ArrowSchema));
     const auto& kc = plan->cols[plan->group_cols[k]];
     make_schema_field(ss->schema.children[f],
-                      (kc.is_bin || kc.numeric_key) ? "l" : "u",
+                      (kc.numeric_key && kc.phys == PT_DOUBLE) ? "g"
+                      : (kc.is_bin || kc.numeric_key) ? "l" : "u",
                       kc.is_bin ? "date_bin" : kc.name.c_str());
     f++;
   }

@@ -1174,20 +1174,32 @@ __global__ void k_pair_lookup(const int32_t* __restrict__ a,
 // strref hashing, numeric keys hash the 64-bit value itself). The all-ones
 // bit pattern is both a valid value (-1) and the EMPTY sentinel, so rows
 // with value -1 claim a dedicated slot at index 2^clog2 instead of probing.
+// f64 keys group by VALUE equality under DataFusion's row-format
+// normalization: -0.0 folds into +0.0 and every NaN into one canonical
+// bit pattern, so bit-equality matches SQL group semantics.
+__device__ inline uint64_t numkey_norm(uint64_t key, int is_f64) {
+  if (!is_f64) return key;
+  if ((key & 0x7fffffffffffffffull) == 0) return 0;              // -0.0 -> +0.0
+  if ((key & 0x7ff0000000000000ull) == 0x7ff0000000000000ull &&
+      (key & 0x000fffffffffffffull) != 0)
+    return 0x7ff8000000000000ull;                                // canonical NaN
+  return key;
+}
+
 __global__ void k_numhash_build(const int64_t* __restrict__ vals,
                                 const uint8_t* __restrict__ valid,
                                 int64_t n_rows, uint64_t* __restrict__ hkeys,
                                 int32_t* __restrict__ hgids, int clog2,
                                 uint32_t* counter,
                                 uint64_t* __restrict__ gid2key,
-                                int32_t gid_cap, int32_t* d_error) {
+                                int32_t gid_cap, int is_f64, int32_t* d_error) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const uint64_t mask = (1ull << clog2) - 1;
   const uint64_t cap_slot = 1ull << clog2;
   for (; i < n_rows; i += stride) {
     if (valid && !valid[i]) continue;
-    uint64_t key = (uint64_t)vals[i];
+    uint64_t key = numkey_norm((uint64_t)vals[i], is_f64);
     if (key == HREF_EMPTY) {  // value -1: the dedicated overflow slot
       uint64_t old = atomicCAS((unsigned long long*)&hkeys[cap_slot],
                                (unsigned long long)HREF_EMPTY, 0ull);
@@ -1222,14 +1234,14 @@ __global__ void k_numhash_lookup(const int64_t* __restrict__ vals,
                                  int64_t n_rows,
                                  const uint64_t* __restrict__ hkeys,
                                  const int32_t* __restrict__ hgids, int clog2,
-                                 int32_t* __restrict__ out_gid) {
+                                 int is_f64, int32_t* __restrict__ out_gid) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const uint64_t mask = (1ull << clog2) - 1;
   const uint64_t cap_slot = 1ull << clog2;
   for (; i < n_rows; i += stride) {
     if (valid && !valid[i]) { out_gid[i] = 0; continue; }
-    uint64_t key = (uint64_t)vals[i];
+    uint64_t key = numkey_norm((uint64_t)vals[i], is_f64);
     if (key == HREF_EMPTY) { out_gid[i] = hgids[cap_slot] + 1; continue; }
     uint64_t slot = mix64(key) & mask;
     for (;;) {
@@ -2297,23 +2309,23 @@ void launch_numhash_build(hipStream_t st, const int64_t* vals,
                           const uint8_t* valid, int64_t n_rows,
                           uint64_t* hkeys, int32_t* hgids, int clog2,
                           uint32_t* counter, uint64_t* gid2key,
-                          int32_t gid_cap, int32_t* d_err) {
+                          int32_t gid_cap, int is_f64, int32_t* d_err) {
   if (!n_rows) return;
   int blocks = (int)((n_rows + 255) / 256);
   if (blocks > 8192) blocks = 8192;
   hipLaunchKernelGGL(k_numhash_build, dim3(blocks), dim3(256), 0, st, vals,
                      valid, n_rows, hkeys, hgids, clog2, counter, gid2key,
-                     gid_cap, d_err);
+                     gid_cap, is_f64, d_err);
 }
 void launch_numhash_lookup(hipStream_t st, const int64_t* vals,
                            const uint8_t* valid, int64_t n_rows,
                            const uint64_t* hkeys, const int32_t* hgids,
-                           int clog2, int32_t* out_gid) {
+                           int clog2, int is_f64, int32_t* out_gid) {
   if (!n_rows) return;
   int blocks = (int)((n_rows + 255) / 256);
   if (blocks > 8192) blocks = 8192;
   hipLaunchKernelGGL(k_numhash_lookup, dim3(blocks), dim3(256), 0, st, vals,
-                     valid, n_rows, hkeys, hgids, clog2, out_gid);
+                     valid, n_rows, hkeys, hgids, clog2, is_f64, out_gid);
 }
 void launch_cmp_str(hipStream_t st, const uint8_t* dec, const int64_t* refs,
                     const uint8_t* valid, const uint8_t* lit, uint32_t lit_len,

@@ -86,11 +86,11 @@ void launch_numhash_build(hipStream_t, const int64_t* vals,
                           const uint8_t* valid, int64_t n_rows,
                           uint64_t* hkeys, int32_t* hgids, int clog2,
                           uint32_t* counter, uint64_t* gid2key,
-                          int32_t gid_cap, int32_t* d_err);
+                          int32_t gid_cap, int is_f64, int32_t* d_err);
 void launch_numhash_lookup(hipStream_t, const int64_t* vals,
                            const uint8_t* valid, int64_t n_rows,
                            const uint64_t* hkeys, const int32_t* hgids,
-                           int clog2, int32_t* out_gid);
+                           int clog2, int is_f64, int32_t* out_gid);
 void launch_cmp_str(hipStream_t, const uint8_t* dec, const int64_t* refs,
                     const uint8_t* valid, const uint8_t* lit, uint32_t lit_len,
                     int op, uint8_t* mask, int64_t n_rows);

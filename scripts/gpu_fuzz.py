@@ -19,7 +19,7 @@ from oracle.compare import rows_equal  # noqa: E402
 STREAMS = {
     "c1": {"rows": 500_000, "keys": ["level", "host", "f_str1", "f_str2"],
            "i64": ["latency", "f_i64"], "f64": ["f_f64"], "contains": None,
-           "ikeys": ["latency", "f_i64"]},
+           "ikeys": ["latency", "f_i64", "f_f64"]},
     "c3": {"rows": 300_000, "keys": ["level", "host", "f_str1"],
            "i64": ["latency", "f_i64"], "f64": ["f_f64"],
            "contains": "message"},

@@ -118,6 +118,12 @@ GOLDEN_QUERIES = {
             "group_by": ["level", "latency"],
             "preds": [{"col": "latency", "op": "lt", "lit": 800}],
         }),
+        ("f64_key_group", {
+            "ext": True,
+            "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
+            "group_by": ["f_f64"],
+            "preds": [{"col": "f_f64", "op": "lt", "lit": 0.002}],
+        }),
         ("ts_key_group", {
             "ext": True,
             "select": [{"agg": "count_star"}],
