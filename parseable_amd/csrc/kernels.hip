@@ -2076,6 +2076,91 @@ k_agg(AggArgs a) {
   }
 }
 
+// Specialized aggregation for the headline shape: ONE key, aggregates =
+// {COUNT_STAR [, one i64 sum/min/max over a null-free column]} — ~1/3 the
+// instructions of the general kernel (no per-agg dispatch, no validity
+// reads, vectorized 16 B gid loads). Dispatch in launch_agg.
+#define AGGF_NONE -1
+template <bool USE_LDS, int K1>
+__global__ void __launch_bounds__(256)
+k_agg_fast(AggArgs a) {
+  extern __shared__ uint64_t lt[];
+  const int slots = 1 + 2 * a.n_aggs;
+  const int64_t tsz = (int64_t)a.n_groups * slots;
+  uint64_t* tab;
+  if (USE_LDS) {
+    tab = lt;
+    for (int64_t i = threadIdx.x; i < tsz; i += blockDim.x) {
+      uint64_t init = 0;
+      int sl = (int)(i % slots);
+      if (sl == 3) {  // the K1 agg's value slot (agg index 1)
+        if (K1 == AGGK_MIN_I64) init = (uint64_t)INT64_MAX;
+        else if (K1 == AGGK_MAX_I64) init = (uint64_t)INT64_MIN;
+      }
+      lt[i] = init;
+    }
+    __syncthreads();
+  } else {
+    tab = a.table;
+  }
+  const int32_t* __restrict__ gid = a.key_gid[0];
+  const int64_t* __restrict__ val = a.agg_val[K1 == AGGF_NONE ? 0 : 1];
+  const int64_t tid0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride4 = (int64_t)gridDim.x * blockDim.x * 4;
+  const int64_t n4 = a.n_rows & ~3ll;
+  for (int64_t i4 = tid0 * 4; i4 < n4; i4 += stride4) {
+    uint32_t m4 = 0x01010101u;
+    if (a.mask) __builtin_memcpy(&m4, a.mask + i4, 4);
+    int32_t g4[4];
+    __builtin_memcpy(g4, gid + i4, 16);
+    int64_t v4[4];
+    if (K1 != AGGF_NONE) __builtin_memcpy(v4, val + i4, 32);
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      if (!((m4 >> (r * 8)) & 0xff)) continue;
+      uint64_t* row = tab + (int64_t)g4[r] * slots;
+      atomicAdd((unsigned long long*)&row[0], 1ull);
+      if (K1 == AGGK_SUM_I64)
+        atomicAdd((unsigned long long*)&row[3], (unsigned long long)v4[r]);
+      else if (K1 == AGGK_MIN_I64)
+        atomic_min_i64(&row[3], v4[r]);
+      else if (K1 == AGGK_MAX_I64)
+        atomic_max_i64(&row[3], v4[r]);
+    }
+  }
+  for (int64_t i = n4 + tid0; i < a.n_rows; i += stride4 / 4) {
+    if (a.mask && !a.mask[i]) continue;
+    uint64_t* row = tab + (int64_t)gid[i] * slots;
+    atomicAdd((unsigned long long*)&row[0], 1ull);
+    if (K1 != AGGF_NONE) {
+      int64_t v = val[i];
+      if (K1 == AGGK_SUM_I64)
+        atomicAdd((unsigned long long*)&row[3], (unsigned long long)v);
+      else if (K1 == AGGK_MIN_I64) atomic_min_i64(&row[3], v);
+      else if (K1 == AGGK_MAX_I64) atomic_max_i64(&row[3], v);
+    }
+  }
+  if (USE_LDS) {
+    __syncthreads();
+    for (int64_t t = threadIdx.x; t < tsz; t += blockDim.x) {
+      int sl = (int)(t % slots);
+      uint64_t v = tab[t];
+      uint64_t* g = &a.table[t];
+      if (sl == 0) {
+        if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v);
+      } else if (sl == 3 && K1 != AGGF_NONE) {
+        if (K1 == AGGK_SUM_I64) {
+          if (v) atomicAdd((unsigned long long*)g, (unsigned long long)v);
+        } else if (K1 == AGGK_MIN_I64) {
+          if ((int64_t)v != INT64_MAX) atomic_min_i64(g, (int64_t)v);
+        } else if ((int64_t)v != INT64_MIN) {
+          atomic_max_i64(g, (int64_t)v);
+        }
+      }
+    }
+  }
+}
+
 __global__ void k_init_table(uint64_t* table, int32_t n_groups, int n_aggs,
                              const int32_t* agg_kind) {
   int slots = 1 + 2 * n_aggs;
@@ -2248,9 +2333,43 @@ void launch_agg(hipStream_t st, const AggArgs& a) {
   int blocks = (int)((a.n_rows + 255) / 256);
   if (blocks > 2048) blocks = 2048;
   if (blocks < 1) blocks = 1;
+  // fast path: one key, {COUNT_STAR [, one null-free i64 sum/min/max]}
+  // (the c2 headline shape); validity reads and the per-agg dispatch
+  // vanish, gid/val loads vectorize
+  bool fast = a.n_keys == 1 && a.n_aggs >= 1 && a.n_aggs <= 2 &&
+              a.agg_kind[0] == AGGK_COUNT_STAR && a.fsum_n == 0;
+  int k1 = AGGF_NONE;
+  if (fast && a.n_aggs == 2) {
+    k1 = a.agg_kind[1];
+    fast = (k1 == AGGK_SUM_I64 || k1 == AGGK_MIN_I64 || k1 == AGGK_MAX_I64) &&
+           a.cnt_skip[1] && a.agg_valid[1] == nullptr &&
+           a.agg_val[1] != nullptr;
+  }
   // 64 KiB still admits 2 blocks/CU and avoids the global-atomic cliff on
   // ~1000-group tables (c2: GROUP BY host was 100x slower via global atomics)
-  if (lds <= 64 * 1024) {
+  bool use_lds = lds <= 64 * 1024;
+  if (fast) {
+    auto launch = [&](auto kern) {
+      hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), use_lds ? lds : 0, st, a);
+    };
+    if (use_lds) {
+      switch (k1) {
+        case AGGK_SUM_I64: launch(k_agg_fast<true, AGGK_SUM_I64>); break;
+        case AGGK_MIN_I64: launch(k_agg_fast<true, AGGK_MIN_I64>); break;
+        case AGGK_MAX_I64: launch(k_agg_fast<true, AGGK_MAX_I64>); break;
+        default: launch(k_agg_fast<true, AGGF_NONE>); break;
+      }
+    } else {
+      switch (k1) {
+        case AGGK_SUM_I64: launch(k_agg_fast<false, AGGK_SUM_I64>); break;
+        case AGGK_MIN_I64: launch(k_agg_fast<false, AGGK_MIN_I64>); break;
+        case AGGK_MAX_I64: launch(k_agg_fast<false, AGGK_MAX_I64>); break;
+        default: launch(k_agg_fast<false, AGGF_NONE>); break;
+      }
+    }
+    return;
+  }
+  if (use_lds) {
     hipLaunchKernelGGL(k_agg<true>, dim3(blocks), dim3(256), lds, st, a);
   } else {
     hipLaunchKernelGGL(k_agg<false>, dim3(blocks), dim3(256), 0, st, a);
