@@ -124,3 +124,73 @@ def test_gloo_world2_merge():
         assert results[r]["int"] == EXPECTED
         assert results[r]["f64"] == EXPECTED_F
         assert results[r]["str"] == EXPECTED_S
+
+
+def _rank_main4(rank, world, port, q):
+    import torch.distributed as dist
+
+    from parseable_amd.dist import DistMerger
+
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank, world_size=world
+    )
+    try:
+        # world-4: mixed avg/f64-sum/utf8-max slots, one empty rank, disjoint
+        # and overlapping key spaces
+        qm = {"select": [{"agg": "count_star"}, {"agg": "avg", "col": "latency"},
+                         {"agg": "sum", "col": "f_f64"},
+                         {"agg": "max", "col": "host"}],
+              "group_by": ["level"]}
+        batches = [
+            _partial2(["INFO", "WARN"], [2, 1],
+                      [([2, 1], [2, 1], pa.int64()),     # count_star mirror
+                       ([10, 7], [2, 1], pa.int64()),    # avg: sums
+                       ([0.5, 0.25], [2, 1], pa.float64()),
+                       (["h-b", "h-a"], [2, 1], pa.string())]),
+            _partial2(["INFO"], [3],
+                      [([3], [3], pa.int64()),
+                       ([30], [3], pa.int64()),
+                       ([1.5], [3], pa.float64()),
+                       (["h-z"], [3], pa.string())]),
+            None,
+            _partial2(["ERROR"], [4],
+                      [([4], [4], pa.int64()),
+                       ([100], [4], pa.int64()),
+                       ([2.0], [4], pa.float64()),
+                       (["h-m"], [4], pa.string())]),
+        ]
+        m = DistMerger(qm, device="cpu")
+        m.setup(batches[rank])
+        rows = m.step(batches[rank])
+        q.put((rank, rows))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_gloo_world4_mixed_aggs():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_rank_main4, args=(r, 4, port, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, rows = q.get()
+        results[rank] = rows
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+    expected = [
+        ["ERROR", 4, 100 / 4, 2.0, "h-m"],
+        ["INFO", 5, 40 / 5, 2.0, "h-z"],
+        ["WARN", 1, 7 / 1, 0.25, "h-a"],
+    ]
+    for r in range(4):
+        assert results[r] == expected
