@@ -978,6 +978,15 @@ extern "C" gpuq_plan* gpuq_plan_build(
   // decompress, LZ4 structure walks) runs parallel across host cores, with
   // serial phases only for global-dict id assignment and the
   // order-preserving merges into the device pools. ---
+  const bool plan_dbg = getenv("GPUQ_PLAN_DEBUG") != nullptr;
+  int64_t tphase = now_ns();
+  auto phase_mark = [&](const char* name) {
+    if (!plan_dbg) return;
+    int64_t t = now_ns();
+    fprintf(stderr, "[gpuq plan] %-22s %7.1f ms\n", name,
+            (t - tphase) / 1e6);
+    tphase = t;
+  };
   for (auto& part : plan->parts) {
     uint32_t row_cursor = 0;
     for (auto& r : part.rgs) {
@@ -1134,6 +1143,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
                                  "outside hash mode — " + c.name);
       part.chunks[i] = std::move(t);
     });
+    phase_mark("phase1 walk+dicts");
 
     // phase 2 (serial): global-dictionary gid assignment per chunk, in order
     for (size_t i = 0; i < part.chunks.size(); i++) {
@@ -1150,6 +1160,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       dict_strs[i].clear();
       dict_strs[i].shrink_to_fit();
     }
+    phase_mark("phase2 gids");
 
     // phase 3 (serial): pool / page-id / dec-arena bases per chunk
     struct Bases {
@@ -1215,6 +1226,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
         }
       }
     }
+
+    phase_mark("phase3 pools+cache");
 
     // phase 4 (parallel): per-chunk device images — DevPage descriptors,
     // LZ4 structure walks, seg/lit/backref records, task-list routing
@@ -1475,6 +1488,8 @@ extern "C" gpuq_plan* gpuq_plan_build(
         throw std::runtime_error("page rows mismatch");
     });
 
+    phase_mark("phase4 lz4 walks");
+
     // phase 5 (serial): order-preserving merge of the per-chunk images
     for (size_t i = 0; i < cbs.size(); i++) {
       auto& cb = cbs[i];
@@ -1509,6 +1524,7 @@ extern "C" gpuq_plan* gpuq_plan_build(
       citems.insert(citems.end(), cb.citems.begin(), cb.citems.end());
       for (auto& h : cb.himgs) part.himgs.emplace_back(h.first, std::move(h.second));
     }
+    phase_mark("phase5 merge");
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
 
     // CONTAINS window build: decompress each PLAIN byte-array page once on
