@@ -1490,39 +1490,83 @@ extern "C" gpuq_plan* gpuq_plan_build(
 
     phase_mark("phase4 lz4 walks");
 
-    // phase 5 (serial): order-preserving merge of the per-chunk images
-    for (size_t i = 0; i < cbs.size(); i++) {
-      auto& cb = cbs[i];
-      part.pages.insert(part.pages.end(), cb.pages.begin(), cb.pages.end());
-      for (auto& [key, ids] : cb.tasks) {
-        auto& dstv = part.tasks[key];
-        dstv.insert(dstv.end(), ids.begin(), ids.end());
+    // phase 5: order-preserving merge of the per-chunk images, PARALLEL —
+    // the record streams total ~8 GB at 1 B rows and serial vector inserts
+    // took ~3 s of the plan build. Offsets by prefix sum, then every chunk
+    // copies (with its piece/backref index bases applied) into its slice.
+    {
+      const size_t nc = cbs.size();
+      std::vector<size_t> o_pages(nc + 1, 0), o_segs(nc + 1, 0),
+          o_brs(nc + 1, 0), o_pagebrs(nc + 1, 0), o_rl(nc + 1, 0),
+          o_rw(nc + 1, 0), o_pp(nc + 1, 0), o_bi(nc + 1, 0),
+          o_ll(nc + 1, 0), o_lw(nc + 1, 0);
+      for (size_t i = 0; i < nc; i++) {
+        const auto& cb = cbs[i];
+        o_pages[i + 1] = o_pages[i] + cb.pages.size();
+        o_segs[i + 1] = o_segs[i] + cb.segs.size();
+        o_brs[i + 1] = o_brs[i] + cb.brs.size();
+        o_pagebrs[i + 1] = o_pagebrs[i] + cb.pagebrs.size();
+        o_rl[i + 1] = o_rl[i] + cb.res_lane.size();
+        o_rw[i + 1] = o_rw[i] + cb.res_wave.size();
+        o_pp[i + 1] = o_pp[i] + cb.piece_pool.size();
+        o_bi[i + 1] = o_bi[i] + cb.brinl.size();
+        o_ll[i + 1] = o_ll[i] + cb.lits_lane.size();
+        o_lw[i + 1] = o_lw[i] + cb.lits_wave.size();
       }
-      part.segs.insert(part.segs.end(), cb.segs.begin(), cb.segs.end());
-      uint32_t br_base = (uint32_t)part.brs.size();
-      part.brs.insert(part.brs.end(), cb.brs.begin(), cb.brs.end());
-      for (DevPageBr pb : cb.pagebrs) {
-        pb.start += br_base;
-        part.pagebrs.push_back(pb);
+      part.pages.resize(o_pages[nc]);
+      part.segs.resize(o_segs[nc]);
+      part.brs.resize(o_brs[nc]);
+      part.pagebrs.resize(o_pagebrs[nc]);
+      part.res_lane.resize(o_rl[nc]);
+      part.res_wave.resize(o_rw[nc]);
+      part.piece_pool.resize(o_pp[nc]);
+      part.brinl.resize(o_bi[nc]);
+      part.lits_lane.resize(o_ll[nc]);
+      part.lits_wave.resize(o_lw[nc]);
+      parallel_for(nc, [&](size_t i) {
+        auto& cb = cbs[i];
+        std::copy(cb.pages.begin(), cb.pages.end(),
+                  part.pages.begin() + o_pages[i]);
+        std::copy(cb.segs.begin(), cb.segs.end(),
+                  part.segs.begin() + o_segs[i]);
+        std::copy(cb.brs.begin(), cb.brs.end(), part.brs.begin() + o_brs[i]);
+        const uint32_t br_base = (uint32_t)o_brs[i];
+        for (size_t j = 0; j < cb.pagebrs.size(); j++) {
+          DevPageBr pb = cb.pagebrs[j];
+          pb.start += br_base;
+          part.pagebrs[o_pagebrs[i] + j] = pb;
+        }
+        std::copy(cb.piece_pool.begin(), cb.piece_pool.end(),
+                  part.piece_pool.begin() + o_pp[i]);
+        const uint32_t piece_base = (uint32_t)o_pp[i];
+        for (size_t j = 0; j < cb.res_lane.size(); j++) {
+          DevBrRes r2 = cb.res_lane[j];
+          r2.piece_start += piece_base;
+          part.res_lane[o_rl[i] + j] = r2;
+        }
+        for (size_t j = 0; j < cb.res_wave.size(); j++) {
+          DevBrRes r2 = cb.res_wave[j];
+          r2.piece_start += piece_base;
+          part.res_wave[o_rw[i] + j] = r2;
+        }
+        std::copy(cb.brinl.begin(), cb.brinl.end(),
+                  part.brinl.begin() + o_bi[i]);
+        std::copy(cb.lits_lane.begin(), cb.lits_lane.end(),
+                  part.lits_lane.begin() + o_ll[i]);
+        std::copy(cb.lits_wave.begin(), cb.lits_wave.end(),
+                  part.lits_wave.begin() + o_lw[i]);
+      });
+      // small, order-sensitive leftovers stay serial
+      for (size_t i = 0; i < nc; i++) {
+        auto& cb = cbs[i];
+        for (auto& [key, ids] : cb.tasks) {
+          auto& dstv = part.tasks[key];
+          dstv.insert(dstv.end(), ids.begin(), ids.end());
+        }
+        citems.insert(citems.end(), cb.citems.begin(), cb.citems.end());
+        for (auto& h : cb.himgs)
+          part.himgs.emplace_back(h.first, std::move(h.second));
       }
-      uint32_t piece_base = (uint32_t)part.piece_pool.size();
-      part.piece_pool.insert(part.piece_pool.end(), cb.piece_pool.begin(),
-                             cb.piece_pool.end());
-      for (DevBrRes r2 : cb.res_lane) {
-        r2.piece_start += piece_base;
-        part.res_lane.push_back(r2);
-      }
-      for (DevBrRes r2 : cb.res_wave) {
-        r2.piece_start += piece_base;
-        part.res_wave.push_back(r2);
-      }
-      part.brinl.insert(part.brinl.end(), cb.brinl.begin(), cb.brinl.end());
-      part.lits_lane.insert(part.lits_lane.end(), cb.lits_lane.begin(),
-                            cb.lits_lane.end());
-      part.lits_wave.insert(part.lits_wave.end(), cb.lits_wave.begin(),
-                            cb.lits_wave.end());
-      citems.insert(citems.end(), cb.citems.begin(), cb.citems.end());
-      for (auto& h : cb.himgs) part.himgs.emplace_back(h.first, std::move(h.second));
     }
     phase_mark("phase5 merge");
     part.dec_bytes += 16384 + 64;  // over-read pad: contains window + unpackers
