@@ -240,3 +240,44 @@ def test_hot_tier_layout_modes_do_not_collide(hash_stream):
     # and back to hash layout again (its own cache entry, now warm)
     rows3, _ = Query(prov).execute(dict(q1))
     assert rows3 == rows1
+
+
+def test_numeric_key_sentinel_minus_one(tmp_path_factory):
+    """Value -1 shares the bit pattern of the numeric-key hash's EMPTY
+    sentinel (k_numhash_* route it to a dedicated overflow slot); nulls and
+    zeros ride along."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from oracle import query_oracle as qo
+    from oracle.compare import assert_rows_equal
+    from parseable_amd import GpuSession
+    from parseable_amd.provider import GpuExecutionPlan, merge_partials
+
+    td = tmp_path_factory.mktemp("sentinel")
+    n = 50_000
+    rng = np.random.default_rng(7)
+    k = rng.integers(-2, 3, n)          # includes -1 heavily
+    k = np.where(rng.random(n) < 0.1, None, k)
+    tbl = pa.table({
+        "p_timestamp": pa.array(np.sort(rng.integers(0, 10**6, n))[::-1],
+                                type=pa.timestamp("ms")),
+        "k": pa.array([None if x is None else int(x) for x in k],
+                      type=pa.int64()),
+        "v": pa.array(rng.integers(0, 100, n), type=pa.int64()),
+    })
+    path = str(td / "f.parquet")
+    pq.write_table(tbl, path, row_group_size=262_144, compression="lz4",
+                   use_dictionary=False, data_page_version="1.0",
+                   write_statistics=True)
+    q = {"select": [{"agg": "count_star"}, {"agg": "sum", "col": "v"}],
+         "group_by": ["k"]}
+    plan = GpuExecutionPlan(GpuSession(), [path], dict(q))
+    plan.load()
+    rows = merge_partials([plan.execute(0)], q)
+    want = qo.execute([path], dict(q))["rows"]
+    assert_rows_equal(rows, want, "numeric key -1 sentinel")
+    assert any(r[0] == -1 for r in rows)
+    assert any(r[0] is None for r in rows)
+    plan.close()
