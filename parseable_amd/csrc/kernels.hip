@@ -1559,21 +1559,7 @@ k_contains_win(const uint8_t* __restrict__ dec,
     for (int k = 0; k < CWIN / (CTHREADS * 4); k++)
       *(uint32_t*)&win[threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u)] = v[k];
     if (nlen) {
-      // reserve queue slots with ONE LDS atomic per thread: per-candidate
-      // atomicAdd on the single counter serialized ~hundreds deep for
-      // common needle bytes (~3.5 us of the ~7 us window). The candidate
-      // masks are recomputed from the register-resident words on the
-      // second pass — no extra arrays, no register-pressure regression.
       const uint32_t pat = 0x01010101u * needle[0];
-      uint32_t mine = 0;
-#pragma unroll
-      for (int k = 0; k < CWIN / (CTHREADS * 4); k++) {
-        uint32_t p = threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u);
-        if (p >= W.nbytes) continue;
-        uint32_t x = v[k] ^ pat;
-        mine += __builtin_popcount((x - 0x01010101u) & ~x & 0x80808080u);
-      }
-      uint32_t base = mine ? atomicAdd(&qn, mine) : 0;
 #pragma unroll
       for (int k = 0; k < CWIN / (CTHREADS * 4); k++) {
         uint32_t p = threadIdx.x * 4u + (uint32_t)k * (CTHREADS * 4u);
@@ -1583,8 +1569,8 @@ k_contains_win(const uint8_t* __restrict__ dec,
         while (cand) {
           int b = (__builtin_ctz(cand)) >> 3;
           cand &= cand - 1;
-          if (base < CQMAX) q[base] = (uint16_t)(p + b);
-          base++;
+          uint32_t qi = atomicAdd(&qn, 1u);
+          if (qi < CQMAX) q[qi] = (uint16_t)(p + b);
         }
       }
     }
