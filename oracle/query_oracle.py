@@ -432,6 +432,12 @@ def execute_acero(files: list[str], query: dict) -> dict:
     else:
         tbl2 = tbl.append_column("__g", pa.array(np.zeros(tbl.num_rows, dtype=np.int8)))
         res = pa.TableGroupBy(tbl2, ["__g"]).aggregate(agglist).drop_columns(["__g"])
+    # normalize timestamp-typed aggregate results (min/max over
+    # p_timestamp) to i64 ms like the numpy oracle
+    for name in res.column_names:
+        idx = res.schema.get_field_index(name)
+        if pa.types.is_timestamp(res.schema.field(name).type):
+            res = res.set_column(idx, name, res.column(name).cast(pa.int64()))
     aggcols = [c for c in res.column_names if c not in group_by]
     rows = []
     for i in range(res.num_rows):

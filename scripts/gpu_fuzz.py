@@ -62,7 +62,7 @@ def gen_query(rng: random.Random, cfg: dict, n_rows: int):
         keys.append(rng.choice(cfg["ikeys"]))  # numeric group key
     q["group_by"] = keys
     aggs = [{"agg": "count_star"}]
-    numcols = cfg["i64"] + cfg["f64"]
+    numcols = cfg["i64"] + cfg["f64"] + ["p_timestamp"]
     for _ in range(rng.randint(0, 3)):
         col = rng.choice(numcols)
         op = rng.choice(["sum", "avg", "min", "max", "count"])

@@ -118,6 +118,15 @@ GOLDEN_QUERIES = {
             "group_by": ["level", "latency"],
             "preds": [{"col": "latency", "op": "lt", "lit": 800}],
         }),
+        ("ts_minmax", {
+            # time span per group: min/max over the timestamp column
+            # (i64 ms; DELTA_BINARY_PACKED pages)
+            "select": [{"agg": "min", "col": "p_timestamp"},
+                       {"agg": "max", "col": "p_timestamp"},
+                       {"agg": "count_star"}],
+            "group_by": ["level"],
+            "preds": [{"col": "host", "op": "eq", "lit": "host-0007"}],
+        }),
         ("f64_key_group", {
             "ext": True,
             "select": [{"agg": "count_star"}, {"agg": "max", "col": "latency"}],
